@@ -1,0 +1,172 @@
+// nexmark_gen.hpp — deterministic Nexmark event generator (host side).
+//
+// Mirrors the structure of the reference generator
+// (crates/nexmark/src/generator/{mod,auctions,people,bids,config}.rs and
+// crates/nexmark/src/config.rs), which is itself a port of the Java Nexmark
+// generator.  The DETERMINISTIC parts — event-kind schedule, person/auction id
+// arithmetic, timestamps — follow the reference exactly:
+//   - event mix person:auction:bid = 1:3:46 (config.rs:128-143)
+//   - last/next person and auction id arithmetic
+//     (generator/people.rs:93-118, generator/auctions.rs:85-123)
+//   - timestamp_for_event = base + n * inter_event_delay_us/1000
+//     (generator/config.rs:118-120; default first_event_rate = 10M events/s,
+//      config.rs:50)
+// The RANDOM draws (hot-seller/bidder/auction choice, category, state, city,
+// name, price) come from a documented splitmix64 stream seeded per run:
+// event-stream parity with the reference's Rust SmallRng is unpinned
+// (SURVEY.md §8c), so the oracle and the GPU engine are fed IDENTICAL streams
+// from this generator and parity is asserted between them (plus against the
+// reference's own golden query vectors for hand-written inputs).
+//
+// Strings are dictionary ids end-to-end: state ids index the reference's
+// US_STATES table (generator/people.rs:18-25: AZ,CA,ID,OR,WA,WY = 0..5), city
+// ids its US_CITIES table (people.rs:27-38, Phoenix=0), name ids are opaque
+// u32s standing for the generated first+last name combination.
+#pragma once
+#include <cstdint>
+#include "../../include/dbsp_hip.h"
+
+namespace nexgen {
+
+// config.rs:119-143 defaults
+constexpr uint64_t PERSON_PROP = 1, AUCTION_PROP = 3, BID_PROP = 46;
+constexpr uint64_t TOTAL_PROP = PERSON_PROP + AUCTION_PROP + BID_PROP;  // 50
+constexpr uint64_t FIRST_PERSON_ID = 1000;    // generator/config.rs:5
+constexpr uint64_t FIRST_AUCTION_ID = 1000;   // generator/config.rs:6
+constexpr uint64_t FIRST_CATEGORY_ID = 10;    // generator/config.rs:7
+constexpr uint64_t NUM_CATEGORIES = 5;        // generator/auctions.rs:19
+constexpr uint64_t HOT_SELLER_RATIO_CONST = 100;   // auctions.rs:23
+constexpr uint64_t HOT_AUCTION_RATIO_CONST = 100;  // bids.rs:17
+constexpr uint64_t HOT_BIDDER_RATIO_CONST = 100;   // bids.rs:18
+constexpr uint64_t CFG_HOT_SELLERS_RATIO = 4;   // config.rs:137
+constexpr uint64_t CFG_HOT_AUCTION_RATIO = 2;   // config.rs:135
+constexpr uint64_t CFG_HOT_BIDDERS_RATIO = 4;   // config.rs:136
+constexpr uint64_t NUM_ACTIVE_PEOPLE = 1000;    // config.rs:139
+constexpr uint64_t PERSON_ID_LEAD = 10;         // config.rs:11
+constexpr uint64_t NUM_IN_FLIGHT_AUCTIONS = 100;  // config.rs:141
+constexpr uint64_t NUM_US_STATES = 6, NUM_US_CITIES = 10;
+constexpr double DEFAULT_RATE = 10'000'000.0;   // events/s, config.rs:50
+
+struct Rng {  // splitmix64 (documented; replaces the reference's SmallRng)
+    uint64_t s;
+    uint64_t next() {
+        uint64_t z = (s += 0x9E3779B97F4A7C15ull);
+        z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+        z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+        return z ^ (z >> 31);
+    }
+    uint64_t range(uint64_t n) { return n ? next() % n : 0; }
+};
+
+struct Generator {
+    uint64_t seed;
+    uint64_t base_time;     // ms; reference benches use wallclock; we fix it
+    double inter_delay_us;  // generator/config.rs:62
+    uint64_t count = 0;     // events generated so far
+    Rng rng;
+
+    explicit Generator(uint64_t seed_ = 1, uint64_t base_time_ms = 10'000'000,
+                       double rate = DEFAULT_RATE)
+        : seed(seed_), base_time(base_time_ms),
+          inter_delay_us(1'000'000.0 / rate), rng{seed_ * 0x9E3779B97F4A7C15ull + 1} {}
+
+    // generator/config.rs:118-120
+    uint64_t timestamp_for(uint64_t event_number) const {
+        return base_time + (uint64_t)(inter_delay_us * (double)event_number) / 1000;
+    }
+
+    // generator/people.rs:105-118
+    static uint64_t last_base0_person_id(uint64_t event_id) {
+        uint64_t epoch = event_id / TOTAL_PROP;
+        uint64_t offset = event_id % TOTAL_PROP;
+        if (offset >= PERSON_PROP) offset = PERSON_PROP - 1;
+        return epoch * PERSON_PROP + offset;
+    }
+
+    // generator/people.rs:93-103
+    uint64_t next_base0_person_id(uint64_t event_id) {
+        uint64_t num_people = last_base0_person_id(event_id) + 1;
+        uint64_t active = num_people < NUM_ACTIVE_PEOPLE ? num_people : NUM_ACTIVE_PEOPLE;
+        uint64_t n = rng.range(active + PERSON_ID_LEAD);
+        return num_people - active + n;
+    }
+
+    // generator/auctions.rs:85-110
+    static uint64_t last_base0_auction_id(uint64_t event_id) {
+        uint64_t epoch = event_id / TOTAL_PROP;
+        uint64_t offset = event_id % TOTAL_PROP;
+        if (offset < PERSON_PROP) {
+            if (epoch == 0) return 0;
+            epoch -= 1;
+            offset = AUCTION_PROP - 1;
+        } else if (offset >= PERSON_PROP + AUCTION_PROP) {
+            offset = AUCTION_PROP - 1;
+        } else {
+            offset -= PERSON_PROP;
+        }
+        return epoch * AUCTION_PROP + offset;
+    }
+
+    // generator/auctions.rs:112-123
+    uint64_t next_base0_auction_id(uint64_t event_id) {
+        uint64_t last = last_base0_auction_id(event_id);
+        uint64_t min_a = last > NUM_IN_FLIGHT_AUCTIONS ? last - NUM_IN_FLIGHT_AUCTIONS : 0;
+        return min_a + rng.range(last - min_a + 1);
+    }
+
+    // One event.  kind/f* layout per include/dbsp_hip.h.
+    dbsp_event next() {
+        uint64_t event_id = count;  // single generator: first_event_id = 0
+        uint64_t ts = timestamp_for(event_id);
+        uint64_t rem = event_id % TOTAL_PROP;  // generator/mod.rs:78-88
+        dbsp_event e{};
+        e.w = 1;
+        if (rem < PERSON_PROP) {
+            // generator/people.rs:50-79
+            e.kind = 0;
+            e.f0 = last_base0_person_id(event_id) + FIRST_PERSON_ID;
+            e.f1 = (uint32_t)rng.next();          // name id
+            e.f2 = rng.range(NUM_US_CITIES);      // city id
+            e.f3 = rng.range(NUM_US_STATES);      // state id
+            e.f4 = ts;
+        } else if (rem < PERSON_PROP + AUCTION_PROP) {
+            // generator/auctions.rs:26-82
+            e.kind = 1;
+            e.f0 = last_base0_auction_id(event_id) + FIRST_AUCTION_ID;
+            uint64_t seller;
+            if (rng.range(CFG_HOT_SELLERS_RATIO) == 0)
+                seller = next_base0_person_id(event_id);
+            else
+                seller = (last_base0_person_id(event_id) / HOT_SELLER_RATIO_CONST) *
+                         HOT_SELLER_RATIO_CONST;
+            e.f1 = seller + FIRST_PERSON_ID;
+            e.f2 = FIRST_CATEGORY_ID + rng.range(NUM_CATEGORIES);
+            e.f3 = ts;
+            e.f4 = ts + 10'000;  // expires; unused by q0/q3/q5/q8
+        } else {
+            // generator/bids.rs:61-100
+            e.kind = 2;
+            uint64_t auction;
+            if (rng.range(CFG_HOT_AUCTION_RATIO) == 0)
+                auction = next_base0_auction_id(event_id);
+            else
+                auction = (last_base0_auction_id(event_id) / HOT_AUCTION_RATIO_CONST) *
+                          HOT_AUCTION_RATIO_CONST;
+            e.f0 = auction + FIRST_AUCTION_ID;
+            uint64_t bidder;
+            if (rng.range(CFG_HOT_BIDDERS_RATIO) == 0)
+                bidder = next_base0_person_id(event_id);
+            else
+                bidder = (last_base0_person_id(event_id) / HOT_BIDDER_RATIO_CONST) *
+                             HOT_BIDDER_RATIO_CONST + 1;
+            e.f1 = bidder + FIRST_PERSON_ID;
+            e.f2 = rng.range(10'000);  // price stand-in (price.rs)
+            e.f3 = ts;
+            e.f4 = 0;
+        }
+        count++;
+        return e;
+    }
+};
+
+}  // namespace nexgen

@@ -1,0 +1,241 @@
+"""CPU oracle vs the reference's own golden vectors (tests/golden/) plus
+randomized model tests following the reference's TestBatch pattern
+(reference trace/test_batch.rs:1-60)."""
+import numpy as np
+import pytest
+
+from dbsp_amd import ROW_DT
+from dbsp_amd import oracle
+from helpers import (CITY_IDS, Intern, auction_event, bid_event, events,
+                     load_golden, np_consolidate, pack_person, person_event,
+                     rows_of, state_id, zset)
+
+# ---------------------------------------------------------------------------
+# golden: join_test (reference operator/join.rs:886-1017)
+# ---------------------------------------------------------------------------
+
+PROJ_HI_K_LO_V1V2 = 2
+PROJ_HI_K_LO_V2V1 = 7
+
+
+def _tick_rows(tick, intern):
+    return rows_of([(k, intern(s), w) for k, s, w in tick])
+
+
+def test_golden_stream_join():
+    g = load_golden("join_test.json")
+    intern = Intern()
+    for t1, t2, exp in zip(g["input1"], g["input2"], g["stream_join"]):
+        a = oracle.consolidate(_tick_rows(t1, intern))
+        b = oracle.consolidate(_tick_rows(t2, intern))
+        out = oracle.consolidate(oracle.join_raw(a, b, PROJ_HI_K_LO_V1V2))
+        expected = zset(rows_of([
+            (k, (intern(s1) << 32) | intern(s2), w) for k, s1, s2, w in exp]))
+        assert zset(out) == expected
+
+
+def test_golden_incremental_join():
+    """Incremental join decomposition dA ⋈ B_prev + A_cur ⋈ dB vs the
+    reference's expected inc_outputs (join.rs:938-962)."""
+    g = load_golden("join_test.json")
+    intern = Intern()
+    a_int = np.empty(0, dtype=ROW_DT)
+    b_int = np.empty(0, dtype=ROW_DT)
+    for t1, t2, exp in zip(g["input1"], g["input2"], g["incremental"]):
+        da = oracle.consolidate(_tick_rows(t1, intern))
+        db = oracle.consolidate(_tick_rows(t2, intern))
+        out1 = oracle.join_raw(da, b_int, PROJ_HI_K_LO_V1V2)
+        a_int = oracle.merge(a_int, da)
+        out2 = oracle.join_raw(db, a_int, PROJ_HI_K_LO_V2V1)
+        b_int = oracle.merge(b_int, db)
+        out = oracle.consolidate(np.concatenate([out1, out2]))
+        expected = zset(rows_of([
+            (k, (intern(s1) << 32) | intern(s2), w) for k, s1, s2, w in exp]))
+        assert zset(out) == expected
+
+
+# ---------------------------------------------------------------------------
+# golden: q3 (reference nexmark queries/q3.rs tests)
+# ---------------------------------------------------------------------------
+
+def test_golden_q3():
+    g = load_golden("q3_people.json")
+    intern = Intern()
+    q = oracle.Query(3)
+    for tick in g["ticks"]:
+        evs = []
+        for p in tick["persons"]:
+            evs.append(person_event(p["id"], intern(p["name"]),
+                                    CITY_IDS[p["city"]], state_id(p["state"], intern)))
+        for a in tick["auctions"]:
+            evs.append(auction_event(a["id"], a["seller"], a["category"]))
+        out = q.step(events(*evs))
+        expected = zset(rows_of([
+            (pack_person(intern(name), CITY_IDS[city], state_id(st, intern)), aid, w)
+            for name, city, st, aid, w in tick["expected"]]))
+        assert zset(out) == expected
+
+
+# ---------------------------------------------------------------------------
+# golden: q5 (reference nexmark queries/q5.rs tests)
+# ---------------------------------------------------------------------------
+
+def test_golden_q5():
+    g = load_golden("q5_hot_items.json")
+    for case in g["cases"]:
+        q = oracle.Query(5)
+        for b1, b2, exp in zip(case["auction1_batches"], case["auction2_batches"],
+                               case["expected"]):
+            evs = [bid_event(1, dt) for dt in b1] + [bid_event(2, dt) for dt in b2]
+            out = q.step(events(*evs))
+            expected = zset(rows_of([(a, n, w) for a, n, w in exp]))
+            assert zset(out) == expected, case["name"]
+
+
+# ---------------------------------------------------------------------------
+# golden: q8 (reference nexmark queries/q8.rs tests)
+# ---------------------------------------------------------------------------
+
+def test_golden_q8():
+    g = load_golden("q8_monitor_new_users.json")
+    for case in g["cases"]:
+        intern = Intern()
+        q = oracle.Query(8)
+        for pb, ab, exp in zip(case["people_batches"], case["auction_batches"],
+                               case["expected"]):
+            evs = [person_event(pid, intern(name), 0, 3, dt=dt)
+                   for pid, name, dt in pb]
+            evs += [auction_event(1, seller, 1, dt=dt) for seller, dt in ab]
+            out = q.step(events(*evs))
+            expected = zset(rows_of([
+                (pid, (intern(name) << 32) | stime, w)
+                for pid, name, stime, w in exp]))
+            assert zset(out) == expected, case["name"]
+
+
+# ---------------------------------------------------------------------------
+# model tests (reference TestBatch pattern: random inputs vs naive model)
+# ---------------------------------------------------------------------------
+
+def _random_rows(rng, n, key_range=50, val_range=8, w_range=3):
+    out = np.empty(n, dtype=ROW_DT)
+    out["k"] = rng.integers(0, key_range, n)
+    out["v"] = rng.integers(0, val_range, n)
+    out["w"] = rng.integers(-w_range, w_range + 1, n)
+    return out
+
+
+def test_consolidate_model():
+    rng = np.random.default_rng(7)
+    for n in [0, 1, 2, 17, 1000, 20000]:
+        r = _random_rows(rng, n)
+        got = oracle.consolidate(r)
+        exp = np_consolidate(r)
+        assert np.array_equal(got, exp)
+        # output invariants: sorted by (k,v), unique, no zero weights
+        if len(got):
+            kv = np.stack([got["k"], got["v"]], axis=1)
+            assert (np.lexsort((got["v"], got["k"])) == np.arange(len(got))).all()
+            assert (kv[1:] != kv[:-1]).any(axis=1).all()
+            assert (got["w"] != 0).all()
+
+
+def test_merge_model():
+    rng = np.random.default_rng(8)
+    for n in [0, 1, 5, 300, 5000]:
+        a = oracle.consolidate(_random_rows(rng, n))
+        b = oracle.consolidate(_random_rows(rng, n))
+        got = oracle.merge(a, b)
+        exp = np_consolidate(np.concatenate([a, b]))
+        assert np.array_equal(got, exp)
+        # adversarial: disjoint, identical, cancelling
+        c = a.copy()
+        c["w"] = -c["w"]
+        assert len(oracle.merge(a, c)) == 0
+
+
+def test_join_model():
+    rng = np.random.default_rng(9)
+    for n in [0, 3, 100, 2000]:
+        d = oracle.consolidate(_random_rows(rng, n, key_range=20))
+        t = oracle.consolidate(_random_rows(rng, n, key_range=20))
+        got = zset(oracle.join_raw(d, t, PROJ_HI_K_LO_V1V2))
+        exp = {}
+        for rd in d:
+            for rt in t:
+                if rd["k"] == rt["k"]:
+                    key = (int(rd["k"]), (int(rd["v"]) << 32) | int(rt["v"]))
+                    exp[key] = exp.get(key, 0) + int(rd["w"]) * int(rt["w"])
+        exp = {k: w for k, w in exp.items() if w != 0}
+        assert got == exp
+
+
+def test_agg_linear_model():
+    rng = np.random.default_rng(10)
+    in_trace = oracle.consolidate(_random_rows(rng, 500, key_range=30, val_range=1))
+    out_trace = oracle.consolidate(_random_rows(rng, 60, key_range=30, val_range=4,
+                                                w_range=1))
+    keys = np.unique(in_trace["k"])[:10]
+    got = oracle.agg_linear_upsert(keys, in_trace, out_trace)
+    exp = []
+    for k in keys:
+        upd = {}
+        s = int(in_trace["w"][in_trace["k"] == k].sum())
+        if s != 0:
+            upd[(int(k), s)] = upd.get((int(k), s), 0) + 1
+        for r in out_trace[out_trace["k"] == k]:
+            if r["w"] != 0:
+                kk = (int(k), int(r["v"]))
+                upd[kk] = upd.get(kk, 0) - int(r["w"])
+        exp.extend((k_, v_, w_) for (k_, v_), w_ in upd.items() if w_ != 0)
+    assert zset(got) == zset(rows_of(exp))
+
+
+def test_window_model():
+    rng = np.random.default_rng(11)
+    trace = oracle.consolidate(_random_rows(rng, 400, key_range=100, w_range=2))
+    batch = oracle.consolidate(_random_rows(rng, 50, key_range=100, w_range=2))
+    s0, e0, s1, e1 = 10, 50, 30, 80
+    got = zset(oracle.window(trace, batch, True, s0, e0, s1, e1))
+    exp = {}
+
+    def add(r, sign):
+        key = (int(r["k"]), int(r["v"]))
+        exp[key] = exp.get(key, 0) + sign * int(r["w"])
+
+    for r in trace:
+        k = int(r["k"])
+        if s0 <= k < min(s1, e0):
+            add(r, -1)
+        if e1 < e0 and e1 <= k < e0:
+            add(r, -1)
+        if max(e0, s1) <= k < e1:
+            add(r, 1)
+    for r in batch:
+        if s1 <= int(r["k"]) < e1:
+            add(r, 1)
+    exp = {k: w for k, w in exp.items() if w != 0}
+    assert got == exp
+
+
+def test_xxh3_against_reference_library():
+    """Pin the xxh3 restatement to the real xxhash library (same algorithm the
+    reference's xxhash-rust crate implements) with the reference's seed
+    (hash.rs:6)."""
+    xxhash = pytest.importorskip("xxhash")
+    seed = 0x7F95EF85BE33C337
+    rng = np.random.default_rng(12)
+    keys = [0, 1, 2**63, 2**64 - 1] + [int(x) for x in
+                                       rng.integers(0, 2**63, 50)]
+    for k in keys:
+        expected = xxhash.xxh3_64_intdigest(int(k).to_bytes(8, "little"), seed=seed)
+        assert oracle.xxh3_u64(k) == expected
+
+
+def test_q0_consolidates_events():
+    e1 = auction_event(1, 99, 1, dt=0, expires=10000)
+    e2 = bid_event(1, 1000, price=80)
+    evs = events(e1, e2, e2)
+    out = oracle.q0_step(evs)
+    assert len(out) == 2
+    assert out["w"].tolist() == [2, 1] or out["w"].tolist() == [1, 2]
