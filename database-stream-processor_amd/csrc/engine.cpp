@@ -1,0 +1,1013 @@
+// engine.cpp — host-side mirror of the reference's Circuit/Stream API driving
+// the CDNA4 kernels (kernels.hip) over HBM-resident batches.
+//
+// The reference keeps one OS thread per worker, each evaluating an identical
+// operator DAG once per tick under a static scheduler
+// (circuit/circuit_builder.rs:1403, circuit/schedule/static_scheduler.rs,
+// dbsp_handle.rs:246).  Here: one PROCESS per GPU (rank), one HIP stream per
+// rank as the scheduler's execution lane, and the per-query operator DAG is
+// evaluated in construction (= topological) order by dbsp_engine_step.
+// Cross-worker exchange (operator/communication/exchange.rs:45-251, an N^2
+// mailbox) is an RCCL grouped send/recv (all-to-all-v) over xGMI.
+//
+// Trace state: each Z1Trace/TraceAppend feedback loop (operator/trace.rs:173-460)
+// is a Spine — a stack of consolidated batches with geometric sizes
+// (spine_fueled.rs:107-119).  Unlike the reference's fueled incremental merging
+// (spine_fueled.rs:856: merges amortised across ticks because a CPU merge is
+// slow), GPU merges run to completion per launch — the merge-path kernel moves
+// GB/ms, so the spine policy only decides WHEN to merge (same power-of-two
+// level policy), not how much fuel each tick contributes.
+//
+// Linear operators (join, window, linear aggregate, upsert retraction) are
+// evaluated per spine batch and summed — the reference reads spines through a
+// k-way CursorList (trace/cursor/cursor_list.rs); linearity makes per-batch
+// evaluation + one consolidate equivalent and keeps every kernel a flat
+// sorted-array pass.
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <algorithm>
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <vector>
+
+#include "../../include/dbsp_hip.h"
+#include "kernels_iface.hpp"
+
+#define HIP_CHECK_ST(x)                                                   \
+    do {                                                                  \
+        hipError_t err_ = (x);                                            \
+        if (err_ != hipSuccess) {                                         \
+            fprintf(stderr, "HIP error %s at %s:%d\n",                    \
+                    hipGetErrorString(err_), __FILE__, __LINE__);         \
+            return DBSP_ERR_INTERNAL;                                     \
+        }                                                                 \
+    } while (0)
+
+#define TRY(x)                                                            \
+    do {                                                                  \
+        dbsp_status st_ = (x);                                            \
+        if (st_ != DBSP_OK) return st_;                                   \
+    } while (0)
+
+// ---------------------------------------------------------------------------
+// context
+// ---------------------------------------------------------------------------
+
+struct KernelStat {
+    double ms = 0;
+    double bytes = 0;
+    int64_t launches = 0;
+};
+
+struct dbsp_ctx {
+    int device = 0;
+    hipStream_t stream = nullptr;
+    bool profile = false;
+    hipEvent_t ev0 = nullptr, ev1 = nullptr;
+    KernelStat stats[6];
+    // RCCL
+    ncclComm_t comm = nullptr;
+    int rank = 0, world = 1;
+};
+
+extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
+    int ndev = 0;
+    hipError_t e = hipGetDeviceCount(&ndev);
+    if (e != hipSuccess || ndev == 0) {
+        fprintf(stderr,
+                "dbsp: NO GPU AVAILABLE (hipGetDeviceCount: %s). The product "
+                "path requires an MI355X; there is no CPU fallback.\n",
+                hipGetErrorString(e));
+        return DBSP_ERR_NOGPU;
+    }
+    dbsp_ctx *c = new dbsp_ctx();
+    c->device = device;
+    HIP_CHECK_ST(hipSetDevice(device));
+    HIP_CHECK_ST(hipStreamCreate(&c->stream));
+    HIP_CHECK_ST(hipEventCreate(&c->ev0));
+    HIP_CHECK_ST(hipEventCreate(&c->ev1));
+    const char *p = getenv("DBSP_PROFILE");
+    c->profile = p && p[0] == '1';
+    *out = c;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_ctx_destroy(dbsp_ctx *c) {
+    if (!c) return DBSP_OK;
+    if (c->comm) ncclCommDestroy(c->comm);
+    hipStreamSynchronize(c->stream);
+    hipEventDestroy(c->ev0);
+    hipEventDestroy(c->ev1);
+    hipStreamDestroy(c->stream);
+    delete c;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_ctx_sync(dbsp_ctx *c) {
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_dev_alloc(dbsp_ctx *c, size_t bytes, void **out) {
+    HIP_CHECK_ST(hipMallocAsync(out, bytes, c->stream));
+    return DBSP_OK;
+}
+extern "C" dbsp_status dbsp_dev_free(dbsp_ctx *c, void *p) {
+    HIP_CHECK_ST(hipFreeAsync(p, c->stream));
+    return DBSP_OK;
+}
+extern "C" dbsp_status dbsp_h2d(dbsp_ctx *c, void *dst, const void *src,
+                                size_t bytes) {
+    HIP_CHECK_ST(hipMemcpyAsync(dst, src, bytes, hipMemcpyHostToDevice, c->stream));
+    return DBSP_OK;
+}
+extern "C" dbsp_status dbsp_d2h(dbsp_ctx *c, void *dst, const void *src,
+                                size_t bytes) {
+    HIP_CHECK_ST(hipMemcpyAsync(dst, src, bytes, hipMemcpyDeviceToHost, c->stream));
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    return DBSP_OK;
+}
+
+// profiling helpers: ops sync internally, so record/elapse inline
+struct ScopedTimer {
+    dbsp_ctx *c;
+    int cls;
+    double bytes;
+    ScopedTimer(dbsp_ctx *c_, int cls_, double bytes_) : c(c_), cls(cls_), bytes(bytes_) {
+        if (c->profile) hipEventRecord(c->ev0, c->stream);
+    }
+    ~ScopedTimer() {
+        if (c->profile) {
+            hipEventRecord(c->ev1, c->stream);
+            hipEventSynchronize(c->ev1);
+            float ms = 0;
+            hipEventElapsedTime(&ms, c->ev0, c->ev1);
+            c->stats[cls].ms += ms;
+            c->stats[cls].bytes += bytes;
+            c->stats[cls].launches += 1;
+        }
+    }
+};
+
+// ---------------------------------------------------------------------------
+// device batches + spine
+// ---------------------------------------------------------------------------
+
+struct DevBatch {
+    uint64_t *k = nullptr, *v = nullptr;
+    int64_t *w = nullptr;
+    int64_t n = 0;
+};
+
+static void free_batch(dbsp_ctx *c, DevBatch &b) {
+    if (b.k) hipFreeAsync(b.k, c->stream);
+    if (b.v) hipFreeAsync(b.v, c->stream);
+    if (b.w) hipFreeAsync(b.w, c->stream);
+    b = DevBatch{};
+}
+
+static dbsp_status alloc_batch(dbsp_ctx *c, int64_t n, DevBatch &b) {
+    b.n = n;
+    HIP_CHECK_ST(hipMallocAsync(&b.k, n * sizeof(uint64_t) + 8, c->stream));
+    HIP_CHECK_ST(hipMallocAsync(&b.v, n * sizeof(uint64_t) + 8, c->stream));
+    HIP_CHECK_ST(hipMallocAsync(&b.w, n * sizeof(int64_t) + 8, c->stream));
+    return DBSP_OK;
+}
+
+// concat batches into one (raw, unsorted use)
+static dbsp_status concat_batches(dbsp_ctx *c, const std::vector<DevBatch> &in,
+                                  DevBatch &out) {
+    int64_t total = 0;
+    for (auto &b : in) total += b.n;
+    TRY(alloc_batch(c, total, out));
+    int64_t off = 0;
+    for (auto &b : in) {
+        if (b.n == 0) continue;
+        HIP_CHECK_ST(hipMemcpyAsync(out.k + off, b.k, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(out.v + off, b.v, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(out.w + off, b.w, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
+        off += b.n;
+    }
+    out.n = total;
+    return DBSP_OK;
+}
+
+// sort + consolidate a RAW batch (consumes `raw`)
+static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &out) {
+    if (raw.n == 0) {
+        free_batch(c, raw);
+        out = DevBatch{};
+        return DBSP_OK;
+    }
+    ScopedTimer t(c, 0, (double)raw.n * 48.0);
+    DevBatch scratch;
+    TRY(alloc_batch(c, raw.n, scratch));
+    bool in_scratch = false;
+    TRY(dbspk::sort_rows(c->stream, raw.k, raw.v, raw.w, raw.n, scratch.k,
+                         scratch.v, scratch.w, &in_scratch));
+    DevBatch &sorted = in_scratch ? scratch : raw;
+    TRY(dbspk::consolidate_sorted(c->stream, sorted.k, sorted.v, sorted.w, raw.n,
+                                  &out.k, &out.v, &out.w, &out.n));
+    free_batch(c, raw);
+    free_batch(c, scratch);
+    return DBSP_OK;
+}
+
+static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
+                                 const DevBatch &b, DevBatch &out) {
+    ScopedTimer t(c, 1, (double)(a.n + b.n) * 48.0);
+    TRY(dbspk::merge_rows(c->stream, a.k, a.v, a.w, a.n, b.k, b.v, b.w, b.n,
+                          &out.k, &out.v, &out.w, &out.n));
+    return DBSP_OK;
+}
+
+// Spine: stack of consolidated batches; invariant size[i] >= 2*size[i+1]
+// (power-of-two leveling, spine_fueled.rs:107-119; merges run to completion)
+struct Spine {
+    std::vector<DevBatch> batches;  // largest first
+
+    int64_t total() const {
+        int64_t t = 0;
+        for (auto &b : batches) t += b.n;
+        return t;
+    }
+
+    dbsp_status insert(dbsp_ctx *c, DevBatch b) {
+        if (b.n == 0) {
+            free_batch(c, b);
+            return DBSP_OK;
+        }
+        batches.push_back(b);
+        while (batches.size() >= 2) {
+            DevBatch &top = batches[batches.size() - 1];
+            DevBatch &below = batches[batches.size() - 2];
+            if (top.n * 2 < below.n) break;  // geometric invariant holds
+            DevBatch merged;
+            TRY(merge_batches(c, below, top, merged));
+            free_batch(c, top);
+            free_batch(c, below);
+            batches.pop_back();
+            batches.pop_back();
+            if (merged.n > 0)
+                batches.push_back(merged);
+            else
+                free_batch(c, merged);
+        }
+        return DBSP_OK;
+    }
+
+    // exhaustive merge to a single batch (consolidate.rs:33-47 semantics)
+    dbsp_status consolidate_all(dbsp_ctx *c) {
+        while (batches.size() >= 2) {
+            DevBatch top = batches.back();
+            batches.pop_back();
+            DevBatch below = batches.back();
+            batches.pop_back();
+            DevBatch merged;
+            TRY(merge_batches(c, below, top, merged));
+            free_batch(c, top);
+            free_batch(c, below);
+            if (merged.n > 0) batches.push_back(merged);
+        }
+        return DBSP_OK;
+    }
+
+    void clear(dbsp_ctx *c) {
+        for (auto &b : batches) free_batch(c, b);
+        batches.clear();
+    }
+};
+
+// ---------------------------------------------------------------------------
+// kernel-level C ABI (device-pointer primitives; used by GPU parity tests)
+// ---------------------------------------------------------------------------
+
+extern "C" dbsp_status dbsp_sort_consolidate(dbsp_ctx *c, const uint64_t *k_in,
+                                             const uint64_t *v_in,
+                                             const int64_t *w_in, int64_t n,
+                                             dbsp_batch *out) {
+    DevBatch raw;
+    TRY(alloc_batch(c, n, raw));
+    if (n > 0) {
+        HIP_CHECK_ST(hipMemcpyAsync(raw.k, k_in, n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(raw.v, v_in, n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(raw.w, w_in, n * 8, hipMemcpyDeviceToDevice, c->stream));
+    }
+    DevBatch res;
+    TRY(sort_consolidate_batch(c, raw, res));
+    out->k = res.k;
+    out->v = res.v;
+    out->w = res.w;
+    out->len = res.n;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_merge(dbsp_ctx *c, const dbsp_batch *a,
+                                  const dbsp_batch *b, dbsp_batch *out) {
+    DevBatch ab{a->k, a->v, a->w, a->len};
+    DevBatch bb{b->k, b->v, b->w, b->len};
+    DevBatch res;
+    TRY(merge_batches(c, ab, bb, res));
+    TRY(dbsp_ctx_sync(c));
+    out->k = res.k;
+    out->v = res.v;
+    out->w = res.w;
+    out->len = res.n;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_join(dbsp_ctx *c, const dbsp_batch *delta,
+                                 const dbsp_batch *trace, dbsp_proj proj,
+                                 uint64_t param, dbsp_batch *out) {
+    DevBatch res;
+    {
+        ScopedTimer t(c, 2, (double)delta->len * 24.0);
+        TRY(dbspk::join_rows(c->stream, delta->k, delta->v, delta->w, delta->len,
+                             trace->k, trace->v, trace->w, trace->len, proj,
+                             param, &res.k, &res.v, &res.w, &res.n));
+    }
+    out->k = res.k;
+    out->v = res.v;
+    out->w = res.w;
+    out->len = res.n;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_agg_linear_upsert(dbsp_ctx *c,
+                                              const uint64_t *delta_keys,
+                                              int64_t nd,
+                                              const dbsp_batch *in_trace,
+                                              const dbsp_batch *out_trace,
+                                              dbsp_batch *out) {
+    DevBatch res;
+    TRY(dbspk::agg_linear_upsert_rows(c->stream, delta_keys, nd, in_trace->k,
+                                      in_trace->v, in_trace->w, in_trace->len,
+                                      out_trace->k, out_trace->v, out_trace->w,
+                                      out_trace->len, &res.k, &res.v, &res.w,
+                                      &res.n));
+    out->k = res.k; out->v = res.v; out->w = res.w; out->len = res.n;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_agg_max_upsert(dbsp_ctx *c,
+                                           const uint64_t *delta_keys,
+                                           int64_t nd,
+                                           const dbsp_batch *in_trace,
+                                           const dbsp_batch *out_trace,
+                                           dbsp_batch *out) {
+    DevBatch res;
+    TRY(dbspk::agg_max_upsert_rows(c->stream, delta_keys, nd, in_trace->k,
+                                   in_trace->v, in_trace->w, in_trace->len,
+                                   out_trace->k, out_trace->v, out_trace->w,
+                                   out_trace->len, &res.k, &res.v, &res.w,
+                                   &res.n));
+    out->k = res.k; out->v = res.v; out->w = res.w; out->len = res.n;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_window(dbsp_ctx *c, const dbsp_batch *trace,
+                                   const dbsp_batch *batch, int have_prev,
+                                   uint64_t s0, uint64_t e0, uint64_t s1,
+                                   uint64_t e1, dbsp_batch *out) {
+    DevBatch res;
+    {
+        ScopedTimer t(c, 4, 0.0);
+        TRY(dbspk::window_rows(c->stream, trace->k, trace->v, trace->w,
+                               trace->len, batch->k, batch->v, batch->w,
+                               batch->len, have_prev, s0, e0, s1, e1, &res.k,
+                               &res.v, &res.w, &res.n));
+    }
+    out->k = res.k; out->v = res.v; out->w = res.w; out->len = res.n;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_shard_partition(dbsp_ctx *c, const dbsp_batch *in,
+                                            int nshards, dbsp_batch *out,
+                                            int64_t *offsets_host) {
+    DevBatch res;
+    TRY(alloc_batch(c, in->len, res));
+    TRY(dbspk::shard_rows(c->stream, in->k, in->v, in->w, in->len, nshards,
+                          res.k, res.v, res.w, offsets_host));
+    out->k = res.k; out->v = res.v; out->w = res.w; out->len = in->len;
+    return DBSP_OK;
+}
+
+extern "C" uint64_t dbsp_xxh3_u64(uint64_t key, uint64_t seed) {
+    return dbspk::host_xxh3_u64(key, seed);
+}
+
+// ---------------------------------------------------------------------------
+// RCCL exchange (replaces exchange.rs:45-251 over xGMI)
+// ---------------------------------------------------------------------------
+
+extern "C" dbsp_status dbsp_comm_unique_id(void *out128) {
+    ncclUniqueId id;
+    if (ncclGetUniqueId(&id) != ncclSuccess) return DBSP_ERR_INTERNAL;
+    memcpy(out128, &id, sizeof(id));
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_comm_init(dbsp_ctx *c, int rank, int world,
+                                      const void *nccl_id) {
+    ncclUniqueId id;
+    memcpy(&id, nccl_id, sizeof(id));
+    if (ncclCommInitRank(&c->comm, world, id, rank) != ncclSuccess)
+        return DBSP_ERR_INTERNAL;
+    c->rank = rank;
+    c->world = world;
+    return DBSP_OK;
+}
+
+// all-to-all-v of the three row columns; counts exchanged first
+static dbsp_status alltoallv_cols(dbsp_ctx *c, const DevBatch &send,
+                                  const int64_t *send_counts, DevBatch &recv,
+                                  int64_t *recv_counts) {
+    int world = c->world;
+    // exchange counts (device staging for RCCL)
+    int64_t *d_send_cnt, *d_recv_cnt;
+    HIP_CHECK_ST(hipMallocAsync(&d_send_cnt, world * 8, c->stream));
+    HIP_CHECK_ST(hipMallocAsync(&d_recv_cnt, world * 8, c->stream));
+    HIP_CHECK_ST(hipMemcpyAsync(d_send_cnt, send_counts, world * 8,
+                                hipMemcpyHostToDevice, c->stream));
+    ncclGroupStart();
+    for (int r = 0; r < world; r++) {
+        ncclSend(d_send_cnt + r, 1, ncclInt64, r, c->comm, c->stream);
+        ncclRecv(d_recv_cnt + r, 1, ncclInt64, r, c->comm, c->stream);
+    }
+    ncclGroupEnd();
+    HIP_CHECK_ST(hipMemcpyAsync(recv_counts, d_recv_cnt, world * 8,
+                                hipMemcpyDeviceToHost, c->stream));
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    int64_t recv_total = 0;
+    for (int r = 0; r < world; r++) recv_total += recv_counts[r];
+    TRY(alloc_batch(c, recv_total, recv));
+    // data columns
+    int64_t soff = 0, roff = 0;
+    ncclGroupStart();
+    soff = 0; roff = 0;
+    for (int r = 0; r < world; r++) {
+        if (send_counts[r] > 0) {
+            ncclSend(send.k + soff, send_counts[r], ncclUint64, r, c->comm, c->stream);
+            ncclSend(send.v + soff, send_counts[r], ncclUint64, r, c->comm, c->stream);
+            ncclSend(send.w + soff, send_counts[r], ncclInt64, r, c->comm, c->stream);
+        }
+        if (recv_counts[r] > 0) {
+            ncclRecv(recv.k + roff, recv_counts[r], ncclUint64, r, c->comm, c->stream);
+            ncclRecv(recv.v + roff, recv_counts[r], ncclUint64, r, c->comm, c->stream);
+            ncclRecv(recv.w + roff, recv_counts[r], ncclInt64, r, c->comm, c->stream);
+        }
+        soff += send_counts[r];
+        roff += recv_counts[r];
+    }
+    ncclGroupEnd();
+    HIP_CHECK_ST(hipFreeAsync(d_send_cnt, c->stream));
+    HIP_CHECK_ST(hipFreeAsync(d_recv_cnt, c->stream));
+    recv.n = recv_total;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_comm_alltoallv(dbsp_ctx *c, const dbsp_batch *send,
+                                           const int64_t *send_counts,
+                                           dbsp_batch *recv, int64_t recv_cap,
+                                           int64_t *recv_counts_host) {
+    (void)recv_cap;
+    DevBatch s{send->k, send->v, send->w, send->len};
+    DevBatch r;
+    TRY(alltoallv_cols(c, s, send_counts, r, recv_counts_host));
+    recv->k = r.k; recv->v = r.v; recv->w = r.w; recv->len = r.n;
+    return DBSP_OK;
+}
+
+// shard + exchange + rebuild: the full shard() operator
+// (shard.rs:88-199: hash-split, exchange, re-consolidate)
+static dbsp_status shard_exchange(dbsp_ctx *c, DevBatch local, DevBatch &out) {
+    if (c->world <= 1) {
+        out = local;
+        return DBSP_OK;
+    }
+    DevBatch parts;
+    TRY(alloc_batch(c, local.n, parts));
+    int64_t offsets[65];
+    TRY(dbspk::shard_rows(c->stream, local.k, local.v, local.w, local.n,
+                          c->world, parts.k, parts.v, parts.w, offsets));
+    free_batch(c, local);
+    int64_t send_counts[64];
+    for (int r = 0; r < c->world; r++) send_counts[r] = offsets[r + 1] - offsets[r];
+    DevBatch recv;
+    int64_t recv_counts[64];
+    TRY(alltoallv_cols(c, parts, send_counts, recv, recv_counts));
+    free_batch(c, parts);
+    TRY(sort_consolidate_batch(c, recv, out));
+    return DBSP_OK;
+}
+
+// ---------------------------------------------------------------------------
+// engine: per-query operator DAGs (mirrors nexmark/src/queries/q{0,3,5,8}.rs)
+// ---------------------------------------------------------------------------
+
+struct dbsp_engine {
+    dbsp_ctx *ctx = nullptr;
+    int query = 0;
+    int rank = 0, world = 1;
+
+    // staged input events (resident in HBM before the timed region)
+    dbsp_event *d_events = nullptr;
+    int64_t n_events = 0;
+    std::vector<dbsp_event> h_events;  // host copy for the q0 CPU path
+
+    // q3 state
+    Spine a_int, p_int;
+    // q8 state
+    Spine pt_int, at_int, wp_int, wa_int;
+    bool q8_have_prev = false;
+    uint64_t q8_s0 = 0, q8_e0 = 0, q8_wm = 0;
+    // q5 state
+    Spine bt_int, wb_int, counts_int, bc_int;
+    DevBatch maxin_int, maxout_int, maxz_int;
+    bool q5_have_prev = false;
+    uint64_t q5_s0 = 0, q5_e0 = 0, q5_wm = 0;
+
+    // last tick's output
+    DevBatch output;
+    std::vector<dbsp_event> q0_output;  // q0 CPU path
+};
+
+extern "C" dbsp_status dbsp_engine_create(dbsp_engine **out, dbsp_ctx *ctx,
+                                          int query, int rank, int world) {
+    if (query != 0 && query != 3 && query != 5 && query != 8)
+        return DBSP_ERR_INVALID;
+    if (world > 1 && query != 3 && query != 0)
+        return DBSP_ERR_INVALID;  // q5/q8 multi-rank: next round (watermark allreduce)
+    dbsp_engine *e = new dbsp_engine();
+    e->ctx = ctx;
+    e->query = query;
+    e->rank = rank;
+    e->world = world;
+    *out = e;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
+    if (!e) return DBSP_OK;
+    dbsp_ctx *c = e->ctx;
+    for (Spine *s : {&e->a_int, &e->p_int, &e->pt_int, &e->at_int, &e->wp_int,
+                     &e->wa_int, &e->bt_int, &e->wb_int, &e->counts_int,
+                     &e->bc_int})
+        s->clear(c);
+    free_batch(c, e->maxin_int);
+    free_batch(c, e->maxout_int);
+    free_batch(c, e->maxz_int);
+    free_batch(c, e->output);
+    if (e->d_events) hipFreeAsync(e->d_events, c->stream);
+    hipStreamSynchronize(c->stream);
+    delete e;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_engine_stage_events(dbsp_engine *e,
+                                                const dbsp_event *events,
+                                                int64_t n) {
+    dbsp_ctx *c = e->ctx;
+    if (e->d_events) {
+        HIP_CHECK_ST(hipFreeAsync(e->d_events, c->stream));
+        e->d_events = nullptr;
+    }
+    HIP_CHECK_ST(hipMallocAsync(&e->d_events, n * sizeof(dbsp_event) + 64, c->stream));
+    HIP_CHECK_ST(hipMemcpyAsync(e->d_events, events, n * sizeof(dbsp_event),
+                                hipMemcpyHostToDevice, c->stream));
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    e->n_events = n;
+    if (e->query == 0)
+        e->h_events.assign(events, events + n);
+    return DBSP_OK;
+}
+
+// ---- query 0: CPU-path plumbing config (BASELINE configs[0]) ----
+// The reference's q0 is the identity circuit (queries/q0.rs:6-8); the measured
+// work is the input path's OrdZSet::from_tuples consolidation
+// (input.rs:664, trace/mod.rs:259-263), which this host path mirrors.
+static void q0_step_host(dbsp_engine *e, const dbsp_event *ev, int64_t n) {
+    std::vector<dbsp_event> v(ev, ev + n);
+    std::sort(v.begin(), v.end(), [](const dbsp_event &a, const dbsp_event &b) {
+        if (a.kind != b.kind) return a.kind < b.kind;
+        if (a.f0 != b.f0) return a.f0 < b.f0;
+        if (a.f1 != b.f1) return a.f1 < b.f1;
+        if (a.f2 != b.f2) return a.f2 < b.f2;
+        if (a.f3 != b.f3) return a.f3 < b.f3;
+        return a.f4 < b.f4;
+    });
+    size_t off = 0;
+    auto eq = [](const dbsp_event &a, const dbsp_event &b) {
+        return a.kind == b.kind && a.f0 == b.f0 && a.f1 == b.f1 &&
+               a.f2 == b.f2 && a.f3 == b.f3 && a.f4 == b.f4;
+    };
+    if (!v.empty()) {
+        for (size_t i = 1; i < v.size(); i++) {
+            if (eq(v[off], v[i])) v[off].w += v[i].w;
+            else {
+                if (v[off].w != 0) off++;
+                v[off] = v[i];
+            }
+        }
+        if (v[off].w != 0) off++;
+        v.resize(off);
+    }
+    e->q0_output = std::move(v);
+}
+
+// ---- shared helpers ----
+
+// flatmap events slice into up to two raw streams, then sort+consolidate
+static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
+                                int64_t n, DevBatch &d0, DevBatch &d1,
+                                bool want_two) {
+    dbsp_ctx *c = e->ctx;
+    DevBatch raw0, raw1;
+    TRY(alloc_batch(c, n > 0 ? n : 1, raw0));
+    TRY(alloc_batch(c, n > 0 ? n : 1, raw1));
+    int64_t n0 = 0, n1 = 0;
+    TRY(dbspk::flatmap_events(c->stream, d_ev, n, e->query, raw0.k, raw0.v,
+                              raw0.w, &n0, raw1.k, raw1.v, raw1.w, &n1));
+    raw0.n = n0;
+    raw1.n = n1;
+    TRY(sort_consolidate_batch(c, raw0, d0));
+    if (want_two)
+        TRY(sort_consolidate_batch(c, raw1, d1));
+    else
+        free_batch(c, raw1);
+    // worker sharding: co-locate keys across ranks (shard.rs:88)
+    if (e->world > 1) {
+        DevBatch s0;
+        TRY(shard_exchange(c, d0, s0));
+        d0 = s0;
+        if (want_two) {
+            DevBatch s1;
+            TRY(shard_exchange(c, d1, s1));
+            d1 = s1;
+        }
+    }
+    return DBSP_OK;
+}
+
+// join delta against every batch of a spine (join is linear in the trace)
+static dbsp_status join_vs_spine(dbsp_ctx *c, const DevBatch &delta,
+                                 const Spine &spine, int proj, uint64_t param,
+                                 std::vector<DevBatch> &outs) {
+    for (auto &b : spine.batches) {
+        if (delta.n == 0 || b.n == 0) continue;
+        DevBatch o;
+        ScopedTimer t(c, 2, (double)delta.n * 24.0);
+        TRY(dbspk::join_rows(c->stream, delta.k, delta.v, delta.w, delta.n, b.k,
+                             b.v, b.w, b.n, proj, param, &o.k, &o.v, &o.w, &o.n));
+        if (o.n > 0) outs.push_back(o);
+        else free_batch(c, o);
+    }
+    return DBSP_OK;
+}
+
+// consolidate a list of raw result batches into one batch
+static dbsp_status finalize_raw(dbsp_ctx *c, std::vector<DevBatch> &outs,
+                                DevBatch &out) {
+    DevBatch cat;
+    TRY(concat_batches(c, outs, cat));
+    for (auto &b : outs) free_batch(c, b);
+    outs.clear();
+    TRY(sort_consolidate_batch(c, cat, out));
+    return DBSP_OK;
+}
+
+// last key of a consolidated batch (watermark: fast_forward_keys,
+// watermark.rs:38-45)
+static dbsp_status last_key(dbsp_ctx *c, const DevBatch &b, uint64_t *out,
+                            bool *has) {
+    if (b.n == 0) {
+        *has = false;
+        return DBSP_OK;
+    }
+    TRY(dbsp_d2h(c, out, b.k + (b.n - 1), 8));
+    *has = true;
+    return DBSP_OK;
+}
+
+// window over a spine: region scans per trace batch + the batch region once
+static dbsp_status window_vs_spine(dbsp_ctx *c, const Spine &trace,
+                                   const DevBatch &batch, bool have_prev,
+                                   uint64_t s0, uint64_t e0, uint64_t s1,
+                                   uint64_t e1, std::vector<DevBatch> &outs) {
+    DevBatch empty;
+    for (auto &b : trace.batches) {
+        DevBatch o;
+        ScopedTimer t(c, 4, 0.0);
+        TRY(dbspk::window_rows(c->stream, b.k, b.v, b.w, b.n, empty.k, empty.v,
+                               empty.w, 0, have_prev ? 1 : 0, s0, e0, s1, e1,
+                               &o.k, &o.v, &o.w, &o.n));
+        if (o.n > 0) outs.push_back(o);
+        else free_batch(c, o);
+    }
+    {
+        DevBatch o;
+        ScopedTimer t(c, 4, 0.0);
+        TRY(dbspk::window_rows(c->stream, empty.k, empty.v, empty.w, 0, batch.k,
+                               batch.v, batch.w, batch.n, 0, 0, 0, s1, e1, &o.k,
+                               &o.v, &o.w, &o.n));
+        if (o.n > 0) outs.push_back(o);
+        else free_batch(c, o);
+    }
+    return DBSP_OK;
+}
+
+// map + sort/consolidate (copies; input retained)
+static dbsp_status map_sorted(dbsp_ctx *c, const DevBatch &in, int mode,
+                              DevBatch &out) {
+    if (in.n == 0) {
+        out = DevBatch{};
+        return DBSP_OK;
+    }
+    DevBatch raw;
+    TRY(alloc_batch(c, in.n, raw));
+    TRY(dbspk::map_rows(c->stream, in.k, in.v, in.w, in.n, mode, raw.k, raw.v, raw.w));
+    raw.n = in.n;
+    TRY(sort_consolidate_batch(c, raw, out));
+    return DBSP_OK;
+}
+
+// copy a batch (for inserting a delta into a spine while retaining it)
+static dbsp_status copy_batch(dbsp_ctx *c, const DevBatch &in, DevBatch &out) {
+    TRY(alloc_batch(c, in.n > 0 ? in.n : 1, out));
+    out.n = in.n;
+    if (in.n > 0) {
+        HIP_CHECK_ST(hipMemcpyAsync(out.k, in.k, in.n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(out.v, in.v, in.n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(out.w, in.w, in.n * 8, hipMemcpyDeviceToDevice, c->stream));
+    }
+    return DBSP_OK;
+}
+
+// linear aggregate + upsert over spines (aggregate/mod.rs:479-547 +
+// upsert.rs:161-208; linear in both traces, so evaluated per spine batch)
+static dbsp_status agg_linear_spine(dbsp_ctx *c, const DevBatch &delta,
+                                    const Spine &in_trace,
+                                    const Spine &out_trace, DevBatch &out) {
+    if (delta.n == 0) {
+        out = DevBatch{};
+        return DBSP_OK;
+    }
+    ScopedTimer timer(c, 3, 0.0);
+    uint64_t *keys = nullptr;
+    int64_t nk = 0;
+    TRY(dbspk::unique_keys(c->stream, delta.k, delta.n, &keys, &nk));
+    // per-key weight sums across in_trace batches
+    int64_t *acc;
+    HIP_CHECK_ST(hipMallocAsync(&acc, nk * 8 + 8, c->stream));
+    HIP_CHECK_ST(hipMemsetAsync(acc, 0, nk * 8, c->stream));
+    for (auto &b : in_trace.batches)
+        TRY(dbspk::agg_sum_batch(c->stream, keys, nk, b.k, b.w, b.n, acc));
+    std::vector<DevBatch> outs;
+    // inserts
+    DevBatch ins;
+    TRY(alloc_batch(c, nk, ins));
+    int64_t n_ins = 0;
+    TRY(dbspk::emit_nonzero(c->stream, keys, acc, nk, ins.k, ins.v, ins.w, &n_ins));
+    ins.n = n_ins;
+    if (ins.n > 0) outs.push_back(ins);
+    else free_batch(c, ins);
+    // retractions: agg kernel against each out_trace batch with an empty input
+    // trace emits exactly the upsert retractions
+    for (auto &b : out_trace.batches) {
+        DevBatch o;
+        TRY(dbspk::agg_linear_upsert_rows(c->stream, keys, nk, nullptr, nullptr,
+                                          nullptr, 0, b.k, b.v, b.w, b.n, &o.k,
+                                          &o.v, &o.w, &o.n));
+        if (o.n > 0) outs.push_back(o);
+        else free_batch(c, o);
+    }
+    HIP_CHECK_ST(hipFreeAsync(acc, c->stream));
+    HIP_CHECK_ST(hipFreeAsync(keys, c->stream));
+    TRY(finalize_raw(c, outs, out));
+    return DBSP_OK;
+}
+
+// ---- q3 tick (queries/q3.rs:35-63) ----
+static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
+    dbsp_ctx *c = e->ctx;
+    DevBatch dA, dP;
+    TRY(build_deltas(e, d_ev, n, dA, dP, true));
+    std::vector<DevBatch> outs;
+    // dA ⋈ P_{t-1}
+    TRY(join_vs_spine(c, dA, e->p_int, DBSP_PROJ_HI_V2_LO_V1, 0, outs));
+    TRY(e->a_int.insert(c, dA));
+    // A_t ⋈ dP
+    TRY(join_vs_spine(c, dP, e->a_int, DBSP_PROJ_HI_V1_LO_V2, 0, outs));
+    TRY(e->p_int.insert(c, dP));
+    free_batch(c, e->output);
+    TRY(finalize_raw(c, outs, e->output));
+    return DBSP_OK;
+}
+
+// ---- q8 tick (queries/q8.rs:48-93) ----
+static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
+    dbsp_ctx *c = e->ctx;
+    constexpr uint64_t TUMBLE_MS = 10000;
+    DevBatch dPT, dAT;
+    TRY(build_deltas(e, d_ev, n, dPT, dAT, true));
+    // watermark over auctions (q8.rs:63-65)
+    uint64_t lk = 0;
+    bool has = false;
+    TRY(last_key(c, dAT, &lk, &has));
+    if (has) e->q8_wm = std::max(e->q8_wm, lk - TUMBLE_MS);
+    uint64_t rounded = e->q8_wm - e->q8_wm % TUMBLE_MS;
+    uint64_t s1 = rounded >= TUMBLE_MS ? rounded - TUMBLE_MS : 0;
+    uint64_t e1 = rounded;
+    // windowed people / auctions
+    std::vector<DevBatch> wp_raw, wa_raw;
+    TRY(window_vs_spine(c, e->pt_int, dPT, e->q8_have_prev, e->q8_s0, e->q8_e0,
+                        s1, e1, wp_raw));
+    TRY(window_vs_spine(c, e->at_int, dAT, e->q8_have_prev, e->q8_s0, e->q8_e0,
+                        s1, e1, wa_raw));
+    e->q8_have_prev = true;
+    e->q8_s0 = s1;
+    e->q8_e0 = e1;
+    TRY(e->pt_int.insert(c, dPT));
+    TRY(e->at_int.insert(c, dAT));
+    // map_index / map + consolidate
+    DevBatch wpr, war, dWP, dWA;
+    TRY(finalize_raw(c, wp_raw, wpr));
+    TRY(finalize_raw(c, wa_raw, war));
+    TRY(map_sorted(c, wpr, 0, dWP));
+    TRY(map_sorted(c, war, 1, dWA));
+    free_batch(c, wpr);
+    free_batch(c, war);
+    std::vector<DevBatch> outs;
+    TRY(join_vs_spine(c, dWP, e->wa_int, DBSP_PROJ_HI_K_LO_V1RND, TUMBLE_MS, outs));
+    TRY(e->wp_int.insert(c, dWP));
+    TRY(join_vs_spine(c, dWA, e->wp_int, DBSP_PROJ_HI_K_LO_V2RND, TUMBLE_MS, outs));
+    TRY(e->wa_int.insert(c, dWA));
+    free_batch(c, e->output);
+    TRY(finalize_raw(c, outs, e->output));
+    return DBSP_OK;
+}
+
+// ---- q5 tick (queries/q5.rs:77-121) ----
+static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
+    dbsp_ctx *c = e->ctx;
+    constexpr uint64_t WIDTH_MS = 10000, TUMBLE_MS = 2000, WM_LAG_MS = 4000;
+    DevBatch dBT, dummy;
+    TRY(build_deltas(e, d_ev, n, dBT, dummy, false));
+    uint64_t lk = 0;
+    bool has = false;
+    TRY(last_key(c, dBT, &lk, &has));
+    if (has) e->q5_wm = std::max(e->q5_wm, lk - WM_LAG_MS);
+    uint64_t rounded = e->q5_wm - e->q5_wm % TUMBLE_MS;
+    uint64_t s1 = rounded >= WIDTH_MS ? rounded - WIDTH_MS : 0;
+    uint64_t e1 = rounded;
+    std::vector<DevBatch> wb_raw;
+    TRY(window_vs_spine(c, e->bt_int, dBT, e->q5_have_prev, e->q5_s0, e->q5_e0,
+                        s1, e1, wb_raw));
+    e->q5_have_prev = true;
+    e->q5_s0 = s1;
+    e->q5_e0 = e1;
+    TRY(e->bt_int.insert(c, dBT));
+    DevBatch wbr, dWB;
+    TRY(finalize_raw(c, wb_raw, wbr));
+    TRY(map_sorted(c, wbr, 1, dWB));  // (time,auction) -> (auction,()); weigh(|_|1)
+    free_batch(c, wbr);
+    // aggregate_linear: input trace includes this tick (trace.rs TraceAppend)
+    DevBatch wb_copy;
+    TRY(copy_batch(c, dWB, wb_copy));
+    TRY(e->wb_int.insert(c, wb_copy));
+    DevBatch dCounts;
+    TRY(agg_linear_spine(c, dWB, e->wb_int, e->counts_int, dCounts));
+    free_batch(c, dWB);
+    // max side (tiny; consolidated single-batch traces)
+    DevBatch dMaxIn;
+    TRY(map_sorted(c, dCounts, 2, dMaxIn));
+    DevBatch dMaxOut;
+    if (dMaxIn.n > 0) {
+        DevBatch m;
+        TRY(merge_batches(c, e->maxin_int, dMaxIn, m));
+        free_batch(c, e->maxin_int);
+        free_batch(c, dMaxIn);
+        e->maxin_int = m;
+        // delta key is the unit key ()
+        uint64_t *d_unit;
+        HIP_CHECK_ST(hipMallocAsync(&d_unit, 8, c->stream));
+        HIP_CHECK_ST(hipMemsetAsync(d_unit, 0, 8, c->stream));
+        DevBatch raw;
+        TRY(dbspk::agg_max_upsert_rows(
+            c->stream, d_unit, 1, e->maxin_int.k, e->maxin_int.v, e->maxin_int.w,
+            e->maxin_int.n, e->maxout_int.k, e->maxout_int.v, e->maxout_int.w,
+            e->maxout_int.n, &raw.k, &raw.v, &raw.w, &raw.n));
+        HIP_CHECK_ST(hipFreeAsync(d_unit, c->stream));
+        TRY(sort_consolidate_batch(c, raw, dMaxOut));
+        DevBatch m2;
+        TRY(merge_batches(c, e->maxout_int, dMaxOut, m2));
+        free_batch(c, e->maxout_int);
+        e->maxout_int = m2;
+    } else {
+        free_batch(c, dMaxIn);
+    }
+    DevBatch dMaxZ, dBC;
+    TRY(map_sorted(c, dMaxOut, 1, dMaxZ));  // ((),m) -> (m,())
+    free_batch(c, dMaxOut);
+    TRY(map_sorted(c, dCounts, 3, dBC));    // (auction,count) -> (count,auction)
+    // final incremental join (q5.rs:118-120)
+    std::vector<DevBatch> outs;
+    TRY(join_vs_spine(c, dMaxZ, e->bc_int, DBSP_PROJ_HI_V2_LO_K, 0, outs));
+    {
+        DevBatch m;
+        TRY(merge_batches(c, e->maxz_int, dMaxZ, m));
+        free_batch(c, e->maxz_int);
+        free_batch(c, dMaxZ);
+        e->maxz_int = m;
+    }
+    if (dBC.n > 0 && e->maxz_int.n > 0) {
+        DevBatch o;
+        TRY(dbspk::join_rows(c->stream, dBC.k, dBC.v, dBC.w, dBC.n,
+                             e->maxz_int.k, e->maxz_int.v, e->maxz_int.w,
+                             e->maxz_int.n, DBSP_PROJ_HI_V1_LO_K, 0, &o.k, &o.v,
+                             &o.w, &o.n));
+        if (o.n > 0) outs.push_back(o);
+        else free_batch(c, o);
+    }
+    TRY(e->bc_int.insert(c, dBC));
+    TRY(e->counts_int.insert(c, dCounts));
+    free_batch(c, e->output);
+    TRY(finalize_raw(c, outs, e->output));
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_engine_step_staged(dbsp_engine *e, int64_t lo,
+                                               int64_t hi) {
+    if (lo < 0 || hi > e->n_events || lo > hi) return DBSP_ERR_INVALID;
+    if (e->query == 0) {
+        q0_step_host(e, e->h_events.data() + lo, hi - lo);
+        return DBSP_OK;
+    }
+    const dbsp_event *d_ev = e->d_events + lo;
+    switch (e->query) {
+        case 3: return q3_step(e, d_ev, hi - lo);
+        case 5: return q5_step(e, d_ev, hi - lo);
+        case 8: return q8_step(e, d_ev, hi - lo);
+    }
+    return DBSP_ERR_INVALID;
+}
+
+extern "C" dbsp_status dbsp_engine_step(dbsp_engine *e, const dbsp_event *events,
+                                        int64_t n) {
+    if (e->query == 0) {
+        q0_step_host(e, events, n);
+        return DBSP_OK;
+    }
+    dbsp_ctx *c = e->ctx;
+    dbsp_event *d_ev;
+    HIP_CHECK_ST(hipMallocAsync(&d_ev, n * sizeof(dbsp_event) + 64, c->stream));
+    HIP_CHECK_ST(hipMemcpyAsync(d_ev, events, n * sizeof(dbsp_event),
+                                hipMemcpyHostToDevice, c->stream));
+    dbsp_status st = DBSP_OK;
+    switch (e->query) {
+        case 3: st = q3_step(e, d_ev, n); break;
+        case 5: st = q5_step(e, d_ev, n); break;
+        case 8: st = q8_step(e, d_ev, n); break;
+        default: st = DBSP_ERR_INVALID;
+    }
+    hipFreeAsync(d_ev, c->stream);
+    return st;
+}
+
+extern "C" dbsp_status dbsp_engine_output(dbsp_engine *e, dbsp_row *out,
+                                          int64_t cap, int64_t *n_out) {
+    dbsp_ctx *c = e->ctx;
+    *n_out = e->output.n;
+    if (e->output.n > cap) return DBSP_ERR_OVERFLOW;
+    if (e->output.n == 0) return DBSP_OK;
+    std::vector<uint64_t> k(e->output.n), v(e->output.n);
+    std::vector<int64_t> w(e->output.n);
+    TRY(dbsp_d2h(c, k.data(), e->output.k, e->output.n * 8));
+    TRY(dbsp_d2h(c, v.data(), e->output.v, e->output.n * 8));
+    TRY(dbsp_d2h(c, w.data(), e->output.w, e->output.n * 8));
+    for (int64_t i = 0; i < e->output.n; i++) out[i] = {k[i], v[i], w[i]};
+    return DBSP_OK;
+}
+
+// q0's output is the consolidated event zset itself
+extern "C" dbsp_status dbsp_engine_output_events(dbsp_engine *e, dbsp_event *out,
+                                                 int64_t cap, int64_t *n_out) {
+    *n_out = (int64_t)e->q0_output.size();
+    if ((int64_t)e->q0_output.size() > cap) return DBSP_ERR_OVERFLOW;
+    memcpy(out, e->q0_output.data(), e->q0_output.size() * sizeof(dbsp_event));
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_engine_kernel_stats(dbsp_engine *e, int kind,
+                                                double *total_ms,
+                                                double *algo_bytes,
+                                                int64_t *launches) {
+    if (kind < 0 || kind > 5) return DBSP_ERR_INVALID;
+    *total_ms = e->ctx->stats[kind].ms;
+    *algo_bytes = e->ctx->stats[kind].bytes;
+    *launches = e->ctx->stats[kind].launches;
+    return DBSP_OK;
+}

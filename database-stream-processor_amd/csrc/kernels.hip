@@ -1,0 +1,1251 @@
+// kernels.hip — hand-written CDNA4 (gfx950) kernels for the DBSP Z-set hot path.
+//
+// MI355X-native design notes (see DESIGN.md for the full rationale):
+//  * Every op here is HBM-bound integer/byte work (no dense contraction ⇒ no
+//    MFMA).  The levers are coalescing (SoA u64 columns, 8B/lane loads),
+//    LDS staging for the block-local sort/merge phases, and ≫256-workgroup
+//    launches with grid-stride loops (256 CUs / 8 XCDs want >2048 blocks
+//    before per-XCD L2 effects matter).
+//  * Wave width is 64; block size 256 (4 waves) everywhere.
+//  * Output sizes are data-dependent (zero-weight elimination,
+//    trace/consolidation/mod.rs:32-51) ⇒ count→scan→emit two-phase kernels.
+//
+// Reference mapping:
+//   radix sort + consolidate  <- consolidate_slice / quicksort
+//                                (trace/consolidation/mod.rs:91-110, quicksort.rs)
+//   merge-path merge          <- ColumnLayerBuilder::push_merge
+//                                (trace/layers/column_layer/builders.rs:98-169)
+//                                + OrderedBuilder::merge_step (ordered/mod.rs:344-396)
+//   join probe/expand         <- Join::eval / JoinTrace::eval
+//                                (operator/join.rs:436-473,751-787)
+//   aggregate + upsert        <- aggregate/mod.rs:479-547 + upsert.rs:161-208
+//   window                    <- time_series/window.rs:144-220
+//   shard partition           <- communication/shard.rs:165-199 + hash.rs:9-13
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+
+#include "../../include/dbsp_hip.h"
+
+#define WAVE 64
+#define BLK 256
+#define SORT_ITEMS 8
+#define SORT_TILE (BLK * SORT_ITEMS)  // 2048 rows per block
+
+#define HIP_CHECK(x)                                                      \
+    do {                                                                  \
+        hipError_t err_ = (x);                                            \
+        if (err_ != hipSuccess) {                                         \
+            fprintf(stderr, "HIP error %s at %s:%d\n",                    \
+                    hipGetErrorString(err_), __FILE__, __LINE__);         \
+            return DBSP_ERR_INTERNAL;                                     \
+        }                                                                 \
+    } while (0)
+
+static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
+
+// ---------------------------------------------------------------------------
+// generic grid-stride helpers
+// ---------------------------------------------------------------------------
+
+static inline dim3 grid_for(int64_t n, int per_block = BLK) {
+    int64_t b = ceil_div(n, per_block);
+    if (b < 1) b = 1;
+    if (b > 16384) b = 16384;  // grid-stride the rest (Guideline 11)
+    return dim3((uint32_t)b);
+}
+
+__global__ void k_fill_u64(uint64_t *p, uint64_t v, int64_t n) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        p[i] = v;
+}
+
+// ---------------------------------------------------------------------------
+// exclusive scan (u64) — 3-kernel recursive device scan
+// ---------------------------------------------------------------------------
+
+__device__ inline uint64_t wave_scan_excl(uint64_t x, uint64_t &total) {
+    // inclusive shuffle scan over the 64-lane wave, then convert
+    uint64_t v = x;
+    for (int d = 1; d < WAVE; d <<= 1) {
+        uint64_t up = __shfl_up(v, d, WAVE);
+        if ((threadIdx.x & (WAVE - 1)) >= d) v += up;
+    }
+    total = __shfl(v, WAVE - 1, WAVE);
+    return v - x;
+}
+
+// per-block scan of SORT_TILE elements; writes per-block total
+__global__ void k_scan_block(const uint64_t *in, uint64_t *out,
+                             uint64_t *block_totals, int64_t n) {
+    __shared__ uint64_t wave_tot[BLK / WAVE];
+    int64_t base = (int64_t)blockIdx.x * SORT_TILE;
+    uint64_t vals[SORT_ITEMS];
+    uint64_t thread_sum = 0;
+    for (int i = 0; i < SORT_ITEMS; i++) {
+        int64_t idx = base + threadIdx.x * SORT_ITEMS + i;
+        vals[i] = idx < n ? in[idx] : 0;
+        thread_sum += vals[i];
+    }
+    uint64_t wave_total;
+    uint64_t thread_off = wave_scan_excl(thread_sum, wave_total);
+    int wid = threadIdx.x / WAVE;
+    if ((threadIdx.x & (WAVE - 1)) == WAVE - 1) wave_tot[wid] = wave_total;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint64_t acc = 0;
+        for (int w = 0; w < BLK / WAVE; w++) {
+            uint64_t t = wave_tot[w];
+            wave_tot[w] = acc;
+            acc += t;
+        }
+        if (block_totals) block_totals[blockIdx.x] = acc;
+    }
+    __syncthreads();
+    uint64_t off = wave_tot[wid] + thread_off;
+    for (int i = 0; i < SORT_ITEMS; i++) {
+        int64_t idx = base + threadIdx.x * SORT_ITEMS + i;
+        if (idx < n) out[idx] = off;
+        off += vals[i];
+    }
+}
+
+__global__ void k_scan_add(uint64_t *out, const uint64_t *block_offsets,
+                           int64_t n) {
+    int64_t base = (int64_t)blockIdx.x * SORT_TILE;
+    uint64_t off = block_offsets[blockIdx.x];
+    for (int i = threadIdx.x; i < SORT_TILE; i += BLK) {
+        int64_t idx = base + i;
+        if (idx < n) out[idx] += off;
+    }
+}
+
+// host-side recursive exclusive scan; returns total via d_total (device, may be
+// null).  in and out may alias.
+struct ScanTemp {
+    // enough levels for 2048^4 = 1.7e13 elements
+    uint64_t *lvl[4] = {nullptr, nullptr, nullptr, nullptr};
+    int64_t lvl_n[4] = {0, 0, 0, 0};
+};
+
+static dbsp_status scan_exclusive(hipStream_t s, const uint64_t *in,
+                                  uint64_t *out, int64_t n, uint64_t *h_total) {
+    if (n == 0) {
+        if (h_total) *h_total = 0;
+        return DBSP_OK;
+    }
+    // collect level sizes
+    int64_t sizes[5];
+    int levels = 0;
+    int64_t m = n;
+    while (true) {
+        sizes[levels] = m;
+        m = ceil_div(m, SORT_TILE);
+        if (sizes[levels] <= SORT_TILE) break;
+        levels++;
+        if (levels >= 4) return DBSP_ERR_INVALID;
+    }
+    // allocate block-total arrays per level
+    uint64_t *tot[5] = {nullptr};
+    for (int l = 0; l <= levels; l++) {
+        int64_t nb = ceil_div(sizes[l], SORT_TILE);
+        HIP_CHECK(hipMallocAsync(&tot[l], (nb + 1) * sizeof(uint64_t), s));
+    }
+    // down-sweep: scan each level, producing block totals
+    const uint64_t *src = in;
+    uint64_t *dst = out;
+    for (int l = 0; l <= levels; l++) {
+        int64_t nb = ceil_div(sizes[l], SORT_TILE);
+        k_scan_block<<<dim3((uint32_t)nb), BLK, 0, s>>>(src, dst, tot[l], sizes[l]);
+        src = tot[l];
+        dst = tot[l];  // scan totals in place at next level
+    }
+    // top level: tot[levels] has <= SORT_TILE entries, already scanned in the
+    // loop? no — the loop scanned level l's DATA and wrote raw totals to tot[l].
+    // The next iteration scans tot[l] (as data) into itself and writes raw
+    // totals to tot[l+1].  After the loop, tot[levels] holds RAW totals of the
+    // last scanned array; scan it with a single block.
+    {
+        int64_t nb_last = ceil_div(sizes[levels], SORT_TILE);
+        k_scan_block<<<dim3(1), BLK, 0, s>>>(tot[levels], tot[levels], tot[levels] + nb_last,
+                                             nb_last);
+        // up-sweep: add scanned block offsets back down
+        for (int l = levels; l >= 0; l--) {
+            int64_t nb = ceil_div(sizes[l], SORT_TILE);
+            uint64_t *data = (l == 0) ? out : tot[l - 1];
+            if (nb > 1 || l > 0) {
+                uint64_t *offs = tot[l];
+                k_scan_add<<<dim3((uint32_t)nb), BLK, 0, s>>>(data, offs, sizes[l]);
+            }
+        }
+        // total = tot[levels][nb_last] (one past the scanned entries)
+        if (h_total) {
+            HIP_CHECK(hipMemcpyAsync(h_total, tot[levels] + nb_last, sizeof(uint64_t),
+                                     hipMemcpyDeviceToHost, s));
+            HIP_CHECK(hipStreamSynchronize(s));
+        }
+    }
+    for (int l = 0; l <= levels; l++) HIP_CHECK(hipFreeAsync(tot[l], s));
+    return DBSP_OK;
+}
+
+// ---------------------------------------------------------------------------
+// LSD radix sort of (k, v, w) rows by composite (k major, v minor)
+// 8-bit digits; bytes 0..7 = v, 8..15 = k; passes above the detected
+// significant byte of max(v)/max(k) are skipped.
+// ---------------------------------------------------------------------------
+
+__global__ void k_minmax_u64(const uint64_t *a, const uint64_t *b, int64_t n,
+                             uint64_t *out_max /* [2] */) {
+    __shared__ uint64_t smax[2];
+    if (threadIdx.x == 0) { smax[0] = 0; smax[1] = 0; }
+    __syncthreads();
+    uint64_t ma = 0, mb = 0;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        ma = max(ma, a[i]);
+        mb = max(mb, b[i]);
+    }
+    atomicMax((unsigned long long *)&smax[0], (unsigned long long)ma);
+    atomicMax((unsigned long long *)&smax[1], (unsigned long long)mb);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        atomicMax((unsigned long long *)&out_max[0], (unsigned long long)smax[0]);
+        atomicMax((unsigned long long *)&out_max[1], (unsigned long long)smax[1]);
+    }
+}
+
+__device__ inline uint32_t sort_digit(uint64_t k, uint64_t v, int byte) {
+    uint64_t limb = byte < 8 ? v : k;
+    int sh = (byte & 7) * 8;
+    return (uint32_t)((limb >> sh) & 0xFF);
+}
+
+// histogram: counts[d * nblocks + blk] (digit-major so one exclusive scan of
+// the flat array yields combined digit-base + block-offset)
+__global__ void k_radix_hist(const uint64_t *k, const uint64_t *v, int64_t n,
+                             int byte, int64_t nblocks, uint64_t *counts) {
+    __shared__ uint32_t hist[256];
+    for (int i = threadIdx.x; i < 256; i += BLK) hist[i] = 0;
+    __syncthreads();
+    int64_t base = (int64_t)blockIdx.x * SORT_TILE;
+    for (int i = threadIdx.x; i < SORT_TILE; i += BLK) {
+        int64_t idx = base + i;
+        if (idx < n) atomicAdd(&hist[sort_digit(k[idx], v[idx], byte)], 1u);
+    }
+    __syncthreads();
+    for (int d = threadIdx.x; d < 256; d += BLK)
+        counts[(int64_t)d * nblocks + blockIdx.x] = hist[d];
+}
+
+// stable scatter: in-LDS 8 x 1-bit split of (digit, local idx) pairs, then
+// rank within digit run + scanned global base.
+__global__ void k_radix_scatter(const uint64_t *k_in, const uint64_t *v_in,
+                                const int64_t *w_in, int64_t n, int byte,
+                                int64_t nblocks, const uint64_t *scanned,
+                                uint64_t *k_out, uint64_t *v_out,
+                                int64_t *w_out) {
+    __shared__ uint32_t buf_a[SORT_TILE];
+    __shared__ uint32_t buf_b[SORT_TILE];
+    __shared__ uint32_t run_start[256];
+    __shared__ uint64_t wave_tot[BLK / WAVE];
+
+    int64_t base = (int64_t)blockIdx.x * SORT_TILE;
+    int64_t tile_n = min((int64_t)SORT_TILE, n - base);
+    if (tile_n <= 0) return;
+
+    // load packed (digit << 16 | idx)
+    for (int i = threadIdx.x; i < SORT_TILE; i += BLK) {
+        uint32_t d = 0xFF;  // pad with max digit so pads sort last
+        if (i < tile_n) d = sort_digit(k_in[base + i], v_in[base + i], byte);
+        buf_a[i] = (d << 16) | (uint32_t)i;
+    }
+    __syncthreads();
+
+    uint32_t *src = buf_a, *dst = buf_b;
+    for (int bit = 0; bit < 8; bit++) {
+        // stable 1-bit split: zeros first (keep order), then ones
+        // per-thread sequential items for stable ordering
+        uint64_t flags[SORT_ITEMS];
+        uint64_t tsum = 0;
+        for (int i = 0; i < SORT_ITEMS; i++) {
+            int idx = threadIdx.x * SORT_ITEMS + i;
+            flags[i] = ((src[idx] >> (16 + bit)) & 1) ? 0 : 1;  // 1 == is-zero
+            tsum += flags[i];
+        }
+        uint64_t wtotal;
+        uint64_t toff = wave_scan_excl(tsum, wtotal);
+        int wid = threadIdx.x / WAVE;
+        if ((threadIdx.x & (WAVE - 1)) == WAVE - 1) wave_tot[wid] = wtotal;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            uint64_t acc = 0;
+            for (int w = 0; w < BLK / WAVE; w++) {
+                uint64_t t = wave_tot[w];
+                wave_tot[w] = acc;
+                acc += t;
+            }
+            run_start[0] = (uint32_t)acc;  // total zeros, reuse smem slot
+        }
+        __syncthreads();
+        uint64_t zeros_before = wave_tot[wid] + toff;
+        uint32_t total_zeros = run_start[0];
+        __syncthreads();
+        for (int i = 0; i < SORT_ITEMS; i++) {
+            int idx = threadIdx.x * SORT_ITEMS + i;
+            uint32_t e = src[idx];
+            uint32_t pos;
+            if (flags[i]) {
+                pos = (uint32_t)zeros_before;
+                zeros_before++;
+            } else {
+                pos = total_zeros + (uint32_t)(idx - zeros_before);
+            }
+            dst[pos] = e;
+        }
+        __syncthreads();
+        uint32_t *t = src; src = dst; dst = t;
+    }
+
+    // run starts per digit
+    for (int i = threadIdx.x; i < 256; i += BLK) run_start[i] = 0xFFFFFFFFu;
+    __syncthreads();
+    for (int i = threadIdx.x; i < tile_n; i += BLK) {
+        uint32_t d = src[i] >> 16;
+        if (i == 0 || (src[i - 1] >> 16) != d) run_start[d] = i;
+    }
+    __syncthreads();
+
+    // scatter rows to scanned global positions
+    for (int i = threadIdx.x; i < tile_n; i += BLK) {
+        uint32_t e = src[i];
+        uint32_t d = e >> 16;
+        uint32_t local = (uint32_t)i - run_start[d];
+        uint64_t pos = scanned[(int64_t)d * nblocks + blockIdx.x] + local;
+        int64_t gi = base + (e & 0xFFFF);
+        k_out[pos] = k_in[gi];
+        v_out[pos] = v_in[gi];
+        w_out[pos] = w_in[gi];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// consolidate sorted rows: head flags -> scan -> segment-sum -> nonzero compact
+// ---------------------------------------------------------------------------
+
+__global__ void k_head_flags(const uint64_t *k, const uint64_t *v, int64_t n,
+                             uint64_t *flags) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        flags[i] = (i == 0) || k[i] != k[i - 1] || v[i] != v[i - 1];
+}
+
+// seg[i] implied by scanned flags; accumulate weights and write unique (k,v)
+__global__ void k_seg_accum(const uint64_t *k, const uint64_t *v,
+                            const int64_t *w, const uint64_t *flag_scan,
+                            const uint64_t *flags, int64_t n, uint64_t *ok,
+                            uint64_t *ov, int64_t *ow) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t seg = flags[i] ? flag_scan[i] : flag_scan[i] - 1;
+        atomicAdd((unsigned long long *)&ow[seg], (unsigned long long)w[i]);
+        if (flags[i]) {
+            ok[seg] = k[i];
+            ov[seg] = v[i];
+        }
+    }
+}
+
+__global__ void k_nonzero_flags(const int64_t *w, int64_t n, uint64_t *flags) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        flags[i] = w[i] != 0;
+}
+
+__global__ void k_compact(const uint64_t *k, const uint64_t *v, const int64_t *w,
+                          const uint64_t *flags, const uint64_t *flag_scan,
+                          int64_t n, uint64_t *ok, uint64_t *ov, int64_t *ow) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        if (flags[i]) {
+            uint64_t pos = flag_scan[i];
+            ok[pos] = k[i];
+            ov[pos] = v[i];
+            ow[pos] = w[i];
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// merge-path merge of two consolidated batches
+// (the trace-merge hot kernel; replaces push_merge/merge_step)
+// ---------------------------------------------------------------------------
+
+__device__ inline bool row_lt(uint64_t ak, uint64_t av, uint64_t bk, uint64_t bv) {
+    return ak != bk ? ak < bk : av < bv;
+}
+__device__ inline bool row_eq(uint64_t ak, uint64_t av, uint64_t bk, uint64_t bv) {
+    return ak == bk && av == bv;
+}
+
+// merge-path split: find (ai, bi), ai + bi = diag, such that
+// a[0..ai) and b[0..bi) are exactly the first `diag` outputs
+// (tie-break: a before b).
+__device__ inline void merge_path(const uint64_t *ak, const uint64_t *av,
+                                  int64_t na, const uint64_t *bk,
+                                  const uint64_t *bv, int64_t nb, int64_t diag,
+                                  int64_t &ai, int64_t &bi) {
+    int64_t lo = max((int64_t)0, diag - nb);
+    int64_t hi = min(diag, na);
+    while (lo < hi) {
+        int64_t mid = (lo + hi) / 2;         // candidate ai
+        int64_t j = diag - mid - 1;          // b index to compare
+        // a[mid] vs b[j]: if a[mid] <= b[j] (tie to a), we can take more a
+        if (!row_lt(bk[j], bv[j], ak[mid], av[mid]))  // b[j] >= a[mid]
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    ai = lo;
+    bi = diag - lo;
+}
+
+// adjust a split so an equal (k,v) pair is never separated: with a-before-b
+// tie-breaking, the pair (a[ai-1], b[bi]) is the only possible split pair.
+__device__ inline void adjust_split(const uint64_t *ak, const uint64_t *av,
+                                    const uint64_t *bk, const uint64_t *bv,
+                                    int64_t na, int64_t nb, int64_t &ai,
+                                    int64_t &bi) {
+    if (ai > 0 && bi < nb && row_eq(ak[ai - 1], av[ai - 1], bk[bi], bv[bi])) bi++;
+}
+
+#define MERGE_PER_THREAD 8
+#define MERGE_TILE (BLK * MERGE_PER_THREAD)
+
+// one thread merges its sub-range, counting or emitting
+template <bool EMIT>
+__global__ void k_merge_pass(const uint64_t *ak, const uint64_t *av,
+                             const int64_t *aw, int64_t na, const uint64_t *bk,
+                             const uint64_t *bv, const int64_t *bw, int64_t nb,
+                             uint64_t *thread_counts,  // count phase out / emit phase in (scanned)
+                             uint64_t *ok, uint64_t *ov, int64_t *ow) {
+    int64_t total = na + nb;
+    int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
+    int64_t tid = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+    // per-thread diagonal range (grid sized so nthreads*MERGE_PER_THREAD >= total)
+    int64_t d0 = min(tid * MERGE_PER_THREAD, total);
+    int64_t d1 = min(d0 + MERGE_PER_THREAD, total);
+    int64_t ai, bi, ae, be;
+    merge_path(ak, av, na, bk, bv, nb, d0, ai, bi);
+    adjust_split(ak, av, bk, bv, na, nb, ai, bi);
+    merge_path(ak, av, na, bk, bv, nb, d1, ae, be);
+    adjust_split(ak, av, bk, bv, na, nb, ae, be);
+    uint64_t cnt = 0;
+    uint64_t opos = EMIT ? thread_counts[tid] : 0;
+    while (ai < ae || bi < be) {
+        bool take_a;
+        bool combine = false;
+        if (ai < ae && bi < be) {
+            if (row_eq(ak[ai], av[ai], bk[bi], bv[bi])) {
+                combine = true;
+                take_a = true;
+            } else {
+                take_a = row_lt(ak[ai], av[ai], bk[bi], bv[bi]);
+            }
+        } else {
+            take_a = ai < ae;
+        }
+        if (combine) {
+            int64_t s = aw[ai] + bw[bi];
+            if (s != 0) {
+                if (EMIT) { ok[opos] = ak[ai]; ov[opos] = av[ai]; ow[opos] = s; }
+                cnt++; opos++;
+            }
+            ai++; bi++;
+        } else if (take_a) {
+            if (EMIT) { ok[opos] = ak[ai]; ov[opos] = av[ai]; ow[opos] = aw[ai]; }
+            cnt++; opos++; ai++;
+        } else {
+            if (EMIT) { ok[opos] = bk[bi]; ov[opos] = bv[bi]; ow[opos] = bw[bi]; }
+            cnt++; opos++; bi++;
+        }
+    }
+    if (!EMIT) thread_counts[tid] = cnt;
+}
+
+// ---------------------------------------------------------------------------
+// join: delta x trace with projection (count/emit)
+// ---------------------------------------------------------------------------
+
+__device__ inline int64_t lower_bound_k(const uint64_t *k, int64_t n,
+                                        uint64_t key) {
+    int64_t lo = 0, hi = n;
+    while (lo < hi) {
+        int64_t mid = (lo + hi) / 2;
+        if (k[mid] < key) lo = mid + 1; else hi = mid;
+    }
+    return lo;
+}
+__device__ inline int64_t upper_bound_k(const uint64_t *k, int64_t n,
+                                        uint64_t key) {
+    int64_t lo = 0, hi = n;
+    while (lo < hi) {
+        int64_t mid = (lo + hi) / 2;
+        if (k[mid] <= key) lo = mid + 1; else hi = mid;
+    }
+    return lo;
+}
+
+__global__ void k_join_count(const uint64_t *dk, int64_t nd, const uint64_t *tk,
+                             int64_t nt, uint64_t *counts, uint64_t *starts) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nd;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t key = dk[i];
+        int64_t lo = lower_bound_k(tk, nt, key);
+        int64_t hi = lo;
+        // galloping upper bound (advance.rs:11-60 analog): trace runs per key
+        // are usually short; probe forward before a full binary search
+        if (lo < nt && tk[lo] == key) {
+            int64_t step = 1;
+            hi = lo;
+            while (hi + step < nt && tk[hi + step] == key) { hi += step; step <<= 1; }
+            hi = upper_bound_k(tk + hi, min(step, nt - hi), key) + hi;
+        }
+        counts[i] = hi - lo;
+        starts[i] = lo;
+    }
+}
+
+__device__ inline void proj_out(int proj, uint64_t param, uint64_t k,
+                                uint64_t v1, uint64_t v2, uint64_t &hi,
+                                uint64_t &lo) {
+    switch (proj) {
+        case DBSP_PROJ_HI_V2_LO_V1: hi = v2; lo = v1; break;
+        case DBSP_PROJ_HI_V1_LO_V2: hi = v1; lo = v2; break;
+        case DBSP_PROJ_HI_K_LO_V1V2: hi = k; lo = (v1 << 32) | (v2 & 0xFFFFFFFFull); break;
+        case DBSP_PROJ_HI_K_LO_V1RND: {
+            uint64_t dt = v1 & 0xFFFFFFFFull;
+            hi = k; lo = (v1 & 0xFFFFFFFF00000000ull) | (dt - dt % param);
+            break;
+        }
+        case DBSP_PROJ_HI_K_LO_V2RND: {
+            uint64_t dt = v2 & 0xFFFFFFFFull;
+            hi = k; lo = (v2 & 0xFFFFFFFF00000000ull) | (dt - dt % param);
+            break;
+        }
+        case DBSP_PROJ_HI_V2_LO_K: hi = v2; lo = k; break;
+        case DBSP_PROJ_HI_V1_LO_K: hi = v1; lo = k; break;
+        case DBSP_PROJ_HI_K_LO_V2V1: hi = k; lo = (v2 << 32) | (v1 & 0xFFFFFFFFull); break;
+        default: hi = 0; lo = 0; break;
+    }
+}
+
+// emit over the OUTPUT index space: balanced regardless of per-key skew
+// (hot-seller keys have thousands of matches; one-thread-per-delta-row would
+// serialize them)
+__global__ void k_join_emit(const uint64_t *dk, const uint64_t *dv,
+                            const int64_t *dw, int64_t nd, const uint64_t *tv,
+                            const int64_t *tw, const uint64_t *offsets,
+                            const uint64_t *starts, int64_t n_out, int proj,
+                            uint64_t param, uint64_t *ok, uint64_t *ov,
+                            int64_t *ow) {
+    for (int64_t o = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; o < n_out;
+         o += (int64_t)gridDim.x * blockDim.x) {
+        // find delta row: offsets is the exclusive scan of counts (length nd)
+        int64_t lo = 0, hi = nd;
+        while (lo < hi) {  // upper_bound(offsets, o) - 1
+            int64_t mid = (lo + hi) / 2;
+            if (offsets[mid] <= (uint64_t)o) lo = mid + 1; else hi = mid;
+        }
+        int64_t i = lo - 1;
+        int64_t j = o - (int64_t)offsets[i];
+        int64_t t = (int64_t)starts[i] + j;
+        uint64_t hi_o, lo_o;
+        proj_out(proj, param, dk[i], dv[i], tv[t], hi_o, lo_o);
+        ok[o] = hi_o;
+        ov[o] = lo_o;
+        ow[o] = dw[i] * tw[t];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// aggregate (linear / max) + upsert  (count/emit over delta keys)
+// ---------------------------------------------------------------------------
+
+template <bool MAX>
+__global__ void k_agg_count(const uint64_t *keys, int64_t nd,
+                            const uint64_t *ik, const int64_t *iw, int64_t ni,
+                            const uint64_t *ok_, int64_t no,
+                            uint64_t *counts, uint64_t *istart, uint64_t *ostart,
+                            uint64_t *newval, uint64_t *has_new) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nd;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t key = keys[i];
+        int64_t ilo = lower_bound_k(ik, ni, key);
+        int64_t ihi = upper_bound_k(ik, ni, key);
+        uint64_t nv = 0;
+        uint64_t hn = 0;
+        if (MAX) {
+            // consolidated trace: any present val has w != 0; max = last val
+            if (ihi > ilo) { nv = 1; hn = 1; }
+            if (hn) nv = 0;  // placeholder; real val read in emit via istart/iend
+        } else {
+            int64_t s = 0;
+            for (int64_t t = ilo; t < ihi; t++) s += iw[t];
+            if (s != 0) { nv = (uint64_t)s; hn = 1; }
+        }
+        int64_t olo = lower_bound_k(ok_, no, key);
+        int64_t ohi = upper_bound_k(ok_, no, key);
+        counts[i] = (hn ? 1 : 0) + (ohi - olo);
+        istart[i] = (uint64_t)ilo;
+        ostart[i] = (uint64_t)olo;
+        newval[i] = MAX ? (ihi > ilo ? /* iend */ (uint64_t)ihi : 0) : nv;
+        has_new[i] = hn;
+    }
+}
+
+template <bool MAX>
+__global__ void k_agg_emit(const uint64_t *keys, int64_t nd,
+                           const uint64_t *iv, const uint64_t *ov_,
+                           const int64_t *ow_, const uint64_t *offsets,
+                           const uint64_t *counts, const uint64_t *istart,
+                           const uint64_t *ostart, const uint64_t *newval,
+                           const uint64_t *has_new, uint64_t *rk, uint64_t *rv,
+                           int64_t *rw) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nd;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t key = keys[i];
+        uint64_t pos = offsets[i];
+        uint64_t cnt = counts[i];
+        uint64_t emitted = 0;
+        if (has_new[i]) {
+            uint64_t nv;
+            if (MAX) {
+                nv = iv[newval[i] - 1];  // last val of key's run (max.rs:40-52)
+            } else {
+                nv = newval[i];
+            }
+            rk[pos] = key; rv[pos] = nv; rw[pos] = 1;
+            pos++; emitted++;
+        }
+        // retractions from the output trace (upsert.rs:180-195)
+        for (uint64_t t = ostart[i]; emitted < cnt; t++, emitted++, pos++) {
+            rk[pos] = key; rv[pos] = ov_[t]; rw[pos] = -ow_[t];
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// window ranges (window.rs:144-220): 4 slice copies with sign
+// ---------------------------------------------------------------------------
+
+// compute the 4 region ranges on device: [r0lo,r0hi) retract, [r1lo,r1hi)
+// retract-shrink, [r2lo,r2hi) insert (trace), [r3lo,r3hi) insert (batch)
+__global__ void k_window_ranges(const uint64_t *tk, int64_t nt,
+                                const uint64_t *bk, int64_t nb, int have_prev,
+                                uint64_t s0, uint64_t e0, uint64_t s1,
+                                uint64_t e1, int64_t *ranges /* [8] */) {
+    if (threadIdx.x != 0 || blockIdx.x != 0) return;
+    int64_t r0lo = 0, r0hi = 0, r1lo = 0, r1hi = 0, r2lo = 0, r2hi = 0;
+    if (have_prev) {
+        uint64_t r0end = min(s1, e0);
+        r0lo = lower_bound_k(tk, nt, s0);
+        r0hi = max(r0lo, (int64_t)lower_bound_k(tk, nt, r0end));
+        if (e1 < e0) {
+            r1lo = lower_bound_k(tk, nt, e1);
+            r1hi = max(r1lo, (int64_t)lower_bound_k(tk, nt, e0));
+        }
+        uint64_t r2start = max(e0, s1);
+        r2lo = lower_bound_k(tk, nt, r2start);
+        r2hi = max(r2lo, (int64_t)lower_bound_k(tk, nt, e1));
+    }
+    int64_t r3lo = lower_bound_k(bk, nb, s1);
+    int64_t r3hi = max(r3lo, (int64_t)lower_bound_k(bk, nb, e1));
+    ranges[0] = r0lo; ranges[1] = r0hi; ranges[2] = r1lo; ranges[3] = r1hi;
+    ranges[4] = r2lo; ranges[5] = r2hi; ranges[6] = r3lo; ranges[7] = r3hi;
+}
+
+__global__ void k_window_emit(const uint64_t *tk, const uint64_t *tv,
+                              const int64_t *tw, const uint64_t *bk,
+                              const uint64_t *bv, const int64_t *bw,
+                              const int64_t *ranges, uint64_t *ok, uint64_t *ov,
+                              int64_t *ow) {
+    int64_t len0 = ranges[1] - ranges[0];
+    int64_t len1 = ranges[3] - ranges[2];
+    int64_t len2 = ranges[5] - ranges[4];
+    int64_t len3 = ranges[7] - ranges[6];
+    int64_t total = len0 + len1 + len2 + len3;
+    for (int64_t o = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; o < total;
+         o += (int64_t)gridDim.x * blockDim.x) {
+        int64_t src;
+        int64_t sign;
+        const uint64_t *k;
+        const uint64_t *v;
+        const int64_t *w;
+        if (o < len0) { src = ranges[0] + o; sign = -1; k = tk; v = tv; w = tw; }
+        else if (o < len0 + len1) { src = ranges[2] + (o - len0); sign = -1; k = tk; v = tv; w = tw; }
+        else if (o < len0 + len1 + len2) { src = ranges[4] + (o - len0 - len1); sign = 1; k = tk; v = tv; w = tw; }
+        else { src = ranges[6] + (o - len0 - len1 - len2); sign = 1; k = bk; v = bv; w = bw; }
+        ok[o] = k[src];
+        ov[o] = v[src];
+        ow[o] = sign * w[src];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// shard partition: xxh3(key) % nshards (shard.rs:165-199, hash.rs:9-13)
+// ---------------------------------------------------------------------------
+
+__device__ inline uint64_t dev_xxh3_u64(uint64_t key, uint64_t seed) {
+    const uint64_t SEC8_16 = 0x1cad21f72c81017cull ^ 0xdb979083e96dd4deull;
+    // read64(kSecret+8) = 0x1cad21f72c81017c, read64(kSecret+16) = 0xdb979083e96dd4de
+    uint64_t s = seed ^ ((uint64_t)__builtin_bswap32((uint32_t)seed) << 32);
+    uint32_t in_lo = (uint32_t)key;
+    uint32_t in_hi = (uint32_t)(key >> 32);
+    uint64_t bitflip = SEC8_16 - s;
+    uint64_t input64 = (uint64_t)in_hi + ((uint64_t)in_lo << 32);
+    uint64_t h = input64 ^ bitflip;
+    uint64_t r49 = (h << 49) | (h >> 15);
+    uint64_t r24 = (h << 24) | (h >> 40);
+    h ^= r49 ^ r24;
+    h *= 0x9FB21C651E98DF25ull;
+    h ^= (h >> 35) + 8;
+    h *= 0x9FB21C651E98DF25ull;
+    h ^= h >> 28;
+    return h;
+}
+
+#define DBSP_HASH_SEED 0x7f95ef85be33c337ull  /* hash.rs:6 */
+
+__global__ void k_shard_hist(const uint64_t *k, int64_t n, int nshards,
+                             uint64_t *hist) {
+    __shared__ uint64_t sh[64];
+    for (int i = threadIdx.x; i < nshards; i += BLK) sh[i] = 0;
+    __syncthreads();
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        atomicAdd((unsigned long long *)&sh[dev_xxh3_u64(k[i], DBSP_HASH_SEED) % nshards], 1ull);
+    __syncthreads();
+    for (int i = threadIdx.x; i < nshards; i += BLK)
+        atomicAdd((unsigned long long *)&hist[i], (unsigned long long)sh[i]);
+}
+
+// unstable within a shard (receivers re-consolidate; Z-set semantics are
+// order-invariant — shard.rs re-assembles via Spine + consolidate anyway)
+__global__ void k_shard_scatter(const uint64_t *k, const uint64_t *v,
+                                const int64_t *w, int64_t n, int nshards,
+                                uint64_t *cursor, uint64_t *ok, uint64_t *ov,
+                                int64_t *ow) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int s = (int)(dev_xxh3_u64(k[i], DBSP_HASH_SEED) % nshards);
+        uint64_t pos = atomicAdd((unsigned long long *)&cursor[s], 1ull);
+        ok[pos] = k[i];
+        ov[pos] = v[i];
+        ow[pos] = w[i];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Nexmark flat_map_index kernels (ingress: input.rs:591-721 + per-query
+// flat_map_index in queries/q{3,5,8}.rs); ticket-append (order-free, the
+// consolidate that follows sorts)
+// ---------------------------------------------------------------------------
+
+__global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
+                          uint64_t *c0, uint64_t *k0, uint64_t *v0, int64_t *w0,
+                          uint64_t *c1, uint64_t *k1, uint64_t *v1, int64_t *w1) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        dbsp_event e = ev[i];
+        if (query == 3) {
+            // q3.rs:37-49
+            if (e.kind == 1 && e.f2 == 10) {
+                uint64_t p = atomicAdd((unsigned long long *)c0, 1ull);
+                k0[p] = e.f1; v0[p] = e.f0; w0[p] = e.w;
+            } else if (e.kind == 0 &&
+                       (e.f3 == 1 || e.f3 == 2 || e.f3 == 3)) {  // CA, ID, OR
+                uint64_t p = atomicAdd((unsigned long long *)c1, 1ull);
+                k1[p] = e.f0;
+                v1[p] = (e.f1 << 16) | ((e.f2 & 0xFF) << 8) | (e.f3 & 0xFF);
+                w1[p] = e.w;
+            }
+        } else if (query == 5) {
+            // q5.rs:79-83: bids by time
+            if (e.kind == 2) {
+                uint64_t p = atomicAdd((unsigned long long *)c0, 1ull);
+                k0[p] = e.f3; v0[p] = e.f0; w0[p] = e.w;
+            }
+        } else if (query == 8) {
+            // q8.rs:50-60
+            if (e.kind == 0) {
+                uint64_t p = atomicAdd((unsigned long long *)c0, 1ull);
+                k0[p] = e.f4; v0[p] = (e.f0 << 32) | (e.f1 & 0xFFFFFFFFull);
+                w0[p] = e.w;
+            } else if (e.kind == 1) {
+                uint64_t p = atomicAdd((unsigned long long *)c1, 1ull);
+                k1[p] = e.f3; v1[p] = e.f1; w1[p] = e.w;
+            }
+        }
+    }
+}
+
+// generic per-row map for the small derived streams (q5/q8):
+// mode 0: (k,v) -> (v>>32, (v&lo32)<<32 | (k&lo32))   [q8 people_by_id map_index]
+// mode 1: (k,v) -> (v, 0)                             [q8 auctions map / q5 windowed-bids map]
+// mode 2: (k,v) -> (0, v)                             [q5 map_index ((),count)]
+// mode 3: (k,v) -> (v, k)                             [q5 by_count map_index]
+__global__ void k_map(const uint64_t *k, const uint64_t *v, const int64_t *w,
+                      int64_t n, int mode, uint64_t *ok, uint64_t *ov,
+                      int64_t *ow) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t kk = k[i], vv = v[i];
+        uint64_t rk, rv;
+        switch (mode) {
+            case 0: rk = vv >> 32; rv = ((vv & 0xFFFFFFFFull) << 32) | (kk & 0xFFFFFFFFull); break;
+            case 1: rk = vv; rv = 0; break;
+            case 2: rk = 0; rv = vv; break;
+            default: rk = vv; rv = kk; break;
+        }
+        ok[i] = rk;
+        ov[i] = rv;
+        ow[i] = w[i];
+    }
+}
+
+// ===========================================================================
+// host-side primitive layer (exported via engine.cpp which owns dbsp_ctx);
+// the functions below are internal helpers shared with engine.cpp
+// ===========================================================================
+
+#include "kernels_iface.hpp"
+
+namespace dbspk {
+
+dbsp_status scan_excl(hipStream_t s, const uint64_t *in, uint64_t *out,
+                      int64_t n, uint64_t *h_total) {
+    return scan_exclusive(s, in, out, n, h_total);
+}
+
+void fill_u64(hipStream_t s, uint64_t *p, uint64_t v, int64_t n) {
+    k_fill_u64<<<grid_for(n), BLK, 0, s>>>(p, v, n);
+}
+
+// sort rows by (k major, v minor); ping-pong scratch must hold n rows
+// (kk2/vv2/ww2).  Skips byte passes above the significant bytes of max(k)/max(v).
+dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
+                      int64_t n, uint64_t *kk2, uint64_t *vv2, int64_t *ww2,
+                      bool *result_in_scratch) {
+    *result_in_scratch = false;
+    if (n <= 1) return DBSP_OK;
+    // significant bytes from max values
+    uint64_t *d_max;
+    HIP_CHECK(hipMallocAsync(&d_max, 2 * sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(d_max, 0, 2 * sizeof(uint64_t), s));
+    k_minmax_u64<<<grid_for(n), BLK, 0, s>>>(kk, vv, n, d_max);
+    uint64_t h_max[2];
+    HIP_CHECK(hipMemcpyAsync(h_max, d_max, 2 * sizeof(uint64_t),
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    HIP_CHECK(hipFreeAsync(d_max, s));
+    int kbytes = 0, vbytes = 0;
+    while (kbytes < 8 && (h_max[0] >> (8 * kbytes)) != 0) kbytes++;
+    while (vbytes < 8 && (h_max[1] >> (8 * vbytes)) != 0) vbytes++;
+
+    int64_t nblocks = ceil_div(n, SORT_TILE);
+    uint64_t *counts;
+    HIP_CHECK(hipMallocAsync(&counts, (int64_t)256 * nblocks * sizeof(uint64_t), s));
+
+    uint64_t *src_k = kk, *src_v = vv; int64_t *src_w = ww;
+    uint64_t *dst_k = kk2, *dst_v = vv2; int64_t *dst_w = ww2;
+    for (int byte = 0; byte < 16; byte++) {
+        bool is_v = byte < 8;
+        if (is_v && (byte & 7) >= vbytes) continue;
+        if (!is_v && (byte & 7) >= kbytes) continue;
+        k_radix_hist<<<dim3((uint32_t)nblocks), BLK, 0, s>>>(src_k, src_v, n, byte,
+                                                            nblocks, counts);
+        dbsp_status st = scan_exclusive(s, counts, counts, 256 * nblocks, nullptr);
+        if (st != DBSP_OK) return st;
+        k_radix_scatter<<<dim3((uint32_t)nblocks), BLK, 0, s>>>(
+            src_k, src_v, src_w, n, byte, nblocks, counts, dst_k, dst_v, dst_w);
+        uint64_t *t;
+        int64_t *tw;
+        t = src_k; src_k = dst_k; dst_k = t;
+        t = src_v; src_v = dst_v; dst_v = t;
+        tw = src_w; src_w = dst_w; dst_w = tw;
+    }
+    HIP_CHECK(hipFreeAsync(counts, s));
+    *result_in_scratch = (src_k != kk);
+    return DBSP_OK;
+}
+
+// consolidate SORTED rows into freshly allocated output; returns exact length
+dbsp_status consolidate_sorted(hipStream_t s, const uint64_t *kk,
+                               const uint64_t *vv, const int64_t *ww, int64_t n,
+                               uint64_t **ok, uint64_t **ov, int64_t **ow,
+                               int64_t *out_n) {
+    if (n == 0) {
+        *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
+        return DBSP_OK;
+    }
+    uint64_t *flags, *fscan;
+    HIP_CHECK(hipMallocAsync(&flags, n * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&fscan, n * sizeof(uint64_t), s));
+    k_head_flags<<<grid_for(n), BLK, 0, s>>>(kk, vv, n, flags);
+    uint64_t nseg = 0;
+    dbsp_status st = scan_exclusive(s, flags, fscan, n, &nseg);
+    if (st != DBSP_OK) return st;
+    uint64_t *sk, *sv; int64_t *sw;
+    HIP_CHECK(hipMallocAsync(&sk, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&sv, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&sw, nseg * sizeof(int64_t) + 8, s));
+    HIP_CHECK(hipMemsetAsync(sw, 0, nseg * sizeof(int64_t), s));
+    k_seg_accum<<<grid_for(n), BLK, 0, s>>>(kk, vv, ww, fscan, flags, n, sk, sv, sw);
+    // drop zero-weight segments
+    uint64_t *nzflags, *nzscan;
+    HIP_CHECK(hipMallocAsync(&nzflags, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&nzscan, nseg * sizeof(uint64_t) + 8, s));
+    k_nonzero_flags<<<grid_for(nseg), BLK, 0, s>>>(sw, nseg, nzflags);
+    uint64_t nout = 0;
+    st = scan_exclusive(s, nzflags, nzscan, nseg, &nout);
+    if (st != DBSP_OK) return st;
+    uint64_t *rk, *rv; int64_t *rw;
+    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    k_compact<<<grid_for(nseg), BLK, 0, s>>>(sk, sv, sw, nzflags, nzscan, nseg,
+                                             rk, rv, rw);
+    HIP_CHECK(hipFreeAsync(flags, s));
+    HIP_CHECK(hipFreeAsync(fscan, s));
+    HIP_CHECK(hipFreeAsync(sk, s));
+    HIP_CHECK(hipFreeAsync(sv, s));
+    HIP_CHECK(hipFreeAsync(sw, s));
+    HIP_CHECK(hipFreeAsync(nzflags, s));
+    HIP_CHECK(hipFreeAsync(nzscan, s));
+    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
+    return DBSP_OK;
+}
+
+dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
+                       const int64_t *aw, int64_t na, const uint64_t *bk,
+                       const uint64_t *bv, const int64_t *bw, int64_t nb,
+                       uint64_t **ok, uint64_t **ov, int64_t **ow,
+                       int64_t *out_n) {
+    int64_t total = na + nb;
+    if (total == 0) {
+        *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
+        return DBSP_OK;
+    }
+    int64_t nthreads = ceil_div(total, MERGE_PER_THREAD);
+    int64_t nblocks = ceil_div(nthreads, BLK);
+    int64_t padded_threads = nblocks * BLK;
+    uint64_t *tcounts;
+    HIP_CHECK(hipMallocAsync(&tcounts, (padded_threads + 1) * sizeof(uint64_t), s));
+    k_merge_pass<false><<<dim3((uint32_t)nblocks), BLK, 0, s>>>(
+        ak, av, aw, na, bk, bv, bw, nb, tcounts, nullptr, nullptr, nullptr);
+    uint64_t nout = 0;
+    dbsp_status st = scan_exclusive(s, tcounts, tcounts, padded_threads, &nout);
+    if (st != DBSP_OK) return st;
+    uint64_t *rk, *rv; int64_t *rw;
+    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    k_merge_pass<true><<<dim3((uint32_t)nblocks), BLK, 0, s>>>(
+        ak, av, aw, na, bk, bv, bw, nb, tcounts, rk, rv, rw);
+    HIP_CHECK(hipFreeAsync(tcounts, s));
+    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
+    return DBSP_OK;
+}
+
+dbsp_status join_rows(hipStream_t s, const uint64_t *dk, const uint64_t *dv,
+                      const int64_t *dw, int64_t nd, const uint64_t *tk,
+                      const uint64_t *tv, const int64_t *tw, int64_t nt,
+                      int proj, uint64_t param, uint64_t **ok, uint64_t **ov,
+                      int64_t **ow, int64_t *out_n) {
+    if (nd == 0 || nt == 0) {
+        *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
+        return DBSP_OK;
+    }
+    uint64_t *counts, *starts;
+    HIP_CHECK(hipMallocAsync(&counts, (nd + 1) * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&starts, nd * sizeof(uint64_t), s));
+    k_join_count<<<grid_for(nd), BLK, 0, s>>>(dk, nd, tk, nt, counts, starts);
+    uint64_t nout = 0;
+    dbsp_status st = scan_exclusive(s, counts, counts, nd, &nout);
+    if (st != DBSP_OK) return st;
+    uint64_t *rk, *rv; int64_t *rw;
+    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    if (nout > 0)
+        k_join_emit<<<grid_for((int64_t)nout), BLK, 0, s>>>(
+            dk, dv, dw, nd, tv, tw, counts, starts, (int64_t)nout, proj, param,
+            rk, rv, rw);
+    HIP_CHECK(hipFreeAsync(counts, s));
+    HIP_CHECK(hipFreeAsync(starts, s));
+    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
+    return DBSP_OK;
+}
+
+template <bool MAX>
+static dbsp_status agg_upsert_impl(hipStream_t s, const uint64_t *keys,
+                                   int64_t nd, const uint64_t *ik,
+                                   const uint64_t *iv, const int64_t *iw,
+                                   int64_t ni, const uint64_t *tok,
+                                   const uint64_t *tov, const int64_t *tow,
+                                   int64_t no, uint64_t **ok, uint64_t **ov,
+                                   int64_t **ow, int64_t *out_n) {
+    if (nd == 0) {
+        *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
+        return DBSP_OK;
+    }
+    uint64_t *counts, *istart, *ostart, *newval, *hasnew, *offsets;
+    HIP_CHECK(hipMallocAsync(&counts, nd * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&istart, nd * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&ostart, nd * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&newval, nd * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&hasnew, nd * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&offsets, nd * sizeof(uint64_t), s));
+    k_agg_count<MAX><<<grid_for(nd), BLK, 0, s>>>(keys, nd, ik, iw, ni, tok, no,
+                                                  counts, istart, ostart, newval,
+                                                  hasnew);
+    uint64_t nout = 0;
+    dbsp_status st = scan_exclusive(s, counts, offsets, nd, &nout);
+    if (st != DBSP_OK) return st;
+    uint64_t *rk, *rv; int64_t *rw;
+    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    if (nout > 0)
+        k_agg_emit<MAX><<<grid_for(nd), BLK, 0, s>>>(keys, nd, iv, tov, tow,
+                                                     offsets, counts, istart,
+                                                     ostart, newval, hasnew, rk,
+                                                     rv, rw);
+    HIP_CHECK(hipFreeAsync(counts, s));
+    HIP_CHECK(hipFreeAsync(istart, s));
+    HIP_CHECK(hipFreeAsync(ostart, s));
+    HIP_CHECK(hipFreeAsync(newval, s));
+    HIP_CHECK(hipFreeAsync(hasnew, s));
+    HIP_CHECK(hipFreeAsync(offsets, s));
+    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
+    return DBSP_OK;
+}
+
+dbsp_status agg_linear_upsert_rows(hipStream_t s, const uint64_t *keys,
+                                   int64_t nd, const uint64_t *ik,
+                                   const uint64_t *iv, const int64_t *iw,
+                                   int64_t ni, const uint64_t *tok,
+                                   const uint64_t *tov, const int64_t *tow,
+                                   int64_t no, uint64_t **ok, uint64_t **ov,
+                                   int64_t **ow, int64_t *out_n) {
+    return agg_upsert_impl<false>(s, keys, nd, ik, iv, iw, ni, tok, tov, tow, no,
+                                  ok, ov, ow, out_n);
+}
+
+dbsp_status agg_max_upsert_rows(hipStream_t s, const uint64_t *keys, int64_t nd,
+                                const uint64_t *ik, const uint64_t *iv,
+                                const int64_t *iw, int64_t ni,
+                                const uint64_t *tok, const uint64_t *tov,
+                                const int64_t *tow, int64_t no, uint64_t **ok,
+                                uint64_t **ov, int64_t **ow, int64_t *out_n) {
+    return agg_upsert_impl<true>(s, keys, nd, ik, iv, iw, ni, tok, tov, tow, no,
+                                 ok, ov, ow, out_n);
+}
+
+dbsp_status window_rows(hipStream_t s, const uint64_t *tk, const uint64_t *tv,
+                        const int64_t *tw, int64_t nt, const uint64_t *bk,
+                        const uint64_t *bv, const int64_t *bw, int64_t nb,
+                        int have_prev, uint64_t s0, uint64_t e0, uint64_t s1,
+                        uint64_t e1, uint64_t **ok, uint64_t **ov, int64_t **ow,
+                        int64_t *out_n) {
+    int64_t *d_ranges;
+    HIP_CHECK(hipMallocAsync(&d_ranges, 8 * sizeof(int64_t), s));
+    k_window_ranges<<<1, 1, 0, s>>>(tk, nt, bk, nb, have_prev, s0, e0, s1, e1,
+                                    d_ranges);
+    int64_t h_ranges[8];
+    HIP_CHECK(hipMemcpyAsync(h_ranges, d_ranges, sizeof(h_ranges),
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    int64_t total = (h_ranges[1] - h_ranges[0]) + (h_ranges[3] - h_ranges[2]) +
+                    (h_ranges[5] - h_ranges[4]) + (h_ranges[7] - h_ranges[6]);
+    uint64_t *rk, *rv; int64_t *rw;
+    HIP_CHECK(hipMallocAsync(&rk, total * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rv, total * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rw, total * sizeof(int64_t) + 8, s));
+    if (total > 0)
+        k_window_emit<<<grid_for(total), BLK, 0, s>>>(tk, tv, tw, bk, bv, bw,
+                                                      d_ranges, rk, rv, rw);
+    HIP_CHECK(hipFreeAsync(d_ranges, s));
+    *ok = rk; *ov = rv; *ow = rw; *out_n = total;
+    return DBSP_OK;
+}
+
+dbsp_status shard_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
+                       const int64_t *w, int64_t n, int nshards, uint64_t *ok,
+                       uint64_t *ov, int64_t *ow, int64_t *h_offsets) {
+    uint64_t *hist;
+    HIP_CHECK(hipMallocAsync(&hist, (nshards + 1) * sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(hist, 0, (nshards + 1) * sizeof(uint64_t), s));
+    if (n > 0) k_shard_hist<<<grid_for(n), BLK, 0, s>>>(k, n, nshards, hist);
+    uint64_t h_hist[64];
+    HIP_CHECK(hipMemcpyAsync(h_hist, hist, nshards * sizeof(uint64_t),
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    int64_t acc = 0;
+    for (int i = 0; i < nshards; i++) {
+        h_offsets[i] = acc;
+        acc += (int64_t)h_hist[i];
+    }
+    h_offsets[nshards] = acc;
+    // cursor initialised to shard starts
+    HIP_CHECK(hipMemcpyAsync(hist, h_offsets, nshards * sizeof(int64_t),
+                             hipMemcpyHostToDevice, s));
+    if (n > 0)
+        k_shard_scatter<<<grid_for(n), BLK, 0, s>>>(k, v, w, n, nshards, hist,
+                                                    ok, ov, ow);
+    HIP_CHECK(hipFreeAsync(hist, s));
+    return DBSP_OK;
+}
+
+dbsp_status flatmap_events(hipStream_t s, const dbsp_event *ev, int64_t n,
+                           int query, uint64_t *k0, uint64_t *v0, int64_t *w0,
+                           int64_t *n0, uint64_t *k1, uint64_t *v1, int64_t *w1,
+                           int64_t *n1) {
+    uint64_t *ctr;
+    HIP_CHECK(hipMallocAsync(&ctr, 2 * sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(ctr, 0, 2 * sizeof(uint64_t), s));
+    if (n > 0)
+        k_flatmap<<<grid_for(n), BLK, 0, s>>>(ev, n, query, ctr, k0, v0, w0,
+                                              ctr + 1, k1, v1, w1);
+    uint64_t h_ctr[2];
+    HIP_CHECK(hipMemcpyAsync(h_ctr, ctr, sizeof(h_ctr), hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    HIP_CHECK(hipFreeAsync(ctr, s));
+    *n0 = (int64_t)h_ctr[0];
+    if (n1) *n1 = (int64_t)h_ctr[1];
+    return DBSP_OK;
+}
+
+dbsp_status map_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
+                     const int64_t *w, int64_t n, int mode, uint64_t *ok,
+                     uint64_t *ov, int64_t *ow) {
+    if (n > 0)
+        k_map<<<grid_for(n), BLK, 0, s>>>(k, v, w, n, mode, ok, ov, ow);
+    return DBSP_OK;
+}
+
+}  // namespace dbspk (reopened below after the extra kernels)
+
+// ---- multi-batch linear aggregate (spine batches summed; linear op) ----
+
+__global__ void k_agg_sum(const uint64_t *keys, int64_t nd, const uint64_t *ik,
+                          const int64_t *iw, int64_t ni, int64_t *acc) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nd;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t key = keys[i];
+        int64_t lo = lower_bound_k(ik, ni, key);
+        int64_t hi = upper_bound_k(ik, ni, key);
+        int64_t s = 0;
+        for (int64_t t = lo; t < hi; t++) s += iw[t];
+        acc[i] += s;
+    }
+}
+
+__global__ void k_emit_nonzero(const uint64_t *keys, const int64_t *acc,
+                               int64_t nd, uint64_t *ctr, uint64_t *ok,
+                               uint64_t *ov, int64_t *ow) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nd;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        if (acc[i] != 0) {
+            uint64_t p = atomicAdd((unsigned long long *)ctr, 1ull);
+            ok[p] = keys[i];
+            ov[p] = (uint64_t)acc[i];
+            ow[p] = 1;
+        }
+    }
+}
+
+__global__ void k_key_head_flags(const uint64_t *k, int64_t n, uint64_t *flags) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        flags[i] = (i == 0) || k[i] != k[i - 1];
+}
+
+__global__ void k_compact_keys(const uint64_t *k, const uint64_t *flags,
+                               const uint64_t *fscan, int64_t n, uint64_t *ok) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        if (flags[i]) ok[fscan[i]] = k[i];
+}
+
+namespace dbspk {
+
+dbsp_status agg_sum_batch(hipStream_t s, const uint64_t *keys, int64_t nd,
+                          const uint64_t *ik, const int64_t *iw, int64_t ni,
+                          int64_t *acc) {
+    if (nd > 0 && ni > 0)
+        k_agg_sum<<<grid_for(nd), BLK, 0, s>>>(keys, nd, ik, iw, ni, acc);
+    return DBSP_OK;
+}
+
+dbsp_status emit_nonzero(hipStream_t s, const uint64_t *keys,
+                         const int64_t *acc, int64_t nd, uint64_t *ok,
+                         uint64_t *ov, int64_t *ow, int64_t *h_count) {
+    uint64_t *ctr;
+    HIP_CHECK(hipMallocAsync(&ctr, sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(ctr, 0, sizeof(uint64_t), s));
+    if (nd > 0)
+        k_emit_nonzero<<<grid_for(nd), BLK, 0, s>>>(keys, acc, nd, ctr, ok, ov, ow);
+    uint64_t h = 0;
+    HIP_CHECK(hipMemcpyAsync(&h, ctr, sizeof(uint64_t), hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    HIP_CHECK(hipFreeAsync(ctr, s));
+    *h_count = (int64_t)h;
+    return DBSP_OK;
+}
+
+dbsp_status unique_keys(hipStream_t s, const uint64_t *kk, int64_t n,
+                        uint64_t **okeys, int64_t *out_n) {
+    if (n == 0) {
+        *okeys = nullptr;
+        *out_n = 0;
+        return DBSP_OK;
+    }
+    uint64_t *flags, *fscan;
+    HIP_CHECK(hipMallocAsync(&flags, n * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&fscan, n * sizeof(uint64_t), s));
+    k_key_head_flags<<<grid_for(n), BLK, 0, s>>>(kk, n, flags);
+    uint64_t nk = 0;
+    dbsp_status st = scan_exclusive(s, flags, fscan, n, &nk);
+    if (st != DBSP_OK) return st;
+    uint64_t *out;
+    HIP_CHECK(hipMallocAsync(&out, nk * sizeof(uint64_t) + 8, s));
+    k_compact_keys<<<grid_for(n), BLK, 0, s>>>(kk, flags, fscan, n, out);
+    HIP_CHECK(hipFreeAsync(flags, s));
+    HIP_CHECK(hipFreeAsync(fscan, s));
+    *okeys = out;
+    *out_n = (int64_t)nk;
+    return DBSP_OK;
+}
+
+uint64_t host_xxh3_u64(uint64_t key, uint64_t seed) {
+    // host copy of dev_xxh3_u64 (kept in sync; parity-tested against the oracle)
+    const uint64_t SEC8_16 = 0x1cad21f72c81017cull ^ 0xdb979083e96dd4deull;
+    uint64_t s = seed ^ ((uint64_t)__builtin_bswap32((uint32_t)seed) << 32);
+    uint64_t input64 = (uint64_t)(uint32_t)(key >> 32) + ((uint64_t)(uint32_t)key << 32);
+    uint64_t h = input64 ^ (SEC8_16 - s);
+    uint64_t r49 = (h << 49) | (h >> 15);
+    uint64_t r24 = (h << 24) | (h >> 40);
+    h ^= r49 ^ r24;
+    h *= 0x9FB21C651E98DF25ull;
+    h ^= (h >> 35) + 8;
+    h *= 0x9FB21C651E98DF25ull;
+    h ^= h >> 28;
+    return h;
+}
+
+}  // namespace dbspk

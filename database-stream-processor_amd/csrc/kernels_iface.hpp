@@ -1,0 +1,87 @@
+// kernels_iface.hpp — internal interface between kernels.hip (device code)
+// and engine.cpp (host-side Circuit/Stream mirror).  Not part of the C ABI.
+#pragma once
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include "../../include/dbsp_hip.h"
+
+namespace dbspk {
+
+dbsp_status scan_excl(hipStream_t s, const uint64_t *in, uint64_t *out,
+                      int64_t n, uint64_t *h_total);
+void fill_u64(hipStream_t s, uint64_t *p, uint64_t v, int64_t n);
+
+dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
+                      int64_t n, uint64_t *kk2, uint64_t *vv2, int64_t *ww2,
+                      bool *result_in_scratch);
+
+dbsp_status consolidate_sorted(hipStream_t s, const uint64_t *kk,
+                               const uint64_t *vv, const int64_t *ww, int64_t n,
+                               uint64_t **ok, uint64_t **ov, int64_t **ow,
+                               int64_t *out_n);
+
+dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
+                       const int64_t *aw, int64_t na, const uint64_t *bk,
+                       const uint64_t *bv, const int64_t *bw, int64_t nb,
+                       uint64_t **ok, uint64_t **ov, int64_t **ow,
+                       int64_t *out_n);
+
+dbsp_status join_rows(hipStream_t s, const uint64_t *dk, const uint64_t *dv,
+                      const int64_t *dw, int64_t nd, const uint64_t *tk,
+                      const uint64_t *tv, const int64_t *tw, int64_t nt,
+                      int proj, uint64_t param, uint64_t **ok, uint64_t **ov,
+                      int64_t **ow, int64_t *out_n);
+
+dbsp_status agg_linear_upsert_rows(hipStream_t s, const uint64_t *keys,
+                                   int64_t nd, const uint64_t *ik,
+                                   const uint64_t *iv, const int64_t *iw,
+                                   int64_t ni, const uint64_t *tok,
+                                   const uint64_t *tov, const int64_t *tow,
+                                   int64_t no, uint64_t **ok, uint64_t **ov,
+                                   int64_t **ow, int64_t *out_n);
+
+dbsp_status agg_max_upsert_rows(hipStream_t s, const uint64_t *keys, int64_t nd,
+                                const uint64_t *ik, const uint64_t *iv,
+                                const int64_t *iw, int64_t ni,
+                                const uint64_t *tok, const uint64_t *tov,
+                                const int64_t *tow, int64_t no, uint64_t **ok,
+                                uint64_t **ov, int64_t **ow, int64_t *out_n);
+
+// multi-batch linear aggregate: accumulate per-delta-key weight sums of one
+// trace batch into acc[nd] (aggregation is linear in the trace, so spines are
+// summed batch by batch)
+dbsp_status agg_sum_batch(hipStream_t s, const uint64_t *keys, int64_t nd,
+                          const uint64_t *ik, const int64_t *iw, int64_t ni,
+                          int64_t *acc);
+// emit (key, acc, +1) for acc != 0 (ticket append; caller consolidates)
+dbsp_status emit_nonzero(hipStream_t s, const uint64_t *keys,
+                         const int64_t *acc, int64_t nd, uint64_t *ok,
+                         uint64_t *ov, int64_t *ow, int64_t *h_count);
+
+dbsp_status window_rows(hipStream_t s, const uint64_t *tk, const uint64_t *tv,
+                        const int64_t *tw, int64_t nt, const uint64_t *bk,
+                        const uint64_t *bv, const int64_t *bw, int64_t nb,
+                        int have_prev, uint64_t s0, uint64_t e0, uint64_t s1,
+                        uint64_t e1, uint64_t **ok, uint64_t **ov, int64_t **ow,
+                        int64_t *out_n);
+
+dbsp_status shard_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
+                       const int64_t *w, int64_t n, int nshards, uint64_t *ok,
+                       uint64_t *ov, int64_t *ow, int64_t *h_offsets);
+
+dbsp_status flatmap_events(hipStream_t s, const dbsp_event *ev, int64_t n,
+                           int query, uint64_t *k0, uint64_t *v0, int64_t *w0,
+                           int64_t *n0, uint64_t *k1, uint64_t *v1, int64_t *w1,
+                           int64_t *n1);
+
+dbsp_status map_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
+                     const int64_t *w, int64_t n, int mode, uint64_t *ok,
+                     uint64_t *ov, int64_t *ow);
+
+// distinct keys of a consolidated (sorted) batch
+dbsp_status unique_keys(hipStream_t s, const uint64_t *kk, int64_t n,
+                        uint64_t **okeys, int64_t *out_n);
+
+uint64_t host_xxh3_u64(uint64_t key, uint64_t seed);
+
+}  // namespace dbspk
