@@ -1,0 +1,230 @@
+#!/usr/bin/env python3
+"""bench.py — Nexmark events/s on the MI355X-native DBSP engine.
+
+A "step" is one clock tick of the flagship query (default q3, the workload
+BASELINE.json's metric is quoted on for one GPU: "Nexmark q3 stream-table
+join, 10M synthetic events, i64 keys, 1 MI355X") over one tick_events-sized
+batch of synthetic events — the reference feeds 40k-event batches per tick
+(crates/nexmark/src/config.rs:108-109).  All (warmup+steps) ticks' events are
+generated up front and staged to HBM before the timed region; the tick
+pipeline (flat_map build, sort+consolidate, trace merges, joins, output
+consolidate) runs entirely on-GPU inside the timed region.
+
+Single GPU:   python bench.py --steps 250 --warmup 10
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N ...
+(one rank per GPU; deltas are key-sharded across ranks with an RCCL
+all-to-all-v over xGMI; weak scaling: each rank processes tick_events per
+tick, so the whole job covers tick_events*N events per step).
+
+Rank 0 prints ONE JSON line (driver contract).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+ROOT = Path(__file__).resolve().parent
+sys.path.insert(0, str(ROOT / "database-stream-processor_amd" / "python"))
+
+HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+# Reference's published q3 number (9,936,407 events/s on an
+# unnamed 16-core CPU — the only published figure for this metric).
+PUBLISHED = {3: 9_936_407.0, 5: 9_906_875.0, 8: 9_380_863.0, 0: 9_926_544.0}
+
+
+def generate_events(total, seed=1):
+    from dbsp_amd import gen
+    return gen.generate(total, seed=seed)
+
+
+def run_engine(ctx_dev, query, events, n_ticks, tick, world, rank, nccl_id,
+               timed_ticks, dist=None):
+    """Stage events; run warmup then timed ticks; return (elapsed_s, engine)."""
+    import torch
+    from dbsp_amd.engine import Ctx, Engine
+
+    ctx = Ctx(ctx_dev)
+    if world > 1:
+        import ctypes
+        ctx._lib.dbsp_comm_init.restype = ctypes.c_int32
+        st = ctx._lib.dbsp_comm_init(ctx._h, rank, world,
+                                     nccl_id.ctypes.data_as(ctypes.c_void_p))
+        assert st == 0, "dbsp_comm_init failed"
+    eng = Engine(ctx, query=query, rank=rank, world=world)
+    eng.stage(events)
+
+    def tick_range(t):
+        # global stream laid out tick-major: tick t covers
+        # [t*tick*world, (t+1)*tick*world); rank r takes its tick-sized slice
+        base = t * tick * world + rank * tick
+        return base, base + tick
+
+    warmup_ticks = n_ticks - timed_ticks
+    for t in range(warmup_ticks):
+        lo, hi = tick_range(t)
+        eng.step_staged(lo, hi)
+    ctx.sync()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    if dist is not None:
+        dist.barrier()
+    t0 = time.perf_counter()
+    for t in range(warmup_ticks, n_ticks):
+        lo, hi = tick_range(t)
+        eng.step_staged(lo, hi)
+    ctx.sync()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if dist is not None:
+        dist.barrier()
+    return elapsed, eng, ctx
+
+
+def cpu_baseline_leg(query, tick, budget_s=15.0):
+    """Time the CPU oracle (the restatement of the crates/dbsp algorithms —
+    kind 'port', single-threaded) on a bounded sample of the same workload."""
+    from dbsp_amd import oracle
+    sample_events = 40 * tick  # ~1.6M events at the default tick; ~10-20 s
+    evs = generate_events(sample_events, seed=1)
+    q = oracle.Query(query)
+    t0 = time.perf_counter()
+    done = 0
+    for lo in range(0, sample_events, tick):
+        q.step(evs[lo:lo + tick], cap=1 << 22)
+        done = lo + tick
+        if time.perf_counter() - t0 > budget_s:
+            break
+    dt = time.perf_counter() - t0
+    q.close()
+    return {
+        "value": done / dt,
+        "unit": "events/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"oracle q{query} over {done} events in {dt:.1f}s "
+                  f"(single-threaded C++ restatement)",
+    }
+
+
+def roofline_leg(query, events, n_ticks, tick):
+    """Re-run the same workload with DBSP_PROFILE=1 (hipEvent timing per
+    kernel class on the engine stream) and derive achieved algorithmic GB/s of
+    the dominant kernel class (trace merge)."""
+    os.environ["DBSP_PROFILE"] = "1"
+    elapsed, eng, ctx = run_engine(0, query, events, n_ticks, tick, 1, 0, None,
+                                   timed_ticks=n_ticks)
+    classes = {0: "sort_consolidate", 1: "merge", 2: "join", 3: "aggregate",
+               4: "window"}
+    stats = {name: eng.kernel_stats(k) for k, name in classes.items()}
+    eng.close()
+    ctx.close()
+    os.environ.pop("DBSP_PROFILE", None)
+    # dominant class by device ms
+    dom = max(stats.items(), key=lambda kv: kv[1][0])
+    name, (ms, by, ln) = dom
+    achieved = (by / 1e9) / (ms / 1e3) if ms > 0 else 0.0
+    return {
+        "bound": "hbm",
+        "kernel_class": name,
+        "achieved": round(achieved, 2),
+        "peak": HBM_PEAK_GBS,
+        "unit": "GB/s",
+        "frac": round(achieved / HBM_PEAK_GBS, 4),
+        "traffic": None,  # PMC pass collected separately (profiles/)
+        "detail": {k: {"ms": round(v[0], 2), "algo_GB": round(v[1] / 1e9, 3),
+                       "launches": v[2]} for k, v in stats.items()},
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=250)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--query", type=int, default=3, choices=[0, 3, 5, 8])
+    ap.add_argument("--tick", type=int, default=40_000)
+    ap.add_argument("--seed", type=int, default=1)
+    ap.add_argument("--no-extras", action="store_true",
+                    help="skip cpu_baseline and roofline legs")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world > 1 and args.query not in (3,):
+        raise SystemExit("multi-GPU sharding is implemented for q3 this round")
+
+    dist = None
+    nccl_id = None
+    if world > 1:
+        import numpy as np
+        import torch.distributed as tdist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        tdist.init_process_group("gloo")
+        dist = tdist
+        import ctypes
+        from dbsp_amd.engine import _L
+        buf = np.zeros(128, dtype=np.uint8)
+        if rank == 0:
+            L = _L()
+            L.dbsp_comm_unique_id.restype = ctypes.c_int32
+            assert L.dbsp_comm_unique_id(
+                buf.ctypes.data_as(ctypes.c_void_p)) == 0
+        import torch
+        t = torch.from_numpy(buf)
+        tdist.broadcast(t, src=0)
+        nccl_id = t.numpy()
+
+    n_ticks = args.warmup + args.steps
+    total_events = n_ticks * args.tick * world
+    events = generate_events(total_events, seed=args.seed)
+
+    elapsed, eng, ctx = run_engine(local_rank, args.query, events, n_ticks,
+                                   args.tick, world, rank, nccl_id,
+                                   timed_ticks=args.steps, dist=dist)
+    # MAX over ranks
+    if dist is not None:
+        import torch
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    eng.close()
+    ctx.close()
+
+    if rank != 0:
+        return
+
+    timed_events = args.steps * args.tick * world
+    value = timed_events / elapsed
+    result = {
+        "metric": "Nexmark events/s",
+        "value": round(value, 1),
+        "unit": "events/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1000, 4),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": round(value / PUBLISHED[args.query], 3),
+        "dtype": "int64",
+        "data": "synthetic",
+        "config": {
+            "workload": f"nexmark-q{args.query}",
+            "events_timed": timed_events,
+            "tick_events": args.tick,
+            "parallelism": f"keyshard{world}" if world > 1 else "single",
+        },
+    }
+    if not args.no_extras and world == 1 and args.query != 0:
+        result["cpu_baseline"] = cpu_baseline_leg(args.query, args.tick)
+        result["roofline"] = roofline_leg(args.query, events, n_ticks, args.tick)
+    print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()
