@@ -71,6 +71,9 @@ struct dbsp_ctx {
     // RCCL
     ncclComm_t comm = nullptr;
     int rank = 0, world = 1;
+    // persistent length scratch (device + pinned host)
+    int64_t *d_len = nullptr;
+    int64_t *h_len = nullptr;
 };
 
 extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
@@ -91,6 +94,8 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     HIP_CHECK_ST(hipEventCreate(&c->ev1));
     const char *p = getenv("DBSP_PROFILE");
     c->profile = p && p[0] == '1';
+    HIP_CHECK_ST(hipMalloc(&c->d_len, sizeof(int64_t)));
+    HIP_CHECK_ST(hipHostMalloc(&c->h_len, sizeof(int64_t)));
     *out = c;
     return DBSP_OK;
 }
@@ -99,6 +104,8 @@ extern "C" dbsp_status dbsp_ctx_destroy(dbsp_ctx *c) {
     if (!c) return DBSP_OK;
     if (c->comm) ncclCommDestroy(c->comm);
     hipStreamSynchronize(c->stream);
+    if (c->d_len) (void)hipFree(c->d_len);
+    if (c->h_len) (void)hipHostFree(c->h_len);
     hipEventDestroy(c->ev0);
     hipEventDestroy(c->ev1);
     hipStreamDestroy(c->stream);
@@ -205,6 +212,22 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
     ScopedTimer t(c, 0, (double)raw.n * 48.0);
     DevBatch scratch;
     TRY(alloc_batch(c, raw.n, scratch));
+    if (raw.n <= 16384) {
+        // fused single-workgroup path: one launch + one length readback
+        DevBatch res;
+        TRY(alloc_batch(c, raw.n, res));
+        TRY(dbspk::sort_cons_small(c->stream, raw.k, raw.v, raw.w, raw.n,
+                                   scratch.k, scratch.v, scratch.w, res.k,
+                                   res.v, res.w, c->d_len));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, sizeof(int64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        res.n = *c->h_len;
+        free_batch(c, raw);
+        free_batch(c, scratch);
+        out = res;
+        return DBSP_OK;
+    }
     bool in_scratch = false;
     TRY(dbspk::sort_rows(c->stream, raw.k, raw.v, raw.w, raw.n, scratch.k,
                          scratch.v, scratch.w, &in_scratch));
@@ -652,19 +675,30 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
     return DBSP_OK;
 }
 
-// join delta against every batch of a spine (join is linear in the trace)
+// join delta against a whole spine in one count/emit pair (join is linear in
+// the trace; spine read path = the reference's CursorList)
 static dbsp_status join_vs_spine(dbsp_ctx *c, const DevBatch &delta,
-                                 const Spine &spine, int proj, uint64_t param,
+                                 Spine &spine, int proj, uint64_t param,
                                  std::vector<DevBatch> &outs) {
+    if (delta.n == 0 || spine.batches.empty()) return DBSP_OK;
+    if ((int)spine.batches.size() > MAX_TRACE_BATCHES)
+        TRY(spine.consolidate_all(c));
+    TraceArgs t{};
     for (auto &b : spine.batches) {
-        if (delta.n == 0 || b.n == 0) continue;
-        DevBatch o;
-        ScopedTimer t(c, 2, (double)delta.n * 24.0);
-        TRY(dbspk::join_rows(c->stream, delta.k, delta.v, delta.w, delta.n, b.k,
-                             b.v, b.w, b.n, proj, param, &o.k, &o.v, &o.w, &o.n));
-        if (o.n > 0) outs.push_back(o);
-        else free_batch(c, o);
+        if (b.n == 0) continue;
+        t.k[t.nb] = b.k;
+        t.v[t.nb] = b.v;
+        t.w[t.nb] = b.w;
+        t.n[t.nb] = b.n;
+        t.nb++;
     }
+    if (t.nb == 0) return DBSP_OK;
+    DevBatch o;
+    ScopedTimer timer(c, 2, (double)delta.n * 24.0);
+    TRY(dbspk::join_spine_rows(c->stream, delta.k, delta.v, delta.w, delta.n, t,
+                               proj, param, &o.k, &o.v, &o.w, &o.n));
+    if (o.n > 0) outs.push_back(o);
+    else free_batch(c, o);
     return DBSP_OK;
 }
 

@@ -28,6 +28,7 @@
 #include <cstdio>
 
 #include "../../include/dbsp_hip.h"
+#include "kernels_iface.hpp"
 
 #define WAVE 64
 #define BLK 256
@@ -333,6 +334,198 @@ __global__ void k_radix_scatter(const uint64_t *k_in, const uint64_t *v_in,
 }
 
 // ---------------------------------------------------------------------------
+// fused single-workgroup sort+consolidate for small batches (n <= 16384)
+//
+// A 40k-event tick produces per-stream deltas of a few thousand rows; the
+// multi-kernel radix path costs ~30 launches + 3 host syncs there, which
+// dominates the tick (measured: 70% of wall in inter-dispatch gaps).  This
+// kernel does the whole consolidate_slice job — LSD radix over the permutation
+// in LDS, then dedup-accumulate and zero-drop — in ONE launch with the output
+// length left in device memory.  1024 threads = 16 waves on one CU; the
+// permutation (digit<<14|idx u32) ping-pongs between two 64 KiB LDS arrays.
+// ---------------------------------------------------------------------------
+
+#define FUSE_MAX 16384
+#define FUSE_THREADS 1024
+#define FUSE_ITEMS (FUSE_MAX / FUSE_THREADS)  // 16
+
+// block-wide exclusive scan over FUSE_MAX u32 flags held per-thread
+// (16 consecutive items per thread); returns thread's exclusive offset and
+// writes the block total to *total (all threads).
+__device__ inline uint32_t fuse_scan(uint32_t thread_sum, uint32_t *wave_tot,
+                                     uint32_t *total) {
+    uint32_t v = thread_sum;
+    for (int d = 1; d < WAVE; d <<= 1) {
+        uint32_t up = __shfl_up(v, d, WAVE);
+        if ((threadIdx.x & (WAVE - 1)) >= d) v += up;
+    }
+    int wid = threadIdx.x / WAVE;  // 16 waves
+    if ((threadIdx.x & (WAVE - 1)) == WAVE - 1) wave_tot[wid] = v;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint32_t acc = 0;
+        for (int w = 0; w < FUSE_THREADS / WAVE; w++) {
+            uint32_t t = wave_tot[w];
+            wave_tot[w] = acc;
+            acc += t;
+        }
+        wave_tot[FUSE_THREADS / WAVE] = acc;
+    }
+    __syncthreads();
+    uint32_t r = wave_tot[wid] + (v - thread_sum);
+    *total = wave_tot[FUSE_THREADS / WAVE];
+    __syncthreads();
+    return r;
+}
+
+__global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(
+    const uint64_t *kin, const uint64_t *vin, const int64_t *win, int64_t n,
+    uint64_t *tk, uint64_t *tv, int64_t *tw,  // scratch (>= n rows)
+    uint64_t *ok, uint64_t *ov, int64_t *ow,  // output (cap >= n rows)
+    int64_t *out_len) {
+    __shared__ uint32_t bufA[FUSE_MAX];
+    __shared__ uint32_t bufB[FUSE_MAX];
+    __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
+    __shared__ uint64_t smax[2];
+    const int tid = threadIdx.x;
+
+    if (n == 0) {
+        if (tid == 0) *out_len = 0;
+        return;
+    }
+
+    // ---- significant bytes of max(k), max(v) ----
+    if (tid == 0) { smax[0] = 0; smax[1] = 0; }
+    __syncthreads();
+    {
+        uint64_t mk = 0, mv = 0;
+        for (int64_t i = tid; i < n; i += FUSE_THREADS) {
+            mk = max(mk, kin[i]);
+            mv = max(mv, vin[i]);
+        }
+        atomicMax((unsigned long long *)&smax[0], (unsigned long long)mk);
+        atomicMax((unsigned long long *)&smax[1], (unsigned long long)mv);
+    }
+    __syncthreads();
+    int kbytes = 0, vbytes = 0;
+    while (kbytes < 8 && (smax[0] >> (8 * kbytes)) != 0) kbytes++;
+    while (vbytes < 8 && (smax[1] >> (8 * vbytes)) != 0) vbytes++;
+
+    // ---- init permutation ----
+    for (int i = tid; i < FUSE_MAX; i += FUSE_THREADS)
+        bufA[i] = (uint32_t)i | (i >= n ? (0xFFu << 18) : 0);  // pads sort last
+    __syncthreads();
+
+    uint32_t *src = bufA, *dst = bufB;
+    const int total_bytes = vbytes + kbytes;
+    for (int pass = 0; pass < total_bytes; pass++) {
+        const bool is_v = pass < vbytes;
+        const int sh = 8 * (is_v ? pass : pass - vbytes);
+        // annotate digits: entry = digit<<18 | idx (idx < 16384 fits 14 bits)
+        for (int i = tid; i < FUSE_MAX; i += FUSE_THREADS) {
+            uint32_t idx = src[i] & 0x3FFFu;
+            uint32_t d = 0xFFu;  // pads (idx >= n) keep the max digit
+            if ((int64_t)idx < n) {
+                uint64_t limb = is_v ? vin[idx] : kin[idx];
+                d = (uint32_t)((limb >> sh) & 0xFF);
+            }
+            src[i] = (d << 18) | idx;
+        }
+        __syncthreads();
+        // 8 stable 1-bit splits
+        for (int bit = 0; bit < 8; bit++) {
+            uint32_t flags[FUSE_ITEMS];
+            uint32_t tsum = 0;
+            for (int j = 0; j < FUSE_ITEMS; j++) {
+                int i = tid * FUSE_ITEMS + j;
+                flags[j] = ((src[i] >> (18 + bit)) & 1) ? 0 : 1;
+                tsum += flags[j];
+            }
+            uint32_t total0;
+            uint32_t zeros_before = fuse_scan(tsum, wave_tot, &total0);
+            for (int j = 0; j < FUSE_ITEMS; j++) {
+                int i = tid * FUSE_ITEMS + j;
+                uint32_t e = src[i];
+                uint32_t pos;
+                if (flags[j]) {
+                    pos = zeros_before++;
+                } else {
+                    pos = total0 + (uint32_t)i - zeros_before;
+                }
+                dst[pos] = e;
+            }
+            __syncthreads();
+            uint32_t *t = src; src = dst; dst = t;
+        }
+    }
+
+    // ---- consolidate: head flags over sorted permutation ----
+    // seg index via scan; weights accumulated into tw (global scratch)
+    uint32_t head[FUSE_ITEMS];
+    uint32_t tsum = 0;
+    for (int j = 0; j < FUSE_ITEMS; j++) {
+        int i = tid * FUSE_ITEMS + j;
+        uint32_t h = 0;
+        if (i < n) {
+            uint32_t idx = src[i] & 0x3FFFu;
+            if (i == 0) h = 1;
+            else {
+                uint32_t pidx = src[i - 1] & 0x3FFFu;
+                h = (kin[idx] != kin[pidx]) || (vin[idx] != vin[pidx]);
+            }
+        }
+        head[j] = h;
+        tsum += h;
+    }
+    uint32_t nseg;
+    uint32_t heads_before = fuse_scan(tsum, wave_tot, &nseg);
+    // store seg ids in dst (reuse) and zero the weight scratch
+    {
+        uint32_t hb = heads_before;
+        for (int j = 0; j < FUSE_ITEMS; j++) {
+            int i = tid * FUSE_ITEMS + j;
+            hb += head[j];
+            if (i < n) dst[i] = hb - 1;  // segment of element i
+        }
+    }
+    for (uint32_t i = tid; i < nseg; i += FUSE_THREADS) tw[i] = 0;
+    __syncthreads();
+    for (int j = 0; j < FUSE_ITEMS; j++) {
+        int i = tid * FUSE_ITEMS + j;
+        if (i < n) {
+            uint32_t idx = src[i] & 0x3FFFu;
+            uint32_t seg = dst[i];
+            atomicAdd((unsigned long long *)&tw[seg], (unsigned long long)win[idx]);
+            if (head[j]) {
+                tk[seg] = kin[idx];
+                tv[seg] = vin[idx];
+            }
+        }
+    }
+    __syncthreads();
+    // ---- drop zero-weight segments (compact into output) ----
+    uint32_t nz[FUSE_ITEMS];
+    uint32_t zsum = 0;
+    for (int j = 0; j < FUSE_ITEMS; j++) {
+        int i = tid * FUSE_ITEMS + j;
+        nz[j] = (i < (int)nseg && tw[i] != 0) ? 1 : 0;
+        zsum += nz[j];
+    }
+    uint32_t nout;
+    uint32_t before = fuse_scan(zsum, wave_tot, &nout);
+    for (int j = 0; j < FUSE_ITEMS; j++) {
+        int i = tid * FUSE_ITEMS + j;
+        if (nz[j]) {
+            uint32_t p = before++;
+            ok[p] = tk[i];
+            ov[p] = tv[i];
+            ow[p] = tw[i];
+        }
+    }
+    if (tid == 0) *out_len = (int64_t)nout;
+}
+
+// ---------------------------------------------------------------------------
 // consolidate sorted rows: head flags -> scan -> segment-sum -> nonzero compact
 // ---------------------------------------------------------------------------
 
@@ -568,6 +761,77 @@ __global__ void k_join_emit(const uint64_t *dk, const uint64_t *dv,
         ok[o] = hi_o;
         ov[o] = lo_o;
         ow[o] = dw[i] * tw[t];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// multi-batch spine join: the reference reads a spine through a k-way
+// CursorList (trace/cursor/cursor_list.rs); join is linear in the trace, so
+// one kernel probes every spine batch (descriptors passed by value as kernel
+// arguments) — one count+emit pair per tick per side regardless of spine depth.
+// ---------------------------------------------------------------------------
+
+__device__ inline int64_t gallop_run(const uint64_t *tk, int64_t nt, uint64_t key,
+                                     int64_t lo) {
+    // length of the key's run starting at lo (gallop + bounded binary search,
+    // the advance.rs:11-60 idiom)
+    if (lo >= nt || tk[lo] != key) return 0;
+    int64_t hi = lo, step = 1;
+    while (hi + step < nt && tk[hi + step] == key) { hi += step; step <<= 1; }
+    int64_t rem = min(step, nt - hi);
+    // upper bound within (hi, hi+rem)
+    int64_t a = 0, b = rem;
+    while (a < b) {
+        int64_t m = (a + b) / 2;
+        if (tk[hi + m] <= key) a = m + 1; else b = m;
+    }
+    return hi + a - lo;
+}
+
+__global__ void k_join_count_multi(const uint64_t *dk, int64_t nd, TraceArgs t,
+                                   uint32_t *cnts /* nd*nb */,
+                                   uint64_t *counts /* nd */) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nd;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t key = dk[i];
+        uint64_t total = 0;
+        for (int b = 0; b < t.nb; b++) {
+            int64_t lo = lower_bound_k(t.k[b], t.n[b], key);
+            int64_t c = gallop_run(t.k[b], t.n[b], key, lo);
+            cnts[i * t.nb + b] = (uint32_t)c;
+            total += c;
+        }
+        counts[i] = total;
+    }
+}
+
+__global__ void k_join_emit_multi(const uint64_t *dk, const uint64_t *dv,
+                                  const int64_t *dw, int64_t nd, TraceArgs t,
+                                  const uint32_t *cnts, const uint64_t *offsets,
+                                  int64_t n_out, int proj, uint64_t param,
+                                  uint64_t *ok, uint64_t *ov, int64_t *ow) {
+    for (int64_t o = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; o < n_out;
+         o += (int64_t)gridDim.x * blockDim.x) {
+        int64_t lo = 0, hi = nd;
+        while (lo < hi) {
+            int64_t mid = (lo + hi) / 2;
+            if (offsets[mid] <= (uint64_t)o) lo = mid + 1; else hi = mid;
+        }
+        int64_t i = lo - 1;
+        int64_t j = o - (int64_t)offsets[i];
+        int b = 0;
+        while (j >= (int64_t)cnts[i * t.nb + b]) {
+            j -= cnts[i * t.nb + b];
+            b++;
+        }
+        uint64_t key = dk[i];
+        int64_t start = lower_bound_k(t.k[b], t.n[b], key);
+        int64_t ti = start + j;
+        uint64_t hi_o, lo_o;
+        proj_out(proj, param, key, dv[i], t.v[b][ti], hi_o, lo_o);
+        ok[o] = hi_o;
+        ov[o] = lo_o;
+        ow[o] = dw[i] * t.w[b][ti];
     }
 }
 
@@ -957,6 +1221,48 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
     k_merge_pass<true><<<dim3((uint32_t)nblocks), BLK, 0, s>>>(
         ak, av, aw, na, bk, bv, bw, nb, tcounts, rk, rv, rw);
     HIP_CHECK(hipFreeAsync(tcounts, s));
+    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
+    return DBSP_OK;
+}
+
+dbsp_status sort_cons_small(hipStream_t s, const uint64_t *kin,
+                            const uint64_t *vin, const int64_t *win, int64_t n,
+                            uint64_t *tk, uint64_t *tv, int64_t *tw,
+                            uint64_t *ok, uint64_t *ov, int64_t *ow,
+                            int64_t *d_len) {
+    if (n > FUSE_MAX) return DBSP_ERR_INVALID;
+    k_sort_cons_small<<<1, FUSE_THREADS, 0, s>>>(kin, vin, win, n, tk, tv, tw,
+                                                 ok, ov, ow, d_len);
+    return DBSP_OK;
+}
+
+dbsp_status join_spine_rows(hipStream_t s, const uint64_t *dk,
+                            const uint64_t *dv, const int64_t *dw, int64_t nd,
+                            const TraceArgs &t, int proj, uint64_t param,
+                            uint64_t **ok, uint64_t **ov, int64_t **ow,
+                            int64_t *out_n) {
+    if (nd == 0 || t.nb == 0) {
+        *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
+        return DBSP_OK;
+    }
+    uint32_t *cnts;
+    uint64_t *counts;
+    HIP_CHECK(hipMallocAsync(&cnts, (int64_t)nd * t.nb * sizeof(uint32_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&counts, (nd + 1) * sizeof(uint64_t), s));
+    k_join_count_multi<<<grid_for(nd), BLK, 0, s>>>(dk, nd, t, cnts, counts);
+    uint64_t nout = 0;
+    dbsp_status st = scan_exclusive(s, counts, counts, nd, &nout);
+    if (st != DBSP_OK) return st;
+    uint64_t *rk, *rv; int64_t *rw;
+    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    if (nout > 0)
+        k_join_emit_multi<<<grid_for((int64_t)nout), BLK, 0, s>>>(
+            dk, dv, dw, nd, t, cnts, counts, (int64_t)nout, proj, param, rk, rv,
+            rw);
+    HIP_CHECK(hipFreeAsync(cnts, s));
+    HIP_CHECK(hipFreeAsync(counts, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
     return DBSP_OK;
 }

@@ -5,6 +5,16 @@
 #include <cstdint>
 #include "../../include/dbsp_hip.h"
 
+// spine-as-kernel-argument descriptor (multi-batch join/probe kernels)
+#define MAX_TRACE_BATCHES 24
+struct TraceArgs {
+    int nb;
+    const uint64_t *k[MAX_TRACE_BATCHES];
+    const uint64_t *v[MAX_TRACE_BATCHES];
+    const int64_t *w[MAX_TRACE_BATCHES];
+    int64_t n[MAX_TRACE_BATCHES];
+};
+
 namespace dbspk {
 
 dbsp_status scan_excl(hipStream_t s, const uint64_t *in, uint64_t *out,
@@ -25,6 +35,21 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
                        const uint64_t *bv, const int64_t *bw, int64_t nb,
                        uint64_t **ok, uint64_t **ov, int64_t **ow,
                        int64_t *out_n);
+
+// fused single-workgroup sort+consolidate (n <= 16384): one launch, length
+// left in *d_len (device)
+dbsp_status sort_cons_small(hipStream_t s, const uint64_t *kin,
+                            const uint64_t *vin, const int64_t *win, int64_t n,
+                            uint64_t *tk, uint64_t *tv, int64_t *tw,
+                            uint64_t *ok, uint64_t *ov, int64_t *ow,
+                            int64_t *d_len);
+
+// join delta against a whole spine (TraceArgs) in one count/emit pair
+dbsp_status join_spine_rows(hipStream_t s, const uint64_t *dk,
+                            const uint64_t *dv, const int64_t *dw, int64_t nd,
+                            const TraceArgs &t, int proj, uint64_t param,
+                            uint64_t **ok, uint64_t **ov, int64_t **ow,
+                            int64_t *out_n);
 
 dbsp_status join_rows(hipStream_t s, const uint64_t *dk, const uint64_t *dv,
                       const int64_t *dw, int64_t nd, const uint64_t *tk,
