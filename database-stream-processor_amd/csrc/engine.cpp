@@ -212,7 +212,7 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
     ScopedTimer t(c, 0, (double)raw.n * 48.0);
     DevBatch scratch;
     TRY(alloc_batch(c, raw.n, scratch));
-    if (raw.n <= 16384) {
+    if (raw.n <= 8192) {
         // fused single-workgroup path: one launch + one length readback
         DevBatch res;
         TRY(alloc_batch(c, raw.n, res));

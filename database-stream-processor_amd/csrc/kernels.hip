@@ -345,9 +345,10 @@ __global__ void k_radix_scatter(const uint64_t *k_in, const uint64_t *v_in,
 // permutation (digit<<14|idx u32) ping-pongs between two 64 KiB LDS arrays.
 // ---------------------------------------------------------------------------
 
-#define FUSE_MAX 16384
+#define FUSE_MAX 8192
 #define FUSE_THREADS 1024
-#define FUSE_ITEMS (FUSE_MAX / FUSE_THREADS)  // 16
+#define FUSE_ITEMS (FUSE_MAX / FUSE_THREADS)  // 8
+#define FUSE_DIGITS 16                        // 4-bit digits, ranked per pass
 
 // block-wide exclusive scan over FUSE_MAX u32 flags held per-thread
 // (16 consecutive items per thread); returns thread's exclusive offset and
@@ -385,6 +386,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(
     int64_t *out_len) {
     __shared__ uint32_t bufA[FUSE_MAX];
     __shared__ uint32_t bufB[FUSE_MAX];
+    __shared__ uint32_t cnt[FUSE_DIGITS * FUSE_THREADS];  // rank counters
     __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
     __shared__ uint64_t smax[2];
     const int tid = threadIdx.x;
@@ -394,7 +396,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(
         return;
     }
 
-    // ---- significant bytes of max(k), max(v) ----
+    // ---- significant bits of max(k), max(v) ----
     if (tid == 0) { smax[0] = 0; smax[1] = 0; }
     __syncthreads();
     {
@@ -407,56 +409,66 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(
         atomicMax((unsigned long long *)&smax[1], (unsigned long long)mv);
     }
     __syncthreads();
-    int kbytes = 0, vbytes = 0;
-    while (kbytes < 8 && (smax[0] >> (8 * kbytes)) != 0) kbytes++;
-    while (vbytes < 8 && (smax[1] >> (8 * vbytes)) != 0) vbytes++;
+    int knibs = 0, vnibs = 0;
+    while (knibs < 16 && (smax[0] >> (4 * knibs)) != 0) knibs++;
+    while (vnibs < 16 && (smax[1] >> (4 * vnibs)) != 0) vnibs++;
 
-    // ---- init permutation ----
+    // ---- init permutation (entry = digit<<13 | idx; idx < 8192) ----
     for (int i = tid; i < FUSE_MAX; i += FUSE_THREADS)
-        bufA[i] = (uint32_t)i | (i >= n ? (0xFFu << 18) : 0);  // pads sort last
+        bufA[i] = (uint32_t)i;
     __syncthreads();
 
+    // LSD over 4-bit digits: v nibbles (minor) then k nibbles (major).
+    // Stable per pass via per-(digit,thread) rank counters: counter cell
+    // cnt[d*THREADS + tid] is exclusive to its thread, so count, scan the
+    // digit-major flat array block-wide, then post-increment to rank.
     uint32_t *src = bufA, *dst = bufB;
-    const int total_bytes = vbytes + kbytes;
-    for (int pass = 0; pass < total_bytes; pass++) {
-        const bool is_v = pass < vbytes;
-        const int sh = 8 * (is_v ? pass : pass - vbytes);
-        // annotate digits: entry = digit<<18 | idx (idx < 16384 fits 14 bits)
-        for (int i = tid; i < FUSE_MAX; i += FUSE_THREADS) {
-            uint32_t idx = src[i] & 0x3FFFu;
-            uint32_t d = 0xFFu;  // pads (idx >= n) keep the max digit
+    const int total_nibs = vnibs + knibs;
+    for (int pass = 0; pass < total_nibs; pass++) {
+        const bool is_v = pass < vnibs;
+        const int sh = 4 * (is_v ? pass : pass - vnibs);
+        // zero own counter column (no barrier needed: cells are per-thread)
+        for (int d = 0; d < FUSE_DIGITS; d++) cnt[d * FUSE_THREADS + tid] = 0;
+        // annotate digit + count (pads: digit 15, stable-after by idx order)
+        for (int j = 0; j < FUSE_ITEMS; j++) {
+            int i = tid * FUSE_ITEMS + j;
+            uint32_t idx = src[i] & 0x1FFFu;
+            uint32_t d = 15;
             if ((int64_t)idx < n) {
                 uint64_t limb = is_v ? vin[idx] : kin[idx];
-                d = (uint32_t)((limb >> sh) & 0xFF);
+                d = (uint32_t)((limb >> sh) & 0xF);
             }
-            src[i] = (d << 18) | idx;
+            src[i] = (d << 13) | idx;
+            cnt[d * FUSE_THREADS + tid]++;
         }
         __syncthreads();
-        // 8 stable 1-bit splits
-        for (int bit = 0; bit < 8; bit++) {
-            uint32_t flags[FUSE_ITEMS];
+        // exclusive scan of the flat counter array (16 cells per thread)
+        {
+            uint32_t local[FUSE_DIGITS];
             uint32_t tsum = 0;
-            for (int j = 0; j < FUSE_ITEMS; j++) {
-                int i = tid * FUSE_ITEMS + j;
-                flags[j] = ((src[i] >> (18 + bit)) & 1) ? 0 : 1;
-                tsum += flags[j];
+            for (int j = 0; j < FUSE_DIGITS; j++) {
+                local[j] = cnt[tid * FUSE_DIGITS + j];
+                tsum += local[j];
             }
-            uint32_t total0;
-            uint32_t zeros_before = fuse_scan(tsum, wave_tot, &total0);
-            for (int j = 0; j < FUSE_ITEMS; j++) {
-                int i = tid * FUSE_ITEMS + j;
-                uint32_t e = src[i];
-                uint32_t pos;
-                if (flags[j]) {
-                    pos = zeros_before++;
-                } else {
-                    pos = total0 + (uint32_t)i - zeros_before;
-                }
-                dst[pos] = e;
+            uint32_t total_unused;
+            uint32_t off = fuse_scan(tsum, wave_tot, &total_unused);
+            for (int j = 0; j < FUSE_DIGITS; j++) {
+                uint32_t c = local[j];
+                cnt[tid * FUSE_DIGITS + j] = off;
+                off += c;
             }
-            __syncthreads();
-            uint32_t *t = src; src = dst; dst = t;
         }
+        __syncthreads();
+        // rank (post-increment own cells in item order) + scatter
+        for (int j = 0; j < FUSE_ITEMS; j++) {
+            int i = tid * FUSE_ITEMS + j;
+            uint32_t e = src[i];
+            uint32_t d = e >> 13;
+            uint32_t r = cnt[d * FUSE_THREADS + tid]++;
+            dst[r] = e;
+        }
+        __syncthreads();
+        uint32_t *t = src; src = dst; dst = t;
     }
 
     // ---- consolidate: head flags over sorted permutation ----
@@ -467,10 +479,10 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(
         int i = tid * FUSE_ITEMS + j;
         uint32_t h = 0;
         if (i < n) {
-            uint32_t idx = src[i] & 0x3FFFu;
+            uint32_t idx = src[i] & 0x1FFFu;
             if (i == 0) h = 1;
             else {
-                uint32_t pidx = src[i - 1] & 0x3FFFu;
+                uint32_t pidx = src[i - 1] & 0x1FFFu;
                 h = (kin[idx] != kin[pidx]) || (vin[idx] != vin[pidx]);
             }
         }
@@ -493,7 +505,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(
     for (int j = 0; j < FUSE_ITEMS; j++) {
         int i = tid * FUSE_ITEMS + j;
         if (i < n) {
-            uint32_t idx = src[i] & 0x3FFFu;
+            uint32_t idx = src[i] & 0x1FFFu;
             uint32_t seg = dst[i];
             atomicAdd((unsigned long long *)&tw[seg], (unsigned long long)win[idx]);
             if (head[j]) {
