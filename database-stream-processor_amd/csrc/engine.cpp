@@ -88,7 +88,21 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     }
     dbsp_ctx *c = new dbsp_ctx();
     c->device = device;
+    // spin-wait syncs: the tick pipeline syncs on data-dependent lengths a few
+    // times per tick; yield-based waits cost 10-20 us each
+    (void)hipSetDeviceFlags(hipDeviceScheduleSpin);
     HIP_CHECK_ST(hipSetDevice(device));
+    // keep freed stream-ordered allocations in the pool (the tick pipeline
+    // allocates/frees ~20 buffers per tick; without this the pool trims back
+    // to the OS between ticks)
+    {
+        hipMemPool_t mp;
+        if (hipDeviceGetDefaultMemPool(&mp, device) == hipSuccess) {
+            uint64_t thresh = UINT64_MAX;
+            (void)hipMemPoolSetAttribute(mp, hipMemPoolAttrReleaseThreshold,
+                                         &thresh);
+        }
+    }
     HIP_CHECK_ST(hipStreamCreate(&c->stream));
     HIP_CHECK_ST(hipEventCreate(&c->ev0));
     HIP_CHECK_ST(hipEventCreate(&c->ev1));
@@ -241,6 +255,13 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
 
 static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
                                  const DevBatch &b, DevBatch &out) {
+    if (a.n + b.n <= 8192) {
+        // small merges (tops of spines) go through the fused one-launch
+        // sort+consolidate instead of the count/scan/emit merge-path pipeline
+        DevBatch cat;
+        TRY(concat_batches(c, {a, b}, cat));
+        return sort_consolidate_batch(c, cat, out);
+    }
     ScopedTimer t(c, 1, (double)(a.n + b.n) * 48.0);
     TRY(dbspk::merge_rows(c->stream, a.k, a.v, a.w, a.n, b.k, b.v, b.w, b.n,
                           &out.k, &out.v, &out.w, &out.n));
@@ -705,6 +726,11 @@ static dbsp_status join_vs_spine(dbsp_ctx *c, const DevBatch &delta,
 // consolidate a list of raw result batches into one batch
 static dbsp_status finalize_raw(dbsp_ctx *c, std::vector<DevBatch> &outs,
                                 DevBatch &out) {
+    if (outs.size() == 1) {
+        DevBatch only = outs[0];
+        outs.clear();
+        return sort_consolidate_batch(c, only, out);
+    }
     DevBatch cat;
     TRY(concat_batches(c, outs, cat));
     for (auto &b : outs) free_batch(c, b);
