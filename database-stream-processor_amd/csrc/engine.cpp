@@ -108,8 +108,8 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     HIP_CHECK_ST(hipEventCreate(&c->ev1));
     const char *p = getenv("DBSP_PROFILE");
     c->profile = p && p[0] == '1';
-    HIP_CHECK_ST(hipMalloc(&c->d_len, sizeof(int64_t)));
-    HIP_CHECK_ST(hipHostMalloc(&c->h_len, sizeof(int64_t)));
+    HIP_CHECK_ST(hipMalloc(&c->d_len, 8 * sizeof(int64_t)));
+    HIP_CHECK_ST(hipHostMalloc(&c->h_len, 8 * sizeof(int64_t)));
     *out = c;
     return DBSP_OK;
 }
@@ -230,9 +230,13 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
         // fused single-workgroup path: one launch + one length readback
         DevBatch res;
         TRY(alloc_batch(c, raw.n, res));
-        TRY(dbspk::sort_cons_small(c->stream, raw.k, raw.v, raw.w, raw.n,
-                                   scratch.k, scratch.v, scratch.w, res.k,
-                                   res.v, res.w, c->d_len));
+        SortArgs sa{};
+        sa.nb = 1;
+        sa.kin[0] = raw.k; sa.vin[0] = raw.v; sa.win[0] = raw.w; sa.n[0] = raw.n;
+        sa.tk[0] = scratch.k; sa.tv[0] = scratch.v; sa.tw[0] = scratch.w;
+        sa.ok[0] = res.k; sa.ov[0] = res.v; sa.ow[0] = res.w;
+        sa.d_len = c->d_len;
+        TRY(dbspk::sort_cons_small_batch(c->stream, sa));
         HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -255,14 +259,20 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
 
 static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
                                  const DevBatch &b, DevBatch &out) {
-    if (a.n + b.n <= 8192) {
-        // small merges (tops of spines) go through the fused one-launch
-        // sort+consolidate instead of the count/scan/emit merge-path pipeline
-        DevBatch cat;
-        TRY(concat_batches(c, {a, b}, cat));
-        return sort_consolidate_batch(c, cat, out);
-    }
     ScopedTimer t(c, 1, (double)(a.n + b.n) * 48.0);
+    if (a.n + b.n <= 8192) {
+        // one launch + one length readback (merge-path diagonals in one WG)
+        DevBatch res;
+        TRY(alloc_batch(c, a.n + b.n, res));
+        TRY(dbspk::merge_small(c->stream, a.k, a.v, a.w, a.n, b.k, b.v, b.w,
+                               b.n, res.k, res.v, res.w, c->d_len));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, sizeof(int64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        res.n = *c->h_len;
+        out = res;
+        return DBSP_OK;
+    }
     TRY(dbspk::merge_rows(c->stream, a.k, a.v, a.w, a.n, b.k, b.v, b.w, b.n,
                           &out.k, &out.v, &out.w, &out.n));
     return DBSP_OK;
@@ -664,6 +674,40 @@ static void q0_step_host(dbsp_engine *e, const dbsp_event *ev, int64_t n) {
 
 // ---- shared helpers ----
 
+// sort+consolidate two independent small raw batches in ONE launch
+// (consumes both raws)
+static dbsp_status sort_two_small(dbsp_ctx *c, DevBatch rawA, DevBatch rawB,
+                                  DevBatch &dA, DevBatch &dB) {
+    ScopedTimer t(c, 0, (double)(rawA.n + rawB.n) * 48.0);
+    DevBatch sA, sB, oA, oB;
+    TRY(alloc_batch(c, rawA.n > 0 ? rawA.n : 1, sA));
+    TRY(alloc_batch(c, rawB.n > 0 ? rawB.n : 1, sB));
+    TRY(alloc_batch(c, rawA.n > 0 ? rawA.n : 1, oA));
+    TRY(alloc_batch(c, rawB.n > 0 ? rawB.n : 1, oB));
+    SortArgs sa{};
+    sa.nb = 2;
+    sa.kin[0] = rawA.k; sa.vin[0] = rawA.v; sa.win[0] = rawA.w; sa.n[0] = rawA.n;
+    sa.kin[1] = rawB.k; sa.vin[1] = rawB.v; sa.win[1] = rawB.w; sa.n[1] = rawB.n;
+    sa.tk[0] = sA.k; sa.tv[0] = sA.v; sa.tw[0] = sA.w;
+    sa.tk[1] = sB.k; sa.tv[1] = sB.v; sa.tw[1] = sB.w;
+    sa.ok[0] = oA.k; sa.ov[0] = oA.v; sa.ow[0] = oA.w;
+    sa.ok[1] = oB.k; sa.ov[1] = oB.v; sa.ow[1] = oB.w;
+    sa.d_len = c->d_len;
+    TRY(dbspk::sort_cons_small_batch(c->stream, sa));
+    HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 2 * sizeof(int64_t),
+                                hipMemcpyDeviceToHost, c->stream));
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    oA.n = c->h_len[0];
+    oB.n = c->h_len[1];
+    free_batch(c, rawA);
+    free_batch(c, rawB);
+    free_batch(c, sA);
+    free_batch(c, sB);
+    dA = oA;
+    dB = oB;
+    return DBSP_OK;
+}
+
 // flatmap events slice into up to two raw streams, then sort+consolidate
 static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
                                 int64_t n, DevBatch &d0, DevBatch &d1,
@@ -677,11 +721,15 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
                               raw0.w, &n0, raw1.k, raw1.v, raw1.w, &n1));
     raw0.n = n0;
     raw1.n = n1;
-    TRY(sort_consolidate_batch(c, raw0, d0));
-    if (want_two)
-        TRY(sort_consolidate_batch(c, raw1, d1));
-    else
-        free_batch(c, raw1);
+    if (want_two && n0 <= 8192 && n1 <= 8192) {
+        TRY(sort_two_small(c, raw0, raw1, d0, d1));
+    } else {
+        TRY(sort_consolidate_batch(c, raw0, d0));
+        if (want_two)
+            TRY(sort_consolidate_batch(c, raw1, d1));
+        else
+            free_batch(c, raw1);
+    }
     // worker sharding: co-locate keys across ranks (shard.rs:88)
     if (e->world > 1) {
         DevBatch s0;
@@ -855,12 +903,27 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     dbsp_ctx *c = e->ctx;
     DevBatch dA, dP;
     TRY(build_deltas(e, d_ev, n, dA, dP, true));
+    // Bilinear expansion against PREVIOUS traces only, so all three joins are
+    // independent of the spine inserts:
+    //   out = dA join P_prev + A_prev join dP + dA join dP
+    // — identical to the reference's dA join trace(P) + trace(A) delayed join dP
+    // (operator/join.rs:217-292) by bilinearity.
     std::vector<DevBatch> outs;
-    // dA ⋈ P_{t-1}
     TRY(join_vs_spine(c, dA, e->p_int, DBSP_PROJ_HI_V2_LO_V1, 0, outs));
-    TRY(e->a_int.insert(c, dA));
-    // A_t ⋈ dP
     TRY(join_vs_spine(c, dP, e->a_int, DBSP_PROJ_HI_V1_LO_V2, 0, outs));
+    if (dA.n > 0 && dP.n > 0) {
+        TraceArgs t{};
+        t.nb = 1;
+        t.k[0] = dP.k; t.v[0] = dP.v; t.w[0] = dP.w; t.n[0] = dP.n;
+        DevBatch o;
+        ScopedTimer timer(c, 2, (double)dA.n * 24.0);
+        TRY(dbspk::join_spine_rows(c->stream, dA.k, dA.v, dA.w, dA.n, t,
+                                   DBSP_PROJ_HI_V2_LO_V1, 0, &o.k, &o.v, &o.w,
+                                   &o.n));
+        if (o.n > 0) outs.push_back(o);
+        else free_batch(c, o);
+    }
+    TRY(e->a_int.insert(c, dA));
     TRY(e->p_int.insert(c, dP));
     free_batch(c, e->output);
     TRY(finalize_raw(c, outs, e->output));

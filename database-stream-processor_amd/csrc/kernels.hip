@@ -379,11 +379,19 @@ __device__ inline uint32_t fuse_scan(uint32_t thread_sum, uint32_t *wave_tot,
     return r;
 }
 
-__global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(
-    const uint64_t *kin, const uint64_t *vin, const int64_t *win, int64_t n,
-    uint64_t *tk, uint64_t *tv, int64_t *tw,  // scratch (>= n rows)
-    uint64_t *ok, uint64_t *ov, int64_t *ow,  // output (cap >= n rows)
-    int64_t *out_len) {
+__global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs args) {
+    const int batch = blockIdx.x;
+    const uint64_t *kin = args.kin[batch];
+    const uint64_t *vin = args.vin[batch];
+    const int64_t *win = args.win[batch];
+    const int64_t n = args.n[batch];
+    uint64_t *tk = args.tk[batch];   // scratch (>= n rows)
+    uint64_t *tv = args.tv[batch];
+    int64_t *tw = args.tw[batch];
+    uint64_t *ok = args.ok[batch];   // output (cap >= n rows)
+    uint64_t *ov = args.ov[batch];
+    int64_t *ow = args.ow[batch];
+    int64_t *out_len = args.d_len + batch;
     __shared__ uint32_t bufA[FUSE_MAX];
     __shared__ uint32_t bufB[FUSE_MAX];
     __shared__ uint32_t cnt[FUSE_DIGITS * FUSE_THREADS];  // rank counters
@@ -774,6 +782,55 @@ __global__ void k_join_emit(const uint64_t *dk, const uint64_t *dv,
         ov[o] = lo_o;
         ow[o] = dw[i] * tw[t];
     }
+}
+
+// single-workgroup merge for small inputs (na+nb <= FUSE_MAX): merge-path
+// thread diagonals + LDS scan; one launch, no host sync, length on device
+__global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_small(
+    const uint64_t *ak, const uint64_t *av, const int64_t *aw, int64_t na,
+    const uint64_t *bk, const uint64_t *bv, const int64_t *bw, int64_t nb,
+    uint64_t *ok, uint64_t *ov, int64_t *ow, int64_t *out_len) {
+    __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
+    const int tid = threadIdx.x;
+    const int64_t total = na + nb;
+    int64_t d0 = min((int64_t)tid * FUSE_ITEMS, total);
+    int64_t d1 = min(d0 + FUSE_ITEMS, total);
+    int64_t ai, bi, ae, be;
+    merge_path(ak, av, na, bk, bv, nb, d0, ai, bi);
+    adjust_split(ak, av, bk, bv, na, nb, ai, bi);
+    merge_path(ak, av, na, bk, bv, nb, d1, ae, be);
+    adjust_split(ak, av, bk, bv, na, nb, ae, be);
+    uint32_t cnt = 0;
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(ak[i], av[i], bk[j], bv[j])) {
+                if (aw[i] + bw[j] != 0) cnt++;
+                i++; j++;
+            } else if (j >= be || (i < ae && row_lt(ak[i], av[i], bk[j], bv[j]))) {
+                cnt++; i++;
+            } else {
+                cnt++; j++;
+            }
+        }
+    }
+    uint32_t tot_out;
+    uint32_t off = fuse_scan(cnt, wave_tot, &tot_out);
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(ak[i], av[i], bk[j], bv[j])) {
+                int64_t s = aw[i] + bw[j];
+                if (s != 0) { ok[off] = ak[i]; ov[off] = av[i]; ow[off] = s; off++; }
+                i++; j++;
+            } else if (j >= be || (i < ae && row_lt(ak[i], av[i], bk[j], bv[j]))) {
+                ok[off] = ak[i]; ov[off] = av[i]; ow[off] = aw[i]; off++; i++;
+            } else {
+                ok[off] = bk[j]; ov[off] = bv[j]; ow[off] = bw[j]; off++; j++;
+            }
+        }
+    }
+    if (tid == 0) *out_len = (int64_t)tot_out;
 }
 
 // ---------------------------------------------------------------------------
@@ -1237,14 +1294,22 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
     return DBSP_OK;
 }
 
-dbsp_status sort_cons_small(hipStream_t s, const uint64_t *kin,
-                            const uint64_t *vin, const int64_t *win, int64_t n,
-                            uint64_t *tk, uint64_t *tv, int64_t *tw,
-                            uint64_t *ok, uint64_t *ov, int64_t *ow,
-                            int64_t *d_len) {
-    if (n > FUSE_MAX) return DBSP_ERR_INVALID;
-    k_sort_cons_small<<<1, FUSE_THREADS, 0, s>>>(kin, vin, win, n, tk, tv, tw,
-                                                 ok, ov, ow, d_len);
+dbsp_status sort_cons_small_batch(hipStream_t s, const SortArgs &args) {
+    for (int b = 0; b < args.nb; b++)
+        if (args.n[b] > FUSE_MAX) return DBSP_ERR_INVALID;
+    if (args.nb == 0) return DBSP_OK;
+    k_sort_cons_small<<<dim3((uint32_t)args.nb), FUSE_THREADS, 0, s>>>(args);
+    return DBSP_OK;
+}
+
+dbsp_status merge_small(hipStream_t s, const uint64_t *ak, const uint64_t *av,
+                        const int64_t *aw, int64_t na, const uint64_t *bk,
+                        const uint64_t *bv, const int64_t *bw, int64_t nb,
+                        uint64_t *ok, uint64_t *ov, int64_t *ow,
+                        int64_t *d_len) {
+    if (na + nb > FUSE_MAX) return DBSP_ERR_INVALID;
+    k_merge_small<<<1, FUSE_THREADS, 0, s>>>(ak, av, aw, na, bk, bv, bw, nb, ok,
+                                             ov, ow, d_len);
     return DBSP_OK;
 }
 

@@ -15,6 +15,21 @@ struct TraceArgs {
     int64_t n[MAX_TRACE_BATCHES];
 };
 
+struct SortArgs {
+    int nb;
+    const uint64_t *kin[4];
+    const uint64_t *vin[4];
+    const int64_t *win[4];
+    int64_t n[4];
+    uint64_t *tk[4];
+    uint64_t *tv[4];
+    int64_t *tw[4];
+    uint64_t *ok[4];
+    uint64_t *ov[4];
+    int64_t *ow[4];
+    int64_t *d_len;  // device array, one length per batch
+};
+
 namespace dbspk {
 
 dbsp_status scan_excl(hipStream_t s, const uint64_t *in, uint64_t *out,
@@ -36,13 +51,18 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
                        uint64_t **ok, uint64_t **ov, int64_t **ow,
                        int64_t *out_n);
 
-// fused single-workgroup sort+consolidate (n <= 16384): one launch, length
-// left in *d_len (device)
-dbsp_status sort_cons_small(hipStream_t s, const uint64_t *kin,
-                            const uint64_t *vin, const int64_t *win, int64_t n,
-                            uint64_t *tk, uint64_t *tv, int64_t *tw,
-                            uint64_t *ok, uint64_t *ov, int64_t *ow,
-                            int64_t *d_len);
+// fused single-workgroup sort+consolidate batch descriptors: up to 4
+// independent small (n <= 8192) raw batches sorted+consolidated concurrently,
+// one workgroup each, lengths left in d_len[i] (device)
+dbsp_status sort_cons_small_batch(hipStream_t s, const SortArgs &args);
+
+// single-workgroup merge of two consolidated batches (na+nb <= 8192):
+// one launch, no host sync; length left in *d_len (device)
+dbsp_status merge_small(hipStream_t s, const uint64_t *ak, const uint64_t *av,
+                        const int64_t *aw, int64_t na, const uint64_t *bk,
+                        const uint64_t *bv, const int64_t *bw, int64_t nb,
+                        uint64_t *ok, uint64_t *ov, int64_t *ow,
+                        int64_t *d_len);
 
 // join delta against a whole spine (TraceArgs) in one count/emit pair
 dbsp_status join_spine_rows(hipStream_t s, const uint64_t *dk,

@@ -125,7 +125,8 @@ struct Generator {
             // generator/people.rs:50-79
             e.kind = 0;
             e.f0 = last_base0_person_id(event_id) + FIRST_PERSON_ID;
-            e.f1 = (uint32_t)rng.next();          // name id
+            e.f1 = rng.range(1000);  // name id (dictionary of ~1000 generated
+                                     // first+last combinations, people.rs:40-46)
             e.f2 = rng.range(NUM_US_CITIES);      // city id
             e.f3 = rng.range(NUM_US_STATES);      // state id
             e.f4 = ts;
