@@ -421,59 +421,73 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     while (knibs < 16 && (smax[0] >> (4 * knibs)) != 0) knibs++;
     while (vnibs < 16 && (smax[1] >> (4 * vnibs)) != 0) vnibs++;
 
+    // adaptive sizing: only t_act threads carry items (wave-rounded), so the
+    // per-pass counter array is 16*t_act cells, not 16*1024 — small deltas
+    // (hundreds of rows) pay for their size, not for FUSE_MAX
+    const int t_act = (int)min((int64_t)FUSE_THREADS,
+                               ((n + FUSE_ITEMS - 1) / FUSE_ITEMS + WAVE - 1) /
+                                   WAVE * WAVE);
+    const int n_pad = t_act * FUSE_ITEMS;
+    const int scan_cells = FUSE_DIGITS * t_act;
+    const int cells_per_thread = (scan_cells + FUSE_THREADS - 1) / FUSE_THREADS;
+
     // ---- init permutation (entry = digit<<13 | idx; idx < 8192) ----
-    for (int i = tid; i < FUSE_MAX; i += FUSE_THREADS)
-        bufA[i] = (uint32_t)i;
+    for (int i = tid; i < n_pad; i += FUSE_THREADS) bufA[i] = (uint32_t)i;
     __syncthreads();
 
     // LSD over 4-bit digits: v nibbles (minor) then k nibbles (major).
     // Stable per pass via per-(digit,thread) rank counters: counter cell
-    // cnt[d*THREADS + tid] is exclusive to its thread, so count, scan the
+    // cnt[d*t_act + tid] is exclusive to its thread, so count, scan the
     // digit-major flat array block-wide, then post-increment to rank.
     uint32_t *src = bufA, *dst = bufB;
     const int total_nibs = vnibs + knibs;
     for (int pass = 0; pass < total_nibs; pass++) {
         const bool is_v = pass < vnibs;
         const int sh = 4 * (is_v ? pass : pass - vnibs);
-        // zero own counter column (no barrier needed: cells are per-thread)
-        for (int d = 0; d < FUSE_DIGITS; d++) cnt[d * FUSE_THREADS + tid] = 0;
-        // annotate digit + count (pads: digit 15, stable-after by idx order)
-        for (int j = 0; j < FUSE_ITEMS; j++) {
-            int i = tid * FUSE_ITEMS + j;
-            uint32_t idx = src[i] & 0x1FFFu;
-            uint32_t d = 15;
-            if ((int64_t)idx < n) {
-                uint64_t limb = is_v ? vin[idx] : kin[idx];
-                d = (uint32_t)((limb >> sh) & 0xF);
+        if (tid < t_act) {
+            // zero own counter column (no barrier: cells are per-thread)
+            for (int d = 0; d < FUSE_DIGITS; d++) cnt[d * t_act + tid] = 0;
+            // annotate digit + count (pads: digit 15, stable-after by idx order)
+            for (int j = 0; j < FUSE_ITEMS; j++) {
+                int i = tid * FUSE_ITEMS + j;
+                uint32_t idx = src[i] & 0x1FFFu;
+                uint32_t d = 15;
+                if ((int64_t)idx < n) {
+                    uint64_t limb = is_v ? vin[idx] : kin[idx];
+                    d = (uint32_t)((limb >> sh) & 0xF);
+                }
+                src[i] = (d << 13) | idx;
+                cnt[d * t_act + tid]++;
             }
-            src[i] = (d << 13) | idx;
-            cnt[d * FUSE_THREADS + tid]++;
         }
         __syncthreads();
-        // exclusive scan of the flat counter array (16 cells per thread)
+        // exclusive scan of the flat counter array
         {
             uint32_t local[FUSE_DIGITS];
             uint32_t tsum = 0;
-            for (int j = 0; j < FUSE_DIGITS; j++) {
-                local[j] = cnt[tid * FUSE_DIGITS + j];
+            for (int j = 0; j < cells_per_thread; j++) {
+                int f = tid * cells_per_thread + j;
+                local[j] = f < scan_cells ? cnt[f] : 0;
                 tsum += local[j];
             }
             uint32_t total_unused;
             uint32_t off = fuse_scan(tsum, wave_tot, &total_unused);
-            for (int j = 0; j < FUSE_DIGITS; j++) {
-                uint32_t c = local[j];
-                cnt[tid * FUSE_DIGITS + j] = off;
-                off += c;
+            for (int j = 0; j < cells_per_thread; j++) {
+                int f = tid * cells_per_thread + j;
+                if (f < scan_cells) cnt[f] = off;
+                off += local[j];
             }
         }
         __syncthreads();
         // rank (post-increment own cells in item order) + scatter
-        for (int j = 0; j < FUSE_ITEMS; j++) {
-            int i = tid * FUSE_ITEMS + j;
-            uint32_t e = src[i];
-            uint32_t d = e >> 13;
-            uint32_t r = cnt[d * FUSE_THREADS + tid]++;
-            dst[r] = e;
+        if (tid < t_act) {
+            for (int j = 0; j < FUSE_ITEMS; j++) {
+                int i = tid * FUSE_ITEMS + j;
+                uint32_t e = src[i];
+                uint32_t d = e >> 13;
+                uint32_t r = cnt[d * t_act + tid]++;
+                dst[r] = e;
+            }
         }
         __syncthreads();
         uint32_t *t = src; src = dst; dst = t;
