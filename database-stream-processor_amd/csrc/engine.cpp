@@ -74,7 +74,24 @@ struct dbsp_ctx {
     // persistent length scratch (device + pinned host)
     int64_t *d_len = nullptr;
     int64_t *h_len = nullptr;
+    // per-tick transient bump arena (reset at each engine tick; falls back to
+    // the stream-ordered pool when exhausted)
+    uint8_t *arena = nullptr;
+    size_t arena_sz = 0;
+    size_t arena_off = 0;
 };
+
+static void *arena_alloc(dbsp_ctx *c, size_t bytes) {
+    bytes = (bytes + 255) & ~(size_t)255;
+    if (!c->arena || c->arena_off + bytes > c->arena_sz) return nullptr;
+    void *p = c->arena + c->arena_off;
+    c->arena_off += bytes;
+    return p;
+}
+
+static inline bool in_arena(dbsp_ctx *c, const void *p) {
+    return c->arena && p >= c->arena && p < c->arena + c->arena_sz;
+}
 
 extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     int ndev = 0;
@@ -109,6 +126,11 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     const char *p = getenv("DBSP_PROFILE");
     c->profile = p && p[0] == '1';
     HIP_CHECK_ST(hipMalloc(&c->d_len, 8 * sizeof(int64_t)));
+    c->arena_sz = (size_t)512 << 20;
+    if (hipMalloc(&c->arena, c->arena_sz) != hipSuccess) {
+        c->arena = nullptr;
+        c->arena_sz = 0;
+    }
     HIP_CHECK_ST(hipHostMalloc(&c->h_len, 8 * sizeof(int64_t)));
     *out = c;
     return DBSP_OK;
@@ -119,6 +141,7 @@ extern "C" dbsp_status dbsp_ctx_destroy(dbsp_ctx *c) {
     if (c->comm) ncclCommDestroy(c->comm);
     hipStreamSynchronize(c->stream);
     if (c->d_len) (void)hipFree(c->d_len);
+    if (c->arena) (void)hipFree(c->arena);
     if (c->h_len) (void)hipHostFree(c->h_len);
     hipEventDestroy(c->ev0);
     hipEventDestroy(c->ev1);
@@ -184,14 +207,25 @@ struct DevBatch {
 };
 
 static void free_batch(dbsp_ctx *c, DevBatch &b) {
-    if (b.k) hipFreeAsync(b.k, c->stream);
-    if (b.v) hipFreeAsync(b.v, c->stream);
-    if (b.w) hipFreeAsync(b.w, c->stream);
+    if (b.k && !in_arena(c, b.k)) hipFreeAsync(b.k, c->stream);
+    if (b.v && !in_arena(c, b.v)) hipFreeAsync(b.v, c->stream);
+    if (b.w && !in_arena(c, b.w)) hipFreeAsync(b.w, c->stream);
     b = DevBatch{};
 }
 
-static dbsp_status alloc_batch(dbsp_ctx *c, int64_t n, DevBatch &b) {
+static dbsp_status alloc_batch(dbsp_ctx *c, int64_t n, DevBatch &b,
+                               bool transient = false) {
     b.n = n;
+    if (transient) {
+        size_t one = (size_t)(n * 8 + 8 + 255) & ~(size_t)255;
+        uint8_t *blk = (uint8_t *)arena_alloc(c, 3 * one);
+        if (blk) {
+            b.k = (uint64_t *)blk;
+            b.v = (uint64_t *)(blk + one);
+            b.w = (int64_t *)(blk + 2 * one);
+            return DBSP_OK;
+        }
+    }
     HIP_CHECK_ST(hipMallocAsync(&b.k, n * sizeof(uint64_t) + 8, c->stream));
     HIP_CHECK_ST(hipMallocAsync(&b.v, n * sizeof(uint64_t) + 8, c->stream));
     HIP_CHECK_ST(hipMallocAsync(&b.w, n * sizeof(int64_t) + 8, c->stream));
@@ -225,7 +259,7 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
     }
     ScopedTimer t(c, 0, (double)raw.n * 48.0);
     DevBatch scratch;
-    TRY(alloc_batch(c, raw.n, scratch));
+    TRY(alloc_batch(c, raw.n, scratch, true));
     if (raw.n <= 8192) {
         // fused single-workgroup path: one launch + one length readback
         DevBatch res;
@@ -680,8 +714,8 @@ static dbsp_status sort_two_small(dbsp_ctx *c, DevBatch rawA, DevBatch rawB,
                                   DevBatch &dA, DevBatch &dB) {
     ScopedTimer t(c, 0, (double)(rawA.n + rawB.n) * 48.0);
     DevBatch sA, sB, oA, oB;
-    TRY(alloc_batch(c, rawA.n > 0 ? rawA.n : 1, sA));
-    TRY(alloc_batch(c, rawB.n > 0 ? rawB.n : 1, sB));
+    TRY(alloc_batch(c, rawA.n > 0 ? rawA.n : 1, sA, true));
+    TRY(alloc_batch(c, rawB.n > 0 ? rawB.n : 1, sB, true));
     TRY(alloc_batch(c, rawA.n > 0 ? rawA.n : 1, oA));
     TRY(alloc_batch(c, rawB.n > 0 ? rawB.n : 1, oB));
     SortArgs sa{};
@@ -714,8 +748,8 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
                                 bool want_two) {
     dbsp_ctx *c = e->ctx;
     DevBatch raw0, raw1;
-    TRY(alloc_batch(c, n > 0 ? n : 1, raw0));
-    TRY(alloc_batch(c, n > 0 ? n : 1, raw1));
+    TRY(alloc_batch(c, n > 0 ? n : 1, raw0, true));
+    TRY(alloc_batch(c, n > 0 ? n : 1, raw1, true));
     int64_t n0 = 0, n1 = 0;
     TRY(dbspk::flatmap_events(c->stream, d_ev, n, e->query, raw0.k, raw0.v,
                               raw0.w, &n0, raw1.k, raw1.v, raw1.w, &n1));
@@ -780,7 +814,20 @@ static dbsp_status finalize_raw(dbsp_ctx *c, std::vector<DevBatch> &outs,
         return sort_consolidate_batch(c, only, out);
     }
     DevBatch cat;
-    TRY(concat_batches(c, outs, cat));
+    {
+        int64_t total = 0;
+        for (auto &b : outs) total += b.n;
+        TRY(alloc_batch(c, total, cat, true));
+        int64_t off = 0;
+        for (auto &b : outs) {
+            if (b.n == 0) continue;
+            HIP_CHECK_ST(hipMemcpyAsync(cat.k + off, b.k, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
+            HIP_CHECK_ST(hipMemcpyAsync(cat.v + off, b.v, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
+            HIP_CHECK_ST(hipMemcpyAsync(cat.w + off, b.w, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
+            off += b.n;
+        }
+        cat.n = total;
+    }
     for (auto &b : outs) free_batch(c, b);
     outs.clear();
     TRY(sort_consolidate_batch(c, cat, out));
@@ -909,19 +956,84 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     // — identical to the reference's dA join trace(P) + trace(A) delayed join dP
     // (operator/join.rs:217-292) by bilinearity.
     std::vector<DevBatch> outs;
-    TRY(join_vs_spine(c, dA, e->p_int, DBSP_PROJ_HI_V2_LO_V1, 0, outs));
-    TRY(join_vs_spine(c, dP, e->a_int, DBSP_PROJ_HI_V1_LO_V2, 0, outs));
-    if (dA.n > 0 && dP.n > 0) {
-        TraceArgs t{};
-        t.nb = 1;
-        t.k[0] = dP.k; t.v[0] = dP.v; t.w[0] = dP.w; t.n[0] = dP.n;
-        DevBatch o;
-        ScopedTimer timer(c, 2, (double)dA.n * 24.0);
-        TRY(dbspk::join_spine_rows(c->stream, dA.k, dA.v, dA.w, dA.n, t,
-                                   DBSP_PROJ_HI_V2_LO_V1, 0, &o.k, &o.v, &o.w,
-                                   &o.n));
-        if (o.n > 0) outs.push_back(o);
-        else free_batch(c, o);
+    {
+        // plan the three joins; small deltas use the one-launch count+scan
+        // kernel so a SINGLE sync covers all three output sizes
+        ScopedTimer timer(c, 2, (double)(2 * dA.n + dP.n) * 24.0);
+        struct Plan {
+            const DevBatch *delta;
+            TraceArgs t;
+            int proj;
+            uint32_t *cnts;
+            uint64_t *offsets;
+            bool small;
+        } plans[3];
+        int np = 0;
+        auto spine_args = [](Spine &sp) {
+            TraceArgs t{};
+            for (auto &b : sp.batches) {
+                if (b.n == 0) continue;
+                t.k[t.nb] = b.k; t.v[t.nb] = b.v; t.w[t.nb] = b.w;
+                t.n[t.nb] = b.n; t.nb++;
+            }
+            return t;
+        };
+        if ((int)e->p_int.batches.size() > MAX_TRACE_BATCHES)
+            TRY(e->p_int.consolidate_all(c));
+        if ((int)e->a_int.batches.size() > MAX_TRACE_BATCHES)
+            TRY(e->a_int.consolidate_all(c));
+        if (dA.n > 0 && !e->p_int.batches.empty())
+            plans[np++] = {&dA, spine_args(e->p_int), DBSP_PROJ_HI_V2_LO_V1,
+                           nullptr, nullptr, false};
+        if (dP.n > 0 && !e->a_int.batches.empty())
+            plans[np++] = {&dP, spine_args(e->a_int), DBSP_PROJ_HI_V1_LO_V2,
+                           nullptr, nullptr, false};
+        if (dA.n > 0 && dP.n > 0) {
+            TraceArgs t{};
+            t.nb = 1;
+            t.k[0] = dP.k; t.v[0] = dP.v; t.w[0] = dP.w; t.n[0] = dP.n;
+            plans[np++] = {&dA, t, DBSP_PROJ_HI_V2_LO_V1, nullptr, nullptr,
+                           false};
+        }
+        for (int i = 0; i < np; i++) {
+            Plan &pl = plans[i];
+            if (pl.t.nb == 0) continue;
+            int64_t nd = pl.delta->n;
+            pl.cnts = (uint32_t *)arena_alloc(c, (size_t)nd * pl.t.nb * 4 + 8);
+            pl.offsets = (uint64_t *)arena_alloc(c, (size_t)(nd + 1) * 8);
+            if (nd <= 8192 && pl.cnts && pl.offsets) {
+                pl.small = true;
+                TRY(dbspk::join_count_scan_small(c->stream, pl.delta->k, nd,
+                                                 pl.t, pl.cnts, pl.offsets,
+                                                 c->d_len + i));
+            }
+        }
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 8 * sizeof(int64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        for (int i = 0; i < np; i++) {
+            Plan &pl = plans[i];
+            if (pl.t.nb == 0) continue;
+            if (pl.small) {
+                int64_t total = c->h_len[i];
+                if (total <= 0) continue;
+                DevBatch o;
+                TRY(alloc_batch(c, total, o, true));
+                TRY(dbspk::join_emit_prepared(c->stream, pl.delta->k,
+                                              pl.delta->v, pl.delta->w,
+                                              pl.delta->n, pl.t, pl.cnts,
+                                              pl.offsets, total, pl.proj, 0,
+                                              o.k, o.v, o.w));
+                outs.push_back(o);
+            } else {
+                DevBatch o;
+                TRY(dbspk::join_spine_rows(c->stream, pl.delta->k, pl.delta->v,
+                                           pl.delta->w, pl.delta->n, pl.t,
+                                           pl.proj, 0, &o.k, &o.v, &o.w, &o.n));
+                if (o.n > 0) outs.push_back(o);
+                else free_batch(c, o);
+            }
+        }
     }
     TRY(e->a_int.insert(c, dA));
     TRY(e->p_int.insert(c, dP));
@@ -1065,6 +1177,7 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
 extern "C" dbsp_status dbsp_engine_step_staged(dbsp_engine *e, int64_t lo,
                                                int64_t hi) {
     if (lo < 0 || hi > e->n_events || lo > hi) return DBSP_ERR_INVALID;
+    e->ctx->arena_off = 0;
     if (e->query == 0) {
         q0_step_host(e, e->h_events.data() + lo, hi - lo);
         return DBSP_OK;
@@ -1085,6 +1198,7 @@ extern "C" dbsp_status dbsp_engine_step(dbsp_engine *e, const dbsp_event *events
         return DBSP_OK;
     }
     dbsp_ctx *c = e->ctx;
+    c->arena_off = 0;
     dbsp_event *d_ev;
     HIP_CHECK_ST(hipMallocAsync(&d_ev, n * sizeof(dbsp_event) + 64, c->stream));
     HIP_CHECK_ST(hipMemcpyAsync(d_ev, events, n * sizeof(dbsp_event),

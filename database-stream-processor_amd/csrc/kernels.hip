@@ -918,6 +918,41 @@ __global__ void k_join_emit_multi(const uint64_t *dk, const uint64_t *dv,
     }
 }
 
+// single-workgroup join count + scan (nd <= 8192): per-row match counts over
+// every spine batch, offsets via in-LDS scan, total to *d_total — one launch,
+// no host sync, no device scan pipeline
+__global__ __launch_bounds__(FUSE_THREADS, 4) void k_join_count_scan_small(
+    const uint64_t *dk, int64_t nd, TraceArgs t, uint32_t *cnts,
+    uint64_t *offsets, int64_t *d_total) {
+    __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
+    const int tid = threadIdx.x;
+    uint32_t row_tot[FUSE_ITEMS];
+    uint32_t tsum = 0;
+    for (int j = 0; j < FUSE_ITEMS; j++) {
+        int i = tid * FUSE_ITEMS + j;
+        uint32_t rt = 0;
+        if (i < nd) {
+            uint64_t key = dk[i];
+            for (int b = 0; b < t.nb; b++) {
+                int64_t lo = lower_bound_k(t.k[b], t.n[b], key);
+                int64_t c = gallop_run(t.k[b], t.n[b], key, lo);
+                cnts[(int64_t)i * t.nb + b] = (uint32_t)c;
+                rt += (uint32_t)c;
+            }
+        }
+        row_tot[j] = rt;
+        tsum += rt;
+    }
+    uint32_t total;
+    uint32_t off = fuse_scan(tsum, wave_tot, &total);
+    for (int j = 0; j < FUSE_ITEMS; j++) {
+        int i = tid * FUSE_ITEMS + j;
+        if (i < nd) offsets[i] = off;
+        off += row_tot[j];
+    }
+    if (tid == 0) *d_total = (int64_t)total;
+}
+
 // ---------------------------------------------------------------------------
 // aggregate (linear / max) + upsert  (count/emit over delta keys)
 // ---------------------------------------------------------------------------
@@ -1355,6 +1390,27 @@ dbsp_status join_spine_rows(hipStream_t s, const uint64_t *dk,
     HIP_CHECK(hipFreeAsync(cnts, s));
     HIP_CHECK(hipFreeAsync(counts, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
+    return DBSP_OK;
+}
+
+dbsp_status join_count_scan_small(hipStream_t s, const uint64_t *dk, int64_t nd,
+                                  const TraceArgs &t, uint32_t *cnts,
+                                  uint64_t *offsets, int64_t *d_total) {
+    if (nd > FUSE_MAX) return DBSP_ERR_INVALID;
+    k_join_count_scan_small<<<1, FUSE_THREADS, 0, s>>>(dk, nd, t, cnts, offsets,
+                                                       d_total);
+    return DBSP_OK;
+}
+
+dbsp_status join_emit_prepared(hipStream_t s, const uint64_t *dk,
+                               const uint64_t *dv, const int64_t *dw,
+                               int64_t nd, const TraceArgs &t,
+                               const uint32_t *cnts, const uint64_t *offsets,
+                               int64_t n_out, int proj, uint64_t param,
+                               uint64_t *ok, uint64_t *ov, int64_t *ow) {
+    if (n_out > 0)
+        k_join_emit_multi<<<grid_for(n_out), BLK, 0, s>>>(
+            dk, dv, dw, nd, t, cnts, offsets, n_out, proj, param, ok, ov, ow);
     return DBSP_OK;
 }
 

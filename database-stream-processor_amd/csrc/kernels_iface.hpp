@@ -64,6 +64,19 @@ dbsp_status merge_small(hipStream_t s, const uint64_t *ak, const uint64_t *av,
                         uint64_t *ok, uint64_t *ov, int64_t *ow,
                         int64_t *d_len);
 
+// single-workgroup join count+scan (nd <= 8192): writes per-row/per-batch
+// cnts, exclusive offsets, and the output total to *d_total (device)
+dbsp_status join_count_scan_small(hipStream_t s, const uint64_t *dk, int64_t nd,
+                                  const TraceArgs &t, uint32_t *cnts,
+                                  uint64_t *offsets, int64_t *d_total);
+// emit phase over precomputed cnts/offsets
+dbsp_status join_emit_prepared(hipStream_t s, const uint64_t *dk,
+                               const uint64_t *dv, const int64_t *dw,
+                               int64_t nd, const TraceArgs &t,
+                               const uint32_t *cnts, const uint64_t *offsets,
+                               int64_t n_out, int proj, uint64_t param,
+                               uint64_t *ok, uint64_t *ov, int64_t *ow);
+
 // join delta against a whole spine (TraceArgs) in one count/emit pair
 dbsp_status join_spine_rows(hipStream_t s, const uint64_t *dk,
                             const uint64_t *dv, const int64_t *dw, int64_t nd,
