@@ -84,11 +84,13 @@ def run_engine(ctx_dev, query, events, n_ticks, tick, world, rank, nccl_id,
     return elapsed, eng, ctx
 
 
-def cpu_baseline_leg(query, tick, budget_s=15.0):
+def cpu_baseline_leg(query, tick, budget_s=30.0):
     """Time the CPU oracle (the restatement of the crates/dbsp algorithms —
-    kind 'port', single-threaded) on a bounded sample of the same workload."""
+    kind 'port', single-threaded) on a bounded sample of the same workload.
+    The sample is the same 10M-event stream the GPU runs (trace growth makes
+    late ticks the expensive ones), cut off at ~budget_s of CPU work."""
     from dbsp_amd import oracle
-    sample_events = 40 * tick  # ~1.6M events at the default tick; ~10-20 s
+    sample_events = 250 * tick  # the full default workload (10M events)
     evs = generate_events(sample_events, seed=1)
     q = oracle.Query(query)
     t0 = time.perf_counter()
