@@ -847,30 +847,46 @@ static dbsp_status last_key(dbsp_ctx *c, const DevBatch &b, uint64_t *out,
     return DBSP_OK;
 }
 
-// window over a spine: region scans per trace batch + the batch region once
-static dbsp_status window_vs_spine(dbsp_ctx *c, const Spine &trace,
+// window over a spine: one ranges launch covers every spine batch's three
+// regions plus the tick-batch region, one emit launch copies them all
+static dbsp_status window_vs_spine(dbsp_ctx *c, Spine &trace,
                                    const DevBatch &batch, bool have_prev,
                                    uint64_t s0, uint64_t e0, uint64_t s1,
                                    uint64_t e1, std::vector<DevBatch> &outs) {
-    DevBatch empty;
+    ScopedTimer t(c, 4, 0.0);
+    if ((int)trace.batches.size() > MAX_TRACE_BATCHES)
+        TRY(trace.consolidate_all(c));
+    TraceArgs ta{};
     for (auto &b : trace.batches) {
-        DevBatch o;
-        ScopedTimer t(c, 4, 0.0);
-        TRY(dbspk::window_rows(c->stream, b.k, b.v, b.w, b.n, empty.k, empty.v,
-                               empty.w, 0, have_prev ? 1 : 0, s0, e0, s1, e1,
-                               &o.k, &o.v, &o.w, &o.n));
-        if (o.n > 0) outs.push_back(o);
-        else free_batch(c, o);
+        if (b.n == 0) continue;
+        ta.k[ta.nb] = b.k;
+        ta.v[ta.nb] = b.v;
+        ta.w[ta.nb] = b.w;
+        ta.n[ta.nb] = b.n;
+        ta.nb++;
     }
-    {
-        DevBatch o;
-        ScopedTimer t(c, 4, 0.0);
-        TRY(dbspk::window_rows(c->stream, empty.k, empty.v, empty.w, 0, batch.k,
-                               batch.v, batch.w, batch.n, 0, 0, 0, s1, e1, &o.k,
-                               &o.v, &o.w, &o.n));
-        if (o.n > 0) outs.push_back(o);
-        else free_batch(c, o);
+    int nreg = 3 * ta.nb + 1;
+    int64_t *table = (int64_t *)arena_alloc(c, (size_t)nreg * 5 * 8 + 8);
+    DevBatch tmp_table;
+    if (!table) {
+        TRY(alloc_batch(c, nreg * 2, tmp_table));  // fallback scratch
+        table = (int64_t *)tmp_table.k;
     }
+    TRY(dbspk::window_ranges_multi(c->stream, ta, batch.k, batch.n,
+                                   have_prev ? 1 : 0, s0, e0, s1, e1, table,
+                                   c->d_len + 7));
+    HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 7, c->d_len + 7, sizeof(int64_t),
+                                hipMemcpyDeviceToHost, c->stream));
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    int64_t total = c->h_len[7];
+    if (total > 0) {
+        DevBatch o;
+        TRY(alloc_batch(c, total, o, true));
+        TRY(dbspk::window_emit_multi(c->stream, ta, batch.k, batch.v, batch.w,
+                                     table, nreg, total, o.k, o.v, o.w));
+        outs.push_back(o);
+    }
+    if (tmp_table.k) free_batch(c, tmp_table);
     return DBSP_OK;
 }
 

@@ -200,27 +200,38 @@ static dbsp_status scan_exclusive(hipStream_t s, const uint64_t *in,
 // ---------------------------------------------------------------------------
 
 __global__ void k_minmax_u64(const uint64_t *a, const uint64_t *b, int64_t n,
-                             uint64_t *out_max /* [2] */) {
+                             uint64_t *out /* [4]: maxA,maxB,minA,minB */) {
     __shared__ uint64_t smax[2];
-    if (threadIdx.x == 0) { smax[0] = 0; smax[1] = 0; }
+    __shared__ uint64_t smin[2];
+    if (threadIdx.x == 0) {
+        smax[0] = 0; smax[1] = 0;
+        smin[0] = ~0ull; smin[1] = ~0ull;
+    }
     __syncthreads();
-    uint64_t ma = 0, mb = 0;
+    uint64_t ma = 0, mb = 0, na = ~0ull, nb = ~0ull;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += (int64_t)gridDim.x * blockDim.x) {
         ma = max(ma, a[i]);
         mb = max(mb, b[i]);
+        na = min(na, a[i]);
+        nb = min(nb, b[i]);
     }
     atomicMax((unsigned long long *)&smax[0], (unsigned long long)ma);
     atomicMax((unsigned long long *)&smax[1], (unsigned long long)mb);
+    atomicMin((unsigned long long *)&smin[0], (unsigned long long)na);
+    atomicMin((unsigned long long *)&smin[1], (unsigned long long)nb);
     __syncthreads();
     if (threadIdx.x == 0) {
-        atomicMax((unsigned long long *)&out_max[0], (unsigned long long)smax[0]);
-        atomicMax((unsigned long long *)&out_max[1], (unsigned long long)smax[1]);
+        atomicMax((unsigned long long *)&out[0], (unsigned long long)smax[0]);
+        atomicMax((unsigned long long *)&out[1], (unsigned long long)smax[1]);
+        atomicMin((unsigned long long *)&out[2], (unsigned long long)smin[0]);
+        atomicMin((unsigned long long *)&out[3], (unsigned long long)smin[1]);
     }
 }
 
-__device__ inline uint32_t sort_digit(uint64_t k, uint64_t v, int byte) {
-    uint64_t limb = byte < 8 ? v : k;
+__device__ inline uint32_t sort_digit(uint64_t k, uint64_t v, int byte,
+                                      uint64_t kbase, uint64_t vbase) {
+    uint64_t limb = byte < 8 ? v - vbase : k - kbase;
     int sh = (byte & 7) * 8;
     return (uint32_t)((limb >> sh) & 0xFF);
 }
@@ -228,14 +239,15 @@ __device__ inline uint32_t sort_digit(uint64_t k, uint64_t v, int byte) {
 // histogram: counts[d * nblocks + blk] (digit-major so one exclusive scan of
 // the flat array yields combined digit-base + block-offset)
 __global__ void k_radix_hist(const uint64_t *k, const uint64_t *v, int64_t n,
-                             int byte, int64_t nblocks, uint64_t *counts) {
+                             int byte, uint64_t kbase, uint64_t vbase,
+                             int64_t nblocks, uint64_t *counts) {
     __shared__ uint32_t hist[256];
     for (int i = threadIdx.x; i < 256; i += BLK) hist[i] = 0;
     __syncthreads();
     int64_t base = (int64_t)blockIdx.x * SORT_TILE;
     for (int i = threadIdx.x; i < SORT_TILE; i += BLK) {
         int64_t idx = base + i;
-        if (idx < n) atomicAdd(&hist[sort_digit(k[idx], v[idx], byte)], 1u);
+        if (idx < n) atomicAdd(&hist[sort_digit(k[idx], v[idx], byte, kbase, vbase)], 1u);
     }
     __syncthreads();
     for (int d = threadIdx.x; d < 256; d += BLK)
@@ -246,6 +258,7 @@ __global__ void k_radix_hist(const uint64_t *k, const uint64_t *v, int64_t n,
 // rank within digit run + scanned global base.
 __global__ void k_radix_scatter(const uint64_t *k_in, const uint64_t *v_in,
                                 const int64_t *w_in, int64_t n, int byte,
+                                uint64_t kbase, uint64_t vbase,
                                 int64_t nblocks, const uint64_t *scanned,
                                 uint64_t *k_out, uint64_t *v_out,
                                 int64_t *w_out) {
@@ -261,7 +274,7 @@ __global__ void k_radix_scatter(const uint64_t *k_in, const uint64_t *v_in,
     // load packed (digit << 16 | idx)
     for (int i = threadIdx.x; i < SORT_TILE; i += BLK) {
         uint32_t d = 0xFF;  // pad with max digit so pads sort last
-        if (i < tile_n) d = sort_digit(k_in[base + i], v_in[base + i], byte);
+        if (i < tile_n) d = sort_digit(k_in[base + i], v_in[base + i], byte, kbase, vbase);
         buf_a[i] = (d << 16) | (uint32_t)i;
     }
     __syncthreads();
@@ -397,6 +410,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     __shared__ uint32_t cnt[FUSE_DIGITS * FUSE_THREADS];  // rank counters
     __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
     __shared__ uint64_t smax[2];
+    __shared__ uint64_t smin[2];
     const int tid = threadIdx.x;
 
     if (n == 0) {
@@ -404,22 +418,33 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
         return;
     }
 
-    // ---- significant bits of max(k), max(v) ----
-    if (tid == 0) { smax[0] = 0; smax[1] = 0; }
+    // ---- significant bits of (max-min) for k and v: sorting by (limb - min)
+    // preserves order and collapses narrow ranges (per-tick timestamps span a
+    // few ms; ids cluster) to far fewer digit passes ----
+    if (tid == 0) {
+        smax[0] = 0; smax[1] = 0;
+        smin[0] = ~0ull; smin[1] = ~0ull;
+    }
     __syncthreads();
     {
-        uint64_t mk = 0, mv = 0;
+        uint64_t mk = 0, mv = 0, nk = ~0ull, nv = ~0ull;
         for (int64_t i = tid; i < n; i += FUSE_THREADS) {
             mk = max(mk, kin[i]);
             mv = max(mv, vin[i]);
+            nk = min(nk, kin[i]);
+            nv = min(nv, vin[i]);
         }
         atomicMax((unsigned long long *)&smax[0], (unsigned long long)mk);
         atomicMax((unsigned long long *)&smax[1], (unsigned long long)mv);
+        atomicMin((unsigned long long *)&smin[0], (unsigned long long)nk);
+        atomicMin((unsigned long long *)&smin[1], (unsigned long long)nv);
     }
     __syncthreads();
+    const uint64_t kbase = smin[0], vbase = smin[1];
+    const uint64_t krange = smax[0] - kbase, vrange = smax[1] - vbase;
     int knibs = 0, vnibs = 0;
-    while (knibs < 16 && (smax[0] >> (4 * knibs)) != 0) knibs++;
-    while (vnibs < 16 && (smax[1] >> (4 * vnibs)) != 0) vnibs++;
+    while (knibs < 16 && (krange >> (4 * knibs)) != 0) knibs++;
+    while (vnibs < 16 && (vrange >> (4 * vnibs)) != 0) vnibs++;
 
     // adaptive sizing: only t_act threads carry items (wave-rounded), so the
     // per-pass counter array is 16*t_act cells, not 16*1024 — small deltas
@@ -453,7 +478,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
                 uint32_t idx = src[i] & 0x1FFFu;
                 uint32_t d = 15;
                 if ((int64_t)idx < n) {
-                    uint64_t limb = is_v ? vin[idx] : kin[idx];
+                    uint64_t limb = is_v ? vin[idx] - vbase : kin[idx] - kbase;
                     d = (uint32_t)((limb >> sh) & 0xF);
                 }
                 src[i] = (d << 13) | idx;
@@ -1077,6 +1102,86 @@ __global__ void k_window_emit(const uint64_t *tk, const uint64_t *tv,
     }
 }
 
+// multi-batch window: all three regions of every spine batch plus the tick's
+// batch region in one ranges launch + one emit launch (window.rs:144-220
+// evaluated per spine batch; the trace excludes the current tick).
+// Region table rows: [src_batch(-1 = tick batch), lo, len, sign, goff]
+__global__ void k_window_ranges_multi(TraceArgs t, const uint64_t *bk,
+                                      int64_t bn, int have_prev, uint64_t s0,
+                                      uint64_t e0, uint64_t s1, uint64_t e1,
+                                      int64_t *table, int64_t *d_total) {
+    const int nreg = 3 * t.nb + 1;
+    for (int r = threadIdx.x; r < nreg; r += blockDim.x) {
+        int64_t src = -1, lo = 0, len = 0, sign = 1;
+        if (r < 3 * t.nb) {
+            int b = r / 3, which = r % 3;
+            const uint64_t *k = t.k[b];
+            int64_t n = t.n[b];
+            src = b;
+            if (have_prev) {
+                if (which == 0) {  // region 1: [s0, min(s1,e0)) retract
+                    uint64_t end = min(s1, e0);
+                    lo = lower_bound_k(k, n, s0);
+                    len = max((int64_t)0, lower_bound_k(k, n, end) - lo);
+                    sign = -1;
+                } else if (which == 1) {  // shrink: [e1, e0) retract
+                    if (e1 < e0) {
+                        lo = lower_bound_k(k, n, e1);
+                        len = max((int64_t)0, lower_bound_k(k, n, e0) - lo);
+                    }
+                    sign = -1;
+                } else {  // region 3: [max(e0,s1), e1) insert
+                    uint64_t st = max(e0, s1);
+                    lo = lower_bound_k(k, n, st);
+                    len = max((int64_t)0, lower_bound_k(k, n, e1) - lo);
+                }
+            }
+        } else {  // batch region: [s1, e1) insert
+            lo = lower_bound_k(bk, bn, s1);
+            len = max((int64_t)0, lower_bound_k(bk, bn, e1) - lo);
+        }
+        table[r * 5 + 0] = src;
+        table[r * 5 + 1] = lo;
+        table[r * 5 + 2] = len;
+        table[r * 5 + 3] = sign;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int64_t acc = 0;
+        for (int r = 0; r < nreg; r++) {
+            table[r * 5 + 4] = acc;
+            acc += table[r * 5 + 2];
+        }
+        *d_total = acc;
+    }
+}
+
+__global__ void k_window_emit_multi(TraceArgs t, const uint64_t *bk,
+                                    const uint64_t *bv, const int64_t *bw,
+                                    const int64_t *table, int nreg,
+                                    int64_t total, uint64_t *ok, uint64_t *ov,
+                                    int64_t *ow) {
+    for (int64_t o = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; o < total;
+         o += (int64_t)gridDim.x * blockDim.x) {
+        // binary search the region by goff
+        int lo = 0, hi = nreg;
+        while (lo < hi) {
+            int mid = (lo + hi) / 2;
+            if (table[mid * 5 + 4] <= o) lo = mid + 1; else hi = mid;
+        }
+        int r = lo - 1;
+        int64_t src = table[r * 5 + 0];
+        int64_t i = table[r * 5 + 1] + (o - table[r * 5 + 4]);
+        int64_t sign = table[r * 5 + 3];
+        const uint64_t *k = src < 0 ? bk : t.k[src];
+        const uint64_t *v = src < 0 ? bv : t.v[src];
+        const int64_t *w = src < 0 ? bw : t.w[src];
+        ok[o] = k[i];
+        ov[o] = v[i];
+        ow[o] = sign * w[i];
+    }
+}
+
 // ---------------------------------------------------------------------------
 // shard partition: xxh3(key) % nshards (shard.rs:165-199, hash.rs:9-13)
 // ---------------------------------------------------------------------------
@@ -1226,17 +1331,20 @@ dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
     if (n <= 1) return DBSP_OK;
     // significant bytes from max values
     uint64_t *d_max;
-    HIP_CHECK(hipMallocAsync(&d_max, 2 * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&d_max, 4 * sizeof(uint64_t), s));
     HIP_CHECK(hipMemsetAsync(d_max, 0, 2 * sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(d_max + 2, 0xFF, 2 * sizeof(uint64_t), s));
     k_minmax_u64<<<grid_for(n), BLK, 0, s>>>(kk, vv, n, d_max);
-    uint64_t h_max[2];
-    HIP_CHECK(hipMemcpyAsync(h_max, d_max, 2 * sizeof(uint64_t),
+    uint64_t h_max[4];
+    HIP_CHECK(hipMemcpyAsync(h_max, d_max, 4 * sizeof(uint64_t),
                              hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
     HIP_CHECK(hipFreeAsync(d_max, s));
+    const uint64_t kbase = h_max[2], vbase = h_max[3];
+    const uint64_t krange = h_max[0] - kbase, vrange = h_max[1] - vbase;
     int kbytes = 0, vbytes = 0;
-    while (kbytes < 8 && (h_max[0] >> (8 * kbytes)) != 0) kbytes++;
-    while (vbytes < 8 && (h_max[1] >> (8 * vbytes)) != 0) vbytes++;
+    while (kbytes < 8 && (krange >> (8 * kbytes)) != 0) kbytes++;
+    while (vbytes < 8 && (vrange >> (8 * vbytes)) != 0) vbytes++;
 
     int64_t nblocks = ceil_div(n, SORT_TILE);
     uint64_t *counts;
@@ -1248,12 +1356,13 @@ dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
         bool is_v = byte < 8;
         if (is_v && (byte & 7) >= vbytes) continue;
         if (!is_v && (byte & 7) >= kbytes) continue;
-        k_radix_hist<<<dim3((uint32_t)nblocks), BLK, 0, s>>>(src_k, src_v, n, byte,
-                                                            nblocks, counts);
+        k_radix_hist<<<dim3((uint32_t)nblocks), BLK, 0, s>>>(
+            src_k, src_v, n, byte, kbase, vbase, nblocks, counts);
         dbsp_status st = scan_exclusive(s, counts, counts, 256 * nblocks, nullptr);
         if (st != DBSP_OK) return st;
         k_radix_scatter<<<dim3((uint32_t)nblocks), BLK, 0, s>>>(
-            src_k, src_v, src_w, n, byte, nblocks, counts, dst_k, dst_v, dst_w);
+            src_k, src_v, src_w, n, byte, kbase, vbase, nblocks, counts, dst_k,
+            dst_v, dst_w);
         uint64_t *t;
         int64_t *tw;
         t = src_k; src_k = dst_k; dst_k = t;
@@ -1534,6 +1643,27 @@ dbsp_status window_rows(hipStream_t s, const uint64_t *tk, const uint64_t *tv,
                                                       d_ranges, rk, rv, rw);
     HIP_CHECK(hipFreeAsync(d_ranges, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = total;
+    return DBSP_OK;
+}
+
+dbsp_status window_ranges_multi(hipStream_t s, const TraceArgs &t,
+                                const uint64_t *bk, int64_t bn, int have_prev,
+                                uint64_t s0, uint64_t e0, uint64_t s1,
+                                uint64_t e1, int64_t *table, int64_t *d_total) {
+    k_window_ranges_multi<<<1, BLK, 0, s>>>(t, bk, bn, have_prev, s0, e0, s1,
+                                            e1, table, d_total);
+    return DBSP_OK;
+}
+
+dbsp_status window_emit_multi(hipStream_t s, const TraceArgs &t,
+                              const uint64_t *bk, const uint64_t *bv,
+                              const int64_t *bw, const int64_t *table,
+                              int nreg, int64_t total, uint64_t *ok,
+                              uint64_t *ov, int64_t *ow) {
+    if (total > 0)
+        k_window_emit_multi<<<grid_for(total), BLK, 0, s>>>(t, bk, bv, bw,
+                                                            table, nreg, total,
+                                                            ok, ov, ow);
     return DBSP_OK;
 }
 

@@ -123,6 +123,19 @@ dbsp_status window_rows(hipStream_t s, const uint64_t *tk, const uint64_t *tv,
                         uint64_t e1, uint64_t **ok, uint64_t **ov, int64_t **ow,
                         int64_t *out_n);
 
+// multi-batch window: ranges for all spine batches + tick batch in one launch
+// (region table rows [src,lo,len,sign,goff]; total left in *d_total), then one
+// grid-stride emit
+dbsp_status window_ranges_multi(hipStream_t s, const TraceArgs &t,
+                                const uint64_t *bk, int64_t bn, int have_prev,
+                                uint64_t s0, uint64_t e0, uint64_t s1,
+                                uint64_t e1, int64_t *table, int64_t *d_total);
+dbsp_status window_emit_multi(hipStream_t s, const TraceArgs &t,
+                              const uint64_t *bk, const uint64_t *bv,
+                              const int64_t *bw, const int64_t *table,
+                              int nreg, int64_t total, uint64_t *ok,
+                              uint64_t *ov, int64_t *ow);
+
 dbsp_status shard_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
                        const int64_t *w, int64_t n, int nshards, uint64_t *ok,
                        uint64_t *ov, int64_t *ow, int64_t *h_offsets);
