@@ -294,7 +294,7 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
 static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
                                  const DevBatch &b, DevBatch &out) {
     ScopedTimer t(c, 1, (double)(a.n + b.n) * 48.0);
-    if (a.n + b.n <= 8192) {
+    if (a.n + b.n <= 32768) {
         // one launch + one length readback (merge-path diagonals in one WG)
         DevBatch res;
         TRY(alloc_batch(c, a.n + b.n, res));
@@ -983,6 +983,7 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             uint32_t *cnts;
             uint64_t *offsets;
             bool small;
+            int slot;
         } plans[3];
         int np = 0;
         auto spine_args = [](Spine &sp) {
@@ -1011,19 +1012,27 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             plans[np++] = {&dA, t, DBSP_PROJ_HI_V2_LO_V1, nullptr, nullptr,
                            false};
         }
+        JoinCountArgs jca{};
         for (int i = 0; i < np; i++) {
             Plan &pl = plans[i];
+            pl.slot = -1;
             if (pl.t.nb == 0) continue;
             int64_t nd = pl.delta->n;
             pl.cnts = (uint32_t *)arena_alloc(c, (size_t)nd * pl.t.nb * 4 + 8);
             pl.offsets = (uint64_t *)arena_alloc(c, (size_t)(nd + 1) * 8);
             if (nd <= 8192 && pl.cnts && pl.offsets) {
                 pl.small = true;
-                TRY(dbspk::join_count_scan_small(c->stream, pl.delta->k, nd,
-                                                 pl.t, pl.cnts, pl.offsets,
-                                                 c->d_len + i));
+                pl.slot = jca.np;
+                jca.dk[jca.np] = pl.delta->k;
+                jca.nd[jca.np] = nd;
+                jca.t[jca.np] = pl.t;
+                jca.cnts[jca.np] = pl.cnts;
+                jca.offsets[jca.np] = pl.offsets;
+                jca.np++;
             }
         }
+        jca.d_total = c->d_len;
+        if (jca.np > 0) TRY(dbspk::join_count_scan_batch(c->stream, jca));
         HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 8 * sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -1031,7 +1040,7 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             Plan &pl = plans[i];
             if (pl.t.nb == 0) continue;
             if (pl.small) {
-                int64_t total = c->h_len[i];
+                int64_t total = c->h_len[pl.slot];
                 if (total <= 0) continue;
                 DevBatch o;
                 TRY(alloc_batch(c, total, o, true));

@@ -832,8 +832,9 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_small(
     __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
     const int tid = threadIdx.x;
     const int64_t total = na + nb;
-    int64_t d0 = min((int64_t)tid * FUSE_ITEMS, total);
-    int64_t d1 = min(d0 + FUSE_ITEMS, total);
+    const int64_t items = (total + FUSE_THREADS - 1) / FUSE_THREADS;
+    int64_t d0 = min((int64_t)tid * items, total);
+    int64_t d1 = min(d0 + items, total);
     int64_t ai, bi, ae, be;
     merge_path(ak, av, na, bk, bv, nb, d0, ai, bi);
     adjust_split(ak, av, bk, bv, na, nb, ai, bi);
@@ -947,9 +948,15 @@ __global__ void k_join_emit_multi(const uint64_t *dk, const uint64_t *dv,
 // every spine batch, offsets via in-LDS scan, total to *d_total — one launch,
 // no host sync, no device scan pipeline
 __global__ __launch_bounds__(FUSE_THREADS, 4) void k_join_count_scan_small(
-    const uint64_t *dk, int64_t nd, TraceArgs t, uint32_t *cnts,
-    uint64_t *offsets, int64_t *d_total) {
+    JoinCountArgs args) {
     __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
+    const int plan = blockIdx.x;
+    const uint64_t *dk = args.dk[plan];
+    const int64_t nd = args.nd[plan];
+    const TraceArgs &t = args.t[plan];
+    uint32_t *cnts = args.cnts[plan];
+    uint64_t *offsets = args.offsets[plan];
+    int64_t *d_total = args.d_total + plan;
     const int tid = threadIdx.x;
     uint32_t row_tot[FUSE_ITEMS];
     uint32_t tsum = 0;
@@ -1465,7 +1472,7 @@ dbsp_status merge_small(hipStream_t s, const uint64_t *ak, const uint64_t *av,
                         const uint64_t *bv, const int64_t *bw, int64_t nb,
                         uint64_t *ok, uint64_t *ov, int64_t *ow,
                         int64_t *d_len) {
-    if (na + nb > FUSE_MAX) return DBSP_ERR_INVALID;
+    if (na + nb > 4 * FUSE_MAX) return DBSP_ERR_INVALID;
     k_merge_small<<<1, FUSE_THREADS, 0, s>>>(ak, av, aw, na, bk, bv, bw, nb, ok,
                                              ov, ow, d_len);
     return DBSP_OK;
@@ -1502,12 +1509,11 @@ dbsp_status join_spine_rows(hipStream_t s, const uint64_t *dk,
     return DBSP_OK;
 }
 
-dbsp_status join_count_scan_small(hipStream_t s, const uint64_t *dk, int64_t nd,
-                                  const TraceArgs &t, uint32_t *cnts,
-                                  uint64_t *offsets, int64_t *d_total) {
-    if (nd > FUSE_MAX) return DBSP_ERR_INVALID;
-    k_join_count_scan_small<<<1, FUSE_THREADS, 0, s>>>(dk, nd, t, cnts, offsets,
-                                                       d_total);
+dbsp_status join_count_scan_batch(hipStream_t s, const JoinCountArgs &args) {
+    if (args.np == 0) return DBSP_OK;
+    for (int i = 0; i < args.np; i++)
+        if (args.nd[i] > FUSE_MAX) return DBSP_ERR_INVALID;
+    k_join_count_scan_small<<<dim3((uint32_t)args.np), FUSE_THREADS, 0, s>>>(args);
     return DBSP_OK;
 }
 

@@ -30,6 +30,16 @@ struct SortArgs {
     int64_t *d_len;  // device array, one length per batch
 };
 
+struct JoinCountArgs {
+    int np;
+    const uint64_t *dk[3];
+    int64_t nd[3];
+    TraceArgs t[3];
+    uint32_t *cnts[3];
+    uint64_t *offsets[3];
+    int64_t *d_total;
+};
+
 namespace dbspk {
 
 dbsp_status scan_excl(hipStream_t s, const uint64_t *in, uint64_t *out,
@@ -64,11 +74,10 @@ dbsp_status merge_small(hipStream_t s, const uint64_t *ak, const uint64_t *av,
                         uint64_t *ok, uint64_t *ov, int64_t *ow,
                         int64_t *d_len);
 
-// single-workgroup join count+scan (nd <= 8192): writes per-row/per-batch
-// cnts, exclusive offsets, and the output total to *d_total (device)
-dbsp_status join_count_scan_small(hipStream_t s, const uint64_t *dk, int64_t nd,
-                                  const TraceArgs &t, uint32_t *cnts,
-                                  uint64_t *offsets, int64_t *d_total);
+// up to 3 single-workgroup join count+scan plans in ONE launch (nd <= 8192
+// each): per-plan per-row/per-batch cnts, exclusive offsets, totals to
+// d_total[i] (device)
+dbsp_status join_count_scan_batch(hipStream_t s, const JoinCountArgs &args);
 // emit phase over precomputed cnts/offsets
 dbsp_status join_emit_prepared(hipStream_t s, const uint64_t *dk,
                                const uint64_t *dv, const int64_t *dw,
