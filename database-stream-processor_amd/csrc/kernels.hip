@@ -1264,7 +1264,7 @@ __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
                        (e.f3 == 1 || e.f3 == 2 || e.f3 == 3)) {  // CA, ID, OR
                 uint64_t p = atomicAdd((unsigned long long *)c1, 1ull);
                 k1[p] = e.f0;
-                v1[p] = (e.f1 << 16) | ((e.f2 & 0xFF) << 8) | (e.f3 & 0xFF);
+                v1[p] = (e.f1 << 8) | ((e.f2 & 0xF) << 4) | (e.f3 & 0xF);
                 w1[p] = e.w;
             }
         } else if (query == 5) {

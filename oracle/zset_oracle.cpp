@@ -289,7 +289,7 @@ inline bool q3_state_of_interest(uint64_t state_id) {
 constexpr uint64_t Q3_CATEGORY = 10;  // queries/q3.rs:31 (FIRST_CATEGORY_ID=10)
 
 inline uint64_t pack_person(uint64_t name_id, uint64_t city_id, uint64_t state_id) {
-    return (name_id << 16) | ((city_id & 0xFF) << 8) | (state_id & 0xFF);
+    return (name_id << 8) | ((city_id & 0xF) << 4) | (state_id & 0xF);
 }
 
 struct Oracle {

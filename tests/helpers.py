@@ -33,7 +33,9 @@ class Intern:
 def state_id(s, intern=None):
     if s in STATE_IDS:
         return STATE_IDS[s]
-    return 90 + (intern(s) if intern else hash(s) % 100)
+    # out-of-dictionary states (e.g. "NL" in the q3 golden test) map into the
+    # unused 4-bit id range 6..15 — must only be distinct from CA/ID/OR
+    return 6 + ((intern(s) if intern else hash(s)) % 10)
 
 
 def person_event(pid, name_id, city_id, sid, dt=0, w=1):
@@ -65,7 +67,9 @@ def events(*evs):
 
 
 def pack_person(name_id, city_id, sid):
-    return (name_id << 16) | ((city_id & 0xFF) << 8) | (sid & 0xFF)
+    # compact person tuple id: name(10b) | city(4b) | state(4b) — short keys
+    # mean fewer radix digit passes (helpers mirror oracle/kernels exactly)
+    return (name_id << 8) | ((city_id & 0xF) << 4) | (sid & 0xF)
 
 
 def zset(rows_arr):

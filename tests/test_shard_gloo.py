@@ -86,7 +86,7 @@ def _worker(rank, world, events_by_tick, out_q):
                 dA.append((e["f1"], e["f0"], e["w"]))
             if e["kind"] == 0 and e["f3"] in (1, 2, 3):
                 dP.append((e["f0"],
-                           (int(e["f1"]) << 16) | (int(e["f2"]) << 8) | int(e["f3"]),
+                           (int(e["f1"]) << 8) | (int(e["f2"]) << 4) | int(e["f3"]),
                            e["w"]))
         dA = np.array(dA, dtype=ROW_DT) if dA else np.empty(0, dtype=ROW_DT)
         dP = np.array(dP, dtype=ROW_DT) if dP else np.empty(0, dtype=ROW_DT)
