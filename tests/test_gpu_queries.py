@@ -108,6 +108,29 @@ def test_golden_q8_on_gpu(ctx):
         eng.close()
 
 
+@pytest.mark.parametrize("query", [3, 5, 8])
+def test_query_parity_with_retractions(ctx, query):
+    """Streams carrying deletions: every third event of earlier ticks is
+    re-issued later with weight -1 (the reference's Z-set deltas carry
+    insertions AND retractions; generated streams alone only exercise w=+1)."""
+    evs = gen.generate(30_000, seed=17)
+    retract = evs[::3].copy()
+    retract["w"] = -1
+    mixed = np.concatenate([evs, retract])
+    _run_parity(ctx, query, mixed, tick=6000, seed_note="+retractions")
+
+
+def test_q3_duplicate_and_zero_weight_events(ctx):
+    """Duplicate events (weights accumulate) and exact cancellations inside
+    one tick (weight-zero elimination end to end)."""
+    evs = gen.generate(5_000, seed=19)
+    dup = evs[:2000].copy()
+    neg = evs[:1000].copy()
+    neg["w"] = -2
+    mixed = np.concatenate([evs, dup, neg])
+    _run_parity(ctx, 3, mixed, tick=1000, seed_note="+dups")
+
+
 def test_q0_host_path(ctx):
     """q0 runs the host plumbing path (BASELINE configs[0]: CPU, 1 worker)."""
     from dbsp_amd.engine import Engine
