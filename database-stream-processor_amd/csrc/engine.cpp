@@ -71,6 +71,8 @@ struct dbsp_ctx {
     // RCCL
     ncclComm_t comm = nullptr;
     int rank = 0, world = 1;
+    bool force_shard = false;  // exercise the full partition+alltoallv path
+                               // even at world=1 (self-exchange; test hook)
     // persistent length scratch (device + pinned host)
     int64_t *d_len = nullptr;
     int64_t *h_len = nullptr;
@@ -125,6 +127,8 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     HIP_CHECK_ST(hipEventCreate(&c->ev1));
     const char *p = getenv("DBSP_PROFILE");
     c->profile = p && p[0] == '1';
+    const char *fs = getenv("DBSP_FORCE_SHARD");
+    c->force_shard = fs && fs[0] == '1';
     HIP_CHECK_ST(hipMalloc(&c->d_len, 8 * sizeof(int64_t)));
     c->arena_sz = (size_t)512 << 20;
     if (hipMalloc(&c->arena, c->arena_sz) != hipSuccess) {
@@ -572,7 +576,7 @@ extern "C" dbsp_status dbsp_comm_alltoallv(dbsp_ctx *c, const dbsp_batch *send,
 // shard + exchange + rebuild: the full shard() operator
 // (shard.rs:88-199: hash-split, exchange, re-consolidate)
 static dbsp_status shard_exchange(dbsp_ctx *c, DevBatch local, DevBatch &out) {
-    if (c->world <= 1) {
+    if (c->world <= 1 && !(c->force_shard && c->comm)) {
         out = local;
         return DBSP_OK;
     }
@@ -765,7 +769,7 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
             free_batch(c, raw1);
     }
     // worker sharding: co-locate keys across ranks (shard.rs:88)
-    if (e->world > 1) {
+    if (e->world > 1 || (c->force_shard && c->comm)) {
         DevBatch s0;
         TRY(shard_exchange(c, d0, s0));
         d0 = s0;
