@@ -76,6 +76,7 @@ struct dbsp_ctx {
     // persistent length scratch (device + pinned host)
     int64_t *d_len = nullptr;
     int64_t *h_len = nullptr;
+    int timer_depth = 0;  // ScopedTimer nesting guard (shared event pair)
     // per-tick transient bump arena (reset at each engine tick; falls back to
     // the stream-ordered pool when exhausted)
     uint8_t *arena = nullptr;
@@ -184,11 +185,17 @@ struct ScopedTimer {
     dbsp_ctx *c;
     int cls;
     double bytes;
+    bool active;
     ScopedTimer(dbsp_ctx *c_, int cls_, double bytes_) : c(c_), cls(cls_), bytes(bytes_) {
-        if (c->profile) hipEventRecord(c->ev0, c->stream);
+        // nested ops (e.g. the merge tree inside a medium sort) attribute to
+        // the OUTER class — one shared event pair
+        active = c->profile && c->timer_depth == 0;
+        c->timer_depth++;
+        if (active) hipEventRecord(c->ev0, c->stream);
     }
     ~ScopedTimer() {
-        if (c->profile) {
+        c->timer_depth--;
+        if (active) {
             hipEventRecord(c->ev1, c->stream);
             hipEventSynchronize(c->ev1);
             float ms = 0;
@@ -254,6 +261,80 @@ static dbsp_status concat_batches(dbsp_ctx *c, const std::vector<DevBatch> &in,
     return DBSP_OK;
 }
 
+static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
+                                 const DevBatch &b, DevBatch &out);
+
+// medium raw batches (8192 < n <= 64k): chunk into <=8 fused single-WG sorts
+// in ONE launch, then a batched single-WG merge tree (3 launches, 1 sync per
+// round) — replaces ~25 launches of the global radix pipeline at these sizes
+static dbsp_status sort_medium(dbsp_ctx *c, DevBatch raw, DevBatch &out) {
+    int nch = (int)((raw.n + 8191) / 8192);
+    SortArgs sa{};
+    sa.nb = nch;
+    std::vector<DevBatch> chunks(nch), scratch(nch);
+    for (int i = 0; i < nch; i++) {
+        int64_t lo = (int64_t)i * 8192;
+        int64_t len = std::min<int64_t>(8192, raw.n - lo);
+        TRY(alloc_batch(c, len, scratch[i], true));
+        TRY(alloc_batch(c, len, chunks[i], true));
+        sa.kin[i] = raw.k + lo; sa.vin[i] = raw.v + lo; sa.win[i] = raw.w + lo;
+        sa.n[i] = len;
+        sa.tk[i] = scratch[i].k; sa.tv[i] = scratch[i].v; sa.tw[i] = scratch[i].w;
+        sa.ok[i] = chunks[i].k; sa.ov[i] = chunks[i].v; sa.ow[i] = chunks[i].w;
+    }
+    sa.d_len = c->d_len;
+    TRY(dbspk::sort_cons_small_batch(c->stream, sa));
+    HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, nch * sizeof(int64_t),
+                                hipMemcpyDeviceToHost, c->stream));
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    for (int i = 0; i < nch; i++) chunks[i].n = c->h_len[i];
+    free_batch(c, raw);
+    for (auto &sc : scratch) free_batch(c, sc);
+    // batched pairwise merge rounds
+    std::vector<DevBatch> cur(chunks.begin(), chunks.end());
+    while (cur.size() > 1) {
+        std::vector<DevBatch> next;
+        MergeArgs ma{};
+        std::vector<DevBatch> results;
+        size_t i = 0;
+        for (; i + 1 < cur.size() && ma.np < MERGE_BATCH_MAX; i += 2) {
+            DevBatch &a = cur[i], &b = cur[i + 1];
+            if (a.n + b.n > 32768) break;  // fall through to merge_batches
+            DevBatch res;
+            bool final_round = (cur.size() == 2);
+            TRY(alloc_batch(c, a.n + b.n, res, !final_round));
+            int p = ma.np++;
+            ma.ak[p] = a.k; ma.av[p] = a.v; ma.aw[p] = a.w; ma.na[p] = a.n;
+            ma.bk[p] = b.k; ma.bv[p] = b.v; ma.bw[p] = b.w; ma.nb[p] = b.n;
+            ma.ok[p] = res.k; ma.ov[p] = res.v; ma.ow[p] = res.w;
+            results.push_back(res);
+        }
+        if (ma.np > 0) {
+            ma.d_len = c->d_len;
+            TRY(dbspk::merge_small_batch(c->stream, ma));
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
+                                        ma.np * sizeof(int64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            for (int p = 0; p < ma.np; p++) results[p].n = c->h_len[p];
+            for (size_t j = 0; j < (size_t)(2 * ma.np); j++) free_batch(c, cur[j]);
+            next.insert(next.end(), results.begin(), results.end());
+        }
+        // remaining (odd leftover or >32k pairs) via the generic path
+        for (; i + 1 < cur.size(); i += 2) {
+            DevBatch res;
+            TRY(merge_batches(c, cur[i], cur[i + 1], res));
+            free_batch(c, cur[i]);
+            free_batch(c, cur[i + 1]);
+            next.push_back(res);
+        }
+        if (i < cur.size()) next.push_back(cur[i]);  // odd carry
+        cur = std::move(next);
+    }
+    out = cur.empty() ? DevBatch{} : cur[0];
+    return DBSP_OK;
+}
+
 // sort + consolidate a RAW batch (consumes `raw`)
 static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &out) {
     if (raw.n == 0) {
@@ -262,6 +343,7 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
         return DBSP_OK;
     }
     ScopedTimer t(c, 0, (double)raw.n * 48.0);
+    if (raw.n > 8192 && raw.n <= 65536) return sort_medium(c, raw, out);
     DevBatch scratch;
     TRY(alloc_batch(c, raw.n, scratch, true));
     if (raw.n <= 8192) {
@@ -302,8 +384,13 @@ static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
         // one launch + one length readback (merge-path diagonals in one WG)
         DevBatch res;
         TRY(alloc_batch(c, a.n + b.n, res));
-        TRY(dbspk::merge_small(c->stream, a.k, a.v, a.w, a.n, b.k, b.v, b.w,
-                               b.n, res.k, res.v, res.w, c->d_len));
+        MergeArgs ma{};
+        ma.np = 1;
+        ma.ak[0] = a.k; ma.av[0] = a.v; ma.aw[0] = a.w; ma.na[0] = a.n;
+        ma.bk[0] = b.k; ma.bv[0] = b.v; ma.bw[0] = b.w; ma.nb[0] = b.n;
+        ma.ok[0] = res.k; ma.ov[0] = res.v; ma.ow[0] = res.w;
+        ma.d_len = c->d_len;
+        TRY(dbspk::merge_small_batch(c->stream, ma));
         HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));

@@ -937,11 +937,18 @@ __global__ __launch_bounds__(MP_THREADS, 2) void k_mp_merge(
 
 // single-workgroup merge for small inputs (na+nb <= FUSE_MAX): merge-path
 // thread diagonals + LDS scan; one launch, no host sync, length on device
-__global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_small(
-    const uint64_t *ak, const uint64_t *av, const int64_t *aw, int64_t na,
-    const uint64_t *bk, const uint64_t *bv, const int64_t *bw, int64_t nb,
-    uint64_t *ok, uint64_t *ov, int64_t *ow, int64_t *out_len) {
+__global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_small(MergeArgs args) {
     __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
+    const int p = blockIdx.x;
+    const uint64_t *ak = args.ak[p], *av = args.av[p];
+    const int64_t *aw = args.aw[p];
+    const int64_t na = args.na[p];
+    const uint64_t *bk = args.bk[p], *bv = args.bv[p];
+    const int64_t *bw = args.bw[p];
+    const int64_t nb = args.nb[p];
+    uint64_t *ok = args.ok[p], *ov = args.ov[p];
+    int64_t *ow = args.ow[p];
+    int64_t *out_len = args.d_len + p;
     const int tid = threadIdx.x;
     const int64_t total = na + nb;
     const int64_t items = (total + FUSE_THREADS - 1) / FUSE_THREADS;
@@ -1589,14 +1596,11 @@ dbsp_status sort_cons_small_batch(hipStream_t s, const SortArgs &args) {
     return DBSP_OK;
 }
 
-dbsp_status merge_small(hipStream_t s, const uint64_t *ak, const uint64_t *av,
-                        const int64_t *aw, int64_t na, const uint64_t *bk,
-                        const uint64_t *bv, const int64_t *bw, int64_t nb,
-                        uint64_t *ok, uint64_t *ov, int64_t *ow,
-                        int64_t *d_len) {
-    if (na + nb > 4 * FUSE_MAX) return DBSP_ERR_INVALID;
-    k_merge_small<<<1, FUSE_THREADS, 0, s>>>(ak, av, aw, na, bk, bv, bw, nb, ok,
-                                             ov, ow, d_len);
+dbsp_status merge_small_batch(hipStream_t s, const MergeArgs &args) {
+    if (args.np == 0) return DBSP_OK;
+    for (int i = 0; i < args.np; i++)
+        if (args.na[i] + args.nb[i] > 4 * FUSE_MAX) return DBSP_ERR_INVALID;
+    k_merge_small<<<dim3((uint32_t)args.np), FUSE_THREADS, 0, s>>>(args);
     return DBSP_OK;
 }
 

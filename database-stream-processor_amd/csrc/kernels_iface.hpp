@@ -15,19 +15,38 @@ struct TraceArgs {
     int64_t n[MAX_TRACE_BATCHES];
 };
 
+#define SORT_BATCH_MAX 8
 struct SortArgs {
     int nb;
-    const uint64_t *kin[4];
-    const uint64_t *vin[4];
-    const int64_t *win[4];
-    int64_t n[4];
-    uint64_t *tk[4];
-    uint64_t *tv[4];
-    int64_t *tw[4];
-    uint64_t *ok[4];
-    uint64_t *ov[4];
-    int64_t *ow[4];
+    const uint64_t *kin[SORT_BATCH_MAX];
+    const uint64_t *vin[SORT_BATCH_MAX];
+    const int64_t *win[SORT_BATCH_MAX];
+    int64_t n[SORT_BATCH_MAX];
+    uint64_t *tk[SORT_BATCH_MAX];
+    uint64_t *tv[SORT_BATCH_MAX];
+    int64_t *tw[SORT_BATCH_MAX];
+    uint64_t *ok[SORT_BATCH_MAX];
+    uint64_t *ov[SORT_BATCH_MAX];
+    int64_t *ow[SORT_BATCH_MAX];
     int64_t *d_len;  // device array, one length per batch
+};
+
+// batched single-workgroup merges (one pair per workgroup; each na+nb <= 32768)
+#define MERGE_BATCH_MAX 4
+struct MergeArgs {
+    int np;
+    const uint64_t *ak[MERGE_BATCH_MAX];
+    const uint64_t *av[MERGE_BATCH_MAX];
+    const int64_t *aw[MERGE_BATCH_MAX];
+    int64_t na[MERGE_BATCH_MAX];
+    const uint64_t *bk[MERGE_BATCH_MAX];
+    const uint64_t *bv[MERGE_BATCH_MAX];
+    const int64_t *bw[MERGE_BATCH_MAX];
+    int64_t nb[MERGE_BATCH_MAX];
+    uint64_t *ok[MERGE_BATCH_MAX];
+    uint64_t *ov[MERGE_BATCH_MAX];
+    int64_t *ow[MERGE_BATCH_MAX];
+    int64_t *d_len;
 };
 
 struct JoinCountArgs {
@@ -68,11 +87,7 @@ dbsp_status sort_cons_small_batch(hipStream_t s, const SortArgs &args);
 
 // single-workgroup merge of two consolidated batches (na+nb <= 8192):
 // one launch, no host sync; length left in *d_len (device)
-dbsp_status merge_small(hipStream_t s, const uint64_t *ak, const uint64_t *av,
-                        const int64_t *aw, int64_t na, const uint64_t *bk,
-                        const uint64_t *bv, const int64_t *bw, int64_t nb,
-                        uint64_t *ok, uint64_t *ov, int64_t *ow,
-                        int64_t *d_len);
+dbsp_status merge_small_batch(hipStream_t s, const MergeArgs &args);
 
 // up to 3 single-workgroup join count+scan plans in ONE launch (nd <= 8192
 // each): per-plan per-row/per-batch cnts, exclusive offsets, totals to
