@@ -1052,6 +1052,81 @@ static dbsp_status agg_linear_spine(dbsp_ctx *c, const DevBatch &delta,
     return DBSP_OK;
 }
 
+static bool spine_needs_merge(Spine &s) {
+    size_t m = s.batches.size();
+    return m >= 2 && s.batches[m - 1].n * 2 >= s.batches[m - 2].n;
+}
+
+// insert one delta into each of two spines, batching the two sides' pending
+// merges into single launches with one shared length sync per round
+static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
+                                      Spine &s2, DevBatch b2) {
+    ScopedTimer t(c, 1, (double)(b1.n + b2.n) * 48.0);
+    if (b1.n > 0) s1.batches.push_back(b1);
+    else free_batch(c, b1);
+    if (b2.n > 0) s2.batches.push_back(b2);
+    else free_batch(c, b2);
+    while (true) {
+        Spine *pending[2];
+        int nps = 0;
+        if (spine_needs_merge(s1)) pending[nps++] = &s1;
+        if (spine_needs_merge(s2)) pending[nps++] = &s2;
+        if (nps == 0) break;
+        MergeArgs ma{};
+        DevBatch results[2];
+        Spine *owners[2];
+        int nbatched = 0;
+        for (int i = 0; i < nps; i++) {
+            Spine &sp = *pending[i];
+            DevBatch &top = sp.batches.back();
+            DevBatch &below = sp.batches[sp.batches.size() - 2];
+            if (top.n + below.n <= 32768) {
+                DevBatch res;
+                TRY(alloc_batch(c, top.n + below.n, res));
+                int p = ma.np++;
+                ma.ak[p] = below.k; ma.av[p] = below.v; ma.aw[p] = below.w;
+                ma.na[p] = below.n;
+                ma.bk[p] = top.k; ma.bv[p] = top.v; ma.bw[p] = top.w;
+                ma.nb[p] = top.n;
+                ma.ok[p] = res.k; ma.ov[p] = res.v; ma.ow[p] = res.w;
+                results[nbatched] = res;
+                owners[nbatched] = &sp;
+                nbatched++;
+            } else {
+                DevBatch res;
+                TRY(merge_batches(c, below, top, res));
+                free_batch(c, top);
+                free_batch(c, below);
+                sp.batches.pop_back();
+                sp.batches.pop_back();
+                if (res.n > 0) sp.batches.push_back(res);
+                else free_batch(c, res);
+            }
+        }
+        if (ma.np > 0) {
+            ma.d_len = c->d_len;
+            TRY(dbspk::merge_small_batch(c->stream, ma));
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
+                                        ma.np * sizeof(int64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            for (int j = 0; j < nbatched; j++) {
+                Spine &sp = *owners[j];
+                results[j].n = c->h_len[j];
+                DevBatch top = sp.batches.back();
+                sp.batches.pop_back();
+                DevBatch below = sp.batches.back();
+                sp.batches.pop_back();
+                free_batch(c, top);
+                free_batch(c, below);
+                if (results[j].n > 0) sp.batches.push_back(results[j]);
+                else free_batch(c, results[j]);
+            }
+        }
+    }
+    return DBSP_OK;
+}
+
 // ---- q3 tick (queries/q3.rs:35-63) ----
 static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     dbsp_ctx *c = e->ctx;
@@ -1127,20 +1202,28 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 8 * sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        // all small plans emit into ONE combined raw buffer at their base
+        // offsets — the downstream consolidate then needs no concat copies
+        int64_t total_small = 0;
+        for (int i = 0; i < np; i++)
+            if (plans[i].t.nb > 0 && plans[i].small)
+                total_small += c->h_len[plans[i].slot];
+        DevBatch comb;
+        if (total_small > 0) TRY(alloc_batch(c, total_small, comb, true));
+        int64_t base = 0;
         for (int i = 0; i < np; i++) {
             Plan &pl = plans[i];
             if (pl.t.nb == 0) continue;
             if (pl.small) {
                 int64_t total = c->h_len[pl.slot];
                 if (total <= 0) continue;
-                DevBatch o;
-                TRY(alloc_batch(c, total, o, true));
                 TRY(dbspk::join_emit_prepared(c->stream, pl.delta->k,
                                               pl.delta->v, pl.delta->w,
                                               pl.delta->n, pl.t, pl.cnts,
                                               pl.offsets, total, pl.proj, 0,
-                                              o.k, o.v, o.w));
-                outs.push_back(o);
+                                              comb.k + base, comb.v + base,
+                                              comb.w + base));
+                base += total;
             } else {
                 DevBatch o;
                 TRY(dbspk::join_spine_rows(c->stream, pl.delta->k, pl.delta->v,
@@ -1149,6 +1232,12 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                 if (o.n > 0) outs.push_back(o);
                 else free_batch(c, o);
             }
+        }
+        if (total_small > 0) {
+            comb.n = total_small;
+            outs.push_back(comb);
+        } else if (comb.k) {
+            free_batch(c, comb);
         }
     }
     // Launch the output consolidate asynchronously, overlap it with the spine
@@ -1161,17 +1250,22 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     if (async_final) {
         ScopedTimer t0(c, 0, (double)cat_n * 48.0);
         DevBatch cat, scratch;
-        TRY(alloc_batch(c, cat_n, cat, true));
-        int64_t off = 0;
-        for (auto &b : outs) {
-            if (b.n == 0) continue;
-            HIP_CHECK_ST(hipMemcpyAsync(cat.k + off, b.k, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
-            HIP_CHECK_ST(hipMemcpyAsync(cat.v + off, b.v, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
-            HIP_CHECK_ST(hipMemcpyAsync(cat.w + off, b.w, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
-            off += b.n;
+        if (outs.size() == 1) {
+            cat = outs[0];  // the usual case: one combined emit buffer
+            outs.clear();
+        } else {
+            TRY(alloc_batch(c, cat_n, cat, true));
+            int64_t off = 0;
+            for (auto &b : outs) {
+                if (b.n == 0) continue;
+                HIP_CHECK_ST(hipMemcpyAsync(cat.k + off, b.k, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
+                HIP_CHECK_ST(hipMemcpyAsync(cat.v + off, b.v, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
+                HIP_CHECK_ST(hipMemcpyAsync(cat.w + off, b.w, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
+                off += b.n;
+            }
+            for (auto &b : outs) free_batch(c, b);
+            outs.clear();
         }
-        for (auto &b : outs) free_batch(c, b);
-        outs.clear();
         TRY(alloc_batch(c, cat_n, scratch, true));
         TRY(alloc_batch(c, cat_n, res));
         SortArgs sa{};
@@ -1183,8 +1277,7 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         TRY(dbspk::sort_cons_small_batch(c->stream, sa));
         // no sync yet — the spine inserts below sync the stream
     }
-    TRY(e->a_int.insert(c, dA));
-    TRY(e->p_int.insert(c, dP));
+    TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
     if (async_final) {
         HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6, sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
