@@ -1313,8 +1313,7 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     e->q8_have_prev = true;
     e->q8_s0 = s1;
     e->q8_e0 = e1;
-    TRY(e->pt_int.insert(c, dPT));
-    TRY(e->at_int.insert(c, dAT));
+    TRY(spines_insert_pair(c, e->pt_int, dPT, e->at_int, dAT));
     // map_index / map + consolidate
     DevBatch wpr, war, dWP, dWA;
     TRY(finalize_raw(c, wp_raw, wpr));
@@ -1323,11 +1322,24 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     TRY(map_sorted(c, war, 1, dWA));
     free_batch(c, wpr);
     free_batch(c, war);
+    // bilinear expansion against PREVIOUS traces (as q3): joins independent
+    // of the inserts, inserts paired into shared launches
     std::vector<DevBatch> outs;
     TRY(join_vs_spine(c, dWP, e->wa_int, DBSP_PROJ_HI_K_LO_V1RND, TUMBLE_MS, outs));
-    TRY(e->wp_int.insert(c, dWP));
     TRY(join_vs_spine(c, dWA, e->wp_int, DBSP_PROJ_HI_K_LO_V2RND, TUMBLE_MS, outs));
-    TRY(e->wa_int.insert(c, dWA));
+    if (dWP.n > 0 && dWA.n > 0) {
+        TraceArgs t{};
+        t.nb = 1;
+        t.k[0] = dWA.k; t.v[0] = dWA.v; t.w[0] = dWA.w; t.n[0] = dWA.n;
+        DevBatch o;
+        ScopedTimer timer(c, 2, (double)dWP.n * 24.0);
+        TRY(dbspk::join_spine_rows(c->stream, dWP.k, dWP.v, dWP.w, dWP.n, t,
+                                   DBSP_PROJ_HI_K_LO_V1RND, TUMBLE_MS, &o.k,
+                                   &o.v, &o.w, &o.n));
+        if (o.n > 0) outs.push_back(o);
+        else free_batch(c, o);
+    }
+    TRY(spines_insert_pair(c, e->wp_int, dWP, e->wa_int, dWA));
     free_batch(c, e->output);
     TRY(finalize_raw(c, outs, e->output));
     return DBSP_OK;
