@@ -135,6 +135,66 @@ def bench_sort(ctx, L, n, reps=3):
     }
 
 
+def bench_incremental(ctx, L, n_trace, n_delta, steps=8):
+    """C5-class incremental loop (SURVEY.md §8 config 5, i64-weight variant;
+    the f64 sum aggregate is a named next step): per step, a 10M-row sorted
+    delta joins against the 1B-row trace spine (join is linear: one probe per
+    spine batch) and is then merged into the spine under the power-of-two
+    leveling — the steady-state trace-maintenance loop at HBM scale."""
+    kt, vt, wt = sorted_unique_batch(n_trace, 7)
+    wt.fill_(1)  # trace weights +1 so delta inserts never fully cancel
+    spine = [(as_batch(kt, vt, wt), None)]  # (batch, dbsp-owned BatchStruct?)
+    keep_alive = [(kt, vt, wt)]
+    gap = max(2, (9 * n_trace) // (2 * n_delta))
+    step_ms = []
+    joined_total = 0
+    for it in range(steps):
+        kd, vd, wd = sorted_unique_batch(n_delta, 100 + it, max_gap=gap)
+        wd.fill_(1)
+        keep_alive.append((kd, vd, wd))
+        d = as_batch(kd, vd, wd)
+        ctx.sync()
+        t0 = time.perf_counter()
+        # join delta against every spine batch (linear in the trace)
+        outs = []
+        for b, _ in spine:
+            o = BatchStruct()
+            assert L.dbsp_join(ctx._h, ctypes.byref(d), ctypes.byref(b), 0, 0,
+                               ctypes.byref(o)) == 0
+            outs.append(o)
+        joined = sum(o.len for o in outs)
+        for o in outs:
+            ctx.free_batch(o)
+        # insert the delta into the spine (power-of-two leveling)
+        spine.append((d, None))
+        while len(spine) >= 2 and spine[-1][0].len * 2 >= spine[-2][0].len:
+            (b_top, own_top) = spine.pop()
+            (b_below, own_below) = spine.pop()
+            m = BatchStruct()
+            assert L.dbsp_merge(ctx._h, ctypes.byref(b_below),
+                                ctypes.byref(b_top), ctypes.byref(m)) == 0
+            for bb, own in ((b_top, own_top), (b_below, own_below)):
+                if own is not None:
+                    ctx.free_batch(bb)
+            spine.append((m, True))
+        ctx.sync()
+        step_ms.append((time.perf_counter() - t0) * 1e3)
+        joined_total += joined
+    trace_rows = sum(b.len for b, _ in spine)
+    for b, own in spine:
+        if own is not None:
+            ctx.free_batch(b)
+    avg = sum(step_ms[1:]) / max(1, len(step_ms) - 1)
+    return {
+        "primitive": "incremental join + spine maintenance (C5-class, i64)",
+        "trace_rows_initial": n_trace, "delta_rows": n_delta, "steps": steps,
+        "trace_rows_final": trace_rows, "join_matches_total": joined_total,
+        "ms_per_step": [round(x, 2) for x in step_ms],
+        "avg_ms_per_step": round(avg, 2),
+        "delta_rows_per_s": round(n_delta / (avg / 1e3), 0),
+    }
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--rows", type=int, default=500_000_000,
@@ -147,6 +207,7 @@ def main():
     print(json.dumps(bench_merge(ctx, L, args.rows)))
     print(json.dumps(bench_join(ctx, L, 2 * args.rows, args.delta)))
     print(json.dumps(bench_sort(ctx, L, args.sort_rows)))
+    print(json.dumps(bench_incremental(ctx, L, 2 * args.rows, args.delta)))
     ctx.close()
 
 
