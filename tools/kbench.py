@@ -68,6 +68,7 @@ def bench_merge(ctx, L, n_per_side, reps=3):
     algo_gb = 24.0 * (2 * n_per_side + n_out) / 1e9
     return {
         "primitive": "dbsp_merge (merge-path trace merge)",
+        "rep_ms": [round(t * 1e3, 1) for t in times],
         "rows_in": 2 * n_per_side, "rows_out": n_out,
         "ms": round(dt * 1e3, 2), "algo_GB": round(algo_gb, 2),
         "achieved_GBs": round(algo_gb / dt, 1),
