@@ -779,7 +779,7 @@ __global__ void k_join_emit(const uint64_t *dk, const uint64_t *dv,
 // (measured 53 GB/s).
 // ---------------------------------------------------------------------------
 
-#define MP_TILE 4096
+#define MP_TILE 2048
 #define MP_THREADS 256
 #define MP_ITEMS (MP_TILE / MP_THREADS)  // 16 diagonals per thread
 
@@ -923,6 +923,152 @@ __global__ __launch_bounds__(MP_THREADS, 2) void k_mp_merge(
             if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
                 int64_t s = lw[i] + lw[naL + j];
                 if (s != 0) { ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = s; gpos++; }
+                i++; j++;
+            } else if (j >= be ||
+                       (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
+                ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = lw[i]; gpos++; i++;
+            } else {
+                ok[gpos] = lk[naL + j]; ov[gpos] = lv[naL + j]; ow[gpos] = lw[naL + j];
+                gpos++; j++;
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// single-pass large merge: decoupled-lookback offsets replace the
+// count->scan->emit pipeline (removes the 16 B/row count re-read, the device
+// scan, and the mid-pipeline host sync).  Blocks take a ticket (atomic) so
+// virtual block order equals launch order (forward progress for lookback);
+// per-block state is one 8-byte {flag,value} granule written/read with
+// agent-scope relaxed atomics (single-granule R2 form of the CDNA4
+// inter-workgroup recipe — no fences needed for an 8-byte payload-is-flag).
+// State layout: state[0] = ticket; state[1+vb] = granule
+//   granule = value<<2 | flag;  flag: 0 invalid, 1 aggregate, 2 prefix
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((address_space(1))) unsigned long long gu64_t;
+
+__global__ __launch_bounds__(MP_THREADS, 3) void k_mp_merge_onepass(
+    const uint64_t *ak, const uint64_t *av, const int64_t *aw, int64_t na,
+    const uint64_t *bk, const uint64_t *bv, const int64_t *bw, int64_t nb,
+    const int64_t *pa, const int64_t *pb, unsigned long long *state,
+    int64_t nblocks, uint64_t *ok, uint64_t *ov, int64_t *ow) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    uint64_t *lk = (uint64_t *)smem;
+    uint64_t *lv = lk + (MP_TILE + 2);
+    int64_t *lw = (int64_t *)(lv + (MP_TILE + 2));
+    __shared__ uint32_t wt[MP_THREADS / WAVE + 1];
+    __shared__ unsigned long long sh_vb;
+    __shared__ unsigned long long sh_prefix;
+    const int tid = threadIdx.x;
+    // ticket: virtual block id in launch order
+    if (tid == 0) sh_vb = atomicAdd(state, 1ull);
+    __syncthreads();
+    const int64_t vb = (int64_t)sh_vb;
+    const int64_t pa0 = pa[vb], pa1 = pa[vb + 1];
+    const int64_t pb0 = pb[vb], pb1 = pb[vb + 1];
+    const int64_t naL = pa1 - pa0, nbL = pb1 - pb0;
+    const int64_t totL = naL + nbL;
+    for (int64_t i = tid; i < naL; i += MP_THREADS) {
+        lk[i] = ak[pa0 + i];
+        lv[i] = av[pa0 + i];
+        lw[i] = aw[pa0 + i];
+    }
+    for (int64_t i = tid; i < nbL; i += MP_THREADS) {
+        lk[naL + i] = bk[pb0 + i];
+        lv[naL + i] = bv[pb0 + i];
+        lw[naL + i] = bw[pb0 + i];
+    }
+    __syncthreads();
+    const int64_t items = (totL + MP_THREADS - 1) / MP_THREADS;
+    int64_t d0 = min((int64_t)tid * items, totL);
+    int64_t d1 = min(d0 + items, totL);
+    int64_t ai, bi, ae, be;
+    merge_path_lds(lk, lv, naL, nbL, d0, ai, bi);
+    adjust_split_lds(lk, lv, naL, nbL, ai, bi);
+    merge_path_lds(lk, lv, naL, nbL, d1, ae, be);
+    adjust_split_lds(lk, lv, naL, nbL, ae, be);
+    // count from LDS
+    uint32_t cnt = 0;
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
+                if (lw[i] + lw[naL + j] != 0) cnt++;
+                i++; j++;
+            } else if (j >= be ||
+                       (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
+                cnt++; i++;
+            } else {
+                cnt++; j++;
+            }
+        }
+    }
+    // block-exclusive scan of thread counts
+    uint32_t thread_off;
+    uint32_t block_cnt;
+    {
+        uint32_t v = cnt;
+        for (int d = 1; d < WAVE; d <<= 1) {
+            uint32_t up = __shfl_up(v, d, WAVE);
+            if ((tid & (WAVE - 1)) >= d) v += up;
+        }
+        if ((tid & (WAVE - 1)) == WAVE - 1) wt[tid / WAVE] = v;
+        __syncthreads();
+        if (tid == 0) {
+            uint32_t acc = 0;
+            for (int w = 0; w < MP_THREADS / WAVE; w++) {
+                uint32_t t = wt[w];
+                wt[w] = acc;
+                acc += t;
+            }
+            wt[MP_THREADS / WAVE] = acc;
+        }
+        __syncthreads();
+        thread_off = wt[tid / WAVE] + (v - cnt);
+        block_cnt = wt[MP_THREADS / WAVE];
+    }
+    // decoupled lookback (lane 0 of the block)
+    if (tid == 0) {
+        gu64_t *g = (gu64_t *)(state + 2);
+        // publish aggregate
+        __hip_atomic_store(&g[vb], ((unsigned long long)block_cnt << 2) | 1ull,
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        unsigned long long prefix = 0;
+        unsigned spins = 0;
+        for (int64_t p = vb - 1; p >= 0;) {
+            unsigned long long e =
+                __hip_atomic_load(&g[p], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            unsigned long long flag = e & 3ull;
+            if (flag == 2ull) {
+                prefix += e >> 2;
+                break;
+            } else if (flag == 1ull) {
+                prefix += e >> 2;
+                p--;
+            } else {
+                __builtin_amdgcn_s_sleep(1);
+                if (++spins > (1u << 24)) {  // bounded spin: poison, don't hang
+                    __hip_atomic_store((gu64_t *)(state + 1), 1ull,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    break;
+                }
+            }
+        }
+        __hip_atomic_store(&g[vb],
+                           ((prefix + block_cnt) << 2) | 2ull,
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        sh_prefix = prefix;
+    }
+    __syncthreads();
+    uint64_t gpos = (uint64_t)sh_prefix + thread_off;
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
+                int64_t sum = lw[i] + lw[naL + j];
+                if (sum != 0) { ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = sum; gpos++; }
                 i++; j++;
             } else if (j >= be ||
                        (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
@@ -1559,32 +1705,39 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
     }
     int64_t nblocks = ceil_div(total, MP_TILE);
     int64_t *pa, *pb;
-    uint64_t *counts;
+    unsigned long long *state;
     HIP_CHECK(hipMallocAsync(&pa, (nblocks + 1) * sizeof(int64_t), s));
     HIP_CHECK(hipMallocAsync(&pb, (nblocks + 1) * sizeof(int64_t), s));
-    HIP_CHECK(hipMallocAsync(&counts, (nblocks + 1) * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&state, (nblocks + 2) * sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(state, 0, (nblocks + 2) * sizeof(uint64_t), s));
     k_mp_partition<<<grid_for(nblocks + 1), BLK, 0, s>>>(ak, av, na, bk, bv, nb,
                                                          nblocks, pa, pb);
-    const size_t smem_count = 2 * (MP_TILE + 2) * sizeof(uint64_t);
-    k_mp_merge<false><<<dim3((uint32_t)nblocks), MP_THREADS, smem_count, s>>>(
-        ak, av, aw, na, bk, bv, bw, nb, pa, pb, counts, nullptr, nullptr,
-        nullptr);
-    uint64_t nout = 0;
-    dbsp_status st = scan_exclusive(s, counts, counts, nblocks, &nout);
-    if (st != DBSP_OK) return st;
     uint64_t *rk, *rv;
     int64_t *rw;
-    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
-    const size_t smem_emit = 3 * (MP_TILE + 2) * sizeof(uint64_t);
-    if (nout > 0)
-        k_mp_merge<true><<<dim3((uint32_t)nblocks), MP_THREADS, smem_emit, s>>>(
-            ak, av, aw, na, bk, bv, bw, nb, pa, pb, counts, rk, rv, rw);
+    HIP_CHECK(hipMallocAsync(&rk, total * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rv, total * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rw, total * sizeof(int64_t) + 8, s));
+    const size_t smem = 3 * (MP_TILE + 2) * sizeof(uint64_t);
+    k_mp_merge_onepass<<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
+        ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
+    // total = inclusive prefix of the last virtual block; error word poisoned
+    // if a lookback spun out (never observed; bounded to avoid hangs)
+    unsigned long long h_state[2];
+    HIP_CHECK(hipMemcpyAsync(&h_state[0], state + 1, sizeof(uint64_t),
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipMemcpyAsync(&h_state[1], state + 2 + (nblocks - 1),
+                             sizeof(uint64_t), hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
     HIP_CHECK(hipFreeAsync(pa, s));
     HIP_CHECK(hipFreeAsync(pb, s));
-    HIP_CHECK(hipFreeAsync(counts, s));
-    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
+    HIP_CHECK(hipFreeAsync(state, s));
+    if (h_state[0] != 0 || (h_state[1] & 3ull) != 2ull) {
+        hipFreeAsync(rk, s);
+        hipFreeAsync(rv, s);
+        hipFreeAsync(rw, s);
+        return DBSP_ERR_INTERNAL;
+    }
+    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)(h_state[1] >> 2);
     return DBSP_OK;
 }
 
