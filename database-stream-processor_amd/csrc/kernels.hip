@@ -822,118 +822,6 @@ __device__ inline void adjust_split_lds(const uint64_t *lk, const uint64_t *lv,
         bi++;
 }
 
-template <bool EMIT>
-__global__ __launch_bounds__(MP_THREADS, 2) void k_mp_merge(
-    const uint64_t *ak, const uint64_t *av, const int64_t *aw, int64_t na,
-    const uint64_t *bk, const uint64_t *bv, const int64_t *bw, int64_t nb,
-    const int64_t *pa, const int64_t *pb,
-    uint64_t *counts,  // COUNT: per-block totals out; EMIT: scanned offsets in
-    uint64_t *ok, uint64_t *ov, int64_t *ow) {
-    extern __shared__ __attribute__((aligned(16))) char smem[];
-    uint64_t *lk = (uint64_t *)smem;          // tile + 2: the split adjustment
-    uint64_t *lv = lk + (MP_TILE + 2);        // can grow a block by one row
-    int64_t *lw = (int64_t *)(lv + (MP_TILE + 2));  // staged only when EMIT
-    __shared__ uint32_t wt[MP_THREADS / WAVE + 1];
-    const int tid = threadIdx.x;
-    const int64_t blk = blockIdx.x;
-    const int64_t pa0 = pa[blk], pa1 = pa[blk + 1];
-    const int64_t pb0 = pb[blk], pb1 = pb[blk + 1];
-    const int64_t naL = pa1 - pa0, nbL = pb1 - pb0;
-    const int64_t totL = naL + nbL;
-    // stage (coalesced 8B per lane)
-    for (int64_t i = tid; i < naL; i += MP_THREADS) {
-        lk[i] = ak[pa0 + i];
-        lv[i] = av[pa0 + i];
-        if (EMIT) lw[i] = aw[pa0 + i];
-    }
-    for (int64_t i = tid; i < nbL; i += MP_THREADS) {
-        lk[naL + i] = bk[pb0 + i];
-        lv[naL + i] = bv[pb0 + i];
-        if (EMIT) lw[naL + i] = bw[pb0 + i];
-    }
-    __syncthreads();
-    // per-thread diagonals within the tile
-    const int64_t items = (totL + MP_THREADS - 1) / MP_THREADS;
-    int64_t d0 = min((int64_t)tid * items, totL);
-    int64_t d1 = min(d0 + items, totL);
-    int64_t ai, bi, ae, be;
-    merge_path_lds(lk, lv, naL, nbL, d0, ai, bi);
-    adjust_split_lds(lk, lv, naL, nbL, ai, bi);
-    merge_path_lds(lk, lv, naL, nbL, d1, ae, be);
-    adjust_split_lds(lk, lv, naL, nbL, ae, be);
-    // count walk (weights touched only on equal pairs in COUNT: rare, global)
-    uint32_t cnt = 0;
-    {
-        int64_t i = ai, j = bi;
-        while (i < ae || j < be) {
-            if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
-                int64_t s = EMIT ? (lw[i] + lw[naL + j])
-                                 : (aw[pa0 + i] + bw[pb0 + j]);
-                if (s != 0) cnt++;
-                i++; j++;
-            } else if (j >= be ||
-                       (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
-                cnt++; i++;
-            } else {
-                cnt++; j++;
-            }
-        }
-    }
-    if (!EMIT) {
-        // block total
-        uint32_t v = cnt;
-        for (int d = 1; d < WAVE; d <<= 1) {
-            uint32_t up = __shfl_up(v, d, WAVE);
-            if ((tid & (WAVE - 1)) >= d) v += up;
-        }
-        if ((tid & (WAVE - 1)) == WAVE - 1) wt[tid / WAVE] = v;
-        __syncthreads();
-        if (tid == 0) {
-            uint32_t acc = 0;
-            for (int w = 0; w < MP_THREADS / WAVE; w++) acc += wt[w];
-            counts[blk] = acc;
-        }
-        return;
-    }
-    // EMIT: exclusive scan of thread counts -> local offsets
-    uint32_t off;
-    {
-        uint32_t v = cnt;
-        for (int d = 1; d < WAVE; d <<= 1) {
-            uint32_t up = __shfl_up(v, d, WAVE);
-            if ((tid & (WAVE - 1)) >= d) v += up;
-        }
-        if ((tid & (WAVE - 1)) == WAVE - 1) wt[tid / WAVE] = v;
-        __syncthreads();
-        if (tid == 0) {
-            uint32_t acc = 0;
-            for (int w = 0; w < MP_THREADS / WAVE; w++) {
-                uint32_t t = wt[w];
-                wt[w] = acc;
-                acc += t;
-            }
-        }
-        __syncthreads();
-        off = wt[tid / WAVE] + (v - cnt);
-    }
-    uint64_t gpos = counts[blk] + off;
-    {
-        int64_t i = ai, j = bi;
-        while (i < ae || j < be) {
-            if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
-                int64_t s = lw[i] + lw[naL + j];
-                if (s != 0) { ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = s; gpos++; }
-                i++; j++;
-            } else if (j >= be ||
-                       (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
-                ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = lw[i]; gpos++; i++;
-            } else {
-                ok[gpos] = lk[naL + j]; ov[gpos] = lv[naL + j]; ow[gpos] = lw[naL + j];
-                gpos++; j++;
-            }
-        }
-    }
-}
 
 // ---------------------------------------------------------------------------
 // single-pass large merge: decoupled-lookback offsets replace the
