@@ -40,6 +40,24 @@ def generate_events(total, seed=1):
     return gen.generate(total, seed=seed)
 
 
+def generate_rank_slices(n_ticks, tick, world, rank, seed=1):
+    """Each rank keeps only its own tick-sized slice of the global stream
+    (tick t covers global events [t*tick*world, (t+1)*tick*world); rank r owns
+    the r-th tick-sized slice).  Generation itself is sequential and cheap;
+    storing the full stream at N=8 would cost ~5 GB/rank."""
+    import numpy as np
+    from dbsp_amd import gen
+    if world == 1:
+        return gen.generate(n_ticks * tick, seed=seed)
+    st = gen.Stream(seed=seed)
+    out = []
+    for _ in range(n_ticks):
+        chunk = st.next(tick * world)
+        out.append(chunk[rank * tick:(rank + 1) * tick].copy())
+    st.close()
+    return np.concatenate(out)
+
+
 def run_engine(ctx_dev, query, events, n_ticks, tick, world, rank, nccl_id,
                timed_ticks, dist=None):
     """Stage events; run warmup then timed ticks; return (elapsed_s, engine)."""
@@ -57,9 +75,8 @@ def run_engine(ctx_dev, query, events, n_ticks, tick, world, rank, nccl_id,
     eng.stage(events)
 
     def tick_range(t):
-        # global stream laid out tick-major: tick t covers
-        # [t*tick*world, (t+1)*tick*world); rank r takes its tick-sized slice
-        base = t * tick * world + rank * tick
+        # events are the rank's own pre-sliced stream (tick-major)
+        base = t * tick
         return base, base + tick
 
     warmup_ticks = n_ticks - timed_ticks
@@ -182,8 +199,8 @@ def main():
         nccl_id = t.numpy()
 
     n_ticks = args.warmup + args.steps
-    total_events = n_ticks * args.tick * world
-    events = generate_events(total_events, seed=args.seed)
+    events = generate_rank_slices(n_ticks, args.tick, world, rank,
+                                  seed=args.seed)
 
     elapsed, eng, ctx = run_engine(local_rank, args.query, events, n_ticks,
                                    args.tick, world, rank, nccl_id,
