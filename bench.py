@@ -174,8 +174,6 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    if world > 1 and args.query not in (3,):
-        raise SystemExit("multi-GPU sharding is implemented for q3 this round")
 
     dist = None
     nccl_id = None

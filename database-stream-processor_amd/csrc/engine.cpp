@@ -660,10 +660,32 @@ extern "C" dbsp_status dbsp_comm_alltoallv(dbsp_ctx *c, const dbsp_batch *send,
     return DBSP_OK;
 }
 
+static inline bool sharding_on(dbsp_ctx *c) {
+    return c->world > 1 || (c->force_shard && c->comm);
+}
+
+// global max over ranks (the multi-worker watermark exchange,
+// watermark.rs:47-73, as an RCCL allreduce of one u64)
+static dbsp_status allreduce_max_u64(dbsp_ctx *c, uint64_t *h_val) {
+    if (!sharding_on(c)) return DBSP_OK;
+    uint64_t *d;
+    HIP_CHECK_ST(hipMallocAsync(&d, sizeof(uint64_t), c->stream));
+    HIP_CHECK_ST(hipMemcpyAsync(d, h_val, sizeof(uint64_t),
+                                hipMemcpyHostToDevice, c->stream));
+    if (ncclAllReduce(d, d, 1, ncclUint64, ncclMax, c->comm, c->stream) !=
+        ncclSuccess)
+        return DBSP_ERR_INTERNAL;
+    HIP_CHECK_ST(hipMemcpyAsync(h_val, d, sizeof(uint64_t),
+                                hipMemcpyDeviceToHost, c->stream));
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    HIP_CHECK_ST(hipFreeAsync(d, c->stream));
+    return DBSP_OK;
+}
+
 // shard + exchange + rebuild: the full shard() operator
 // (shard.rs:88-199: hash-split, exchange, re-consolidate)
 static dbsp_status shard_exchange(dbsp_ctx *c, DevBatch local, DevBatch &out) {
-    if (c->world <= 1 && !(c->force_shard && c->comm)) {
+    if (!sharding_on(c)) {
         out = local;
         return DBSP_OK;
     }
@@ -718,8 +740,7 @@ extern "C" dbsp_status dbsp_engine_create(dbsp_engine **out, dbsp_ctx *ctx,
                                           int query, int rank, int world) {
     if (query != 0 && query != 3 && query != 5 && query != 8)
         return DBSP_ERR_INVALID;
-    if (world > 1 && query != 3 && query != 0)
-        return DBSP_ERR_INVALID;  // q5/q8 multi-rank: next round (watermark allreduce)
+
     dbsp_engine *e = new dbsp_engine();
     e->ctx = ctx;
     e->query = query;
@@ -856,7 +877,7 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
             free_batch(c, raw1);
     }
     // worker sharding: co-locate keys across ranks (shard.rs:88)
-    if (e->world > 1 || (c->force_shard && c->comm)) {
+    if (sharding_on(c)) {
         DevBatch s0;
         TRY(shard_exchange(c, d0, s0));
         d0 = s0;
@@ -1300,7 +1321,9 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     uint64_t lk = 0;
     bool has = false;
     TRY(last_key(c, dAT, &lk, &has));
-    if (has) e->q8_wm = std::max(e->q8_wm, lk - TUMBLE_MS);
+    uint64_t gmax = has ? lk : 0;
+    TRY(allreduce_max_u64(c, &gmax));
+    if (gmax > 0) e->q8_wm = std::max(e->q8_wm, gmax - TUMBLE_MS);
     uint64_t rounded = e->q8_wm - e->q8_wm % TUMBLE_MS;
     uint64_t s1 = rounded >= TUMBLE_MS ? rounded - TUMBLE_MS : 0;
     uint64_t e1 = rounded;
@@ -1322,6 +1345,13 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     TRY(map_sorted(c, war, 1, dWA));
     free_batch(c, wpr);
     free_batch(c, war);
+    if (sharding_on(c)) {
+        DevBatch t1, t2;
+        TRY(shard_exchange(c, dWP, t1));
+        TRY(shard_exchange(c, dWA, t2));
+        dWP = t1;
+        dWA = t2;
+    }
     // bilinear expansion against PREVIOUS traces (as q3): joins independent
     // of the inserts, inserts paired into shared launches
     std::vector<DevBatch> outs;
@@ -1354,7 +1384,9 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     uint64_t lk = 0;
     bool has = false;
     TRY(last_key(c, dBT, &lk, &has));
-    if (has) e->q5_wm = std::max(e->q5_wm, lk - WM_LAG_MS);
+    uint64_t gmax = has ? lk : 0;
+    TRY(allreduce_max_u64(c, &gmax));
+    if (gmax > 0) e->q5_wm = std::max(e->q5_wm, gmax - WM_LAG_MS);
     uint64_t rounded = e->q5_wm - e->q5_wm % TUMBLE_MS;
     uint64_t s1 = rounded >= WIDTH_MS ? rounded - WIDTH_MS : 0;
     uint64_t e1 = rounded;
@@ -1369,6 +1401,11 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     TRY(finalize_raw(c, wb_raw, wbr));
     TRY(map_sorted(c, wbr, 1, dWB));  // (time,auction) -> (auction,()); weigh(|_|1)
     free_batch(c, wbr);
+    if (sharding_on(c)) {
+        DevBatch t1;
+        TRY(shard_exchange(c, dWB, t1));
+        dWB = t1;
+    }
     // aggregate_linear: input trace includes this tick (trace.rs TraceAppend)
     DevBatch wb_copy;
     TRY(copy_batch(c, dWB, wb_copy));
@@ -1379,6 +1416,11 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     // max side (tiny; consolidated single-batch traces)
     DevBatch dMaxIn;
     TRY(map_sorted(c, dCounts, 2, dMaxIn));
+    if (sharding_on(c)) {
+        DevBatch t1;
+        TRY(shard_exchange(c, dMaxIn, t1));  // unit key () lives on one rank
+        dMaxIn = t1;
+    }
     DevBatch dMaxOut;
     if (dMaxIn.n > 0) {
         DevBatch m;
@@ -1408,6 +1450,13 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     TRY(map_sorted(c, dMaxOut, 1, dMaxZ));  // ((),m) -> (m,())
     free_batch(c, dMaxOut);
     TRY(map_sorted(c, dCounts, 3, dBC));    // (auction,count) -> (count,auction)
+    if (sharding_on(c)) {
+        DevBatch t1, t2;
+        TRY(shard_exchange(c, dMaxZ, t1));
+        TRY(shard_exchange(c, dBC, t2));
+        dMaxZ = t1;
+        dBC = t2;
+    }
     // final incremental join (q5.rs:118-120)
     std::vector<DevBatch> outs;
     TRY(join_vs_spine(c, dMaxZ, e->bc_int, DBSP_PROJ_HI_V2_LO_K, 0, outs));

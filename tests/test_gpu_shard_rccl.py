@@ -57,15 +57,17 @@ def test_alltoallv_self_roundtrip():
     ctx.close()
 
 
-def test_q3_engine_with_forced_shard_exchange():
-    """Full q3 through the partition + RCCL alltoallv + re-consolidate path
-    (world=1 self-exchange), tick-by-tick parity vs the oracle."""
+@pytest.mark.parametrize("query", [3, 5, 8])
+def test_engine_with_forced_shard_exchange(query):
+    """Full q3/q5/q8 through the partition + RCCL alltoallv + re-consolidate
+    path, including the watermark allreduce and the aggregate/join reshard
+    points (world=1 self-exchange), tick-by-tick parity vs the oracle."""
     os.environ["DBSP_FORCE_SHARD"] = "1"
     try:
         from dbsp_amd.engine import Engine
         ctx, _ = _make_ctx_with_comm()
-        eng = Engine(ctx, query=3)
-        q = oracle.Query(3)
+        eng = Engine(ctx, query=query)
+        q = oracle.Query(query)
         evs = gen.generate(50_000, seed=23)
         eng.stage(evs)
         for lo in range(0, len(evs), 10_000):
@@ -73,7 +75,7 @@ def test_q3_engine_with_forced_shard_exchange():
             eng.step_staged(lo, hi)
             got = eng.output()
             exp = q.step(evs[lo:hi])
-            assert zset(got) == zset(exp), f"tick [{lo},{hi})"
+            assert zset(got) == zset(exp), f"q{query} tick [{lo},{hi})"
         eng.close()
         ctx.close()
     finally:
