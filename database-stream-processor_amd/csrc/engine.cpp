@@ -574,6 +574,156 @@ extern "C" dbsp_status dbsp_shard_partition(dbsp_ctx *c, const dbsp_batch *in,
     return DBSP_OK;
 }
 
+// ---- f64-weight C ABI (config C5) ----
+
+extern "C" dbsp_status dbsp_sort_consolidate_f64(dbsp_ctx *c,
+                                                 const uint64_t *k_in,
+                                                 const uint64_t *v_in,
+                                                 const double *w_in, int64_t n,
+                                                 dbsp_batch *out) {
+    DevBatch raw;
+    TRY(alloc_batch(c, n > 0 ? n : 1, raw, true));
+    if (n > 0) {
+        HIP_CHECK_ST(hipMemcpyAsync(raw.k, k_in, n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(raw.v, v_in, n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(raw.w, w_in, n * 8, hipMemcpyDeviceToDevice, c->stream));
+    }
+    raw.n = n;
+    if (n == 0) {
+        out->k = nullptr; out->v = nullptr; out->w = nullptr; out->len = 0;
+        return DBSP_OK;
+    }
+    if (n <= 8192) {
+        DevBatch scratch, res;
+        TRY(alloc_batch(c, n, scratch, true));
+        TRY(alloc_batch(c, n, res));
+        SortArgs sa{};
+        sa.nb = 1;
+        sa.kin[0] = raw.k; sa.vin[0] = raw.v; sa.win[0] = raw.w; sa.n[0] = n;
+        sa.tk[0] = scratch.k; sa.tv[0] = scratch.v; sa.tw[0] = scratch.w;
+        sa.ok[0] = res.k; sa.ov[0] = res.v; sa.ow[0] = res.w;
+        sa.d_len = c->d_len;
+        TRY(dbspk::sort_cons_small_batch_f64(c->stream, sa));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, sizeof(int64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        out->k = res.k; out->v = res.v; out->w = res.w; out->len = *c->h_len;
+        return DBSP_OK;
+    }
+    // big path: the radix sort moves the weight column bitwise; the
+    // consolidate uses the deterministic segmented tree
+    DevBatch scratch;
+    TRY(alloc_batch(c, n, scratch, true));
+    bool in_scratch = false;
+    TRY(dbspk::sort_rows(c->stream, raw.k, raw.v, raw.w, n, scratch.k,
+                         scratch.v, scratch.w, &in_scratch));
+    DevBatch &sorted = in_scratch ? scratch : raw;
+    uint64_t *rk, *rv;
+    double *rw;
+    int64_t nout = 0;
+    TRY(dbspk::consolidate_sorted_f64(c->stream, sorted.k, sorted.v,
+                                      (double *)sorted.w, n, &rk, &rv, &rw,
+                                      &nout));
+    out->k = rk; out->v = rv; out->w = (int64_t *)rw; out->len = nout;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_merge_f64(dbsp_ctx *c, const dbsp_batch *a,
+                                      const dbsp_batch *b, dbsp_batch *out) {
+    if (a->len + b->len <= 32768) {
+        DevBatch res;
+        TRY(alloc_batch(c, a->len + b->len, res));
+        MergeArgs ma{};
+        ma.np = 1;
+        ma.ak[0] = a->k; ma.av[0] = a->v; ma.aw[0] = a->w; ma.na[0] = a->len;
+        ma.bk[0] = b->k; ma.bv[0] = b->v; ma.bw[0] = b->w; ma.nb[0] = b->len;
+        ma.ok[0] = res.k; ma.ov[0] = res.v; ma.ow[0] = res.w;
+        ma.d_len = c->d_len;
+        TRY(dbspk::merge_small_batch_f64(c->stream, ma));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, sizeof(int64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        out->k = res.k; out->v = res.v; out->w = res.w; out->len = *c->h_len;
+        return DBSP_OK;
+    }
+    uint64_t *rk, *rv;
+    double *rw;
+    int64_t nout = 0;
+    TRY(dbspk::merge_rows_f64(c->stream, a->k, a->v, (const double *)a->w,
+                              a->len, b->k, b->v, (const double *)b->w, b->len,
+                              &rk, &rv, &rw, &nout));
+    out->k = rk; out->v = rv; out->w = (int64_t *)rw; out->len = nout;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_weigh_f64(dbsp_ctx *c, const dbsp_batch *in,
+                                      dbsp_batch *out) {
+    DevBatch res;
+    TRY(alloc_batch(c, in->len > 0 ? in->len : 1, res));
+    TRY(dbspk::map_rows(c->stream, in->k, in->v, in->w, in->len, 4, res.k,
+                        res.v, res.w));
+    out->k = res.k; out->v = res.v; out->w = res.w; out->len = in->len;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_agg_linear_upsert_f64(dbsp_ctx *c,
+                                                  const uint64_t *delta_keys,
+                                                  int64_t nd,
+                                                  const dbsp_batch *in_trace,
+                                                  const dbsp_batch *out_trace,
+                                                  dbsp_batch *out) {
+    if (nd == 0) {
+        out->k = nullptr; out->v = nullptr; out->w = nullptr; out->len = 0;
+        return DBSP_OK;
+    }
+    double *acc;
+    HIP_CHECK_ST(hipMallocAsync(&acc, nd * 8 + 8, c->stream));
+    HIP_CHECK_ST(hipMemsetAsync(acc, 0, nd * 8, c->stream));
+    TRY(dbspk::agg_sum_batch_f64(c->stream, delta_keys, nd, in_trace->k,
+                                 (const double *)in_trace->w, in_trace->len,
+                                 acc));
+    DevBatch ins;
+    TRY(alloc_batch(c, nd, ins, true));
+    int64_t n_ins = 0;
+    TRY(dbspk::emit_nonzero_f64(c->stream, delta_keys, acc, nd, ins.k, ins.v,
+                                ins.w, &n_ins));
+    ins.n = n_ins;
+    HIP_CHECK_ST(hipFreeAsync(acc, c->stream));
+    // retractions against the (i64-weighted) output trace
+    DevBatch retr;
+    TRY(dbspk::agg_linear_upsert_rows(c->stream, delta_keys, nd, nullptr,
+                                      nullptr, nullptr, 0, out_trace->k,
+                                      out_trace->v, out_trace->w,
+                                      out_trace->len, &retr.k, &retr.v,
+                                      &retr.w, &retr.n));
+    DevBatch res;
+    TRY(alloc_batch(c, ins.n + retr.n, res));
+    res.n = 0;
+    if (ins.n > 0) {
+        HIP_CHECK_ST(hipMemcpyAsync(res.k, ins.k, ins.n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(res.v, ins.v, ins.n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(res.w, ins.w, ins.n * 8, hipMemcpyDeviceToDevice, c->stream));
+    }
+    if (retr.n > 0) {
+        HIP_CHECK_ST(hipMemcpyAsync(res.k + ins.n, retr.k, retr.n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(res.v + ins.n, retr.v, retr.n * 8, hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(res.w + ins.n, retr.w, retr.n * 8, hipMemcpyDeviceToDevice, c->stream));
+    }
+    res.n = ins.n + retr.n;
+    free_batch(c, ins);
+    free_batch(c, retr);
+    TRY(dbsp_ctx_sync(c));
+    out->k = res.k; out->v = res.v; out->w = res.w; out->len = res.n;
+    return DBSP_OK;
+}
+
+extern "C" dbsp_status dbsp_unique_keys(dbsp_ctx *c, const dbsp_batch *in,
+                                        uint64_t **out_keys, int64_t *n_out) {
+    TRY(dbspk::unique_keys(c->stream, in->k, in->len, out_keys, n_out));
+    TRY(dbsp_ctx_sync(c));
+    return DBSP_OK;
+}
+
 extern "C" uint64_t dbsp_xxh3_u64(uint64_t key, uint64_t seed) {
     return dbspk::host_xxh3_u64(key, seed);
 }

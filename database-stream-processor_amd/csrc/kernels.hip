@@ -392,18 +392,19 @@ __device__ inline uint32_t fuse_scan(uint32_t thread_sum, uint32_t *wave_tot,
     return r;
 }
 
+template <typename W>
 __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs args) {
     const int batch = blockIdx.x;
     const uint64_t *kin = args.kin[batch];
     const uint64_t *vin = args.vin[batch];
-    const int64_t *win = args.win[batch];
+    const W *win = (const W *)args.win[batch];
     const int64_t n = args.n[batch];
     uint64_t *tk = args.tk[batch];   // scratch (>= n rows)
     uint64_t *tv = args.tv[batch];
-    int64_t *tw = args.tw[batch];
+    W *tw = (W *)args.tw[batch];
     uint64_t *ok = args.ok[batch];   // output (cap >= n rows)
     uint64_t *ov = args.ov[batch];
-    int64_t *ow = args.ow[batch];
+    W *ow = (W *)args.ow[batch];
     int64_t *out_len = args.d_len + batch;
     __shared__ uint32_t bufA[FUSE_MAX];
     __shared__ uint32_t bufB[FUSE_MAX];
@@ -547,27 +548,72 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
             if (i < n) dst[i] = hb - 1;  // segment of element i
         }
     }
-    for (uint32_t i = tid; i < nseg; i += FUSE_THREADS) tw[i] = 0;
-    __syncthreads();
-    for (int j = 0; j < FUSE_ITEMS; j++) {
-        int i = tid * FUSE_ITEMS + j;
-        if (i < n) {
-            uint32_t idx = src[i] & 0x1FFFu;
-            uint32_t seg = dst[i];
-            atomicAdd((unsigned long long *)&tw[seg], (unsigned long long)win[idx]);
-            if (head[j]) {
-                tk[seg] = kin[idx];
-                tv[seg] = vin[idx];
+    if constexpr (sizeof(W) == 8 && (W)0.5 == (W)0) {  // integer weights
+        for (uint32_t i = tid; i < nseg; i += FUSE_THREADS) tw[i] = 0;
+        __syncthreads();
+        for (int j = 0; j < FUSE_ITEMS; j++) {
+            int i = tid * FUSE_ITEMS + j;
+            if (i < n) {
+                uint32_t idx = src[i] & 0x1FFFu;
+                uint32_t seg = dst[i];
+                atomicAdd((unsigned long long *)&tw[seg],
+                          (unsigned long long)win[idx]);
+                if (head[j]) {
+                    tk[seg] = kin[idx];
+                    tv[seg] = vin[idx];
+                }
             }
         }
+        __syncthreads();
+    } else {
+        // f64: deterministic segmented tree sum in sorted-position order.
+        // Stage weights in sorted order (tw[i] = w of sorted element i), record
+        // each segment's head position (cnt LDS is free after sorting), then
+        // combine (j % 2d == 0) pairs within segments — fixed reduction order,
+        // |err| <= 2 ulp * ceil(log2(run_len)) vs a sequential sum.
+        for (int j = 0; j < FUSE_ITEMS; j++) {
+            int i = tid * FUSE_ITEMS + j;
+            if (i < n) {
+                uint32_t idx = src[i] & 0x1FFFu;
+                tw[i] = win[idx];
+                if (head[j]) {
+                    uint32_t seg = dst[i];
+                    cnt[seg] = (uint32_t)i;  // head position of the segment
+                    tk[seg] = kin[idx];
+                    tv[seg] = vin[idx];
+                }
+            }
+        }
+        __syncthreads();
+        for (int64_t d = 1; d < n; d <<= 1) {
+            for (int j = 0; j < FUSE_ITEMS; j++) {
+                int i = tid * FUSE_ITEMS + j;
+                if (i + d < n && dst[i] == dst[i + d]) {
+                    uint32_t rel = (uint32_t)i - cnt[dst[i]];
+                    if ((rel & (2 * d - 1)) == 0) tw[i] += tw[i + d];
+                }
+            }
+            __syncthreads();
+        }
+        // move each segment's total (at its head position) to tw[seg]
+        W my_tot[FUSE_ITEMS];
+        for (int j = 0; j < FUSE_ITEMS; j++) {
+            int i = tid * FUSE_ITEMS + j;
+            my_tot[j] = (i < n && head[j]) ? tw[i] : (W)0;
+        }
+        __syncthreads();
+        for (int j = 0; j < FUSE_ITEMS; j++) {
+            int i = tid * FUSE_ITEMS + j;
+            if (i < n && head[j]) tw[dst[i]] = my_tot[j];
+        }
+        __syncthreads();
     }
-    __syncthreads();
     // ---- drop zero-weight segments (compact into output) ----
     uint32_t nz[FUSE_ITEMS];
     uint32_t zsum = 0;
     for (int j = 0; j < FUSE_ITEMS; j++) {
         int i = tid * FUSE_ITEMS + j;
-        nz[j] = (i < (int)nseg && tw[i] != 0) ? 1 : 0;
+        nz[j] = (i < (int)nseg && tw[i] != (W)0) ? 1 : 0;
         zsum += nz[j];
     }
     uint32_t nout;
@@ -609,6 +655,56 @@ __global__ void k_seg_accum(const uint64_t *k, const uint64_t *v,
             ov[seg] = v[i];
         }
     }
+}
+
+// f64 big-path consolidate support: deterministic segmented tree sum over the
+// sorted weight array (fixed position order; |err| <= 2 ulp * ceil(log2 run)).
+__device__ inline uint64_t seg_of(const uint64_t *fscan, const uint64_t *flags,
+                                  int64_t i) {
+    return flags[i] ? fscan[i] : fscan[i] - 1;
+}
+
+__global__ void k_seg_headpos(const uint64_t *flags, const uint64_t *fscan,
+                              int64_t n, uint64_t *hp) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        if (flags[i]) hp[fscan[i]] = (uint64_t)i;
+}
+
+__global__ void k_seg_tree_round(double *w, const uint64_t *flags,
+                                 const uint64_t *fscan, const uint64_t *hp,
+                                 int64_t n, int64_t d) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        if (i + d < n) {
+            uint64_t sg = seg_of(fscan, flags, i);
+            if (sg == seg_of(fscan, flags, i + d) &&
+                (((uint64_t)i - hp[sg]) & (uint64_t)(2 * d - 1)) == 0)
+                w[i] += w[i + d];
+        }
+    }
+}
+
+__global__ void k_seg_collect_f64(const uint64_t *k, const uint64_t *v,
+                                  const double *w, const uint64_t *flags,
+                                  const uint64_t *fscan, int64_t n,
+                                  uint64_t *ok, uint64_t *ov, double *ow) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        if (flags[i]) {
+            uint64_t sg = fscan[i];
+            ok[sg] = k[i];
+            ov[sg] = v[i];
+            ow[sg] = w[i];  // tree total sits at the head position
+        }
+    }
+}
+
+__global__ void k_nonzero_flags_f64(const double *w, int64_t n,
+                                    uint64_t *flags) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        flags[i] = w[i] != 0.0;  // -0.0 compares equal to 0.0: dropped
 }
 
 __global__ void k_nonzero_flags(const int64_t *w, int64_t n, uint64_t *flags) {
@@ -738,6 +834,7 @@ __device__ inline void proj_out(int proj, uint64_t param, uint64_t k,
         case DBSP_PROJ_HI_V2_LO_K: hi = v2; lo = k; break;
         case DBSP_PROJ_HI_V1_LO_K: hi = v1; lo = k; break;
         case DBSP_PROJ_HI_K_LO_V2V1: hi = k; lo = (v2 << 32) | (v1 & 0xFFFFFFFFull); break;
+        case DBSP_PROJ_HI_K_LO_V2: hi = k; lo = v2; break;
         default: hi = 0; lo = 0; break;
     }
 }
@@ -837,15 +934,16 @@ __device__ inline void adjust_split_lds(const uint64_t *lk, const uint64_t *lv,
 
 typedef __attribute__((address_space(1))) unsigned long long gu64_t;
 
+template <typename W>
 __global__ __launch_bounds__(MP_THREADS, 3) void k_mp_merge_onepass(
-    const uint64_t *ak, const uint64_t *av, const int64_t *aw, int64_t na,
-    const uint64_t *bk, const uint64_t *bv, const int64_t *bw, int64_t nb,
+    const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
+    const uint64_t *bk, const uint64_t *bv, const W *bw, int64_t nb,
     const int64_t *pa, const int64_t *pb, unsigned long long *state,
-    int64_t nblocks, uint64_t *ok, uint64_t *ov, int64_t *ow) {
+    int64_t nblocks, uint64_t *ok, uint64_t *ov, W *ow) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     uint64_t *lk = (uint64_t *)smem;
     uint64_t *lv = lk + (MP_TILE + 2);
-    int64_t *lw = (int64_t *)(lv + (MP_TILE + 2));
+    W *lw = (W *)(lv + (MP_TILE + 2));
     __shared__ uint32_t wt[MP_THREADS / WAVE + 1];
     __shared__ unsigned long long sh_vb;
     __shared__ unsigned long long sh_prefix;
@@ -883,7 +981,7 @@ __global__ __launch_bounds__(MP_THREADS, 3) void k_mp_merge_onepass(
         int64_t i = ai, j = bi;
         while (i < ae || j < be) {
             if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
-                if (lw[i] + lw[naL + j] != 0) cnt++;
+                if (lw[i] + lw[naL + j] != (W)0) cnt++;
                 i++; j++;
             } else if (j >= be ||
                        (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
@@ -955,8 +1053,8 @@ __global__ __launch_bounds__(MP_THREADS, 3) void k_mp_merge_onepass(
         int64_t i = ai, j = bi;
         while (i < ae || j < be) {
             if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
-                int64_t sum = lw[i] + lw[naL + j];
-                if (sum != 0) { ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = sum; gpos++; }
+                W sum = lw[i] + lw[naL + j];
+                if (sum != (W)0) { ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = sum; gpos++; }
                 i++; j++;
             } else if (j >= be ||
                        (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
@@ -971,17 +1069,18 @@ __global__ __launch_bounds__(MP_THREADS, 3) void k_mp_merge_onepass(
 
 // single-workgroup merge for small inputs (na+nb <= FUSE_MAX): merge-path
 // thread diagonals + LDS scan; one launch, no host sync, length on device
+template <typename W>
 __global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_small(MergeArgs args) {
     __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
     const int p = blockIdx.x;
     const uint64_t *ak = args.ak[p], *av = args.av[p];
-    const int64_t *aw = args.aw[p];
+    const W *aw = (const W *)args.aw[p];
     const int64_t na = args.na[p];
     const uint64_t *bk = args.bk[p], *bv = args.bv[p];
-    const int64_t *bw = args.bw[p];
+    const W *bw = (const W *)args.bw[p];
     const int64_t nb = args.nb[p];
     uint64_t *ok = args.ok[p], *ov = args.ov[p];
-    int64_t *ow = args.ow[p];
+    W *ow = (W *)args.ow[p];
     int64_t *out_len = args.d_len + p;
     const int tid = threadIdx.x;
     const int64_t total = na + nb;
@@ -998,7 +1097,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_small(MergeArgs args)
         int64_t i = ai, j = bi;
         while (i < ae || j < be) {
             if (i < ae && j < be && row_eq(ak[i], av[i], bk[j], bv[j])) {
-                if (aw[i] + bw[j] != 0) cnt++;
+                if (aw[i] + bw[j] != (W)0) cnt++;
                 i++; j++;
             } else if (j >= be || (i < ae && row_lt(ak[i], av[i], bk[j], bv[j]))) {
                 cnt++; i++;
@@ -1013,8 +1112,8 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_small(MergeArgs args)
         int64_t i = ai, j = bi;
         while (i < ae || j < be) {
             if (i < ae && j < be && row_eq(ak[i], av[i], bk[j], bv[j])) {
-                int64_t s = aw[i] + bw[j];
-                if (s != 0) { ok[off] = ak[i]; ov[off] = av[i]; ow[off] = s; off++; }
+                W s = aw[i] + bw[j];
+                if (s != (W)0) { ok[off] = ak[i]; ov[off] = av[i]; ow[off] = s; off++; }
                 i++; j++;
             } else if (j >= be || (i < ae && row_lt(ak[i], av[i], bk[j], bv[j]))) {
                 ok[off] = ak[i]; ov[off] = av[i]; ow[off] = aw[i]; off++; i++;
@@ -1452,15 +1551,23 @@ __global__ void k_map(const uint64_t *k, const uint64_t *v, const int64_t *w,
          i += (int64_t)gridDim.x * blockDim.x) {
         uint64_t kk = k[i], vv = v[i];
         uint64_t rk, rv;
+        int64_t rw = w[i];
         switch (mode) {
             case 0: rk = vv >> 32; rv = ((vv & 0xFFFFFFFFull) << 32) | (kk & 0xFFFFFFFFull); break;
             case 1: rk = vv; rv = 0; break;
             case 2: rk = 0; rv = vv; break;
+            case 4: {  // weigh (aggregate/mod.rs:297-323): f(k,v)=f64(v); w' = f(k,v)*w
+                rk = kk; rv = 0;
+                double f = *(const double *)&vv;
+                double wp = f * (double)w[i];
+                rw = *(const int64_t *)&wp;
+                break;
+            }
             default: rk = vv; rv = kk; break;
         }
         ok[i] = rk;
         ov[i] = rv;
-        ow[i] = w[i];
+        ow[i] = rw;
     }
 }
 
@@ -1581,11 +1688,12 @@ dbsp_status consolidate_sorted(hipStream_t s, const uint64_t *kk,
     return DBSP_OK;
 }
 
-dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
-                       const int64_t *aw, int64_t na, const uint64_t *bk,
-                       const uint64_t *bv, const int64_t *bw, int64_t nb,
-                       uint64_t **ok, uint64_t **ov, int64_t **ow,
-                       int64_t *out_n) {
+template <typename W>
+static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
+                                const uint64_t *av, const W *aw, int64_t na,
+                                const uint64_t *bk, const uint64_t *bv,
+                                const W *bw, int64_t nb, uint64_t **ok,
+                                uint64_t **ov, W **ow, int64_t *out_n) {
     int64_t total = na + nb;
     if (total == 0) {
         *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
@@ -1601,12 +1709,12 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
     k_mp_partition<<<grid_for(nblocks + 1), BLK, 0, s>>>(ak, av, na, bk, bv, nb,
                                                          nblocks, pa, pb);
     uint64_t *rk, *rv;
-    int64_t *rw;
+    W *rw;
     HIP_CHECK(hipMallocAsync(&rk, total * sizeof(uint64_t) + 8, s));
     HIP_CHECK(hipMallocAsync(&rv, total * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rw, total * sizeof(int64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rw, total * sizeof(W) + 8, s));
     const size_t smem = 3 * (MP_TILE + 2) * sizeof(uint64_t);
-    k_mp_merge_onepass<<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
+    k_mp_merge_onepass<W><<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
         ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
     // total = inclusive prefix of the last virtual block; error word poisoned
     // if a lookback spun out (never observed; bounded to avoid hangs)
@@ -1629,11 +1737,39 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
     return DBSP_OK;
 }
 
+dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
+                       const int64_t *aw, int64_t na, const uint64_t *bk,
+                       const uint64_t *bv, const int64_t *bw, int64_t nb,
+                       uint64_t **ok, uint64_t **ov, int64_t **ow,
+                       int64_t *out_n) {
+    return merge_rows_t<int64_t>(s, ak, av, aw, na, bk, bv, bw, nb, ok, ov, ow,
+                                 out_n);
+}
+
+dbsp_status merge_rows_f64(hipStream_t s, const uint64_t *ak, const uint64_t *av,
+                           const double *aw, int64_t na, const uint64_t *bk,
+                           const uint64_t *bv, const double *bw, int64_t nb,
+                           uint64_t **ok, uint64_t **ov, double **ow,
+                           int64_t *out_n) {
+    return merge_rows_t<double>(s, ak, av, aw, na, bk, bv, bw, nb, ok, ov, ow,
+                                out_n);
+}
+
 dbsp_status sort_cons_small_batch(hipStream_t s, const SortArgs &args) {
     for (int b = 0; b < args.nb; b++)
         if (args.n[b] > FUSE_MAX) return DBSP_ERR_INVALID;
     if (args.nb == 0) return DBSP_OK;
-    k_sort_cons_small<<<dim3((uint32_t)args.nb), FUSE_THREADS, 0, s>>>(args);
+    k_sort_cons_small<int64_t>
+        <<<dim3((uint32_t)args.nb), FUSE_THREADS, 0, s>>>(args);
+    return DBSP_OK;
+}
+
+dbsp_status sort_cons_small_batch_f64(hipStream_t s, const SortArgs &args) {
+    for (int b = 0; b < args.nb; b++)
+        if (args.n[b] > FUSE_MAX) return DBSP_ERR_INVALID;
+    if (args.nb == 0) return DBSP_OK;
+    k_sort_cons_small<double>
+        <<<dim3((uint32_t)args.nb), FUSE_THREADS, 0, s>>>(args);
     return DBSP_OK;
 }
 
@@ -1641,7 +1777,15 @@ dbsp_status merge_small_batch(hipStream_t s, const MergeArgs &args) {
     if (args.np == 0) return DBSP_OK;
     for (int i = 0; i < args.np; i++)
         if (args.na[i] + args.nb[i] > 4 * FUSE_MAX) return DBSP_ERR_INVALID;
-    k_merge_small<<<dim3((uint32_t)args.np), FUSE_THREADS, 0, s>>>(args);
+    k_merge_small<int64_t><<<dim3((uint32_t)args.np), FUSE_THREADS, 0, s>>>(args);
+    return DBSP_OK;
+}
+
+dbsp_status merge_small_batch_f64(hipStream_t s, const MergeArgs &args) {
+    if (args.np == 0) return DBSP_OK;
+    for (int i = 0; i < args.np; i++)
+        if (args.na[i] + args.nb[i] > 4 * FUSE_MAX) return DBSP_ERR_INVALID;
+    k_merge_small<double><<<dim3((uint32_t)args.np), FUSE_THREADS, 0, s>>>(args);
     return DBSP_OK;
 }
 
@@ -1898,6 +2042,38 @@ dbsp_status map_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
 
 // ---- multi-batch linear aggregate (spine batches summed; linear op) ----
 
+// f64 linear aggregate: per-key sums in batch-position order (deterministic)
+__global__ void k_agg_sum_f64(const uint64_t *keys, int64_t nd,
+                              const uint64_t *ik, const double *iw, int64_t ni,
+                              double *acc) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nd;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        uint64_t key = keys[i];
+        int64_t lo = lower_bound_k(ik, ni, key);
+        int64_t hi = upper_bound_k(ik, ni, key);
+        double sum = 0.0;
+        for (int64_t t = lo; t < hi; t++) sum += iw[t];
+        acc[i] += sum;
+    }
+}
+
+// emit (key, bits(sum), +1) for sum != 0.0 (WeightedCount returns None for a
+// zero f64 aggregate — aggregate/mod.rs:129-156 with R = F64)
+__global__ void k_emit_nonzero_f64(const uint64_t *keys, const double *acc,
+                                   int64_t nd, uint64_t *ctr, uint64_t *ok,
+                                   uint64_t *ov, int64_t *ow) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nd;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        if (acc[i] != 0.0) {
+            uint64_t p = atomicAdd((unsigned long long *)ctr, 1ull);
+            ok[p] = keys[i];
+            double a = acc[i];
+            ov[p] = *(const uint64_t *)&a;
+            ow[p] = 1;
+        }
+    }
+}
+
 __global__ void k_agg_sum(const uint64_t *keys, int64_t nd, const uint64_t *ik,
                           const int64_t *iw, int64_t ni, int64_t *acc) {
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nd;
@@ -1939,6 +2115,87 @@ __global__ void k_compact_keys(const uint64_t *k, const uint64_t *flags,
 }
 
 namespace dbspk {
+
+// consolidate SORTED rows with f64 weights: deterministic segmented tree
+// (position-fixed order; tolerance 2 ulp * ceil(log2 run) vs sequential)
+dbsp_status consolidate_sorted_f64(hipStream_t s, const uint64_t *kk,
+                                   const uint64_t *vv, double *ww, int64_t n,
+                                   uint64_t **ok, uint64_t **ov, double **ow,
+                                   int64_t *out_n) {
+    if (n == 0) {
+        *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
+        return DBSP_OK;
+    }
+    uint64_t *flags, *fscan, *hp;
+    HIP_CHECK(hipMallocAsync(&flags, n * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&fscan, n * sizeof(uint64_t), s));
+    k_head_flags<<<grid_for(n), BLK, 0, s>>>(kk, vv, n, flags);
+    uint64_t nseg = 0;
+    dbsp_status st = scan_exclusive(s, flags, fscan, n, &nseg);
+    if (st != DBSP_OK) return st;
+    HIP_CHECK(hipMallocAsync(&hp, nseg * sizeof(uint64_t) + 8, s));
+    k_seg_headpos<<<grid_for(n), BLK, 0, s>>>(flags, fscan, n, hp);
+    for (int64_t d = 1; d < n; d <<= 1)
+        k_seg_tree_round<<<grid_for(n), BLK, 0, s>>>(ww, flags, fscan, hp, n, d);
+    uint64_t *sk, *sv;
+    double *sw;
+    HIP_CHECK(hipMallocAsync(&sk, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&sv, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&sw, nseg * sizeof(double) + 8, s));
+    k_seg_collect_f64<<<grid_for(n), BLK, 0, s>>>(kk, vv, ww, flags, fscan, n,
+                                                  sk, sv, sw);
+    uint64_t *nzflags, *nzscan;
+    HIP_CHECK(hipMallocAsync(&nzflags, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&nzscan, nseg * sizeof(uint64_t) + 8, s));
+    k_nonzero_flags_f64<<<grid_for(nseg), BLK, 0, s>>>(sw, nseg, nzflags);
+    uint64_t nout = 0;
+    st = scan_exclusive(s, nzflags, nzscan, nseg, &nout);
+    if (st != DBSP_OK) return st;
+    uint64_t *rk, *rv;
+    double *rw;
+    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(double) + 8, s));
+    // k_compact copies the weight column bitwise (8 B) — valid for f64
+    k_compact<<<grid_for(nseg), BLK, 0, s>>>(sk, sv, (const int64_t *)sw,
+                                             nzflags, nzscan, nseg, rk, rv,
+                                             (int64_t *)rw);
+    HIP_CHECK(hipFreeAsync(flags, s));
+    HIP_CHECK(hipFreeAsync(fscan, s));
+    HIP_CHECK(hipFreeAsync(hp, s));
+    HIP_CHECK(hipFreeAsync(sk, s));
+    HIP_CHECK(hipFreeAsync(sv, s));
+    HIP_CHECK(hipFreeAsync(sw, s));
+    HIP_CHECK(hipFreeAsync(nzflags, s));
+    HIP_CHECK(hipFreeAsync(nzscan, s));
+    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
+    return DBSP_OK;
+}
+
+dbsp_status agg_sum_batch_f64(hipStream_t s, const uint64_t *keys, int64_t nd,
+                              const uint64_t *ik, const double *iw, int64_t ni,
+                              double *acc) {
+    if (nd > 0 && ni > 0)
+        k_agg_sum_f64<<<grid_for(nd), BLK, 0, s>>>(keys, nd, ik, iw, ni, acc);
+    return DBSP_OK;
+}
+
+dbsp_status emit_nonzero_f64(hipStream_t s, const uint64_t *keys,
+                             const double *acc, int64_t nd, uint64_t *ok,
+                             uint64_t *ov, int64_t *ow, int64_t *h_count) {
+    uint64_t *ctr;
+    HIP_CHECK(hipMallocAsync(&ctr, sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(ctr, 0, sizeof(uint64_t), s));
+    if (nd > 0)
+        k_emit_nonzero_f64<<<grid_for(nd), BLK, 0, s>>>(keys, acc, nd, ctr, ok,
+                                                        ov, ow);
+    uint64_t h = 0;
+    HIP_CHECK(hipMemcpyAsync(&h, ctr, sizeof(uint64_t), hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    HIP_CHECK(hipFreeAsync(ctr, s));
+    *h_count = (int64_t)h;
+    return DBSP_OK;
+}
 
 dbsp_status agg_sum_batch(hipStream_t s, const uint64_t *keys, int64_t nd,
                           const uint64_t *ik, const int64_t *iw, int64_t ni,

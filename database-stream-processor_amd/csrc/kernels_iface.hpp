@@ -177,6 +177,26 @@ dbsp_status map_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
 dbsp_status unique_keys(hipStream_t s, const uint64_t *kk, int64_t n,
                         uint64_t **okeys, int64_t *out_n);
 
+// ---- f64-weight variants (config C5: f64 sum aggregate; deterministic
+// position-fixed reduction order, tolerance 2 ulp * reduction depth) ----
+dbsp_status sort_cons_small_batch_f64(hipStream_t s, const SortArgs &args);
+dbsp_status merge_small_batch_f64(hipStream_t s, const MergeArgs &args);
+dbsp_status merge_rows_f64(hipStream_t s, const uint64_t *ak,
+                           const uint64_t *av, const double *aw, int64_t na,
+                           const uint64_t *bk, const uint64_t *bv,
+                           const double *bw, int64_t nb, uint64_t **ok,
+                           uint64_t **ov, double **ow, int64_t *out_n);
+dbsp_status consolidate_sorted_f64(hipStream_t s, const uint64_t *kk,
+                                   const uint64_t *vv, double *ww, int64_t n,
+                                   uint64_t **ok, uint64_t **ov, double **ow,
+                                   int64_t *out_n);
+dbsp_status agg_sum_batch_f64(hipStream_t s, const uint64_t *keys, int64_t nd,
+                              const uint64_t *ik, const double *iw, int64_t ni,
+                              double *acc);
+dbsp_status emit_nonzero_f64(hipStream_t s, const uint64_t *keys,
+                             const double *acc, int64_t nd, uint64_t *ok,
+                             uint64_t *ov, int64_t *ow, int64_t *h_count);
+
 uint64_t host_xxh3_u64(uint64_t key, uint64_t seed);
 
 }  // namespace dbspk

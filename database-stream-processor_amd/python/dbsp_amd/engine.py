@@ -68,6 +68,25 @@ def _L():
                                            ctypes.POINTER(BatchStruct), vp]
         L.dbsp_xxh3_u64.restype = u64
         L.dbsp_xxh3_u64.argtypes = [u64, u64]
+        L.dbsp_sort_consolidate_f64.restype = i32
+        L.dbsp_sort_consolidate_f64.argtypes = [vp, vp, vp, vp, i64,
+                                                ctypes.POINTER(BatchStruct)]
+        L.dbsp_merge_f64.restype = i32
+        L.dbsp_merge_f64.argtypes = [vp, ctypes.POINTER(BatchStruct),
+                                     ctypes.POINTER(BatchStruct),
+                                     ctypes.POINTER(BatchStruct)]
+        L.dbsp_weigh_f64.restype = i32
+        L.dbsp_weigh_f64.argtypes = [vp, ctypes.POINTER(BatchStruct),
+                                     ctypes.POINTER(BatchStruct)]
+        L.dbsp_agg_linear_upsert_f64.restype = i32
+        L.dbsp_agg_linear_upsert_f64.argtypes = [vp, vp, i64,
+                                                 ctypes.POINTER(BatchStruct),
+                                                 ctypes.POINTER(BatchStruct),
+                                                 ctypes.POINTER(BatchStruct)]
+        L.dbsp_unique_keys.restype = i32
+        L.dbsp_unique_keys.argtypes = [vp, ctypes.POINTER(BatchStruct),
+                                       ctypes.POINTER(vp),
+                                       ctypes.POINTER(i64)]
         L.dbsp_comm_unique_id.restype = i32
         L.dbsp_comm_unique_id.argtypes = [vp]
         L.dbsp_comm_init.restype = i32
@@ -247,6 +266,62 @@ class Ctx:
         self.sync()
         res = self.download_rows(out)
         for x in (t, b, out):
+            self.free_batch(x)
+        return res
+
+    def sort_consolidate_f64(self, rows: np.ndarray) -> np.ndarray:
+        raw = self.upload_rows(rows)
+        out = BatchStruct()
+        _check(self._lib.dbsp_sort_consolidate_f64(self._h, raw.k, raw.v, raw.w,
+                                                   raw.len, ctypes.byref(out)),
+               "sort_consolidate_f64")
+        self.sync()
+        res = self.download_rows(out)
+        self.free_batch(raw)
+        self.free_batch(out)
+        return res
+
+    def merge_f64(self, a_rows, b_rows) -> np.ndarray:
+        a = self.upload_rows(a_rows)
+        b = self.upload_rows(b_rows)
+        out = BatchStruct()
+        _check(self._lib.dbsp_merge_f64(self._h, ctypes.byref(a),
+                                        ctypes.byref(b), ctypes.byref(out)),
+               "merge_f64")
+        res = self.download_rows(out)
+        for x in (a, b, out):
+            self.free_batch(x)
+        return res
+
+    def weigh_f64(self, rows) -> np.ndarray:
+        inp = self.upload_rows(rows)
+        out = BatchStruct()
+        _check(self._lib.dbsp_weigh_f64(self._h, ctypes.byref(inp),
+                                        ctypes.byref(out)), "weigh_f64")
+        self.sync()
+        res = self.download_rows(out)
+        for x in (inp, out):
+            self.free_batch(x)
+        return res
+
+    def agg_linear_upsert_f64(self, keys, in_rows, out_rows) -> np.ndarray:
+        keys = np.ascontiguousarray(keys, dtype=np.uint64)
+        dk = ctypes.c_void_p()
+        _check(self._lib.dbsp_dev_alloc(self._h, max(len(keys), 1) * 8,
+                                        ctypes.byref(dk)), "alloc")
+        if len(keys):
+            _check(self._lib.dbsp_h2d(self._h, dk, _p(keys), len(keys) * 8), "h2d")
+        it = self.upload_rows(in_rows)
+        ot = self.upload_rows(out_rows)
+        out = BatchStruct()
+        _check(self._lib.dbsp_agg_linear_upsert_f64(self._h, dk, len(keys),
+                                                    ctypes.byref(it),
+                                                    ctypes.byref(ot),
+                                                    ctypes.byref(out)), "aggf64")
+        self.sync()
+        res = self.download_rows(out)
+        self._lib.dbsp_dev_free(self._h, dk)
+        for x in (it, ot, out):
             self.free_batch(x)
         return res
 

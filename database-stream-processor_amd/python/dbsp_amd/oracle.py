@@ -40,6 +40,14 @@ def _o():
         L.oracle_query_step.argtypes = [vp, vp, i64, vp, i64]
         L.oracle_q0_step.restype = i64
         L.oracle_q0_step.argtypes = [vp, i64, vp, i64]
+        L.oracle_consolidate_f64.restype = i64
+        L.oracle_consolidate_f64.argtypes = [vp, i64]
+        L.oracle_merge_f64.restype = i64
+        L.oracle_merge_f64.argtypes = [vp, i64, vp, i64, vp]
+        L.oracle_weigh_f64.restype = i64
+        L.oracle_weigh_f64.argtypes = [vp, i64, vp]
+        L.oracle_agg_linear_upsert_f64.restype = i64
+        L.oracle_agg_linear_upsert_f64.argtypes = [vp, i64, vp, i64, vp, i64, vp, i64]
     return _lib
 
 
@@ -129,5 +137,35 @@ def q0_step(events: np.ndarray, cap=None) -> np.ndarray:
     cap = cap or len(events) + 16
     out = np.empty(cap, dtype=EVENT_DT)
     n = _o().oracle_q0_step(_p(events), len(events), _p(out), cap)
+    assert n >= 0
+    return out[:n].copy()
+
+
+# ---- f64-weight variants (weights are f64 bit patterns in the w column) ----
+
+def consolidate_f64(r: np.ndarray) -> np.ndarray:
+    r = np.array(r, dtype=ROW_DT)
+    n = _o().oracle_consolidate_f64(_p(r), len(r))
+    return r[:n].copy()
+
+
+def merge_f64(a: np.ndarray, b: np.ndarray) -> np.ndarray:
+    out = np.empty(len(a) + len(b), dtype=ROW_DT)
+    n = _o().oracle_merge_f64(_p(a), len(a), _p(b), len(b), _p(out))
+    return out[:n].copy()
+
+
+def weigh_f64(r: np.ndarray) -> np.ndarray:
+    out = np.empty(len(r), dtype=ROW_DT)
+    n = _o().oracle_weigh_f64(_p(np.ascontiguousarray(r, dtype=ROW_DT)), len(r), _p(out))
+    return out[:n].copy()
+
+
+def agg_linear_upsert_f64(keys, in_trace, out_trace, cap=1 << 20):
+    keys = np.asarray(keys, dtype=np.uint64)
+    out = np.empty(cap, dtype=ROW_DT)
+    n = _o().oracle_agg_linear_upsert_f64(_p(keys), len(keys), _p(in_trace),
+                                          len(in_trace), _p(out_trace),
+                                          len(out_trace), _p(out), cap)
     assert n >= 0
     return out[:n].copy()

@@ -89,6 +89,8 @@ typedef enum {
     DBSP_PROJ_HI_V2_LO_K     = 5, /* q5 final join */
     DBSP_PROJ_HI_V1_LO_K     = 6, /* q5 final join swapped side */
     DBSP_PROJ_HI_K_LO_V2V1   = 7, /* generic pair join, swapped side: lo=v2<<32|v1 */
+    DBSP_PROJ_HI_K_LO_V2     = 8, /* (k, v2): keeps the join key, takes the trace val
+                                     (full 64-bit; the C5 join -> f64-sum pipeline) */
 } dbsp_proj;
 
 /* ======================================================================
@@ -191,6 +193,33 @@ dbsp_status dbsp_window(dbsp_ctx *ctx, const dbsp_batch *trace,
 dbsp_status dbsp_shard_partition(dbsp_ctx *ctx, const dbsp_batch *in,
                                  int nshards, dbsp_batch *out,
                                  int64_t *offsets_host);
+
+/* ---- f64-weight variants (config C5: the f64 sum aggregate path).
+ * Weights travel in the same 8-byte column as f64 bit patterns.  All f64
+ * reductions use a position-fixed order (segmented tree / batch order);
+ * documented tolerance vs a sequential sum: |err| <= 2 ulp * reduction depth
+ * (SURVEY.md §8d).  Zero-weight elimination follows the reference's F64
+ * is_zero (== 0.0, so -0.0 is eliminated; algebra/floats.rs:24). */
+dbsp_status dbsp_sort_consolidate_f64(dbsp_ctx *ctx, const uint64_t *k_in,
+                                      const uint64_t *v_in, const double *w_in,
+                                      int64_t n, dbsp_batch *out);
+dbsp_status dbsp_merge_f64(dbsp_ctx *ctx, const dbsp_batch *a,
+                           const dbsp_batch *b, dbsp_batch *out);
+/* weigh (aggregate/mod.rs:297-323) with f(k, v) = f64_from_bits(v):
+ * (k, v, w) -> raw rows (k, (), f64(v) * w); caller consolidates. */
+dbsp_status dbsp_weigh_f64(dbsp_ctx *ctx, const dbsp_batch *in,
+                           dbsp_batch *out_raw);
+/* linear aggregate over an f64-weighted input trace + upsert against an
+ * i64-weighted output trace whose values are f64 bit patterns. */
+dbsp_status dbsp_agg_linear_upsert_f64(dbsp_ctx *ctx,
+                                       const uint64_t *delta_keys, int64_t nd,
+                                       const dbsp_batch *in_trace,
+                                       const dbsp_batch *out_trace,
+                                       dbsp_batch *out_raw);
+
+/* Distinct keys of a consolidated batch (device out, caller frees). */
+dbsp_status dbsp_unique_keys(dbsp_ctx *ctx, const dbsp_batch *in,
+                             uint64_t **out_keys, int64_t *n_out);
 
 /* Reference hash (host-callable too, for tests): xxh3_64(le_bytes(key), seed). */
 uint64_t dbsp_xxh3_u64(uint64_t key, uint64_t seed);

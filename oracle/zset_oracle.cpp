@@ -111,6 +111,7 @@ inline void proj_emit(std::vector<Row> &out, dbsp_proj proj, uint64_t param,
         case DBSP_PROJ_HI_V2_LO_K: hi = v2; lo = k; break;
         case DBSP_PROJ_HI_V1_LO_K: hi = v1; lo = k; break;
         case DBSP_PROJ_HI_K_LO_V2V1: hi = k; lo = (v2 << 32) | (v1 & 0xFFFFFFFFull); break;
+        case DBSP_PROJ_HI_K_LO_V2: hi = k; lo = v2; break;
     }
     out.push_back({hi, lo, w});
 }
@@ -543,6 +544,97 @@ int64_t oracle_window(const dbsp_row *trace, int64_t nt, const dbsp_row *batch,
 }
 
 uint64_t oracle_xxh3_u64(uint64_t key, uint64_t seed) { return xxh3_u64(key, seed); }
+
+// ---- f64-weight variants (config C5).  The oracle is the SEQUENTIAL
+// reference: sums accumulate left-to-right in sorted order; the GPU's
+// position-fixed tree order is compared against it within the stated
+// tolerance (|err| <= 2 ulp * reduction depth).  Zero
+// elimination follows F64 is_zero (== 0.0; algebra/floats.rs:24). ----
+
+static inline double bits_to_f64(int64_t w) {
+    double d;
+    std::memcpy(&d, &w, 8);
+    return d;
+}
+static inline int64_t f64_to_bits(double d) {
+    int64_t w;
+    std::memcpy(&w, &d, 8);
+    return w;
+}
+
+int64_t oracle_consolidate_f64(dbsp_row *rows, int64_t n) {
+    std::vector<Row> v(n);
+    for (int64_t i = 0; i < n; i++) v[i] = {rows[i].k, rows[i].v, rows[i].w};
+    std::sort(v.begin(), v.end(), row_lt);
+    std::vector<Row> out;
+    size_t i = 0;
+    while (i < v.size()) {
+        size_t j = i;
+        double sum = 0.0;
+        while (j < v.size() && v[j].k == v[i].k && v[j].v == v[i].v) {
+            sum += bits_to_f64(v[j].w);
+            j++;
+        }
+        if (sum != 0.0) out.push_back({v[i].k, v[i].v, f64_to_bits(sum)});
+        i = j;
+    }
+    for (size_t t = 0; t < out.size(); t++)
+        rows[t] = {out[t].k, out[t].v, out[t].w};
+    return (int64_t)out.size();
+}
+
+int64_t oracle_merge_f64(const dbsp_row *a, int64_t na, const dbsp_row *b,
+                         int64_t nb, dbsp_row *out) {
+    size_t i = 0, j = 0, o = 0;
+    while (i < (size_t)na && j < (size_t)nb) {
+        Row ra{a[i].k, a[i].v, a[i].w}, rb{b[j].k, b[j].v, b[j].w};
+        if (row_lt(ra, rb)) {
+            out[o++] = a[i++];
+        } else if (row_lt(rb, ra)) {
+            out[o++] = b[j++];
+        } else {
+            double s = bits_to_f64(a[i].w) + bits_to_f64(b[j].w);
+            if (s != 0.0) out[o++] = {a[i].k, a[i].v, f64_to_bits(s)};
+            i++; j++;
+        }
+    }
+    while (i < (size_t)na) out[o++] = a[i++];
+    while (j < (size_t)nb) out[o++] = b[j++];
+    return (int64_t)o;
+}
+
+// weigh (aggregate/mod.rs:297-323) with f(k,v) = f64_from_bits(v)
+int64_t oracle_weigh_f64(const dbsp_row *in, int64_t n, dbsp_row *out) {
+    for (int64_t i = 0; i < n; i++) {
+        double wp = bits_to_f64((int64_t)in[i].v) * (double)in[i].w;
+        out[i] = {in[i].k, 0, f64_to_bits(wp)};
+    }
+    return n;
+}
+
+int64_t oracle_agg_linear_upsert_f64(const uint64_t *keys, int64_t nk,
+                                     const dbsp_row *in_trace, int64_t nin,
+                                     const dbsp_row *out_trace, int64_t nout_t,
+                                     dbsp_row *out, int64_t cap) {
+    int64_t o = 0;
+    for (int64_t t = 0; t < nk; t++) {
+        uint64_t key = keys[t];
+        double s = 0.0;
+        for (int64_t i = 0; i < nin; i++)
+            if (in_trace[i].k == key) s += bits_to_f64(in_trace[i].w);
+        std::vector<Row> upd;
+        if (s != 0.0) upd.push_back({key, (uint64_t)f64_to_bits(s), 1});
+        for (int64_t i = 0; i < nout_t; i++)
+            if (out_trace[i].k == key && out_trace[i].w != 0)
+                upd.push_back({key, out_trace[i].v, -out_trace[i].w});
+        consolidate(upd);
+        for (auto &r : upd) {
+            if (o >= cap) return -1;
+            out[o++] = {r.k, r.v, r.w};
+        }
+    }
+    return o;
+}
 
 // --- query drivers ---
 void *oracle_query_new(int query) {

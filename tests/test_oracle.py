@@ -239,3 +239,34 @@ def test_q0_consolidates_events():
     out = oracle.q0_step(evs)
     assert len(out) == 2
     assert out["w"].tolist() == [2, 1] or out["w"].tolist() == [1, 2]
+
+
+def test_oracle_f64_model():
+    """f64-weight oracle vs a numpy restatement (sequential sums, ==0.0 drop)."""
+    rng = np.random.default_rng(21)
+    n = 5000
+    r = np.empty(n, dtype=ROW_DT)
+    r["k"] = rng.integers(0, 100, n)
+    r["v"] = rng.integers(0, 4, n)
+    r["w"] = (rng.integers(-8, 9, n).astype(np.float64) * 0.25).view(np.int64)
+    got = oracle.consolidate_f64(r)
+    d = {}
+    order = np.lexsort((r["v"], r["k"]))
+    for i in order:
+        key = (int(r["k"][i]), int(r["v"][i]))
+        d[key] = d.get(key, 0.0) + float(np.int64(r["w"][i]).view(np.float64))
+    exp = {k: w for k, w in d.items() if w != 0.0}
+    gotd = {(int(x["k"]), int(x["v"])): float(np.int64(x["w"]).view(np.float64))
+            for x in got}
+    assert gotd == exp
+    # merge_f64 == consolidate_f64 of concat on consolidated inputs
+    a = oracle.consolidate_f64(r[: n // 2])
+    b = oracle.consolidate_f64(r[n // 2:])
+    m = oracle.merge_f64(a, b)
+    c = oracle.consolidate_f64(np.concatenate([a, b]))
+    assert np.array_equal(m, c)
+    # -0.0 elimination
+    z = np.zeros(2, dtype=ROW_DT)
+    z["k"] = [1, 1]
+    z["w"] = np.array([0.5, -0.5]).view(np.int64)
+    assert len(oracle.consolidate_f64(z)) == 0
