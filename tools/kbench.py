@@ -196,6 +196,125 @@ def bench_incremental(ctx, L, n_trace, n_delta, steps=8):
     }
 
 
+def bench_c5(ctx, L, n_trace, n_delta, steps=6):
+    """Config C5 end-to-end: 1B-row indexed trace
+    (1 val/key, f64 vals) x 10M-row delta incremental join, f64 vals weighed
+    (f(k,v)=v) into f64 weights, consolidated deterministically, and
+    sum-aggregated per key with upsert against the running output trace.
+    i64 structural weights bit-exact; f64 sums within 2 ulp * depth."""
+    g = torch.Generator(device="cuda").manual_seed(41)
+    gaps = torch.randint(1, 8, (n_trace,), generator=g, dtype=torch.int64,
+                         device="cuda")
+    kt = torch.cumsum(gaps, 0)
+    vt = torch.rand(n_trace, generator=g, dtype=torch.float64,
+                    device="cuda").view(torch.int64)
+    wt = torch.ones(n_trace, dtype=torch.int64, device="cuda")
+    torch.cuda.synchronize()
+    spine = [(as_batch(kt, vt, wt), None)]
+    kmax = int(kt[-1].item())
+    wb = None        # f64-weighted integral (auction-sum input trace)
+    out_tr = None    # aggregate output trace (k -> f64 bits, i64 weights)
+    step_ms = []
+    gap = max(2, (9 * n_trace) // (2 * n_delta))
+    for it in range(steps):
+        gd = torch.Generator(device="cuda").manual_seed(500 + it)
+        kd = torch.cumsum(torch.randint(1, gap, (n_delta,), generator=gd,
+                                        dtype=torch.int64, device="cuda"), 0)
+        vd = torch.zeros(n_delta, dtype=torch.int64, device="cuda")
+        wd = torch.ones(n_delta, dtype=torch.int64, device="cuda")
+        torch.cuda.synchronize()
+        d = as_batch(kd, vd, wd)
+        ctx.sync()
+        t0 = time.perf_counter()
+        # join vs every spine batch (proj (k, v2): carry the f64 val)
+        outs = []
+        for b, _ in spine:
+            o = BatchStruct()
+            assert L.dbsp_join(ctx._h, ctypes.byref(d), ctypes.byref(b), 8, 0,
+                               ctypes.byref(o)) == 0
+            outs.append(o)
+        # weigh + consolidate each join output, fold into the f64 integral
+        for o in outs:
+            if o.len > 0:
+                wre = BatchStruct()
+                assert L.dbsp_weigh_f64(ctx._h, ctypes.byref(o),
+                                        ctypes.byref(wre)) == 0
+                dwb = BatchStruct()
+                assert L.dbsp_sort_consolidate_f64(ctx._h, wre.k, wre.v, wre.w,
+                                                   wre.len,
+                                                   ctypes.byref(dwb)) == 0
+                ctx.free_batch(wre)
+                if wb is None:
+                    wb = dwb
+                else:
+                    m = BatchStruct()
+                    assert L.dbsp_merge_f64(ctx._h, ctypes.byref(wb),
+                                            ctypes.byref(dwb),
+                                            ctypes.byref(m)) == 0
+                    ctx.free_batch(wb)
+                    ctx.free_batch(dwb)
+                    wb = m
+            ctx.free_batch(o)
+        # aggregate affected keys + upsert
+        if wb is not None and wb.len > 0:
+            keys = ctypes.c_void_p()
+            nk = ctypes.c_int64()
+            assert L.dbsp_unique_keys(ctx._h, ctypes.byref(wb),
+                                      ctypes.byref(keys),
+                                      ctypes.byref(nk)) == 0
+            empty = BatchStruct()
+            upd = BatchStruct()
+            assert L.dbsp_agg_linear_upsert_f64(
+                ctx._h, keys, nk.value, ctypes.byref(wb),
+                ctypes.byref(out_tr if out_tr else empty),
+                ctypes.byref(upd)) == 0
+            L.dbsp_dev_free(ctx._h, keys)
+            cons = BatchStruct()
+            assert L.dbsp_sort_consolidate(ctx._h, upd.k, upd.v, upd.w,
+                                           upd.len, ctypes.byref(cons)) == 0
+            ctx.free_batch(upd)
+            if out_tr is None:
+                out_tr = cons
+            else:
+                m = BatchStruct()
+                assert L.dbsp_merge(ctx._h, ctypes.byref(out_tr),
+                                    ctypes.byref(cons), ctypes.byref(m)) == 0
+                ctx.free_batch(out_tr)
+                ctx.free_batch(cons)
+                out_tr = m
+        # insert the delta into the join trace spine
+        spine.append((d, None))
+        while len(spine) >= 2 and spine[-1][0].len * 2 >= spine[-2][0].len:
+            (b_top, own_top) = spine.pop()
+            (b_below, own_below) = spine.pop()
+            m = BatchStruct()
+            assert L.dbsp_merge(ctx._h, ctypes.byref(b_below),
+                                ctypes.byref(b_top), ctypes.byref(m)) == 0
+            for bb, own in ((b_top, own_top), (b_below, own_below)):
+                if own is not None:
+                    ctx.free_batch(bb)
+            spine.append((m, True))
+        ctx.sync()
+        step_ms.append((time.perf_counter() - t0) * 1e3)
+    res = {
+        "primitive": "C5 end-to-end: 1B trace x 10M delta incremental join "
+                     "+ f64 weigh/consolidate/sum-aggregate (i64 join, f64 agg)",
+        "trace_rows": n_trace, "delta_rows": n_delta, "steps": steps,
+        "agg_out_rows": out_tr.len if out_tr else 0,
+        "wb_rows": wb.len if wb else 0,
+        "ms_per_step": [round(x, 2) for x in step_ms],
+        "avg_ms_per_step": round(sum(step_ms[1:]) / max(1, len(step_ms) - 1), 2),
+    }
+    for b, own in spine:
+        if own is not None:
+            ctx.free_batch(b)
+    if wb is not None:
+        ctx.free_batch(wb)
+    if out_tr is not None:
+        ctx.free_batch(out_tr)
+    return res
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--rows", type=int, default=500_000_000,
@@ -209,6 +328,7 @@ def main():
     print(json.dumps(bench_join(ctx, L, 2 * args.rows, args.delta)))
     print(json.dumps(bench_sort(ctx, L, args.sort_rows)))
     print(json.dumps(bench_incremental(ctx, L, 2 * args.rows, args.delta)))
+    print(json.dumps(bench_c5(ctx, L, 2 * args.rows, args.delta)))
     ctx.close()
 
 
