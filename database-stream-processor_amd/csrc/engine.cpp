@@ -144,13 +144,13 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
 extern "C" dbsp_status dbsp_ctx_destroy(dbsp_ctx *c) {
     if (!c) return DBSP_OK;
     if (c->comm) ncclCommDestroy(c->comm);
-    hipStreamSynchronize(c->stream);
+    (void)hipStreamSynchronize(c->stream);
     if (c->d_len) (void)hipFree(c->d_len);
     if (c->arena) (void)hipFree(c->arena);
     if (c->h_len) (void)hipHostFree(c->h_len);
-    hipEventDestroy(c->ev0);
-    hipEventDestroy(c->ev1);
-    hipStreamDestroy(c->stream);
+    (void)hipEventDestroy(c->ev0);
+    (void)hipEventDestroy(c->ev1);
+    (void)hipStreamDestroy(c->stream);
     delete c;
     return DBSP_OK;
 }
@@ -196,10 +196,10 @@ struct ScopedTimer {
     ~ScopedTimer() {
         c->timer_depth--;
         if (active) {
-            hipEventRecord(c->ev1, c->stream);
-            hipEventSynchronize(c->ev1);
+            (void)hipEventRecord(c->ev1, c->stream);
+            (void)hipEventSynchronize(c->ev1);
             float ms = 0;
-            hipEventElapsedTime(&ms, c->ev0, c->ev1);
+            (void)hipEventElapsedTime(&ms, c->ev0, c->ev1);
             c->stats[cls].ms += ms;
             c->stats[cls].bytes += bytes;
             c->stats[cls].launches += 1;
@@ -218,9 +218,9 @@ struct DevBatch {
 };
 
 static void free_batch(dbsp_ctx *c, DevBatch &b) {
-    if (b.k && !in_arena(c, b.k)) hipFreeAsync(b.k, c->stream);
-    if (b.v && !in_arena(c, b.v)) hipFreeAsync(b.v, c->stream);
-    if (b.w && !in_arena(c, b.w)) hipFreeAsync(b.w, c->stream);
+    if (b.k && !in_arena(c, b.k)) (void)hipFreeAsync(b.k, c->stream);
+    if (b.v && !in_arena(c, b.v)) (void)hipFreeAsync(b.v, c->stream);
+    if (b.w && !in_arena(c, b.w)) (void)hipFreeAsync(b.w, c->stream);
     b = DevBatch{};
 }
 
@@ -911,8 +911,8 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
     free_batch(c, e->maxout_int);
     free_batch(c, e->maxz_int);
     free_batch(c, e->output);
-    if (e->d_events) hipFreeAsync(e->d_events, c->stream);
-    hipStreamSynchronize(c->stream);
+    if (e->d_events) (void)hipFreeAsync(e->d_events, c->stream);
+    (void)hipStreamSynchronize(c->stream);
     delete e;
     return DBSP_OK;
 }
@@ -1765,7 +1765,7 @@ extern "C" dbsp_status dbsp_engine_step(dbsp_engine *e, const dbsp_event *events
         case 8: st = q8_step(e, d_ev, n); break;
         default: st = DBSP_ERR_INVALID;
     }
-    hipFreeAsync(d_ev, c->stream);
+    (void)hipFreeAsync(d_ev, c->stream);
     return st;
 }
 

@@ -1728,9 +1728,9 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
     HIP_CHECK(hipFreeAsync(pb, s));
     HIP_CHECK(hipFreeAsync(state, s));
     if (h_state[0] != 0 || (h_state[1] & 3ull) != 2ull) {
-        hipFreeAsync(rk, s);
-        hipFreeAsync(rv, s);
-        hipFreeAsync(rw, s);
+        (void)hipFreeAsync(rk, s);
+        (void)hipFreeAsync(rv, s);
+        (void)hipFreeAsync(rw, s);
         return DBSP_ERR_INTERNAL;
     }
     *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)(h_state[1] >> 2);
