@@ -191,7 +191,7 @@ struct ScopedTimer {
         // the OUTER class — one shared event pair
         active = c->profile && c->timer_depth == 0;
         c->timer_depth++;
-        if (active) hipEventRecord(c->ev0, c->stream);
+        if (active) (void)hipEventRecord(c->ev0, c->stream);
     }
     ~ScopedTimer() {
         c->timer_depth--;
