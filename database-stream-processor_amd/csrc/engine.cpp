@@ -1278,8 +1278,8 @@ static dbsp_status copy_batch(dbsp_ctx *c, const DevBatch &in, DevBatch &out) {
 // linear aggregate + upsert over spines (aggregate/mod.rs:479-547 +
 // upsert.rs:161-208; linear in both traces, so evaluated per spine batch)
 static dbsp_status agg_linear_spine(dbsp_ctx *c, const DevBatch &delta,
-                                    const Spine &in_trace,
-                                    const Spine &out_trace, DevBatch &out) {
+                                    const Spine &in_trace, Spine &out_trace,
+                                    DevBatch &out) {
     if (delta.n == 0) {
         out = DevBatch{};
         return DBSP_OK;
@@ -1297,21 +1297,79 @@ static dbsp_status agg_linear_spine(dbsp_ctx *c, const DevBatch &delta,
     std::vector<DevBatch> outs;
     // inserts
     DevBatch ins;
-    TRY(alloc_batch(c, nk, ins));
+    TRY(alloc_batch(c, nk, ins, true));
     int64_t n_ins = 0;
     TRY(dbspk::emit_nonzero(c->stream, keys, acc, nk, ins.k, ins.v, ins.w, &n_ins));
     ins.n = n_ins;
     if (ins.n > 0) outs.push_back(ins);
-    else free_batch(c, ins);
-    // retractions: agg kernel against each out_trace batch with an empty input
-    // trace emits exactly the upsert retractions
-    for (auto &b : out_trace.batches) {
-        DevBatch o;
-        TRY(dbspk::agg_linear_upsert_rows(c->stream, keys, nk, nullptr, nullptr,
-                                          nullptr, 0, b.k, b.v, b.w, b.n, &o.k,
-                                          &o.v, &o.w, &o.n));
-        if (o.n > 0) outs.push_back(o);
-        else free_batch(c, o);
+    // retractions ARE a join: probing the output trace with (key, _, -1) rows
+    // under proj (k, v2) emits exactly (key, old_val, -old_w) — the upsert
+    // contract (upsert.rs:180-195) — in one count/emit pair over the whole
+    // spine instead of a scan pipeline per batch.
+    if (!out_trace.batches.empty()) {
+        if ((int)out_trace.batches.size() > MAX_TRACE_BATCHES)
+            TRY(out_trace.consolidate_all(c));
+        TraceArgs t{};
+        for (auto &b : out_trace.batches) {
+            if (b.n == 0) continue;
+            t.k[t.nb] = b.k; t.v[t.nb] = b.v; t.w[t.nb] = b.w;
+            t.n[t.nb] = b.n; t.nb++;
+        }
+        if (t.nb > 0) {
+            // synthetic delta columns: v = 0, w = -1
+            uint64_t *dv = (uint64_t *)arena_alloc(c, (size_t)nk * 8 + 8);
+            int64_t *dw = (int64_t *)arena_alloc(c, (size_t)nk * 8 + 8);
+            DevBatch tmp;
+            if (!dv || !dw) {
+                TRY(alloc_batch(c, nk, tmp));
+                dv = tmp.k;
+                dw = (int64_t *)tmp.v;
+            }
+            HIP_CHECK_ST(hipMemsetAsync(dv, 0, nk * 8, c->stream));
+            HIP_CHECK_ST(hipMemsetAsync(dw, 0xFF, nk * 8, c->stream));  // -1
+            if (nk <= 8192) {
+                uint32_t *cnts = (uint32_t *)arena_alloc(c, (size_t)nk * t.nb * 4 + 8);
+                uint64_t *offsets = (uint64_t *)arena_alloc(c, (size_t)(nk + 1) * 8);
+                DevBatch tmp2;
+                if (!cnts || !offsets) {
+                    TRY(alloc_batch(c, nk * (t.nb + 2), tmp2));
+                    cnts = (uint32_t *)tmp2.k;
+                    offsets = (uint64_t *)tmp2.v;
+                }
+                JoinCountArgs jca{};
+                jca.np = 1;
+                jca.dk[0] = keys;
+                jca.nd[0] = nk;
+                jca.t[0] = t;
+                jca.cnts[0] = cnts;
+                jca.offsets[0] = offsets;
+                jca.d_total = c->d_len;
+                TRY(dbspk::join_count_scan_batch(c->stream, jca));
+                HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 8,
+                                            hipMemcpyDeviceToHost, c->stream));
+                HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+                int64_t total = c->h_len[0];
+                if (total > 0) {
+                    DevBatch o;
+                    TRY(alloc_batch(c, total, o, true));
+                    TRY(dbspk::join_emit_prepared(c->stream, keys, dv, dw, nk,
+                                                  t, cnts, offsets, total,
+                                                  DBSP_PROJ_HI_K_LO_V2, 0, o.k,
+                                                  o.v, o.w));
+                    o.n = total;
+                    outs.push_back(o);
+                }
+                free_batch(c, tmp2);
+            } else {
+                DevBatch o;
+                TRY(dbspk::join_spine_rows(c->stream, keys, dv, dw, nk, t,
+                                           DBSP_PROJ_HI_K_LO_V2, 0, &o.k, &o.v,
+                                           &o.w, &o.n));
+                if (o.n > 0) outs.push_back(o);
+                else free_batch(c, o);
+            }
+            free_batch(c, tmp);
+        }
     }
     HIP_CHECK_ST(hipFreeAsync(acc, c->stream));
     HIP_CHECK_ST(hipFreeAsync(keys, c->stream));
