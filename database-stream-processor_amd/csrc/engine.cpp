@@ -717,6 +717,26 @@ extern "C" dbsp_status dbsp_agg_linear_upsert_f64(dbsp_ctx *c,
     return DBSP_OK;
 }
 
+extern "C" dbsp_status dbsp_distinct_inc(dbsp_ctx *c, const dbsp_batch *delta,
+                                         const dbsp_batch *trace_batches,
+                                         int n_batches, dbsp_batch *out) {
+    if (n_batches > MAX_TRACE_BATCHES) return DBSP_ERR_INVALID;
+    TraceArgs t{};
+    for (int i = 0; i < n_batches; i++) {
+        if (trace_batches[i].len == 0) continue;
+        t.k[t.nb] = trace_batches[i].k;
+        t.v[t.nb] = trace_batches[i].v;
+        t.w[t.nb] = trace_batches[i].w;
+        t.n[t.nb] = trace_batches[i].len;
+        t.nb++;
+    }
+    DevBatch res;
+    TRY(dbspk::distinct_inc_rows(c->stream, delta->k, delta->v, delta->w,
+                                 delta->len, t, &res.k, &res.v, &res.w, &res.n));
+    out->k = res.k; out->v = res.v; out->w = res.w; out->len = res.n;
+    return DBSP_OK;
+}
+
 extern "C" dbsp_status dbsp_unique_keys(dbsp_ctx *c, const dbsp_batch *in,
                                         uint64_t **out_keys, int64_t *n_out) {
     TRY(dbspk::unique_keys(c->stream, in->k, in->len, out_keys, n_out));

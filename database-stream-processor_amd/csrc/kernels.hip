@@ -1238,6 +1238,47 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_join_count_scan_small(
 }
 
 // ---------------------------------------------------------------------------
+// incremental distinct (operator/distinct.rs:273,404-462 at root scope):
+// for each delta pair (k,v):  out = [w_before + dw > 0] - [w_before > 0]
+// where w_before is the pair's total weight in the delayed integral.  NOT
+// linear in the trace (the indicator needs the total), so one kernel probes
+// every spine batch.  Output positions via flags+scan keep delta order, so
+// the result is consolidated by construction.
+// ---------------------------------------------------------------------------
+
+__device__ inline int64_t find_pair_weight(const uint64_t *tk,
+                                           const uint64_t *tv,
+                                           const int64_t *tw, int64_t n,
+                                           uint64_t key, uint64_t val) {
+    int64_t lo = lower_bound_k(tk, n, key);
+    int64_t hi = lo + gallop_run(tk, n, key, lo);
+    // binary search the val within the key's run
+    while (lo < hi) {
+        int64_t mid = (lo + hi) / 2;
+        if (tv[mid] < val) lo = mid + 1;
+        else hi = mid;
+    }
+    if (lo < n && tk[lo] == key && tv[lo] == val) return tw[lo];
+    return 0;
+}
+
+__global__ void k_distinct_count(const uint64_t *dk, const uint64_t *dv,
+                                 const int64_t *dw, int64_t nd, TraceArgs t,
+                                 uint64_t *flags, int64_t *outw) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < nd;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int64_t before = 0;
+        for (int b = 0; b < t.nb; b++)
+            before += find_pair_weight(t.k[b], t.v[b], t.w[b], t.n[b], dk[i],
+                                       dv[i]);
+        int64_t after = before + dw[i];
+        int64_t o = (int64_t)(after > 0) - (int64_t)(before > 0);
+        flags[i] = o != 0;
+        outw[i] = o;
+    }
+}
+
+// ---------------------------------------------------------------------------
 // aggregate (linear / max) + upsert  (count/emit over delta keys)
 // ---------------------------------------------------------------------------
 
@@ -1910,6 +1951,37 @@ static dbsp_status agg_upsert_impl(hipStream_t s, const uint64_t *keys,
     HIP_CHECK(hipFreeAsync(newval, s));
     HIP_CHECK(hipFreeAsync(hasnew, s));
     HIP_CHECK(hipFreeAsync(offsets, s));
+    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
+    return DBSP_OK;
+}
+
+dbsp_status distinct_inc_rows(hipStream_t s, const uint64_t *dk,
+                              const uint64_t *dv, const int64_t *dw,
+                              int64_t nd, const TraceArgs &t, uint64_t **ok,
+                              uint64_t **ov, int64_t **ow, int64_t *out_n) {
+    if (nd == 0) {
+        *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
+        return DBSP_OK;
+    }
+    uint64_t *flags, *fscan;
+    int64_t *outw;
+    HIP_CHECK(hipMallocAsync(&flags, nd * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&fscan, nd * sizeof(uint64_t), s));
+    HIP_CHECK(hipMallocAsync(&outw, nd * sizeof(int64_t), s));
+    k_distinct_count<<<grid_for(nd), BLK, 0, s>>>(dk, dv, dw, nd, t, flags, outw);
+    uint64_t nout = 0;
+    dbsp_status st = scan_exclusive(s, flags, fscan, nd, &nout);
+    if (st != DBSP_OK) return st;
+    uint64_t *rk, *rv;
+    int64_t *rw;
+    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    k_compact<<<grid_for(nd), BLK, 0, s>>>(dk, dv, outw, flags, fscan, nd, rk,
+                                           rv, rw);
+    HIP_CHECK(hipFreeAsync(flags, s));
+    HIP_CHECK(hipFreeAsync(fscan, s));
+    HIP_CHECK(hipFreeAsync(outw, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
     return DBSP_OK;
 }

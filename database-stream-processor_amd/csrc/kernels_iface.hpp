@@ -114,6 +114,13 @@ dbsp_status join_rows(hipStream_t s, const uint64_t *dk, const uint64_t *dv,
                       int proj, uint64_t param, uint64_t **ok, uint64_t **ov,
                       int64_t **ow, int64_t *out_n);
 
+// incremental distinct over a whole spine (non-linear: the per-pair total
+// must be summed across batches before the indicator)
+dbsp_status distinct_inc_rows(hipStream_t s, const uint64_t *dk,
+                              const uint64_t *dv, const int64_t *dw,
+                              int64_t nd, const TraceArgs &t, uint64_t **ok,
+                              uint64_t **ov, int64_t **ow, int64_t *out_n);
+
 dbsp_status agg_linear_upsert_rows(hipStream_t s, const uint64_t *keys,
                                    int64_t nd, const uint64_t *ik,
                                    const uint64_t *iv, const int64_t *iw,

@@ -83,6 +83,11 @@ def _L():
                                                  ctypes.POINTER(BatchStruct),
                                                  ctypes.POINTER(BatchStruct),
                                                  ctypes.POINTER(BatchStruct)]
+        L.dbsp_distinct_inc.restype = i32
+        L.dbsp_distinct_inc.argtypes = [vp, ctypes.POINTER(BatchStruct),
+                                        ctypes.POINTER(BatchStruct),
+                                        ctypes.c_int,
+                                        ctypes.POINTER(BatchStruct)]
         L.dbsp_unique_keys.restype = i32
         L.dbsp_unique_keys.argtypes = [vp, ctypes.POINTER(BatchStruct),
                                        ctypes.POINTER(vp),
@@ -323,6 +328,22 @@ class Ctx:
         self._lib.dbsp_dev_free(self._h, dk)
         for x in (it, ot, out):
             self.free_batch(x)
+        return res
+
+    def distinct_inc(self, delta_rows, trace_batch_rows_list):
+        d = self.upload_rows(delta_rows)
+        batches = [self.upload_rows(b) for b in trace_batch_rows_list]
+        arr = (BatchStruct * max(len(batches), 1))(*batches)
+        out = BatchStruct()
+        _check(self._lib.dbsp_distinct_inc(self._h, ctypes.byref(d), arr,
+                                           len(batches), ctypes.byref(out)),
+               "distinct")
+        self.sync()
+        res = self.download_rows(out)
+        self.free_batch(d)
+        for b in batches:
+            self.free_batch(b)
+        self.free_batch(out)
         return res
 
     def shard_partition(self, rows, nshards):

@@ -48,6 +48,8 @@ def _o():
         L.oracle_weigh_f64.argtypes = [vp, i64, vp]
         L.oracle_agg_linear_upsert_f64.restype = i64
         L.oracle_agg_linear_upsert_f64.argtypes = [vp, i64, vp, i64, vp, i64, vp, i64]
+        L.oracle_distinct_inc.restype = i64
+        L.oracle_distinct_inc.argtypes = [vp, i64, vp, i64, vp]
     return _lib
 
 
@@ -168,4 +170,13 @@ def agg_linear_upsert_f64(keys, in_trace, out_trace, cap=1 << 20):
                                           len(in_trace), _p(out_trace),
                                           len(out_trace), _p(out), cap)
     assert n >= 0
+    return out[:n].copy()
+
+
+def distinct_inc(delta, trace):
+    out = np.empty(max(len(delta), 1), dtype=ROW_DT)
+    n = _o().oracle_distinct_inc(_p(np.ascontiguousarray(delta, dtype=ROW_DT)),
+                                 len(delta),
+                                 _p(np.ascontiguousarray(trace, dtype=ROW_DT)),
+                                 len(trace), _p(out))
     return out[:n].copy()

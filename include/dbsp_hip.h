@@ -217,6 +217,16 @@ dbsp_status dbsp_agg_linear_upsert_f64(dbsp_ctx *ctx,
                                        const dbsp_batch *out_trace,
                                        dbsp_batch *out_raw);
 
+/* Incremental distinct (operator/distinct.rs:273: DistinctIncremental at
+ * root scope): for each delta pair, out = [w_before + dw > 0] - [w_before > 0]
+ * where w_before is the pair's total weight in the delayed integral (passed
+ * as its spine batches: distinct is NOT linear in the trace).  The output is
+ * consolidated by construction.  Unlocks antijoin/outer-join and Nexmark
+ * q4/q6/q9. */
+dbsp_status dbsp_distinct_inc(dbsp_ctx *ctx, const dbsp_batch *delta,
+                              const dbsp_batch *trace_batches, int n_batches,
+                              dbsp_batch *out);
+
 /* Distinct keys of a consolidated batch (device out, caller frees). */
 dbsp_status dbsp_unique_keys(dbsp_ctx *ctx, const dbsp_batch *in,
                              uint64_t **out_keys, int64_t *n_out);

@@ -545,6 +545,23 @@ int64_t oracle_window(const dbsp_row *trace, int64_t nt, const dbsp_row *batch,
 
 uint64_t oracle_xxh3_u64(uint64_t key, uint64_t seed) { return xxh3_u64(key, seed); }
 
+// Incremental distinct (operator/distinct.rs:404-462 at root scope, depth 1:
+// distinct_vals = {f(t), f(t-1)}; partial derivative = dist(after)-dist(before))
+int64_t oracle_distinct_inc(const dbsp_row *delta, int64_t nd,
+                            const dbsp_row *trace, int64_t nt, dbsp_row *out) {
+    int64_t o = 0;
+    for (int64_t i = 0; i < nd; i++) {
+        int64_t before = 0;
+        for (int64_t j = 0; j < nt; j++)
+            if (trace[j].k == delta[i].k && trace[j].v == delta[i].v)
+                before += trace[j].w;
+        int64_t after = before + delta[i].w;
+        int64_t w = (int64_t)(after > 0) - (int64_t)(before > 0);
+        if (w != 0) out[o++] = {delta[i].k, delta[i].v, w};
+    }
+    return o;
+}
+
 // ---- f64-weight variants (config C5).  The oracle is the SEQUENTIAL
 // reference: sums accumulate left-to-right in sorted order; the GPU's
 // position-fixed tree order is compared against it within the stated

@@ -163,3 +163,25 @@ def test_device_xxh3_matches_oracle(ctx):
     rng = np.random.default_rng(8)
     for k in [0, 1, 2**63] + [int(x) for x in rng.integers(0, 2**63, 20)]:
         assert xxh3_u64(k) == oracle.xxh3_u64(k)
+
+
+def test_distinct_inc_parity(ctx):
+    """Incremental distinct (operator/distinct.rs root scope): delta vs a
+    multi-batch integral whose per-pair totals only emerge across batches."""
+    rng = np.random.default_rng(41)
+    b1 = oracle.consolidate(_random_rows(rng, 3000, key_range=100, val_range=4))
+    b2 = oracle.consolidate(_random_rows(rng, 3000, key_range=100, val_range=4))
+    delta = oracle.consolidate(_random_rows(rng, 2000, key_range=100, val_range=4))
+    merged = oracle.merge(b1, b2)
+    got = ctx.distinct_inc(delta, [b1, b2])
+    exp = oracle.distinct_inc(delta, merged)
+    assert np.array_equal(got, exp)
+    # adversarial: weights cancelling across batches (before == 0), and
+    # transitions in both directions
+    neg = b1.copy()
+    neg["w"] = -neg["w"]
+    got = ctx.distinct_inc(delta, [b1, neg])
+    exp = oracle.distinct_inc(delta, np.empty(0, dtype=ROW_DT))
+    assert np.array_equal(got, exp)
+    # empty trace and empty delta
+    assert len(ctx.distinct_inc(np.empty(0, dtype=ROW_DT), [b1])) == 0

@@ -270,3 +270,22 @@ def test_oracle_f64_model():
     z["k"] = [1, 1]
     z["w"] = np.array([0.5, -0.5]).view(np.int64)
     assert len(oracle.consolidate_f64(z)) == 0
+
+
+def test_oracle_distinct_model():
+    """Distinct indicator-difference vs a brute-force model
+    (reference distinct.rs proptests pattern)."""
+    rng = np.random.default_rng(42)
+    trace = oracle.consolidate(_random_rows(rng, 500, key_range=30, val_range=3))
+    delta = oracle.consolidate(_random_rows(rng, 200, key_range=30, val_range=3))
+    got = zset(oracle.distinct_inc(delta, trace))
+    exp = {}
+    tr = {(int(r["k"]), int(r["v"])): int(r["w"]) for r in trace}
+    for r in delta:
+        key = (int(r["k"]), int(r["v"]))
+        before = tr.get(key, 0)
+        after = before + int(r["w"])
+        w = int(after > 0) - int(before > 0)
+        if w:
+            exp[key] = w
+    assert got == exp
