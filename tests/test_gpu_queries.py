@@ -131,6 +131,15 @@ def test_q3_duplicate_and_zero_weight_events(ctx):
     _run_parity(ctx, 3, mixed, tick=1000, seed_note="+dups")
 
 
+@pytest.mark.parametrize("query", [3, 5, 8])
+def test_query_parity_at_scale(ctx, query):
+    """2M events (50 reference-sized ticks): traces reach hundreds of
+    thousands of rows, exercising spine cascades, the medium sort path and
+    multi-batch probes end to end."""
+    evs = gen.generate(2_000_000, seed=29)
+    _run_parity(ctx, query, evs, tick=40_000)
+
+
 def test_q0_host_path(ctx):
     """q0 runs the host plumbing path (BASELINE configs[0]: CPU, 1 worker)."""
     from dbsp_amd.engine import Engine
