@@ -146,6 +146,8 @@ extern "C" dbsp_status dbsp_ctx_destroy(dbsp_ctx *c) {
     if (c->comm) ncclCommDestroy(c->comm);
     (void)hipStreamSynchronize(c->stream);
     if (c->d_len) (void)hipFree(c->d_len);
+    dbspk::cache_trim(c->stream);
+    (void)hipStreamSynchronize(c->stream);
     if (c->arena) (void)hipFree(c->arena);
     if (c->h_len) (void)hipHostFree(c->h_len);
     (void)hipEventDestroy(c->ev0);
@@ -165,7 +167,7 @@ extern "C" dbsp_status dbsp_dev_alloc(dbsp_ctx *c, size_t bytes, void **out) {
     return DBSP_OK;
 }
 extern "C" dbsp_status dbsp_dev_free(dbsp_ctx *c, void *p) {
-    HIP_CHECK_ST(hipFreeAsync(p, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(p, c->stream));
     return DBSP_OK;
 }
 extern "C" dbsp_status dbsp_h2d(dbsp_ctx *c, void *dst, const void *src,
@@ -218,9 +220,9 @@ struct DevBatch {
 };
 
 static void free_batch(dbsp_ctx *c, DevBatch &b) {
-    if (b.k && !in_arena(c, b.k)) (void)hipFreeAsync(b.k, c->stream);
-    if (b.v && !in_arena(c, b.v)) (void)hipFreeAsync(b.v, c->stream);
-    if (b.w && !in_arena(c, b.w)) (void)hipFreeAsync(b.w, c->stream);
+    if (b.k && !in_arena(c, b.k)) (void)dbspk::cache_free(b.k, c->stream);
+    if (b.v && !in_arena(c, b.v)) (void)dbspk::cache_free(b.v, c->stream);
+    if (b.w && !in_arena(c, b.w)) (void)dbspk::cache_free(b.w, c->stream);
     b = DevBatch{};
 }
 
@@ -237,9 +239,9 @@ static dbsp_status alloc_batch(dbsp_ctx *c, int64_t n, DevBatch &b,
             return DBSP_OK;
         }
     }
-    HIP_CHECK_ST(hipMallocAsync(&b.k, n * sizeof(uint64_t) + 8, c->stream));
-    HIP_CHECK_ST(hipMallocAsync(&b.v, n * sizeof(uint64_t) + 8, c->stream));
-    HIP_CHECK_ST(hipMallocAsync(&b.w, n * sizeof(int64_t) + 8, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&b.k, n * sizeof(uint64_t) + 8, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&b.v, n * sizeof(uint64_t) + 8, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&b.w, n * sizeof(int64_t) + 8, c->stream));
     return DBSP_OK;
 }
 
@@ -677,7 +679,7 @@ extern "C" dbsp_status dbsp_agg_linear_upsert_f64(dbsp_ctx *c,
         return DBSP_OK;
     }
     double *acc;
-    HIP_CHECK_ST(hipMallocAsync(&acc, nd * 8 + 8, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&acc, nd * 8 + 8, c->stream));
     HIP_CHECK_ST(hipMemsetAsync(acc, 0, nd * 8, c->stream));
     TRY(dbspk::agg_sum_batch_f64(c->stream, delta_keys, nd, in_trace->k,
                                  (const double *)in_trace->w, in_trace->len,
@@ -688,7 +690,7 @@ extern "C" dbsp_status dbsp_agg_linear_upsert_f64(dbsp_ctx *c,
     TRY(dbspk::emit_nonzero_f64(c->stream, delta_keys, acc, nd, ins.k, ins.v,
                                 ins.w, &n_ins));
     ins.n = n_ins;
-    HIP_CHECK_ST(hipFreeAsync(acc, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(acc, c->stream));
     // retractions against the (i64-weighted) output trace
     DevBatch retr;
     TRY(dbspk::agg_linear_upsert_rows(c->stream, delta_keys, nd, nullptr,
@@ -777,8 +779,8 @@ static dbsp_status alltoallv_cols(dbsp_ctx *c, const DevBatch &send,
     int world = c->world;
     // exchange counts (device staging for RCCL)
     int64_t *d_send_cnt, *d_recv_cnt;
-    HIP_CHECK_ST(hipMallocAsync(&d_send_cnt, world * 8, c->stream));
-    HIP_CHECK_ST(hipMallocAsync(&d_recv_cnt, world * 8, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_send_cnt, world * 8, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_recv_cnt, world * 8, c->stream));
     HIP_CHECK_ST(hipMemcpyAsync(d_send_cnt, send_counts, world * 8,
                                 hipMemcpyHostToDevice, c->stream));
     ncclGroupStart();
@@ -812,8 +814,8 @@ static dbsp_status alltoallv_cols(dbsp_ctx *c, const DevBatch &send,
         roff += recv_counts[r];
     }
     ncclGroupEnd();
-    HIP_CHECK_ST(hipFreeAsync(d_send_cnt, c->stream));
-    HIP_CHECK_ST(hipFreeAsync(d_recv_cnt, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(d_send_cnt, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(d_recv_cnt, c->stream));
     recv.n = recv_total;
     return DBSP_OK;
 }
@@ -839,7 +841,7 @@ static inline bool sharding_on(dbsp_ctx *c) {
 static dbsp_status allreduce_max_u64(dbsp_ctx *c, uint64_t *h_val) {
     if (!sharding_on(c)) return DBSP_OK;
     uint64_t *d;
-    HIP_CHECK_ST(hipMallocAsync(&d, sizeof(uint64_t), c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&d, sizeof(uint64_t), c->stream));
     HIP_CHECK_ST(hipMemcpyAsync(d, h_val, sizeof(uint64_t),
                                 hipMemcpyHostToDevice, c->stream));
     if (ncclAllReduce(d, d, 1, ncclUint64, ncclMax, c->comm, c->stream) !=
@@ -848,7 +850,7 @@ static dbsp_status allreduce_max_u64(dbsp_ctx *c, uint64_t *h_val) {
     HIP_CHECK_ST(hipMemcpyAsync(h_val, d, sizeof(uint64_t),
                                 hipMemcpyDeviceToHost, c->stream));
     HIP_CHECK_ST(hipStreamSynchronize(c->stream));
-    HIP_CHECK_ST(hipFreeAsync(d, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(d, c->stream));
     return DBSP_OK;
 }
 
@@ -931,7 +933,7 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
     free_batch(c, e->maxout_int);
     free_batch(c, e->maxz_int);
     free_batch(c, e->output);
-    if (e->d_events) (void)hipFreeAsync(e->d_events, c->stream);
+    if (e->d_events) (void)dbspk::cache_free(e->d_events, c->stream);
     (void)hipStreamSynchronize(c->stream);
     delete e;
     return DBSP_OK;
@@ -942,10 +944,10 @@ extern "C" dbsp_status dbsp_engine_stage_events(dbsp_engine *e,
                                                 int64_t n) {
     dbsp_ctx *c = e->ctx;
     if (e->d_events) {
-        HIP_CHECK_ST(hipFreeAsync(e->d_events, c->stream));
+        HIP_CHECK_ST(dbspk::cache_free(e->d_events, c->stream));
         e->d_events = nullptr;
     }
-    HIP_CHECK_ST(hipMallocAsync(&e->d_events, n * sizeof(dbsp_event) + 64, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&e->d_events, n * sizeof(dbsp_event) + 64, c->stream));
     HIP_CHECK_ST(hipMemcpyAsync(e->d_events, events, n * sizeof(dbsp_event),
                                 hipMemcpyHostToDevice, c->stream));
     HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -1054,8 +1056,8 @@ static dbsp_status shard_exchange_pair(dbsp_ctx *c, DevBatch l0, DevBatch l1,
         send_cnt[2 * r + 1] = off1[r + 1] - off1[r];
     }
     int64_t *d_snd, *d_rcv;
-    HIP_CHECK_ST(hipMallocAsync(&d_snd, 2 * world * 8, c->stream));
-    HIP_CHECK_ST(hipMallocAsync(&d_rcv, 2 * world * 8, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_snd, 2 * world * 8, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_rcv, 2 * world * 8, c->stream));
     HIP_CHECK_ST(hipMemcpyAsync(d_snd, send_cnt, 2 * world * 8,
                                 hipMemcpyHostToDevice, c->stream));
     ncclGroupStart();
@@ -1105,8 +1107,8 @@ static dbsp_status shard_exchange_pair(dbsp_ctx *c, DevBatch l0, DevBatch l1,
         ro1 += recv_cnt[2 * r + 1];
     }
     ncclGroupEnd();
-    HIP_CHECK_ST(hipFreeAsync(d_snd, c->stream));
-    HIP_CHECK_ST(hipFreeAsync(d_rcv, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(d_snd, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(d_rcv, c->stream));
     r0.n = tot0;
     r1.n = tot1;
     if (r0.n <= 8192 && r1.n <= 8192) {
@@ -1310,7 +1312,7 @@ static dbsp_status agg_linear_spine(dbsp_ctx *c, const DevBatch &delta,
     TRY(dbspk::unique_keys(c->stream, delta.k, delta.n, &keys, &nk));
     // per-key weight sums across in_trace batches
     int64_t *acc;
-    HIP_CHECK_ST(hipMallocAsync(&acc, nk * 8 + 8, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&acc, nk * 8 + 8, c->stream));
     HIP_CHECK_ST(hipMemsetAsync(acc, 0, nk * 8, c->stream));
     for (auto &b : in_trace.batches)
         TRY(dbspk::agg_sum_batch(c->stream, keys, nk, b.k, b.w, b.n, acc));
@@ -1391,8 +1393,8 @@ static dbsp_status agg_linear_spine(dbsp_ctx *c, const DevBatch &delta,
             free_batch(c, tmp);
         }
     }
-    HIP_CHECK_ST(hipFreeAsync(acc, c->stream));
-    HIP_CHECK_ST(hipFreeAsync(keys, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(acc, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(keys, c->stream));
     TRY(finalize_raw(c, outs, out));
     return DBSP_OK;
 }
@@ -1754,14 +1756,14 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         e->maxin_int = m;
         // delta key is the unit key ()
         uint64_t *d_unit;
-        HIP_CHECK_ST(hipMallocAsync(&d_unit, 8, c->stream));
+        HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_unit, 8, c->stream));
         HIP_CHECK_ST(hipMemsetAsync(d_unit, 0, 8, c->stream));
         DevBatch raw;
         TRY(dbspk::agg_max_upsert_rows(
             c->stream, d_unit, 1, e->maxin_int.k, e->maxin_int.v, e->maxin_int.w,
             e->maxin_int.n, e->maxout_int.k, e->maxout_int.v, e->maxout_int.w,
             e->maxout_int.n, &raw.k, &raw.v, &raw.w, &raw.n));
-        HIP_CHECK_ST(hipFreeAsync(d_unit, c->stream));
+        HIP_CHECK_ST(dbspk::cache_free(d_unit, c->stream));
         TRY(sort_consolidate_batch(c, raw, dMaxOut));
         DevBatch m2;
         TRY(merge_batches(c, e->maxout_int, dMaxOut, m2));
@@ -1833,7 +1835,7 @@ extern "C" dbsp_status dbsp_engine_step(dbsp_engine *e, const dbsp_event *events
     dbsp_ctx *c = e->ctx;
     c->arena_off = 0;
     dbsp_event *d_ev;
-    HIP_CHECK_ST(hipMallocAsync(&d_ev, n * sizeof(dbsp_event) + 64, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_ev, n * sizeof(dbsp_event) + 64, c->stream));
     HIP_CHECK_ST(hipMemcpyAsync(d_ev, events, n * sizeof(dbsp_event),
                                 hipMemcpyHostToDevice, c->stream));
     dbsp_status st = DBSP_OK;
@@ -1843,7 +1845,7 @@ extern "C" dbsp_status dbsp_engine_step(dbsp_engine *e, const dbsp_event *events
         case 8: st = q8_step(e, d_ev, n); break;
         default: st = DBSP_ERR_INVALID;
     }
-    (void)hipFreeAsync(d_ev, c->stream);
+    (void)dbspk::cache_free(d_ev, c->stream);
     return st;
 }
 

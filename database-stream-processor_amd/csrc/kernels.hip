@@ -30,6 +30,83 @@
 #include "../../include/dbsp_hip.h"
 #include "kernels_iface.hpp"
 
+// ---------------------------------------------------------------------------
+// Big-buffer cache.  hipMallocAsync's pool hands a freed multi-GB block back
+// to the SAME size-class only a couple of allocations later, so trace-scale
+// merges (6.5 GB per output column at 1B rows) were re-carving physical
+// pages on almost every call at ~9 GB/s — a 720 ms host stall per column
+// (profiles/r01_merge_host_stall.txt).  All product allocations go through
+// this free-list instead: blocks >= CACHE_MIN bytes are kept and reused
+// best-fit (everything runs on the one ctx stream, so reuse is ordered by
+// stream order alone); smaller ones pass through to the pool, which handles
+// tick-scale buffers fine.  On allocation failure the cache is dropped and
+// the carve retried.
+// ---------------------------------------------------------------------------
+#include <map>
+#include <mutex>
+#include <unordered_map>
+
+namespace dbspk {
+namespace {
+constexpr size_t CACHE_MIN = 32u << 20;
+std::mutex g_cache_mu;
+std::unordered_map<void *, size_t> g_cache_live;
+std::multimap<size_t, void *> g_cache_free;
+
+void cache_drop_locked(hipStream_t s) {
+    for (auto &e : g_cache_free) (void)hipFreeAsync(e.second, s);
+    g_cache_free.clear();
+}
+}  // namespace
+
+hipError_t cache_malloc(void **out, size_t bytes, hipStream_t s) {
+    if (bytes >= CACHE_MIN) {
+        std::lock_guard<std::mutex> g(g_cache_mu);
+        auto it = g_cache_free.lower_bound(bytes);
+        // accept up to 50% + 64 MB waste; beyond that carve fresh
+        if (it != g_cache_free.end() &&
+            it->first <= bytes + bytes / 2 + (64u << 20)) {
+            *out = it->second;
+            g_cache_live.emplace(it->second, it->first);
+            g_cache_free.erase(it);
+            return hipSuccess;
+        }
+    }
+    hipError_t e = hipMallocAsync(out, bytes, s);
+    if (e != hipSuccess) {
+        std::lock_guard<std::mutex> g(g_cache_mu);
+        cache_drop_locked(s);
+        (void)hipStreamSynchronize(s);
+        e = hipMallocAsync(out, bytes, s);
+    }
+    if (e == hipSuccess && bytes >= CACHE_MIN) {
+        std::lock_guard<std::mutex> g(g_cache_mu);
+        g_cache_live.emplace(*out, bytes);
+    }
+    return e;
+}
+
+hipError_t cache_free(void *p, hipStream_t s) {
+    if (!p) return hipSuccess;
+    {
+        std::lock_guard<std::mutex> g(g_cache_mu);
+        auto it = g_cache_live.find(p);
+        if (it != g_cache_live.end()) {
+            g_cache_free.emplace(it->second, p);
+            g_cache_live.erase(it);
+            return hipSuccess;
+        }
+    }
+    return hipFreeAsync(p, s);
+}
+
+void cache_trim(hipStream_t s) {
+    std::lock_guard<std::mutex> g(g_cache_mu);
+    cache_drop_locked(s);
+}
+}  // namespace dbspk
+
+
 #define WAVE 64
 #define BLK 256
 #define SORT_ITEMS 8
@@ -153,7 +230,7 @@ static dbsp_status scan_exclusive(hipStream_t s, const uint64_t *in,
     uint64_t *tot[5] = {nullptr};
     for (int l = 0; l <= levels; l++) {
         int64_t nb = ceil_div(sizes[l], SORT_TILE);
-        HIP_CHECK(hipMallocAsync(&tot[l], (nb + 1) * sizeof(uint64_t), s));
+        HIP_CHECK(dbspk::cache_malloc((void **)&tot[l], (nb + 1) * sizeof(uint64_t), s));
     }
     // down-sweep: scan each level, producing block totals
     const uint64_t *src = in;
@@ -189,7 +266,7 @@ static dbsp_status scan_exclusive(hipStream_t s, const uint64_t *in,
             HIP_CHECK(hipStreamSynchronize(s));
         }
     }
-    for (int l = 0; l <= levels; l++) HIP_CHECK(hipFreeAsync(tot[l], s));
+    for (int l = 0; l <= levels; l++) HIP_CHECK(dbspk::cache_free(tot[l], s));
     return DBSP_OK;
 }
 
@@ -932,6 +1009,119 @@ __device__ inline void adjust_split_lds(const uint64_t *lk, const uint64_t *lv,
 //   granule = value<<2 | flag;  flag: 0 invalid, 1 aggregate, 2 prefix
 // ---------------------------------------------------------------------------
 
+template <bool EMIT, typename W>
+__global__ __launch_bounds__(MP_THREADS, 2) void k_mp_merge(
+    const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
+    const uint64_t *bk, const uint64_t *bv, const W *bw, int64_t nb,
+    const int64_t *pa, const int64_t *pb,
+    uint64_t *counts,  // COUNT: per-block totals out; EMIT: scanned offsets in
+    uint64_t *ok, uint64_t *ov, W *ow) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    uint64_t *lk = (uint64_t *)smem;          // tile + 2: the split adjustment
+    uint64_t *lv = lk + (MP_TILE + 2);        // can grow a block by one row
+    W *lw = (W *)(lv + (MP_TILE + 2));        // staged only when EMIT
+    __shared__ uint32_t wt[MP_THREADS / WAVE + 1];
+    const int tid = threadIdx.x;
+    const int64_t blk = blockIdx.x;
+    const int64_t pa0 = pa[blk], pa1 = pa[blk + 1];
+    const int64_t pb0 = pb[blk], pb1 = pb[blk + 1];
+    const int64_t naL = pa1 - pa0, nbL = pb1 - pb0;
+    const int64_t totL = naL + nbL;
+    // stage (coalesced 8B per lane)
+    for (int64_t i = tid; i < naL; i += MP_THREADS) {
+        lk[i] = ak[pa0 + i];
+        lv[i] = av[pa0 + i];
+        if (EMIT) lw[i] = aw[pa0 + i];
+    }
+    for (int64_t i = tid; i < nbL; i += MP_THREADS) {
+        lk[naL + i] = bk[pb0 + i];
+        lv[naL + i] = bv[pb0 + i];
+        if (EMIT) lw[naL + i] = bw[pb0 + i];
+    }
+    __syncthreads();
+    // per-thread diagonals within the tile
+    const int64_t items = (totL + MP_THREADS - 1) / MP_THREADS;
+    int64_t d0 = min((int64_t)tid * items, totL);
+    int64_t d1 = min(d0 + items, totL);
+    int64_t ai, bi, ae, be;
+    merge_path_lds(lk, lv, naL, nbL, d0, ai, bi);
+    adjust_split_lds(lk, lv, naL, nbL, ai, bi);
+    merge_path_lds(lk, lv, naL, nbL, d1, ae, be);
+    adjust_split_lds(lk, lv, naL, nbL, ae, be);
+    // count walk (weights touched only on equal pairs in COUNT: rare, global)
+    uint32_t cnt = 0;
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
+                W s = EMIT ? (lw[i] + lw[naL + j])
+                           : (aw[pa0 + i] + bw[pb0 + j]);
+                if (s != (W)0) cnt++;
+                i++; j++;
+            } else if (j >= be ||
+                       (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
+                cnt++; i++;
+            } else {
+                cnt++; j++;
+            }
+        }
+    }
+    if (!EMIT) {
+        // block total
+        uint32_t v = cnt;
+        for (int d = 1; d < WAVE; d <<= 1) {
+            uint32_t up = __shfl_up(v, d, WAVE);
+            if ((tid & (WAVE - 1)) >= d) v += up;
+        }
+        if ((tid & (WAVE - 1)) == WAVE - 1) wt[tid / WAVE] = v;
+        __syncthreads();
+        if (tid == 0) {
+            uint32_t acc = 0;
+            for (int w = 0; w < MP_THREADS / WAVE; w++) acc += wt[w];
+            counts[blk] = acc;
+        }
+        return;
+    }
+    // EMIT: exclusive scan of thread counts -> local offsets
+    uint32_t off;
+    {
+        uint32_t v = cnt;
+        for (int d = 1; d < WAVE; d <<= 1) {
+            uint32_t up = __shfl_up(v, d, WAVE);
+            if ((tid & (WAVE - 1)) >= d) v += up;
+        }
+        if ((tid & (WAVE - 1)) == WAVE - 1) wt[tid / WAVE] = v;
+        __syncthreads();
+        if (tid == 0) {
+            uint32_t acc = 0;
+            for (int w = 0; w < MP_THREADS / WAVE; w++) {
+                uint32_t t = wt[w];
+                wt[w] = acc;
+                acc += t;
+            }
+        }
+        __syncthreads();
+        off = wt[tid / WAVE] + (v - cnt);
+    }
+    uint64_t gpos = counts[blk] + off;
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
+                W s = lw[i] + lw[naL + j];
+                if (s != (W)0) { ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = s; gpos++; }
+                i++; j++;
+            } else if (j >= be ||
+                       (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
+                ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = lw[i]; gpos++; i++;
+            } else {
+                ok[gpos] = lk[naL + j]; ov[gpos] = lv[naL + j]; ow[gpos] = lw[naL + j];
+                gpos++; j++;
+            }
+        }
+    }
+}
+
 typedef __attribute__((address_space(1))) unsigned long long gu64_t;
 
 template <typename W>
@@ -1034,8 +1224,15 @@ __global__ __launch_bounds__(MP_THREADS, 3) void k_mp_merge_onepass(
                 prefix += e >> 2;
                 p--;
             } else {
-                __builtin_amdgcn_s_sleep(1);
-                if (++spins > (1u << 24)) {  // bounded spin: poison, don't hang
+                // Exponential backoff: with ~1k resident pollers a hot
+                // (64-cycle) poll loop throttles the L2/fabric enough to
+                // stall the very aggregates being waited for — observed as a
+                // bimodal ~30x collapse of the whole merge.
+                ++spins;
+                if (spins < 4) __builtin_amdgcn_s_sleep(1);
+                else if (spins < 32) __builtin_amdgcn_s_sleep(16);
+                else __builtin_amdgcn_s_sleep(64);
+                if (spins > (1u << 22)) {  // bounded spin: poison, don't hang
                     __hip_atomic_store((gu64_t *)(state + 1), 1ull,
                                        __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                     break;
@@ -1639,7 +1836,7 @@ dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
     if (n <= 1) return DBSP_OK;
     // significant bytes from max values
     uint64_t *d_max;
-    HIP_CHECK(hipMallocAsync(&d_max, 4 * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&d_max, 4 * sizeof(uint64_t), s));
     HIP_CHECK(hipMemsetAsync(d_max, 0, 2 * sizeof(uint64_t), s));
     HIP_CHECK(hipMemsetAsync(d_max + 2, 0xFF, 2 * sizeof(uint64_t), s));
     k_minmax_u64<<<grid_for(n), BLK, 0, s>>>(kk, vv, n, d_max);
@@ -1647,7 +1844,7 @@ dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
     HIP_CHECK(hipMemcpyAsync(h_max, d_max, 4 * sizeof(uint64_t),
                              hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
-    HIP_CHECK(hipFreeAsync(d_max, s));
+    HIP_CHECK(dbspk::cache_free(d_max, s));
     const uint64_t kbase = h_max[2], vbase = h_max[3];
     const uint64_t krange = h_max[0] - kbase, vrange = h_max[1] - vbase;
     int kbytes = 0, vbytes = 0;
@@ -1656,7 +1853,7 @@ dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
 
     int64_t nblocks = ceil_div(n, SORT_TILE);
     uint64_t *counts;
-    HIP_CHECK(hipMallocAsync(&counts, (int64_t)256 * nblocks * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&counts, (int64_t)256 * nblocks * sizeof(uint64_t), s));
 
     uint64_t *src_k = kk, *src_v = vv; int64_t *src_w = ww;
     uint64_t *dst_k = kk2, *dst_v = vv2; int64_t *dst_w = ww2;
@@ -1677,7 +1874,7 @@ dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
         t = src_v; src_v = dst_v; dst_v = t;
         tw = src_w; src_w = dst_w; dst_w = tw;
     }
-    HIP_CHECK(hipFreeAsync(counts, s));
+    HIP_CHECK(dbspk::cache_free(counts, s));
     *result_in_scratch = (src_k != kk);
     return DBSP_OK;
 }
@@ -1692,39 +1889,39 @@ dbsp_status consolidate_sorted(hipStream_t s, const uint64_t *kk,
         return DBSP_OK;
     }
     uint64_t *flags, *fscan;
-    HIP_CHECK(hipMallocAsync(&flags, n * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&fscan, n * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&flags, n * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&fscan, n * sizeof(uint64_t), s));
     k_head_flags<<<grid_for(n), BLK, 0, s>>>(kk, vv, n, flags);
     uint64_t nseg = 0;
     dbsp_status st = scan_exclusive(s, flags, fscan, n, &nseg);
     if (st != DBSP_OK) return st;
     uint64_t *sk, *sv; int64_t *sw;
-    HIP_CHECK(hipMallocAsync(&sk, nseg * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&sv, nseg * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&sw, nseg * sizeof(int64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&sk, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&sv, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&sw, nseg * sizeof(int64_t) + 8, s));
     HIP_CHECK(hipMemsetAsync(sw, 0, nseg * sizeof(int64_t), s));
     k_seg_accum<<<grid_for(n), BLK, 0, s>>>(kk, vv, ww, fscan, flags, n, sk, sv, sw);
     // drop zero-weight segments
     uint64_t *nzflags, *nzscan;
-    HIP_CHECK(hipMallocAsync(&nzflags, nseg * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&nzscan, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&nzflags, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&nzscan, nseg * sizeof(uint64_t) + 8, s));
     k_nonzero_flags<<<grid_for(nseg), BLK, 0, s>>>(sw, nseg, nzflags);
     uint64_t nout = 0;
     st = scan_exclusive(s, nzflags, nzscan, nseg, &nout);
     if (st != DBSP_OK) return st;
     uint64_t *rk, *rv; int64_t *rw;
-    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rw, nout * sizeof(int64_t) + 8, s));
     k_compact<<<grid_for(nseg), BLK, 0, s>>>(sk, sv, sw, nzflags, nzscan, nseg,
                                              rk, rv, rw);
-    HIP_CHECK(hipFreeAsync(flags, s));
-    HIP_CHECK(hipFreeAsync(fscan, s));
-    HIP_CHECK(hipFreeAsync(sk, s));
-    HIP_CHECK(hipFreeAsync(sv, s));
-    HIP_CHECK(hipFreeAsync(sw, s));
-    HIP_CHECK(hipFreeAsync(nzflags, s));
-    HIP_CHECK(hipFreeAsync(nzscan, s));
+    HIP_CHECK(dbspk::cache_free(flags, s));
+    HIP_CHECK(dbspk::cache_free(fscan, s));
+    HIP_CHECK(dbspk::cache_free(sk, s));
+    HIP_CHECK(dbspk::cache_free(sv, s));
+    HIP_CHECK(dbspk::cache_free(sw, s));
+    HIP_CHECK(dbspk::cache_free(nzflags, s));
+    HIP_CHECK(dbspk::cache_free(nzscan, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
     return DBSP_OK;
 }
@@ -1740,41 +1937,92 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
         *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
         return DBSP_OK;
     }
+    // Default: robust two-pass (count -> scan -> emit).  The single-pass
+    // decoupled-lookback variant is ~2x faster at 1B rows when healthy but
+    // can collapse ~30x when the prefix chain serializes under fabric
+    // pressure (profiles/r01_merge_pmc.txt discussion) — opt in with
+    // DBSP_MERGE_ONEPASS=1 until that is fixed.
+    static const bool onepass = []() {
+        const char *e = getenv("DBSP_MERGE_ONEPASS");
+        return e && e[0] == '1';
+    }();
     int64_t nblocks = ceil_div(total, MP_TILE);
     int64_t *pa, *pb;
-    unsigned long long *state;
-    HIP_CHECK(hipMallocAsync(&pa, (nblocks + 1) * sizeof(int64_t), s));
-    HIP_CHECK(hipMallocAsync(&pb, (nblocks + 1) * sizeof(int64_t), s));
-    HIP_CHECK(hipMallocAsync(&state, (nblocks + 2) * sizeof(uint64_t), s));
-    HIP_CHECK(hipMemsetAsync(state, 0, (nblocks + 2) * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&pa, (nblocks + 1) * sizeof(int64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&pb, (nblocks + 1) * sizeof(int64_t), s));
     k_mp_partition<<<grid_for(nblocks + 1), BLK, 0, s>>>(ak, av, na, bk, bv, nb,
                                                          nblocks, pa, pb);
     uint64_t *rk, *rv;
     W *rw;
-    HIP_CHECK(hipMallocAsync(&rk, total * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rv, total * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rw, total * sizeof(W) + 8, s));
-    const size_t smem = 3 * (MP_TILE + 2) * sizeof(uint64_t);
-    k_mp_merge_onepass<W><<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
-        ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
-    // total = inclusive prefix of the last virtual block; error word poisoned
-    // if a lookback spun out (never observed; bounded to avoid hangs)
-    unsigned long long h_state[2];
-    HIP_CHECK(hipMemcpyAsync(&h_state[0], state + 1, sizeof(uint64_t),
-                             hipMemcpyDeviceToHost, s));
-    HIP_CHECK(hipMemcpyAsync(&h_state[1], state + 2 + (nblocks - 1),
-                             sizeof(uint64_t), hipMemcpyDeviceToHost, s));
-    HIP_CHECK(hipStreamSynchronize(s));
-    HIP_CHECK(hipFreeAsync(pa, s));
-    HIP_CHECK(hipFreeAsync(pb, s));
-    HIP_CHECK(hipFreeAsync(state, s));
-    if (h_state[0] != 0 || (h_state[1] & 3ull) != 2ull) {
-        (void)hipFreeAsync(rk, s);
-        (void)hipFreeAsync(rv, s);
-        (void)hipFreeAsync(rw, s);
-        return DBSP_ERR_INTERNAL;
+    if (onepass) {
+        unsigned long long *state;
+        HIP_CHECK(dbspk::cache_malloc((void **)&state, (nblocks + 2) * sizeof(uint64_t), s));
+        HIP_CHECK(hipMemsetAsync(state, 0, (nblocks + 2) * sizeof(uint64_t), s));
+        HIP_CHECK(dbspk::cache_malloc((void **)&rk, total * sizeof(uint64_t) + 8, s));
+        HIP_CHECK(dbspk::cache_malloc((void **)&rv, total * sizeof(uint64_t) + 8, s));
+        HIP_CHECK(dbspk::cache_malloc((void **)&rw, total * sizeof(W) + 8, s));
+        const size_t smem = 3 * (MP_TILE + 2) * sizeof(uint64_t);
+        k_mp_merge_onepass<W><<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
+            ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
+        unsigned long long h_state[2];
+        HIP_CHECK(hipMemcpyAsync(&h_state[0], state + 1, sizeof(uint64_t),
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipMemcpyAsync(&h_state[1], state + 2 + (nblocks - 1),
+                                 sizeof(uint64_t), hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        HIP_CHECK(dbspk::cache_free(pa, s));
+        HIP_CHECK(dbspk::cache_free(pb, s));
+        HIP_CHECK(dbspk::cache_free(state, s));
+        if (h_state[0] != 0 || (h_state[1] & 3ull) != 2ull) {
+            (void)dbspk::cache_free(rk, s);
+            (void)dbspk::cache_free(rv, s);
+            (void)dbspk::cache_free(rw, s);
+            return DBSP_ERR_INTERNAL;
+        }
+        *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)(h_state[1] >> 2);
+        return DBSP_OK;
     }
-    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)(h_state[1] >> 2);
+    uint64_t *counts;
+    HIP_CHECK(dbspk::cache_malloc((void **)&counts, (nblocks + 1) * sizeof(uint64_t), s));
+    const size_t smem_count = 2 * (MP_TILE + 2) * sizeof(uint64_t);
+    k_mp_merge<false, W><<<dim3((uint32_t)nblocks), MP_THREADS, smem_count, s>>>(
+        ak, av, aw, na, bk, bv, bw, nb, pa, pb, counts, nullptr, nullptr,
+        nullptr);
+    uint64_t nout = 0;
+    static const bool host_prof = []() {
+        const char *e = getenv("DBSP_PROFILE");
+        return e && e[0] == '2';
+    }();
+    auto tick_us = []() {
+        struct timespec ts;
+        clock_gettime(CLOCK_MONOTONIC, &ts);
+        return ts.tv_sec * 1000000.0 + ts.tv_nsec / 1000.0;
+    };
+    double t_scan0 = host_prof ? tick_us() : 0;
+    dbsp_status st = scan_exclusive(s, counts, counts, nblocks, &nout);
+    if (st != DBSP_OK) return st;
+    double t_scan1 = host_prof ? tick_us() : 0;
+    HIP_CHECK(dbspk::cache_malloc((void **)&rk, nout * sizeof(uint64_t) + 8, s));
+    double t_a1 = host_prof ? tick_us() : 0;
+    HIP_CHECK(dbspk::cache_malloc((void **)&rv, nout * sizeof(uint64_t) + 8, s));
+    double t_a2 = host_prof ? tick_us() : 0;
+    HIP_CHECK(dbspk::cache_malloc((void **)&rw, nout * sizeof(W) + 8, s));
+    double t_a3 = host_prof ? tick_us() : 0;
+    if (host_prof && na + nb > (int64_t)100000000)
+        fprintf(stderr,
+                "[merge host] scan+sync %.1f ms  mallocA(k) %.1f ms  "
+                "mallocA(v) %.1f ms  mallocA(w) %.1f ms  (nout=%llu)\n",
+                (t_scan1 - t_scan0) / 1000.0, (t_a1 - t_scan1) / 1000.0,
+                (t_a2 - t_a1) / 1000.0, (t_a3 - t_a2) / 1000.0,
+                (unsigned long long)nout);
+    const size_t smem_emit = 3 * (MP_TILE + 2) * sizeof(uint64_t);
+    if (nout > 0)
+        k_mp_merge<true, W><<<dim3((uint32_t)nblocks), MP_THREADS, smem_emit, s>>>(
+            ak, av, aw, na, bk, bv, bw, nb, pa, pb, counts, rk, rv, rw);
+    HIP_CHECK(dbspk::cache_free(pa, s));
+    HIP_CHECK(dbspk::cache_free(pb, s));
+    HIP_CHECK(dbspk::cache_free(counts, s));
+    *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
     return DBSP_OK;
 }
 
@@ -1841,22 +2089,22 @@ dbsp_status join_spine_rows(hipStream_t s, const uint64_t *dk,
     }
     uint32_t *cnts;
     uint64_t *counts;
-    HIP_CHECK(hipMallocAsync(&cnts, (int64_t)nd * t.nb * sizeof(uint32_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&counts, (nd + 1) * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&cnts, (int64_t)nd * t.nb * sizeof(uint32_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&counts, (nd + 1) * sizeof(uint64_t), s));
     k_join_count_multi<<<grid_for(nd), BLK, 0, s>>>(dk, nd, t, cnts, counts);
     uint64_t nout = 0;
     dbsp_status st = scan_exclusive(s, counts, counts, nd, &nout);
     if (st != DBSP_OK) return st;
     uint64_t *rk, *rv; int64_t *rw;
-    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rw, nout * sizeof(int64_t) + 8, s));
     if (nout > 0)
         k_join_emit_multi<<<grid_for((int64_t)nout), BLK, 0, s>>>(
             dk, dv, dw, nd, t, cnts, counts, (int64_t)nout, proj, param, rk, rv,
             rw);
-    HIP_CHECK(hipFreeAsync(cnts, s));
-    HIP_CHECK(hipFreeAsync(counts, s));
+    HIP_CHECK(dbspk::cache_free(cnts, s));
+    HIP_CHECK(dbspk::cache_free(counts, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
     return DBSP_OK;
 }
@@ -1891,22 +2139,22 @@ dbsp_status join_rows(hipStream_t s, const uint64_t *dk, const uint64_t *dv,
         return DBSP_OK;
     }
     uint64_t *counts, *starts;
-    HIP_CHECK(hipMallocAsync(&counts, (nd + 1) * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&starts, nd * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&counts, (nd + 1) * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&starts, nd * sizeof(uint64_t), s));
     k_join_count<<<grid_for(nd), BLK, 0, s>>>(dk, nd, tk, nt, counts, starts);
     uint64_t nout = 0;
     dbsp_status st = scan_exclusive(s, counts, counts, nd, &nout);
     if (st != DBSP_OK) return st;
     uint64_t *rk, *rv; int64_t *rw;
-    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rw, nout * sizeof(int64_t) + 8, s));
     if (nout > 0)
         k_join_emit<<<grid_for((int64_t)nout), BLK, 0, s>>>(
             dk, dv, dw, nd, tv, tw, counts, starts, (int64_t)nout, proj, param,
             rk, rv, rw);
-    HIP_CHECK(hipFreeAsync(counts, s));
-    HIP_CHECK(hipFreeAsync(starts, s));
+    HIP_CHECK(dbspk::cache_free(counts, s));
+    HIP_CHECK(dbspk::cache_free(starts, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
     return DBSP_OK;
 }
@@ -1924,12 +2172,12 @@ static dbsp_status agg_upsert_impl(hipStream_t s, const uint64_t *keys,
         return DBSP_OK;
     }
     uint64_t *counts, *istart, *ostart, *newval, *hasnew, *offsets;
-    HIP_CHECK(hipMallocAsync(&counts, nd * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&istart, nd * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&ostart, nd * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&newval, nd * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&hasnew, nd * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&offsets, nd * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&counts, nd * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&istart, nd * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&ostart, nd * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&newval, nd * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&hasnew, nd * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&offsets, nd * sizeof(uint64_t), s));
     k_agg_count<MAX><<<grid_for(nd), BLK, 0, s>>>(keys, nd, ik, iw, ni, tok, no,
                                                   counts, istart, ostart, newval,
                                                   hasnew);
@@ -1937,20 +2185,20 @@ static dbsp_status agg_upsert_impl(hipStream_t s, const uint64_t *keys,
     dbsp_status st = scan_exclusive(s, counts, offsets, nd, &nout);
     if (st != DBSP_OK) return st;
     uint64_t *rk, *rv; int64_t *rw;
-    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rw, nout * sizeof(int64_t) + 8, s));
     if (nout > 0)
         k_agg_emit<MAX><<<grid_for(nd), BLK, 0, s>>>(keys, nd, iv, tov, tow,
                                                      offsets, counts, istart,
                                                      ostart, newval, hasnew, rk,
                                                      rv, rw);
-    HIP_CHECK(hipFreeAsync(counts, s));
-    HIP_CHECK(hipFreeAsync(istart, s));
-    HIP_CHECK(hipFreeAsync(ostart, s));
-    HIP_CHECK(hipFreeAsync(newval, s));
-    HIP_CHECK(hipFreeAsync(hasnew, s));
-    HIP_CHECK(hipFreeAsync(offsets, s));
+    HIP_CHECK(dbspk::cache_free(counts, s));
+    HIP_CHECK(dbspk::cache_free(istart, s));
+    HIP_CHECK(dbspk::cache_free(ostart, s));
+    HIP_CHECK(dbspk::cache_free(newval, s));
+    HIP_CHECK(dbspk::cache_free(hasnew, s));
+    HIP_CHECK(dbspk::cache_free(offsets, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
     return DBSP_OK;
 }
@@ -1965,23 +2213,23 @@ dbsp_status distinct_inc_rows(hipStream_t s, const uint64_t *dk,
     }
     uint64_t *flags, *fscan;
     int64_t *outw;
-    HIP_CHECK(hipMallocAsync(&flags, nd * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&fscan, nd * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&outw, nd * sizeof(int64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&flags, nd * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&fscan, nd * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&outw, nd * sizeof(int64_t), s));
     k_distinct_count<<<grid_for(nd), BLK, 0, s>>>(dk, dv, dw, nd, t, flags, outw);
     uint64_t nout = 0;
     dbsp_status st = scan_exclusive(s, flags, fscan, nd, &nout);
     if (st != DBSP_OK) return st;
     uint64_t *rk, *rv;
     int64_t *rw;
-    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(int64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rw, nout * sizeof(int64_t) + 8, s));
     k_compact<<<grid_for(nd), BLK, 0, s>>>(dk, dv, outw, flags, fscan, nd, rk,
                                            rv, rw);
-    HIP_CHECK(hipFreeAsync(flags, s));
-    HIP_CHECK(hipFreeAsync(fscan, s));
-    HIP_CHECK(hipFreeAsync(outw, s));
+    HIP_CHECK(dbspk::cache_free(flags, s));
+    HIP_CHECK(dbspk::cache_free(fscan, s));
+    HIP_CHECK(dbspk::cache_free(outw, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
     return DBSP_OK;
 }
@@ -2014,7 +2262,7 @@ dbsp_status window_rows(hipStream_t s, const uint64_t *tk, const uint64_t *tv,
                         uint64_t e1, uint64_t **ok, uint64_t **ov, int64_t **ow,
                         int64_t *out_n) {
     int64_t *d_ranges;
-    HIP_CHECK(hipMallocAsync(&d_ranges, 8 * sizeof(int64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&d_ranges, 8 * sizeof(int64_t), s));
     k_window_ranges<<<1, 1, 0, s>>>(tk, nt, bk, nb, have_prev, s0, e0, s1, e1,
                                     d_ranges);
     int64_t h_ranges[8];
@@ -2024,13 +2272,13 @@ dbsp_status window_rows(hipStream_t s, const uint64_t *tk, const uint64_t *tv,
     int64_t total = (h_ranges[1] - h_ranges[0]) + (h_ranges[3] - h_ranges[2]) +
                     (h_ranges[5] - h_ranges[4]) + (h_ranges[7] - h_ranges[6]);
     uint64_t *rk, *rv; int64_t *rw;
-    HIP_CHECK(hipMallocAsync(&rk, total * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rv, total * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rw, total * sizeof(int64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rk, total * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rv, total * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rw, total * sizeof(int64_t) + 8, s));
     if (total > 0)
         k_window_emit<<<grid_for(total), BLK, 0, s>>>(tk, tv, tw, bk, bv, bw,
                                                       d_ranges, rk, rv, rw);
-    HIP_CHECK(hipFreeAsync(d_ranges, s));
+    HIP_CHECK(dbspk::cache_free(d_ranges, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = total;
     return DBSP_OK;
 }
@@ -2060,7 +2308,7 @@ dbsp_status shard_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
                        const int64_t *w, int64_t n, int nshards, uint64_t *ok,
                        uint64_t *ov, int64_t *ow, int64_t *h_offsets) {
     uint64_t *hist;
-    HIP_CHECK(hipMallocAsync(&hist, (nshards + 1) * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&hist, (nshards + 1) * sizeof(uint64_t), s));
     HIP_CHECK(hipMemsetAsync(hist, 0, (nshards + 1) * sizeof(uint64_t), s));
     if (n > 0) k_shard_hist<<<grid_for(n), BLK, 0, s>>>(k, n, nshards, hist);
     uint64_t h_hist[64];
@@ -2079,7 +2327,7 @@ dbsp_status shard_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
     if (n > 0)
         k_shard_scatter<<<grid_for(n), BLK, 0, s>>>(k, v, w, n, nshards, hist,
                                                     ok, ov, ow);
-    HIP_CHECK(hipFreeAsync(hist, s));
+    HIP_CHECK(dbspk::cache_free(hist, s));
     return DBSP_OK;
 }
 
@@ -2088,7 +2336,7 @@ dbsp_status flatmap_events(hipStream_t s, const dbsp_event *ev, int64_t n,
                            int64_t *n0, uint64_t *k1, uint64_t *v1, int64_t *w1,
                            int64_t *n1) {
     uint64_t *ctr;
-    HIP_CHECK(hipMallocAsync(&ctr, 2 * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&ctr, 2 * sizeof(uint64_t), s));
     HIP_CHECK(hipMemsetAsync(ctr, 0, 2 * sizeof(uint64_t), s));
     if (n > 0)
         k_flatmap<<<grid_for(n), BLK, 0, s>>>(ev, n, query, ctr, k0, v0, w0,
@@ -2096,7 +2344,7 @@ dbsp_status flatmap_events(hipStream_t s, const dbsp_event *ev, int64_t n,
     uint64_t h_ctr[2];
     HIP_CHECK(hipMemcpyAsync(h_ctr, ctr, sizeof(h_ctr), hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
-    HIP_CHECK(hipFreeAsync(ctr, s));
+    HIP_CHECK(dbspk::cache_free(ctr, s));
     *n0 = (int64_t)h_ctr[0];
     if (n1) *n1 = (int64_t)h_ctr[1];
     return DBSP_OK;
@@ -2199,47 +2447,47 @@ dbsp_status consolidate_sorted_f64(hipStream_t s, const uint64_t *kk,
         return DBSP_OK;
     }
     uint64_t *flags, *fscan, *hp;
-    HIP_CHECK(hipMallocAsync(&flags, n * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&fscan, n * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&flags, n * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&fscan, n * sizeof(uint64_t), s));
     k_head_flags<<<grid_for(n), BLK, 0, s>>>(kk, vv, n, flags);
     uint64_t nseg = 0;
     dbsp_status st = scan_exclusive(s, flags, fscan, n, &nseg);
     if (st != DBSP_OK) return st;
-    HIP_CHECK(hipMallocAsync(&hp, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&hp, nseg * sizeof(uint64_t) + 8, s));
     k_seg_headpos<<<grid_for(n), BLK, 0, s>>>(flags, fscan, n, hp);
     for (int64_t d = 1; d < n; d <<= 1)
         k_seg_tree_round<<<grid_for(n), BLK, 0, s>>>(ww, flags, fscan, hp, n, d);
     uint64_t *sk, *sv;
     double *sw;
-    HIP_CHECK(hipMallocAsync(&sk, nseg * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&sv, nseg * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&sw, nseg * sizeof(double) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&sk, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&sv, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&sw, nseg * sizeof(double) + 8, s));
     k_seg_collect_f64<<<grid_for(n), BLK, 0, s>>>(kk, vv, ww, flags, fscan, n,
                                                   sk, sv, sw);
     uint64_t *nzflags, *nzscan;
-    HIP_CHECK(hipMallocAsync(&nzflags, nseg * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&nzscan, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&nzflags, nseg * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&nzscan, nseg * sizeof(uint64_t) + 8, s));
     k_nonzero_flags_f64<<<grid_for(nseg), BLK, 0, s>>>(sw, nseg, nzflags);
     uint64_t nout = 0;
     st = scan_exclusive(s, nzflags, nzscan, nseg, &nout);
     if (st != DBSP_OK) return st;
     uint64_t *rk, *rv;
     double *rw;
-    HIP_CHECK(hipMallocAsync(&rk, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rv, nout * sizeof(uint64_t) + 8, s));
-    HIP_CHECK(hipMallocAsync(&rw, nout * sizeof(double) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rk, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rv, nout * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&rw, nout * sizeof(double) + 8, s));
     // k_compact copies the weight column bitwise (8 B) — valid for f64
     k_compact<<<grid_for(nseg), BLK, 0, s>>>(sk, sv, (const int64_t *)sw,
                                              nzflags, nzscan, nseg, rk, rv,
                                              (int64_t *)rw);
-    HIP_CHECK(hipFreeAsync(flags, s));
-    HIP_CHECK(hipFreeAsync(fscan, s));
-    HIP_CHECK(hipFreeAsync(hp, s));
-    HIP_CHECK(hipFreeAsync(sk, s));
-    HIP_CHECK(hipFreeAsync(sv, s));
-    HIP_CHECK(hipFreeAsync(sw, s));
-    HIP_CHECK(hipFreeAsync(nzflags, s));
-    HIP_CHECK(hipFreeAsync(nzscan, s));
+    HIP_CHECK(dbspk::cache_free(flags, s));
+    HIP_CHECK(dbspk::cache_free(fscan, s));
+    HIP_CHECK(dbspk::cache_free(hp, s));
+    HIP_CHECK(dbspk::cache_free(sk, s));
+    HIP_CHECK(dbspk::cache_free(sv, s));
+    HIP_CHECK(dbspk::cache_free(sw, s));
+    HIP_CHECK(dbspk::cache_free(nzflags, s));
+    HIP_CHECK(dbspk::cache_free(nzscan, s));
     *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)nout;
     return DBSP_OK;
 }
@@ -2256,7 +2504,7 @@ dbsp_status emit_nonzero_f64(hipStream_t s, const uint64_t *keys,
                              const double *acc, int64_t nd, uint64_t *ok,
                              uint64_t *ov, int64_t *ow, int64_t *h_count) {
     uint64_t *ctr;
-    HIP_CHECK(hipMallocAsync(&ctr, sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&ctr, sizeof(uint64_t), s));
     HIP_CHECK(hipMemsetAsync(ctr, 0, sizeof(uint64_t), s));
     if (nd > 0)
         k_emit_nonzero_f64<<<grid_for(nd), BLK, 0, s>>>(keys, acc, nd, ctr, ok,
@@ -2264,7 +2512,7 @@ dbsp_status emit_nonzero_f64(hipStream_t s, const uint64_t *keys,
     uint64_t h = 0;
     HIP_CHECK(hipMemcpyAsync(&h, ctr, sizeof(uint64_t), hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
-    HIP_CHECK(hipFreeAsync(ctr, s));
+    HIP_CHECK(dbspk::cache_free(ctr, s));
     *h_count = (int64_t)h;
     return DBSP_OK;
 }
@@ -2281,14 +2529,14 @@ dbsp_status emit_nonzero(hipStream_t s, const uint64_t *keys,
                          const int64_t *acc, int64_t nd, uint64_t *ok,
                          uint64_t *ov, int64_t *ow, int64_t *h_count) {
     uint64_t *ctr;
-    HIP_CHECK(hipMallocAsync(&ctr, sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&ctr, sizeof(uint64_t), s));
     HIP_CHECK(hipMemsetAsync(ctr, 0, sizeof(uint64_t), s));
     if (nd > 0)
         k_emit_nonzero<<<grid_for(nd), BLK, 0, s>>>(keys, acc, nd, ctr, ok, ov, ow);
     uint64_t h = 0;
     HIP_CHECK(hipMemcpyAsync(&h, ctr, sizeof(uint64_t), hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
-    HIP_CHECK(hipFreeAsync(ctr, s));
+    HIP_CHECK(dbspk::cache_free(ctr, s));
     *h_count = (int64_t)h;
     return DBSP_OK;
 }
@@ -2301,17 +2549,17 @@ dbsp_status unique_keys(hipStream_t s, const uint64_t *kk, int64_t n,
         return DBSP_OK;
     }
     uint64_t *flags, *fscan;
-    HIP_CHECK(hipMallocAsync(&flags, n * sizeof(uint64_t), s));
-    HIP_CHECK(hipMallocAsync(&fscan, n * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&flags, n * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&fscan, n * sizeof(uint64_t), s));
     k_key_head_flags<<<grid_for(n), BLK, 0, s>>>(kk, n, flags);
     uint64_t nk = 0;
     dbsp_status st = scan_exclusive(s, flags, fscan, n, &nk);
     if (st != DBSP_OK) return st;
     uint64_t *out;
-    HIP_CHECK(hipMallocAsync(&out, nk * sizeof(uint64_t) + 8, s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&out, nk * sizeof(uint64_t) + 8, s));
     k_compact_keys<<<grid_for(n), BLK, 0, s>>>(kk, flags, fscan, n, out);
-    HIP_CHECK(hipFreeAsync(flags, s));
-    HIP_CHECK(hipFreeAsync(fscan, s));
+    HIP_CHECK(dbspk::cache_free(flags, s));
+    HIP_CHECK(dbspk::cache_free(fscan, s));
     *okeys = out;
     *out_n = (int64_t)nk;
     return DBSP_OK;

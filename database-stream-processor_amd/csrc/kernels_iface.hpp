@@ -59,6 +59,15 @@ struct JoinCountArgs {
     int64_t *d_total;
 };
 
+
+namespace dbspk {
+// big-buffer free-list over hipMallocAsync (see kernels.hip): same signature
+// shape as hipMallocAsync/hipFreeAsync so call sites stay HIP_CHECK-able
+hipError_t cache_malloc(void **out, size_t bytes, hipStream_t s);
+hipError_t cache_free(void *p, hipStream_t s);
+void cache_trim(hipStream_t s);
+}  // namespace dbspk
+
 namespace dbspk {
 
 dbsp_status scan_excl(hipStream_t s, const uint64_t *in, uint64_t *out,

@@ -11,6 +11,7 @@ Prints one JSON line per primitive.
 import argparse
 import ctypes
 import json
+import os
 import sys
 import time
 from pathlib import Path
@@ -49,6 +50,7 @@ def as_batch(k, v, w):
 
 
 def bench_merge(ctx, L, n_per_side, reps=3):
+    reps = int(os.environ.get("KB_REPS", reps))
     ka, va, wa = sorted_unique_batch(n_per_side, 1)
     kb, vb, wb = sorted_unique_batch(n_per_side, 2)
     a, b = as_batch(ka, va, wa), as_batch(kb, vb, wb)
@@ -321,14 +323,22 @@ def main():
                     help="rows per merge side (trace = 2x this)")
     ap.add_argument("--delta", type=int, default=10_000_000)
     ap.add_argument("--sort-rows", type=int, default=50_000_000)
+    ap.add_argument("--only", default=None,
+                    choices=["merge", "join", "sort", "incremental", "c5"])
     args = ap.parse_args()
     ctx = Ctx(0)
     L = _L()
-    print(json.dumps(bench_merge(ctx, L, args.rows)))
-    print(json.dumps(bench_join(ctx, L, 2 * args.rows, args.delta)))
-    print(json.dumps(bench_sort(ctx, L, args.sort_rows)))
-    print(json.dumps(bench_incremental(ctx, L, 2 * args.rows, args.delta)))
-    print(json.dumps(bench_c5(ctx, L, 2 * args.rows, args.delta)))
+    legs = {
+        "merge": lambda: bench_merge(ctx, L, args.rows),
+        "join": lambda: bench_join(ctx, L, 2 * args.rows, args.delta),
+        "sort": lambda: bench_sort(ctx, L, args.sort_rows),
+        "incremental": lambda: bench_incremental(ctx, L, 2 * args.rows,
+                                                 args.delta),
+        "c5": lambda: bench_c5(ctx, L, 2 * args.rows, args.delta),
+    }
+    for name, leg in legs.items():
+        if args.only is None or args.only == name:
+            print(json.dumps(leg()))
     ctx.close()
 
 
