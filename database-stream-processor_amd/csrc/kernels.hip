@@ -61,6 +61,12 @@ void cache_drop_locked(hipStream_t s) {
 
 hipError_t cache_malloc(void **out, size_t bytes, hipStream_t s) {
     if (bytes >= CACHE_MIN) {
+        // round carves up to a quarter-power-of-two class so a workload with
+        // slowly growing buffers (spine merges) re-hits earlier blocks
+        // instead of carving a fresh one per size (waste < 25%)
+        size_t msb = (size_t)1 << (63 - __builtin_clzll(bytes));
+        size_t gran = msb >> 2;
+        bytes = (bytes + gran - 1) / gran * gran;
         std::lock_guard<std::mutex> g(g_cache_mu);
         auto it = g_cache_free.lower_bound(bytes);
         // accept up to 50% + 64 MB waste; beyond that carve fresh
