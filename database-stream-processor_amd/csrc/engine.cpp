@@ -345,6 +345,29 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
         return DBSP_OK;
     }
     ScopedTimer t(c, 0, (double)raw.n * 48.0);
+    if (raw.n > 8192) {
+        // dense-range probe: a tick's delta usually covers a tiny key box
+        // (q5 bids: ~4-8 ms timestamps x the ~100-auction in-flight window),
+        // where a weight histogram over (krange+1)*(vrange+1) cells replaces
+        // the whole sort.  Worth one extra sync only above the fused size.
+        uint64_t mm[4];
+        TRY(dbspk::minmax_rows(c->stream, raw.k, raw.v, raw.n, mm));
+        const uint64_t kr = mm[2] - mm[0], vr = mm[3] - mm[1];
+        const int64_t dense_cap =
+            std::min<int64_t>((int64_t)1 << 22, 8 * raw.n);
+        if (kr < (uint64_t)dense_cap && vr < (uint64_t)dense_cap &&
+            (int64_t)((kr + 1) * (vr + 1)) <= dense_cap) {
+            DevBatch res;
+            TRY(alloc_batch(c, raw.n, res));
+            TRY(dbspk::sort_cons_dense(c->stream, raw.k, raw.v, raw.w, raw.n,
+                                       mm[0], mm[1], (int64_t)(kr + 1),
+                                       (int64_t)(vr + 1), res.k, res.v, res.w,
+                                       &res.n));
+            free_batch(c, raw);
+            out = res;
+            return DBSP_OK;
+        }
+    }
     if (raw.n > 8192 && raw.n <= 65536) return sort_medium(c, raw, out);
     DevBatch scratch;
     TRY(alloc_batch(c, raw.n, scratch, true));

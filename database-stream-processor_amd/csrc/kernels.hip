@@ -713,6 +713,77 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     if (tid == 0) *out_len = (int64_t)nout;
 }
 
+
+
+
+// ---------------------------------------------------------------------------
+// dense-range sort+consolidate.  A Nexmark tick's delta usually has a TINY
+// key box: q5 bids cover ~4-8 distinct ms timestamps x the ~100-auction
+// in-flight window (generator/auctions.rs:112-123), so instead of sorting
+// 37k rows we scatter weights into a (krange+1)*(vrange+1) histogram with
+// many blocks and emit the nonzero cells in index order — already sorted and
+// consolidated.  Integer weights only (atomic accumulation).
+// ---------------------------------------------------------------------------
+
+__global__ void k_minmax_rows(const uint64_t *k, const uint64_t *v, int64_t n,
+                              unsigned long long *mm /* kmin,vmin,kmax,vmax */) {
+    uint64_t mk = 0, mv = 0, nk = ~0ull, nv = ~0ull;
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        mk = max(mk, k[i]);
+        mv = max(mv, v[i]);
+        nk = min(nk, k[i]);
+        nv = min(nv, v[i]);
+    }
+    __shared__ unsigned long long s[4];
+    if (threadIdx.x == 0) { s[0] = ~0ull; s[1] = ~0ull; s[2] = 0; s[3] = 0; }
+    __syncthreads();
+    atomicMin(&s[0], (unsigned long long)nk);
+    atomicMin(&s[1], (unsigned long long)nv);
+    atomicMax(&s[2], (unsigned long long)mk);
+    atomicMax(&s[3], (unsigned long long)mv);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        atomicMin(&mm[0], s[0]);
+        atomicMin(&mm[1], s[1]);
+        atomicMax(&mm[2], s[2]);
+        atomicMax(&mm[3], s[3]);
+    }
+}
+
+__global__ void k_hist_add(const uint64_t *k, const uint64_t *v,
+                           const int64_t *w, int64_t n, uint64_t kbase,
+                           uint64_t vbase, uint64_t vspan,
+                           unsigned long long *wbuf) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x)
+        atomicAdd(&wbuf[(k[i] - kbase) * vspan + (v[i] - vbase)],
+                  (unsigned long long)w[i]);
+}
+
+__global__ void k_hist_flags(const unsigned long long *wbuf, int64_t r,
+                             uint64_t *flags) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < r;
+         i += (int64_t)gridDim.x * blockDim.x)
+        flags[i] = wbuf[i] != 0;
+}
+
+__global__ void k_hist_emit(const unsigned long long *wbuf,
+                            const uint64_t *flag_scan, int64_t r,
+                            uint64_t kbase, uint64_t vbase, uint64_t vspan,
+                            uint64_t *ok, uint64_t *ov, int64_t *ow) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < r;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        unsigned long long wv = wbuf[i];
+        if (wv != 0) {
+            uint64_t p = flag_scan[i];
+            ok[p] = kbase + (uint64_t)i / vspan;
+            ov[p] = vbase + (uint64_t)i % vspan;
+            ow[p] = (int64_t)wv;
+        }
+    }
+}
+
 // ---------------------------------------------------------------------------
 // consolidate sorted rows: head flags -> scan -> segment-sum -> nonzero compact
 // ---------------------------------------------------------------------------
@@ -2048,6 +2119,47 @@ dbsp_status merge_rows_f64(hipStream_t s, const uint64_t *ak, const uint64_t *av
                            int64_t *out_n) {
     return merge_rows_t<double>(s, ak, av, aw, na, bk, bv, bw, nb, ok, ov, ow,
                                 out_n);
+}
+
+
+dbsp_status minmax_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
+                        int64_t n, uint64_t mm[4]) {
+    unsigned long long *d;
+    HIP_CHECK(dbspk::cache_malloc((void **)&d, 4 * sizeof(uint64_t), s));
+    const unsigned long long init[4] = {~0ull, ~0ull, 0ull, 0ull};
+    HIP_CHECK(hipMemcpyAsync(d, init, sizeof(init), hipMemcpyHostToDevice, s));
+    k_minmax_rows<<<grid_for(n), BLK, 0, s>>>(k, v, n, d);
+    HIP_CHECK(hipMemcpyAsync(mm, d, 4 * sizeof(uint64_t),
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    HIP_CHECK(dbspk::cache_free(d, s));
+    return DBSP_OK;
+}
+
+dbsp_status sort_cons_dense(hipStream_t s, const uint64_t *k, const uint64_t *v,
+                            const int64_t *w, int64_t n, uint64_t kbase,
+                            uint64_t vbase, int64_t kspan, int64_t vspan,
+                            uint64_t *ok, uint64_t *ov, int64_t *ow,
+                            int64_t *out_n) {
+    const int64_t r = kspan * vspan;
+    unsigned long long *wbuf;
+    uint64_t *flags;
+    HIP_CHECK(dbspk::cache_malloc((void **)&wbuf, r * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&flags, (r + 1) * sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(wbuf, 0, r * sizeof(uint64_t), s));
+    k_hist_add<<<grid_for(n), BLK, 0, s>>>(k, v, w, n, kbase, vbase,
+                                           (uint64_t)vspan, wbuf);
+    k_hist_flags<<<grid_for(r), BLK, 0, s>>>(wbuf, r, flags);
+    uint64_t nout = 0;
+    dbsp_status st = scan_exclusive(s, flags, flags, r, &nout);
+    if (st != DBSP_OK) return st;
+    if (nout > 0)
+        k_hist_emit<<<grid_for(r), BLK, 0, s>>>(wbuf, flags, r, kbase, vbase,
+                                                (uint64_t)vspan, ok, ov, ow);
+    HIP_CHECK(dbspk::cache_free(wbuf, s));
+    HIP_CHECK(dbspk::cache_free(flags, s));
+    *out_n = (int64_t)nout;
+    return DBSP_OK;
 }
 
 dbsp_status sort_cons_small_batch(hipStream_t s, const SortArgs &args) {

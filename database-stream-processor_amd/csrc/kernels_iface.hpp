@@ -93,6 +93,16 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
 // independent small (n <= 8192) raw batches sorted+consolidated concurrently,
 // one workgroup each, lengths left in d_len[i] (device)
 dbsp_status sort_cons_small_batch(hipStream_t s, const SortArgs &args);
+// min/max of k and v (one sync): mm = {kmin, vmin, kmax, vmax}
+dbsp_status minmax_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
+                        int64_t n, uint64_t mm[4]);
+// dense-range consolidate: weights scattered into a kspan*vspan histogram,
+// nonzero cells emitted in index order (sorted); i64 only
+dbsp_status sort_cons_dense(hipStream_t s, const uint64_t *k, const uint64_t *v,
+                            const int64_t *w, int64_t n, uint64_t kbase,
+                            uint64_t vbase, int64_t kspan, int64_t vspan,
+                            uint64_t *ok, uint64_t *ov, int64_t *ow,
+                            int64_t *out_n);
 
 // single-workgroup merge of two consolidated batches (na+nb <= 8192):
 // one launch, no host sync; length left in *d_len (device)

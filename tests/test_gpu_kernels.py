@@ -45,7 +45,17 @@ def test_sort_consolidate_parity(ctx):
         _random_rows(rng, 100_000, key_range=2**40, val_range=2**30),
         _random_rows(rng, 2048),          # exactly one sort tile
         _random_rows(rng, 2049),
+        _random_rows(rng, 65_536),        # dense-range path (small key box)
+        _random_rows(rng, 65_537),
+        _random_rows(rng, 20_000, key_range=5, val_range=200),  # q5 tick shape
+        _random_rows(rng, 40_000, key_range=2**40, val_range=2**30),  # chunked
     ]
+    # medium-path all-equal keys, zero-sum weights (single segment, nseg=1)
+    z40 = np.zeros(40_000, dtype=ROW_DT)
+    z40["k"] = 3
+    z40["v"] = 4
+    z40["w"] = np.where(np.arange(40_000) % 2 == 0, 1, -1)
+    cases.append(z40)
     # all-equal keys, zero-sum weights
     z = np.zeros(1000, dtype=ROW_DT)
     z["k"] = 7
