@@ -74,7 +74,7 @@ struct dbsp_ctx {
     bool force_shard = false;  // exercise the full partition+alltoallv path
                                // even at world=1 (self-exchange; test hook)
     // persistent length scratch (device + pinned host)
-    int64_t *d_len = nullptr;
+    int64_t *d_len = nullptr;   // 16 device length slots (8-15: chained ticks)
     int64_t *h_len = nullptr;
     int timer_depth = 0;  // ScopedTimer nesting guard (shared event pair)
     // per-tick transient bump arena (reset at each engine tick; falls back to
@@ -130,13 +130,13 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     c->profile = p && p[0] == '1';
     const char *fs = getenv("DBSP_FORCE_SHARD");
     c->force_shard = fs && fs[0] == '1';
-    HIP_CHECK_ST(hipMalloc(&c->d_len, 8 * sizeof(int64_t)));
+    HIP_CHECK_ST(hipMalloc(&c->d_len, 16 * sizeof(int64_t)));
     c->arena_sz = (size_t)512 << 20;
     if (hipMalloc(&c->arena, c->arena_sz) != hipSuccess) {
         c->arena = nullptr;
         c->arena_sz = 0;
     }
-    HIP_CHECK_ST(hipHostMalloc(&c->h_len, 8 * sizeof(int64_t)));
+    HIP_CHECK_ST(hipHostMalloc(&c->h_len, 16 * sizeof(int64_t)));
     *out = c;
     return DBSP_OK;
 }
@@ -1181,6 +1181,46 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
     return DBSP_OK;
 }
 
+// Chained variant (single-rank): flatmap counters stay on device
+// (d_len[8..9]), the fused sort launches behind them reading its lengths
+// from the device (outputs at d_len[10..11], -1 on overflow), and no sync
+// happens here — the caller reads everything at its one tick sync and
+// re-sorts through the sized paths if the speculation lost.
+static dbsp_status build_deltas_chain(dbsp_engine *e, const dbsp_event *d_ev,
+                                      int64_t n, DevBatch &rawA, DevBatch &rawB,
+                                      DevBatch &oA, DevBatch &oB) {
+    dbsp_ctx *c = e->ctx;
+    int64_t cap = n > 0 ? n : 1;
+    TRY(alloc_batch(c, cap, rawA, true));
+    TRY(alloc_batch(c, cap, rawB, true));
+    TRY(dbspk::flatmap_events_chain(c->stream, d_ev, n, e->query, rawA.k,
+                                    rawA.v, rawA.w, rawB.k, rawB.v, rawB.w,
+                                    (uint64_t *)(c->d_len + 8)));
+    DevBatch sA, sB;
+    TRY(alloc_batch(c, cap, sA, true));
+    TRY(alloc_batch(c, cap, sB, true));
+    TRY(alloc_batch(c, cap, oA));
+    TRY(alloc_batch(c, cap, oB));
+    SortArgs sa{};
+    sa.nb = 2;
+    sa.kin[0] = rawA.k; sa.vin[0] = rawA.v; sa.win[0] = rawA.w;
+    sa.kin[1] = rawB.k; sa.vin[1] = rawB.v; sa.win[1] = rawB.w;
+    sa.n_dev[0] = c->d_len + 8;
+    sa.n_dev[1] = c->d_len + 9;
+    sa.tk[0] = sA.k; sa.tv[0] = sA.v; sa.tw[0] = sA.w;
+    sa.tk[1] = sB.k; sa.tv[1] = sB.v; sa.tw[1] = sB.w;
+    sa.ok[0] = oA.k; sa.ov[0] = oA.v; sa.ow[0] = oA.w;
+    sa.ok[1] = oB.k; sa.ov[1] = oB.v; sa.ow[1] = oB.w;
+    sa.d_len = c->d_len + 10;
+    {
+        ScopedTimer t(c, 0, (double)n * 48.0);
+        TRY(dbspk::sort_cons_small_batch(c->stream, sa));
+    }
+    oA.n = -1;  // pending: d_len[10], d_len[11]
+    oB.n = -1;
+    return DBSP_OK;
+}
+
 // join delta against a whole spine in one count/emit pair (join is linear in
 // the trace; spine read path = the reference's CursorList)
 static dbsp_status join_vs_spine(dbsp_ctx *c, const DevBatch &delta,
@@ -1500,8 +1540,19 @@ static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
 // ---- q3 tick (queries/q3.rs:35-63) ----
 static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     dbsp_ctx *c = e->ctx;
-    DevBatch dA, dP;
-    TRY(build_deltas(e, d_ev, n, dA, dP, true));
+    // Single-rank ticks chain flatmap -> fused sorts -> join count/scan with
+    // DEVICE lengths and pay ONE sync for the whole front half; if a delta
+    // overflows the fused sort (never at the benchmark tick sizes) the sync
+    // detects the -1 sentinels and the tick re-sorts through the sized
+    // paths.  Sharded ranks keep the explicit path (the exchange needs host
+    // lengths).
+    const bool chain = !sharding_on(c);
+    DevBatch dA, dP, rawA, rawP;
+    if (chain) {
+        TRY(build_deltas_chain(e, d_ev, n, rawA, rawP, dA, dP));
+    } else {
+        TRY(build_deltas(e, d_ev, n, dA, dP, true));
+    }
     // Bilinear expansion against PREVIOUS traces only, so all three joins are
     // independent of the spine inserts:
     //   out = dA join P_prev + A_prev join dP + dA join dP
@@ -1509,9 +1560,7 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     // (operator/join.rs:217-292) by bilinearity.
     std::vector<DevBatch> outs;
     {
-        // plan the three joins; small deltas use the one-launch count+scan
-        // kernel so a SINGLE sync covers all three output sizes
-        ScopedTimer timer(c, 2, (double)(2 * dA.n + dP.n) * 24.0);
+        ScopedTimer timer(c, 2, (double)n * 24.0);
         struct Plan {
             const DevBatch *delta;
             TraceArgs t;
@@ -1519,6 +1568,7 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             uint32_t *cnts;
             uint64_t *offsets;
             bool small;
+            bool dd;  // delta-vs-delta plan (trace length is tick-fresh)
             int slot;
         } plans[3];
         int np = 0;
@@ -1535,45 +1585,93 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             TRY(e->p_int.consolidate_all(c));
         if ((int)e->a_int.batches.size() > MAX_TRACE_BATCHES)
             TRY(e->a_int.consolidate_all(c));
-        if (dA.n > 0 && !e->p_int.batches.empty())
-            plans[np++] = {&dA, spine_args(e->p_int), DBSP_PROJ_HI_V2_LO_V1,
-                           nullptr, nullptr, false};
-        if (dP.n > 0 && !e->a_int.batches.empty())
-            plans[np++] = {&dP, spine_args(e->a_int), DBSP_PROJ_HI_V1_LO_V2,
-                           nullptr, nullptr, false};
-        if (dA.n > 0 && dP.n > 0) {
-            TraceArgs t{};
-            t.nb = 1;
-            t.k[0] = dP.k; t.v[0] = dP.v; t.w[0] = dP.w; t.n[0] = dP.n;
-            plans[np++] = {&dA, t, DBSP_PROJ_HI_V2_LO_V1, nullptr, nullptr,
-                           false};
-        }
-        JoinCountArgs jca{};
-        for (int i = 0; i < np; i++) {
-            Plan &pl = plans[i];
-            pl.slot = -1;
-            if (pl.t.nb == 0) continue;
-            int64_t nd = pl.delta->n;
-            pl.cnts = (uint32_t *)arena_alloc(c, (size_t)nd * pl.t.nb * 4 + 8);
-            pl.offsets = (uint64_t *)arena_alloc(c, (size_t)(nd + 1) * 8);
-            if (nd <= 8192 && pl.cnts && pl.offsets) {
-                pl.small = true;
-                pl.slot = jca.np;
-                jca.dk[jca.np] = pl.delta->k;
-                jca.nd[jca.np] = nd;
-                jca.t[jca.np] = pl.t;
-                jca.cnts[jca.np] = pl.cnts;
-                jca.offsets[jca.np] = pl.offsets;
-                jca.np++;
+        bool spec = chain;  // delta lengths still on the device
+        for (;;) {
+            np = 0;
+            const bool haveA = spec || dA.n > 0, haveP = spec || dP.n > 0;
+            if (haveA && !e->p_int.batches.empty())
+                plans[np++] = {&dA, spine_args(e->p_int), DBSP_PROJ_HI_V2_LO_V1,
+                               nullptr, nullptr, false, false, -1};
+            if (haveP && !e->a_int.batches.empty())
+                plans[np++] = {&dP, spine_args(e->a_int), DBSP_PROJ_HI_V1_LO_V2,
+                               nullptr, nullptr, false, false, -1};
+            if (haveA && haveP) {
+                TraceArgs t{};
+                t.nb = 1;
+                t.k[0] = dP.k; t.v[0] = dP.v; t.w[0] = dP.w;
+                t.n[0] = spec ? 0 : dP.n;
+                plans[np++] = {&dA, t, DBSP_PROJ_HI_V2_LO_V1, nullptr, nullptr,
+                               false, true, -1};
             }
+            JoinCountArgs jca{};
+            bool arena_ok = true;
+            for (int i = 0; i < np; i++) {
+                Plan &pl = plans[i];
+                pl.slot = -1;
+                if (pl.t.nb == 0) continue;
+                const int64_t nd_cap = spec ? n : pl.delta->n;
+                pl.cnts =
+                    (uint32_t *)arena_alloc(c, (size_t)nd_cap * pl.t.nb * 4 + 8);
+                pl.offsets = (uint64_t *)arena_alloc(c, (size_t)(nd_cap + 1) * 8);
+                const bool bufs = pl.cnts && pl.offsets;
+                if (spec && !bufs) {
+                    arena_ok = false;
+                    break;
+                }
+                if (bufs && (spec || nd_cap <= 8192)) {
+                    pl.small = true;
+                    pl.slot = jca.np;
+                    jca.dk[jca.np] = pl.delta->k;
+                    jca.nd[jca.np] = spec ? 0 : pl.delta->n;
+                    if (spec)
+                        jca.nd_dev[jca.np] =
+                            c->d_len + (pl.delta == &dA ? 10 : 11);
+                    if (spec && pl.dd) jca.tn_dev[jca.np] = c->d_len + 11;
+                    jca.t[jca.np] = pl.t;
+                    jca.cnts[jca.np] = pl.cnts;
+                    jca.offsets[jca.np] = pl.offsets;
+                    jca.np++;
+                }
+            }
+            if (spec && !arena_ok) {
+                // arena exhausted mid-speculation: fetch the real lengths and
+                // rebuild the plans the explicit way
+                HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
+                                            16 * sizeof(int64_t),
+                                            hipMemcpyDeviceToHost, c->stream));
+                HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            } else {
+                jca.d_total = c->d_len;
+                if (jca.np > 0) TRY(dbspk::join_count_scan_batch(c->stream, jca));
+                HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
+                                            16 * sizeof(int64_t),
+                                            hipMemcpyDeviceToHost, c->stream));
+                HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            }
+            if (!spec) break;
+            bool lost = !arena_ok || c->h_len[10] < 0 || c->h_len[11] < 0;
+            for (int i = 0; i < np && !lost; i++)
+                if (plans[i].slot >= 0 && c->h_len[plans[i].slot] < 0)
+                    lost = true;
+            if (!lost) {
+                dA.n = c->h_len[10];
+                dP.n = c->h_len[11];
+                break;
+            }
+            // speculation lost: re-sort from the raw flatmap outputs with the
+            // now-known lengths, then redo the planning explicitly
+            rawA.n = c->h_len[8];
+            rawP.n = c->h_len[9];
+            free_batch(c, dA);
+            free_batch(c, dP);
+            TRY(sort_consolidate_batch(c, rawA, dA));
+            TRY(sort_consolidate_batch(c, rawP, dP));
+            rawA = DevBatch{};
+            rawP = DevBatch{};
+            spec = false;
         }
-        jca.d_total = c->d_len;
-        if (jca.np > 0) TRY(dbspk::join_count_scan_batch(c->stream, jca));
-        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 8 * sizeof(int64_t),
-                                    hipMemcpyDeviceToHost, c->stream));
-        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
-        // all small plans emit into ONE combined raw buffer at their base
-        // offsets — the downstream consolidate then needs no concat copies
+        for (int i = 0; i < np; i++)
+            if (plans[i].dd) plans[i].t.n[0] = dP.n;
         int64_t total_small = 0;
         for (int i = 0; i < np; i++)
             if (plans[i].t.nb > 0 && plans[i].small)
@@ -1581,6 +1679,8 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         DevBatch comb;
         if (total_small > 0) TRY(alloc_batch(c, total_small, comb, true));
         int64_t base = 0;
+        // all small plans emit into ONE combined raw buffer at their base
+        // offsets — the downstream consolidate then needs no concat copies
         for (int i = 0; i < np; i++) {
             Plan &pl = plans[i];
             if (pl.t.nb == 0) continue;

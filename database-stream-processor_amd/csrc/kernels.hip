@@ -481,7 +481,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     const uint64_t *kin = args.kin[batch];
     const uint64_t *vin = args.vin[batch];
     const W *win = (const W *)args.win[batch];
-    const int64_t n = args.n[batch];
+    int64_t n = args.n[batch];
     uint64_t *tk = args.tk[batch];   // scratch (>= n rows)
     uint64_t *tv = args.tv[batch];
     W *tw = (W *)args.tw[batch];
@@ -489,6 +489,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     uint64_t *ov = args.ov[batch];
     W *ow = (W *)args.ow[batch];
     int64_t *out_len = args.d_len + batch;
+    const int64_t n_chain = args.n_dev[batch] ? *args.n_dev[batch] : -1;
     __shared__ uint32_t bufA[FUSE_MAX];
     __shared__ uint32_t bufB[FUSE_MAX];
     __shared__ uint32_t cnt[FUSE_DIGITS * FUSE_THREADS];  // rank counters
@@ -497,6 +498,13 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     __shared__ uint64_t smin[2];
     const int tid = threadIdx.x;
 
+    if (args.n_dev[batch]) {
+        if (n_chain > FUSE_MAX) {  // speculation lost: host re-sorts
+            if (tid == 0) *out_len = -1;
+            return;
+        }
+        n = n_chain;
+    }
     if (n == 0) {
         if (tid == 0) *out_len = 0;
         return;
@@ -1478,8 +1486,14 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_join_count_scan_small(
     __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
     const int plan = blockIdx.x;
     const uint64_t *dk = args.dk[plan];
-    const int64_t nd = args.nd[plan];
+    const int64_t nd =
+        args.nd_dev[plan] ? *args.nd_dev[plan] : args.nd[plan];
     const TraceArgs &t = args.t[plan];
+    const int64_t tn0 = args.tn_dev[plan] ? *args.tn_dev[plan] : t.n[0];
+    if (nd < 0 || nd > FUSE_MAX || tn0 < 0) {
+        if (threadIdx.x == 0) args.d_total[plan] = -1;
+        return;
+    }
     uint32_t *cnts = args.cnts[plan];
     uint64_t *offsets = args.offsets[plan];
     int64_t *d_total = args.d_total + plan;
@@ -1492,8 +1506,9 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_join_count_scan_small(
         if (i < nd) {
             uint64_t key = dk[i];
             for (int b = 0; b < t.nb; b++) {
-                int64_t lo = lower_bound_k(t.k[b], t.n[b], key);
-                int64_t c = gallop_run(t.k[b], t.n[b], key, lo);
+                const int64_t tb_n = b == 0 ? tn0 : t.n[b];
+                int64_t lo = lower_bound_k(t.k[b], tb_n, key);
+                int64_t c = gallop_run(t.k[b], tb_n, key, lo);
                 cnts[(int64_t)i * t.nb + b] = (uint32_t)c;
                 rt += (uint32_t)c;
             }
@@ -2446,6 +2461,18 @@ dbsp_status shard_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
         k_shard_scatter<<<grid_for(n), BLK, 0, s>>>(k, v, w, n, nshards, hist,
                                                     ok, ov, ow);
     HIP_CHECK(dbspk::cache_free(hist, s));
+    return DBSP_OK;
+}
+
+dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
+                                 int64_t n, int query, uint64_t *k0,
+                                 uint64_t *v0, int64_t *w0, uint64_t *k1,
+                                 uint64_t *v1, int64_t *w1,
+                                 uint64_t *ctr /* 2 device slots */) {
+    HIP_CHECK(hipMemsetAsync(ctr, 0, 2 * sizeof(uint64_t), s));
+    if (n > 0)
+        k_flatmap<<<grid_for(n), BLK, 0, s>>>(ev, n, query, ctr, k0, v0, w0,
+                                              ctr + 1, k1, v1, w1);
     return DBSP_OK;
 }
 

@@ -29,6 +29,11 @@ struct SortArgs {
     uint64_t *ov[SORT_BATCH_MAX];
     int64_t *ow[SORT_BATCH_MAX];
     int64_t *d_len;  // device array, one length per batch
+    // optional per-batch DEVICE length (overrides n[b]): lets the sort launch
+    // chain behind the producing kernel without a host sync.  If the device
+    // length exceeds the fused capacity the kernel writes out_len = -1 and
+    // the host re-sorts through the sized paths after its next sync.
+    const int64_t *n_dev[SORT_BATCH_MAX];
 };
 
 // batched single-workgroup merges (one pair per workgroup; each na+nb <= 32768)
@@ -57,6 +62,12 @@ struct JoinCountArgs {
     uint32_t *cnts[3];
     uint64_t *offsets[3];
     int64_t *d_total;
+    // optional device lengths: nd_dev overrides nd (the delta length), and
+    // tn_dev overrides t.n[0] for single-batch traces that are themselves
+    // tick-fresh deltas.  Negative device lengths (speculative-sort overflow)
+    // make the plan write d_total = -1 for the host to detect at its sync.
+    const int64_t *nd_dev[3];
+    const int64_t *tn_dev[3];
 };
 
 
@@ -92,6 +103,10 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
 // fused single-workgroup sort+consolidate batch descriptors: up to 4
 // independent small (n <= 8192) raw batches sorted+consolidated concurrently,
 // one workgroup each, lengths left in d_len[i] (device)
+dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
+                                 int64_t n, int query, uint64_t *k0,
+                                 uint64_t *v0, int64_t *w0, uint64_t *k1,
+                                 uint64_t *v1, int64_t *w1, uint64_t *ctr);
 dbsp_status sort_cons_small_batch(hipStream_t s, const SortArgs &args);
 // min/max of k and v (one sync): mm = {kmin, vmin, kmax, vmax}
 dbsp_status minmax_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
