@@ -1038,7 +1038,7 @@ __global__ void k_join_emit(const uint64_t *dk, const uint64_t *dv,
 // (measured 53 GB/s).
 // ---------------------------------------------------------------------------
 
-#define MP_TILE 2048
+#define MP_TILE 1024
 #define MP_THREADS 256
 #define MP_ITEMS (MP_TILE / MP_THREADS)  // 16 diagonals per thread
 
@@ -1094,6 +1094,49 @@ __device__ inline void adjust_split_lds(const uint64_t *lk, const uint64_t *lv,
 //   granule = value<<2 | flag;  flag: 0 invalid, 1 aggregate, 2 prefix
 // ---------------------------------------------------------------------------
 
+
+// stage `n` rows of (k,v[,w]) into LDS at base `off`: pairs of u64 move as
+// ulonglong2 when both sides are 16 B aligned; odd head/tail go scalar
+template <bool EMIT, typename W>
+__device__ inline void stage_run(uint64_t *lk, uint64_t *lv, W *lw,
+                                 int64_t off, const uint64_t *gk,
+                                 const uint64_t *gv, const W *gw, int64_t n,
+                                 int tid) {
+    // align the GLOBAL side to 16 B (columns share base alignment, so one
+    // parity covers k, v and w)
+    const int64_t head = ((uintptr_t)gk & 15) != 0 ? 1 : 0;
+    if (head && n > 0 && tid == 0) {
+        lk[off] = gk[0];
+        lv[off] = gv[0];
+        if (EMIT) lw[off] = gw[0];
+    }
+    const int64_t pairs = n > head ? (n - head) >> 1 : 0;
+    const ulonglong2 *gk2 = (const ulonglong2 *)(gk + head);
+    const ulonglong2 *gv2 = (const ulonglong2 *)(gv + head);
+    for (int64_t p = tid; p < pairs; p += MP_THREADS) {
+        ulonglong2 kk = gk2[p];
+        ulonglong2 vv = gv2[p];
+        int64_t i = off + head + 2 * p;
+        lk[i] = kk.x; lk[i + 1] = kk.y;
+        lv[i] = vv.x; lv[i + 1] = vv.y;
+    }
+    if (EMIT) {
+        const ulonglong2 *gw2 = (const ulonglong2 *)(gw + head);
+        for (int64_t p = tid; p < pairs; p += MP_THREADS) {
+            ulonglong2 ww = gw2[p];
+            int64_t i = off + head + 2 * p;
+            lw[i] = ((const W *)&ww)[0];
+            lw[i + 1] = ((const W *)&ww)[1];
+        }
+    }
+    const int64_t tail = head + 2 * pairs;
+    if (tail < n && tid == 0) {
+        lk[off + tail] = gk[tail];
+        lv[off + tail] = gv[tail];
+        if (EMIT) lw[off + tail] = gw[tail];
+    }
+}
+
 template <bool EMIT, typename W>
 __global__ __launch_bounds__(MP_THREADS, 2) void k_mp_merge(
     const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
@@ -1112,17 +1155,10 @@ __global__ __launch_bounds__(MP_THREADS, 2) void k_mp_merge(
     const int64_t pb0 = pb[blk], pb1 = pb[blk + 1];
     const int64_t naL = pa1 - pa0, nbL = pb1 - pb0;
     const int64_t totL = naL + nbL;
-    // stage (coalesced 8B per lane)
-    for (int64_t i = tid; i < naL; i += MP_THREADS) {
-        lk[i] = ak[pa0 + i];
-        lv[i] = av[pa0 + i];
-        if (EMIT) lw[i] = aw[pa0 + i];
-    }
-    for (int64_t i = tid; i < nbL; i += MP_THREADS) {
-        lk[naL + i] = bk[pb0 + i];
-        lv[naL + i] = bv[pb0 + i];
-        if (EMIT) lw[naL + i] = bw[pb0 + i];
-    }
+    // stage: 16 B vector loads where the global offset is even (the common
+    // case after the head fix-up), scalar head/tail otherwise
+    stage_run<EMIT>(lk, lv, lw, 0, ak + pa0, av + pa0, aw + pa0, naL, tid);
+    stage_run<EMIT>(lk, lv, lw, naL, bk + pb0, bv + pb0, bw + pb0, nbL, tid);
     __syncthreads();
     // per-thread diagonals within the tile
     const int64_t items = (totL + MP_THREADS - 1) / MP_THREADS;
