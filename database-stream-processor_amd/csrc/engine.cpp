@@ -926,6 +926,14 @@ struct dbsp_engine {
     bool q5_have_prev = false;
     uint64_t q5_s0 = 0, q5_e0 = 0, q5_wm = 0;
 
+    // chained-tick device watermark state: {wm, s0, e0, have_prev} and the
+    // published bounds {s0,e0,s1,e1,have_prev,err} (q5/q8 single-rank path)
+    unsigned long long *d_wm = nullptr;     // 4 slots
+    unsigned long long *d_bounds = nullptr; // 6 slots
+    // host mirror read back at each tick sync (only err is load-bearing on
+    // the fast path; the rest feeds the explicit fallback)
+    unsigned long long h_bounds[6] = {0, 0, 0, 0, 0, 0};
+
     // last tick's output
     DevBatch output;
     std::vector<dbsp_event> q0_output;  // q0 CPU path
@@ -941,11 +949,20 @@ extern "C" dbsp_status dbsp_engine_create(dbsp_engine **out, dbsp_ctx *ctx,
     e->query = query;
     e->rank = rank;
     e->world = world;
+    if (ctx) {
+        if (hipMalloc(&e->d_wm, 10 * sizeof(uint64_t)) == hipSuccess) {
+            e->d_bounds = e->d_wm + 4;
+            (void)hipMemset(e->d_wm, 0, 10 * sizeof(uint64_t));
+        } else {
+            e->d_wm = nullptr;
+        }
+    }
     *out = e;
     return DBSP_OK;
 }
 
 extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
+    if (e && e->d_wm) (void)hipFree(e->d_wm);
     if (!e) return DBSP_OK;
     dbsp_ctx *c = e->ctx;
     for (Spine *s : {&e->a_int, &e->p_int, &e->pt_int, &e->at_int, &e->wp_int,
@@ -1537,6 +1554,19 @@ static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
     return DBSP_OK;
 }
 
+static TraceArgs trace_args_of(Spine &sp) {
+    TraceArgs t{};
+    for (auto &b : sp.batches) {
+        if (b.n == 0) continue;
+        t.k[t.nb] = b.k;
+        t.v[t.nb] = b.v;
+        t.w[t.nb] = b.w;
+        t.n[t.nb] = b.n;
+        t.nb++;
+    }
+    return t;
+}
+
 // ---- q3 tick (queries/q3.rs:35-63) ----
 static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     dbsp_ctx *c = e->ctx;
@@ -1764,27 +1794,122 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
 static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     dbsp_ctx *c = e->ctx;
     constexpr uint64_t TUMBLE_MS = 10000;
-    DevBatch dPT, dAT;
-    TRY(build_deltas(e, d_ev, n, dPT, dAT, true));
-    // watermark over auctions (q8.rs:63-65)
-    uint64_t lk = 0;
-    bool has = false;
-    TRY(last_key(c, dAT, &lk, &has));
-    uint64_t gmax = has ? lk : 0;
-    TRY(allreduce_max_u64(c, &gmax));
-    if (gmax > 0) e->q8_wm = std::max(e->q8_wm, gmax - TUMBLE_MS);
-    uint64_t rounded = e->q8_wm - e->q8_wm % TUMBLE_MS;
-    uint64_t s1 = rounded >= TUMBLE_MS ? rounded - TUMBLE_MS : 0;
-    uint64_t e1 = rounded;
-    // windowed people / auctions
+    // Single-rank ticks chain flatmap -> sorts -> device watermark -> window
+    // ranges with ONE sync; the watermark lives in e->d_wm so the bounds
+    // never round-trip to the host on the fast path.  Sharded ranks keep the
+    // explicit path (the watermark allreduce and exchange need host values).
+    const bool chain = !sharding_on(c) && e->d_wm;
+    DevBatch dPT, dAT, rawP, rawA;
     std::vector<DevBatch> wp_raw, wa_raw;
-    TRY(window_vs_spine(c, e->pt_int, dPT, e->q8_have_prev, e->q8_s0, e->q8_e0,
-                        s1, e1, wp_raw));
-    TRY(window_vs_spine(c, e->at_int, dAT, e->q8_have_prev, e->q8_s0, e->q8_e0,
-                        s1, e1, wa_raw));
-    e->q8_have_prev = true;
-    e->q8_s0 = s1;
-    e->q8_e0 = e1;
+    bool front_done = false;
+    if (chain) {
+        TRY(build_deltas_chain(e, d_ev, n, rawP, rawA, dPT, dAT));
+        TRY(dbspk::wm_update(c->stream, dAT.k, c->d_len + 11, TUMBLE_MS,
+                             TUMBLE_MS, TUMBLE_MS, e->d_wm, e->d_bounds));
+        if ((int)e->pt_int.batches.size() > MAX_TRACE_BATCHES)
+            TRY(e->pt_int.consolidate_all(c));
+        if ((int)e->at_int.batches.size() > MAX_TRACE_BATCHES)
+            TRY(e->at_int.consolidate_all(c));
+        TraceArgs taP = trace_args_of(e->pt_int), taA = trace_args_of(e->at_int);
+        const int nregP = 3 * taP.nb + 1, nregA = 3 * taA.nb + 1;
+        int64_t *tableP = (int64_t *)arena_alloc(c, (size_t)nregP * 5 * 8 + 8);
+        int64_t *tableA = (int64_t *)arena_alloc(c, (size_t)nregA * 5 * 8 + 8);
+        bool launched = tableP && tableA;
+        if (launched) {
+            ScopedTimer t(c, 4, 0.0);
+            TRY(dbspk::window_ranges_chain(c->stream, taP, dPT.k,
+                                           c->d_len + 10, e->d_bounds, tableP,
+                                           c->d_len + 12));
+            TRY(dbspk::window_ranges_chain(c->stream, taA, dAT.k,
+                                           c->d_len + 11, e->d_bounds, tableA,
+                                           c->d_len + 13));
+        }
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 16 * sizeof(int64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipMemcpyAsync(e->h_bounds, e->d_bounds,
+                                    6 * sizeof(uint64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        const bool lost = !launched || e->h_bounds[5] != 0 ||
+                          c->h_len[10] < 0 || c->h_len[11] < 0 ||
+                          c->h_len[12] < 0 || c->h_len[13] < 0;
+        if (!lost) {
+            dPT.n = c->h_len[10];
+            dAT.n = c->h_len[11];
+            if (c->h_len[12] > 0) {
+                DevBatch o;
+                TRY(alloc_batch(c, c->h_len[12], o, true));
+                TRY(dbspk::window_emit_multi(c->stream, taP, dPT.k, dPT.v,
+                                             dPT.w, tableP, nregP,
+                                             c->h_len[12], o.k, o.v, o.w));
+                wp_raw.push_back(o);
+            }
+            if (c->h_len[13] > 0) {
+                DevBatch o;
+                TRY(alloc_batch(c, c->h_len[13], o, true));
+                TRY(dbspk::window_emit_multi(c->stream, taA, dAT.k, dAT.v,
+                                             dAT.w, tableA, nregA,
+                                             c->h_len[13], o.k, o.v, o.w));
+                wa_raw.push_back(o);
+            }
+            front_done = true;
+        } else {
+            // speculation lost (delta overflowed the fused sort, or no arena
+            // for the tables): re-sort with real lengths, pull the watermark
+            // state to the host, run the explicit window path, write back
+            rawP.n = c->h_len[8];
+            rawA.n = c->h_len[9];
+            free_batch(c, dPT);
+            free_batch(c, dAT);
+            TRY(sort_consolidate_batch(c, rawP, dPT));
+            TRY(sort_consolidate_batch(c, rawA, dAT));
+            rawP = DevBatch{};
+            rawA = DevBatch{};
+            unsigned long long hw[4];
+            HIP_CHECK_ST(hipMemcpyAsync(hw, e->d_wm, sizeof(hw),
+                                        hipMemcpyDeviceToHost, c->stream));
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            uint64_t wm = hw[0];
+            uint64_t lk = 0;
+            bool has = false;
+            TRY(last_key(c, dAT, &lk, &has));
+            if (has && lk > 0) wm = std::max(wm, lk - TUMBLE_MS);
+            uint64_t rounded = wm - wm % TUMBLE_MS;
+            uint64_t s1 = rounded >= TUMBLE_MS ? rounded - TUMBLE_MS : 0;
+            uint64_t e1 = rounded;
+            TRY(window_vs_spine(c, e->pt_int, dPT, hw[3] != 0, hw[1], hw[2],
+                                s1, e1, wp_raw));
+            TRY(window_vs_spine(c, e->at_int, dAT, hw[3] != 0, hw[1], hw[2],
+                                s1, e1, wa_raw));
+            hw[0] = wm;
+            hw[1] = s1;
+            hw[2] = e1;
+            hw[3] = 1;
+            HIP_CHECK_ST(hipMemcpyAsync(e->d_wm, hw, sizeof(hw),
+                                        hipMemcpyHostToDevice, c->stream));
+            front_done = true;
+        }
+    }
+    if (!front_done) {
+        TRY(build_deltas(e, d_ev, n, dPT, dAT, true));
+        // watermark over auctions (q8.rs:63-65)
+        uint64_t lk = 0;
+        bool has = false;
+        TRY(last_key(c, dAT, &lk, &has));
+        uint64_t gmax = has ? lk : 0;
+        TRY(allreduce_max_u64(c, &gmax));
+        if (gmax > 0) e->q8_wm = std::max(e->q8_wm, gmax - TUMBLE_MS);
+        uint64_t rounded = e->q8_wm - e->q8_wm % TUMBLE_MS;
+        uint64_t s1 = rounded >= TUMBLE_MS ? rounded - TUMBLE_MS : 0;
+        uint64_t e1 = rounded;
+        TRY(window_vs_spine(c, e->pt_int, dPT, e->q8_have_prev, e->q8_s0,
+                            e->q8_e0, s1, e1, wp_raw));
+        TRY(window_vs_spine(c, e->at_int, dAT, e->q8_have_prev, e->q8_s0,
+                            e->q8_e0, s1, e1, wa_raw));
+        e->q8_have_prev = true;
+        e->q8_s0 = s1;
+        e->q8_e0 = e1;
+    }
     TRY(spines_insert_pair(c, e->pt_int, dPT, e->at_int, dAT));
     // map_index / map + consolidate
     DevBatch wpr, war, dWP, dWA;

@@ -1731,10 +1731,63 @@ __global__ void k_window_emit(const uint64_t *tk, const uint64_t *tv,
 // batch region in one ranges launch + one emit launch (window.rs:144-220
 // evaluated per spine batch; the trace excludes the current tick).
 // Region table rows: [src_batch(-1 = tick batch), lo, len, sign, goff]
+
+// ---------------------------------------------------------------------------
+// device-resident watermark (q5.rs:85-90 / q8.rs:63-65 waterline): single
+// thread reads the sorted delta's last key (the tick's max event-time),
+// advances the persistent watermark state and publishes the window bounds
+// for the chained ranges launch.  state = {wm, s0, e0, have_prev};
+// bounds = {s0, e0, s1, e1, have_prev, err}.  err=1 when the delta length is
+// the speculative-sort overflow sentinel — the host then recomputes on its
+// explicit path (the state is left untouched).
+// ---------------------------------------------------------------------------
+__global__ void k_wm_update(const uint64_t *ak, const int64_t *n_dev,
+                            uint64_t width, uint64_t tumble, uint64_t lag,
+                            unsigned long long *state,
+                            unsigned long long *bounds) {
+    const int64_t n = *n_dev;
+    if (n < 0) {
+        bounds[5] = 1;
+        return;
+    }
+    uint64_t wm = state[0];
+    if (n > 0) {
+        uint64_t gmax = ak[n - 1];
+        if (gmax > 0) wm = max(wm, gmax - lag);
+    }
+    uint64_t rounded = wm - wm % tumble;
+    uint64_t s1 = rounded >= width ? rounded - width : 0;
+    uint64_t e1 = rounded;
+    bounds[0] = state[1];
+    bounds[1] = state[2];
+    bounds[2] = s1;
+    bounds[3] = e1;
+    bounds[4] = state[3];
+    bounds[5] = 0;
+    state[0] = wm;
+    state[1] = s1;
+    state[2] = e1;
+    state[3] = 1;
+}
+
 __global__ void k_window_ranges_multi(TraceArgs t, const uint64_t *bk,
                                       int64_t bn, int have_prev, uint64_t s0,
                                       uint64_t e0, uint64_t s1, uint64_t e1,
-                                      int64_t *table, int64_t *d_total) {
+                                      int64_t *table, int64_t *d_total,
+                                      const int64_t *bn_dev,
+                                      const unsigned long long *bounds) {
+    if (bounds) {  // chained: bounds from k_wm_update, delta length from sort
+        if (bounds[5] != 0 || (bn_dev && *bn_dev < 0)) {
+            if (threadIdx.x == 0) *d_total = -1;
+            return;
+        }
+        s0 = bounds[0];
+        e0 = bounds[1];
+        s1 = bounds[2];
+        e1 = bounds[3];
+        have_prev = (int)bounds[4];
+        if (bn_dev) bn = *bn_dev;
+    }
     const int nreg = 3 * t.nb + 1;
     for (int r = threadIdx.x; r < nreg; r += blockDim.x) {
         int64_t src = -1, lo = 0, len = 0, sign = 1;
@@ -2457,7 +2510,24 @@ dbsp_status window_ranges_multi(hipStream_t s, const TraceArgs &t,
                                 uint64_t s0, uint64_t e0, uint64_t s1,
                                 uint64_t e1, int64_t *table, int64_t *d_total) {
     k_window_ranges_multi<<<1, BLK, 0, s>>>(t, bk, bn, have_prev, s0, e0, s1,
-                                            e1, table, d_total);
+                                            e1, table, d_total, nullptr,
+                                            nullptr);
+    return DBSP_OK;
+}
+
+dbsp_status wm_update(hipStream_t s, const uint64_t *ak, const int64_t *n_dev,
+                      uint64_t width, uint64_t tumble, uint64_t lag,
+                      unsigned long long *state, unsigned long long *bounds) {
+    k_wm_update<<<1, 1, 0, s>>>(ak, n_dev, width, tumble, lag, state, bounds);
+    return DBSP_OK;
+}
+
+dbsp_status window_ranges_chain(hipStream_t s, const TraceArgs &t,
+                                const uint64_t *bk, const int64_t *bn_dev,
+                                const unsigned long long *bounds,
+                                int64_t *table, int64_t *d_total) {
+    k_window_ranges_multi<<<1, BLK, 0, s>>>(t, bk, 0, 0, 0, 0, 0, 0, table,
+                                            d_total, bn_dev, bounds);
     return DBSP_OK;
 }
 

@@ -103,6 +103,13 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
 // fused single-workgroup sort+consolidate batch descriptors: up to 4
 // independent small (n <= 8192) raw batches sorted+consolidated concurrently,
 // one workgroup each, lengths left in d_len[i] (device)
+dbsp_status wm_update(hipStream_t s, const uint64_t *ak, const int64_t *n_dev,
+                      uint64_t width, uint64_t tumble, uint64_t lag,
+                      unsigned long long *state, unsigned long long *bounds);
+dbsp_status window_ranges_chain(hipStream_t s, const TraceArgs &t,
+                                const uint64_t *bk, const int64_t *bn_dev,
+                                const unsigned long long *bounds,
+                                int64_t *table, int64_t *d_total);
 dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
                                  int64_t n, int query, uint64_t *k0,
                                  uint64_t *v0, int64_t *w0, uint64_t *k1,
