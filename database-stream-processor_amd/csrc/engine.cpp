@@ -1817,10 +1817,10 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         bool launched = tableP && tableA;
         if (launched) {
             ScopedTimer t(c, 4, 0.0);
-            TRY(dbspk::window_ranges_chain(c->stream, taP, dPT.k,
+            TRY(dbspk::window_ranges_chain(c->stream, taP, dPT.k, 0,
                                            c->d_len + 10, e->d_bounds, tableP,
                                            c->d_len + 12));
-            TRY(dbspk::window_ranges_chain(c->stream, taA, dAT.k,
+            TRY(dbspk::window_ranges_chain(c->stream, taA, dAT.k, 0,
                                            c->d_len + 11, e->d_bounds, tableA,
                                            c->d_len + 13));
         }
@@ -1953,23 +1953,114 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
 static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     dbsp_ctx *c = e->ctx;
     constexpr uint64_t WIDTH_MS = 10000, TUMBLE_MS = 2000, WM_LAG_MS = 4000;
-    DevBatch dBT, dummy;
-    TRY(build_deltas(e, d_ev, n, dBT, dummy, false));
-    uint64_t lk = 0;
-    bool has = false;
-    TRY(last_key(c, dBT, &lk, &has));
-    uint64_t gmax = has ? lk : 0;
-    TRY(allreduce_max_u64(c, &gmax));
-    if (gmax > 0) e->q5_wm = std::max(e->q5_wm, gmax - WM_LAG_MS);
-    uint64_t rounded = e->q5_wm - e->q5_wm % TUMBLE_MS;
-    uint64_t s1 = rounded >= WIDTH_MS ? rounded - WIDTH_MS : 0;
-    uint64_t e1 = rounded;
+    // Single-rank ticks: flatmap and the dense-range min/max probe chain into
+    // ONE sync, the bid delta sorts through the dense path, and the
+    // watermark/window-ranges run device-side as in q8.
+    const bool chain = !sharding_on(c) && e->d_wm;
+    DevBatch dBT;
     std::vector<DevBatch> wb_raw;
-    TRY(window_vs_spine(c, e->bt_int, dBT, e->q5_have_prev, e->q5_s0, e->q5_e0,
-                        s1, e1, wb_raw));
-    e->q5_have_prev = true;
-    e->q5_s0 = s1;
-    e->q5_e0 = e1;
+    bool front_done = false;
+    if (chain) {
+        const int64_t cap = n > 0 ? n : 1;
+        DevBatch raw0, raw1;
+        TRY(alloc_batch(c, cap, raw0, true));
+        TRY(alloc_batch(c, cap, raw1, true));
+        TRY(dbspk::flatmap_events_chain(c->stream, d_ev, n, e->query, raw0.k,
+                                        raw0.v, raw0.w, raw1.k, raw1.v, raw1.w,
+                                        (uint64_t *)(c->d_len + 8)));
+        TRY(dbspk::minmax_rows_chain(c->stream, raw0.k, raw0.v, cap,
+                                     c->d_len + 8,
+                                     (unsigned long long *)(c->d_len + 12)));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 16 * sizeof(int64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        free_batch(c, raw1);
+        const int64_t n0 = c->h_len[8];
+        raw0.n = n0;
+        if (n0 == 0) {
+            dBT = DevBatch{};
+            free_batch(c, raw0);
+        } else {
+            const uint64_t *mm = (const uint64_t *)(c->h_len + 12);
+            const uint64_t kr = mm[2] - mm[0], vr = mm[3] - mm[1];
+            const int64_t dense_cap =
+                std::min<int64_t>((int64_t)1 << 22, 8 * n0);
+            if (n0 > 8192 && kr < (uint64_t)dense_cap &&
+                vr < (uint64_t)dense_cap &&
+                (int64_t)((kr + 1) * (vr + 1)) <= dense_cap) {
+                ScopedTimer t(c, 0, (double)n0 * 48.0);
+                DevBatch res;
+                TRY(alloc_batch(c, n0, res));
+                TRY(dbspk::sort_cons_dense(c->stream, raw0.k, raw0.v, raw0.w,
+                                           n0, mm[0], mm[1], (int64_t)(kr + 1),
+                                           (int64_t)(vr + 1), res.k, res.v,
+                                           res.w, &res.n));
+                free_batch(c, raw0);
+                dBT = res;
+            } else {
+                TRY(sort_consolidate_batch(c, raw0, dBT));
+            }
+        }
+        // device watermark + chained window ranges (q5.rs:85-90)
+        TRY(dbspk::wm_update_n(c->stream, dBT.k, dBT.n, WIDTH_MS, TUMBLE_MS,
+                               WM_LAG_MS, e->d_wm, e->d_bounds));
+        if ((int)e->bt_int.batches.size() > MAX_TRACE_BATCHES)
+            TRY(e->bt_int.consolidate_all(c));
+        TraceArgs taB = trace_args_of(e->bt_int);
+        const int nregB = 3 * taB.nb + 1;
+        int64_t *tableB = (int64_t *)arena_alloc(c, (size_t)nregB * 5 * 8 + 8);
+        if (tableB) {
+            {
+                ScopedTimer t(c, 4, 0.0);
+                TRY(dbspk::window_ranges_chain(c->stream, taB, dBT.k, dBT.n,
+                                               nullptr, e->d_bounds, tableB,
+                                               c->d_len + 7));
+            }
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 7, c->d_len + 7,
+                                        sizeof(int64_t), hipMemcpyDeviceToHost,
+                                        c->stream));
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            if (c->h_len[7] > 0) {
+                DevBatch o;
+                TRY(alloc_batch(c, c->h_len[7], o, true));
+                TRY(dbspk::window_emit_multi(c->stream, taB, dBT.k, dBT.v,
+                                             dBT.w, tableB, nregB, c->h_len[7],
+                                             o.k, o.v, o.w));
+                wb_raw.push_back(o);
+            }
+            front_done = true;
+        } else {
+            // no arena for the table: wm_update already advanced the device
+            // state and published the bounds — read them and run the
+            // explicit window path
+            HIP_CHECK_ST(hipMemcpyAsync(e->h_bounds, e->d_bounds,
+                                        6 * sizeof(uint64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            TRY(window_vs_spine(c, e->bt_int, dBT, e->h_bounds[4] != 0,
+                                e->h_bounds[0], e->h_bounds[1], e->h_bounds[2],
+                                e->h_bounds[3], wb_raw));
+            front_done = true;
+        }
+    }
+    if (!front_done) {
+        DevBatch dummy;
+        TRY(build_deltas(e, d_ev, n, dBT, dummy, false));
+        uint64_t lk = 0;
+        bool has = false;
+        TRY(last_key(c, dBT, &lk, &has));
+        uint64_t gmax = has ? lk : 0;
+        TRY(allreduce_max_u64(c, &gmax));
+        if (gmax > 0) e->q5_wm = std::max(e->q5_wm, gmax - WM_LAG_MS);
+        uint64_t rounded = e->q5_wm - e->q5_wm % TUMBLE_MS;
+        uint64_t s1 = rounded >= WIDTH_MS ? rounded - WIDTH_MS : 0;
+        uint64_t e1 = rounded;
+        TRY(window_vs_spine(c, e->bt_int, dBT, e->q5_have_prev, e->q5_s0,
+                            e->q5_e0, s1, e1, wb_raw));
+        e->q5_have_prev = true;
+        e->q5_s0 = s1;
+        e->q5_e0 = e1;
+    }
     TRY(e->bt_int.insert(c, dBT));
     DevBatch wbr, dWB;
     TRY(finalize_raw(c, wb_raw, wbr));

@@ -734,7 +734,9 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
 // ---------------------------------------------------------------------------
 
 __global__ void k_minmax_rows(const uint64_t *k, const uint64_t *v, int64_t n,
-                              unsigned long long *mm /* kmin,vmin,kmax,vmax */) {
+                              unsigned long long *mm /* kmin,vmin,kmax,vmax */,
+                              const int64_t *n_dev = nullptr) {
+    if (n_dev) n = *n_dev;
     uint64_t mk = 0, mv = 0, nk = ~0ull, nv = ~0ull;
     for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
          i += (int64_t)gridDim.x * blockDim.x) {
@@ -1741,11 +1743,29 @@ __global__ void k_window_emit(const uint64_t *tk, const uint64_t *tv,
 // the speculative-sort overflow sentinel — the host then recomputes on its
 // explicit path (the state is left untouched).
 // ---------------------------------------------------------------------------
+__device__ void wm_update_body(const uint64_t *ak, int64_t n, uint64_t width,
+                               uint64_t tumble, uint64_t lag,
+                               unsigned long long *state,
+                               unsigned long long *bounds);
+
+__global__ void k_wm_update_n(const uint64_t *ak, int64_t n, uint64_t width,
+                              uint64_t tumble, uint64_t lag,
+                              unsigned long long *state,
+                              unsigned long long *bounds) {
+    wm_update_body(ak, n, width, tumble, lag, state, bounds);
+}
+
 __global__ void k_wm_update(const uint64_t *ak, const int64_t *n_dev,
                             uint64_t width, uint64_t tumble, uint64_t lag,
                             unsigned long long *state,
                             unsigned long long *bounds) {
-    const int64_t n = *n_dev;
+    wm_update_body(ak, *n_dev, width, tumble, lag, state, bounds);
+}
+
+__device__ void wm_update_body(const uint64_t *ak, int64_t n, uint64_t width,
+                               uint64_t tumble, uint64_t lag,
+                               unsigned long long *state,
+                               unsigned long long *bounds) {
     if (n < 0) {
         bounds[5] = 1;
         return;
@@ -2240,6 +2260,19 @@ dbsp_status minmax_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
     return DBSP_OK;
 }
 
+// chained variant: n read from the device, results left in mm_dev[0..3]
+// (caller reads them at its own sync); grid sized by the capacity bound
+dbsp_status minmax_rows_chain(hipStream_t s, const uint64_t *k,
+                              const uint64_t *v, int64_t cap,
+                              const int64_t *n_dev,
+                              unsigned long long *mm_dev) {
+    const unsigned long long init[4] = {~0ull, ~0ull, 0ull, 0ull};
+    HIP_CHECK(hipMemcpyAsync(mm_dev, init, sizeof(init), hipMemcpyHostToDevice,
+                             s));
+    k_minmax_rows<<<grid_for(cap), BLK, 0, s>>>(k, v, 0, mm_dev, n_dev);
+    return DBSP_OK;
+}
+
 dbsp_status sort_cons_dense(hipStream_t s, const uint64_t *k, const uint64_t *v,
                             const int64_t *w, int64_t n, uint64_t kbase,
                             uint64_t vbase, int64_t kspan, int64_t vspan,
@@ -2522,11 +2555,20 @@ dbsp_status wm_update(hipStream_t s, const uint64_t *ak, const int64_t *n_dev,
     return DBSP_OK;
 }
 
+dbsp_status wm_update_n(hipStream_t s, const uint64_t *ak, int64_t n,
+                        uint64_t width, uint64_t tumble, uint64_t lag,
+                        unsigned long long *state,
+                        unsigned long long *bounds) {
+    k_wm_update_n<<<1, 1, 0, s>>>(ak, n, width, tumble, lag, state, bounds);
+    return DBSP_OK;
+}
+
 dbsp_status window_ranges_chain(hipStream_t s, const TraceArgs &t,
-                                const uint64_t *bk, const int64_t *bn_dev,
+                                const uint64_t *bk, int64_t bn,
+                                const int64_t *bn_dev,
                                 const unsigned long long *bounds,
                                 int64_t *table, int64_t *d_total) {
-    k_window_ranges_multi<<<1, BLK, 0, s>>>(t, bk, 0, 0, 0, 0, 0, 0, table,
+    k_window_ranges_multi<<<1, BLK, 0, s>>>(t, bk, bn, 0, 0, 0, 0, 0, table,
                                             d_total, bn_dev, bounds);
     return DBSP_OK;
 }

@@ -106,10 +106,18 @@ dbsp_status merge_rows(hipStream_t s, const uint64_t *ak, const uint64_t *av,
 dbsp_status wm_update(hipStream_t s, const uint64_t *ak, const int64_t *n_dev,
                       uint64_t width, uint64_t tumble, uint64_t lag,
                       unsigned long long *state, unsigned long long *bounds);
+dbsp_status wm_update_n(hipStream_t s, const uint64_t *ak, int64_t n,
+                        uint64_t width, uint64_t tumble, uint64_t lag,
+                        unsigned long long *state, unsigned long long *bounds);
 dbsp_status window_ranges_chain(hipStream_t s, const TraceArgs &t,
-                                const uint64_t *bk, const int64_t *bn_dev,
+                                const uint64_t *bk, int64_t bn,
+                                const int64_t *bn_dev,
                                 const unsigned long long *bounds,
                                 int64_t *table, int64_t *d_total);
+dbsp_status minmax_rows_chain(hipStream_t s, const uint64_t *k,
+                              const uint64_t *v, int64_t cap,
+                              const int64_t *n_dev,
+                              unsigned long long *mm_dev);
 dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
                                  int64_t n, int query, uint64_t *k0,
                                  uint64_t *v0, int64_t *w0, uint64_t *k1,
