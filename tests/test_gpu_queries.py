@@ -151,3 +151,26 @@ def test_q0_host_path(ctx):
     exp = oracle.q0_step(evs)
     assert np.array_equal(got, exp)
     eng.close()
+
+
+@pytest.mark.parametrize("query", [3, 5, 8])
+def test_query_parity_oversized_deltas(ctx, query):
+    """Ticks whose per-stream deltas exceed the fused-sort capacity (8192
+    rows) and whose key boxes are too wide for the dense-range path: the
+    chained single-rank tick loses its speculation (the sort kernels write
+    the -1 sentinels) and must recover through the sized sort paths with
+    the lengths read back at the tick sync."""
+    from helpers import events
+    rng = np.random.default_rng(53)
+    evs = []
+    for i in range(12_000):
+        dt = 10_000_000 + i
+        evs.append(person_event(int(rng.integers(0, 1 << 40)),
+                                i % 997, i % 15, i % 6, dt=dt))
+        evs.append(auction_event(int(rng.integers(0, 1 << 40)),
+                                 int(rng.integers(0, 1 << 40)), 10 + i % 5,
+                                 dt=dt))
+        # wide auction ids reject the dense-range consolidate too
+        evs.append(bid_event(int(rng.integers(0, 1 << 40)), dt))
+    _run_parity(ctx, query, events(*evs), tick=30_000,
+                seed_note="+oversized")
