@@ -74,24 +74,15 @@ def run_engine(ctx_dev, query, events, n_ticks, tick, world, rank, nccl_id,
     eng = Engine(ctx, query=query, rank=rank, world=world)
     eng.stage(events)
 
-    def tick_range(t):
-        # events are the rank's own pre-sliced stream (tick-major)
-        base = t * tick
-        return base, base + tick
-
     warmup_ticks = n_ticks - timed_ticks
-    for t in range(warmup_ticks):
-        lo, hi = tick_range(t)
-        eng.step_staged(lo, hi)
+    eng.run_staged(0, warmup_ticks * tick, tick)
     ctx.sync()
     if torch.cuda.is_available():
         torch.cuda.synchronize()
     if dist is not None:
         dist.barrier()
     t0 = time.perf_counter()
-    for t in range(warmup_ticks, n_ticks):
-        lo, hi = tick_range(t)
-        eng.step_staged(lo, hi)
+    eng.run_staged(warmup_ticks * tick, n_ticks * tick, tick)
     ctx.sync()
     if torch.cuda.is_available():
         torch.cuda.synchronize()

@@ -934,8 +934,13 @@ struct dbsp_engine {
     // the fast path; the rest feeds the explicit fallback)
     unsigned long long h_bounds[6] = {0, 0, 0, 0, 0, 0};
 
-    // last tick's output
+    // last tick's output; out_store is a reused capacity buffer so the hot
+    // path allocates nothing for it (output_is_store marks when output
+    // aliases it and must not be freed)
     DevBatch output;
+    DevBatch out_store;
+    int64_t out_cap = 0;
+    bool output_is_store = false;
     std::vector<dbsp_event> q0_output;  // q0 CPU path
 };
 
@@ -961,6 +966,8 @@ extern "C" dbsp_status dbsp_engine_create(dbsp_engine **out, dbsp_ctx *ctx,
     return DBSP_OK;
 }
 
+static void engine_free_output(dbsp_engine *e);
+
 extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
     if (e && e->d_wm) (void)hipFree(e->d_wm);
     if (!e) return DBSP_OK;
@@ -972,7 +979,8 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
     free_batch(c, e->maxin_int);
     free_batch(c, e->maxout_int);
     free_batch(c, e->maxz_int);
-    free_batch(c, e->output);
+    engine_free_output(e);
+    if (e->out_store.k) free_batch(c, e->out_store);
     if (e->d_events) (void)dbspk::cache_free(e->d_events, c->stream);
     (void)hipStreamSynchronize(c->stream);
     delete e;
@@ -1554,6 +1562,12 @@ static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
     return DBSP_OK;
 }
 
+static void engine_free_output(dbsp_engine *e) {
+    if (!e->output_is_store) free_batch(e->ctx, e->output);
+    e->output = DevBatch{};
+    e->output_is_store = false;
+}
+
 static TraceArgs trace_args_of(Spine &sp) {
     TraceArgs t{};
     for (auto &b : sp.batches) {
@@ -1742,7 +1756,7 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     }
     // Launch the output consolidate asynchronously, overlap it with the spine
     // inserts (independent work), and read its length at their sync.
-    free_batch(c, e->output);
+    engine_free_output(e);
     int64_t cat_n = 0;
     for (auto &b : outs) cat_n += b.n;
     bool async_final = cat_n > 0 && cat_n <= 8192;
@@ -1767,7 +1781,12 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             outs.clear();
         }
         TRY(alloc_batch(c, cat_n, scratch, true));
-        TRY(alloc_batch(c, cat_n, res));
+        if (e->out_cap < cat_n) {
+            if (e->out_store.k) free_batch(c, e->out_store);
+            e->out_cap = std::max<int64_t>(2 * cat_n, 8192);
+            TRY(alloc_batch(c, e->out_cap, e->out_store));
+        }
+        res = e->out_store;
         SortArgs sa{};
         sa.nb = 1;
         sa.kin[0] = cat.k; sa.vin[0] = cat.v; sa.win[0] = cat.w; sa.n[0] = cat_n;
@@ -1784,6 +1803,7 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
         res.n = c->h_len[6];
         e->output = res;
+        e->output_is_store = true;
     } else {
         TRY(finalize_raw(c, outs, e->output));
     }
@@ -1944,7 +1964,7 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         else free_batch(c, o);
     }
     TRY(spines_insert_pair(c, e->wp_int, dWP, e->wa_int, dWA));
-    free_batch(c, e->output);
+    engine_free_output(e);
     TRY(finalize_raw(c, outs, e->output));
     return DBSP_OK;
 }
@@ -2143,7 +2163,7 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     }
     TRY(e->bc_int.insert(c, dBC));
     TRY(e->counts_int.insert(c, dCounts));
-    free_batch(c, e->output);
+    engine_free_output(e);
     TRY(finalize_raw(c, outs, e->output));
     return DBSP_OK;
 }
@@ -2163,6 +2183,19 @@ extern "C" dbsp_status dbsp_engine_step_staged(dbsp_engine *e, int64_t lo,
         case 8: return q8_step(e, d_ev, hi - lo);
     }
     return DBSP_ERR_INVALID;
+}
+
+extern "C" dbsp_status dbsp_engine_run_staged(dbsp_engine *e, int64_t lo,
+                                              int64_t hi, int64_t tick) {
+    // the benchmark hot loop: ticks run back-to-back inside one C call so the
+    // per-tick host cost excludes the Python/ctypes round-trip
+    if (tick <= 0) return DBSP_ERR_INVALID;
+    for (int64_t t = lo; t < hi; t += tick) {
+        dbsp_status st =
+            dbsp_engine_step_staged(e, t, std::min(hi, t + tick));
+        if (st != DBSP_OK) return st;
+    }
+    return DBSP_OK;
 }
 
 extern "C" dbsp_status dbsp_engine_step(dbsp_engine *e, const dbsp_event *events,

@@ -106,6 +106,8 @@ def _L():
         L.dbsp_engine_stage_events.argtypes = [vp, vp, i64]
         L.dbsp_engine_step_staged.restype = i32
         L.dbsp_engine_step_staged.argtypes = [vp, i64, i64]
+        L.dbsp_engine_run_staged.restype = i32
+        L.dbsp_engine_run_staged.argtypes = [vp, i64, i64, i64]
         L.dbsp_engine_output.restype = i32
         L.dbsp_engine_output.argtypes = [vp, vp, i64, ctypes.POINTER(i64)]
         L.dbsp_engine_output_events.restype = i32
@@ -393,6 +395,10 @@ class Engine:
 
     def step_staged(self, lo, hi):
         _check(self._lib.dbsp_engine_step_staged(self._h, lo, hi), "step")
+
+    def run_staged(self, lo, hi, tick):
+        """Tick loop inside the C side (the benchmark hot path)."""
+        _check(self._lib.dbsp_engine_run_staged(self._h, lo, hi, tick), "run")
 
     def step(self, events: np.ndarray):
         events = np.ascontiguousarray(events, dtype=EVENT_DT)

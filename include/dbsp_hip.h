@@ -272,6 +272,11 @@ dbsp_status dbsp_engine_step(dbsp_engine *e, const dbsp_event *events, int64_t n
 dbsp_status dbsp_engine_stage_events(dbsp_engine *e, const dbsp_event *events,
                                      int64_t n);
 dbsp_status dbsp_engine_step_staged(dbsp_engine *e, int64_t lo, int64_t hi);
+/* Run staged events [lo,hi) in tick-sized steps inside one call (the
+ * benchmark loop; per-tick outputs are produced and replaced in turn —
+ * read dbsp_engine_output afterwards for the LAST tick only). */
+dbsp_status dbsp_engine_run_staged(dbsp_engine *e, int64_t lo, int64_t hi,
+                                   int64_t tick);
 
 /* Copy last tick's output delta to host rows; returns count via *n_out
  * (capacity cap rows; DBSP_ERR_OVERFLOW if larger). For q0 the output rows are
