@@ -477,6 +477,23 @@ __device__ inline uint32_t fuse_scan(uint32_t thread_sum, uint32_t *wave_tot,
 
 template <typename W>
 __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs args) {
+    // Fused sort+consolidate of one raw batch per workgroup (n <= 8192).
+    //
+    // LSD radix over 4-bit digits of the concatenated (k - kmin, v - vmin)
+    // key, restructured around the 64-wide wavefront:
+    //  - positions are STRIDED (row i handled by lane i%64 of wave (i/64)%16
+    //    in round i/1024), so every LDS access is conflict-free and ranks
+    //    assigned in (digit, round, wave, lane) order equal position order —
+    //    i.e. the sort is stable without per-thread counters;
+    //  - ranks come from 4 bit-sliced wave ballots per round (no 16x1024
+    //    counter array, no per-item LDS counter increments); per-(digit,
+    //    round, wave) totals live in a 16*8*16-cell array scanned by the
+    //    block scan;
+    //  - entries carry the NEXT 16 key bits next to the row index
+    //    (entry = key16 << 13 | idx), so digit passes never touch global
+    //    memory: one gather per 4-pass group rebuilds the entries.
+    // The consolidation reuses the same ballot machinery for head flags and
+    // the nonzero compaction.
     const int batch = blockIdx.x;
     const uint64_t *kin = args.kin[batch];
     const uint64_t *vin = args.vin[batch];
@@ -492,11 +509,14 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     const int64_t n_chain = args.n_dev[batch] ? *args.n_dev[batch] : -1;
     __shared__ uint32_t bufA[FUSE_MAX];
     __shared__ uint32_t bufB[FUSE_MAX];
-    __shared__ uint32_t cnt[FUSE_DIGITS * FUSE_THREADS];  // rank counters
+    __shared__ uint32_t cnt[FUSE_MAX];  // ballot cells / f64 head positions
     __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
     __shared__ uint64_t smax[2];
     __shared__ uint64_t smin[2];
     const int tid = threadIdx.x;
+    const int wid = tid / WAVE;
+    const int lane = tid % WAVE;
+    const uint64_t lt = ((uint64_t)1 << lane) - 1;
 
     if (args.n_dev[batch]) {
         if (n_chain > FUSE_MAX) {  // speculation lost: host re-sorts
@@ -510,9 +530,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
         return;
     }
 
-    // ---- significant bits of (max-min) for k and v: sorting by (limb - min)
-    // preserves order and collapses narrow ranges (per-tick timestamps span a
-    // few ms; ids cluster) to far fewer digit passes ----
+    // ---- significant bits of (max-min) for k and v ----
     if (tid == 0) {
         smax[0] = 0; smax[1] = 0;
         smin[0] = ~0ull; smin[1] = ~0ull;
@@ -537,119 +555,172 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     int knibs = 0, vnibs = 0;
     while (knibs < 16 && (krange >> (4 * knibs)) != 0) knibs++;
     while (vnibs < 16 && (vrange >> (4 * vnibs)) != 0) vnibs++;
-
-    // adaptive sizing: only t_act threads carry items (wave-rounded), so the
-    // per-pass counter array is 16*t_act cells, not 16*1024 — small deltas
-    // (hundreds of rows) pay for their size, not for FUSE_MAX
-    const int t_act = (int)min((int64_t)FUSE_THREADS,
-                               ((n + FUSE_ITEMS - 1) / FUSE_ITEMS + WAVE - 1) /
-                                   WAVE * WAVE);
-    const int n_pad = t_act * FUSE_ITEMS;
-    const int scan_cells = FUSE_DIGITS * t_act;
-    const int cells_per_thread = (scan_cells + FUSE_THREADS - 1) / FUSE_THREADS;
-
-    // ---- init permutation (entry = digit<<13 | idx; idx < 8192) ----
-    for (int i = tid; i < n_pad; i += FUSE_THREADS) bufA[i] = (uint32_t)i;
-    __syncthreads();
-
-    // LSD over 4-bit digits: v nibbles (minor) then k nibbles (major).
-    // Stable per pass via per-(digit,thread) rank counters: counter cell
-    // cnt[d*t_act + tid] is exclusive to its thread, so count, scan the
-    // digit-major flat array block-wide, then post-increment to rank.
-    uint32_t *src = bufA, *dst = bufB;
     const int total_nibs = vnibs + knibs;
-    for (int pass = 0; pass < total_nibs; pass++) {
-        const bool is_v = pass < vnibs;
-        const int sh = 4 * (is_v ? pass : pass - vnibs);
-        if (tid < t_act) {
-            // zero own counter column (no barrier: cells are per-thread)
-            for (int d = 0; d < FUSE_DIGITS; d++) cnt[d * t_act + tid] = 0;
-            // annotate digit + count (pads: digit 15, stable-after by idx order)
+    const int vbits = 4 * vnibs;
+    const int rounds = (int)((n + FUSE_THREADS - 1) / FUSE_THREADS);  // <= 8
+    const int nwaves = FUSE_THREADS / WAVE;
+
+    // bits [s, s+16) of the concatenated key ((k-kbase) << vbits | (v-vbase))
+    auto key16_at = [&](uint32_t idx, int s) -> uint32_t {
+        const uint64_t vp = vin[idx] - vbase;
+        const uint64_t kp = kin[idx] - kbase;
+        uint64_t bits = 0;
+        if (s < vbits) {
+            bits = vp >> s;
+            const int up = vbits - s;
+            if (up < 64) bits |= kp << up;
+        } else {
+            bits = kp >> (s - vbits);
+        }
+        return (uint32_t)(bits & 0xFFFF);
+    };
+
+    // ---- digit passes in groups of 4 sharing one entry packing ----
+    uint32_t *src = bufA, *dst = bufB;
+    for (int base = 0; base < total_nibs; base += 4) {
+        // (re)pack: entry = next-16-key-bits << 13 | idx  (one global gather)
+        for (int64_t i = tid; i < n; i += FUSE_THREADS) {
+            const uint32_t idx =
+                base == 0 ? (uint32_t)i : (src[i] & 0x1FFFu);
+            src[i] = (key16_at(idx, 4 * base) << 13) | idx;
+        }
+        __syncthreads();
+        const int sub_max = min(4, total_nibs - base);
+        for (int sub = 0; sub < sub_max; sub++) {
+            const int sh = 13 + 4 * sub;
+            const int cells_n = FUSE_DIGITS * rounds * nwaves;
+            for (int c = tid; c < cells_n; c += FUSE_THREADS) cnt[c] = 0;
+            __syncthreads();
+            uint32_t myrank[FUSE_ITEMS];
+            uint32_t mydig[FUSE_ITEMS];
+#pragma unroll
             for (int j = 0; j < FUSE_ITEMS; j++) {
-                int i = tid * FUSE_ITEMS + j;
-                uint32_t idx = src[i] & 0x1FFFu;
-                uint32_t d = 15;
-                if ((int64_t)idx < n) {
-                    uint64_t limb = is_v ? vin[idx] - vbase : kin[idx] - kbase;
-                    d = (uint32_t)((limb >> sh) & 0xF);
+                if (j >= rounds) break;
+                const int64_t i = (int64_t)j * FUSE_THREADS + tid;
+                const bool act = i < n;
+                const uint32_t e = act ? src[i] : 0;
+                const uint32_t d = (e >> sh) & 0xF;
+                const uint64_t am = __ballot(act);
+                const uint64_t b0 = __ballot(d & 1);
+                const uint64_t b1 = __ballot(d & 2);
+                const uint64_t b2 = __ballot(d & 4);
+                const uint64_t b3 = __ballot(d & 8);
+                uint64_t m = am;
+                m &= (d & 1) ? b0 : ~b0;
+                m &= (d & 2) ? b1 : ~b1;
+                m &= (d & 4) ? b2 : ~b2;
+                m &= (d & 8) ? b3 : ~b3;
+                myrank[j] = (uint32_t)__popcll(m & lt);
+                mydig[j] = d;
+                if (act && (m & lt) == 0)  // lowest lane of this digit group
+                    cnt[(d * rounds + j) * nwaves + wid] =
+                        (uint32_t)__popcll(m);
+                if (!act) myrank[j] = 0xFFFFFFFFu;
+            }
+            __syncthreads();
+            {   // exclusive scan of the (digit, round, wave) cells
+                uint32_t local[2];
+                const int cpt = (cells_n + FUSE_THREADS - 1) / FUSE_THREADS;
+                uint32_t tsum = 0;
+                for (int j = 0; j < cpt; j++) {
+                    const int f = tid * cpt + j;
+                    local[j] = f < cells_n ? cnt[f] : 0;
+                    tsum += local[j];
                 }
-                src[i] = (d << 13) | idx;
-                cnt[d * t_act + tid]++;
+                uint32_t total_unused;
+                uint32_t off = fuse_scan(tsum, wave_tot, &total_unused);
+                for (int j = 0; j < cpt; j++) {
+                    const int f = tid * cpt + j;
+                    if (f < cells_n) cnt[f] = off;
+                    off += local[j];
+                }
             }
-        }
-        __syncthreads();
-        // exclusive scan of the flat counter array
-        {
-            uint32_t local[FUSE_DIGITS];
-            uint32_t tsum = 0;
-            for (int j = 0; j < cells_per_thread; j++) {
-                int f = tid * cells_per_thread + j;
-                local[j] = f < scan_cells ? cnt[f] : 0;
-                tsum += local[j];
-            }
-            uint32_t total_unused;
-            uint32_t off = fuse_scan(tsum, wave_tot, &total_unused);
-            for (int j = 0; j < cells_per_thread; j++) {
-                int f = tid * cells_per_thread + j;
-                if (f < scan_cells) cnt[f] = off;
-                off += local[j];
-            }
-        }
-        __syncthreads();
-        // rank (post-increment own cells in item order) + scatter
-        if (tid < t_act) {
+            __syncthreads();
+#pragma unroll
             for (int j = 0; j < FUSE_ITEMS; j++) {
-                int i = tid * FUSE_ITEMS + j;
-                uint32_t e = src[i];
-                uint32_t d = e >> 13;
-                uint32_t r = cnt[d * t_act + tid]++;
-                dst[r] = e;
+                if (j >= rounds) break;
+                const int64_t i = (int64_t)j * FUSE_THREADS + tid;
+                if (i < n) {
+                    const uint32_t r =
+                        cnt[(mydig[j] * rounds + j) * nwaves + wid] +
+                        myrank[j];
+                    dst[r] = src[i];
+                }
             }
+            __syncthreads();
+            uint32_t *t = src; src = dst; dst = t;
         }
-        __syncthreads();
-        uint32_t *t = src; src = dst; dst = t;
     }
 
-    // ---- consolidate: head flags over sorted permutation ----
-    // seg index via scan; weights accumulated into tw (global scratch)
-    uint32_t head[FUSE_ITEMS];
-    uint32_t tsum = 0;
-    for (int j = 0; j < FUSE_ITEMS; j++) {
-        int i = tid * FUSE_ITEMS + j;
-        uint32_t h = 0;
-        if (i < n) {
-            uint32_t idx = src[i] & 0x1FFFu;
-            if (i == 0) h = 1;
-            else {
-                uint32_t pidx = src[i - 1] & 0x1FFFu;
-                h = (kin[idx] != kin[pidx]) || (vin[idx] != vin[pidx]);
+    if (total_nibs == 0) {  // all rows share one (k,v): identity entries
+        for (int64_t i = tid; i < n; i += FUSE_THREADS) src[i] = (uint32_t)i;
+        __syncthreads();
+    }
+
+    // ---- consolidate: head flags (full-key compares via one gather) ----
+    // seg ids into dst (the free buffer); ballot cells per (round, wave)
+    uint32_t headmask[FUSE_ITEMS];  // per round: this wave's head ballot
+    uint32_t myhead[FUSE_ITEMS];
+    {
+        const int cells_n = rounds * nwaves;
+        for (int c = tid; c < cells_n; c += FUSE_THREADS) cnt[c] = 0;
+        __syncthreads();
+#pragma unroll
+        for (int j = 0; j < FUSE_ITEMS; j++) {
+            if (j >= rounds) break;
+            const int64_t i = (int64_t)j * FUSE_THREADS + tid;
+            uint32_t h = 0;
+            if (i < n) {
+                const uint32_t idx = src[i] & 0x1FFFu;
+                if (i == 0) h = 1;
+                else {
+                    const uint32_t pidx = src[i - 1] & 0x1FFFu;
+                    h = (kin[idx] != kin[pidx]) || (vin[idx] != vin[pidx]);
+                }
             }
+            myhead[j] = h;
+            const uint64_t m = __ballot(h != 0);
+            headmask[j] = (uint32_t)__popcll(m & lt);  // rank among heads
+            if (i < n && lane == 0)
+                cnt[j * nwaves + wid] = (uint32_t)__popcll(m);
         }
-        head[j] = h;
-        tsum += h;
+        __syncthreads();
     }
     uint32_t nseg;
-    uint32_t heads_before = fuse_scan(tsum, wave_tot, &nseg);
-    // store seg ids in dst (reuse) and zero the weight scratch
-    {
-        uint32_t hb = heads_before;
-        for (int j = 0; j < FUSE_ITEMS; j++) {
-            int i = tid * FUSE_ITEMS + j;
-            hb += head[j];
-            if (i < n) dst[i] = hb - 1;  // segment of element i
+    {   // scan head cells
+        const int cells_n = rounds * nwaves;
+        uint32_t local[1];
+        uint32_t tsum = 0;
+        if (tid < cells_n) {
+            local[0] = cnt[tid];
+            tsum = local[0];
         }
+        uint32_t off = fuse_scan(tsum, wave_tot, &nseg);
+        if (tid < cells_n) cnt[tid] = off;
+        __syncthreads();
     }
+    // seg of row i = heads at-or-before i, minus 1
+#pragma unroll
+    for (int j = 0; j < FUSE_ITEMS; j++) {
+        if (j >= rounds) break;
+        const int64_t i = (int64_t)j * FUSE_THREADS + tid;
+        if (i < n)
+            dst[i] = cnt[j * nwaves + wid] + headmask[j] + myhead[j] - 1;
+    }
+    __syncthreads();
     if constexpr (sizeof(W) == 8 && (W)0.5 == (W)0) {  // integer weights
         for (uint32_t i = tid; i < nseg; i += FUSE_THREADS) tw[i] = 0;
         __syncthreads();
+#pragma unroll
         for (int j = 0; j < FUSE_ITEMS; j++) {
-            int i = tid * FUSE_ITEMS + j;
+            if (j >= rounds) break;
+            const int64_t i = (int64_t)j * FUSE_THREADS + tid;
             if (i < n) {
-                uint32_t idx = src[i] & 0x1FFFu;
-                uint32_t seg = dst[i];
+                const uint32_t idx = src[i] & 0x1FFFu;
+                const uint32_t seg = dst[i];
                 atomicAdd((unsigned long long *)&tw[seg],
                           (unsigned long long)win[idx]);
-                if (head[j]) {
+                if (myhead[j]) {
                     tk[seg] = kin[idx];
                     tv[seg] = vin[idx];
                 }
@@ -657,19 +728,20 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
         }
         __syncthreads();
     } else {
-        // f64: deterministic segmented tree sum in sorted-position order.
-        // Stage weights in sorted order (tw[i] = w of sorted element i), record
-        // each segment's head position (cnt LDS is free after sorting), then
-        // combine (j % 2d == 0) pairs within segments — fixed reduction order,
-        // |err| <= 2 ulp * ceil(log2(run_len)) vs a sequential sum.
+        // f64: deterministic segmented tree sum in sorted-position order
+        // (identical order to the stable sort's positions, so results are
+        // bit-equal to the previous per-thread-counter kernel).  cnt (8192
+        // cells) now holds each segment's head position.
+#pragma unroll
         for (int j = 0; j < FUSE_ITEMS; j++) {
-            int i = tid * FUSE_ITEMS + j;
+            if (j >= rounds) break;
+            const int64_t i = (int64_t)j * FUSE_THREADS + tid;
             if (i < n) {
-                uint32_t idx = src[i] & 0x1FFFu;
+                const uint32_t idx = src[i] & 0x1FFFu;
                 tw[i] = win[idx];
-                if (head[j]) {
-                    uint32_t seg = dst[i];
-                    cnt[seg] = (uint32_t)i;  // head position of the segment
+                if (myhead[j]) {
+                    const uint32_t seg = dst[i];
+                    cnt[seg] = (uint32_t)i;
                     tk[seg] = kin[idx];
                     tv[seg] = vin[idx];
                 }
@@ -677,48 +749,74 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
         }
         __syncthreads();
         for (int64_t d = 1; d < n; d <<= 1) {
+#pragma unroll
             for (int j = 0; j < FUSE_ITEMS; j++) {
-                int i = tid * FUSE_ITEMS + j;
-                if (i + d < n && dst[i] == dst[i + d]) {
-                    uint32_t rel = (uint32_t)i - cnt[dst[i]];
+                if (j >= rounds) break;
+                const int64_t i = (int64_t)j * FUSE_THREADS + tid;
+                if (i < n && i + d < n && dst[i] == dst[i + d]) {
+                    const uint32_t rel = (uint32_t)i - cnt[dst[i]];
                     if ((rel & (2 * d - 1)) == 0) tw[i] += tw[i + d];
                 }
             }
             __syncthreads();
         }
-        // move each segment's total (at its head position) to tw[seg]
         W my_tot[FUSE_ITEMS];
+#pragma unroll
         for (int j = 0; j < FUSE_ITEMS; j++) {
-            int i = tid * FUSE_ITEMS + j;
-            my_tot[j] = (i < n && head[j]) ? tw[i] : (W)0;
+            if (j >= rounds) break;
+            const int64_t i = (int64_t)j * FUSE_THREADS + tid;
+            my_tot[j] = (i < n && myhead[j]) ? tw[i] : (W)0;
         }
         __syncthreads();
+#pragma unroll
         for (int j = 0; j < FUSE_ITEMS; j++) {
-            int i = tid * FUSE_ITEMS + j;
-            if (i < n && head[j]) tw[dst[i]] = my_tot[j];
+            if (j >= rounds) break;
+            const int64_t i = (int64_t)j * FUSE_THREADS + tid;
+            if (i < n && myhead[j]) tw[dst[i]] = my_tot[j];
         }
         __syncthreads();
     }
-    // ---- drop zero-weight segments (compact into output) ----
-    uint32_t nz[FUSE_ITEMS];
-    uint32_t zsum = 0;
-    for (int j = 0; j < FUSE_ITEMS; j++) {
-        int i = tid * FUSE_ITEMS + j;
-        nz[j] = (i < (int)nseg && tw[i] != (W)0) ? 1 : 0;
-        zsum += nz[j];
-    }
-    uint32_t nout;
-    uint32_t before = fuse_scan(zsum, wave_tot, &nout);
-    for (int j = 0; j < FUSE_ITEMS; j++) {
-        int i = tid * FUSE_ITEMS + j;
-        if (nz[j]) {
-            uint32_t p = before++;
-            ok[p] = tk[i];
-            ov[p] = tv[i];
-            ow[p] = tw[i];
+    // ---- drop zero-weight segments (ballot compaction over seg ids) ----
+    {
+        const int seg_rounds = (int)((nseg + FUSE_THREADS - 1) / FUSE_THREADS);
+        const int cells_n = seg_rounds * nwaves;
+        for (int c = tid; c < cells_n + 1; c += FUSE_THREADS) cnt[c] = 0;
+        __syncthreads();
+        uint32_t nzrank[FUSE_ITEMS];
+        uint32_t mynz[FUSE_ITEMS];
+#pragma unroll
+        for (int j = 0; j < FUSE_ITEMS; j++) {
+            if (j >= seg_rounds) break;
+            const uint32_t i = (uint32_t)j * FUSE_THREADS + tid;
+            const uint32_t nz = (i < nseg && tw[i] != (W)0) ? 1 : 0;
+            mynz[j] = nz;
+            const uint64_t m = __ballot(nz != 0);
+            nzrank[j] = (uint32_t)__popcll(m & lt);
+            if (i < nseg && lane == 0)
+                cnt[j * nwaves + wid] = (uint32_t)__popcll(m);
         }
+        __syncthreads();
+        uint32_t nout;
+        {
+            uint32_t local = 0;
+            if (tid < cells_n) local = cnt[tid];
+            uint32_t off = fuse_scan(local, wave_tot, &nout);
+            if (tid < cells_n) cnt[tid] = off;
+            __syncthreads();
+        }
+#pragma unroll
+        for (int j = 0; j < FUSE_ITEMS; j++) {
+            if (j >= seg_rounds) break;
+            const uint32_t i = (uint32_t)j * FUSE_THREADS + tid;
+            if (i < nseg && mynz[j]) {
+                const uint32_t p = cnt[j * nwaves + wid] + nzrank[j];
+                ok[p] = tk[i];
+                ov[p] = tv[i];
+                ow[p] = tw[i];
+            }
+        }
+        if (tid == 0) *out_len = (int64_t)nout;
     }
-    if (tid == 0) *out_len = (int64_t)nout;
 }
 
 
