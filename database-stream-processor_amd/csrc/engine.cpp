@@ -938,6 +938,11 @@ struct dbsp_engine {
     // path allocates nothing for it (output_is_store marks when output
     // aliases it and must not be freed)
     DevBatch output;
+    // chained-tick speculation control: big ticks whose deltas always
+    // overflow the fused sort should not pay the speculative launch + redo
+    // every tick — three consecutive losses switch the engine to the
+    // explicit path until a chained tick wins again
+    int spec_fail = 0;
     DevBatch out_store;
     int64_t out_cap = 0;
     bool output_is_store = false;
@@ -1590,7 +1595,8 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     // detects the -1 sentinels and the tick re-sorts through the sized
     // paths.  Sharded ranks keep the explicit path (the exchange needs host
     // lengths).
-    const bool chain = !sharding_on(c);
+    const bool chain =
+        !sharding_on(c) && n <= 131072 && e->spec_fail < 3;
     DevBatch dA, dP, rawA, rawP;
     if (chain) {
         TRY(build_deltas_chain(e, d_ev, n, rawA, rawP, dA, dP));
@@ -1700,8 +1706,10 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             if (!lost) {
                 dA.n = c->h_len[10];
                 dP.n = c->h_len[11];
+                e->spec_fail = 0;
                 break;
             }
+            e->spec_fail++;
             // speculation lost: re-sort from the raw flatmap outputs with the
             // now-known lengths, then redo the planning explicitly
             rawA.n = c->h_len[8];
@@ -1818,7 +1826,8 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     // ranges with ONE sync; the watermark lives in e->d_wm so the bounds
     // never round-trip to the host on the fast path.  Sharded ranks keep the
     // explicit path (the watermark allreduce and exchange need host values).
-    const bool chain = !sharding_on(c) && e->d_wm;
+    const bool chain = !sharding_on(c) && e->d_wm && n <= 131072 &&
+                       e->spec_fail < 3;
     DevBatch dPT, dAT, rawP, rawA;
     std::vector<DevBatch> wp_raw, wa_raw;
     bool front_done = false;
@@ -1873,7 +1882,9 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                 wa_raw.push_back(o);
             }
             front_done = true;
+            e->spec_fail = 0;
         } else {
+            e->spec_fail++;
             // speculation lost (delta overflowed the fused sort, or no arena
             // for the tables): re-sort with real lengths, pull the watermark
             // state to the host, run the explicit window path, write back
