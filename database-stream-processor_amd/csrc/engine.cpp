@@ -32,6 +32,7 @@
 #include <cstdio>
 #include <cstring>
 #include <vector>
+#include <functional>
 
 #include "../../include/dbsp_hip.h"
 #include "kernels_iface.hpp"
@@ -82,12 +83,19 @@ struct dbsp_ctx {
     uint8_t *arena = nullptr;
     size_t arena_sz = 0;
     size_t arena_off = 0;
+    // ping-pong halves: a pipelined tick launches the NEXT tick's front half
+    // while this tick finishes, so two ticks' transients are live at once
+    size_t arena_base = 0;
+    size_t arena_half = 0;  // half size (arena_sz / 2); 0 = no split
+    hipEvent_t ev_sync = nullptr;  // tick-end event (pre-front-launch point)
 };
 
 static void *arena_alloc(dbsp_ctx *c, size_t bytes) {
     bytes = (bytes + 255) & ~(size_t)255;
-    if (!c->arena || c->arena_off + bytes > c->arena_sz) return nullptr;
-    void *p = c->arena + c->arena_off;
+    size_t limit = c->arena_half ? c->arena_base + c->arena_half : c->arena_sz;
+    if (!c->arena || c->arena_base + c->arena_off + bytes > limit)
+        return nullptr;
+    void *p = c->arena + c->arena_base + c->arena_off;
     c->arena_off += bytes;
     return p;
 }
@@ -126,12 +134,14 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     HIP_CHECK_ST(hipStreamCreate(&c->stream));
     HIP_CHECK_ST(hipEventCreate(&c->ev0));
     HIP_CHECK_ST(hipEventCreate(&c->ev1));
+    HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_sync, hipEventDisableTiming));
     const char *p = getenv("DBSP_PROFILE");
     c->profile = p && p[0] == '1';
     const char *fs = getenv("DBSP_FORCE_SHARD");
     c->force_shard = fs && fs[0] == '1';
     HIP_CHECK_ST(hipMalloc(&c->d_len, 16 * sizeof(int64_t)));
     c->arena_sz = (size_t)512 << 20;
+    c->arena_half = c->arena_sz / 2;
     if (hipMalloc(&c->arena, c->arena_sz) != hipSuccess) {
         c->arena = nullptr;
         c->arena_sz = 0;
@@ -143,6 +153,7 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
 
 extern "C" dbsp_status dbsp_ctx_destroy(dbsp_ctx *c) {
     if (!c) return DBSP_OK;
+    if (c->ev_sync) (void)hipEventDestroy(c->ev_sync);
     if (c->comm) ncclCommDestroy(c->comm);
     (void)hipStreamSynchronize(c->stream);
     if (c->d_len) (void)hipFree(c->d_len);
@@ -938,6 +949,20 @@ struct dbsp_engine {
     // path allocates nothing for it (output_is_store marks when output
     // aliases it and must not be freed)
     DevBatch output;
+    // pipelined front half of the NEXT tick (launched during this tick's
+    // tail via the spines_insert_pair hook, consumed by the next step; the
+    // (ev, n) pair is verified at consumption so an out-of-band step never
+    // uses a stale front)
+    struct {
+        bool pending = false;
+        const dbsp_event *ev = nullptr;
+        int64_t n = -1;
+        DevBatch rawA, rawB, oA, oB;
+        size_t arena_base = 0, arena_off = 0;
+    } front;
+    const dbsp_event *next_ev = nullptr;  // set by dbsp_engine_run_staged
+    int64_t next_n = -1;
+
     // chained-tick speculation control: big ticks whose deltas always
     // overflow the fused sort should not pay the speculative launch + redo
     // every tick — three consecutive losses switch the engine to the
@@ -985,6 +1010,10 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
     free_batch(c, e->maxout_int);
     free_batch(c, e->maxz_int);
     engine_free_output(e);
+    if (e->front.pending) {
+        free_batch(c, e->front.oA);
+        free_batch(c, e->front.oB);
+    }
     if (e->out_store.k) free_batch(c, e->out_store);
     if (e->d_events) (void)dbspk::cache_free(e->d_events, c->stream);
     (void)hipStreamSynchronize(c->stream);
@@ -1499,9 +1528,17 @@ static bool spine_needs_merge(Spine &s) {
 
 // insert one delta into each of two spines, batching the two sides' pending
 // merges into single launches with one shared length sync per round
+// `hook` (optional) is the tick-pipelining point: it is invoked exactly once,
+// after this tick's remaining work and length copies are queued and an event
+// recorded, and the first wait is hipEventSynchronize on that event — so
+// whatever the hook launches (the NEXT tick's front half) overlaps the
+// host-side tail of this tick instead of extending its sync.
 static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
-                                      Spine &s2, DevBatch b2) {
+                                      Spine &s2, DevBatch b2,
+                                      const std::function<dbsp_status()> *hook
+                                      = nullptr) {
     ScopedTimer t(c, 1, (double)(b1.n + b2.n) * 48.0);
+    bool hook_fired = false;
     if (b1.n > 0) s1.batches.push_back(b1);
     else free_batch(c, b1);
     if (b2.n > 0) s2.batches.push_back(b2);
@@ -1549,7 +1586,14 @@ static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
             HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
                                         ma.np * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
-            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            if (hook && !hook_fired) {
+                (void)hipEventRecord(c->ev_sync, c->stream);
+                TRY((*hook)());
+                hook_fired = true;
+                (void)hipEventSynchronize(c->ev_sync);
+            } else {
+                HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            }
             for (int j = 0; j < nbatched; j++) {
                 Spine &sp = *owners[j];
                 results[j].n = c->h_len[j];
@@ -1563,6 +1607,11 @@ static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
                 else free_batch(c, results[j]);
             }
         }
+    }
+    if (hook && !hook_fired) {
+        (void)hipEventRecord(c->ev_sync, c->stream);
+        TRY((*hook)());
+        (void)hipEventSynchronize(c->ev_sync);
     }
     return DBSP_OK;
 }
@@ -1595,10 +1644,24 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     // detects the -1 sentinels and the tick re-sorts through the sized
     // paths.  Sharded ranks keep the explicit path (the exchange needs host
     // lengths).
+    bool use_front = e->front.pending && e->front.ev == d_ev &&
+                     e->front.n == n;
+    if (e->front.pending && !use_front) {  // stale pipelined front: discard
+        free_batch(c, e->front.oA);
+        free_batch(c, e->front.oB);
+        e->front.pending = false;
+    }
     const bool chain =
-        !sharding_on(c) && n <= 131072 && e->spec_fail < 3;
+        use_front ||
+        (!sharding_on(c) && n <= 131072 && e->spec_fail < 3);
     DevBatch dA, dP, rawA, rawP;
-    if (chain) {
+    if (use_front) {
+        e->front.pending = false;
+        rawA = e->front.rawA;
+        rawP = e->front.rawB;
+        dA = e->front.oA;
+        dP = e->front.oB;
+    } else if (chain) {
         TRY(build_deltas_chain(e, d_ev, n, rawA, rawP, dA, dP));
     } else {
         TRY(build_deltas(e, d_ev, n, dA, dP, true));
@@ -1804,16 +1867,57 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         TRY(dbspk::sort_cons_small_batch(c->stream, sa));
         // no sync yet — the spine inserts below sync the stream
     }
-    TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
-    if (async_final) {
-        HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6, sizeof(int64_t),
-                                    hipMemcpyDeviceToHost, c->stream));
-        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
-        res.n = c->h_len[6];
-        e->output = res;
-        e->output_is_store = true;
+    if (!sharding_on(c)) {
+        // queue the output length readback now (stream-ordered after the
+        // output sort) so the insert hook's event covers it, then let the
+        // inserts' first wait double as the pipelining point
+        if (async_final)
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6,
+                                        sizeof(int64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
+        std::function<dbsp_status()> hook = [&]() -> dbsp_status {
+            if (!e->next_ev || e->next_n < 0 || e->next_n > 131072 ||
+                e->spec_fail >= 3)
+                return DBSP_OK;
+            const size_t save_base = c->arena_base, save_off = c->arena_off;
+            c->arena_base =
+                c->arena_half ? (save_base ? 0 : c->arena_half) : save_base;
+            c->arena_off = 0;
+            dbsp_status st = build_deltas_chain(e, e->next_ev, e->next_n,
+                                                e->front.rawA, e->front.rawB,
+                                                e->front.oA, e->front.oB);
+            e->front.arena_base = c->arena_base;
+            e->front.arena_off = c->arena_off;
+            c->arena_base = save_base;
+            c->arena_off = save_off;
+            if (st == DBSP_OK) {
+                e->front.pending = true;
+                e->front.ev = e->next_ev;
+                e->front.n = e->next_n;
+            }
+            return st;
+        };
+        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
+        if (async_final) {
+            res.n = c->h_len[6];
+            e->output = res;
+            e->output_is_store = true;
+        } else {
+            TRY(finalize_raw(c, outs, e->output));
+        }
     } else {
-        TRY(finalize_raw(c, outs, e->output));
+        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
+        if (async_final) {
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6,
+                                        sizeof(int64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            res.n = c->h_len[6];
+            e->output = res;
+            e->output_is_store = true;
+        } else {
+            TRY(finalize_raw(c, outs, e->output));
+        }
     }
     return DBSP_OK;
 }
@@ -2182,7 +2286,18 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
 extern "C" dbsp_status dbsp_engine_step_staged(dbsp_engine *e, int64_t lo,
                                                int64_t hi) {
     if (lo < 0 || hi > e->n_events || lo > hi) return DBSP_ERR_INVALID;
-    e->ctx->arena_off = 0;
+    {
+        dbsp_ctx *cc = e->ctx;
+        if (e->front.pending) {
+            // continue in the half the pipelined front allocated from
+            cc->arena_base = e->front.arena_base;
+            cc->arena_off = e->front.arena_off;
+        } else {
+            cc->arena_base =
+                cc->arena_half ? (cc->arena_base ? 0 : cc->arena_half) : 0;
+            cc->arena_off = 0;
+        }
+    }
     if (e->query == 0) {
         q0_step_host(e, e->h_events.data() + lo, hi - lo);
         return DBSP_OK;
@@ -2202,10 +2317,20 @@ extern "C" dbsp_status dbsp_engine_run_staged(dbsp_engine *e, int64_t lo,
     // per-tick host cost excludes the Python/ctypes round-trip
     if (tick <= 0) return DBSP_ERR_INVALID;
     for (int64_t t = lo; t < hi; t += tick) {
-        dbsp_status st =
-            dbsp_engine_step_staged(e, t, std::min(hi, t + tick));
+        const int64_t t_hi = std::min(hi, t + tick);
+        const int64_t n_lo = t_hi, n_hi = std::min(hi, t_hi + tick);
+        if (n_lo < hi) {
+            e->next_ev = e->d_events + n_lo;
+            e->next_n = n_hi - n_lo;
+        } else {
+            e->next_ev = nullptr;
+            e->next_n = -1;
+        }
+        dbsp_status st = dbsp_engine_step_staged(e, t, t_hi);
         if (st != DBSP_OK) return st;
     }
+    e->next_ev = nullptr;
+    e->next_n = -1;
     return DBSP_OK;
 }
 
