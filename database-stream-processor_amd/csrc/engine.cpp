@@ -1930,13 +1930,29 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     // ranges with ONE sync; the watermark lives in e->d_wm so the bounds
     // never round-trip to the host on the fast path.  Sharded ranks keep the
     // explicit path (the watermark allreduce and exchange need host values).
-    const bool chain = !sharding_on(c) && e->d_wm && n <= 131072 &&
-                       e->spec_fail < 3;
+    bool use_front = e->front.pending && e->front.ev == d_ev &&
+                     e->front.n == n;
+    if (e->front.pending && !use_front) {  // stale pipelined front: discard
+        free_batch(c, e->front.oA);
+        free_batch(c, e->front.oB);
+        e->front.pending = false;
+    }
+    const bool chain = use_front ||
+                       (!sharding_on(c) && e->d_wm && n <= 131072 &&
+                        e->spec_fail < 3);
     DevBatch dPT, dAT, rawP, rawA;
     std::vector<DevBatch> wp_raw, wa_raw;
     bool front_done = false;
     if (chain) {
-        TRY(build_deltas_chain(e, d_ev, n, rawP, rawA, dPT, dAT));
+        if (use_front) {
+            e->front.pending = false;
+            rawP = e->front.rawA;
+            rawA = e->front.rawB;
+            dPT = e->front.oA;
+            dAT = e->front.oB;
+        } else {
+            TRY(build_deltas_chain(e, d_ev, n, rawP, rawA, dPT, dAT));
+        }
         TRY(dbspk::wm_update(c->stream, dAT.k, c->d_len + 11, TUMBLE_MS,
                              TUMBLE_MS, TUMBLE_MS, e->d_wm, e->d_bounds));
         if ((int)e->pt_int.batches.size() > MAX_TRACE_BATCHES)
@@ -2078,9 +2094,88 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         if (o.n > 0) outs.push_back(o);
         else free_batch(c, o);
     }
-    TRY(spines_insert_pair(c, e->wp_int, dWP, e->wa_int, dWA));
     engine_free_output(e);
-    TRY(finalize_raw(c, outs, e->output));
+    // consolidate the joined output asynchronously (as q3): sort launched to
+    // the reused store before the inserts, length read at their sync
+    int64_t cat_n = 0;
+    for (auto &b : outs) cat_n += b.n;
+    const bool async_final = !sharding_on(c) && cat_n > 0 && cat_n <= 8192;
+    DevBatch res;
+    if (async_final) {
+        ScopedTimer t0(c, 0, (double)cat_n * 48.0);
+        DevBatch cat, scratch;
+        if (outs.size() == 1) {
+            cat = outs[0];
+            outs.clear();
+        } else {
+            TRY(alloc_batch(c, cat_n, cat, true));
+            int64_t off = 0;
+            for (auto &b : outs) {
+                if (b.n == 0) continue;
+                HIP_CHECK_ST(hipMemcpyAsync(cat.k + off, b.k, b.n * 8,
+                                            hipMemcpyDeviceToDevice, c->stream));
+                HIP_CHECK_ST(hipMemcpyAsync(cat.v + off, b.v, b.n * 8,
+                                            hipMemcpyDeviceToDevice, c->stream));
+                HIP_CHECK_ST(hipMemcpyAsync(cat.w + off, b.w, b.n * 8,
+                                            hipMemcpyDeviceToDevice, c->stream));
+                off += b.n;
+            }
+            for (auto &b : outs) free_batch(c, b);
+            outs.clear();
+        }
+        TRY(alloc_batch(c, cat_n, scratch, true));
+        if (e->out_cap < cat_n) {
+            if (e->out_store.k) free_batch(c, e->out_store);
+            e->out_cap = std::max<int64_t>(2 * cat_n, 8192);
+            TRY(alloc_batch(c, e->out_cap, e->out_store));
+        }
+        res = e->out_store;
+        SortArgs sa{};
+        sa.nb = 1;
+        sa.kin[0] = cat.k; sa.vin[0] = cat.v; sa.win[0] = cat.w;
+        sa.n[0] = cat_n;
+        sa.tk[0] = scratch.k; sa.tv[0] = scratch.v; sa.tw[0] = scratch.w;
+        sa.ok[0] = res.k; sa.ov[0] = res.v; sa.ow[0] = res.w;
+        sa.d_len = c->d_len + 6;
+        TRY(dbspk::sort_cons_small_batch(c->stream, sa));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6,
+                                    sizeof(int64_t), hipMemcpyDeviceToHost,
+                                    c->stream));
+    }
+    if (!sharding_on(c)) {
+        std::function<dbsp_status()> hook = [&]() -> dbsp_status {
+            if (!e->next_ev || e->next_n < 0 || e->next_n > 131072 ||
+                e->spec_fail >= 3 || !e->d_wm)
+                return DBSP_OK;
+            const size_t save_base = c->arena_base, save_off = c->arena_off;
+            c->arena_base =
+                c->arena_half ? (save_base ? 0 : c->arena_half) : save_base;
+            c->arena_off = 0;
+            dbsp_status st = build_deltas_chain(e, e->next_ev, e->next_n,
+                                                e->front.rawA, e->front.rawB,
+                                                e->front.oA, e->front.oB);
+            e->front.arena_base = c->arena_base;
+            e->front.arena_off = c->arena_off;
+            c->arena_base = save_base;
+            c->arena_off = save_off;
+            if (st == DBSP_OK) {
+                e->front.pending = true;
+                e->front.ev = e->next_ev;
+                e->front.n = e->next_n;
+            }
+            return st;
+        };
+        TRY(spines_insert_pair(c, e->wp_int, dWP, e->wa_int, dWA, &hook));
+    } else {
+        TRY(spines_insert_pair(c, e->wp_int, dWP, e->wa_int, dWA));
+    }
+    if (async_final) {
+        res.n = c->h_len[6];
+        e->output = res;
+        e->output_is_store = true;
+    } else {
+        TRY(finalize_raw(c, outs, e->output));
+    }
     return DBSP_OK;
 }
 
@@ -2095,17 +2190,29 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     DevBatch dBT;
     std::vector<DevBatch> wb_raw;
     bool front_done = false;
+    bool use_front = e->front.pending && e->front.ev == d_ev &&
+                     e->front.n == n;
+    if (e->front.pending && !use_front) {  // stale pipelined front: discard
+        e->front.pending = false;  // q5 fronts hold only arena transients
+    }
     if (chain) {
         const int64_t cap = n > 0 ? n : 1;
         DevBatch raw0, raw1;
-        TRY(alloc_batch(c, cap, raw0, true));
-        TRY(alloc_batch(c, cap, raw1, true));
-        TRY(dbspk::flatmap_events_chain(c->stream, d_ev, n, e->query, raw0.k,
-                                        raw0.v, raw0.w, raw1.k, raw1.v, raw1.w,
-                                        (uint64_t *)(c->d_len + 8)));
-        TRY(dbspk::minmax_rows_chain(c->stream, raw0.k, raw0.v, cap,
-                                     c->d_len + 8,
-                                     (unsigned long long *)(c->d_len + 12)));
+        if (use_front) {
+            e->front.pending = false;
+            raw0 = e->front.rawA;
+            raw1 = e->front.rawB;
+        } else {
+            TRY(alloc_batch(c, cap, raw0, true));
+            TRY(alloc_batch(c, cap, raw1, true));
+            TRY(dbspk::flatmap_events_chain(c->stream, d_ev, n, e->query,
+                                            raw0.k, raw0.v, raw0.w, raw1.k,
+                                            raw1.v, raw1.w,
+                                            (uint64_t *)(c->d_len + 8)));
+            TRY(dbspk::minmax_rows_chain(c->stream, raw0.k, raw0.v, cap,
+                                         c->d_len + 8,
+                                         (unsigned long long *)(c->d_len + 12)));
+        }
         HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 16 * sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -2276,10 +2383,98 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         if (o.n > 0) outs.push_back(o);
         else free_batch(c, o);
     }
-    TRY(e->bc_int.insert(c, dBC));
-    TRY(e->counts_int.insert(c, dCounts));
     engine_free_output(e);
-    TRY(finalize_raw(c, outs, e->output));
+    int64_t cat_n = 0;
+    for (auto &b : outs) cat_n += b.n;
+    const bool async_final = !sharding_on(c) && cat_n > 0 && cat_n <= 8192;
+    DevBatch res;
+    if (async_final) {
+        ScopedTimer t0(c, 0, (double)cat_n * 48.0);
+        DevBatch cat, scratch;
+        if (outs.size() == 1) {
+            cat = outs[0];
+            outs.clear();
+        } else {
+            TRY(alloc_batch(c, cat_n, cat, true));
+            int64_t off = 0;
+            for (auto &b : outs) {
+                if (b.n == 0) continue;
+                HIP_CHECK_ST(hipMemcpyAsync(cat.k + off, b.k, b.n * 8,
+                                            hipMemcpyDeviceToDevice, c->stream));
+                HIP_CHECK_ST(hipMemcpyAsync(cat.v + off, b.v, b.n * 8,
+                                            hipMemcpyDeviceToDevice, c->stream));
+                HIP_CHECK_ST(hipMemcpyAsync(cat.w + off, b.w, b.n * 8,
+                                            hipMemcpyDeviceToDevice, c->stream));
+                off += b.n;
+            }
+            for (auto &b : outs) free_batch(c, b);
+            outs.clear();
+        }
+        TRY(alloc_batch(c, cat_n, scratch, true));
+        if (e->out_cap < cat_n) {
+            if (e->out_store.k) free_batch(c, e->out_store);
+            e->out_cap = std::max<int64_t>(2 * cat_n, 8192);
+            TRY(alloc_batch(c, e->out_cap, e->out_store));
+        }
+        res = e->out_store;
+        SortArgs sa{};
+        sa.nb = 1;
+        sa.kin[0] = cat.k; sa.vin[0] = cat.v; sa.win[0] = cat.w;
+        sa.n[0] = cat_n;
+        sa.tk[0] = scratch.k; sa.tv[0] = scratch.v; sa.tw[0] = scratch.w;
+        sa.ok[0] = res.k; sa.ov[0] = res.v; sa.ow[0] = res.w;
+        sa.d_len = c->d_len + 6;
+        TRY(dbspk::sort_cons_small_batch(c->stream, sa));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6,
+                                    sizeof(int64_t), hipMemcpyDeviceToHost,
+                                    c->stream));
+    }
+    if (!sharding_on(c)) {
+        std::function<dbsp_status()> hook = [&]() -> dbsp_status {
+            if (!e->next_ev || e->next_n < 0 || !e->d_wm) return DBSP_OK;
+            const size_t save_base = c->arena_base, save_off = c->arena_off;
+            c->arena_base =
+                c->arena_half ? (save_base ? 0 : c->arena_half) : save_base;
+            c->arena_off = 0;
+            const int64_t ncap = e->next_n > 0 ? e->next_n : 1;
+            dbsp_status st = alloc_batch(c, ncap, e->front.rawA, true);
+            if (st == DBSP_OK) st = alloc_batch(c, ncap, e->front.rawB, true);
+            if (st == DBSP_OK)
+                st = dbspk::flatmap_events_chain(
+                    c->stream, e->next_ev, e->next_n, e->query,
+                    e->front.rawA.k, e->front.rawA.v, e->front.rawA.w,
+                    e->front.rawB.k, e->front.rawB.v, e->front.rawB.w,
+                    (uint64_t *)(c->d_len + 8));
+            if (st == DBSP_OK)
+                st = dbspk::minmax_rows_chain(
+                    c->stream, e->front.rawA.k, e->front.rawA.v, ncap,
+                    c->d_len + 8, (unsigned long long *)(c->d_len + 12));
+            e->front.arena_base = c->arena_base;
+            e->front.arena_off = c->arena_off;
+            c->arena_base = save_base;
+            c->arena_off = save_off;
+            if (st == DBSP_OK) {
+                e->front.pending = true;
+                e->front.ev = e->next_ev;
+                e->front.n = e->next_n;
+                e->front.oA = DevBatch{};
+                e->front.oB = DevBatch{};
+            }
+            return st;
+        };
+        TRY(spines_insert_pair(c, e->bc_int, dBC, e->counts_int, dCounts,
+                               &hook));
+    } else {
+        TRY(e->bc_int.insert(c, dBC));
+        TRY(e->counts_int.insert(c, dCounts));
+    }
+    if (async_final) {
+        res.n = c->h_len[6];
+        e->output = res;
+        e->output_is_store = true;
+    } else {
+        TRY(finalize_raw(c, outs, e->output));
+    }
     return DBSP_OK;
 }
 
