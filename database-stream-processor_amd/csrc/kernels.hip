@@ -61,17 +61,17 @@ void cache_drop_locked(hipStream_t s) {
 
 hipError_t cache_malloc(void **out, size_t bytes, hipStream_t s) {
     if (bytes >= CACHE_MIN) {
-        // round carves up to a quarter-power-of-two class so a workload with
-        // slowly growing buffers (spine merges) re-hits earlier blocks
-        // instead of carving a fresh one per size (waste < 25%)
+        // round carves up to the next power of two and accept cached blocks
+        // up to 2x the request: a workload with growing buffers (spine
+        // merges, trace cascades) then carves only log2(max/min) times
+        // instead of once per size, at <= 2x footprint — cheap against
+        // 288 GB of HBM, and the page-mapping stalls dominate otherwise
         size_t msb = (size_t)1 << (63 - __builtin_clzll(bytes));
-        size_t gran = msb >> 2;
-        bytes = (bytes + gran - 1) / gran * gran;
+        if (msb < bytes) bytes = msb << 1;
         std::lock_guard<std::mutex> g(g_cache_mu);
         auto it = g_cache_free.lower_bound(bytes);
-        // accept up to 50% + 64 MB waste; beyond that carve fresh
         if (it != g_cache_free.end() &&
-            it->first <= bytes + bytes / 2 + (64u << 20)) {
+            it->first <= 2 * bytes + (64u << 20)) {
             *out = it->second;
             g_cache_live.emplace(it->second, it->first);
             g_cache_free.erase(it);

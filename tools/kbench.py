@@ -138,7 +138,7 @@ def bench_sort(ctx, L, n, reps=3):
     }
 
 
-def bench_incremental(ctx, L, n_trace, n_delta, steps=8):
+def bench_incremental(ctx, L, n_trace, n_delta, steps=12):
     """C5-class incremental loop (SURVEY.md §8 config 5, i64-weight variant;
     the f64 sum aggregate is a named next step): per step, a 10M-row sorted
     delta joins against the 1B-row trace spine (join is linear: one probe per
@@ -198,7 +198,7 @@ def bench_incremental(ctx, L, n_trace, n_delta, steps=8):
     }
 
 
-def bench_c5(ctx, L, n_trace, n_delta, steps=6):
+def bench_c5(ctx, L, n_trace, n_delta, steps=10):
     """Config C5 end-to-end: 1B-row indexed trace
     (1 val/key, f64 vals) x 10M-row delta incremental join, f64 vals weighed
     (f(k,v)=v) into f64 weights, consolidated deterministically, and
