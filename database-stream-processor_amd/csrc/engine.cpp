@@ -87,7 +87,8 @@ struct dbsp_ctx {
     // while this tick finishes, so two ticks' transients are live at once
     size_t arena_base = 0;
     size_t arena_half = 0;  // half size (arena_sz / 2); 0 = no split
-    hipEvent_t ev_sync = nullptr;  // tick-end event (pre-front-launch point)
+    hipEvent_t ev_sync = nullptr;   // tick-end event (pre-front-launch point)
+    hipEvent_t ev_sync2 = nullptr;  // early-wake event (post-count readback)
 };
 
 static void *arena_alloc(dbsp_ctx *c, size_t bytes) {
@@ -135,6 +136,7 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     HIP_CHECK_ST(hipEventCreate(&c->ev0));
     HIP_CHECK_ST(hipEventCreate(&c->ev1));
     HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_sync, hipEventDisableTiming));
+    HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_sync2, hipEventDisableTiming));
     const char *p = getenv("DBSP_PROFILE");
     c->profile = p && p[0] == '1';
     const char *fs = getenv("DBSP_FORCE_SHARD");
@@ -154,6 +156,7 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
 extern "C" dbsp_status dbsp_ctx_destroy(dbsp_ctx *c) {
     if (!c) return DBSP_OK;
     if (c->ev_sync) (void)hipEventDestroy(c->ev_sync);
+    if (c->ev_sync2) (void)hipEventDestroy(c->ev_sync2);
     if (c->comm) ncclCommDestroy(c->comm);
     (void)hipStreamSynchronize(c->stream);
     if (c->d_len) (void)hipFree(c->d_len);
@@ -1636,14 +1639,23 @@ static TraceArgs trace_args_of(Spine &sp) {
 }
 
 // ---- q3 tick (queries/q3.rs:35-63) ----
+//
+// Single-rank ticks run almost entirely as ONE chained launch train with
+// device-side lengths:
+//   flatmap -> fused delta sorts (speculative, -1 sentinel over 8192 rows)
+//   -> 3-plan join count/scan -> per-plan emit bases -> chained emits into a
+//   capacity buffer -> fused output consolidate into the reused store.
+// The host wakes at an EVENT recorded right after the count totals + delta
+// lengths copy — while the GPU still runs the emits and the output sort —
+// checks the speculation, and immediately queues the spine inserts (whose
+// first wait doubles as the pipelining point launching the NEXT tick's
+// front).  Emit overflow (combined output beyond the capacity buffer) and
+// output overflow (beyond the fused sort) are detected after the inserts
+// from the second readback and replayed explicitly; a lost sort speculation
+// re-sorts through the sized paths.  Sharded ranks keep the explicit path
+// (the exchange needs host lengths).
 static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     dbsp_ctx *c = e->ctx;
-    // Single-rank ticks chain flatmap -> fused sorts -> join count/scan with
-    // DEVICE lengths and pay ONE sync for the whole front half; if a delta
-    // overflows the fused sort (never at the benchmark tick sizes) the sync
-    // detects the -1 sentinels and the tick re-sorts through the sized
-    // paths.  Sharded ranks keep the explicit path (the exchange needs host
-    // lengths).
     bool use_front = e->front.pending && e->front.ev == d_ev &&
                      e->front.n == n;
     if (e->front.pending && !use_front) {  // stale pipelined front: discard
@@ -1672,28 +1684,23 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     // — identical to the reference's dA join trace(P) + trace(A) delayed join dP
     // (operator/join.rs:217-292) by bilinearity.
     std::vector<DevBatch> outs;
+    bool chain_emits = false;       // emits + output sort queued pre-event
+    DevBatch comb_chain{};          // capacity emit buffer (arena)
+    int64_t slot_totals[3] = {0, 0, 0};
+    constexpr int64_t EMIT_CAP = 32768;
+    struct Plan {
+        const DevBatch *delta;
+        TraceArgs t;
+        int proj;
+        uint32_t *cnts;
+        uint64_t *offsets;
+        bool small;
+        bool dd;  // delta-vs-delta plan (trace length is tick-fresh)
+        int slot;
+    } plans[3];
+    int np = 0;
     {
         ScopedTimer timer(c, 2, (double)n * 24.0);
-        struct Plan {
-            const DevBatch *delta;
-            TraceArgs t;
-            int proj;
-            uint32_t *cnts;
-            uint64_t *offsets;
-            bool small;
-            bool dd;  // delta-vs-delta plan (trace length is tick-fresh)
-            int slot;
-        } plans[3];
-        int np = 0;
-        auto spine_args = [](Spine &sp) {
-            TraceArgs t{};
-            for (auto &b : sp.batches) {
-                if (b.n == 0) continue;
-                t.k[t.nb] = b.k; t.v[t.nb] = b.v; t.w[t.nb] = b.w;
-                t.n[t.nb] = b.n; t.nb++;
-            }
-            return t;
-        };
         if ((int)e->p_int.batches.size() > MAX_TRACE_BATCHES)
             TRY(e->p_int.consolidate_all(c));
         if ((int)e->a_int.batches.size() > MAX_TRACE_BATCHES)
@@ -1703,11 +1710,13 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             np = 0;
             const bool haveA = spec || dA.n > 0, haveP = spec || dP.n > 0;
             if (haveA && !e->p_int.batches.empty())
-                plans[np++] = {&dA, spine_args(e->p_int), DBSP_PROJ_HI_V2_LO_V1,
-                               nullptr, nullptr, false, false, -1};
+                plans[np++] = {&dA, trace_args_of(e->p_int),
+                               DBSP_PROJ_HI_V2_LO_V1, nullptr, nullptr,
+                               false, false, -1};
             if (haveP && !e->a_int.batches.empty())
-                plans[np++] = {&dP, spine_args(e->a_int), DBSP_PROJ_HI_V1_LO_V2,
-                               nullptr, nullptr, false, false, -1};
+                plans[np++] = {&dP, trace_args_of(e->a_int),
+                               DBSP_PROJ_HI_V1_LO_V2, nullptr, nullptr,
+                               false, false, -1};
             if (haveA && haveP) {
                 TraceArgs t{};
                 t.nb = 1;
@@ -1759,8 +1768,65 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                 HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
                                             16 * sizeof(int64_t),
                                             hipMemcpyDeviceToHost, c->stream));
-                HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+                if (spec && jca.np > 0) {
+                    (void)hipEventRecord(c->ev_sync2, c->stream);
+                    // chain the emits + output consolidate behind the
+                    // device-side totals (the host wakes at the event above,
+                    // before these run); the second readback lands their
+                    // verdicts at h_len[8..15): totals, combined total,
+                    // overflow flag, output length
+                    int64_t *bases =
+                        (int64_t *)arena_alloc(c, (jca.np + 1) * 8 + 8);
+                    DevBatch scr;
+                    if (bases &&
+                        alloc_batch(c, EMIT_CAP, comb_chain, true) == DBSP_OK &&
+                        alloc_batch(c, EMIT_CAP, scr, true) == DBSP_OK) {
+                        TRY(dbspk::emit_bases(c->stream, c->d_len, jca.np,
+                                              EMIT_CAP, bases, c->d_len + 3,
+                                              c->d_len + 4));
+                        for (int i = 0; i < np; i++) {
+                            Plan &pl = plans[i];
+                            if (pl.slot < 0) continue;
+                            TRY(dbspk::join_emit_chain(
+                                c->stream, pl.delta->k, pl.delta->v,
+                                pl.delta->w, jca.nd_dev[pl.slot], pl.t,
+                                jca.tn_dev[pl.slot], pl.cnts, pl.offsets,
+                                c->d_len + pl.slot, bases + pl.slot,
+                                c->d_len + 4, EMIT_CAP, pl.proj, 0,
+                                comb_chain.k, comb_chain.v, comb_chain.w));
+                        }
+                        if (!e->out_store.k) {
+                            e->out_cap = 8192;
+                            TRY(alloc_batch(c, e->out_cap, e->out_store));
+                        }
+                        SortArgs sa{};
+                        sa.nb = 1;
+                        sa.kin[0] = comb_chain.k;
+                        sa.vin[0] = comb_chain.v;
+                        sa.win[0] = comb_chain.w;
+                        sa.n_dev[0] = c->d_len + 3;
+                        sa.tk[0] = scr.k; sa.tv[0] = scr.v; sa.tw[0] = scr.w;
+                        sa.ok[0] = e->out_store.k;
+                        sa.ov[0] = e->out_store.v;
+                        sa.ow[0] = e->out_store.w;
+                        sa.d_len = c->d_len + 6;
+                        TRY(dbspk::sort_cons_small_batch(c->stream, sa));
+                        chain_emits = true;
+                    }
+                    // second readback: emit total / overflow flag / output
+                    // length land at h_len[12..15] (clear of the flatmap
+                    // counts and sort lengths the lost-speculation path
+                    // reads from the first copy)
+                    HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 12, c->d_len + 3,
+                                                4 * sizeof(int64_t),
+                                                hipMemcpyDeviceToHost,
+                                                c->stream));
+                    (void)hipEventSynchronize(c->ev_sync2);
+                } else {
+                    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+                }
             }
+            for (int i = 0; i < np && i < 3; i++) slot_totals[i] = c->h_len[i];
             if (!spec) break;
             bool lost = !arena_ok || c->h_len[10] < 0 || c->h_len[11] < 0;
             for (int i = 0; i < np && !lost; i++)
@@ -1773,6 +1839,7 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                 break;
             }
             e->spec_fail++;
+            chain_emits = false;  // the chained emits bailed on the flag
             // speculation lost: re-sort from the raw flatmap outputs with the
             // now-known lengths, then redo the planning explicitly
             rawA.n = c->h_len[8];
@@ -1787,20 +1854,21 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         }
         for (int i = 0; i < np; i++)
             if (plans[i].dd) plans[i].t.n[0] = dP.n;
+    }
+    auto emit_explicit = [&](void) -> dbsp_status {
+        ScopedTimer timer(c, 2, 0.0);
         int64_t total_small = 0;
         for (int i = 0; i < np; i++)
             if (plans[i].t.nb > 0 && plans[i].small)
-                total_small += c->h_len[plans[i].slot];
+                total_small += slot_totals[plans[i].slot];
         DevBatch comb;
         if (total_small > 0) TRY(alloc_batch(c, total_small, comb, true));
         int64_t base = 0;
-        // all small plans emit into ONE combined raw buffer at their base
-        // offsets — the downstream consolidate then needs no concat copies
         for (int i = 0; i < np; i++) {
             Plan &pl = plans[i];
             if (pl.t.nb == 0) continue;
             if (pl.small) {
-                int64_t total = c->h_len[pl.slot];
+                int64_t total = slot_totals[pl.slot];
                 if (total <= 0) continue;
                 TRY(dbspk::join_emit_prepared(c->stream, pl.delta->k,
                                               pl.delta->v, pl.delta->w,
@@ -1824,10 +1892,58 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         } else if (comb.k) {
             free_batch(c, comb);
         }
-    }
-    // Launch the output consolidate asynchronously, overlap it with the spine
-    // inserts (independent work), and read its length at their sync.
+        return DBSP_OK;
+    };
+    std::function<dbsp_status()> hook = [&]() -> dbsp_status {
+        if (!e->next_ev || e->next_n < 0 || e->next_n > 131072 ||
+            e->spec_fail >= 3)
+            return DBSP_OK;
+        const size_t save_base = c->arena_base, save_off = c->arena_off;
+        c->arena_base =
+            c->arena_half ? (save_base ? 0 : c->arena_half) : save_base;
+        c->arena_off = 0;
+        dbsp_status st = build_deltas_chain(e, e->next_ev, e->next_n,
+                                            e->front.rawA, e->front.rawB,
+                                            e->front.oA, e->front.oB);
+        e->front.arena_base = c->arena_base;
+        e->front.arena_off = c->arena_off;
+        c->arena_base = save_base;
+        c->arena_off = save_off;
+        if (st == DBSP_OK) {
+            e->front.pending = true;
+            e->front.ev = e->next_ev;
+            e->front.n = e->next_n;
+        }
+        return st;
+    };
     engine_free_output(e);
+    if (chain_emits) {
+        // inserts first — they overlap the still-running emits/output sort —
+        // then read the emit verdicts from the second readback
+        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
+        const int64_t flag = c->h_len[13];
+        if (flag == 0) {
+            const int64_t out_n = c->h_len[15];
+            if (out_n >= 0) {
+                e->output = e->out_store;
+                e->output.n = out_n;
+                e->output_is_store = true;
+            } else {
+                // output larger than the fused sort: the capacity buffer
+                // holds the raw rows
+                comb_chain.n = c->h_len[12];
+                TRY(sort_consolidate_batch(c, comb_chain, e->output));
+            }
+        } else {
+            // combined emit output exceeded the capacity buffer: replay the
+            // emits explicitly (counts/offsets are still valid)
+            TRY(emit_explicit());
+            TRY(finalize_raw(c, outs, e->output));
+        }
+        return DBSP_OK;
+    }
+    // explicit path (sharded ranks and lost speculations)
+    TRY(emit_explicit());
     int64_t cat_n = 0;
     for (auto &b : outs) cat_n += b.n;
     bool async_final = cat_n > 0 && cat_n <= 8192;
@@ -1843,9 +1959,12 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             int64_t off = 0;
             for (auto &b : outs) {
                 if (b.n == 0) continue;
-                HIP_CHECK_ST(hipMemcpyAsync(cat.k + off, b.k, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
-                HIP_CHECK_ST(hipMemcpyAsync(cat.v + off, b.v, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
-                HIP_CHECK_ST(hipMemcpyAsync(cat.w + off, b.w, b.n * 8, hipMemcpyDeviceToDevice, c->stream));
+                HIP_CHECK_ST(hipMemcpyAsync(cat.k + off, b.k, b.n * 8,
+                                            hipMemcpyDeviceToDevice, c->stream));
+                HIP_CHECK_ST(hipMemcpyAsync(cat.v + off, b.v, b.n * 8,
+                                            hipMemcpyDeviceToDevice, c->stream));
+                HIP_CHECK_ST(hipMemcpyAsync(cat.w + off, b.w, b.n * 8,
+                                            hipMemcpyDeviceToDevice, c->stream));
                 off += b.n;
             }
             for (auto &b : outs) free_batch(c, b);
@@ -1868,35 +1987,10 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         // no sync yet — the spine inserts below sync the stream
     }
     if (!sharding_on(c)) {
-        // queue the output length readback now (stream-ordered after the
-        // output sort) so the insert hook's event covers it, then let the
-        // inserts' first wait double as the pipelining point
         if (async_final)
             HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6,
                                         sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
-        std::function<dbsp_status()> hook = [&]() -> dbsp_status {
-            if (!e->next_ev || e->next_n < 0 || e->next_n > 131072 ||
-                e->spec_fail >= 3)
-                return DBSP_OK;
-            const size_t save_base = c->arena_base, save_off = c->arena_off;
-            c->arena_base =
-                c->arena_half ? (save_base ? 0 : c->arena_half) : save_base;
-            c->arena_off = 0;
-            dbsp_status st = build_deltas_chain(e, e->next_ev, e->next_n,
-                                                e->front.rawA, e->front.rawB,
-                                                e->front.oA, e->front.oB);
-            e->front.arena_base = c->arena_base;
-            e->front.arena_off = c->arena_off;
-            c->arena_base = save_base;
-            c->arena_off = save_off;
-            if (st == DBSP_OK) {
-                e->front.pending = true;
-                e->front.ev = e->next_ev;
-                e->front.n = e->next_n;
-            }
-            return st;
-        };
         TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
         if (async_final) {
             res.n = c->h_len[6];

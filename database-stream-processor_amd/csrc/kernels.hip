@@ -519,7 +519,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     const uint64_t lt = ((uint64_t)1 << lane) - 1;
 
     if (args.n_dev[batch]) {
-        if (n_chain > FUSE_MAX) {  // speculation lost: host re-sorts
+        if (n_chain > FUSE_MAX || n_chain < 0) {  // speculation lost
             if (tid == 0) *out_len = -1;
             return;
         }
@@ -1584,6 +1584,72 @@ __global__ void k_join_count_multi(const uint64_t *dk, int64_t nd, TraceArgs t,
     }
 }
 
+
+// ---------------------------------------------------------------------------
+// chained join emit: per-plan output bases from the device-side count totals
+// (so the emits and the output consolidate launch before the tick sync), and
+// the emit kernel itself reads every size from device memory.  flag != 0
+// (a lost speculation upstream, or the combined output exceeding the
+// capacity buffer) makes every consumer bail; the host re-emits explicitly.
+// ---------------------------------------------------------------------------
+__global__ void k_emit_bases(const int64_t *totals, int np, int64_t cap,
+                             int64_t *bases, int64_t *out_total,
+                             int64_t *out_flag) {
+    int64_t acc = 0;
+    int ok = 1;
+    for (int i = 0; i < np; i++) {
+        const int64_t t = totals[i];
+        if (t < 0) { ok = 0; break; }
+        bases[i] = acc;
+        acc += t;
+    }
+    if (acc > cap) ok = 0;
+    *out_total = ok ? acc : -1;
+    *out_flag = ok ? 0 : 1;
+}
+
+__global__ void k_join_emit_chain(const uint64_t *dk, const uint64_t *dv,
+                                  const int64_t *dw, const int64_t *nd_dev,
+                                  TraceArgs t, const int64_t *tn_dev,
+                                  const uint32_t *cnts,
+                                  const uint64_t *offsets,
+                                  const int64_t *total_dev,
+                                  const int64_t *base_dev,
+                                  const int64_t *flag_dev, int proj,
+                                  uint64_t param, uint64_t *ok, uint64_t *ov,
+                                  int64_t *ow) {
+    if (*flag_dev != 0) return;
+    const int64_t nd = *nd_dev;
+    const int64_t n_out = *total_dev;
+    if (nd <= 0 || n_out <= 0) return;
+    const int64_t base = *base_dev;
+    const int64_t tn0 = tn_dev ? *tn_dev : t.n[0];
+    for (int64_t o = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; o < n_out;
+         o += (int64_t)gridDim.x * blockDim.x) {
+        int64_t lo = 0, hi = nd;
+        while (lo < hi) {
+            int64_t mid = (lo + hi) / 2;
+            if (offsets[mid] <= (uint64_t)o) lo = mid + 1; else hi = mid;
+        }
+        int64_t i = lo - 1;
+        int64_t j = o - (int64_t)offsets[i];
+        int b = 0;
+        while (j >= (int64_t)cnts[i * t.nb + b]) {
+            j -= cnts[i * t.nb + b];
+            b++;
+        }
+        uint64_t key = dk[i];
+        const int64_t tb_n = b == 0 ? tn0 : t.n[b];
+        int64_t start = lower_bound_k(t.k[b], tb_n, key);
+        int64_t ti = start + j;
+        uint64_t hi_o, lo_o;
+        proj_out(proj, param, key, dv[i], t.v[b][ti], hi_o, lo_o);
+        ok[base + o] = hi_o;
+        ov[base + o] = lo_o;
+        ow[base + o] = dw[i] * t.w[b][ti];
+    }
+}
+
 __global__ void k_join_emit_multi(const uint64_t *dk, const uint64_t *dv,
                                   const int64_t *dw, int64_t nd, TraceArgs t,
                                   const uint32_t *cnts, const uint64_t *offsets,
@@ -2467,6 +2533,28 @@ dbsp_status join_count_scan_batch(hipStream_t s, const JoinCountArgs &args) {
     for (int i = 0; i < args.np; i++)
         if (args.nd[i] > FUSE_MAX) return DBSP_ERR_INVALID;
     k_join_count_scan_small<<<dim3((uint32_t)args.np), FUSE_THREADS, 0, s>>>(args);
+    return DBSP_OK;
+}
+
+
+dbsp_status emit_bases(hipStream_t s, const int64_t *totals, int np,
+                       int64_t cap, int64_t *bases, int64_t *out_total,
+                       int64_t *out_flag) {
+    k_emit_bases<<<1, 1, 0, s>>>(totals, np, cap, bases, out_total, out_flag);
+    return DBSP_OK;
+}
+
+dbsp_status join_emit_chain(hipStream_t s, const uint64_t *dk,
+                            const uint64_t *dv, const int64_t *dw,
+                            const int64_t *nd_dev, const TraceArgs &t,
+                            const int64_t *tn_dev, const uint32_t *cnts,
+                            const uint64_t *offsets, const int64_t *total_dev,
+                            const int64_t *base_dev, const int64_t *flag_dev,
+                            int64_t grid_cap, int proj, uint64_t param,
+                            uint64_t *ok, uint64_t *ov, int64_t *ow) {
+    k_join_emit_chain<<<grid_for(grid_cap), BLK, 0, s>>>(
+        dk, dv, dw, nd_dev, t, tn_dev, cnts, offsets, total_dev, base_dev,
+        flag_dev, proj, param, ok, ov, ow);
     return DBSP_OK;
 }
 

@@ -142,6 +142,19 @@ dbsp_status merge_small_batch(hipStream_t s, const MergeArgs &args);
 // each): per-plan per-row/per-batch cnts, exclusive offsets, totals to
 // d_total[i] (device)
 dbsp_status join_count_scan_batch(hipStream_t s, const JoinCountArgs &args);
+// chained emit machinery: per-plan bases from device totals, emit with every
+// size read device-side (flag short-circuits on lost speculation/overflow)
+dbsp_status emit_bases(hipStream_t s, const int64_t *totals, int np,
+                       int64_t cap, int64_t *bases, int64_t *out_total,
+                       int64_t *out_flag);
+dbsp_status join_emit_chain(hipStream_t s, const uint64_t *dk,
+                            const uint64_t *dv, const int64_t *dw,
+                            const int64_t *nd_dev, const TraceArgs &t,
+                            const int64_t *tn_dev, const uint32_t *cnts,
+                            const uint64_t *offsets, const int64_t *total_dev,
+                            const int64_t *base_dev, const int64_t *flag_dev,
+                            int64_t grid_cap, int proj, uint64_t param,
+                            uint64_t *ok, uint64_t *ov, int64_t *ow);
 // emit phase over precomputed cnts/offsets
 dbsp_status join_emit_prepared(hipStream_t s, const uint64_t *dk,
                                const uint64_t *dv, const int64_t *dw,
