@@ -1536,31 +1536,36 @@ static bool spine_needs_merge(Spine &s) {
 // recorded, and the first wait is hipEventSynchronize on that event — so
 // whatever the hook launches (the NEXT tick's front half) overlaps the
 // host-side tail of this tick instead of extending its sync.
-static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
-                                      Spine &s2, DevBatch b2,
-                                      const std::function<dbsp_status()> *hook
-                                      = nullptr) {
-    ScopedTimer t(c, 1, (double)(b1.n + b2.n) * 48.0);
+static dbsp_status spines_insert_multi(dbsp_ctx *c, Spine *const *sps,
+                                       DevBatch *bs, int ns,
+                                       const std::function<dbsp_status()> *hook
+                                       = nullptr) {
+    double bytes = 0;
+    for (int i = 0; i < ns; i++) bytes += (double)bs[i].n * 48.0;
+    ScopedTimer t(c, 1, bytes);
     bool hook_fired = false;
-    if (b1.n > 0) s1.batches.push_back(b1);
-    else free_batch(c, b1);
-    if (b2.n > 0) s2.batches.push_back(b2);
-    else free_batch(c, b2);
+    for (int i = 0; i < ns; i++) {
+        if (bs[i].n > 0) sps[i]->batches.push_back(bs[i]);
+        else free_batch(c, bs[i]);
+    }
     while (true) {
-        Spine *pending[2];
+        Spine *pending[4];
         int nps = 0;
-        if (spine_needs_merge(s1)) pending[nps++] = &s1;
-        if (spine_needs_merge(s2)) pending[nps++] = &s2;
+        for (int i = 0; i < ns && nps < 4; i++) {
+            bool dup = false;
+            for (int j = 0; j < nps; j++) dup |= pending[j] == sps[i];
+            if (!dup && spine_needs_merge(*sps[i])) pending[nps++] = sps[i];
+        }
         if (nps == 0) break;
         MergeArgs ma{};
-        DevBatch results[2];
-        Spine *owners[2];
+        DevBatch results[4];
+        Spine *owners[4];
         int nbatched = 0;
         for (int i = 0; i < nps; i++) {
             Spine &sp = *pending[i];
             DevBatch &top = sp.batches.back();
             DevBatch &below = sp.batches[sp.batches.size() - 2];
-            if (top.n + below.n <= 32768) {
+            if (top.n + below.n <= 32768 && ma.np < MERGE_BATCH_MAX) {
                 DevBatch res;
                 TRY(alloc_batch(c, top.n + below.n, res));
                 int p = ma.np++;
@@ -1617,6 +1622,15 @@ static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
         (void)hipEventSynchronize(c->ev_sync);
     }
     return DBSP_OK;
+}
+
+static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
+                                      Spine &s2, DevBatch b2,
+                                      const std::function<dbsp_status()> *hook
+                                      = nullptr) {
+    Spine *sps[2] = {&s1, &s2};
+    DevBatch bs[2] = {b1, b2};
+    return spines_insert_multi(c, sps, bs, 2, hook);
 }
 
 static void engine_free_output(dbsp_engine *e) {
@@ -2155,7 +2169,10 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         e->q8_s0 = s1;
         e->q8_e0 = e1;
     }
-    TRY(spines_insert_pair(c, e->pt_int, dPT, e->at_int, dAT));
+    if (sharding_on(c)) {
+        TRY(spines_insert_pair(c, e->pt_int, dPT, e->at_int, dAT));
+    }  // single-rank: deferred into the tail multi-insert (the traces are
+       // only read next tick, and one batched call saves a sync round)
     // map_index / map + consolidate
     DevBatch wpr, war, dWP, dWA;
     TRY(finalize_raw(c, wp_raw, wpr));
@@ -2259,7 +2276,9 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             }
             return st;
         };
-        TRY(spines_insert_pair(c, e->wp_int, dWP, e->wa_int, dWA, &hook));
+        Spine *sps[4] = {&e->wp_int, &e->wa_int, &e->pt_int, &e->at_int};
+        DevBatch bs[4] = {dWP, dWA, dPT, dAT};
+        TRY(spines_insert_multi(c, sps, bs, 4, &hook));
     } else {
         TRY(spines_insert_pair(c, e->wp_int, dWP, e->wa_int, dWA));
     }
@@ -2397,7 +2416,9 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         e->q5_s0 = s1;
         e->q5_e0 = e1;
     }
-    TRY(e->bt_int.insert(c, dBT));
+    if (sharding_on(c)) TRY(e->bt_int.insert(c, dBT));
+    // single-rank: bt_int is only read next tick — inserted in the tail
+    // multi-insert
     DevBatch wbr, dWB;
     TRY(finalize_raw(c, wb_raw, wbr));
     TRY(map_sorted(c, wbr, 1, dWB));  // (time,auction) -> (auction,()); weigh(|_|1)
@@ -2556,8 +2577,9 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             }
             return st;
         };
-        TRY(spines_insert_pair(c, e->bc_int, dBC, e->counts_int, dCounts,
-                               &hook));
+        Spine *sps[3] = {&e->bc_int, &e->counts_int, &e->bt_int};
+        DevBatch bs[3] = {dBC, dCounts, dBT};
+        TRY(spines_insert_multi(c, sps, bs, 3, &hook));
     } else {
         TRY(e->bc_int.insert(c, dBC));
         TRY(e->counts_int.insert(c, dCounts));
