@@ -1367,16 +1367,8 @@ __global__ __launch_bounds__(MP_THREADS, 3) void k_mp_merge_onepass(
     const int64_t pb0 = pb[vb], pb1 = pb[vb + 1];
     const int64_t naL = pa1 - pa0, nbL = pb1 - pb0;
     const int64_t totL = naL + nbL;
-    for (int64_t i = tid; i < naL; i += MP_THREADS) {
-        lk[i] = ak[pa0 + i];
-        lv[i] = av[pa0 + i];
-        lw[i] = aw[pa0 + i];
-    }
-    for (int64_t i = tid; i < nbL; i += MP_THREADS) {
-        lk[naL + i] = bk[pb0 + i];
-        lv[naL + i] = bv[pb0 + i];
-        lw[naL + i] = bw[pb0 + i];
-    }
+    stage_run<true>(lk, lv, lw, 0, ak + pa0, av + pa0, aw + pa0, naL, tid);
+    stage_run<true>(lk, lv, lw, naL, bk + pb0, bv + pb0, bw + pb0, nbL, tid);
     __syncthreads();
     const int64_t items = (totL + MP_THREADS - 1) / MP_THREADS;
     int64_t d0 = min((int64_t)tid * items, totL);
@@ -2132,7 +2124,12 @@ __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
             // q8.rs:50-60
             if (e.kind == 0) {
                 uint64_t p = atomicAdd((unsigned long long *)c0, 1ull);
-                k0[p] = e.f4; v0[p] = (e.f0 << 32) | (e.f1 & 0xFFFFFFFFull);
+                // (id, name) packed ORDER-PRESERVING as id*1024+name: the
+                // name dictionary is the generator's ~1000-value space
+                // (people.rs name draw), so 10 bits suffice — and the packed
+                // range stays ~2^20 per tick instead of id<<32 spanning 2^42,
+                // which halves the delta sort's digit passes
+                k0[p] = e.f4; v0[p] = (e.f0 << 10) | (e.f1 & 0x3FFull);
                 w0[p] = e.w;
             } else if (e.kind == 1) {
                 uint64_t p = atomicAdd((unsigned long long *)c1, 1ull);
@@ -2143,7 +2140,7 @@ __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
 }
 
 // generic per-row map for the small derived streams (q5/q8):
-// mode 0: (k,v) -> (v>>32, (v&lo32)<<32 | (k&lo32))   [q8 people_by_id map_index]
+// mode 0: (k,v) -> (v>>10, (v&0x3FF)<<32 | (k&lo32))  [q8 people_by_id map_index]
 // mode 1: (k,v) -> (v, 0)                             [q8 auctions map / q5 windowed-bids map]
 // mode 2: (k,v) -> (0, v)                             [q5 map_index ((),count)]
 // mode 3: (k,v) -> (v, k)                             [q5 by_count map_index]
@@ -2156,7 +2153,7 @@ __global__ void k_map(const uint64_t *k, const uint64_t *v, const int64_t *w,
         uint64_t rk, rv;
         int64_t rw = w[i];
         switch (mode) {
-            case 0: rk = vv >> 32; rv = ((vv & 0xFFFFFFFFull) << 32) | (kk & 0xFFFFFFFFull); break;
+            case 0: rk = vv >> 10; rv = ((vv & 0x3FFull) << 32) | (kk & 0xFFFFFFFFull); break;
             case 1: rk = vv; rv = 0; break;
             case 2: rk = 0; rv = vv; break;
             case 4: {  // weigh (aggregate/mod.rs:297-323): f(k,v)=f64(v); w' = f(k,v)*w

@@ -369,7 +369,8 @@ std::vector<Row> q8_step(Oracle &o, const std::vector<dbsp_event> &ev) {
     constexpr uint64_t TUMBLE_MS = 10'000;  // queries/q8.rs:46
     std::vector<Row> dPT, dAT;
     for (auto &e : ev) {
-        if (e.kind == 0) dPT.push_back({e.f4, (e.f0 << 32) | (e.f1 & 0xFFFFFFFFull), e.w});
+        if (e.kind == 0)  // (id, name) packed order-preserving as id*1024+name
+            dPT.push_back({e.f4, (e.f0 << 10) | (e.f1 & 0x3FFull), e.w});
         if (e.kind == 1) dAT.push_back({e.f3, e.f1, e.w});
     }
     consolidate(dPT);
@@ -391,7 +392,7 @@ std::vector<Row> q8_step(Oracle &o, const std::vector<dbsp_event> &ev) {
     // map_index (q8.rs:84): (dt,(id,name)) -> (id, name<<32|dt)
     std::vector<Row> dWP, dWA;
     for (auto &r : dWPraw)
-        dWP.push_back({r.v >> 32, ((r.v & 0xFFFFFFFFull) << 32) | (r.k & 0xFFFFFFFFull), r.w});
+        dWP.push_back({r.v >> 10, ((r.v & 0x3FFull) << 32) | (r.k & 0xFFFFFFFFull), r.w});
     // map (q8.rs:87): (dt,seller) -> (seller,())
     for (auto &r : dWAraw) dWA.push_back({r.v, 0, r.w});
     consolidate(dWP);
