@@ -2096,44 +2096,75 @@ __global__ void k_shard_scatter(const uint64_t *k, const uint64_t *v,
 // consolidate that follows sorts)
 // ---------------------------------------------------------------------------
 
+// wave-aggregated ticket append: one atomicAdd per wave per predicate
+// instead of one per matching row (the flatmap's counters were the hottest
+// atomics in the tick)
+__device__ inline uint64_t wave_append(unsigned long long *ctr, bool pred) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const uint64_t m = __ballot(pred);
+    if (m == 0) return 0;
+    const uint64_t lt = ((uint64_t)1 << lane) - 1;
+    unsigned long long base = 0;
+    const int leader = __ffsll((unsigned long long)m) - 1;
+    if (lane == leader)
+        base = atomicAdd(ctr, (unsigned long long)__popcll(m));
+    base = (unsigned long long)__shfl((long long)base, leader, WAVE);
+    return base + (uint64_t)__popcll(m & lt);
+}
+
 __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
                           uint64_t *c0, uint64_t *k0, uint64_t *v0, int64_t *w0,
                           uint64_t *c1, uint64_t *k1, uint64_t *v1, int64_t *w1) {
-    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
-         i += (int64_t)gridDim.x * blockDim.x) {
-        dbsp_event e = ev[i];
+    // wave-uniform outer loop (wb is the wave's first row) so the ballots in
+    // wave_append see every lane
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t wb = blockIdx.x * (int64_t)blockDim.x +
+                      (threadIdx.x & ~(int64_t)(WAVE - 1));
+         wb < n; wb += stride) {
+        const int64_t i = wb + lane;
+        dbsp_event e{};
+        const bool act = i < n;
+        if (act) e = ev[i];
         if (query == 3) {
             // q3.rs:37-49
-            if (e.kind == 1 && e.f2 == 10) {
-                uint64_t p = atomicAdd((unsigned long long *)c0, 1ull);
-                k0[p] = e.f1; v0[p] = e.f0; w0[p] = e.w;
-            } else if (e.kind == 0 &&
-                       (e.f3 == 1 || e.f3 == 2 || e.f3 == 3)) {  // CA, ID, OR
-                uint64_t p = atomicAdd((unsigned long long *)c1, 1ull);
-                k1[p] = e.f0;
-                v1[p] = (e.f1 << 8) | ((e.f2 & 0xF) << 4) | (e.f3 & 0xF);
-                w1[p] = e.w;
+            const bool p0 = act && e.kind == 1 && e.f2 == 10;
+            const uint64_t pos0 = wave_append((unsigned long long *)c0, p0);
+            if (p0) {
+                k0[pos0] = e.f1; v0[pos0] = e.f0; w0[pos0] = e.w;
+            }
+            const bool p1 = act && e.kind == 0 &&
+                            (e.f3 == 1 || e.f3 == 2 || e.f3 == 3);  // CA,ID,OR
+            const uint64_t pos1 = wave_append((unsigned long long *)c1, p1);
+            if (p1) {
+                k1[pos1] = e.f0;
+                v1[pos1] = (e.f1 << 8) | ((e.f2 & 0xF) << 4) | (e.f3 & 0xF);
+                w1[pos1] = e.w;
             }
         } else if (query == 5) {
             // q5.rs:79-83: bids by time
-            if (e.kind == 2) {
-                uint64_t p = atomicAdd((unsigned long long *)c0, 1ull);
-                k0[p] = e.f3; v0[p] = e.f0; w0[p] = e.w;
+            const bool p0 = act && e.kind == 2;
+            const uint64_t pos0 = wave_append((unsigned long long *)c0, p0);
+            if (p0) {
+                k0[pos0] = e.f3; v0[pos0] = e.f0; w0[pos0] = e.w;
             }
         } else if (query == 8) {
             // q8.rs:50-60
-            if (e.kind == 0) {
-                uint64_t p = atomicAdd((unsigned long long *)c0, 1ull);
+            const bool p0 = act && e.kind == 0;
+            const uint64_t pos0 = wave_append((unsigned long long *)c0, p0);
+            if (p0) {
                 // (id, name) packed ORDER-PRESERVING as id*1024+name: the
                 // name dictionary is the generator's ~1000-value space
                 // (people.rs name draw), so 10 bits suffice — and the packed
                 // range stays ~2^20 per tick instead of id<<32 spanning 2^42,
                 // which halves the delta sort's digit passes
-                k0[p] = e.f4; v0[p] = (e.f0 << 10) | (e.f1 & 0x3FFull);
-                w0[p] = e.w;
-            } else if (e.kind == 1) {
-                uint64_t p = atomicAdd((unsigned long long *)c1, 1ull);
-                k1[p] = e.f3; v1[p] = e.f1; w1[p] = e.w;
+                k0[pos0] = e.f4; v0[pos0] = (e.f0 << 10) | (e.f1 & 0x3FFull);
+                w0[pos0] = e.w;
+            }
+            const bool p1 = act && e.kind == 1;
+            const uint64_t pos1 = wave_append((unsigned long long *)c1, p1);
+            if (p1) {
+                k1[pos1] = e.f3; v1[pos1] = e.f1; w1[pos1] = e.w;
             }
         }
     }
