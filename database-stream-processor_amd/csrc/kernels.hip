@@ -1139,16 +1139,17 @@ __global__ void k_join_emit(const uint64_t *dk, const uint64_t *dv,
 // ---------------------------------------------------------------------------
 
 #define MP_TILE 1024
+#define MP_TILE_OP 2048  // single-pass lookback: wider tiles than the two-pass shorten the chain
 #define MP_THREADS 256
 #define MP_ITEMS (MP_TILE / MP_THREADS)  // 16 diagonals per thread
 
 __global__ void k_mp_partition(const uint64_t *ak, const uint64_t *av,
                                int64_t na, const uint64_t *bk,
                                const uint64_t *bv, int64_t nb, int64_t nblocks,
-                               int64_t *pa, int64_t *pb) {
+                               int64_t tile, int64_t *pa, int64_t *pb) {
     int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
     if (i > nblocks) return;
-    int64_t d = min(i * MP_TILE, na + nb);
+    int64_t d = min(i * tile, na + nb);
     int64_t ai, bi;
     merge_path(ak, av, na, bk, bv, nb, d, ai, bi);
     adjust_split(ak, av, bk, bv, na, nb, ai, bi);
@@ -1353,8 +1354,8 @@ __global__ __launch_bounds__(MP_THREADS, 3) void k_mp_merge_onepass(
     int64_t nblocks, uint64_t *ok, uint64_t *ov, W *ow) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     uint64_t *lk = (uint64_t *)smem;
-    uint64_t *lv = lk + (MP_TILE + 2);
-    W *lw = (W *)(lv + (MP_TILE + 2));
+    uint64_t *lv = lk + (MP_TILE_OP + 2);
+    W *lw = (W *)(lv + (MP_TILE_OP + 2));
     __shared__ uint32_t wt[MP_THREADS / WAVE + 1];
     __shared__ unsigned long long sh_vb;
     __shared__ unsigned long long sh_prefix;
@@ -2339,12 +2340,13 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
         const char *e = getenv("DBSP_MERGE_ONEPASS");
         return e && e[0] == '1';
     }();
-    int64_t nblocks = ceil_div(total, MP_TILE);
+    const int64_t tile = onepass ? MP_TILE_OP : MP_TILE;
+    int64_t nblocks = ceil_div(total, tile);
     int64_t *pa, *pb;
     HIP_CHECK(dbspk::cache_malloc((void **)&pa, (nblocks + 1) * sizeof(int64_t), s));
     HIP_CHECK(dbspk::cache_malloc((void **)&pb, (nblocks + 1) * sizeof(int64_t), s));
     k_mp_partition<<<grid_for(nblocks + 1), BLK, 0, s>>>(ak, av, na, bk, bv, nb,
-                                                         nblocks, pa, pb);
+                                                         nblocks, tile, pa, pb);
     uint64_t *rk, *rv;
     W *rw;
     if (onepass) {
@@ -2354,7 +2356,7 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
         HIP_CHECK(dbspk::cache_malloc((void **)&rk, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rv, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rw, total * sizeof(W) + 8, s));
-        const size_t smem = 3 * (MP_TILE + 2) * sizeof(uint64_t);
+        const size_t smem = 3 * (MP_TILE_OP + 2) * sizeof(uint64_t);
         k_mp_merge_onepass<W><<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
             ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         unsigned long long h_state[2];
