@@ -174,3 +174,18 @@ def test_query_parity_oversized_deltas(ctx, query):
         evs.append(bid_event(int(rng.integers(0, 1 << 40)), dt))
     _run_parity(ctx, query, events(*evs), tick=30_000,
                 seed_note="+oversized")
+
+
+@pytest.mark.parametrize("query", [5, 8])
+def test_query_parity_window_slides(ctx, query):
+    """Accelerated event time: 200k events spanning 20s of event time, so the
+    q5/q8 tumbling windows slide on ~every tick (the generated parity streams
+    at the 10M/s rate never cross a 2s tumble inside 100k events). Exercises
+    the 3-region retract/insert path and the device watermark end to end."""
+    evs = gen.generate(200_000, seed=37)
+    dt = (10_000_000 + np.arange(len(evs)) // 10).astype(np.uint64)
+    person = evs["kind"] == 0
+    bid_or_auction = ~person
+    evs["f4"] = np.where(person, dt, evs["f4"])     # person dt
+    evs["f3"] = np.where(bid_or_auction, dt, evs["f3"])  # auction/bid dt
+    _run_parity(ctx, query, evs, tick=20_000, seed_note="+slides")
