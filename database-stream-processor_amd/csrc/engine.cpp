@@ -2332,6 +2332,7 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         free_batch(c, raw1);
         const int64_t n0 = c->h_len[8];
         raw0.n = n0;
+        bool n_pending = false;  // dBT.n still on the device (d_len[5])
         if (n0 == 0) {
             dBT = DevBatch{};
             free_batch(c, raw0);
@@ -2343,22 +2344,31 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             if (n0 > 8192 && kr < (uint64_t)dense_cap &&
                 vr < (uint64_t)dense_cap &&
                 (int64_t)((kr + 1) * (vr + 1)) <= dense_cap) {
+                // chained: length stays on the device until the window sync
                 ScopedTimer t(c, 0, (double)n0 * 48.0);
                 DevBatch res;
                 TRY(alloc_batch(c, n0, res));
-                TRY(dbspk::sort_cons_dense(c->stream, raw0.k, raw0.v, raw0.w,
-                                           n0, mm[0], mm[1], (int64_t)(kr + 1),
-                                           (int64_t)(vr + 1), res.k, res.v,
-                                           res.w, &res.n));
+                TRY(dbspk::sort_cons_dense_chain(
+                    c->stream, raw0.k, raw0.v, raw0.w, n0, mm[0], mm[1],
+                    (int64_t)(kr + 1), (int64_t)(vr + 1), res.k, res.v, res.w,
+                    c->d_len + 5));
                 free_batch(c, raw0);
                 dBT = res;
+                dBT.n = -1;
+                n_pending = true;
             } else {
                 TRY(sort_consolidate_batch(c, raw0, dBT));
             }
         }
         // device watermark + chained window ranges (q5.rs:85-90)
-        TRY(dbspk::wm_update_n(c->stream, dBT.k, dBT.n, WIDTH_MS, TUMBLE_MS,
-                               WM_LAG_MS, e->d_wm, e->d_bounds));
+        if (n_pending) {
+            TRY(dbspk::wm_update(c->stream, dBT.k, c->d_len + 5, WIDTH_MS,
+                                 TUMBLE_MS, WM_LAG_MS, e->d_wm, e->d_bounds));
+        } else {
+            TRY(dbspk::wm_update_n(c->stream, dBT.k, dBT.n, WIDTH_MS,
+                                   TUMBLE_MS, WM_LAG_MS, e->d_wm,
+                                   e->d_bounds));
+        }
         if ((int)e->bt_int.batches.size() > MAX_TRACE_BATCHES)
             TRY(e->bt_int.consolidate_all(c));
         TraceArgs taB = trace_args_of(e->bt_int);
@@ -2367,14 +2377,19 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         if (tableB) {
             {
                 ScopedTimer t(c, 4, 0.0);
-                TRY(dbspk::window_ranges_chain(c->stream, taB, dBT.k, dBT.n,
-                                               nullptr, e->d_bounds, tableB,
-                                               c->d_len + 7));
+                TRY(dbspk::window_ranges_chain(
+                    c->stream, taB, dBT.k, dBT.n,
+                    n_pending ? c->d_len + 5 : nullptr, e->d_bounds, tableB,
+                    c->d_len + 7));
             }
-            HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 7, c->d_len + 7,
-                                        sizeof(int64_t), hipMemcpyDeviceToHost,
-                                        c->stream));
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
+                                        16 * sizeof(int64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
             HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            if (n_pending) {
+                dBT.n = c->h_len[5];
+                n_pending = false;
+            }
             if (c->h_len[7] > 0) {
                 DevBatch o;
                 TRY(alloc_batch(c, c->h_len[7], o, true));
@@ -2386,12 +2401,19 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             front_done = true;
         } else {
             // no arena for the table: wm_update already advanced the device
-            // state and published the bounds — read them and run the
-            // explicit window path
+            // state and published the bounds — read them (and the pending
+            // dense length) and run the explicit window path
             HIP_CHECK_ST(hipMemcpyAsync(e->h_bounds, e->d_bounds,
                                         6 * sizeof(uint64_t),
                                         hipMemcpyDeviceToHost, c->stream));
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
+                                        16 * sizeof(int64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
             HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            if (n_pending) {
+                dBT.n = c->h_len[5];
+                n_pending = false;
+            }
             TRY(window_vs_spine(c, e->bt_int, dBT, e->h_bounds[4] != 0,
                                 e->h_bounds[0], e->h_bounds[1], e->h_bounds[2],
                                 e->h_bounds[3], wb_raw));

@@ -876,6 +876,12 @@ __global__ void k_hist_flags(const unsigned long long *wbuf, int64_t r,
         flags[i] = wbuf[i] != 0;
 }
 
+__global__ void k_hist_total(const unsigned long long *wbuf,
+                             const uint64_t *flag_scan, int64_t r,
+                             int64_t *out) {
+    *out = (int64_t)(flag_scan[r - 1] + (wbuf[r - 1] != 0 ? 1 : 0));
+}
+
 __global__ void k_hist_emit(const unsigned long long *wbuf,
                             const uint64_t *flag_scan, int64_t r,
                             uint64_t kbase, uint64_t vbase, uint64_t vspan,
@@ -2464,6 +2470,34 @@ dbsp_status minmax_rows_chain(hipStream_t s, const uint64_t *k,
     HIP_CHECK(hipMemcpyAsync(mm_dev, init, sizeof(init), hipMemcpyHostToDevice,
                              s));
     k_minmax_rows<<<grid_for(cap), BLK, 0, s>>>(k, v, 0, mm_dev, n_dev);
+    return DBSP_OK;
+}
+
+// chained variant: no host sync — the consolidated length lands in
+// *out_n_dev for the caller's one tick sync (the emit grid covers the whole
+// histogram range, so nothing here needs the total on the host)
+dbsp_status sort_cons_dense_chain(hipStream_t s, const uint64_t *k,
+                                  const uint64_t *v, const int64_t *w,
+                                  int64_t n, uint64_t kbase, uint64_t vbase,
+                                  int64_t kspan, int64_t vspan, uint64_t *ok,
+                                  uint64_t *ov, int64_t *ow,
+                                  int64_t *out_n_dev) {
+    const int64_t r = kspan * vspan;
+    unsigned long long *wbuf;
+    uint64_t *flags;
+    HIP_CHECK(dbspk::cache_malloc((void **)&wbuf, r * sizeof(uint64_t), s));
+    HIP_CHECK(dbspk::cache_malloc((void **)&flags, (r + 1) * sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(wbuf, 0, r * sizeof(uint64_t), s));
+    k_hist_add<<<grid_for(n), BLK, 0, s>>>(k, v, w, n, kbase, vbase,
+                                           (uint64_t)vspan, wbuf);
+    k_hist_flags<<<grid_for(r), BLK, 0, s>>>(wbuf, r, flags);
+    dbsp_status st = scan_exclusive(s, flags, flags, r, nullptr);
+    if (st != DBSP_OK) return st;
+    k_hist_total<<<1, 1, 0, s>>>(wbuf, flags, r, out_n_dev);
+    k_hist_emit<<<grid_for(r), BLK, 0, s>>>(wbuf, flags, r, kbase, vbase,
+                                            (uint64_t)vspan, ok, ov, ow);
+    HIP_CHECK(dbspk::cache_free(wbuf, s));
+    HIP_CHECK(dbspk::cache_free(flags, s));
     return DBSP_OK;
 }
 

@@ -133,6 +133,13 @@ dbsp_status sort_cons_dense(hipStream_t s, const uint64_t *k, const uint64_t *v,
                             uint64_t vbase, int64_t kspan, int64_t vspan,
                             uint64_t *ok, uint64_t *ov, int64_t *ow,
                             int64_t *out_n);
+// chained dense consolidate: length to *out_n_dev, no sync
+dbsp_status sort_cons_dense_chain(hipStream_t s, const uint64_t *k,
+                                  const uint64_t *v, const int64_t *w,
+                                  int64_t n, uint64_t kbase, uint64_t vbase,
+                                  int64_t kspan, int64_t vspan, uint64_t *ok,
+                                  uint64_t *ov, int64_t *ow,
+                                  int64_t *out_n_dev);
 
 // single-workgroup merge of two consolidated batches (na+nb <= 8192):
 // one launch, no host sync; length left in *d_len (device)
