@@ -289,3 +289,25 @@ def test_oracle_distinct_model():
         if w:
             exp[key] = w
     assert got == exp
+
+
+def test_oracle_q3_tick_deltas_integrate_to_batch_result():
+    """Incremental correctness as an algebraic property: q3 has no windows,
+    so the Z-set SUM of the per-tick output deltas must equal the output of
+    running the whole stream as one tick (the incremental circuit computes
+    d(out) = lift(join)(d(in)) integrated; dbsp paper Thm. 4.4 shape)."""
+    from collections import Counter
+    from dbsp_amd import gen
+    evs = gen.generate(50_000, seed=71)
+    q_inc = oracle.Query(3)
+    integral = Counter()
+    for lo in range(0, len(evs), 5_000):
+        for r in q_inc.step(evs[lo:lo + 5_000], cap=1 << 22):
+            integral[(int(r["k"]), int(r["v"]))] += int(r["w"])
+    q_inc.close()
+    integral = {kv: w for kv, w in integral.items() if w != 0}
+    q_one = oracle.Query(3)
+    single = {(int(r["k"]), int(r["v"])): int(r["w"])
+              for r in q_one.step(evs, cap=1 << 22)}
+    q_one.close()
+    assert integral == single
