@@ -430,15 +430,18 @@ __global__ void k_radix_scatter(const uint64_t *k_in, const uint64_t *v_in,
 }
 
 // ---------------------------------------------------------------------------
-// fused single-workgroup sort+consolidate for small batches (n <= 16384)
+// fused single-workgroup sort+consolidate for small batches (n <= 8192)
 //
 // A 40k-event tick produces per-stream deltas of a few thousand rows; the
 // multi-kernel radix path costs ~30 launches + 3 host syncs there, which
-// dominates the tick (measured: 70% of wall in inter-dispatch gaps).  This
-// kernel does the whole consolidate_slice job — LSD radix over the permutation
-// in LDS, then dedup-accumulate and zero-drop — in ONE launch with the output
-// length left in device memory.  1024 threads = 16 waves on one CU; the
-// permutation (digit<<14|idx u32) ping-pongs between two 64 KiB LDS arrays.
+// dominates the tick.  This kernel does the whole consolidate_slice job —
+// LSD radix over a permutation held in LDS, then dedup-accumulate and
+// zero-drop — in ONE launch with the output length left in device memory.
+// 1024 threads = 16 waves on one CU; the permutation entry
+// (key16 << 13 | idx, u32) carries the next 16 key bits so digit passes are
+// LDS-only, ranks come from bit-sliced wave ballots over strided positions
+// (stable by construction), and the entry arrays ping-pong between two
+// 32 KiB LDS buffers.  See the kernel's own comment for the details.
 // ---------------------------------------------------------------------------
 
 #define FUSE_MAX 8192
@@ -459,14 +462,18 @@ __device__ inline uint32_t fuse_scan(uint32_t thread_sum, uint32_t *wave_tot,
     int wid = threadIdx.x / WAVE;  // 16 waves
     if ((threadIdx.x & (WAVE - 1)) == WAVE - 1) wave_tot[wid] = v;
     __syncthreads();
-    if (threadIdx.x == 0) {
-        uint32_t acc = 0;
-        for (int w = 0; w < FUSE_THREADS / WAVE; w++) {
-            uint32_t t = wave_tot[w];
-            wave_tot[w] = acc;
-            acc += t;
+    if (threadIdx.x < WAVE) {
+        // wave 0 scans the 16 wave totals with shuffles (the serial
+        // thread-0 loop cost ~1 us of LDS latency per call)
+        const int nw = FUSE_THREADS / WAVE;
+        uint32_t t = threadIdx.x < nw ? wave_tot[threadIdx.x] : 0;
+        uint32_t sc = t;
+        for (int d = 1; d < nw; d <<= 1) {
+            uint32_t up = __shfl_up(sc, d, WAVE);
+            if ((int)threadIdx.x >= d) sc += up;
         }
-        wave_tot[FUSE_THREADS / WAVE] = acc;
+        if ((int)threadIdx.x < nw) wave_tot[threadIdx.x] = sc - t;
+        if ((int)threadIdx.x == nw - 1) wave_tot[nw] = sc;
     }
     __syncthreads();
     uint32_t r = wave_tot[wid] + (v - thread_sum);
