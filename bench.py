@@ -65,8 +65,14 @@ def run_engine(ctx_dev, query, events, n_ticks, tick, world, rank, nccl_id,
     from dbsp_amd.engine import Ctx, Engine
 
     ctx = Ctx(ctx_dev)
-    if world > 1:
+    if world > 1 or os.environ.get("DBSP_FORCE_SHARD") == "1":
         import ctypes
+        import numpy as np
+        if nccl_id is None:  # forced single-rank sharding (self-exchange)
+            nccl_id = np.zeros(128, dtype=np.uint8)
+            ctx._lib.dbsp_comm_unique_id.restype = ctypes.c_int32
+            assert ctx._lib.dbsp_comm_unique_id(
+                nccl_id.ctypes.data_as(ctypes.c_void_p)) == 0
         ctx._lib.dbsp_comm_init.restype = ctypes.c_int32
         st = ctx._lib.dbsp_comm_init(ctx._h, rank, world,
                                      nccl_id.ctypes.data_as(ctypes.c_void_p))

@@ -873,24 +873,6 @@ static inline bool sharding_on(dbsp_ctx *c) {
     return c->world > 1 || (c->force_shard && c->comm);
 }
 
-// global max over ranks (the multi-worker watermark exchange,
-// watermark.rs:47-73, as an RCCL allreduce of one u64)
-static dbsp_status allreduce_max_u64(dbsp_ctx *c, uint64_t *h_val) {
-    if (!sharding_on(c)) return DBSP_OK;
-    uint64_t *d;
-    HIP_CHECK_ST(dbspk::cache_malloc((void **)&d, sizeof(uint64_t), c->stream));
-    HIP_CHECK_ST(hipMemcpyAsync(d, h_val, sizeof(uint64_t),
-                                hipMemcpyHostToDevice, c->stream));
-    if (ncclAllReduce(d, d, 1, ncclUint64, ncclMax, c->comm, c->stream) !=
-        ncclSuccess)
-        return DBSP_ERR_INTERNAL;
-    HIP_CHECK_ST(hipMemcpyAsync(h_val, d, sizeof(uint64_t),
-                                hipMemcpyDeviceToHost, c->stream));
-    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
-    HIP_CHECK_ST(dbspk::cache_free(d, c->stream));
-    return DBSP_OK;
-}
-
 // shard + exchange + rebuild: the full shard() operator
 // (shard.rs:88-199: hash-split, exchange, re-consolidate)
 static dbsp_status shard_exchange(dbsp_ctx *c, DevBatch local, DevBatch &out) {
@@ -932,13 +914,9 @@ struct dbsp_engine {
     Spine a_int, p_int;
     // q8 state
     Spine pt_int, at_int, wp_int, wa_int;
-    bool q8_have_prev = false;
-    uint64_t q8_s0 = 0, q8_e0 = 0, q8_wm = 0;
     // q5 state
     Spine bt_int, wb_int, counts_int, bc_int;
     DevBatch maxin_int, maxout_int, maxz_int;
-    bool q5_have_prev = false;
-    uint64_t q5_s0 = 0, q5_e0 = 0, q5_wm = 0;
 
     // chained-tick device watermark state: {wm, s0, e0, have_prev} and the
     // published bounds {s0,e0,s1,e1,have_prev,err} (q5/q8 single-rank path)
@@ -988,12 +966,12 @@ extern "C" dbsp_status dbsp_engine_create(dbsp_engine **out, dbsp_ctx *ctx,
     e->rank = rank;
     e->world = world;
     if (ctx) {
-        if (hipMalloc(&e->d_wm, 10 * sizeof(uint64_t)) == hipSuccess) {
-            e->d_bounds = e->d_wm + 4;
-            (void)hipMemset(e->d_wm, 0, 10 * sizeof(uint64_t));
-        } else {
-            e->d_wm = nullptr;
+        if (hipMalloc(&e->d_wm, 11 * sizeof(uint64_t)) != hipSuccess) {
+            delete e;
+            return DBSP_ERR_OOM;
         }
+        e->d_bounds = e->d_wm + 4;
+        (void)hipMemset(e->d_wm, 0, 11 * sizeof(uint64_t));
     }
     *out = e;
     return DBSP_OK;
@@ -1206,26 +1184,64 @@ static dbsp_status shard_exchange_pair(dbsp_ctx *c, DevBatch l0, DevBatch l1,
 }
 
 // flatmap events slice into up to two raw streams, then sort+consolidate
+static dbsp_status build_deltas_chain(dbsp_engine *e, const dbsp_event *d_ev,
+                                      int64_t n, DevBatch &rawA, DevBatch &rawB,
+                                      DevBatch &oA, DevBatch &oB);
+
 static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
                                 int64_t n, DevBatch &d0, DevBatch &d1,
                                 bool want_two) {
+    // Explicit-path deltas (sharded ranks, oversized ticks): the front still
+    // CHAINS — flatmap and the speculative fused sorts launch back-to-back
+    // and one sync reads counts + lengths; deltas over the fused capacity
+    // re-sort through the sized paths.  Only then does the key exchange run
+    // (it needs host lengths).
     dbsp_ctx *c = e->ctx;
-    DevBatch raw0, raw1;
-    TRY(alloc_batch(c, n > 0 ? n : 1, raw0, true));
-    TRY(alloc_batch(c, n > 0 ? n : 1, raw1, true));
-    int64_t n0 = 0, n1 = 0;
-    TRY(dbspk::flatmap_events(c->stream, d_ev, n, e->query, raw0.k, raw0.v,
-                              raw0.w, &n0, raw1.k, raw1.v, raw1.w, &n1));
-    raw0.n = n0;
-    raw1.n = n1;
-    if (want_two && n0 <= 8192 && n1 <= 8192) {
-        TRY(sort_two_small(c, raw0, raw1, d0, d1));
+    if (n > 16384) {
+        // big ticks: per-stream deltas would lose the sort speculation every
+        // time, so run the explicit front (host counts, sized sorts)
+
+        DevBatch raw0, raw1;
+        TRY(alloc_batch(c, n > 0 ? n : 1, raw0, true));
+        TRY(alloc_batch(c, n > 0 ? n : 1, raw1, true));
+        int64_t n0 = 0, n1 = 0;
+        TRY(dbspk::flatmap_events(c->stream, d_ev, n, e->query, raw0.k, raw0.v,
+                                  raw0.w, &n0, raw1.k, raw1.v, raw1.w, &n1));
+        raw0.n = n0;
+        raw1.n = n1;
+        if (want_two && n0 <= 8192 && n1 <= 8192) {
+            TRY(sort_two_small(c, raw0, raw1, d0, d1));
+        } else {
+            TRY(sort_consolidate_batch(c, raw0, d0));
+            if (want_two)
+                TRY(sort_consolidate_batch(c, raw1, d1));
+            else
+                free_batch(c, raw1);
+        }
     } else {
-        TRY(sort_consolidate_batch(c, raw0, d0));
-        if (want_two)
-            TRY(sort_consolidate_batch(c, raw1, d1));
-        else
-            free_batch(c, raw1);
+        DevBatch rawA, rawB, oA, oB;
+        TRY(build_deltas_chain(e, d_ev, n, rawA, rawB, oA, oB));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 16 * sizeof(int64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        if (c->h_len[10] < 0 || c->h_len[11] < 0) {
+            rawA.n = c->h_len[8];
+            rawB.n = c->h_len[9];
+            free_batch(c, oA);
+            free_batch(c, oB);
+            TRY(sort_consolidate_batch(c, rawA, d0));
+            if (want_two) {
+                TRY(sort_consolidate_batch(c, rawB, d1));
+            } else {
+                free_batch(c, rawB);
+            }
+        } else {
+            oA.n = c->h_len[10];
+            oB.n = c->h_len[11];
+            d0 = oA;
+            if (want_two) d1 = oB;
+            else free_batch(c, oB);
+        }
     }
     // worker sharding: co-locate keys across ranks (shard.rs:88)
     if (sharding_on(c)) {
@@ -2150,24 +2166,79 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         }
     }
     if (!front_done) {
+        // explicit deltas (sharded exchange / chain fallback), but the
+        // watermark and windows stay device-side: the tick's local max
+        // event-time reduces across ranks ON THE STREAM (q8.rs:63-65 as an
+        // RCCL allreduce feeding k_wm_update_g) and the ranges chain behind
+        // it — the same state words the chained path uses, so flipping
+        // between the modes never desyncs the watermark.
         TRY(build_deltas(e, d_ev, n, dPT, dAT, true));
-        // watermark over auctions (q8.rs:63-65)
-        uint64_t lk = 0;
-        bool has = false;
-        TRY(last_key(c, dAT, &lk, &has));
-        uint64_t gmax = has ? lk : 0;
-        TRY(allreduce_max_u64(c, &gmax));
-        if (gmax > 0) e->q8_wm = std::max(e->q8_wm, gmax - TUMBLE_MS);
-        uint64_t rounded = e->q8_wm - e->q8_wm % TUMBLE_MS;
-        uint64_t s1 = rounded >= TUMBLE_MS ? rounded - TUMBLE_MS : 0;
-        uint64_t e1 = rounded;
-        TRY(window_vs_spine(c, e->pt_int, dPT, e->q8_have_prev, e->q8_s0,
-                            e->q8_e0, s1, e1, wp_raw));
-        TRY(window_vs_spine(c, e->at_int, dAT, e->q8_have_prev, e->q8_s0,
-                            e->q8_e0, s1, e1, wa_raw));
-        e->q8_have_prev = true;
-        e->q8_s0 = s1;
-        e->q8_e0 = e1;
+        unsigned long long *gslot = e->d_wm + 10;
+        if (dAT.n > 0) {
+            HIP_CHECK_ST(hipMemcpyAsync(gslot, dAT.k + dAT.n - 1,
+                                        sizeof(uint64_t),
+                                        hipMemcpyDeviceToDevice, c->stream));
+        } else {
+            HIP_CHECK_ST(hipMemsetAsync(gslot, 0, sizeof(uint64_t),
+                                        c->stream));
+        }
+        if (sharding_on(c) &&
+            ncclAllReduce(gslot, gslot, 1, ncclUint64, ncclMax, c->comm,
+                          c->stream) != ncclSuccess)
+            return DBSP_ERR_INTERNAL;
+        TRY(dbspk::wm_update_g(c->stream, gslot, TUMBLE_MS, TUMBLE_MS,
+                               TUMBLE_MS, e->d_wm, e->d_bounds));
+        if ((int)e->pt_int.batches.size() > MAX_TRACE_BATCHES)
+            TRY(e->pt_int.consolidate_all(c));
+        if ((int)e->at_int.batches.size() > MAX_TRACE_BATCHES)
+            TRY(e->at_int.consolidate_all(c));
+        TraceArgs taP = trace_args_of(e->pt_int), taA = trace_args_of(e->at_int);
+        const int nregP = 3 * taP.nb + 1, nregA = 3 * taA.nb + 1;
+        int64_t *tableP = (int64_t *)arena_alloc(c, (size_t)nregP * 5 * 8 + 8);
+        int64_t *tableA = (int64_t *)arena_alloc(c, (size_t)nregA * 5 * 8 + 8);
+        if (tableP && tableA) {
+            {
+                ScopedTimer t(c, 4, 0.0);
+                TRY(dbspk::window_ranges_chain(c->stream, taP, dPT.k, dPT.n,
+                                               nullptr, e->d_bounds, tableP,
+                                               c->d_len + 12));
+                TRY(dbspk::window_ranges_chain(c->stream, taA, dAT.k, dAT.n,
+                                               nullptr, e->d_bounds, tableA,
+                                               c->d_len + 13));
+            }
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
+                                        16 * sizeof(int64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            if (c->h_len[12] > 0) {
+                DevBatch o;
+                TRY(alloc_batch(c, c->h_len[12], o, true));
+                TRY(dbspk::window_emit_multi(c->stream, taP, dPT.k, dPT.v,
+                                             dPT.w, tableP, nregP,
+                                             c->h_len[12], o.k, o.v, o.w));
+                wp_raw.push_back(o);
+            }
+            if (c->h_len[13] > 0) {
+                DevBatch o;
+                TRY(alloc_batch(c, c->h_len[13], o, true));
+                TRY(dbspk::window_emit_multi(c->stream, taA, dAT.k, dAT.v,
+                                             dAT.w, tableA, nregA,
+                                             c->h_len[13], o.k, o.v, o.w));
+                wa_raw.push_back(o);
+            }
+        } else {
+            // no arena: read the published bounds and run the generic path
+            HIP_CHECK_ST(hipMemcpyAsync(e->h_bounds, e->d_bounds,
+                                        6 * sizeof(uint64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            TRY(window_vs_spine(c, e->pt_int, dPT, e->h_bounds[4] != 0,
+                                e->h_bounds[0], e->h_bounds[1], e->h_bounds[2],
+                                e->h_bounds[3], wp_raw));
+            TRY(window_vs_spine(c, e->at_int, dAT, e->h_bounds[4] != 0,
+                                e->h_bounds[0], e->h_bounds[1], e->h_bounds[2],
+                                e->h_bounds[3], wa_raw));
+        }
     }
     if (sharding_on(c)) {
         TRY(spines_insert_pair(c, e->pt_int, dPT, e->at_int, dAT));
@@ -2183,8 +2254,7 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     free_batch(c, war);
     if (sharding_on(c)) {
         DevBatch t1, t2;
-        TRY(shard_exchange(c, dWP, t1));
-        TRY(shard_exchange(c, dWA, t2));
+        TRY(shard_exchange_pair(c, dWP, dWA, t1, t2));
         dWP = t1;
         dWA = t2;
     }
@@ -2421,22 +2491,59 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         }
     }
     if (!front_done) {
+        // explicit deltas (sharded exchange), device watermark: local max
+        // event-time reduced across ranks on the stream, bounds published
+        // device-side into the same state the chained path uses
         DevBatch dummy;
         TRY(build_deltas(e, d_ev, n, dBT, dummy, false));
-        uint64_t lk = 0;
-        bool has = false;
-        TRY(last_key(c, dBT, &lk, &has));
-        uint64_t gmax = has ? lk : 0;
-        TRY(allreduce_max_u64(c, &gmax));
-        if (gmax > 0) e->q5_wm = std::max(e->q5_wm, gmax - WM_LAG_MS);
-        uint64_t rounded = e->q5_wm - e->q5_wm % TUMBLE_MS;
-        uint64_t s1 = rounded >= WIDTH_MS ? rounded - WIDTH_MS : 0;
-        uint64_t e1 = rounded;
-        TRY(window_vs_spine(c, e->bt_int, dBT, e->q5_have_prev, e->q5_s0,
-                            e->q5_e0, s1, e1, wb_raw));
-        e->q5_have_prev = true;
-        e->q5_s0 = s1;
-        e->q5_e0 = e1;
+        unsigned long long *gslot = e->d_wm + 10;
+        if (dBT.n > 0) {
+            HIP_CHECK_ST(hipMemcpyAsync(gslot, dBT.k + dBT.n - 1,
+                                        sizeof(uint64_t),
+                                        hipMemcpyDeviceToDevice, c->stream));
+        } else {
+            HIP_CHECK_ST(hipMemsetAsync(gslot, 0, sizeof(uint64_t),
+                                        c->stream));
+        }
+        if (sharding_on(c) &&
+            ncclAllReduce(gslot, gslot, 1, ncclUint64, ncclMax, c->comm,
+                          c->stream) != ncclSuccess)
+            return DBSP_ERR_INTERNAL;
+        TRY(dbspk::wm_update_g(c->stream, gslot, WIDTH_MS, TUMBLE_MS,
+                               WM_LAG_MS, e->d_wm, e->d_bounds));
+        if ((int)e->bt_int.batches.size() > MAX_TRACE_BATCHES)
+            TRY(e->bt_int.consolidate_all(c));
+        TraceArgs taB = trace_args_of(e->bt_int);
+        const int nregB = 3 * taB.nb + 1;
+        int64_t *tableB = (int64_t *)arena_alloc(c, (size_t)nregB * 5 * 8 + 8);
+        if (tableB) {
+            {
+                ScopedTimer t(c, 4, 0.0);
+                TRY(dbspk::window_ranges_chain(c->stream, taB, dBT.k, dBT.n,
+                                               nullptr, e->d_bounds, tableB,
+                                               c->d_len + 7));
+            }
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 7, c->d_len + 7,
+                                        sizeof(int64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            if (c->h_len[7] > 0) {
+                DevBatch o;
+                TRY(alloc_batch(c, c->h_len[7], o, true));
+                TRY(dbspk::window_emit_multi(c->stream, taB, dBT.k, dBT.v,
+                                             dBT.w, tableB, nregB,
+                                             c->h_len[7], o.k, o.v, o.w));
+                wb_raw.push_back(o);
+            }
+        } else {
+            HIP_CHECK_ST(hipMemcpyAsync(e->h_bounds, e->d_bounds,
+                                        6 * sizeof(uint64_t),
+                                        hipMemcpyDeviceToHost, c->stream));
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            TRY(window_vs_spine(c, e->bt_int, dBT, e->h_bounds[4] != 0,
+                                e->h_bounds[0], e->h_bounds[1], e->h_bounds[2],
+                                e->h_bounds[3], wb_raw));
+        }
     }
     if (sharding_on(c)) TRY(e->bt_int.insert(c, dBT));
     // single-rank: bt_int is only read next tick — inserted in the tail
@@ -2496,8 +2603,7 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     TRY(map_sorted(c, dCounts, 3, dBC));    // (auction,count) -> (count,auction)
     if (sharding_on(c)) {
         DevBatch t1, t2;
-        TRY(shard_exchange(c, dMaxZ, t1));
-        TRY(shard_exchange(c, dBC, t2));
+        TRY(shard_exchange_pair(c, dMaxZ, dBC, t1, t2));
         dMaxZ = t1;
         dBC = t2;
     }

@@ -1918,6 +1918,31 @@ __device__ void wm_update_body(const uint64_t *ak, int64_t n, uint64_t width,
                                unsigned long long *state,
                                unsigned long long *bounds);
 
+// sharded variant: the tick's global max event-time arrives REDUCED in
+// *gmax_dev (local last key -> ncclAllReduce(max) on the stream), so the
+// whole watermark/window-bounds update stays device-side across ranks
+__global__ void k_wm_update_g(const unsigned long long *gmax_dev,
+                              uint64_t width, uint64_t tumble, uint64_t lag,
+                              unsigned long long *state,
+                              unsigned long long *bounds) {
+    const uint64_t g = *gmax_dev;
+    uint64_t wm = state[0];
+    if (g > 0) wm = max(wm, (uint64_t)(g - lag));
+    uint64_t rounded = wm - wm % tumble;
+    uint64_t s1 = rounded >= width ? rounded - width : 0;
+    uint64_t e1 = rounded;
+    bounds[0] = state[1];
+    bounds[1] = state[2];
+    bounds[2] = s1;
+    bounds[3] = e1;
+    bounds[4] = state[3];
+    bounds[5] = 0;
+    state[0] = wm;
+    state[1] = s1;
+    state[2] = e1;
+    state[3] = 1;
+}
+
 __global__ void k_wm_update_n(const uint64_t *ak, int64_t n, uint64_t width,
                               uint64_t tumble, uint64_t lag,
                               unsigned long long *state,
@@ -2817,6 +2842,15 @@ dbsp_status wm_update_n(hipStream_t s, const uint64_t *ak, int64_t n,
                         unsigned long long *state,
                         unsigned long long *bounds) {
     k_wm_update_n<<<1, 1, 0, s>>>(ak, n, width, tumble, lag, state, bounds);
+    return DBSP_OK;
+}
+
+dbsp_status wm_update_g(hipStream_t s, const unsigned long long *gmax_dev,
+                        uint64_t width, uint64_t tumble, uint64_t lag,
+                        unsigned long long *state,
+                        unsigned long long *bounds) {
+    k_wm_update_g<<<1, 1, 0, s>>>(gmax_dev, width, tumble, lag, state,
+                                  bounds);
     return DBSP_OK;
 }
 

@@ -109,6 +109,9 @@ dbsp_status wm_update(hipStream_t s, const uint64_t *ak, const int64_t *n_dev,
 dbsp_status wm_update_n(hipStream_t s, const uint64_t *ak, int64_t n,
                         uint64_t width, uint64_t tumble, uint64_t lag,
                         unsigned long long *state, unsigned long long *bounds);
+dbsp_status wm_update_g(hipStream_t s, const unsigned long long *gmax_dev,
+                        uint64_t width, uint64_t tumble, uint64_t lag,
+                        unsigned long long *state, unsigned long long *bounds);
 dbsp_status window_ranges_chain(hipStream_t s, const TraceArgs &t,
                                 const uint64_t *bk, int64_t bn,
                                 const int64_t *bn_dev,
