@@ -1104,10 +1104,9 @@ static dbsp_status shard_exchange_pair(dbsp_ctx *c, DevBatch l0, DevBatch l1,
     TRY(alloc_batch(c, l0.n > 0 ? l0.n : 1, p0, true));
     TRY(alloc_batch(c, l1.n > 0 ? l1.n : 1, p1, true));
     int64_t off0[65], off1[65];
-    TRY(dbspk::shard_rows(c->stream, l0.k, l0.v, l0.w, l0.n, world, p0.k, p0.v,
-                          p0.w, off0));
-    TRY(dbspk::shard_rows(c->stream, l1.k, l1.v, l1.w, l1.n, world, p1.k, p1.v,
-                          p1.w, off1));
+    TRY(dbspk::shard_rows_pair(c->stream, l0.k, l0.v, l0.w, l0.n, l1.k, l1.v,
+                               l1.w, l1.n, world, p0.k, p0.v, p0.w, p1.k,
+                               p1.v, p1.w, off0, off1));
     p0.n = l0.n;
     p1.n = l1.n;
     free_batch(c, l0);

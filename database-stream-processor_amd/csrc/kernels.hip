@@ -2903,6 +2903,56 @@ dbsp_status shard_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
     return DBSP_OK;
 }
 
+// pair variant: both deltas' histograms land in one readback, so the
+// exchange pays ONE partition sync instead of two
+dbsp_status shard_rows_pair(hipStream_t s, const uint64_t *k0,
+                            const uint64_t *v0, const int64_t *w0, int64_t n0,
+                            const uint64_t *k1, const uint64_t *v1,
+                            const int64_t *w1, int64_t n1, int nshards,
+                            uint64_t *ok0, uint64_t *ov0, int64_t *ow0,
+                            uint64_t *ok1, uint64_t *ov1, int64_t *ow1,
+                            int64_t *h_off0, int64_t *h_off1) {
+    uint64_t *hist;
+    HIP_CHECK(dbspk::cache_malloc((void **)&hist,
+                                  2 * (nshards + 1) * sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(hist, 0, 2 * (nshards + 1) * sizeof(uint64_t), s));
+    uint64_t *hist1 = hist + nshards + 1;
+    if (n0 > 0) k_shard_hist<<<grid_for(n0), BLK, 0, s>>>(k0, n0, nshards, hist);
+    if (n1 > 0)
+        k_shard_hist<<<grid_for(n1), BLK, 0, s>>>(k1, n1, nshards, hist1);
+    uint64_t h_hist[130];
+    HIP_CHECK(hipMemcpyAsync(h_hist, hist, sizeof(h_hist[0]) * (nshards + 1),
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipMemcpyAsync(h_hist + 65, hist1,
+                             sizeof(h_hist[0]) * (nshards + 1),
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    int64_t acc0 = 0, acc1 = 0;
+    int64_t cur[130];
+    for (int i = 0; i < nshards; i++) {
+        h_off0[i] = acc0;
+        h_off1[i] = acc1;
+        cur[i] = acc0;
+        cur[65 + i] = acc1;
+        acc0 += (int64_t)h_hist[i];
+        acc1 += (int64_t)h_hist[65 + i];
+    }
+    h_off0[nshards] = acc0;
+    h_off1[nshards] = acc1;
+    HIP_CHECK(hipMemcpyAsync(hist, cur, nshards * sizeof(int64_t),
+                             hipMemcpyHostToDevice, s));
+    HIP_CHECK(hipMemcpyAsync(hist1, cur + 65, nshards * sizeof(int64_t),
+                             hipMemcpyHostToDevice, s));
+    if (n0 > 0)
+        k_shard_scatter<<<grid_for(n0), BLK, 0, s>>>(k0, v0, w0, n0, nshards,
+                                                     hist, ok0, ov0, ow0);
+    if (n1 > 0)
+        k_shard_scatter<<<grid_for(n1), BLK, 0, s>>>(k1, v1, w1, n1, nshards,
+                                                     hist1, ok1, ov1, ow1);
+    HIP_CHECK(dbspk::cache_free(hist, s));
+    return DBSP_OK;
+}
+
 dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
                                  int64_t n, int query, uint64_t *k0,
                                  uint64_t *v0, int64_t *w0, uint64_t *k1,

@@ -121,6 +121,13 @@ dbsp_status minmax_rows_chain(hipStream_t s, const uint64_t *k,
                               const uint64_t *v, int64_t cap,
                               const int64_t *n_dev,
                               unsigned long long *mm_dev);
+dbsp_status shard_rows_pair(hipStream_t s, const uint64_t *k0,
+                            const uint64_t *v0, const int64_t *w0, int64_t n0,
+                            const uint64_t *k1, const uint64_t *v1,
+                            const int64_t *w1, int64_t n1, int nshards,
+                            uint64_t *ok0, uint64_t *ov0, int64_t *ow0,
+                            uint64_t *ok1, uint64_t *ov1, int64_t *ow1,
+                            int64_t *h_off0, int64_t *h_off1);
 dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
                                  int64_t n, int query, uint64_t *k0,
                                  uint64_t *v0, int64_t *w0, uint64_t *k1,
