@@ -815,22 +815,28 @@ static dbsp_status alltoallv_cols(dbsp_ctx *c, const DevBatch &send,
                                   int64_t *recv_counts) {
     int world = c->world;
     // exchange counts (device staging for RCCL)
-    int64_t *d_send_cnt, *d_recv_cnt;
+    // ONE allgather of every rank's send-count vector beats a grouped
+    // world^2 send/recv mesh on setup latency; rank p's row holds what p
+    // sends to each peer, so what WE receive from p is row[p][rank]
+    int64_t *d_send_cnt, *d_all_cnt;
     HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_send_cnt, world * 8, c->stream));
-    HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_recv_cnt, world * 8, c->stream));
+    HIP_CHECK_ST(
+        dbspk::cache_malloc((void **)&d_all_cnt, (size_t)world * world * 8,
+                            c->stream));
     HIP_CHECK_ST(hipMemcpyAsync(d_send_cnt, send_counts, world * 8,
                                 hipMemcpyHostToDevice, c->stream));
-    ncclGroupStart();
-    for (int r = 0; r < world; r++) {
-        ncclSend(d_send_cnt + r, 1, ncclInt64, r, c->comm, c->stream);
-        ncclRecv(d_recv_cnt + r, 1, ncclInt64, r, c->comm, c->stream);
-    }
-    ncclGroupEnd();
-    HIP_CHECK_ST(hipMemcpyAsync(recv_counts, d_recv_cnt, world * 8,
+    if (ncclAllGather(d_send_cnt, d_all_cnt, world, ncclInt64, c->comm,
+                      c->stream) != ncclSuccess)
+        return DBSP_ERR_INTERNAL;
+    int64_t h_all[64 * 64];
+    HIP_CHECK_ST(hipMemcpyAsync(h_all, d_all_cnt, (size_t)world * world * 8,
                                 hipMemcpyDeviceToHost, c->stream));
     HIP_CHECK_ST(hipStreamSynchronize(c->stream));
     int64_t recv_total = 0;
-    for (int r = 0; r < world; r++) recv_total += recv_counts[r];
+    for (int r = 0; r < world; r++) {
+        recv_counts[r] = h_all[(size_t)r * world + c->rank];
+        recv_total += recv_counts[r];
+    }
     TRY(alloc_batch(c, recv_total, recv));
     // data columns
     int64_t soff = 0, roff = 0;
@@ -852,7 +858,7 @@ static dbsp_status alltoallv_cols(dbsp_ctx *c, const DevBatch &send,
     }
     ncclGroupEnd();
     HIP_CHECK_ST(dbspk::cache_free(d_send_cnt, c->stream));
-    HIP_CHECK_ST(dbspk::cache_free(d_recv_cnt, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(d_all_cnt, c->stream));
     recv.n = recv_total;
     return DBSP_OK;
 }
@@ -1117,20 +1123,24 @@ static dbsp_status shard_exchange_pair(dbsp_ctx *c, DevBatch l0, DevBatch l1,
         send_cnt[2 * r] = off0[r + 1] - off0[r];
         send_cnt[2 * r + 1] = off1[r + 1] - off1[r];
     }
-    int64_t *d_snd, *d_rcv;
+    // one allgather of the 2-per-peer count vectors (see alltoallv_cols)
+    int64_t *d_snd, *d_all;
     HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_snd, 2 * world * 8, c->stream));
-    HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_rcv, 2 * world * 8, c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&d_all,
+                                     (size_t)world * 2 * world * 8, c->stream));
     HIP_CHECK_ST(hipMemcpyAsync(d_snd, send_cnt, 2 * world * 8,
                                 hipMemcpyHostToDevice, c->stream));
-    ncclGroupStart();
-    for (int r = 0; r < world; r++) {
-        ncclSend(d_snd + 2 * r, 2, ncclInt64, r, c->comm, c->stream);
-        ncclRecv(d_rcv + 2 * r, 2, ncclInt64, r, c->comm, c->stream);
-    }
-    ncclGroupEnd();
-    HIP_CHECK_ST(hipMemcpyAsync(recv_cnt, d_rcv, 2 * world * 8,
+    if (ncclAllGather(d_snd, d_all, 2 * world, ncclInt64, c->comm,
+                      c->stream) != ncclSuccess)
+        return DBSP_ERR_INTERNAL;
+    int64_t h_all[2 * 64 * 64];
+    HIP_CHECK_ST(hipMemcpyAsync(h_all, d_all, (size_t)world * 2 * world * 8,
                                 hipMemcpyDeviceToHost, c->stream));
     HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    for (int r = 0; r < world; r++) {
+        recv_cnt[2 * r] = h_all[(size_t)r * 2 * world + 2 * c->rank];
+        recv_cnt[2 * r + 1] = h_all[(size_t)r * 2 * world + 2 * c->rank + 1];
+    }
     int64_t tot0 = 0, tot1 = 0;
     for (int r = 0; r < world; r++) {
         tot0 += recv_cnt[2 * r];
@@ -1170,7 +1180,7 @@ static dbsp_status shard_exchange_pair(dbsp_ctx *c, DevBatch l0, DevBatch l1,
     }
     ncclGroupEnd();
     HIP_CHECK_ST(dbspk::cache_free(d_snd, c->stream));
-    HIP_CHECK_ST(dbspk::cache_free(d_rcv, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(d_all, c->stream));
     r0.n = tot0;
     r1.n = tot1;
     if (r0.n <= 8192 && r1.n <= 8192) {
