@@ -800,6 +800,9 @@ extern "C" dbsp_status dbsp_comm_unique_id(void *out128) {
 
 extern "C" dbsp_status dbsp_comm_init(dbsp_ctx *c, int rank, int world,
                                       const void *nccl_id) {
+    // host-side count/offset arrays are sized for one node's worth of ranks
+    if (world < 1 || world > 64 || rank < 0 || rank >= world)
+        return DBSP_ERR_INVALID;
     ncclUniqueId id;
     memcpy(&id, nccl_id, sizeof(id));
     if (ncclCommInitRank(&c->comm, world, id, rank) != ncclSuccess)
