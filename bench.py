@@ -171,6 +171,11 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    import torch
+    if torch.cuda.is_available():
+        # keep torch's notion of the current device aligned with the rank's
+        # GPU, so the bracketing torch.cuda.synchronize() syncs the right one
+        torch.cuda.set_device(local_rank)
 
     dist = None
     nccl_id = None
