@@ -195,3 +195,27 @@ def test_distinct_inc_parity(ctx):
     assert np.array_equal(got, exp)
     # empty trace and empty delta
     assert len(ctx.distinct_inc(np.empty(0, dtype=ROW_DT), [b1])) == 0
+
+
+def test_window_reference_vectors(ctx):
+    """dbsp_window replayed against the reference's own in-tree window tests
+    (time_series/window.rs:249-455 sliding/tumbling/shrinking, transcribed to
+    tests/golden/window_ops.json): per-tick bounds + input deltas must yield
+    the reference's exact output deltas, including left-shrinks and
+    jump-forward windows."""
+    from helpers import load_golden, rows_of
+    import numpy as np
+    g = load_golden("window_ops.json")
+    for case in g["cases"]:
+        trace = np.empty(0, dtype=ROW_DT)
+        have_prev = False
+        s0 = e0 = 0
+        for tick, (b, inp, exp) in enumerate(
+                zip(case["bounds"], case["inputs"], case["expected"])):
+            s1, e1 = b
+            batch = oracle.consolidate(rows_of([tuple(r) for r in inp]))
+            got = ctx.window(trace, batch, have_prev, s0, e0, s1, e1)
+            assert zset(got) == zset(rows_of([tuple(r) for r in exp])), (
+                f"{case['name']}: tick {tick}")
+            trace = oracle.merge(trace, batch)
+            have_prev, s0, e0 = True, s1, e1

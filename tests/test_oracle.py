@@ -311,3 +311,24 @@ def test_oracle_q3_tick_deltas_integrate_to_batch_result():
               for r in q_one.step(evs, cap=1 << 22)}
     q_one.close()
     assert integral == single
+
+
+def test_oracle_window_reference_vectors():
+    """The window operator replayed against the reference's own in-tree
+    tests (time_series/window.rs:249-455: sliding, tumbling, shrinking) —
+    per-tick bounds, input deltas, and expected output deltas transcribed
+    to tests/golden/window_ops.json."""
+    g = load_golden("window_ops.json")
+    for case in g["cases"]:
+        trace = np.empty(0, dtype=ROW_DT)
+        have_prev = False
+        s0 = e0 = 0
+        for tick, (b, inp, exp) in enumerate(
+                zip(case["bounds"], case["inputs"], case["expected"])):
+            s1, e1 = b
+            batch = oracle.consolidate(rows_of([tuple(r) for r in inp]))
+            got = oracle.window(trace, batch, have_prev, s0, e0, s1, e1)
+            assert zset(got) == zset(rows_of([tuple(r) for r in exp])), (
+                f"{case['name']}: tick {tick}")
+            trace = oracle.merge(trace, batch)
+            have_prev, s0, e0 = True, s1, e1
