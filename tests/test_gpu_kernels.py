@@ -219,3 +219,19 @@ def test_window_reference_vectors(ctx):
                 f"{case['name']}: tick {tick}")
             trace = oracle.merge(trace, batch)
             have_prev, s0, e0 = True, s1, e1
+
+
+def test_distinct_reference_vectors(ctx):
+    """dbsp_distinct_inc replayed against the reference's
+    distinct_indexed_test vectors (operator/distinct.rs:825-891)."""
+    from helpers import load_golden, rows_of
+    g = load_golden("distinct_indexed.json")
+    trace = np.empty(0, dtype=ROW_DT)
+    integral = np.empty(0, dtype=ROW_DT)
+    for tick, t in enumerate(g["ticks"]):
+        delta = oracle.consolidate(rows_of([tuple(r) for r in t["delta"]]))
+        out = ctx.distinct_inc(delta, [trace] if len(trace) else [])
+        integral = oracle.merge(integral, out)
+        assert zset(integral) == zset(rows_of(
+            [tuple(r) for r in t["integral"]])), f"tick {tick}"
+        trace = oracle.merge(trace, delta)

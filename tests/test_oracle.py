@@ -332,3 +332,19 @@ def test_oracle_window_reference_vectors():
                 f"{case['name']}: tick {tick}")
             trace = oracle.merge(trace, batch)
             have_prev, s0, e0 = True, s1, e1
+
+
+def test_oracle_distinct_reference_vectors():
+    """Incremental distinct replayed against the reference's own
+    distinct_indexed_test (operator/distinct.rs:825-891): the integral of the
+    emitted deltas must match the reference's per-tick distinct integrals."""
+    g = load_golden("distinct_indexed.json")
+    trace = np.empty(0, dtype=ROW_DT)
+    integral = np.empty(0, dtype=ROW_DT)
+    for tick, t in enumerate(g["ticks"]):
+        delta = oracle.consolidate(rows_of([tuple(r) for r in t["delta"]]))
+        out = oracle.distinct_inc(delta, trace)
+        integral = oracle.merge(integral, out)
+        assert zset(integral) == zset(rows_of(
+            [tuple(r) for r in t["integral"]])), f"tick {tick}"
+        trace = oracle.merge(trace, delta)
