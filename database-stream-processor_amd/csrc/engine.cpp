@@ -1402,7 +1402,9 @@ static dbsp_status window_vs_spine(dbsp_ctx *c, Spine &trace,
     int64_t *table = (int64_t *)arena_alloc(c, (size_t)nreg * 5 * 8 + 8);
     DevBatch tmp_table;
     if (!table) {
-        TRY(alloc_batch(c, nreg * 2, tmp_table));  // fallback scratch
+        // fallback scratch: the ranges kernel writes 5 int64 per region + 8
+        // bytes through .k, and alloc_batch's k column is n*8+8 bytes
+        TRY(alloc_batch(c, nreg * 5, tmp_table));
         table = (int64_t *)tmp_table.k;
     }
     TRY(dbspk::window_ranges_multi(c->stream, ta, batch.k, batch.n,
