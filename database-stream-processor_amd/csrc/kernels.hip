@@ -1359,53 +1359,83 @@ __global__ __launch_bounds__(MP_THREADS, 2) void k_mp_merge(
 
 typedef __attribute__((address_space(1))) unsigned long long gu64_t;
 
+// ---------------------------------------------------------------------------
+// single-pass merge v2.  One FUSED branchless walk per tile (the round-1
+// variant walked the tile twice: once to count, once to emit); the walk
+// captures per-step decisions (advance-a / equal-pair / keep) as mask bits in
+// one register and the emit replays them with pure LDS reads — no compares.
+// The decoupled lookback runs WAVE-wide: all 64 lanes of wave 0 load a
+// 64-entry predecessor window per round instead of lane 0 walking the chain
+// serially (the round-1 chain at ~1M tiles was the measured bottleneck of
+// the one-pass variant).  Tile = MP_TILE (1024) -> 24.6 KB LDS -> 6
+// blocks/CU, matching the two-pass emit's occupancy.
+// State layout: state[0] = ticket; state[1] = poison; state[2+vb] = granule
+//   granule = value<<2 | flag;  flag: 0 invalid, 1 aggregate, 2 prefix
+// ---------------------------------------------------------------------------
+
+#define OP_ITEMS 5  // ceil((MP_TILE+1)/MP_THREADS); a split adjustment can
+                    // grow a tile by one row past MP_TILE
+
 template <typename W>
-__global__ __launch_bounds__(MP_THREADS, 3) void k_mp_merge_onepass(
+__global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
     const uint64_t *bk, const uint64_t *bv, const W *bw, int64_t nb,
     const int64_t *pa, const int64_t *pb, unsigned long long *state,
     int64_t nblocks, uint64_t *ok, uint64_t *ov, W *ow) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     uint64_t *lk = (uint64_t *)smem;
-    uint64_t *lv = lk + (MP_TILE_OP + 2);
-    W *lw = (W *)(lv + (MP_TILE_OP + 2));
+    uint64_t *lv = lk + (MP_TILE + 2);
+    W *lw = (W *)(lv + (MP_TILE + 2));
     __shared__ uint32_t wt[MP_THREADS / WAVE + 1];
     __shared__ unsigned long long sh_vb;
     __shared__ unsigned long long sh_prefix;
     const int tid = threadIdx.x;
-    // ticket: virtual block id in launch order
+    // ticket: virtual block id in launch order (forward progress for lookback)
     if (tid == 0) sh_vb = atomicAdd(state, 1ull);
     __syncthreads();
     const int64_t vb = (int64_t)sh_vb;
     const int64_t pa0 = pa[vb], pa1 = pa[vb + 1];
     const int64_t pb0 = pb[vb], pb1 = pb[vb + 1];
-    const int64_t naL = pa1 - pa0, nbL = pb1 - pb0;
-    const int64_t totL = naL + nbL;
+    const int naL = (int)(pa1 - pa0), nbL = (int)(pb1 - pb0);
+    const int totL = naL + nbL;
     stage_run<true>(lk, lv, lw, 0, ak + pa0, av + pa0, aw + pa0, naL, tid);
     stage_run<true>(lk, lv, lw, naL, bk + pb0, bv + pb0, bw + pb0, nbL, tid);
     __syncthreads();
-    const int64_t items = (totL + MP_THREADS - 1) / MP_THREADS;
-    int64_t d0 = min((int64_t)tid * items, totL);
-    int64_t d1 = min(d0 + items, totL);
-    int64_t ai, bi, ae, be;
-    merge_path_lds(lk, lv, naL, nbL, d0, ai, bi);
-    adjust_split_lds(lk, lv, naL, nbL, ai, bi);
-    merge_path_lds(lk, lv, naL, nbL, d1, ae, be);
-    adjust_split_lds(lk, lv, naL, nbL, ae, be);
-    // count from LDS
+    const int items = (totL + MP_THREADS - 1) / MP_THREADS;
+    const int d0 = min(tid * items, totL);
+    const int d1 = min(d0 + items, totL);
+    int64_t ai64, bi64, ae64, be64;
+    merge_path_lds(lk, lv, naL, nbL, d0, ai64, bi64);
+    adjust_split_lds(lk, lv, naL, nbL, ai64, bi64);
+    merge_path_lds(lk, lv, naL, nbL, d1, ae64, be64);
+    adjust_split_lds(lk, lv, naL, nbL, ae64, be64);
+    const int ai = (int)ai64, bi = (int)bi64, ae = (int)ae64, be = (int)be64;
+    // fused walk: predicated (no branches), decisions captured as mask bits.
+    // Steps needed <= items: the range can hold items+1 input rows only when
+    // the d1 split adjustment pulled in the b half of an equal pair, and that
+    // pair is consumed in one step.
+    uint32_t m_take = 0, m_eq = 0, m_keep = 0;
     uint32_t cnt = 0;
     {
-        int64_t i = ai, j = bi;
-        while (i < ae || j < be) {
-            if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
-                if (lw[i] + lw[naL + j] != (W)0) cnt++;
-                i++; j++;
-            } else if (j >= be ||
-                       (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
-                cnt++; i++;
-            } else {
-                cnt++; j++;
-            }
+        int i = ai, j = bi;
+#pragma unroll
+        for (int t = 0; t < OP_ITEMS; t++) {
+            const bool a_ok = i < ae, b_ok = j < be;
+            const bool act = a_ok | b_ok;
+            // all indices stay inside the tile+2 LDS arrays even when a side
+            // is exhausted (i <= naL, naL + j <= totL <= MP_TILE + 1)
+            const uint64_t ka = lk[i], va = lv[i];
+            const uint64_t kb = lk[naL + j], vB = lv[naL + j];
+            const bool eq = a_ok & b_ok & row_eq(ka, va, kb, vB);
+            const bool take_a = a_ok & ((!b_ok) | row_lt(ka, va, kb, vB) | eq);
+            const W sum = (W)(lw[i] + lw[naL + j]);
+            const bool keep = act & ((!eq) | (sum != (W)0));
+            m_take |= (uint32_t)take_a << t;
+            m_eq |= (uint32_t)eq << t;
+            m_keep |= (uint32_t)(keep & act) << t;
+            cnt += keep & act;
+            i += (int)(act & (take_a | eq));
+            j += (int)(act & ((!take_a) | eq));
         }
     }
     // block-exclusive scan of thread counts
@@ -1432,61 +1462,90 @@ __global__ __launch_bounds__(MP_THREADS, 3) void k_mp_merge_onepass(
         thread_off = wt[tid / WAVE] + (v - cnt);
         block_cnt = wt[MP_THREADS / WAVE];
     }
-    // decoupled lookback (lane 0 of the block)
-    if (tid == 0) {
+    // wave-parallel decoupled lookback: wave 0 loads 64 predecessor granules
+    // per round; virtual blocks below 0 read as prefix 0 (chain terminator)
+    if (tid < WAVE) {
         gu64_t *g = (gu64_t *)(state + 2);
-        // publish aggregate
-        __hip_atomic_store(&g[vb], ((unsigned long long)block_cnt << 2) | 1ull,
-                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        unsigned long long prefix = 0;
+        if (tid == 0)
+            __hip_atomic_store(&g[vb],
+                               ((unsigned long long)block_cnt << 2) | 1ull,
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        unsigned long long running = 0;
+        int64_t wbase = vb - WAVE;
         unsigned spins = 0;
-        for (int64_t p = vb - 1; p >= 0;) {
-            unsigned long long e =
-                __hip_atomic_load(&g[p], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-            unsigned long long flag = e & 3ull;
-            if (flag == 2ull) {
-                prefix += e >> 2;
-                break;
-            } else if (flag == 1ull) {
-                prefix += e >> 2;
-                p--;
+        bool done = (vb == 0);
+        while (!done) {
+            const int64_t p = wbase + tid;
+            const unsigned long long e =
+                p >= 0 ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT)
+                       : 2ull;  // virtual predecessor: prefix 0
+            const unsigned flag = (unsigned)(e & 3ull);
+            const uint64_t pmask = __ballot(flag == 2u);
+            const uint64_t imask = __ballot(flag == 0u);
+            bool retry;
+            if (pmask != 0) {
+                const int hi = 63 - __clzll(pmask);  // newest prefix lane
+                retry = ((imask >> hi) >> 1) != 0;   // invalid above it
+                if (!retry) {
+                    unsigned long long c = tid >= hi ? (e >> 2) : 0;
+                    for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
+                    running += c;
+                    done = true;
+                }
             } else {
-                // Exponential backoff: with ~1k resident pollers a hot
-                // (64-cycle) poll loop throttles the L2/fabric enough to
-                // stall the very aggregates being waited for — observed as a
-                // bimodal ~30x collapse of the whole merge.
+                retry = imask != 0;
+                if (!retry) {
+                    unsigned long long c = e >> 2;
+                    for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
+                    running += c;
+                    wbase -= WAVE;
+                }
+            }
+            if (retry) {
+                // backoff: a hot poll loop throttles the very L2 traffic the
+                // awaited aggregates need (observed round 1 as a bimodal
+                // ~30x collapse of the whole merge)
                 ++spins;
                 if (spins < 4) __builtin_amdgcn_s_sleep(1);
                 else if (spins < 32) __builtin_amdgcn_s_sleep(16);
                 else __builtin_amdgcn_s_sleep(64);
                 if (spins > (1u << 22)) {  // bounded spin: poison, don't hang
-                    __hip_atomic_store((gu64_t *)(state + 1), 1ull,
-                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                    break;
+                    if (tid == 0)
+                        __hip_atomic_store((gu64_t *)(state + 1), 1ull,
+                                           __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT);
+                    done = true;
                 }
             }
         }
-        __hip_atomic_store(&g[vb],
-                           ((prefix + block_cnt) << 2) | 2ull,
-                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        sh_prefix = prefix;
+        if (tid == 0) {
+            __hip_atomic_store(&g[vb],
+                               ((running + block_cnt) << 2) | 2ull,
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            sh_prefix = running;
+        }
     }
     __syncthreads();
+    // emit: replay the captured decisions with pure LDS reads (no compares)
     uint64_t gpos = (uint64_t)sh_prefix + thread_off;
     {
-        int64_t i = ai, j = bi;
-        while (i < ae || j < be) {
-            if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
-                W sum = lw[i] + lw[naL + j];
-                if (sum != (W)0) { ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = sum; gpos++; }
-                i++; j++;
-            } else if (j >= be ||
-                       (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
-                ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = lw[i]; gpos++; i++;
-            } else {
-                ok[gpos] = lk[naL + j]; ov[gpos] = lv[naL + j]; ow[gpos] = lw[naL + j];
-                gpos++; j++;
+        int i = ai, j = bi;
+#pragma unroll
+        for (int t = 0; t < OP_ITEMS; t++) {
+            const bool take_a = (m_take >> t) & 1u;
+            const bool eq = (m_eq >> t) & 1u;
+            const bool keep = (m_keep >> t) & 1u;
+            const bool act = (i < ae) | (j < be);
+            const int sel = take_a ? i : naL + j;
+            if (keep) {
+                ok[gpos] = lk[sel];
+                ov[gpos] = lv[sel];
+                ow[gpos] = eq ? (W)(lw[i] + lw[naL + j]) : lw[sel];
+                gpos++;
             }
+            i += (int)(act & (take_a | eq));
+            j += (int)(act & ((!take_a) | eq));
         }
     }
 }
@@ -2369,16 +2428,16 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
         *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
         return DBSP_OK;
     }
-    // Default: robust two-pass (count -> scan -> emit).  The single-pass
-    // decoupled-lookback variant is ~2x faster at 1B rows when healthy but
-    // can collapse ~30x when the prefix chain serializes under fabric
-    // pressure (profiles/r01_merge_pmc.txt discussion) — opt in with
-    // DBSP_MERGE_ONEPASS=1 until that is fixed.
-    static const bool onepass = []() {
-        const char *e = getenv("DBSP_MERGE_ONEPASS");
+    // Default: single-pass fused-walk merge with wave-parallel decoupled
+    // lookback (reads each input byte once; see k_mp_merge_onepass).  The
+    // two-pass count->scan->emit pipeline remains as the fallback for a
+    // poisoned lookback (bounded spins exhausted) and as an opt-out
+    // (DBSP_MERGE_TWOPASS=1) for A/B measurement.
+    static const bool twopass = []() {
+        const char *e = getenv("DBSP_MERGE_TWOPASS");
         return e && e[0] == '1';
     }();
-    const int64_t tile = onepass ? MP_TILE_OP : MP_TILE;
+    const int64_t tile = MP_TILE;
     int64_t nblocks = ceil_div(total, tile);
     int64_t *pa, *pb;
     HIP_CHECK(dbspk::cache_malloc((void **)&pa, (nblocks + 1) * sizeof(int64_t), s));
@@ -2387,14 +2446,14 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
                                                          nblocks, tile, pa, pb);
     uint64_t *rk, *rv;
     W *rw;
-    if (onepass) {
+    if (!twopass) {
         unsigned long long *state;
         HIP_CHECK(dbspk::cache_malloc((void **)&state, (nblocks + 2) * sizeof(uint64_t), s));
         HIP_CHECK(hipMemsetAsync(state, 0, (nblocks + 2) * sizeof(uint64_t), s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rk, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rv, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rw, total * sizeof(W) + 8, s));
-        const size_t smem = 3 * (MP_TILE_OP + 2) * sizeof(uint64_t);
+        const size_t smem = 3 * (MP_TILE + 2) * sizeof(uint64_t);
         k_mp_merge_onepass<W><<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
             ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         unsigned long long h_state[2];
@@ -2403,17 +2462,18 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
         HIP_CHECK(hipMemcpyAsync(&h_state[1], state + 2 + (nblocks - 1),
                                  sizeof(uint64_t), hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipStreamSynchronize(s));
-        HIP_CHECK(dbspk::cache_free(pa, s));
-        HIP_CHECK(dbspk::cache_free(pb, s));
         HIP_CHECK(dbspk::cache_free(state, s));
-        if (h_state[0] != 0 || (h_state[1] & 3ull) != 2ull) {
-            (void)dbspk::cache_free(rk, s);
-            (void)dbspk::cache_free(rv, s);
-            (void)dbspk::cache_free(rw, s);
-            return DBSP_ERR_INTERNAL;
+        if (h_state[0] == 0 && (h_state[1] & 3ull) == 2ull) {
+            HIP_CHECK(dbspk::cache_free(pa, s));
+            HIP_CHECK(dbspk::cache_free(pb, s));
+            *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)(h_state[1] >> 2);
+            return DBSP_OK;
         }
-        *ok = rk; *ov = rv; *ow = rw; *out_n = (int64_t)(h_state[1] >> 2);
-        return DBSP_OK;
+        // poisoned lookback: free the optimistic buffers and fall through to
+        // the two-pass pipeline on the partitions already computed
+        (void)dbspk::cache_free(rk, s);
+        (void)dbspk::cache_free(rv, s);
+        (void)dbspk::cache_free(rw, s);
     }
     uint64_t *counts;
     HIP_CHECK(dbspk::cache_malloc((void **)&counts, (nblocks + 1) * sizeof(uint64_t), s));

@@ -348,3 +348,35 @@ def test_oracle_distinct_reference_vectors():
         assert zset(integral) == zset(rows_of(
             [tuple(r) for r in t["integral"]])), f"tick {tick}"
         trace = oracle.merge(trace, delta)
+
+
+def test_oracle_sharded_q3_unions_to_global():
+    """bench.py's all-cores cpu_baseline runs N key-sharded oracle workers
+    (persons by id, auctions by seller — the shard points of queries/q3.rs);
+    their outputs must union to the single-circuit result (the reference's
+    multi-worker equality claim, join.rs:1019-1033 / shard.rs:35-60)."""
+    import sys
+    from pathlib import Path
+    sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+    from bench import _shard_filter
+    from dbsp_amd import gen
+
+    evs = gen.generate(200_000, seed=7)
+    tick = 20_000
+    world = 4
+    single = oracle.Query(3)
+    shards = [oracle.Query(3) for _ in range(world)]
+    for lo in range(0, len(evs), tick):
+        chunk = evs[lo:lo + tick]
+        expect = zset(single.step(chunk, cap=1 << 22))
+        got = {}
+        for r in range(world):
+            for row in shards[r].step(_shard_filter(chunk, 3, world, r),
+                                      cap=1 << 22):
+                key = (int(row["k"]), int(row["v"]))
+                got[key] = got.get(key, 0) + int(row["w"])
+        got = {k: w for k, w in got.items() if w != 0}
+        assert got == expect, f"tick at {lo}"
+    single.close()
+    for q in shards:
+        q.close()
