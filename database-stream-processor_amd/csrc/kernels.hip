@@ -1373,19 +1373,23 @@ typedef __attribute__((address_space(1))) unsigned long long gu64_t;
 //   granule = value<<2 | flag;  flag: 0 invalid, 1 aggregate, 2 prefix
 // ---------------------------------------------------------------------------
 
-#define OP_ITEMS 5  // ceil((MP_TILE+1)/MP_THREADS); a split adjustment can
-                    // grow a tile by one row past MP_TILE
+// lookback window: each lane holds LB_PER_LANE consecutive granules, so one
+// round covers WAVE*LB_PER_LANE predecessors (256 at 4/lane) — the round-1
+// serial-lane walk resolved ~64/round and its latency dominated the kernel
+#define LB_PER_LANE 4
 
-template <typename W>
+template <typename W, int TILE>
 __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
     const uint64_t *bk, const uint64_t *bv, const W *bw, int64_t nb,
     const int64_t *pa, const int64_t *pb, unsigned long long *state,
     int64_t nblocks, uint64_t *ok, uint64_t *ov, W *ow) {
+    // steps per thread: a split adjustment can grow a tile one row past TILE
+    constexpr int OP_ITEMS = (TILE + MP_THREADS) / MP_THREADS;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     uint64_t *lk = (uint64_t *)smem;
-    uint64_t *lv = lk + (MP_TILE + 2);
-    W *lw = (W *)(lv + (MP_TILE + 2));
+    uint64_t *lv = lk + (TILE + 2);
+    W *lw = (W *)(lv + (TILE + 2));
     __shared__ uint32_t wt[MP_THREADS / WAVE + 1];
     __shared__ unsigned long long sh_vb;
     __shared__ unsigned long long sh_prefix;
@@ -1462,33 +1466,52 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
         thread_off = wt[tid / WAVE] + (v - cnt);
         block_cnt = wt[MP_THREADS / WAVE];
     }
-    // wave-parallel decoupled lookback: wave 0 loads 64 predecessor granules
-    // per round; virtual blocks below 0 read as prefix 0 (chain terminator)
+    // wave-parallel decoupled lookback: wave 0 loads WAVE*LB_PER_LANE (256)
+    // consecutive predecessor granules per round — lane t holds positions
+    // wbase + LB_PER_LANE*t + q; virtual blocks below 0 read as prefix 0
+    // (chain terminator)
     if (tid < WAVE) {
         gu64_t *g = (gu64_t *)(state + 2);
         if (tid == 0)
             __hip_atomic_store(&g[vb],
                                ((unsigned long long)block_cnt << 2) | 1ull,
                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        constexpr int LBW = WAVE * LB_PER_LANE;
         unsigned long long running = 0;
-        int64_t wbase = vb - WAVE;
+        int64_t wbase = vb - LBW;
         unsigned spins = 0;
         bool done = (vb == 0);
         while (!done) {
-            const int64_t p = wbase + tid;
-            const unsigned long long e =
-                p >= 0 ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
-                                           __HIP_MEMORY_SCOPE_AGENT)
-                       : 2ull;  // virtual predecessor: prefix 0
-            const unsigned flag = (unsigned)(e & 3ull);
-            const uint64_t pmask = __ballot(flag == 2u);
-            const uint64_t imask = __ballot(flag == 0u);
+            unsigned long long e[LB_PER_LANE];
+            unsigned l_f2 = 0, l_inv = 0;  // per-lane flag bitmasks (bit q)
+#pragma unroll
+            for (int q = 0; q < LB_PER_LANE; q++) {
+                const int64_t p = wbase + LB_PER_LANE * tid + q;
+                e[q] = p >= 0 ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
+                                                  __HIP_MEMORY_SCOPE_AGENT)
+                              : 2ull;  // virtual predecessor: prefix 0
+                const unsigned flag = (unsigned)(e[q] & 3ull);
+                l_f2 |= (unsigned)(flag == 2u) << q;
+                l_inv |= (unsigned)(flag == 0u) << q;
+            }
+            const uint64_t pmask = __ballot(l_f2 != 0);
+            const uint64_t imask = __ballot(l_inv != 0);
             bool retry;
             if (pmask != 0) {
-                const int hi = 63 - __clzll(pmask);  // newest prefix lane
-                retry = ((imask >> hi) >> 1) != 0;   // invalid above it
+                const int hi = 63 - __clzll(pmask);  // newest lane w/ prefix
+                const unsigned hi_f2 = __shfl(l_f2, hi, WAVE);
+                const unsigned hi_inv = __shfl(l_inv, hi, WAVE);
+                const int hi_q = 31 - __builtin_clz(hi_f2);  // newest prefix
+                // invalid strictly above the chosen prefix forces a retry
+                retry = (((imask >> hi) >> 1) != 0) ||
+                        ((hi_inv >> hi_q) >> 1) != 0;
                 if (!retry) {
-                    unsigned long long c = tid >= hi ? (e >> 2) : 0;
+                    unsigned long long c = 0;
+#pragma unroll
+                    for (int q = 0; q < LB_PER_LANE; q++) {
+                        const bool use = tid > hi || (tid == hi && q >= hi_q);
+                        c += use ? (e[q] >> 2) : 0;
+                    }
                     for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
                     running += c;
                     done = true;
@@ -1496,10 +1519,12 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
             } else {
                 retry = imask != 0;
                 if (!retry) {
-                    unsigned long long c = e >> 2;
+                    unsigned long long c = 0;
+#pragma unroll
+                    for (int q = 0; q < LB_PER_LANE; q++) c += e[q] >> 2;
                     for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
                     running += c;
-                    wbase -= WAVE;
+                    wbase -= LBW;
                 }
             }
             if (retry) {
@@ -2437,7 +2462,11 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
         const char *e = getenv("DBSP_MERGE_TWOPASS");
         return e && e[0] == '1';
     }();
-    const int64_t tile = MP_TILE;
+    static const int op_tile = []() {
+        const char *e = getenv("DBSP_MERGE_OP_TILE");
+        return (e && atoi(e) == 2048) ? 2048 : MP_TILE;
+    }();
+    const int64_t tile = twopass ? MP_TILE : op_tile;
     int64_t nblocks = ceil_div(total, tile);
     int64_t *pa, *pb;
     HIP_CHECK(dbspk::cache_malloc((void **)&pa, (nblocks + 1) * sizeof(int64_t), s));
@@ -2453,9 +2482,17 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
         HIP_CHECK(dbspk::cache_malloc((void **)&rk, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rv, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rw, total * sizeof(W) + 8, s));
-        const size_t smem = 3 * (MP_TILE + 2) * sizeof(uint64_t);
-        k_mp_merge_onepass<W><<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
-            ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
+        const size_t smem = 3 * (tile + 2) * sizeof(uint64_t);
+        if (tile == 2048)
+            k_mp_merge_onepass<W, 2048>
+                <<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
+                    ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks,
+                    rk, rv, rw);
+        else
+            k_mp_merge_onepass<W, MP_TILE>
+                <<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
+                    ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks,
+                    rk, rv, rw);
         unsigned long long h_state[2];
         HIP_CHECK(hipMemcpyAsync(&h_state[0], state + 1, sizeof(uint64_t),
                                  hipMemcpyDeviceToHost, s));
