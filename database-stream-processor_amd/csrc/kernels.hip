@@ -1374,11 +1374,9 @@ typedef __attribute__((address_space(1))) unsigned long long gu64_t;
 // ---------------------------------------------------------------------------
 
 // lookback window: each lane holds LB_PER_LANE consecutive granules, so one
-// round covers WAVE*LB_PER_LANE predecessors (256 at 4/lane) — the round-1
-// serial-lane walk resolved ~64/round and its latency dominated the kernel
-#define LB_PER_LANE 4
-
-template <typename W, int TILE>
+// round covers WAVE*LB_PER_LANE predecessors — the round-1 serial-lane walk
+// resolved ~64/round and its latency dominated the kernel
+template <typename W, int TILE, int LB_PER_LANE = 1>
 __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
     const uint64_t *bk, const uint64_t *bv, const W *bw, int64_t nb,
@@ -1528,14 +1526,16 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
                 }
             }
             if (retry) {
-                // backoff: a hot poll loop throttles the very L2 traffic the
-                // awaited aggregates need (observed round 1 as a bimodal
-                // ~30x collapse of the whole merge)
+                // light backoff: near-predecessor aggregates land within a
+                // fraction of a block's own work phase, so long sleeps (the
+                // round-1 s_sleep(64) tier, ~2 us) cost more latency than the
+                // polls they save; wave-windowed polls are already 1/WAVE the
+                // per-granule poll rate of the round-1 serial walk that
+                // caused the fabric-throttling collapse
                 ++spins;
-                if (spins < 4) __builtin_amdgcn_s_sleep(1);
-                else if (spins < 32) __builtin_amdgcn_s_sleep(16);
-                else __builtin_amdgcn_s_sleep(64);
-                if (spins > (1u << 22)) {  // bounded spin: poison, don't hang
+                if (spins < 8) __builtin_amdgcn_s_sleep(1);
+                else __builtin_amdgcn_s_sleep(4);
+                if (spins > (1u << 23)) {  // bounded spin: poison, don't hang
                     if (tid == 0)
                         __hip_atomic_store((gu64_t *)(state + 1), 1ull,
                                            __ATOMIC_RELAXED,
@@ -2482,17 +2482,24 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
         HIP_CHECK(dbspk::cache_malloc((void **)&rk, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rv, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rw, total * sizeof(W) + 8, s));
+        static const int lbw = []() {
+            const char *e = getenv("DBSP_MERGE_LB");
+            return (e && atoi(e) == 4) ? 4 : 1;
+        }();
         const size_t smem = 3 * (tile + 2) * sizeof(uint64_t);
-        if (tile == 2048)
-            k_mp_merge_onepass<W, 2048>
-                <<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
-                    ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks,
-                    rk, rv, rw);
+        const dim3 g((uint32_t)nblocks);
+        if (tile == 2048 && lbw == 4)
+            k_mp_merge_onepass<W, 2048, 4><<<g, MP_THREADS, smem, s>>>(
+                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
+        else if (tile == 2048)
+            k_mp_merge_onepass<W, 2048, 1><<<g, MP_THREADS, smem, s>>>(
+                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
+        else if (lbw == 4)
+            k_mp_merge_onepass<W, MP_TILE, 4><<<g, MP_THREADS, smem, s>>>(
+                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else
-            k_mp_merge_onepass<W, MP_TILE>
-                <<<dim3((uint32_t)nblocks), MP_THREADS, smem, s>>>(
-                    ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks,
-                    rk, rv, rw);
+            k_mp_merge_onepass<W, MP_TILE, 1><<<g, MP_THREADS, smem, s>>>(
+                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         unsigned long long h_state[2];
         HIP_CHECK(hipMemcpyAsync(&h_state[0], state + 1, sizeof(uint64_t),
                                  hipMemcpyDeviceToHost, s));
