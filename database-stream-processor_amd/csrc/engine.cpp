@@ -442,15 +442,52 @@ static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
     return DBSP_OK;
 }
 
+// f64-weighted twin of merge_batches (config C5's weighted integral spine;
+// position-fixed reduction orders, algebra/floats.rs:24 zero elimination)
+static dbsp_status merge_batches_f64(dbsp_ctx *c, const DevBatch &a,
+                                     const DevBatch &b, DevBatch &out) {
+    ScopedTimer t(c, 1, (double)(a.n + b.n) * 48.0);
+    if (a.n + b.n <= 32768) {
+        DevBatch res;
+        TRY(alloc_batch(c, a.n + b.n, res));
+        MergeArgs ma{};
+        ma.np = 1;
+        ma.ak[0] = a.k; ma.av[0] = a.v; ma.aw[0] = a.w; ma.na[0] = a.n;
+        ma.bk[0] = b.k; ma.bv[0] = b.v; ma.bw[0] = b.w; ma.nb[0] = b.n;
+        ma.ok[0] = res.k; ma.ov[0] = res.v; ma.ow[0] = res.w;
+        ma.d_len = c->d_len;
+        TRY(dbspk::merge_small_batch_f64(c->stream, ma));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, sizeof(int64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        res.n = *c->h_len;
+        out = res;
+        return DBSP_OK;
+    }
+    double *rw;
+    TRY(dbspk::merge_rows_f64(c->stream, a.k, a.v, (const double *)a.w, a.n,
+                              b.k, b.v, (const double *)b.w, b.n, &out.k,
+                              &out.v, &rw, &out.n));
+    out.w = (int64_t *)rw;
+    return DBSP_OK;
+}
+
 // Spine: stack of consolidated batches; invariant size[i] >= 2*size[i+1]
 // (power-of-two leveling, spine_fueled.rs:107-119; merges run to completion)
 struct Spine {
     std::vector<DevBatch> batches;  // largest first
+    bool wf64 = false;  // f64-weighted batches (C5's weighted integral)
 
     int64_t total() const {
         int64_t t = 0;
         for (auto &b : batches) t += b.n;
         return t;
+    }
+
+    dbsp_status merge2(dbsp_ctx *c, const DevBatch &a, const DevBatch &b,
+                       DevBatch &out) const {
+        return wf64 ? merge_batches_f64(c, a, b, out)
+                    : merge_batches(c, a, b, out);
     }
 
     dbsp_status insert(dbsp_ctx *c, DevBatch b) {
@@ -464,7 +501,7 @@ struct Spine {
             DevBatch &below = batches[batches.size() - 2];
             if (top.n * 2 < below.n) break;  // geometric invariant holds
             DevBatch merged;
-            TRY(merge_batches(c, below, top, merged));
+            TRY(merge2(c, below, top, merged));
             free_batch(c, top);
             free_batch(c, below);
             batches.pop_back();
@@ -485,7 +522,7 @@ struct Spine {
             DevBatch below = batches.back();
             batches.pop_back();
             DevBatch merged;
-            TRY(merge_batches(c, below, top, merged));
+            TRY(merge2(c, below, top, merged));
             free_batch(c, top);
             free_batch(c, below);
             if (merged.n > 0) batches.push_back(merged);
@@ -1545,6 +1582,72 @@ static dbsp_status agg_linear_spine(dbsp_ctx *c, const DevBatch &delta,
                 if (o.n > 0) outs.push_back(o);
                 else free_batch(c, o);
             }
+            free_batch(c, tmp);
+        }
+    }
+    HIP_CHECK_ST(dbspk::cache_free(acc, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(keys, c->stream));
+    TRY(finalize_raw(c, outs, out));
+    return DBSP_OK;
+}
+
+// f64 twin of agg_linear_spine: per-key f64 sums across the weighted-integral
+// spine (aggregate_linear's sum, aggregate/mod.rs:297-323; batch-order
+// accumulation, tolerance 2 ulp * depth), retractions against the
+// i64-weighted output trace exactly as the i64 path
+static dbsp_status agg_linear_spine_f64(dbsp_ctx *c, const DevBatch &delta,
+                                        const Spine &in_trace,
+                                        Spine &out_trace, DevBatch &out) {
+    if (delta.n == 0) {
+        out = DevBatch{};
+        return DBSP_OK;
+    }
+    ScopedTimer timer(c, 3, 0.0);
+    uint64_t *keys = nullptr;
+    int64_t nk = 0;
+    TRY(dbspk::unique_keys(c->stream, delta.k, delta.n, &keys, &nk));
+    double *acc;
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&acc, nk * 8 + 8, c->stream));
+    HIP_CHECK_ST(hipMemsetAsync(acc, 0, nk * 8, c->stream));
+    for (auto &b : in_trace.batches)
+        TRY(dbspk::agg_sum_batch_f64(c->stream, keys, nk, b.k,
+                                     (const double *)b.w, b.n, acc));
+    std::vector<DevBatch> outs;
+    DevBatch ins;
+    TRY(alloc_batch(c, nk, ins, true));
+    int64_t n_ins = 0;
+    TRY(dbspk::emit_nonzero_f64(c->stream, keys, acc, nk, ins.k, ins.v, ins.w,
+                                &n_ins));
+    ins.n = n_ins;
+    if (ins.n > 0) outs.push_back(ins);
+    // retractions: probe the (i64-weighted) output trace with (key, _, -1)
+    // rows under proj (k, v2) — the upsert contract (upsert.rs:180-195)
+    if (!out_trace.batches.empty()) {
+        if ((int)out_trace.batches.size() > MAX_TRACE_BATCHES)
+            TRY(out_trace.consolidate_all(c));
+        TraceArgs t{};
+        for (auto &b : out_trace.batches) {
+            if (b.n == 0) continue;
+            t.k[t.nb] = b.k; t.v[t.nb] = b.v; t.w[t.nb] = b.w;
+            t.n[t.nb] = b.n; t.nb++;
+        }
+        if (t.nb > 0) {
+            uint64_t *dv = (uint64_t *)arena_alloc(c, (size_t)nk * 8 + 8);
+            int64_t *dw = (int64_t *)arena_alloc(c, (size_t)nk * 8 + 8);
+            DevBatch tmp;
+            if (!dv || !dw) {
+                TRY(alloc_batch(c, nk, tmp));
+                dv = tmp.k;
+                dw = (int64_t *)tmp.v;
+            }
+            HIP_CHECK_ST(hipMemsetAsync(dv, 0, nk * 8, c->stream));
+            HIP_CHECK_ST(hipMemsetAsync(dw, 0xFF, nk * 8, c->stream));  // -1
+            DevBatch o;
+            TRY(dbspk::join_spine_rows(c->stream, keys, dv, dw, nk, t,
+                                       DBSP_PROJ_HI_K_LO_V2, 0, &o.k, &o.v,
+                                       &o.w, &o.n));
+            if (o.n > 0) outs.push_back(o);
+            else free_batch(c, o);
             free_batch(c, tmp);
         }
     }

@@ -1376,7 +1376,7 @@ typedef __attribute__((address_space(1))) unsigned long long gu64_t;
 // lookback window: each lane holds LB_PER_LANE consecutive granules, so one
 // round covers WAVE*LB_PER_LANE predecessors — the round-1 serial-lane walk
 // resolved ~64/round and its latency dominated the kernel
-template <typename W, int TILE, int LB_PER_LANE = 1>
+template <typename W, int TILE, int LB_PER_LANE = 1, bool SKIP_LB = false>
 __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
     const uint64_t *bk, const uint64_t *bv, const W *bw, int64_t nb,
@@ -1468,7 +1468,18 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     // consecutive predecessor granules per round — lane t holds positions
     // wbase + LB_PER_LANE*t + q; virtual blocks below 0 read as prefix 0
     // (chain terminator)
-    if (tid < WAVE) {
+    if (SKIP_LB) {
+        // timing diagnostic ONLY (DBSP_MERGE_NOLB=1): measures the kernel
+        // without the lookback protocol; outputs land at uncompacted per-tile
+        // offsets (store pattern representative, results wrong by design)
+        if (tid == 0) {
+            gu64_t *g = (gu64_t *)(state + 2);
+            __hip_atomic_store(&g[vb],
+                               ((unsigned long long)block_cnt << 2) | 2ull,
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            sh_prefix = (unsigned long long)(vb * TILE);
+        }
+    } else if (tid < WAVE) {
         gu64_t *g = (gu64_t *)(state + 2);
         if (tid == 0)
             __hip_atomic_store(&g[vb],
@@ -2343,6 +2354,48 @@ void fill_u64(hipStream_t s, uint64_t *p, uint64_t v, int64_t n) {
     k_fill_u64<<<grid_for(n), BLK, 0, s>>>(p, v, n);
 }
 
+// ---------------------------------------------------------------------------
+// C5 synthetic operand generator (BASELINE configs[4]: 1B-row OrdIndexedZSet
+// x 10M-row delta).  Counter-based splitmix64 keeps keys strictly increasing
+// by construction (k_i = stride*i + h(i) % jitter, jitter < stride), so the
+// generated batch is sorted-unique with no device scan or sort — the trace
+// never stages through the host (24 GB at 1B rows).
+// ---------------------------------------------------------------------------
+
+__device__ __host__ inline uint64_t c5_mix(uint64_t x) {
+    uint64_t z = x + 0x9E3779B97F4A7C15ull;
+    z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+    z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+    return z ^ (z >> 31);
+}
+
+__global__ void k_c5_gen(int64_t n, uint64_t stride, uint64_t jitter,
+                         uint64_t seed, int val_mode, uint64_t *k, uint64_t *v,
+                         int64_t *w) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * (int64_t)blockDim.x) {
+        const uint64_t h = c5_mix(seed + (uint64_t)i * 0x9E3779B97F4A7C15ull);
+        k[i] = stride * (uint64_t)i + (jitter ? h % jitter : 0);
+        if (val_mode == 0) {
+            // uniform f64 in [0,1) as bit pattern (f64 vals)
+            const double u = (double)(c5_mix(h) >> 11) * 0x1.0p-53;
+            v[i] = __double_as_longlong(u);
+        } else {
+            v[i] = 0;
+        }
+        w[i] = 1;
+    }
+}
+
+dbsp_status c5_gen_rows(hipStream_t s, int64_t n, uint64_t stride,
+                        uint64_t jitter, uint64_t seed, int val_mode,
+                        uint64_t *k, uint64_t *v, int64_t *w) {
+    if (n <= 0) return DBSP_OK;
+    k_c5_gen<<<grid_for(n), BLK, 0, s>>>(n, stride, jitter, seed, val_mode, k,
+                                         v, w);
+    return DBSP_OK;
+}
+
 // sort rows by (k major, v minor); ping-pong scratch must hold n rows
 // (kk2/vv2/ww2).  Skips byte passes above the significant bytes of max(k)/max(v).
 dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
@@ -2486,9 +2539,19 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
             const char *e = getenv("DBSP_MERGE_LB");
             return (e && atoi(e) == 4) ? 4 : 1;
         }();
+        static const bool nolb = []() {  // timing diagnostic (wrong results)
+            const char *e = getenv("DBSP_MERGE_NOLB");
+            return e && e[0] == '1';
+        }();
         const size_t smem = 3 * (tile + 2) * sizeof(uint64_t);
         const dim3 g((uint32_t)nblocks);
-        if (tile == 2048 && lbw == 4)
+        if (nolb && tile == 2048)
+            k_mp_merge_onepass<W, 2048, 1, true><<<g, MP_THREADS, smem, s>>>(
+                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
+        else if (nolb)
+            k_mp_merge_onepass<W, MP_TILE, 1, true><<<g, MP_THREADS, smem, s>>>(
+                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
+        else if (tile == 2048 && lbw == 4)
             k_mp_merge_onepass<W, 2048, 4><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else if (tile == 2048)

@@ -283,6 +283,13 @@ dbsp_status emit_nonzero_f64(hipStream_t s, const uint64_t *keys,
                              const double *acc, int64_t nd, uint64_t *ok,
                              uint64_t *ov, int64_t *ow, int64_t *h_count);
 
+// C5 synthetic operands: sorted-unique rows by construction
+// (k_i = stride*i + h(i)%jitter, jitter < stride; val_mode 0 = uniform f64
+// bits, 1 = zero)
+dbsp_status c5_gen_rows(hipStream_t s, int64_t n, uint64_t stride,
+                        uint64_t jitter, uint64_t seed, int val_mode,
+                        uint64_t *k, uint64_t *v, int64_t *w);
+
 uint64_t host_xxh3_u64(uint64_t key, uint64_t seed);
 
 }  // namespace dbspk
