@@ -963,6 +963,14 @@ struct dbsp_engine {
     // q5 state
     Spine bt_int, wb_int, counts_int, bc_int;
     DevBatch maxin_int, maxout_int, maxz_int;
+    // C5 state (query 100; BASELINE configs[4]: 1B-row indexed trace x
+    // 10M-row delta incremental join + f64 sum aggregate)
+    Spine c5_trace;   // (k, f64-bits val, +1 i64) join-side trace
+    Spine c5_wint;    // f64-WEIGHTED integral of the weighed join stream
+    Spine c5_out;     // aggregate output trace (k -> f64 sum bits, i64 w)
+    int64_t c5_n_delta = 0;
+    uint64_t c5_seed = 0;
+    uint64_t c5_delta_stride = 0;
 
     // chained-tick device watermark state: {wm, s0, e0, have_prev} and the
     // published bounds {s0,e0,s1,e1,have_prev,err} (q5/q8 single-rank path)
@@ -1003,7 +1011,7 @@ struct dbsp_engine {
 
 extern "C" dbsp_status dbsp_engine_create(dbsp_engine **out, dbsp_ctx *ctx,
                                           int query, int rank, int world) {
-    if (query != 0 && query != 3 && query != 5 && query != 8)
+    if (query != 0 && query != 3 && query != 5 && query != 8 && query != 100)
         return DBSP_ERR_INVALID;
 
     dbsp_engine *e = new dbsp_engine();
@@ -1031,7 +1039,7 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
     dbsp_ctx *c = e->ctx;
     for (Spine *s : {&e->a_int, &e->p_int, &e->pt_int, &e->at_int, &e->wp_int,
                      &e->wa_int, &e->bt_int, &e->wb_int, &e->counts_int,
-                     &e->bc_int})
+                     &e->bc_int, &e->c5_trace, &e->c5_wint, &e->c5_out})
         s->clear(c);
     free_batch(c, e->maxin_int);
     free_batch(c, e->maxout_int);

@@ -1373,10 +1373,21 @@ typedef __attribute__((address_space(1))) unsigned long long gu64_t;
 //   granule = value<<2 | flag;  flag: 0 invalid, 1 aggregate, 2 prefix
 // ---------------------------------------------------------------------------
 
-// lookback window: each lane holds LB_PER_LANE consecutive granules, so one
-// round covers WAVE*LB_PER_LANE predecessors — the round-1 serial-lane walk
-// resolved ~64/round and its latency dominated the kernel
-template <typename W, int TILE, int LB_PER_LANE = 1, bool SKIP_LB = false>
+// Hierarchical lookback state (nblocks tiles, groups of LB_GROUP tiles):
+//   state[0] = ticket; state[1] = poison
+//   g[0..nblocks)      individual granules  (value<<2 | flag)
+//   gsum[0..ngroups)   running sum of the group's published aggregates
+//   gcnt[0..ngroups)   how many of the group's tiles have published
+//   gpre[0..ngroups)   group prefix granule (flag2 set by the group's last
+//                      tile with its inclusive prefix)
+// A tile walks individuals only within its own group (<= LB_GROUP/WAVE
+// rounds), then one wave-window over group entries — a complete group
+// (gcnt == group size) contributes its whole sum in ONE load, so the walk
+// no longer scales with the resident-block window (the measured 8.5 ms
+// flat-lookback overhead at ~1M tiles).
+#define LB_GROUP 256
+
+template <typename W, int TILE, bool SKIP_LB = false>
 __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
     const uint64_t *bk, const uint64_t *bv, const W *bw, int64_t nb,
@@ -1384,10 +1395,13 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     int64_t nblocks, uint64_t *ok, uint64_t *ov, W *ow) {
     // steps per thread: a split adjustment can grow a tile one row past TILE
     constexpr int OP_ITEMS = (TILE + MP_THREADS) / MP_THREADS;
+    // LDS stages keys+vals only (the search/compare operands); weights are
+    // read straight from global — touched once per consumed row in
+    // thread-contiguous order, they stream through L2, and the slimmer LDS
+    // footprint (16.4 KB at TILE=1024) lifts occupancy to 8 blocks/CU
     extern __shared__ __attribute__((aligned(16))) char smem[];
     uint64_t *lk = (uint64_t *)smem;
     uint64_t *lv = lk + (TILE + 2);
-    W *lw = (W *)(lv + (TILE + 2));
     __shared__ uint32_t wt[MP_THREADS / WAVE + 1];
     __shared__ unsigned long long sh_vb;
     __shared__ unsigned long long sh_prefix;
@@ -1400,8 +1414,12 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     const int64_t pb0 = pb[vb], pb1 = pb[vb + 1];
     const int naL = (int)(pa1 - pa0), nbL = (int)(pb1 - pb0);
     const int totL = naL + nbL;
-    stage_run<true>(lk, lv, lw, 0, ak + pa0, av + pa0, aw + pa0, naL, tid);
-    stage_run<true>(lk, lv, lw, naL, bk + pb0, bv + pb0, bw + pb0, nbL, tid);
+    const W *Aw = aw + pa0;
+    const W *Bw = bw + pb0;
+    stage_run<false, W>(lk, lv, nullptr, 0, ak + pa0, av + pa0, nullptr, naL,
+                        tid);
+    stage_run<false, W>(lk, lv, nullptr, naL, bk + pb0, bv + pb0, nullptr, nbL,
+                        tid);
     __syncthreads();
     const int items = (totL + MP_THREADS - 1) / MP_THREADS;
     const int d0 = min(tid * items, totL);
@@ -1425,13 +1443,13 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
             const bool a_ok = i < ae, b_ok = j < be;
             const bool act = a_ok | b_ok;
             // all indices stay inside the tile+2 LDS arrays even when a side
-            // is exhausted (i <= naL, naL + j <= totL <= MP_TILE + 1)
+            // is exhausted (i <= naL, naL + j <= totL <= TILE + 1)
             const uint64_t ka = lk[i], va = lv[i];
             const uint64_t kb = lk[naL + j], vB = lv[naL + j];
             const bool eq = a_ok & b_ok & row_eq(ka, va, kb, vB);
             const bool take_a = a_ok & ((!b_ok) | row_lt(ka, va, kb, vB) | eq);
-            const W sum = (W)(lw[i] + lw[naL + j]);
-            const bool keep = act & ((!eq) | (sum != (W)0));
+            bool keep = act;
+            if (eq) keep = (W)(Aw[i] + Bw[j]) != (W)0;  // rare: global reads
             m_take |= (uint32_t)take_a << t;
             m_eq |= (uint32_t)eq << t;
             m_keep |= (uint32_t)(keep & act) << t;
@@ -1464,85 +1482,76 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
         thread_off = wt[tid / WAVE] + (v - cnt);
         block_cnt = wt[MP_THREADS / WAVE];
     }
-    // wave-parallel decoupled lookback: wave 0 loads WAVE*LB_PER_LANE (256)
-    // consecutive predecessor granules per round — lane t holds positions
-    // wbase + LB_PER_LANE*t + q; virtual blocks below 0 read as prefix 0
-    // (chain terminator)
+    // hierarchical decoupled lookback (wave 0)
+    const int64_t ngroups = (nblocks + LB_GROUP - 1) / LB_GROUP;
+    gu64_t *g = (gu64_t *)(state + 2);
+    gu64_t *gsum = g + nblocks;
+    gu64_t *gcnt = gsum + ngroups;
+    gu64_t *gpre = gcnt + ngroups;
+    const int64_t grp = vb / LB_GROUP;
     if (SKIP_LB) {
         // timing diagnostic ONLY (DBSP_MERGE_NOLB=1): measures the kernel
         // without the lookback protocol; outputs land at uncompacted per-tile
         // offsets (store pattern representative, results wrong by design)
         if (tid == 0) {
-            gu64_t *g = (gu64_t *)(state + 2);
             __hip_atomic_store(&g[vb],
                                ((unsigned long long)block_cnt << 2) | 2ull,
                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
             sh_prefix = (unsigned long long)(vb * TILE);
         }
     } else if (tid < WAVE) {
-        gu64_t *g = (gu64_t *)(state + 2);
-        if (tid == 0)
+        if (tid == 0) {
+            // publish the individual aggregate, fold it into the group sum,
+            // and bump the group's publish count LAST (release, so a reader
+            // that sees the count complete also sees every sum contribution)
             __hip_atomic_store(&g[vb],
                                ((unsigned long long)block_cnt << 2) | 1ull,
                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        constexpr int LBW = WAVE * LB_PER_LANE;
+            __hip_atomic_fetch_add(&gsum[grp], (unsigned long long)block_cnt,
+                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            __hip_atomic_fetch_add(&gcnt[grp], 1ull, __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+        }
         unsigned long long running = 0;
-        int64_t wbase = vb - LBW;
         unsigned spins = 0;
         bool done = (vb == 0);
-        while (!done) {
-            unsigned long long e[LB_PER_LANE];
-            unsigned l_f2 = 0, l_inv = 0;  // per-lane flag bitmasks (bit q)
-#pragma unroll
-            for (int q = 0; q < LB_PER_LANE; q++) {
-                const int64_t p = wbase + LB_PER_LANE * tid + q;
-                e[q] = p >= 0 ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
-                                                  __HIP_MEMORY_SCOPE_AGENT)
-                              : 2ull;  // virtual predecessor: prefix 0
-                const unsigned flag = (unsigned)(e[q] & 3ull);
-                l_f2 |= (unsigned)(flag == 2u) << q;
-                l_inv |= (unsigned)(flag == 0u) << q;
-            }
-            const uint64_t pmask = __ballot(l_f2 != 0);
-            const uint64_t imask = __ballot(l_inv != 0);
+        bool found_prefix = done;
+        // phase 1: individuals within this tile's own group, newest window
+        // first; a flag2 inside terminates the whole walk
+        const int64_t gstart = grp * LB_GROUP;
+        int64_t whi = vb;  // window is [max(gstart, whi-WAVE), whi)
+        while (!done && whi > gstart) {
+            const int64_t wlo = whi - WAVE > gstart ? whi - WAVE : gstart;
+            const int64_t p = wlo + tid;
+            // inactive lanes read as neutral aggregates of 0
+            const unsigned long long e =
+                p < whi ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
+                                            __HIP_MEMORY_SCOPE_AGENT)
+                        : 1ull;
+            const unsigned flag = (unsigned)(e & 3ull);
+            const uint64_t pmask = __ballot(flag == 2u);
+            const uint64_t imask = __ballot(flag == 0u);
             bool retry;
             if (pmask != 0) {
-                const int hi = 63 - __clzll(pmask);  // newest lane w/ prefix
-                const unsigned hi_f2 = __shfl(l_f2, hi, WAVE);
-                const unsigned hi_inv = __shfl(l_inv, hi, WAVE);
-                const int hi_q = 31 - __builtin_clz(hi_f2);  // newest prefix
-                // invalid strictly above the chosen prefix forces a retry
-                retry = (((imask >> hi) >> 1) != 0) ||
-                        ((hi_inv >> hi_q) >> 1) != 0;
+                const int hi = 63 - __clzll(pmask);  // newest prefix lane
+                retry = ((imask >> hi) >> 1) != 0;   // invalid above it
                 if (!retry) {
-                    unsigned long long c = 0;
-#pragma unroll
-                    for (int q = 0; q < LB_PER_LANE; q++) {
-                        const bool use = tid > hi || (tid == hi && q >= hi_q);
-                        c += use ? (e[q] >> 2) : 0;
-                    }
+                    unsigned long long c = tid >= hi ? (e >> 2) : 0;
                     for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
                     running += c;
                     done = true;
+                    found_prefix = true;
                 }
             } else {
                 retry = imask != 0;
                 if (!retry) {
-                    unsigned long long c = 0;
-#pragma unroll
-                    for (int q = 0; q < LB_PER_LANE; q++) c += e[q] >> 2;
+                    unsigned long long c = e >> 2;
                     for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
                     running += c;
-                    wbase -= LBW;
+                    whi = wlo;
                 }
             }
             if (retry) {
-                // light backoff: near-predecessor aggregates land within a
-                // fraction of a block's own work phase, so long sleeps (the
-                // round-1 s_sleep(64) tier, ~2 us) cost more latency than the
-                // polls they save; wave-windowed polls are already 1/WAVE the
-                // per-granule poll rate of the round-1 serial walk that
-                // caused the fabric-throttling collapse
                 ++spins;
                 if (spins < 8) __builtin_amdgcn_s_sleep(1);
                 else __builtin_amdgcn_s_sleep(4);
@@ -1555,15 +1564,88 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
                 }
             }
         }
+        // phase 2: whole groups below, one wave-window of group entries per
+        // round; lane t holds group grp-1-round*WAVE-t (ascending age), a
+        // complete group contributes its whole sum in one load
+        int64_t ground = 0;
+        while (!done) {
+            const int64_t gi = grp - 1 - ground * WAVE - (int64_t)tid;
+            unsigned long long ec;  // granule-shaped: value<<2 | flag
+            if (gi < 0) {
+                ec = 2ull;  // below group 0: prefix 0 (chain terminator)
+            } else {
+                ec = __hip_atomic_load(&gpre[gi], __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+                if ((ec & 3ull) != 2ull) {
+                    const unsigned long long gsz =
+                        gi == ngroups - 1
+                            ? (unsigned long long)(nblocks - gi * LB_GROUP)
+                            : (unsigned long long)LB_GROUP;
+                    const unsigned long long have = __hip_atomic_load(
+                        &gcnt[gi], __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
+                    if (have == gsz) {
+                        const unsigned long long s = __hip_atomic_load(
+                            &gsum[gi], __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_AGENT);
+                        ec = (s << 2) | 1ull;  // complete group aggregate
+                    } else {
+                        ec = 0ull;  // incomplete: invalid
+                    }
+                }
+            }
+            const unsigned flag = (unsigned)(ec & 3ull);
+            const uint64_t pmask = __ballot(flag == 2u);
+            const uint64_t imask = __ballot(flag == 0u);
+            bool retry;
+            if (pmask != 0) {
+                // nearest (youngest) prefix group = LOWEST lane with flag2
+                const int lo = __ffsll((unsigned long long)pmask) - 1;
+                retry = (imask & ((1ull << lo) - 1)) != 0;  // invalid below it
+                if (!retry) {
+                    unsigned long long c = tid <= lo ? (ec >> 2) : 0;
+                    for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
+                    running += c;
+                    done = true;
+                    found_prefix = true;
+                }
+            } else {
+                retry = imask != 0;
+                if (!retry) {
+                    unsigned long long c = ec >> 2;
+                    for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
+                    running += c;
+                    ground++;
+                }
+            }
+            if (retry) {
+                ++spins;
+                if (spins < 8) __builtin_amdgcn_s_sleep(1);
+                else __builtin_amdgcn_s_sleep(4);
+                if (spins > (1u << 23)) {
+                    if (tid == 0)
+                        __hip_atomic_store((gu64_t *)(state + 1), 1ull,
+                                           __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT);
+                    done = true;
+                }
+            }
+        }
+        (void)found_prefix;
         if (tid == 0) {
-            __hip_atomic_store(&g[vb],
-                               ((running + block_cnt) << 2) | 2ull,
-                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            const unsigned long long inc = running + block_cnt;
+            __hip_atomic_store(&g[vb], (inc << 2) | 2ull, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+            // the group's last tile publishes the group prefix
+            if (vb == nblocks - 1 || vb % LB_GROUP == LB_GROUP - 1)
+                __hip_atomic_store(&gpre[grp], (inc << 2) | 2ull,
+                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
             sh_prefix = running;
         }
     }
     __syncthreads();
-    // emit: replay the captured decisions with pure LDS reads (no compares)
+    // emit: replay the captured decisions — keys/vals from LDS (no
+    // compares), weights straight from global (one read per consumed row,
+    // thread-contiguous)
     uint64_t gpos = (uint64_t)sh_prefix + thread_off;
     {
         int i = ai, j = bi;
@@ -1577,7 +1659,8 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
             if (keep) {
                 ok[gpos] = lk[sel];
                 ov[gpos] = lv[sel];
-                ow[gpos] = eq ? (W)(lw[i] + lw[naL + j]) : lw[sel];
+                ow[gpos] = eq ? (W)(Aw[i] + Bw[j])
+                              : (take_a ? Aw[i] : Bw[j]);
                 gpos++;
             }
             i += (int)(act & (take_a | eq));
@@ -2529,39 +2612,31 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
     uint64_t *rk, *rv;
     W *rw;
     if (!twopass) {
+        const int64_t ngroups = (nblocks + LB_GROUP - 1) / LB_GROUP;
+        const int64_t nstate = 2 + nblocks + 3 * ngroups;
         unsigned long long *state;
-        HIP_CHECK(dbspk::cache_malloc((void **)&state, (nblocks + 2) * sizeof(uint64_t), s));
-        HIP_CHECK(hipMemsetAsync(state, 0, (nblocks + 2) * sizeof(uint64_t), s));
+        HIP_CHECK(dbspk::cache_malloc((void **)&state, nstate * sizeof(uint64_t), s));
+        HIP_CHECK(hipMemsetAsync(state, 0, nstate * sizeof(uint64_t), s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rk, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rv, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rw, total * sizeof(W) + 8, s));
-        static const int lbw = []() {
-            const char *e = getenv("DBSP_MERGE_LB");
-            return (e && atoi(e) == 4) ? 4 : 1;
-        }();
         static const bool nolb = []() {  // timing diagnostic (wrong results)
             const char *e = getenv("DBSP_MERGE_NOLB");
             return e && e[0] == '1';
         }();
-        const size_t smem = 3 * (tile + 2) * sizeof(uint64_t);
+        const size_t smem = 2 * (tile + 2) * sizeof(uint64_t);
         const dim3 g((uint32_t)nblocks);
         if (nolb && tile == 2048)
-            k_mp_merge_onepass<W, 2048, 1, true><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, 2048, true><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else if (nolb)
-            k_mp_merge_onepass<W, MP_TILE, 1, true><<<g, MP_THREADS, smem, s>>>(
-                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
-        else if (tile == 2048 && lbw == 4)
-            k_mp_merge_onepass<W, 2048, 4><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, MP_TILE, true><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else if (tile == 2048)
-            k_mp_merge_onepass<W, 2048, 1><<<g, MP_THREADS, smem, s>>>(
-                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
-        else if (lbw == 4)
-            k_mp_merge_onepass<W, MP_TILE, 4><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, 2048><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else
-            k_mp_merge_onepass<W, MP_TILE, 1><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, MP_TILE><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         unsigned long long h_state[2];
         HIP_CHECK(hipMemcpyAsync(&h_state[0], state + 1, sizeof(uint64_t),
