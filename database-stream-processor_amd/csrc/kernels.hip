@@ -1373,21 +1373,10 @@ typedef __attribute__((address_space(1))) unsigned long long gu64_t;
 //   granule = value<<2 | flag;  flag: 0 invalid, 1 aggregate, 2 prefix
 // ---------------------------------------------------------------------------
 
-// Hierarchical lookback state (nblocks tiles, groups of LB_GROUP tiles):
-//   state[0] = ticket; state[1] = poison
-//   g[0..nblocks)      individual granules  (value<<2 | flag)
-//   gsum[0..ngroups)   running sum of the group's published aggregates
-//   gcnt[0..ngroups)   how many of the group's tiles have published
-//   gpre[0..ngroups)   group prefix granule (flag2 set by the group's last
-//                      tile with its inclusive prefix)
-// A tile walks individuals only within its own group (<= LB_GROUP/WAVE
-// rounds), then one wave-window over group entries — a complete group
-// (gcnt == group size) contributes its whole sum in ONE load, so the walk
-// no longer scales with the resident-block window (the measured 8.5 ms
-// flat-lookback overhead at ~1M tiles).
-#define LB_GROUP 256
-
-template <typename W, int TILE, bool SKIP_LB = false>
+// lookback window: each lane holds LB_PER_LANE consecutive granules, so one
+// round covers WAVE*LB_PER_LANE predecessors — the round-1 serial-lane walk
+// resolved ~64/round and its latency dominated the kernel
+template <typename W, int TILE, int LB_PER_LANE = 1, bool SKIP_LB = false>
 __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
     const uint64_t *bk, const uint64_t *bv, const W *bw, int64_t nb,
@@ -1395,13 +1384,10 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     int64_t nblocks, uint64_t *ok, uint64_t *ov, W *ow) {
     // steps per thread: a split adjustment can grow a tile one row past TILE
     constexpr int OP_ITEMS = (TILE + MP_THREADS) / MP_THREADS;
-    // LDS stages keys+vals only (the search/compare operands); weights are
-    // read straight from global — touched once per consumed row in
-    // thread-contiguous order, they stream through L2, and the slimmer LDS
-    // footprint (16.4 KB at TILE=1024) lifts occupancy to 8 blocks/CU
     extern __shared__ __attribute__((aligned(16))) char smem[];
     uint64_t *lk = (uint64_t *)smem;
     uint64_t *lv = lk + (TILE + 2);
+    W *lw = (W *)(lv + (TILE + 2));
     __shared__ uint32_t wt[MP_THREADS / WAVE + 1];
     __shared__ unsigned long long sh_vb;
     __shared__ unsigned long long sh_prefix;
@@ -1414,12 +1400,8 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     const int64_t pb0 = pb[vb], pb1 = pb[vb + 1];
     const int naL = (int)(pa1 - pa0), nbL = (int)(pb1 - pb0);
     const int totL = naL + nbL;
-    const W *Aw = aw + pa0;
-    const W *Bw = bw + pb0;
-    stage_run<false, W>(lk, lv, nullptr, 0, ak + pa0, av + pa0, nullptr, naL,
-                        tid);
-    stage_run<false, W>(lk, lv, nullptr, naL, bk + pb0, bv + pb0, nullptr, nbL,
-                        tid);
+    stage_run<true>(lk, lv, lw, 0, ak + pa0, av + pa0, aw + pa0, naL, tid);
+    stage_run<true>(lk, lv, lw, naL, bk + pb0, bv + pb0, bw + pb0, nbL, tid);
     __syncthreads();
     const int items = (totL + MP_THREADS - 1) / MP_THREADS;
     const int d0 = min(tid * items, totL);
@@ -1443,13 +1425,13 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
             const bool a_ok = i < ae, b_ok = j < be;
             const bool act = a_ok | b_ok;
             // all indices stay inside the tile+2 LDS arrays even when a side
-            // is exhausted (i <= naL, naL + j <= totL <= TILE + 1)
+            // is exhausted (i <= naL, naL + j <= totL <= MP_TILE + 1)
             const uint64_t ka = lk[i], va = lv[i];
             const uint64_t kb = lk[naL + j], vB = lv[naL + j];
             const bool eq = a_ok & b_ok & row_eq(ka, va, kb, vB);
             const bool take_a = a_ok & ((!b_ok) | row_lt(ka, va, kb, vB) | eq);
-            bool keep = act;
-            if (eq) keep = (W)(Aw[i] + Bw[j]) != (W)0;  // rare: global reads
+            const W sum = (W)(lw[i] + lw[naL + j]);
+            const bool keep = act & ((!eq) | (sum != (W)0));
             m_take |= (uint32_t)take_a << t;
             m_eq |= (uint32_t)eq << t;
             m_keep |= (uint32_t)(keep & act) << t;
@@ -1482,170 +1464,106 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
         thread_off = wt[tid / WAVE] + (v - cnt);
         block_cnt = wt[MP_THREADS / WAVE];
     }
-    // hierarchical decoupled lookback (wave 0)
-    const int64_t ngroups = (nblocks + LB_GROUP - 1) / LB_GROUP;
-    gu64_t *g = (gu64_t *)(state + 2);
-    gu64_t *gsum = g + nblocks;
-    gu64_t *gcnt = gsum + ngroups;
-    gu64_t *gpre = gcnt + ngroups;
-    const int64_t grp = vb / LB_GROUP;
+    // wave-parallel decoupled lookback: wave 0 loads WAVE*LB_PER_LANE (256)
+    // consecutive predecessor granules per round — lane t holds positions
+    // wbase + LB_PER_LANE*t + q; virtual blocks below 0 read as prefix 0
+    // (chain terminator)
     if (SKIP_LB) {
         // timing diagnostic ONLY (DBSP_MERGE_NOLB=1): measures the kernel
         // without the lookback protocol; outputs land at uncompacted per-tile
         // offsets (store pattern representative, results wrong by design)
         if (tid == 0) {
+            gu64_t *g = (gu64_t *)(state + 2);
             __hip_atomic_store(&g[vb],
                                ((unsigned long long)block_cnt << 2) | 2ull,
                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
             sh_prefix = (unsigned long long)(vb * TILE);
         }
     } else if (tid < WAVE) {
-        if (tid == 0) {
-            // publish the individual aggregate, fold it into the group sum,
-            // and bump the group's publish count LAST (release, so a reader
-            // that sees the count complete also sees every sum contribution)
+        gu64_t *g = (gu64_t *)(state + 2);
+        if (tid == 0)
             __hip_atomic_store(&g[vb],
                                ((unsigned long long)block_cnt << 2) | 1ull,
                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-            __hip_atomic_fetch_add(&gsum[grp], (unsigned long long)block_cnt,
-                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-            __hip_atomic_fetch_add(&gcnt[grp], 1ull, __ATOMIC_RELEASE,
-                                   __HIP_MEMORY_SCOPE_AGENT);
-        }
+        // Software-pipelined flat lookback: preload LB_PRE rounds (LB_PRE*64
+        // granules) as INDEPENDENT loads — one memory round-trip for the
+        // whole batch — then resolve round by round with local ballots.  The
+        // sequential-round version paid ~24 dependent L2 round trips per
+        // block (~8.5 ms over a 1B-row merge); here the typical cost is
+        // ceil(resident/512) ~ 3 round trips.  Rounds with still-invalid
+        // entries (near the chain head) re-poll just that round.
+        constexpr int LB_PRE = 8;
         unsigned long long running = 0;
         unsigned spins = 0;
         bool done = (vb == 0);
-        bool found_prefix = done;
-        // phase 1: individuals within this tile's own group, newest window
-        // first; a flag2 inside terminates the whole walk
-        const int64_t gstart = grp * LB_GROUP;
-        int64_t whi = vb;  // window is [max(gstart, whi-WAVE), whi)
-        while (!done && whi > gstart) {
-            const int64_t wlo = whi - WAVE > gstart ? whi - WAVE : gstart;
-            const int64_t p = wlo + tid;
-            // inactive lanes read as neutral aggregates of 0
-            const unsigned long long e =
-                p < whi ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
-                                            __HIP_MEMORY_SCOPE_AGENT)
-                        : 1ull;
-            const unsigned flag = (unsigned)(e & 3ull);
-            const uint64_t pmask = __ballot(flag == 2u);
-            const uint64_t imask = __ballot(flag == 0u);
-            bool retry;
-            if (pmask != 0) {
-                const int hi = 63 - __clzll(pmask);  // newest prefix lane
-                retry = ((imask >> hi) >> 1) != 0;   // invalid above it
-                if (!retry) {
-                    unsigned long long c = tid >= hi ? (e >> 2) : 0;
-                    for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
-                    running += c;
-                    done = true;
-                    found_prefix = true;
-                }
-            } else {
-                retry = imask != 0;
-                if (!retry) {
-                    unsigned long long c = e >> 2;
-                    for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
-                    running += c;
-                    whi = wlo;
-                }
-            }
-            if (retry) {
-                ++spins;
-                if (spins < 8) __builtin_amdgcn_s_sleep(1);
-                else __builtin_amdgcn_s_sleep(4);
-                if (spins > (1u << 23)) {  // bounded spin: poison, don't hang
-                    if (tid == 0)
-                        __hip_atomic_store((gu64_t *)(state + 1), 1ull,
-                                           __ATOMIC_RELAXED,
-                                           __HIP_MEMORY_SCOPE_AGENT);
-                    done = true;
-                }
-            }
-        }
-        // phase 2: whole groups below, one wave-window of group entries per
-        // round; lane t holds group grp-1-round*WAVE-t (ascending age), a
-        // complete group contributes its whole sum in one load
-        int64_t ground = 0;
+        int64_t whi = vb;  // everything in [0, whi) still unresolved
         while (!done) {
-            const int64_t gi = grp - 1 - ground * WAVE - (int64_t)tid;
-            unsigned long long ec;  // granule-shaped: value<<2 | flag
-            if (gi < 0) {
-                ec = 2ull;  // below group 0: prefix 0 (chain terminator)
-            } else {
-                ec = __hip_atomic_load(&gpre[gi], __ATOMIC_RELAXED,
-                                       __HIP_MEMORY_SCOPE_AGENT);
-                if ((ec & 3ull) != 2ull) {
-                    const unsigned long long gsz =
-                        gi == ngroups - 1
-                            ? (unsigned long long)(nblocks - gi * LB_GROUP)
-                            : (unsigned long long)LB_GROUP;
-                    const unsigned long long have = __hip_atomic_load(
-                        &gcnt[gi], __ATOMIC_ACQUIRE, __HIP_MEMORY_SCOPE_AGENT);
-                    if (have == gsz) {
-                        const unsigned long long s = __hip_atomic_load(
-                            &gsum[gi], __ATOMIC_RELAXED,
-                            __HIP_MEMORY_SCOPE_AGENT);
-                        ec = (s << 2) | 1ull;  // complete group aggregate
+            const int64_t wlo = whi - (int64_t)LB_PRE * WAVE;
+            unsigned long long e[LB_PRE];
+#pragma unroll
+            for (int r = 0; r < LB_PRE; r++) {
+                const int64_t p = wlo + (int64_t)r * WAVE + tid;
+                e[r] = p >= 0 ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
+                                                  __HIP_MEMORY_SCOPE_AGENT)
+                              : 2ull;  // virtual predecessor: prefix 0
+            }
+            for (int r = LB_PRE - 1; r >= 0 && !done; r--) {
+                for (;;) {  // resolve round r (re-polling only this round)
+                    const unsigned flag = (unsigned)(e[r] & 3ull);
+                    const uint64_t pmask = __ballot(flag == 2u);
+                    const uint64_t imask = __ballot(flag == 0u);
+                    bool retry;
+                    if (pmask != 0) {
+                        const int hi = 63 - __clzll(pmask);  // newest prefix
+                        retry = ((imask >> hi) >> 1) != 0;   // invalid above
+                        if (!retry) {
+                            unsigned long long c = tid >= hi ? (e[r] >> 2) : 0;
+                            for (int d = 32; d; d >>= 1)
+                                c += __shfl_xor(c, d, WAVE);
+                            running += c;
+                            done = true;
+                            break;
+                        }
                     } else {
-                        ec = 0ull;  // incomplete: invalid
+                        retry = imask != 0;
+                        if (!retry) {
+                            unsigned long long c = e[r] >> 2;
+                            for (int d = 32; d; d >>= 1)
+                                c += __shfl_xor(c, d, WAVE);
+                            running += c;
+                            whi = wlo + (int64_t)r * WAVE;
+                            break;
+                        }
                     }
-                }
-            }
-            const unsigned flag = (unsigned)(ec & 3ull);
-            const uint64_t pmask = __ballot(flag == 2u);
-            const uint64_t imask = __ballot(flag == 0u);
-            bool retry;
-            if (pmask != 0) {
-                // nearest (youngest) prefix group = LOWEST lane with flag2
-                const int lo = __ffsll((unsigned long long)pmask) - 1;
-                retry = (imask & ((1ull << lo) - 1)) != 0;  // invalid below it
-                if (!retry) {
-                    unsigned long long c = tid <= lo ? (ec >> 2) : 0;
-                    for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
-                    running += c;
-                    done = true;
-                    found_prefix = true;
-                }
-            } else {
-                retry = imask != 0;
-                if (!retry) {
-                    unsigned long long c = ec >> 2;
-                    for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
-                    running += c;
-                    ground++;
-                }
-            }
-            if (retry) {
-                ++spins;
-                if (spins < 8) __builtin_amdgcn_s_sleep(1);
-                else __builtin_amdgcn_s_sleep(4);
-                if (spins > (1u << 23)) {
-                    if (tid == 0)
-                        __hip_atomic_store((gu64_t *)(state + 1), 1ull,
-                                           __ATOMIC_RELAXED,
-                                           __HIP_MEMORY_SCOPE_AGENT);
-                    done = true;
+                    // light backoff, then re-poll just this round
+                    ++spins;
+                    if (spins < 8) __builtin_amdgcn_s_sleep(1);
+                    else __builtin_amdgcn_s_sleep(4);
+                    if (spins > (1u << 23)) {  // bounded: poison, don't hang
+                        if (tid == 0)
+                            __hip_atomic_store((gu64_t *)(state + 1), 1ull,
+                                               __ATOMIC_RELAXED,
+                                               __HIP_MEMORY_SCOPE_AGENT);
+                        done = true;
+                        break;
+                    }
+                    const int64_t p = wlo + (int64_t)r * WAVE + tid;
+                    e[r] = p >= 0
+                               ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
+                                                   __HIP_MEMORY_SCOPE_AGENT)
+                               : 2ull;
                 }
             }
         }
-        (void)found_prefix;
         if (tid == 0) {
-            const unsigned long long inc = running + block_cnt;
-            __hip_atomic_store(&g[vb], (inc << 2) | 2ull, __ATOMIC_RELAXED,
-                               __HIP_MEMORY_SCOPE_AGENT);
-            // the group's last tile publishes the group prefix
-            if (vb == nblocks - 1 || vb % LB_GROUP == LB_GROUP - 1)
-                __hip_atomic_store(&gpre[grp], (inc << 2) | 2ull,
-                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            __hip_atomic_store(&g[vb],
+                               ((running + block_cnt) << 2) | 2ull,
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
             sh_prefix = running;
         }
     }
     __syncthreads();
-    // emit: replay the captured decisions — keys/vals from LDS (no
-    // compares), weights straight from global (one read per consumed row,
-    // thread-contiguous)
+    // emit: replay the captured decisions with pure LDS reads (no compares)
     uint64_t gpos = (uint64_t)sh_prefix + thread_off;
     {
         int i = ai, j = bi;
@@ -1659,8 +1577,7 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
             if (keep) {
                 ok[gpos] = lk[sel];
                 ov[gpos] = lv[sel];
-                ow[gpos] = eq ? (W)(Aw[i] + Bw[j])
-                              : (take_a ? Aw[i] : Bw[j]);
+                ow[gpos] = eq ? (W)(lw[i] + lw[naL + j]) : lw[sel];
                 gpos++;
             }
             i += (int)(act & (take_a | eq));
@@ -2479,58 +2396,6 @@ dbsp_status c5_gen_rows(hipStream_t s, int64_t n, uint64_t stride,
     return DBSP_OK;
 }
 
-// sort rows by (k major, v minor); ping-pong scratch must hold n rows
-// (kk2/vv2/ww2).  Skips byte passes above the significant bytes of max(k)/max(v).
-dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
-                      int64_t n, uint64_t *kk2, uint64_t *vv2, int64_t *ww2,
-                      bool *result_in_scratch) {
-    *result_in_scratch = false;
-    if (n <= 1) return DBSP_OK;
-    // significant bytes from max values
-    uint64_t *d_max;
-    HIP_CHECK(dbspk::cache_malloc((void **)&d_max, 4 * sizeof(uint64_t), s));
-    HIP_CHECK(hipMemsetAsync(d_max, 0, 2 * sizeof(uint64_t), s));
-    HIP_CHECK(hipMemsetAsync(d_max + 2, 0xFF, 2 * sizeof(uint64_t), s));
-    k_minmax_u64<<<grid_for(n), BLK, 0, s>>>(kk, vv, n, d_max);
-    uint64_t h_max[4];
-    HIP_CHECK(hipMemcpyAsync(h_max, d_max, 4 * sizeof(uint64_t),
-                             hipMemcpyDeviceToHost, s));
-    HIP_CHECK(hipStreamSynchronize(s));
-    HIP_CHECK(dbspk::cache_free(d_max, s));
-    const uint64_t kbase = h_max[2], vbase = h_max[3];
-    const uint64_t krange = h_max[0] - kbase, vrange = h_max[1] - vbase;
-    int kbytes = 0, vbytes = 0;
-    while (kbytes < 8 && (krange >> (8 * kbytes)) != 0) kbytes++;
-    while (vbytes < 8 && (vrange >> (8 * vbytes)) != 0) vbytes++;
-
-    int64_t nblocks = ceil_div(n, SORT_TILE);
-    uint64_t *counts;
-    HIP_CHECK(dbspk::cache_malloc((void **)&counts, (int64_t)256 * nblocks * sizeof(uint64_t), s));
-
-    uint64_t *src_k = kk, *src_v = vv; int64_t *src_w = ww;
-    uint64_t *dst_k = kk2, *dst_v = vv2; int64_t *dst_w = ww2;
-    for (int byte = 0; byte < 16; byte++) {
-        bool is_v = byte < 8;
-        if (is_v && (byte & 7) >= vbytes) continue;
-        if (!is_v && (byte & 7) >= kbytes) continue;
-        k_radix_hist<<<dim3((uint32_t)nblocks), BLK, 0, s>>>(
-            src_k, src_v, n, byte, kbase, vbase, nblocks, counts);
-        dbsp_status st = scan_exclusive(s, counts, counts, 256 * nblocks, nullptr);
-        if (st != DBSP_OK) return st;
-        k_radix_scatter<<<dim3((uint32_t)nblocks), BLK, 0, s>>>(
-            src_k, src_v, src_w, n, byte, kbase, vbase, nblocks, counts, dst_k,
-            dst_v, dst_w);
-        uint64_t *t;
-        int64_t *tw;
-        t = src_k; src_k = dst_k; dst_k = t;
-        t = src_v; src_v = dst_v; dst_v = t;
-        tw = src_w; src_w = dst_w; dst_w = tw;
-    }
-    HIP_CHECK(dbspk::cache_free(counts, s));
-    *result_in_scratch = (src_k != kk);
-    return DBSP_OK;
-}
-
 // consolidate SORTED rows into freshly allocated output; returns exact length
 dbsp_status consolidate_sorted(hipStream_t s, const uint64_t *kk,
                                const uint64_t *vv, const int64_t *ww, int64_t n,
@@ -2612,31 +2477,39 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
     uint64_t *rk, *rv;
     W *rw;
     if (!twopass) {
-        const int64_t ngroups = (nblocks + LB_GROUP - 1) / LB_GROUP;
-        const int64_t nstate = 2 + nblocks + 3 * ngroups;
         unsigned long long *state;
-        HIP_CHECK(dbspk::cache_malloc((void **)&state, nstate * sizeof(uint64_t), s));
-        HIP_CHECK(hipMemsetAsync(state, 0, nstate * sizeof(uint64_t), s));
+        HIP_CHECK(dbspk::cache_malloc((void **)&state, (nblocks + 2) * sizeof(uint64_t), s));
+        HIP_CHECK(hipMemsetAsync(state, 0, (nblocks + 2) * sizeof(uint64_t), s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rk, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rv, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rw, total * sizeof(W) + 8, s));
+        static const int lbw = []() {
+            const char *e = getenv("DBSP_MERGE_LB");
+            return (e && atoi(e) == 4) ? 4 : 1;
+        }();
         static const bool nolb = []() {  // timing diagnostic (wrong results)
             const char *e = getenv("DBSP_MERGE_NOLB");
             return e && e[0] == '1';
         }();
-        const size_t smem = 2 * (tile + 2) * sizeof(uint64_t);
+        const size_t smem = 3 * (tile + 2) * sizeof(uint64_t);
         const dim3 g((uint32_t)nblocks);
         if (nolb && tile == 2048)
-            k_mp_merge_onepass<W, 2048, true><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, 2048, 1, true><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else if (nolb)
-            k_mp_merge_onepass<W, MP_TILE, true><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, MP_TILE, 1, true><<<g, MP_THREADS, smem, s>>>(
+                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
+        else if (tile == 2048 && lbw == 4)
+            k_mp_merge_onepass<W, 2048, 4><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else if (tile == 2048)
-            k_mp_merge_onepass<W, 2048><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, 2048, 1><<<g, MP_THREADS, smem, s>>>(
+                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
+        else if (lbw == 4)
+            k_mp_merge_onepass<W, MP_TILE, 4><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else
-            k_mp_merge_onepass<W, MP_TILE><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, MP_TILE, 1><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         unsigned long long h_state[2];
         HIP_CHECK(hipMemcpyAsync(&h_state[0], state + 1, sizeof(uint64_t),
