@@ -2396,6 +2396,58 @@ dbsp_status c5_gen_rows(hipStream_t s, int64_t n, uint64_t stride,
     return DBSP_OK;
 }
 
+// sort rows by (k major, v minor); ping-pong scratch must hold n rows
+// (kk2/vv2/ww2).  Skips byte passes above the significant bytes of max(k)/max(v).
+dbsp_status sort_rows(hipStream_t s, uint64_t *kk, uint64_t *vv, int64_t *ww,
+                      int64_t n, uint64_t *kk2, uint64_t *vv2, int64_t *ww2,
+                      bool *result_in_scratch) {
+    *result_in_scratch = false;
+    if (n <= 1) return DBSP_OK;
+    // significant bytes from max values
+    uint64_t *d_max;
+    HIP_CHECK(dbspk::cache_malloc((void **)&d_max, 4 * sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(d_max, 0, 2 * sizeof(uint64_t), s));
+    HIP_CHECK(hipMemsetAsync(d_max + 2, 0xFF, 2 * sizeof(uint64_t), s));
+    k_minmax_u64<<<grid_for(n), BLK, 0, s>>>(kk, vv, n, d_max);
+    uint64_t h_max[4];
+    HIP_CHECK(hipMemcpyAsync(h_max, d_max, 4 * sizeof(uint64_t),
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    HIP_CHECK(dbspk::cache_free(d_max, s));
+    const uint64_t kbase = h_max[2], vbase = h_max[3];
+    const uint64_t krange = h_max[0] - kbase, vrange = h_max[1] - vbase;
+    int kbytes = 0, vbytes = 0;
+    while (kbytes < 8 && (krange >> (8 * kbytes)) != 0) kbytes++;
+    while (vbytes < 8 && (vrange >> (8 * vbytes)) != 0) vbytes++;
+
+    int64_t nblocks = ceil_div(n, SORT_TILE);
+    uint64_t *counts;
+    HIP_CHECK(dbspk::cache_malloc((void **)&counts, (int64_t)256 * nblocks * sizeof(uint64_t), s));
+
+    uint64_t *src_k = kk, *src_v = vv; int64_t *src_w = ww;
+    uint64_t *dst_k = kk2, *dst_v = vv2; int64_t *dst_w = ww2;
+    for (int byte = 0; byte < 16; byte++) {
+        bool is_v = byte < 8;
+        if (is_v && (byte & 7) >= vbytes) continue;
+        if (!is_v && (byte & 7) >= kbytes) continue;
+        k_radix_hist<<<dim3((uint32_t)nblocks), BLK, 0, s>>>(
+            src_k, src_v, n, byte, kbase, vbase, nblocks, counts);
+        dbsp_status st = scan_exclusive(s, counts, counts, 256 * nblocks, nullptr);
+        if (st != DBSP_OK) return st;
+        k_radix_scatter<<<dim3((uint32_t)nblocks), BLK, 0, s>>>(
+            src_k, src_v, src_w, n, byte, kbase, vbase, nblocks, counts, dst_k,
+            dst_v, dst_w);
+        uint64_t *t;
+        int64_t *tw;
+        t = src_k; src_k = dst_k; dst_k = t;
+        t = src_v; src_v = dst_v; dst_v = t;
+        tw = src_w; src_w = dst_w; dst_w = tw;
+    }
+    HIP_CHECK(dbspk::cache_free(counts, s));
+    *result_in_scratch = (src_k != kk);
+    return DBSP_OK;
+}
+
 // consolidate SORTED rows into freshly allocated output; returns exact length
 dbsp_status consolidate_sorted(hipStream_t s, const uint64_t *kk,
                                const uint64_t *vv, const int64_t *ww, int64_t n,
