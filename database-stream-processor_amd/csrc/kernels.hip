@@ -1412,12 +1412,20 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     merge_path_lds(lk, lv, naL, nbL, d1, ae64, be64);
     adjust_split_lds(lk, lv, naL, nbL, ae64, be64);
     const int ai = (int)ai64, bi = (int)bi64, ae = (int)ae64, be = (int)be64;
-    // fused walk: predicated (no branches), decisions captured as mask bits.
+    // fused walk: predicated (no branches), OUTPUT ROWS captured in
+    // registers (OP_ITEMS x 3 VGPR pairs) so the emit below is pure stores —
+    // no second index-walk, no re-compares, no replay LDS reads.
     // Steps needed <= items: the range can hold items+1 input rows only when
     // the d1 split adjustment pulled in the b half of an equal pair, and that
     // pair is consumed in one step.
+    // register capture fits the VGPR budget only at the small tile (5 steps
+    // = 30 capture VGPRs); the 2048 tile (9 steps) spills and keeps the
+    // decision-mask + replay form instead
+    constexpr bool CAPTURE = OP_ITEMS <= 5;
     uint32_t m_take = 0, m_eq = 0, m_keep = 0;
     uint32_t cnt = 0;
+    uint64_t cap_k[CAPTURE ? OP_ITEMS : 1], cap_v[CAPTURE ? OP_ITEMS : 1];
+    W cap_w[CAPTURE ? OP_ITEMS : 1];
     {
         int i = ai, j = bi;
 #pragma unroll
@@ -1430,10 +1438,17 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
             const uint64_t kb = lk[naL + j], vB = lv[naL + j];
             const bool eq = a_ok & b_ok & row_eq(ka, va, kb, vB);
             const bool take_a = a_ok & ((!b_ok) | row_lt(ka, va, kb, vB) | eq);
-            const W sum = (W)(lw[i] + lw[naL + j]);
+            const W wa = lw[i], wb2 = lw[naL + j];
+            const W sum = eq ? (W)(wa + wb2) : (take_a ? wa : wb2);
             const bool keep = act & ((!eq) | (sum != (W)0));
-            m_take |= (uint32_t)take_a << t;
-            m_eq |= (uint32_t)eq << t;
+            if (CAPTURE) {
+                cap_k[CAPTURE ? t : 0] = take_a ? ka : kb;
+                cap_v[CAPTURE ? t : 0] = take_a ? va : vB;
+                cap_w[CAPTURE ? t : 0] = sum;
+            } else {
+                m_take |= (uint32_t)take_a << t;
+                m_eq |= (uint32_t)eq << t;
+            }
             m_keep |= (uint32_t)(keep & act) << t;
             cnt += keep & act;
             i += (int)(act & (take_a | eq));
@@ -1485,73 +1500,53 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
             __hip_atomic_store(&g[vb],
                                ((unsigned long long)block_cnt << 2) | 1ull,
                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        // Software-pipelined flat lookback: preload LB_PRE rounds (LB_PRE*64
-        // granules) as INDEPENDENT loads — one memory round-trip for the
-        // whole batch — then resolve round by round with local ballots.  The
-        // sequential-round version paid ~24 dependent L2 round trips per
-        // block (~8.5 ms over a 1B-row merge); here the typical cost is
-        // ceil(resident/512) ~ 3 round trips.  Rounds with still-invalid
-        // entries (near the chain head) re-poll just that round.
-        constexpr int LB_PRE = 8;
+        // Flat wave-window lookback (64 granules/round).  Measured notes:
+        // preloading 8 rounds of granules ahead (one round-trip for 512) and
+        // 256-granule windows both came out SLOWER — the protocol is bound
+        // by total granule-load throughput against the small hot region, not
+        // per-round latency — so the minimal-load sequential window wins.
         unsigned long long running = 0;
         unsigned spins = 0;
         bool done = (vb == 0);
-        int64_t whi = vb;  // everything in [0, whi) still unresolved
+        int64_t wbase = vb - WAVE;
         while (!done) {
-            const int64_t wlo = whi - (int64_t)LB_PRE * WAVE;
-            unsigned long long e[LB_PRE];
-#pragma unroll
-            for (int r = 0; r < LB_PRE; r++) {
-                const int64_t p = wlo + (int64_t)r * WAVE + tid;
-                e[r] = p >= 0 ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
-                                                  __HIP_MEMORY_SCOPE_AGENT)
-                              : 2ull;  // virtual predecessor: prefix 0
+            const int64_t p = wbase + tid;
+            const unsigned long long e =
+                p >= 0 ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT)
+                       : 2ull;  // virtual predecessor: prefix 0
+            const unsigned flag = (unsigned)(e & 3ull);
+            const uint64_t pmask = __ballot(flag == 2u);
+            const uint64_t imask = __ballot(flag == 0u);
+            bool retry;
+            if (pmask != 0) {
+                const int hi = 63 - __clzll(pmask);  // newest prefix lane
+                retry = ((imask >> hi) >> 1) != 0;   // invalid above it
+                if (!retry) {
+                    unsigned long long c = tid >= hi ? (e >> 2) : 0;
+                    for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
+                    running += c;
+                    done = true;
+                }
+            } else {
+                retry = imask != 0;
+                if (!retry) {
+                    unsigned long long c = e >> 2;
+                    for (int d = 32; d; d >>= 1) c += __shfl_xor(c, d, WAVE);
+                    running += c;
+                    wbase -= WAVE;
+                }
             }
-            for (int r = LB_PRE - 1; r >= 0 && !done; r--) {
-                for (;;) {  // resolve round r (re-polling only this round)
-                    const unsigned flag = (unsigned)(e[r] & 3ull);
-                    const uint64_t pmask = __ballot(flag == 2u);
-                    const uint64_t imask = __ballot(flag == 0u);
-                    bool retry;
-                    if (pmask != 0) {
-                        const int hi = 63 - __clzll(pmask);  // newest prefix
-                        retry = ((imask >> hi) >> 1) != 0;   // invalid above
-                        if (!retry) {
-                            unsigned long long c = tid >= hi ? (e[r] >> 2) : 0;
-                            for (int d = 32; d; d >>= 1)
-                                c += __shfl_xor(c, d, WAVE);
-                            running += c;
-                            done = true;
-                            break;
-                        }
-                    } else {
-                        retry = imask != 0;
-                        if (!retry) {
-                            unsigned long long c = e[r] >> 2;
-                            for (int d = 32; d; d >>= 1)
-                                c += __shfl_xor(c, d, WAVE);
-                            running += c;
-                            whi = wlo + (int64_t)r * WAVE;
-                            break;
-                        }
-                    }
-                    // light backoff, then re-poll just this round
-                    ++spins;
-                    if (spins < 8) __builtin_amdgcn_s_sleep(1);
-                    else __builtin_amdgcn_s_sleep(4);
-                    if (spins > (1u << 23)) {  // bounded: poison, don't hang
-                        if (tid == 0)
-                            __hip_atomic_store((gu64_t *)(state + 1), 1ull,
-                                               __ATOMIC_RELAXED,
-                                               __HIP_MEMORY_SCOPE_AGENT);
-                        done = true;
-                        break;
-                    }
-                    const int64_t p = wlo + (int64_t)r * WAVE + tid;
-                    e[r] = p >= 0
-                               ? __hip_atomic_load(&g[p], __ATOMIC_RELAXED,
-                                                   __HIP_MEMORY_SCOPE_AGENT)
-                               : 2ull;
+            if (retry) {
+                ++spins;
+                if (spins < 8) __builtin_amdgcn_s_sleep(1);
+                else __builtin_amdgcn_s_sleep(4);
+                if (spins > (1u << 23)) {  // bounded spin: poison, don't hang
+                    if (tid == 0)
+                        __hip_atomic_store((gu64_t *)(state + 1), 1ull,
+                                           __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT);
+                    done = true;
                 }
             }
         }
@@ -1563,9 +1558,20 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
         }
     }
     __syncthreads();
-    // emit: replay the captured decisions with pure LDS reads (no compares)
+    // emit: pure stores from the register capture (small tile), or replay of
+    // the decision masks with LDS reads (large tile)
     uint64_t gpos = (uint64_t)sh_prefix + thread_off;
-    {
+    if (CAPTURE) {
+#pragma unroll
+        for (int t = 0; t < OP_ITEMS; t++) {
+            if ((m_keep >> t) & 1u) {
+                ok[gpos] = cap_k[CAPTURE ? t : 0];
+                ov[gpos] = cap_v[CAPTURE ? t : 0];
+                ow[gpos] = cap_w[CAPTURE ? t : 0];
+                gpos++;
+            }
+        }
+    } else {
         int i = ai, j = bi;
 #pragma unroll
         for (int t = 0; t < OP_ITEMS; t++) {
