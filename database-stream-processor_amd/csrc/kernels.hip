@@ -1251,112 +1251,6 @@ __device__ inline void stage_run(uint64_t *lk, uint64_t *lv, W *lw,
     }
 }
 
-template <bool EMIT, typename W>
-__global__ __launch_bounds__(MP_THREADS, 2) void k_mp_merge(
-    const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
-    const uint64_t *bk, const uint64_t *bv, const W *bw, int64_t nb,
-    const int64_t *pa, const int64_t *pb,
-    uint64_t *counts,  // COUNT: per-block totals out; EMIT: scanned offsets in
-    uint64_t *ok, uint64_t *ov, W *ow) {
-    extern __shared__ __attribute__((aligned(16))) char smem[];
-    uint64_t *lk = (uint64_t *)smem;          // tile + 2: the split adjustment
-    uint64_t *lv = lk + (MP_TILE + 2);        // can grow a block by one row
-    W *lw = (W *)(lv + (MP_TILE + 2));        // staged only when EMIT
-    __shared__ uint32_t wt[MP_THREADS / WAVE + 1];
-    const int tid = threadIdx.x;
-    const int64_t blk = blockIdx.x;
-    const int64_t pa0 = pa[blk], pa1 = pa[blk + 1];
-    const int64_t pb0 = pb[blk], pb1 = pb[blk + 1];
-    const int64_t naL = pa1 - pa0, nbL = pb1 - pb0;
-    const int64_t totL = naL + nbL;
-    // stage: 16 B vector loads where the global offset is even (the common
-    // case after the head fix-up), scalar head/tail otherwise
-    stage_run<EMIT>(lk, lv, lw, 0, ak + pa0, av + pa0, aw + pa0, naL, tid);
-    stage_run<EMIT>(lk, lv, lw, naL, bk + pb0, bv + pb0, bw + pb0, nbL, tid);
-    __syncthreads();
-    // per-thread diagonals within the tile
-    const int64_t items = (totL + MP_THREADS - 1) / MP_THREADS;
-    int64_t d0 = min((int64_t)tid * items, totL);
-    int64_t d1 = min(d0 + items, totL);
-    int64_t ai, bi, ae, be;
-    merge_path_lds(lk, lv, naL, nbL, d0, ai, bi);
-    adjust_split_lds(lk, lv, naL, nbL, ai, bi);
-    merge_path_lds(lk, lv, naL, nbL, d1, ae, be);
-    adjust_split_lds(lk, lv, naL, nbL, ae, be);
-    // count walk (weights touched only on equal pairs in COUNT: rare, global)
-    uint32_t cnt = 0;
-    {
-        int64_t i = ai, j = bi;
-        while (i < ae || j < be) {
-            if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
-                W s = EMIT ? (lw[i] + lw[naL + j])
-                           : (aw[pa0 + i] + bw[pb0 + j]);
-                if (s != (W)0) cnt++;
-                i++; j++;
-            } else if (j >= be ||
-                       (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
-                cnt++; i++;
-            } else {
-                cnt++; j++;
-            }
-        }
-    }
-    if (!EMIT) {
-        // block total
-        uint32_t v = cnt;
-        for (int d = 1; d < WAVE; d <<= 1) {
-            uint32_t up = __shfl_up(v, d, WAVE);
-            if ((tid & (WAVE - 1)) >= d) v += up;
-        }
-        if ((tid & (WAVE - 1)) == WAVE - 1) wt[tid / WAVE] = v;
-        __syncthreads();
-        if (tid == 0) {
-            uint32_t acc = 0;
-            for (int w = 0; w < MP_THREADS / WAVE; w++) acc += wt[w];
-            counts[blk] = acc;
-        }
-        return;
-    }
-    // EMIT: exclusive scan of thread counts -> local offsets
-    uint32_t off;
-    {
-        uint32_t v = cnt;
-        for (int d = 1; d < WAVE; d <<= 1) {
-            uint32_t up = __shfl_up(v, d, WAVE);
-            if ((tid & (WAVE - 1)) >= d) v += up;
-        }
-        if ((tid & (WAVE - 1)) == WAVE - 1) wt[tid / WAVE] = v;
-        __syncthreads();
-        if (tid == 0) {
-            uint32_t acc = 0;
-            for (int w = 0; w < MP_THREADS / WAVE; w++) {
-                uint32_t t = wt[w];
-                wt[w] = acc;
-                acc += t;
-            }
-        }
-        __syncthreads();
-        off = wt[tid / WAVE] + (v - cnt);
-    }
-    uint64_t gpos = counts[blk] + off;
-    {
-        int64_t i = ai, j = bi;
-        while (i < ae || j < be) {
-            if (i < ae && j < be && row_eq(lk[i], lv[i], lk[naL + j], lv[naL + j])) {
-                W s = lw[i] + lw[naL + j];
-                if (s != (W)0) { ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = s; gpos++; }
-                i++; j++;
-            } else if (j >= be ||
-                       (i < ae && row_lt(lk[i], lv[i], lk[naL + j], lv[naL + j]))) {
-                ok[gpos] = lk[i]; ov[gpos] = lv[i]; ow[gpos] = lw[i]; gpos++; i++;
-            } else {
-                ok[gpos] = lk[naL + j]; ov[gpos] = lv[naL + j]; ow[gpos] = lw[naL + j];
-                gpos++; j++;
-            }
-        }
-    }
-}
-
 typedef __attribute__((address_space(1))) unsigned long long gu64_t;
 
 // ---------------------------------------------------------------------------
@@ -1373,10 +1267,15 @@ typedef __attribute__((address_space(1))) unsigned long long gu64_t;
 //   granule = value<<2 | flag;  flag: 0 invalid, 1 aggregate, 2 prefix
 // ---------------------------------------------------------------------------
 
-// lookback window: each lane holds LB_PER_LANE consecutive granules, so one
-// round covers WAVE*LB_PER_LANE predecessors — the round-1 serial-lane walk
-// resolved ~64/round and its latency dominated the kernel
-template <typename W, int TILE, int LB_PER_LANE = 1, bool SKIP_LB = false>
+// One merge-tile kernel, three offset modes (MODE):
+//   0  single-pass: decoupled-lookback prefix (state = ticket/poison/granules)
+//   1  count phase: write the tile's output count to state[vb], no emit
+//   2  emit phase: read the tile's pre-scanned output offset from state[vb]
+// Modes 1+2 form the default two-pass pipeline (count -> device scan ->
+// emit); mode 0 reads each input byte once but its lookback protocol costs
+// more than the count re-read at 1B-row scale (measured: granule-load
+// throughput-bound — wider windows and preloading both made it slower).
+template <typename W, int TILE, int MODE, bool SKIP_LB = false>
 __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     const uint64_t *ak, const uint64_t *av, const W *aw, int64_t na,
     const uint64_t *bk, const uint64_t *bv, const W *bw, int64_t nb,
@@ -1384,6 +1283,7 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     int64_t nblocks, uint64_t *ok, uint64_t *ov, W *ow) {
     // steps per thread: a split adjustment can grow a tile one row past TILE
     constexpr int OP_ITEMS = (TILE + MP_THREADS) / MP_THREADS;
+    constexpr bool STAGE_W = MODE != 1;  // count phase: weights only on eq
     extern __shared__ __attribute__((aligned(16))) char smem[];
     uint64_t *lk = (uint64_t *)smem;
     uint64_t *lv = lk + (TILE + 2);
@@ -1392,16 +1292,22 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     __shared__ unsigned long long sh_vb;
     __shared__ unsigned long long sh_prefix;
     const int tid = threadIdx.x;
-    // ticket: virtual block id in launch order (forward progress for lookback)
-    if (tid == 0) sh_vb = atomicAdd(state, 1ull);
-    __syncthreads();
+    if (MODE == 0) {
+        // ticket: virtual block id in launch order (lookback progress)
+        if (tid == 0) sh_vb = atomicAdd(state, 1ull);
+        __syncthreads();
+    } else if (tid == 0) {
+        sh_vb = (unsigned long long)blockIdx.x;
+    }
+    if (MODE != 0) __syncthreads();
     const int64_t vb = (int64_t)sh_vb;
     const int64_t pa0 = pa[vb], pa1 = pa[vb + 1];
     const int64_t pb0 = pb[vb], pb1 = pb[vb + 1];
     const int naL = (int)(pa1 - pa0), nbL = (int)(pb1 - pb0);
     const int totL = naL + nbL;
-    stage_run<true>(lk, lv, lw, 0, ak + pa0, av + pa0, aw + pa0, naL, tid);
-    stage_run<true>(lk, lv, lw, naL, bk + pb0, bv + pb0, bw + pb0, nbL, tid);
+    stage_run<STAGE_W>(lk, lv, lw, 0, ak + pa0, av + pa0, aw + pa0, naL, tid);
+    stage_run<STAGE_W>(lk, lv, lw, naL, bk + pb0, bv + pb0, bw + pb0, nbL,
+                       tid);
     __syncthreads();
     const int items = (totL + MP_THREADS - 1) / MP_THREADS;
     const int d0 = min(tid * items, totL);
@@ -1421,7 +1327,7 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
     // register capture fits the VGPR budget only at the small tile (5 steps
     // = 30 capture VGPRs); the 2048 tile (9 steps) spills and keeps the
     // decision-mask + replay form instead
-    constexpr bool CAPTURE = OP_ITEMS <= 5;
+    constexpr bool CAPTURE = MODE != 1 && OP_ITEMS <= 5;
     uint32_t m_take = 0, m_eq = 0, m_keep = 0;
     uint32_t cnt = 0;
     uint64_t cap_k[CAPTURE ? OP_ITEMS : 1], cap_v[CAPTURE ? OP_ITEMS : 1];
@@ -1438,14 +1344,22 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
             const uint64_t kb = lk[naL + j], vB = lv[naL + j];
             const bool eq = a_ok & b_ok & row_eq(ka, va, kb, vB);
             const bool take_a = a_ok & ((!b_ok) | row_lt(ka, va, kb, vB) | eq);
-            const W wa = lw[i], wb2 = lw[naL + j];
-            const W sum = eq ? (W)(wa + wb2) : (take_a ? wa : wb2);
+            W sum;
+            if (MODE == 1) {
+                // count phase stages no weights: read both sides only at an
+                // equal pair (rare; the cancellation test needs the sum)
+                sum = (W)0;
+                if (eq) sum = (W)(aw[pa0 + i] + bw[pb0 + j]);
+            } else {
+                const W wa = lw[i], wb2 = lw[naL + j];
+                sum = eq ? (W)(wa + wb2) : (take_a ? wa : wb2);
+            }
             const bool keep = act & ((!eq) | (sum != (W)0));
             if (CAPTURE) {
                 cap_k[CAPTURE ? t : 0] = take_a ? ka : kb;
                 cap_v[CAPTURE ? t : 0] = take_a ? va : vB;
                 cap_w[CAPTURE ? t : 0] = sum;
-            } else {
+            } else if (MODE != 1) {
                 m_take |= (uint32_t)take_a << t;
                 m_eq |= (uint32_t)eq << t;
             }
@@ -1479,11 +1393,15 @@ __global__ __launch_bounds__(MP_THREADS, 4) void k_mp_merge_onepass(
         thread_off = wt[tid / WAVE] + (v - cnt);
         block_cnt = wt[MP_THREADS / WAVE];
     }
-    // wave-parallel decoupled lookback: wave 0 loads WAVE*LB_PER_LANE (256)
-    // consecutive predecessor granules per round — lane t holds positions
-    // wbase + LB_PER_LANE*t + q; virtual blocks below 0 read as prefix 0
-    // (chain terminator)
-    if (SKIP_LB) {
+    if (MODE == 1) {
+        // count phase: publish the tile's count, nothing else to do
+        if (tid == 0) state[vb] = (unsigned long long)block_cnt;
+        return;
+    }
+    if (MODE == 2) {
+        // emit phase: the device scan already turned counts into offsets
+        if (tid == 0) sh_prefix = state[vb];
+    } else if (SKIP_LB) {
         // timing diagnostic ONLY (DBSP_MERGE_NOLB=1): measures the kernel
         // without the lookback protocol; outputs land at uncompacted per-tile
         // offsets (store pattern representative, results wrong by design)
@@ -2512,20 +2430,21 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
         *ok = nullptr; *ov = nullptr; *ow = nullptr; *out_n = 0;
         return DBSP_OK;
     }
-    // Default: single-pass fused-walk merge with wave-parallel decoupled
-    // lookback (reads each input byte once; see k_mp_merge_onepass).  The
-    // two-pass count->scan->emit pipeline remains as the fallback for a
-    // poisoned lookback (bounded spins exhausted) and as an opt-out
-    // (DBSP_MERGE_TWOPASS=1) for A/B measurement.
+    // Default: two-pass count -> device scan -> emit (k_mp_merge_onepass
+    // modes 1+2 — branchless capture walk in both phases).  The single-pass
+    // decoupled-lookback variant (mode 0) reads each input byte once but its
+    // lookback protocol measures SLOWER than the count re-read at 1B rows
+    // (granule-load throughput-bound; see the kernel comment) — opt in with
+    // DBSP_MERGE_ONEPASS=1; a poisoned lookback falls back to two-pass.
     static const bool twopass = []() {
-        const char *e = getenv("DBSP_MERGE_TWOPASS");
-        return e && e[0] == '1';
+        const char *e = getenv("DBSP_MERGE_ONEPASS");
+        return !(e && e[0] == '1');
     }();
     static const int op_tile = []() {
         const char *e = getenv("DBSP_MERGE_OP_TILE");
         return (e && atoi(e) == 2048) ? 2048 : MP_TILE;
     }();
-    const int64_t tile = twopass ? MP_TILE : op_tile;
+    const int64_t tile = op_tile;
     int64_t nblocks = ceil_div(total, tile);
     int64_t *pa, *pb;
     HIP_CHECK(dbspk::cache_malloc((void **)&pa, (nblocks + 1) * sizeof(int64_t), s));
@@ -2534,6 +2453,7 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
                                                          nblocks, tile, pa, pb);
     uint64_t *rk, *rv;
     W *rw;
+    const dim3 g((uint32_t)nblocks);
     if (!twopass) {
         unsigned long long *state;
         HIP_CHECK(dbspk::cache_malloc((void **)&state, (nblocks + 2) * sizeof(uint64_t), s));
@@ -2541,33 +2461,22 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
         HIP_CHECK(dbspk::cache_malloc((void **)&rk, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rv, total * sizeof(uint64_t) + 8, s));
         HIP_CHECK(dbspk::cache_malloc((void **)&rw, total * sizeof(W) + 8, s));
-        static const int lbw = []() {
-            const char *e = getenv("DBSP_MERGE_LB");
-            return (e && atoi(e) == 4) ? 4 : 1;
-        }();
         static const bool nolb = []() {  // timing diagnostic (wrong results)
             const char *e = getenv("DBSP_MERGE_NOLB");
             return e && e[0] == '1';
         }();
         const size_t smem = 3 * (tile + 2) * sizeof(uint64_t);
-        const dim3 g((uint32_t)nblocks);
         if (nolb && tile == 2048)
-            k_mp_merge_onepass<W, 2048, 1, true><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, 2048, 0, true><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else if (nolb)
-            k_mp_merge_onepass<W, MP_TILE, 1, true><<<g, MP_THREADS, smem, s>>>(
-                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
-        else if (tile == 2048 && lbw == 4)
-            k_mp_merge_onepass<W, 2048, 4><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, MP_TILE, 0, true><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else if (tile == 2048)
-            k_mp_merge_onepass<W, 2048, 1><<<g, MP_THREADS, smem, s>>>(
-                ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
-        else if (lbw == 4)
-            k_mp_merge_onepass<W, MP_TILE, 4><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, 2048, 0><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         else
-            k_mp_merge_onepass<W, MP_TILE, 1><<<g, MP_THREADS, smem, s>>>(
+            k_mp_merge_onepass<W, MP_TILE, 0><<<g, MP_THREADS, smem, s>>>(
                 ak, av, aw, na, bk, bv, bw, nb, pa, pb, state, nblocks, rk, rv, rw);
         unsigned long long h_state[2];
         HIP_CHECK(hipMemcpyAsync(&h_state[0], state + 1, sizeof(uint64_t),
@@ -2590,41 +2499,32 @@ static dbsp_status merge_rows_t(hipStream_t s, const uint64_t *ak,
     }
     uint64_t *counts;
     HIP_CHECK(dbspk::cache_malloc((void **)&counts, (nblocks + 1) * sizeof(uint64_t), s));
-    const size_t smem_count = 2 * (MP_TILE + 2) * sizeof(uint64_t);
-    k_mp_merge<false, W><<<dim3((uint32_t)nblocks), MP_THREADS, smem_count, s>>>(
-        ak, av, aw, na, bk, bv, bw, nb, pa, pb, counts, nullptr, nullptr,
-        nullptr);
+    const size_t smem_count = 2 * (tile + 2) * sizeof(uint64_t);
+    if (tile == 2048)
+        k_mp_merge_onepass<W, 2048, 1><<<g, MP_THREADS, smem_count, s>>>(
+            ak, av, aw, na, bk, bv, bw, nb, pa, pb,
+            (unsigned long long *)counts, nblocks, nullptr, nullptr, nullptr);
+    else
+        k_mp_merge_onepass<W, MP_TILE, 1><<<g, MP_THREADS, smem_count, s>>>(
+            ak, av, aw, na, bk, bv, bw, nb, pa, pb,
+            (unsigned long long *)counts, nblocks, nullptr, nullptr, nullptr);
     uint64_t nout = 0;
-    static const bool host_prof = []() {
-        const char *e = getenv("DBSP_PROFILE");
-        return e && e[0] == '2';
-    }();
-    auto tick_us = []() {
-        struct timespec ts;
-        clock_gettime(CLOCK_MONOTONIC, &ts);
-        return ts.tv_sec * 1000000.0 + ts.tv_nsec / 1000.0;
-    };
-    double t_scan0 = host_prof ? tick_us() : 0;
     dbsp_status st = scan_exclusive(s, counts, counts, nblocks, &nout);
     if (st != DBSP_OK) return st;
-    double t_scan1 = host_prof ? tick_us() : 0;
     HIP_CHECK(dbspk::cache_malloc((void **)&rk, nout * sizeof(uint64_t) + 8, s));
-    double t_a1 = host_prof ? tick_us() : 0;
     HIP_CHECK(dbspk::cache_malloc((void **)&rv, nout * sizeof(uint64_t) + 8, s));
-    double t_a2 = host_prof ? tick_us() : 0;
     HIP_CHECK(dbspk::cache_malloc((void **)&rw, nout * sizeof(W) + 8, s));
-    double t_a3 = host_prof ? tick_us() : 0;
-    if (host_prof && na + nb > (int64_t)100000000)
-        fprintf(stderr,
-                "[merge host] scan+sync %.1f ms  mallocA(k) %.1f ms  "
-                "mallocA(v) %.1f ms  mallocA(w) %.1f ms  (nout=%llu)\n",
-                (t_scan1 - t_scan0) / 1000.0, (t_a1 - t_scan1) / 1000.0,
-                (t_a2 - t_a1) / 1000.0, (t_a3 - t_a2) / 1000.0,
-                (unsigned long long)nout);
-    const size_t smem_emit = 3 * (MP_TILE + 2) * sizeof(uint64_t);
-    if (nout > 0)
-        k_mp_merge<true, W><<<dim3((uint32_t)nblocks), MP_THREADS, smem_emit, s>>>(
-            ak, av, aw, na, bk, bv, bw, nb, pa, pb, counts, rk, rv, rw);
+    const size_t smem_emit = 3 * (tile + 2) * sizeof(uint64_t);
+    if (nout > 0) {
+        if (tile == 2048)
+            k_mp_merge_onepass<W, 2048, 2><<<g, MP_THREADS, smem_emit, s>>>(
+                ak, av, aw, na, bk, bv, bw, nb, pa, pb,
+                (unsigned long long *)counts, nblocks, rk, rv, rw);
+        else
+            k_mp_merge_onepass<W, MP_TILE, 2><<<g, MP_THREADS, smem_emit, s>>>(
+                ak, av, aw, na, bk, bv, bw, nb, pa, pb,
+                (unsigned long long *)counts, nblocks, rk, rv, rw);
+    }
     HIP_CHECK(dbspk::cache_free(pa, s));
     HIP_CHECK(dbspk::cache_free(pb, s));
     HIP_CHECK(dbspk::cache_free(counts, s));
