@@ -225,14 +225,139 @@ def roofline_leg(query, events, n_ticks, tick):
     }
 
 
+def c5_cpu_baseline(n_trace, n_delta, seed, budget_s=25.0):
+    """CPU oracle leg for C5 (kind 'port'): the same trace (numpy
+    regeneration of the device generator) probed by a bounded slice of the
+    first delta through the same join -> weigh -> consolidate -> aggregate
+    pipeline, scaled to delta rows/s.  Falls back to a smaller trace if the
+    host can't hold n_trace rows (24 GB at 1B)."""
+    import numpy as np
+    sys.path.insert(0, str(ROOT / "tests"))
+    from test_gpu_c5 import c5_gen, c5_oracle_tick
+    from dbsp_amd import ROW_DT
+    nt = n_trace
+    while True:
+        try:
+            trace = c5_gen(nt, 5, 4, seed, 0)
+            break
+        except MemoryError:
+            nt //= 4
+    stride = max(3, (5 * n_trace) // n_delta)
+    dseed = (seed + 0x9E3779B97F4A7C15) % (1 << 64)
+    # bounded sample: slices of the first tick's delta until ~budget_s
+    wint = np.empty(0, dtype=ROW_DT)
+    out_trace = np.empty(0, dtype=ROW_DT)
+    slice_rows = 1_000_000
+    done = 0
+    t0 = time.perf_counter()
+    full = c5_gen(n_delta, stride, stride - 1, dseed, 1)
+    while done < n_delta and time.perf_counter() - t0 < budget_s:
+        sl = full[done:done + slice_rows]
+        wint, out_trace, _ = c5_oracle_tick(trace, wint, out_trace, sl)
+        done += len(sl)
+    dt = time.perf_counter() - t0
+    return {
+        "value": done / dt,
+        "unit": "delta rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": f"oracle C5 pipeline: {done} delta rows vs a {nt}-row "
+                  f"trace in {dt:.1f}s (single-threaded C++ restatement"
+                  + ("" if nt == n_trace else f"; trace cut to {nt} rows "
+                     "for host RAM") + ")",
+    }
+
+
+def run_c5(args, rank, world, local_rank, dist):
+    """Config C5 (BASELINE configs[4]) as an engine workload: a step = one
+    n_delta-row device-generated delta joined incrementally against the
+    n_trace-row trace with the f64 sum aggregate + upsert.  N>1 = weak
+    scaling over key-disjoint shard replicas (rank-salted seeds; the path
+    partitions with no data-path collective)."""
+    import torch
+    from dbsp_amd.engine import Ctx, Engine
+    n_ticks = args.warmup + args.steps
+    nd = args.tick
+    ctx = Ctx(local_rank)
+    eng = Engine(ctx, query=100, rank=rank, world=world)
+    eng.c5_init(args.c5_trace, nd, seed=args.seed + rank * 7919)
+    eng.run_staged(0, args.warmup * nd, nd)
+    ctx.sync()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    if dist is not None:
+        dist.barrier()
+    t0 = time.perf_counter()
+    eng.run_staged(args.warmup * nd, n_ticks * nd, nd)
+    ctx.sync()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if dist is not None:
+        import torch as _t
+        t = _t.tensor([elapsed], dtype=_t.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    stats = {name: eng.kernel_stats(k) for k, name in
+             {0: "sort_consolidate", 1: "merge", 2: "join",
+              3: "aggregate"}.items()}
+    eng.close()
+    ctx.close()
+    if rank != 0:
+        return
+    timed_rows = args.steps * nd * world
+    value = timed_rows / elapsed
+    dom = max(stats.items(), key=lambda kv: kv[1][0])
+    name, (ms, by, ln) = dom
+    achieved = (by / 1e9) / (ms / 1e3) if ms > 0 else 0.0
+    result = {
+        "metric": "C5 delta rows/s",
+        "value": round(value, 1),
+        "unit": "delta rows/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1000, 4),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "f64",
+        "data": "synthetic",
+        "config": {
+            "workload": "c5-incremental-join-f64-sum",
+            "trace_rows": args.c5_trace,
+            "delta_rows": nd,
+            "parallelism": f"keyshard{world}" if world > 1 else "single",
+        },
+        "roofline": {
+            "bound": "hbm",
+            "kernel_class": name,
+            "achieved": round(achieved, 2),
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": round(achieved / HBM_PEAK_GBS, 4),
+            "traffic": None,
+            "detail": {k: {"ms": round(v[0], 2),
+                           "algo_GB": round(v[1] / 1e9, 3),
+                           "launches": v[2]} for k, v in stats.items()},
+        },
+    }
+    if not args.no_extras and world == 1:
+        result["cpu_baseline"] = c5_cpu_baseline(args.c5_trace, nd, args.seed)
+    print(json.dumps(result))
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=250)
     ap.add_argument("--warmup", type=int, default=10)
-    ap.add_argument("--query", type=int, default=3, choices=[0, 3, 5, 8])
+    ap.add_argument("--query", default="3",
+                    choices=["0", "3", "5", "8", "c5"])
     ap.add_argument("--tick", type=int, default=40_000)
     ap.add_argument("--seed", type=int, default=1)
+    ap.add_argument("--c5-trace", type=int, default=1_000_000_000,
+                    help="C5 trace rows (configs[4]: 1B)")
     ap.add_argument("--no-extras", action="store_true",
                     help="skip cpu_baseline and roofline legs")
     args = ap.parse_args()
@@ -266,6 +391,17 @@ def main():
         t = torch.from_numpy(buf)
         tdist.broadcast(t, src=0)
         nccl_id = t.numpy()
+
+    if args.query == "c5":
+        # configs[4]: a step = one 10M-row delta; --tick keeps its meaning as
+        # rows per step, defaulting to the config's 10M (not the Nexmark 40k)
+        if args.tick == 40_000:
+            args.tick = 10_000_000
+        if args.steps == 250:
+            args.steps = 20
+        run_c5(args, rank, world, local_rank, dist)
+        return
+    args.query = int(args.query)
 
     n_ticks = args.warmup + args.steps
     events = generate_rank_slices(n_ticks, args.tick, world, rank,

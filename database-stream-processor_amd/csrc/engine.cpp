@@ -2847,6 +2847,113 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     return DBSP_OK;
 }
 
+// ---------------------------------------------------------------------------
+// C5 (query 100): synthetic 1B-row OrdIndexedZSet x 10M-row delta incremental
+// join with f64 sum aggregate (BASELINE configs[4]).  Per
+// tick: delta ⋈ trace carrying the trace's f64 val (JoinTrace::eval,
+// join.rs:732-863), weigh f(k,v)=f64(v) (aggregate/mod.rs:297-323),
+// consolidate, fold into the f64-weighted integral spine, re-aggregate the
+// affected keys with upsert retractions against the output trace
+// (aggregate/mod.rs:479-547 + upsert.rs:161-208), then TraceAppend the delta
+// into the join trace.  Both the trace and each tick's delta are generated
+// ON DEVICE (no 24 GB host staging) by the counter-based splitmix64
+// generator in kernels.hip (c5_gen_rows).
+// ---------------------------------------------------------------------------
+
+extern "C" dbsp_status dbsp_engine_c5_init(dbsp_engine *e, int64_t n_trace,
+                                           int64_t n_delta, uint64_t seed) {
+    if (!e || e->query != 100 || n_trace <= 0 || n_delta <= 0)
+        return DBSP_ERR_INVALID;
+    dbsp_ctx *c = e->ctx;
+    e->c5_trace.clear(c);
+    e->c5_wint.clear(c);
+    e->c5_out.clear(c);
+    e->c5_wint.wf64 = true;
+    DevBatch tr;
+    TRY(alloc_batch(c, n_trace, tr));
+    // trace keys 5i + h%4 (sorted unique, ~0.8 keys/int density), f64 vals
+    TRY(dbspk::c5_gen_rows(c->stream, n_trace, 5, 4, seed, 0, tr.k, tr.v,
+                           tr.w));
+    TRY(e->c5_trace.insert(c, tr));
+    e->c5_n_delta = n_delta;
+    e->c5_seed = seed;
+    // delta keys span the same range as the trace's (5*n_trace)
+    uint64_t stride = (uint64_t)((5 * n_trace) / n_delta);
+    if (stride < 3) stride = 3;
+    e->c5_delta_stride = stride;
+    e->n_events = INT64_MAX;  // step ranges are delta-row ids, not events
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    return DBSP_OK;
+}
+
+static dbsp_status c5_step(dbsp_engine *e, int64_t lo, int64_t hi) {
+    dbsp_ctx *c = e->ctx;
+    engine_free_output(e);
+    const int64_t nd = hi - lo;
+    if (nd <= 0) return DBSP_OK;
+    if (e->c5_n_delta <= 0) return DBSP_ERR_INVALID;  // c5_init not called
+    const uint64_t tick = (uint64_t)(lo / e->c5_n_delta);
+    // tick's delta: device-generated sorted-unique rows, v = 0, w = +1
+    DevBatch delta;
+    TRY(alloc_batch(c, nd, delta));
+    TRY(dbspk::c5_gen_rows(c->stream, nd, e->c5_delta_stride,
+                           e->c5_delta_stride - 1,
+                           e->c5_seed + 0x9E3779B97F4A7C15ull * (tick + 1), 1,
+                           delta.k, delta.v, delta.w));
+    // 1. join vs the trace spine, carrying the trace-side f64 val
+    std::vector<DevBatch> outs;
+    TRY(join_vs_spine(c, delta, e->c5_trace, DBSP_PROJ_HI_K_LO_V2, 0, outs));
+    // 2. weigh + consolidate into the tick's f64-weighted delta
+    DevBatch dwb{};
+    if (!outs.empty()) {
+        DevBatch cat;
+        TRY(concat_batches(c, outs, cat));
+        for (auto &b : outs) free_batch(c, b);
+        outs.clear();
+        DevBatch weighed;
+        TRY(alloc_batch(c, cat.n, weighed, true));
+        {
+            ScopedTimer t(c, 3, 0.0);
+            TRY(dbspk::map_rows(c->stream, cat.k, cat.v, cat.w, cat.n, 4,
+                                weighed.k, weighed.v, weighed.w));
+        }
+        weighed.n = cat.n;
+        free_batch(c, cat);
+        dbsp_batch ob{};
+        TRY(dbsp_sort_consolidate_f64(c, weighed.k, weighed.v,
+                                      (const double *)weighed.w, weighed.n,
+                                      &ob));
+        free_batch(c, weighed);
+        dwb = DevBatch{ob.k, ob.v, ob.w, ob.len};
+    }
+    // 3. aggregate the affected keys over the integral (including this
+    // tick's weighted delta, which is inserted first) + upsert retractions
+    DevBatch upd{};
+    if (dwb.n > 0) {
+        DevBatch wcopy;
+        {
+            std::vector<DevBatch> one{dwb};
+            TRY(concat_batches(c, one, wcopy));
+        }
+        TRY(e->c5_wint.insert(c, wcopy));
+        TRY(agg_linear_spine_f64(c, dwb, e->c5_wint, e->c5_out, upd));
+        free_batch(c, dwb);
+    }
+    // 4. output delta + TraceAppend on the output trace
+    if (upd.n > 0) {
+        DevBatch ocopy;
+        std::vector<DevBatch> one{upd};
+        TRY(concat_batches(c, one, ocopy));
+        TRY(e->c5_out.insert(c, ocopy));
+    }
+    e->output = upd;
+    e->output_is_store = false;
+    // 5. TraceAppend the delta into the join trace
+    TRY(e->c5_trace.insert(c, delta));
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    return DBSP_OK;
+}
+
 extern "C" dbsp_status dbsp_engine_step_staged(dbsp_engine *e, int64_t lo,
                                                int64_t hi) {
     if (lo < 0 || hi > e->n_events || lo > hi) return DBSP_ERR_INVALID;
@@ -2866,6 +2973,7 @@ extern "C" dbsp_status dbsp_engine_step_staged(dbsp_engine *e, int64_t lo,
         q0_step_host(e, e->h_events.data() + lo, hi - lo);
         return DBSP_OK;
     }
+    if (e->query == 100) return c5_step(e, lo, hi);
     const dbsp_event *d_ev = e->d_events + lo;
     switch (e->query) {
         case 3: return q3_step(e, d_ev, hi - lo);
@@ -2883,7 +2991,7 @@ extern "C" dbsp_status dbsp_engine_run_staged(dbsp_engine *e, int64_t lo,
     for (int64_t t = lo; t < hi; t += tick) {
         const int64_t t_hi = std::min(hi, t + tick);
         const int64_t n_lo = t_hi, n_hi = std::min(hi, t_hi + tick);
-        if (n_lo < hi) {
+        if (n_lo < hi && e->d_events) {
             e->next_ev = e->d_events + n_lo;
             e->next_n = n_hi - n_lo;
         } else {

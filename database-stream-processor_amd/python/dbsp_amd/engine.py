@@ -112,6 +112,8 @@ def _L():
         L.dbsp_engine_output.argtypes = [vp, vp, i64, ctypes.POINTER(i64)]
         L.dbsp_engine_output_events.restype = i32
         L.dbsp_engine_output_events.argtypes = [vp, vp, i64, ctypes.POINTER(i64)]
+        L.dbsp_engine_c5_init.restype = i32
+        L.dbsp_engine_c5_init.argtypes = [vp, i64, i64, u64]
         L.dbsp_engine_kernel_stats.restype = i32
         L.dbsp_engine_kernel_stats.argtypes = [vp, ctypes.c_int,
                                                ctypes.POINTER(ctypes.c_double),
@@ -392,6 +394,13 @@ class Engine:
         _check(self._lib.dbsp_engine_stage_events(self._h, _p(events),
                                                   len(events)), "stage")
         self._staged = events  # keep alive
+
+    def c5_init(self, n_trace, n_delta, seed=41):
+        """Config C5 (query 100): device-generate the n_trace-row indexed
+        trace and fix the per-tick delta size; step ranges then index delta
+        rows (run_staged(0, steps*n_delta, n_delta) = `steps` ticks)."""
+        _check(self._lib.dbsp_engine_c5_init(self._h, n_trace, n_delta, seed),
+               "c5_init")
 
     def step_staged(self, lo, hi):
         _check(self._lib.dbsp_engine_step_staged(self._h, lo, hi), "step")

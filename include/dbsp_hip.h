@@ -254,11 +254,22 @@ dbsp_status dbsp_comm_alltoallv(dbsp_ctx *ctx, const dbsp_batch *send,
 
 typedef struct dbsp_engine dbsp_engine;
 
-/* query: 0, 3, 5, 8 (Nexmark).  Builds the same operator DAG as
- * crates/nexmark/src/queries/q{0,3,5,8}.rs over the mirrored API. */
+/* query: 0, 3, 5, 8 (Nexmark — the same operator DAG as
+ * crates/nexmark/src/queries/q{0,3,5,8}.rs over the mirrored API), or 100
+ * (config C5: the synthetic 1B-row OrdIndexedZSet x 10M-row delta
+ * incremental join + f64 sum aggregate of BASELINE configs[4]). */
 dbsp_status dbsp_engine_create(dbsp_engine **out, dbsp_ctx *ctx, int query,
                                int rank, int world);
 dbsp_status dbsp_engine_destroy(dbsp_engine *e);
+
+/* C5 (query 100) setup: device-generates the n_trace-row (k, f64-bits, +1)
+ * indexed trace (counter-based splitmix64; sorted-unique by construction)
+ * and fixes the per-tick delta size.  After this, step ranges index delta
+ * ROWS: dbsp_engine_run_staged(e, 0, steps*n_delta, n_delta) runs `steps`
+ * ticks, each one a fresh device-generated n_delta-row delta joined and
+ * aggregated incrementally (operator mapping in the engine source). */
+dbsp_status dbsp_engine_c5_init(dbsp_engine *e, int64_t n_trace,
+                                int64_t n_delta, uint64_t seed);
 
 /* One clock tick: ingest `events` (host array, this tick's input delta) and run
  * every operator once in dependency order — the mirror of DBSPHandle::step
