@@ -278,38 +278,52 @@ def run_c5(args, rank, world, local_rank, dist):
     from dbsp_amd.engine import Ctx, Engine
     n_ticks = args.warmup + args.steps
     nd = args.tick
-    ctx = Ctx(local_rank)
-    eng = Engine(ctx, query=100, rank=rank, world=world)
-    eng.c5_init(args.c5_trace, nd, seed=args.seed + rank * 7919)
-    eng.run_staged(0, args.warmup * nd, nd)
-    ctx.sync()
-    if torch.cuda.is_available():
-        torch.cuda.synchronize()
+
+    def one_run(warm, timed):
+        ctx = Ctx(local_rank)
+        eng = Engine(ctx, query=100, rank=rank, world=world)
+        eng.c5_init(args.c5_trace, nd, seed=args.seed + rank * 7919)
+        eng.run_staged(0, warm * nd, nd)
+        ctx.sync()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        if dist is not None:
+            dist.barrier()
+        t0 = time.perf_counter()
+        eng.run_staged(warm * nd, (warm + timed) * nd, nd)
+        ctx.sync()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        el = time.perf_counter() - t0
+        st = {name: eng.kernel_stats(k) for k, name in
+              {0: "sort_consolidate", 1: "merge", 2: "join",
+               3: "aggregate"}.items()}
+        eng.close()
+        ctx.close()
+        return el, st
+
+    elapsed, _ = one_run(args.warmup, args.steps)
     if dist is not None:
-        dist.barrier()
-    t0 = time.perf_counter()
-    eng.run_staged(args.warmup * nd, n_ticks * nd, nd)
-    ctx.sync()
-    if torch.cuda.is_available():
-        torch.cuda.synchronize()
-    elapsed = time.perf_counter() - t0
-    if dist is not None:
-        import torch as _t
-        t = _t.tensor([elapsed], dtype=_t.float64)
+        t = torch.tensor([elapsed], dtype=torch.float64)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
-    stats = {name: eng.kernel_stats(k) for k, name in
-             {0: "sort_consolidate", 1: "merge", 2: "join",
-              3: "aggregate"}.items()}
-    eng.close()
-    ctx.close()
+    stats = None
+    if not args.no_extras and world == 1:
+        # per-kernel-class hipEvent timing needs DBSP_PROFILE=1 at ctx
+        # creation: collect it from a short separate run
+        os.environ["DBSP_PROFILE"] = "1"
+        _, stats = one_run(1, min(args.steps, 5))
+        os.environ.pop("DBSP_PROFILE", None)
     if rank != 0:
         return
     timed_rows = args.steps * nd * world
     value = timed_rows / elapsed
-    dom = max(stats.items(), key=lambda kv: kv[1][0])
-    name, (ms, by, ln) = dom
-    achieved = (by / 1e9) / (ms / 1e3) if ms > 0 else 0.0
+    if stats:
+        dom = max(stats.items(), key=lambda kv: kv[1][0])
+        name, (ms, by, ln) = dom
+        achieved = (by / 1e9) / (ms / 1e3) if ms > 0 else 0.0
+    else:
+        name, achieved, stats = "unprofiled", 0.0, {}
     result = {
         "metric": "C5 delta rows/s",
         "value": round(value, 1),

@@ -49,14 +49,17 @@ def c5_gen(n, stride, jitter, seed, val_mode):
 def c5_oracle_tick(trace, wint, out_trace, delta):
     """One C5 tick on the CPU oracle; returns (new wint, new out_trace,
     output delta)."""
-    joined = oracle.join_raw(delta, trace, 8)  # (k, v_trace), w = w1*w2
+    # caps sized for this workload (1 val/key: <= 1 match per delta row;
+    # join_raw's default cap is delta*trace — petabytes at C5 scale)
+    joined = oracle.join_raw(delta, trace, 8, cap=4 * len(delta) + 1024)
     if len(joined) == 0:
         return wint, out_trace, np.empty(0, dtype=ROW_DT)
     weighed = oracle.weigh_f64(joined)
     dwb = oracle.consolidate_f64(weighed)
     wint = oracle.merge_f64(wint, dwb) if len(wint) else dwb
     keys = np.unique(dwb["k"])
-    upd = oracle.agg_linear_upsert_f64(keys, wint, out_trace)
+    upd = oracle.agg_linear_upsert_f64(keys, wint, out_trace,
+                                       cap=4 * len(keys) + 1024)
     upd = oracle.consolidate(upd)
     if len(upd):
         out_trace = oracle.merge(out_trace, upd)
