@@ -1705,14 +1705,39 @@ __global__ void k_join_emit_multi(const uint64_t *dk, const uint64_t *dv,
     }
 }
 
-// single-workgroup join count + scan (nd <= 8192): per-row match counts over
-// every spine batch, offsets via in-LDS scan, total to *d_total — one launch,
-// no host sync, no device scan pipeline
-__global__ __launch_bounds__(FUSE_THREADS, 4) void k_join_count_scan_small(
+// small join count + scan (nd <= 8192), split in two chained launches: a
+// multi-block probe phase (the round-1 single-WG version serialized ~2M
+// dependent probe loads on ONE CU — 44 us of a 280 us q3 tick) and a
+// single-WG scan that re-sums the per-row counts from cnts.
+#define JPROBE_BLOCKS 8
+
+__global__ __launch_bounds__(BLK, 8) void k_join_probe_multi(
+    JoinCountArgs args) {
+    const int plan = blockIdx.x / JPROBE_BLOCKS;
+    const int slice = blockIdx.x % JPROBE_BLOCKS;
+    const uint64_t *dk = args.dk[plan];
+    const int64_t nd =
+        args.nd_dev[plan] ? *args.nd_dev[plan] : args.nd[plan];
+    const TraceArgs &t = args.t[plan];
+    const int64_t tn0 = args.tn_dev[plan] ? *args.tn_dev[plan] : t.n[0];
+    if (nd < 0 || nd > FUSE_MAX || tn0 < 0) return;  // scan flags the total
+    uint32_t *cnts = args.cnts[plan];
+    for (int64_t i = (int64_t)slice * BLK + threadIdx.x; i < nd;
+         i += (int64_t)JPROBE_BLOCKS * BLK) {
+        const uint64_t key = dk[i];
+        for (int b = 0; b < t.nb; b++) {
+            const int64_t tb_n = b == 0 ? tn0 : t.n[b];
+            int64_t lo = lower_bound_k(t.k[b], tb_n, key);
+            cnts[(int64_t)i * t.nb + b] =
+                (uint32_t)gallop_run(t.k[b], tb_n, key, lo);
+        }
+    }
+}
+
+__global__ __launch_bounds__(FUSE_THREADS, 4) void k_join_scan_small(
     JoinCountArgs args) {
     __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
     const int plan = blockIdx.x;
-    const uint64_t *dk = args.dk[plan];
     const int64_t nd =
         args.nd_dev[plan] ? *args.nd_dev[plan] : args.nd[plan];
     const TraceArgs &t = args.t[plan];
@@ -1730,16 +1755,9 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_join_count_scan_small(
     for (int j = 0; j < FUSE_ITEMS; j++) {
         int i = tid * FUSE_ITEMS + j;
         uint32_t rt = 0;
-        if (i < nd) {
-            uint64_t key = dk[i];
-            for (int b = 0; b < t.nb; b++) {
-                const int64_t tb_n = b == 0 ? tn0 : t.n[b];
-                int64_t lo = lower_bound_k(t.k[b], tb_n, key);
-                int64_t c = gallop_run(t.k[b], tb_n, key, lo);
-                cnts[(int64_t)i * t.nb + b] = (uint32_t)c;
-                rt += (uint32_t)c;
-            }
-        }
+        if (i < nd)
+            for (int b = 0; b < t.nb; b++)
+                rt += cnts[(int64_t)i * t.nb + b];
         row_tot[j] = rt;
         tsum += rt;
     }
@@ -2701,7 +2719,9 @@ dbsp_status join_count_scan_batch(hipStream_t s, const JoinCountArgs &args) {
     if (args.np == 0) return DBSP_OK;
     for (int i = 0; i < args.np; i++)
         if (args.nd[i] > FUSE_MAX) return DBSP_ERR_INVALID;
-    k_join_count_scan_small<<<dim3((uint32_t)args.np), FUSE_THREADS, 0, s>>>(args);
+    k_join_probe_multi<<<dim3((uint32_t)(args.np * JPROBE_BLOCKS)), BLK, 0,
+                         s>>>(args);
+    k_join_scan_small<<<dim3((uint32_t)args.np), FUSE_THREADS, 0, s>>>(args);
     return DBSP_OK;
 }
 
