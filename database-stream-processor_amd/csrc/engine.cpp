@@ -75,7 +75,8 @@ struct dbsp_ctx {
     bool force_shard = false;  // exercise the full partition+alltoallv path
                                // even at world=1 (self-exchange; test hook)
     // persistent length scratch (device + pinned host)
-    int64_t *d_len = nullptr;   // 16 device length slots (8-15: chained ticks)
+    int64_t *d_len = nullptr;   // 20 device length slots (8-15: chained
+                                // ticks; 16-17: framed-exchange totals)
     int64_t *h_len = nullptr;
     int timer_depth = 0;  // ScopedTimer nesting guard (shared event pair)
     // per-tick transient bump arena (reset at each engine tick; falls back to
@@ -141,14 +142,14 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     c->profile = p && p[0] == '1';
     const char *fs = getenv("DBSP_FORCE_SHARD");
     c->force_shard = fs && fs[0] == '1';
-    HIP_CHECK_ST(hipMalloc(&c->d_len, 16 * sizeof(int64_t)));
+    HIP_CHECK_ST(hipMalloc(&c->d_len, 20 * sizeof(int64_t)));
     c->arena_sz = (size_t)512 << 20;
     c->arena_half = c->arena_sz / 2;
     if (hipMalloc(&c->arena, c->arena_sz) != hipSuccess) {
         c->arena = nullptr;
         c->arena_sz = 0;
     }
-    HIP_CHECK_ST(hipHostMalloc(&c->h_len, 16 * sizeof(int64_t)));
+    HIP_CHECK_ST(hipHostMalloc(&c->h_len, 20 * sizeof(int64_t)));
     *out = c;
     return DBSP_OK;
 }
@@ -920,8 +921,10 @@ static inline bool sharding_on(dbsp_ctx *c) {
 }
 
 // shard + exchange + rebuild: the full shard() operator
-// (shard.rs:88-199: hash-split, exchange, re-consolidate)
-static dbsp_status shard_exchange(dbsp_ctx *c, DevBatch local, DevBatch &out) {
+// (shard.rs:88-199: hash-split, exchange, re-consolidate).  hint > 0
+// enables the fixed-frame fast path (see shard_exchange_pair below).
+static dbsp_status shard_exchange(dbsp_ctx *c, DevBatch local, DevBatch &out,
+                                  int64_t hint = 0) {
     if (!sharding_on(c)) {
         out = local;
         return DBSP_OK;
@@ -932,6 +935,48 @@ static dbsp_status shard_exchange(dbsp_ctx *c, DevBatch local, DevBatch &out) {
     TRY(dbspk::shard_rows(c->stream, local.k, local.v, local.w, local.n,
                           c->world, parts.k, parts.v, parts.w, offsets));
     free_batch(c, local);
+    if (hint > 0 && c->world <= 8 && c->comm) {
+        const int64_t P = 2 * hint / c->world + 512;
+        const int64_t S = 2 + 3 * P + 3;  // degenerate pair: stream1 cap 1
+        uint64_t *fsend, *frecv;
+        HIP_CHECK_ST(dbspk::cache_malloc((void **)&fsend,
+                                         (size_t)c->world * S * 8, c->stream));
+        HIP_CHECK_ST(dbspk::cache_malloc((void **)&frecv,
+                                         (size_t)c->world * S * 8, c->stream));
+        dbspk::FramePairArgs fa{};
+        fa.p0k = parts.k; fa.p0v = parts.v; fa.p0w = parts.w;
+        fa.p1k = parts.k; fa.p1v = parts.v; fa.p1w = parts.w;
+        for (int r = 0; r <= c->world; r++) {
+            fa.off0[r] = offsets[r];
+            fa.off1[r] = 0;  // empty stream1
+        }
+        fa.world = c->world;
+        fa.P0 = P;
+        fa.P1 = 1;
+        fa.frame = fsend;
+        TRY(dbspk::frames_pack_pair(c->stream, fa));
+        if (ncclAllToAll(fsend, frecv, (size_t)S, ncclUint64, c->comm,
+                         c->stream) != ncclSuccess)
+            return DBSP_ERR_INTERNAL;
+        DevBatch r0, r1;
+        TRY(alloc_batch(c, c->world * P, r0, true));
+        TRY(alloc_batch(c, c->world, r1, true));
+        TRY(dbspk::frames_unpack_pair(c->stream, frecv, c->world, P, 1, r0.k,
+                                      r0.v, r0.w, r1.k, r1.v, r1.w,
+                                      c->d_len + 16, c->d_len + 17));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 16, c->d_len + 16, 2 * 8,
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        HIP_CHECK_ST(dbspk::cache_free(fsend, c->stream));
+        HIP_CHECK_ST(dbspk::cache_free(frecv, c->stream));
+        if (c->h_len[16] >= 0) {
+            r0.n = c->h_len[16];
+            free_batch(c, parts);
+            TRY(sort_consolidate_batch(c, r0, out));
+            return DBSP_OK;
+        }
+        // overflow sentinel: dynamic replay below (all ranks together)
+    }
     int64_t send_counts[64];
     for (int r = 0; r < c->world; r++) send_counts[r] = offsets[r + 1] - offsets[r];
     DevBatch recv;
@@ -1147,7 +1192,8 @@ static dbsp_status sort_two_small(dbsp_ctx *c, DevBatch rawA, DevBatch rawB,
 // data phase (the per-tick collective latency is the scaling cost at 40k-event
 // ticks; fusing the two sides halves it)
 static dbsp_status shard_exchange_pair(dbsp_ctx *c, DevBatch l0, DevBatch l1,
-                                       DevBatch &o0, DevBatch &o1) {
+                                       DevBatch &o0, DevBatch &o1,
+                                       int64_t hint0 = 0, int64_t hint1 = 0) {
     if (!sharding_on(c)) {
         o0 = l0;
         o1 = l1;
@@ -1165,6 +1211,64 @@ static dbsp_status shard_exchange_pair(dbsp_ctx *c, DevBatch l0, DevBatch l1,
     p1.n = l1.n;
     free_batch(c, l0);
     free_batch(c, l1);
+    // FAST PATH: fixed-frame exchange.  hint0/hint1 = the caller's
+    // rank-uniform expectation of each stream's GLOBAL rows this tick (from
+    // the tick's event count and the query's mix), so every rank derives
+    // the SAME per-peer capacity and the equal-count ncclAllToAll is a
+    // legal collective.  Overflow (per-peer count above capacity, e.g.
+    // hot-key skew) travels in the frame header; every rank sees -1 totals
+    // at its sync below and replays through the dynamic path — identical
+    // collective sequence on all ranks, so no deadlock.
+    if (hint0 > 0 && hint1 > 0 && world <= 8 && c->comm) {
+        const int64_t P0 = 2 * hint0 / world + 512;
+        const int64_t P1 = 2 * hint1 / world + 512;
+        const int64_t S = 2 + 3 * P0 + 3 * P1;
+        uint64_t *fsend, *frecv;
+        HIP_CHECK_ST(dbspk::cache_malloc((void **)&fsend,
+                                         (size_t)world * S * 8, c->stream));
+        HIP_CHECK_ST(dbspk::cache_malloc((void **)&frecv,
+                                         (size_t)world * S * 8, c->stream));
+        dbspk::FramePairArgs fa{};
+        fa.p0k = p0.k; fa.p0v = p0.v; fa.p0w = p0.w;
+        fa.p1k = p1.k; fa.p1v = p1.v; fa.p1w = p1.w;
+        for (int r = 0; r <= world; r++) {
+            fa.off0[r] = off0[r];
+            fa.off1[r] = off1[r];
+        }
+        fa.world = world;
+        fa.P0 = P0;
+        fa.P1 = P1;
+        fa.frame = fsend;
+        TRY(dbspk::frames_pack_pair(c->stream, fa));
+        if (ncclAllToAll(fsend, frecv, (size_t)S, ncclUint64, c->comm,
+                         c->stream) != ncclSuccess)
+            return DBSP_ERR_INTERNAL;
+        DevBatch r0, r1;
+        TRY(alloc_batch(c, world * P0, r0, true));
+        TRY(alloc_batch(c, world * P1, r1, true));
+        TRY(dbspk::frames_unpack_pair(c->stream, frecv, world, P0, P1, r0.k,
+                                      r0.v, r0.w, r1.k, r1.v, r1.w,
+                                      c->d_len + 16, c->d_len + 17));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 16, c->d_len + 16, 2 * 8,
+                                    hipMemcpyDeviceToHost, c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        HIP_CHECK_ST(dbspk::cache_free(fsend, c->stream));
+        HIP_CHECK_ST(dbspk::cache_free(frecv, c->stream));
+        const int64_t tot0 = c->h_len[16], tot1 = c->h_len[17];
+        if (tot0 >= 0 && tot1 >= 0) {
+            r0.n = tot0;
+            r1.n = tot1;
+            if (r0.n <= 8192 && r1.n <= 8192) {
+                TRY(sort_two_small(c, r0, r1, o0, o1));
+            } else {
+                TRY(sort_consolidate_batch(c, r0, o0));
+                TRY(sort_consolidate_batch(c, r1, o1));
+            }
+            return DBSP_OK;
+        }
+        // overflow sentinel: fall through to the dynamic exchange on the
+        // retained partitioned buffers (all ranks take this same branch)
+    }
     // counts: 2 int64 per pair, one grouped phase
     int64_t send_cnt[128], recv_cnt[128];
     for (int r = 0; r < world; r++) {
@@ -1302,14 +1406,18 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
     }
     // worker sharding: co-locate keys across ranks (shard.rs:88)
     if (sharding_on(c)) {
+        // frame hints: the generator's event-kind schedule is deterministic
+        // (person:auction:bid = 1:3:46, config.rs:128-143), so per-stream
+        // global row counts are known from the tick's event count alone
         if (want_two) {
             DevBatch s0, s1;
-            TRY(shard_exchange_pair(c, d0, d1, s0, s1));
+            TRY(shard_exchange_pair(c, d0, d1, s0, s1, n / 50 + 64,
+                                    3 * n / 50 + 64));
             d0 = s0;
             d1 = s1;
         } else {
             DevBatch s0;
-            TRY(shard_exchange(c, d0, s0));
+            TRY(shard_exchange(c, d0, s0, n));
             d0 = s0;
         }
     }
@@ -2379,7 +2487,8 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     free_batch(c, war);
     if (sharding_on(c)) {
         DevBatch t1, t2;
-        TRY(shard_exchange_pair(c, dWP, dWA, t1, t2));
+        TRY(shard_exchange_pair(c, dWP, dWA, t1, t2, 4 * n / 50 + 64,
+                                12 * n / 50 + 64));
         dWP = t1;
         dWA = t2;
     }
@@ -2679,7 +2788,7 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     free_batch(c, wbr);
     if (sharding_on(c)) {
         DevBatch t1;
-        TRY(shard_exchange(c, dWB, t1));
+        TRY(shard_exchange(c, dWB, t1, 2 * n));
         dWB = t1;
     }
     // aggregate_linear: input trace includes this tick (trace.rs TraceAppend)
@@ -2694,7 +2803,7 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     TRY(map_sorted(c, dCounts, 2, dMaxIn));
     if (sharding_on(c)) {
         DevBatch t1;
-        TRY(shard_exchange(c, dMaxIn, t1));  // unit key () lives on one rank
+        TRY(shard_exchange(c, dMaxIn, t1, n / 8 + 64));  // unit key () on one rank
         dMaxIn = t1;
     }
     DevBatch dMaxOut;
@@ -2728,7 +2837,7 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     TRY(map_sorted(c, dCounts, 3, dBC));    // (auction,count) -> (count,auction)
     if (sharding_on(c)) {
         DevBatch t1, t2;
-        TRY(shard_exchange_pair(c, dMaxZ, dBC, t1, t2));
+        TRY(shard_exchange_pair(c, dMaxZ, dBC, t1, t2, 64, n / 4 + 64));
         dMaxZ = t1;
         dBC = t2;
     }

@@ -3046,6 +3046,132 @@ dbsp_status shard_rows_pair(hipStream_t s, const uint64_t *k0,
     return DBSP_OK;
 }
 
+// ---------------------------------------------------------------------------
+// fixed-frame pair exchange (see kernels_iface.hpp FramePairArgs): ONE pack
+// kernel + ONE equal-count ncclAllToAll + ONE unpack kernel replace the
+// counts-allgather, its host sync, and the ~6*world-call grouped send/recv
+// mesh of the dynamic path (r01_q3_shard_kernel_trace.txt attributed the 2x
+// sharded-tick cost to exactly that host latency).
+// ---------------------------------------------------------------------------
+
+__global__ void k_frames_pack_pair(FramePairArgs a) {
+    const int64_t S = 2 + 3 * a.P0 + 3 * a.P1;  // segment words
+    const int64_t total = (int64_t)a.world * S;
+    for (int64_t x = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         x < total; x += gridDim.x * (int64_t)blockDim.x) {
+        const int r = (int)(x / S);
+        const int64_t q = x % S;
+        const int64_t n0 = a.off0[r + 1] - a.off0[r];
+        const int64_t n1 = a.off1[r + 1] - a.off1[r];
+        uint64_t val = 0;
+        if (q == 0) {
+            val = (uint64_t)n0;  // true count: > P0 flags sender overflow
+        } else if (q == 1) {
+            val = (uint64_t)n1;
+        } else if (q - 2 < 3 * a.P0) {
+            const int64_t p = q - 2;
+            const int col = (int)(p / a.P0);
+            const int64_t i = p % a.P0;
+            if (i < n0 && i < a.P0) {
+                const uint64_t *src = col == 0   ? a.p0k
+                                      : col == 1 ? a.p0v
+                                                 : (const uint64_t *)a.p0w;
+                val = src[a.off0[r] + i];
+            }
+        } else {
+            const int64_t p = q - 2 - 3 * a.P0;
+            const int col = (int)(p / a.P1);
+            const int64_t i = p % a.P1;
+            if (i < n1 && i < a.P1) {
+                const uint64_t *src = col == 0   ? a.p1k
+                                      : col == 1 ? a.p1v
+                                                 : (const uint64_t *)a.p1w;
+                val = src[a.off1[r] + i];
+            }
+        }
+        a.frame[x] = val;
+    }
+}
+
+__global__ void k_frames_unpack_pair(const uint64_t *frame, int world,
+                                     int64_t P0, int64_t P1, uint64_t *r0k,
+                                     uint64_t *r0v, int64_t *r0w,
+                                     uint64_t *r1k, uint64_t *r1v,
+                                     int64_t *r1w, int64_t *d_tot0,
+                                     int64_t *d_tot1) {
+    const int64_t S = 2 + 3 * P0 + 3 * P1;
+    // per-thread recompute of the <=8 headers (cached) + prefixes
+    int64_t c0[8], c1[8], pre0[9], pre1[9];
+    bool lost = false;
+    pre0[0] = 0;
+    pre1[0] = 0;
+    for (int r = 0; r < world; r++) {
+        c0[r] = (int64_t)frame[(int64_t)r * S];
+        c1[r] = (int64_t)frame[(int64_t)r * S + 1];
+        if (c0[r] > P0 || c1[r] > P1 || c0[r] < 0 || c1[r] < 0) lost = true;
+        pre0[r + 1] = pre0[r] + (lost ? 0 : c0[r]);
+        pre1[r + 1] = pre1[r] + (lost ? 0 : c1[r]);
+    }
+    if (lost) {
+        // sender overflow somewhere: publish the sentinel; every rank sees
+        // its own copy at the next sync and replays the dynamic exchange
+        if (blockIdx.x == 0 && threadIdx.x == 0) {
+            *d_tot0 = -1;
+            *d_tot1 = -1;
+        }
+        return;
+    }
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        *d_tot0 = pre0[world];
+        *d_tot1 = pre1[world];
+    }
+    const int64_t total = (int64_t)world * S;
+    for (int64_t x = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         x < total; x += gridDim.x * (int64_t)blockDim.x) {
+        const int r = (int)(x / S);
+        const int64_t q = x % S;
+        if (q < 2) continue;
+        if (q - 2 < 3 * P0) {
+            const int64_t p = q - 2;
+            const int col = (int)(p / P0);
+            const int64_t i = p % P0;
+            if (i < c0[r]) {
+                const uint64_t v = frame[x];
+                if (col == 0) r0k[pre0[r] + i] = v;
+                else if (col == 1) r0v[pre0[r] + i] = v;
+                else r0w[pre0[r] + i] = (int64_t)v;
+            }
+        } else {
+            const int64_t p = q - 2 - 3 * P0;
+            const int col = (int)(p / P1);
+            const int64_t i = p % P1;
+            if (i < c1[r]) {
+                const uint64_t v = frame[x];
+                if (col == 0) r1k[pre1[r] + i] = v;
+                else if (col == 1) r1v[pre1[r] + i] = v;
+                else r1w[pre1[r] + i] = (int64_t)v;
+            }
+        }
+    }
+}
+
+dbsp_status frames_pack_pair(hipStream_t s, const FramePairArgs &a) {
+    const int64_t total = (int64_t)a.world * (2 + 3 * a.P0 + 3 * a.P1);
+    k_frames_pack_pair<<<grid_for(total), BLK, 0, s>>>(a);
+    return DBSP_OK;
+}
+
+dbsp_status frames_unpack_pair(hipStream_t s, const uint64_t *frame,
+                               int world, int64_t P0, int64_t P1,
+                               uint64_t *r0k, uint64_t *r0v, int64_t *r0w,
+                               uint64_t *r1k, uint64_t *r1v, int64_t *r1w,
+                               int64_t *d_tot0, int64_t *d_tot1) {
+    const int64_t total = (int64_t)world * (2 + 3 * P0 + 3 * P1);
+    k_frames_unpack_pair<<<grid_for(total), BLK, 0, s>>>(
+        frame, world, P0, P1, r0k, r0v, r0w, r1k, r1v, r1w, d_tot0, d_tot1);
+    return DBSP_OK;
+}
+
 dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
                                  int64_t n, int query, uint64_t *k0,
                                  uint64_t *v0, int64_t *w0, uint64_t *k1,

@@ -132,6 +132,31 @@ dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
                                  int64_t n, int query, uint64_t *k0,
                                  uint64_t *v0, int64_t *w0, uint64_t *k1,
                                  uint64_t *v1, int64_t *w1, uint64_t *ctr);
+
+// fixed-frame pair exchange (engine exchange fast path): both partitioned
+// streams packed into per-peer segments of fixed capacity P0/P1 rows with a
+// 2-word count header, so ONE equal-count ncclAllToAll replaces the
+// counts-allgather + host sync + grouped send/recv mesh.  Segment layout
+// (u64 words): [cnt0, cnt1, k0(P0), v0(P0), w0(P0), k1(P1), v1(P1), w1(P1)].
+// A count above its capacity travels in the header as-is; the unpack kernel
+// then writes -1 totals (sentinel) and the engine replays the exchange
+// through the dynamic alltoallv path on every rank.
+struct FramePairArgs {
+    const uint64_t *p0k, *p0v;
+    const int64_t *p0w;
+    const uint64_t *p1k, *p1v;
+    const int64_t *p1w;
+    int64_t off0[9], off1[9];  // host partition offsets (nshards <= 8)
+    int world;
+    int64_t P0, P1;
+    uint64_t *frame;
+};
+dbsp_status frames_pack_pair(hipStream_t s, const FramePairArgs &a);
+dbsp_status frames_unpack_pair(hipStream_t s, const uint64_t *frame,
+                               int world, int64_t P0, int64_t P1,
+                               uint64_t *r0k, uint64_t *r0v, int64_t *r0w,
+                               uint64_t *r1k, uint64_t *r1v, int64_t *r1w,
+                               int64_t *d_tot0, int64_t *d_tot1);
 dbsp_status sort_cons_small_batch(hipStream_t s, const SortArgs &args);
 // min/max of k and v (one sync): mm = {kmin, vmin, kmax, vmax}
 dbsp_status minmax_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
