@@ -1382,7 +1382,7 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
     } else {
         DevBatch rawA, rawB, oA, oB;
         TRY(build_deltas_chain(e, d_ev, n, rawA, rawB, oA, oB));
-        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 16 * sizeof(int64_t),
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 18 * sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
         if (c->h_len[10] < 0 || c->h_len[11] < 0) {
@@ -1450,6 +1450,76 @@ static dbsp_status build_deltas_chain(dbsp_engine *e, const dbsp_event *d_ev,
     sa.kin[1] = rawB.k; sa.vin[1] = rawB.v; sa.win[1] = rawB.w;
     sa.n_dev[0] = c->d_len + 8;
     sa.n_dev[1] = c->d_len + 9;
+    sa.tk[0] = sA.k; sa.tv[0] = sA.v; sa.tw[0] = sA.w;
+    sa.tk[1] = sB.k; sa.tv[1] = sB.v; sa.tw[1] = sB.w;
+    sa.ok[0] = oA.k; sa.ov[0] = oA.v; sa.ow[0] = oA.w;
+    sa.ok[1] = oB.k; sa.ov[1] = oB.v; sa.ow[1] = oB.w;
+    sa.d_len = c->d_len + 10;
+    {
+        ScopedTimer t(c, 0, (double)n * 48.0);
+        TRY(dbspk::sort_cons_small_batch(c->stream, sa));
+    }
+    oA.n = -1;  // pending: d_len[10], d_len[11]
+    oB.n = -1;
+    return DBSP_OK;
+}
+
+// Sharded chained front (q3-class pair streams): flatmap -> hash-scatter
+// straight into fixed frames (device counters, no pre-sort — shard.rs
+// consolidates POST-exchange) -> one equal-count ncclAllToAll -> unpack
+// (receive totals at d_len[16..17]) -> speculative fused sorts reading those
+// totals.  The round-1 sharded tick ran the explicit path with ~7 host
+// syncs (partition offsets, counts allgather, totals, sorts...); this one
+// syncs ONCE at the caller like the unsharded chain.  Overflow anywhere
+// (frame capacity, fused-sort capacity) surfaces as -1 lengths and the
+// caller replays dynamically — every rank sees the same sentinel at the
+// same sync, so the collective sequence stays rank-uniform.
+static dbsp_status build_deltas_chain_sharded(
+    dbsp_engine *e, const dbsp_event *d_ev, int64_t n, DevBatch &rawA,
+    DevBatch &rawB, DevBatch &recvA, DevBatch &recvB, DevBatch &oA,
+    DevBatch &oB) {
+    dbsp_ctx *c = e->ctx;
+    const int world = c->world;
+    const int64_t cap = n > 0 ? n : 1;
+    TRY(alloc_batch(c, cap, rawA, true));
+    TRY(alloc_batch(c, cap, rawB, true));
+    TRY(dbspk::flatmap_events_chain(c->stream, d_ev, n, e->query, rawA.k,
+                                    rawA.v, rawA.w, rawB.k, rawB.v, rawB.w,
+                                    (uint64_t *)(c->d_len + 8)));
+    // frame capacities from the deterministic event mix (config.rs:128-143)
+    const int64_t P0 = 2 * (n / 50 + 64) / world + 512;
+    const int64_t P1 = 2 * (3 * n / 50 + 64) / world + 512;
+    const int64_t S = 2 + 3 * P0 + 3 * P1;
+    uint64_t *fsend, *frecv;
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&fsend, (size_t)world * S * 8,
+                                     c->stream));
+    HIP_CHECK_ST(dbspk::cache_malloc((void **)&frecv, (size_t)world * S * 8,
+                                     c->stream));
+    TRY(dbspk::shard_frames_chain(c->stream, rawA.k, rawA.v, rawA.w,
+                                  c->d_len + 8, rawB.k, rawB.v, rawB.w,
+                                  c->d_len + 9, world, P0, P1, cap, fsend));
+    if (ncclAllToAll(fsend, frecv, (size_t)S, ncclUint64, c->comm,
+                     c->stream) != ncclSuccess)
+        return DBSP_ERR_INTERNAL;
+    const int64_t capA = world * P0, capB = world * P1;
+    TRY(alloc_batch(c, capA, recvA, true));
+    TRY(alloc_batch(c, capB, recvB, true));
+    TRY(dbspk::frames_unpack_pair(c->stream, frecv, world, P0, P1, recvA.k,
+                                  recvA.v, recvA.w, recvB.k, recvB.v, recvB.w,
+                                  c->d_len + 16, c->d_len + 17));
+    HIP_CHECK_ST(dbspk::cache_free(fsend, c->stream));
+    HIP_CHECK_ST(dbspk::cache_free(frecv, c->stream));
+    DevBatch sA, sB;
+    TRY(alloc_batch(c, capA, sA, true));
+    TRY(alloc_batch(c, capB, sB, true));
+    TRY(alloc_batch(c, capA, oA));
+    TRY(alloc_batch(c, capB, oB));
+    SortArgs sa{};
+    sa.nb = 2;
+    sa.kin[0] = recvA.k; sa.vin[0] = recvA.v; sa.win[0] = recvA.w;
+    sa.kin[1] = recvB.k; sa.vin[1] = recvB.v; sa.win[1] = recvB.w;
+    sa.n_dev[0] = c->d_len + 16;
+    sa.n_dev[1] = c->d_len + 17;
     sa.tk[0] = sA.k; sa.tv[0] = sA.v; sa.tw[0] = sA.w;
     sa.tk[1] = sB.k; sa.tv[1] = sB.v; sa.tw[1] = sB.w;
     sa.ok[0] = oA.k; sa.ov[0] = oA.v; sa.ow[0] = oA.w;
@@ -1926,16 +1996,21 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         free_batch(c, e->front.oB);
         e->front.pending = false;
     }
+    const bool shard_chain =
+        sharding_on(c) && c->world <= 8 && n <= 131072 && e->spec_fail < 3;
     const bool chain =
-        use_front ||
+        use_front || shard_chain ||
         (!sharding_on(c) && n <= 131072 && e->spec_fail < 3);
-    DevBatch dA, dP, rawA, rawP;
+    DevBatch dA, dP, rawA, rawP, recvA, recvP;
     if (use_front) {
         e->front.pending = false;
         rawA = e->front.rawA;
         rawP = e->front.rawB;
         dA = e->front.oA;
         dP = e->front.oB;
+    } else if (shard_chain) {
+        TRY(build_deltas_chain_sharded(e, d_ev, n, rawA, rawP, recvA, recvP,
+                                       dA, dP));
     } else if (chain) {
         TRY(build_deltas_chain(e, d_ev, n, rawA, rawP, dA, dP));
     } else {
@@ -2022,14 +2097,14 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                 // arena exhausted mid-speculation: fetch the real lengths and
                 // rebuild the plans the explicit way
                 HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
-                                            16 * sizeof(int64_t),
+                                            18 * sizeof(int64_t),
                                             hipMemcpyDeviceToHost, c->stream));
                 HIP_CHECK_ST(hipStreamSynchronize(c->stream));
             } else {
                 jca.d_total = c->d_len;
                 if (jca.np > 0) TRY(dbspk::join_count_scan_batch(c->stream, jca));
                 HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
-                                            16 * sizeof(int64_t),
+                                            18 * sizeof(int64_t),
                                             hipMemcpyDeviceToHost, c->stream));
                 if (spec && jca.np > 0) {
                     (void)hipEventRecord(c->ev_sync2, c->stream);
@@ -2103,12 +2178,33 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             }
             e->spec_fail++;
             chain_emits = false;  // the chained emits bailed on the flag
+            free_batch(c, dA);
+            free_batch(c, dP);
+            if (shard_chain) {
+                // sharded recovery: frame overflow (totals -1, all ranks see
+                // it here) replays the dynamic exchange from the raw flatmap
+                // outputs; a plain sort overflow re-sorts the RECEIVED rows
+                if (c->h_len[16] < 0 || c->h_len[17] < 0) {
+                    rawA.n = c->h_len[8];
+                    rawP.n = c->h_len[9];
+                    TRY(shard_exchange_pair(c, rawA, rawP, dA, dP));
+                } else {
+                    recvA.n = c->h_len[16];
+                    recvP.n = c->h_len[17];
+                    TRY(sort_consolidate_batch(c, recvA, dA));
+                    TRY(sort_consolidate_batch(c, recvP, dP));
+                    recvA = DevBatch{};
+                    recvP = DevBatch{};
+                }
+                rawA = DevBatch{};
+                rawP = DevBatch{};
+                spec = false;
+                continue;
+            }
             // speculation lost: re-sort from the raw flatmap outputs with the
             // now-known lengths, then redo the planning explicitly
             rawA.n = c->h_len[8];
             rawP.n = c->h_len[9];
-            free_batch(c, dA);
-            free_batch(c, dP);
             TRY(sort_consolidate_batch(c, rawA, dA));
             TRY(sort_consolidate_batch(c, rawP, dP));
             rawA = DevBatch{};
@@ -2158,8 +2254,11 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         return DBSP_OK;
     };
     std::function<dbsp_status()> hook = [&]() -> dbsp_status {
-        if (!e->next_ev || e->next_n < 0 || e->next_n > 131072 ||
-            e->spec_fail >= 3)
+        // sharded ranks: no pipelined fronts (the front would enqueue an
+        // alltoall whose cross-rank ordering vs this tick's tail collectives
+        // is not guaranteed rank-uniform)
+        if (sharding_on(c) || !e->next_ev || e->next_n < 0 ||
+            e->next_n > 131072 || e->spec_fail >= 3)
             return DBSP_OK;
         const size_t save_base = c->arena_base, save_off = c->arena_off;
         c->arena_base =
@@ -2330,7 +2429,7 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                                            c->d_len + 11, e->d_bounds, tableA,
                                            c->d_len + 13));
         }
-        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 16 * sizeof(int64_t),
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 18 * sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipMemcpyAsync(e->h_bounds, e->d_bounds,
                                     6 * sizeof(uint64_t),
@@ -2440,7 +2539,7 @@ static dbsp_status q8_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                                                c->d_len + 13));
             }
             HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
-                                        16 * sizeof(int64_t),
+                                        18 * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
             HIP_CHECK_ST(hipStreamSynchronize(c->stream));
             if (c->h_len[12] > 0) {
@@ -2630,7 +2729,7 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                                          c->d_len + 8,
                                          (unsigned long long *)(c->d_len + 12)));
         }
-        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 16 * sizeof(int64_t),
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, 18 * sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
         free_batch(c, raw1);
@@ -2687,7 +2786,7 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                     c->d_len + 7));
             }
             HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
-                                        16 * sizeof(int64_t),
+                                        18 * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
             HIP_CHECK_ST(hipStreamSynchronize(c->stream));
             if (n_pending) {
@@ -2711,7 +2810,7 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                                         6 * sizeof(uint64_t),
                                         hipMemcpyDeviceToHost, c->stream));
             HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
-                                        16 * sizeof(int64_t),
+                                        18 * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
             HIP_CHECK_ST(hipStreamSynchronize(c->stream));
             if (n_pending) {

@@ -3161,6 +3161,58 @@ dbsp_status frames_pack_pair(hipStream_t s, const FramePairArgs &a) {
     return DBSP_OK;
 }
 
+// chained shard-to-frames: hash-partition BOTH raw streams straight into the
+// frame segments with atomic header tickets — no host offsets, no separate
+// partition/pack kernels, lengths read from the device (the chained tick's
+// flatmap counters).  A ticket beyond capacity skips the write; the header
+// keeps the true count, which the receiving unpack flags as the overflow
+// sentinel.
+__global__ void k_frames_init_headers(uint64_t *frame, int world, int64_t S) {
+    const int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < 2 * world) frame[(int64_t)(i / 2) * S + (i & 1)] = 0;
+}
+
+__global__ void k_shard_frames_chain(const uint64_t *k0, const uint64_t *v0,
+                                     const int64_t *w0, const int64_t *n0_dev,
+                                     const uint64_t *k1, const uint64_t *v1,
+                                     const int64_t *w1, const int64_t *n1_dev,
+                                     int world, int64_t P0, int64_t P1,
+                                     uint64_t *frame) {
+    const int64_t S = 2 + 3 * P0 + 3 * P1;
+    const int64_t n0 = *n0_dev, n1 = *n1_dev;
+    const int64_t total = n0 + n1;
+    for (int64_t x = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+         x < total; x += gridDim.x * (int64_t)blockDim.x) {
+        const bool s1 = x >= n0;
+        const int64_t i = s1 ? x - n0 : x;
+        const uint64_t key = s1 ? k1[i] : k0[i];
+        const int peer = (int)(dev_xxh3_u64(key, DBSP_HASH_SEED) % world);
+        uint64_t *seg = frame + (int64_t)peer * S;
+        const int64_t P = s1 ? P1 : P0;
+        const int64_t t = (int64_t)atomicAdd(
+            (unsigned long long *)(seg + (s1 ? 1 : 0)), 1ull);
+        if (t < P) {
+            uint64_t *base = seg + 2 + (s1 ? 3 * P0 : 0);
+            base[t] = key;
+            base[P + t] = s1 ? v1[i] : v0[i];
+            base[2 * P + t] = (uint64_t)(s1 ? w1[i] : w0[i]);
+        }
+    }
+}
+
+dbsp_status shard_frames_chain(hipStream_t s, const uint64_t *k0,
+                               const uint64_t *v0, const int64_t *w0,
+                               const int64_t *n0_dev, const uint64_t *k1,
+                               const uint64_t *v1, const int64_t *w1,
+                               const int64_t *n1_dev, int world, int64_t P0,
+                               int64_t P1, int64_t n_cap, uint64_t *frame) {
+    const int64_t S = 2 + 3 * P0 + 3 * P1;
+    k_frames_init_headers<<<dim3(1), dim3(64), 0, s>>>(frame, world, S);
+    k_shard_frames_chain<<<grid_for(n_cap > 0 ? n_cap : 1), BLK, 0, s>>>(
+        k0, v0, w0, n0_dev, k1, v1, w1, n1_dev, world, P0, P1, frame);
+    return DBSP_OK;
+}
+
 dbsp_status frames_unpack_pair(hipStream_t s, const uint64_t *frame,
                                int world, int64_t P0, int64_t P1,
                                uint64_t *r0k, uint64_t *r0v, int64_t *r0w,

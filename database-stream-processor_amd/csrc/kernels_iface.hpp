@@ -152,6 +152,14 @@ struct FramePairArgs {
     uint64_t *frame;
 };
 dbsp_status frames_pack_pair(hipStream_t s, const FramePairArgs &a);
+// chained variant: hash-partition both RAW streams straight into the frames
+// (lengths read from device counters; n_cap bounds the launch grid)
+dbsp_status shard_frames_chain(hipStream_t s, const uint64_t *k0,
+                               const uint64_t *v0, const int64_t *w0,
+                               const int64_t *n0_dev, const uint64_t *k1,
+                               const uint64_t *v1, const int64_t *w1,
+                               const int64_t *n1_dev, int world, int64_t P0,
+                               int64_t P1, int64_t n_cap, uint64_t *frame);
 dbsp_status frames_unpack_pair(hipStream_t s, const uint64_t *frame,
                                int world, int64_t P0, int64_t P1,
                                uint64_t *r0k, uint64_t *r0v, int64_t *r0w,
