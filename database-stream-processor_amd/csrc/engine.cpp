@@ -1008,6 +1008,10 @@ struct dbsp_engine {
     // q5 state
     Spine bt_int, wb_int, counts_int, bc_int;
     DevBatch maxin_int, maxout_int, maxz_int;
+    // q4 state (queries/q4.rs: join + per-auction Max + per-category Average)
+    Spine q4_a_int, q4_b_int;      // auction / bid traces (keyed by auction)
+    DevBatch q4_maxin, q4_maxout;  // consolidated max in/out integrals
+    Spine q4_avg_int, q4_avgout;   // packed (sum<<20|count) integral + output
     // C5 state (query 100; BASELINE configs[4]: 1B-row indexed trace x
     // 10M-row delta incremental join + f64 sum aggregate)
     Spine c5_trace;   // (k, f64-bits val, +1 i64) join-side trace
@@ -1056,7 +1060,8 @@ struct dbsp_engine {
 
 extern "C" dbsp_status dbsp_engine_create(dbsp_engine **out, dbsp_ctx *ctx,
                                           int query, int rank, int world) {
-    if (query != 0 && query != 3 && query != 5 && query != 8 && query != 100)
+    if (query != 0 && query != 3 && query != 4 && query != 5 &&
+        query != 8 && query != 100)
         return DBSP_ERR_INVALID;
 
     dbsp_engine *e = new dbsp_engine();
@@ -1084,8 +1089,12 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
     dbsp_ctx *c = e->ctx;
     for (Spine *s : {&e->a_int, &e->p_int, &e->pt_int, &e->at_int, &e->wp_int,
                      &e->wa_int, &e->bt_int, &e->wb_int, &e->counts_int,
-                     &e->bc_int, &e->c5_trace, &e->c5_wint, &e->c5_out})
+                     &e->bc_int, &e->c5_trace, &e->c5_wint, &e->c5_out,
+                     &e->q4_a_int, &e->q4_b_int, &e->q4_avg_int,
+                     &e->q4_avgout})
         s->clear(c);
+    free_batch(c, e->q4_maxin);
+    free_batch(c, e->q4_maxout);
     free_batch(c, e->maxin_int);
     free_batch(c, e->maxout_int);
     free_batch(c, e->maxz_int);
@@ -1411,8 +1420,11 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
         // global row counts are known from the tick's event count alone
         if (want_two) {
             DevBatch s0, s1;
-            TRY(shard_exchange_pair(c, d0, d1, s0, s1, n / 50 + 64,
-                                    3 * n / 50 + 64));
+            const int64_t h0 =
+                e->query == 4 ? 3 * n / 50 + 64 : n / 50 + 64;
+            const int64_t h1 =
+                e->query == 4 ? 46 * n / 50 + 64 : 3 * n / 50 + 64;
+            TRY(shard_exchange_pair(c, d0, d1, s0, s1, h0, h1));
             d0 = s0;
             d1 = s1;
         } else {
@@ -3056,6 +3068,117 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
 }
 
 // ---------------------------------------------------------------------------
+// q4 (queries/q4.rs): average winning-bid price per category.
+//   auctions ⋈ bids on auction id with the bid-validity window applied in
+//   the join_func (q4.rs:58-68; filtered pairs emit weight 0) →
+//   Max per (auction, category) over the consolidated integral
+//   (aggregate(Max), max.rs:36-55) with upsert retractions →
+//   Average per category as ONE linear aggregate over the packed value
+//   (sum<<20)+count (price < 2^20, per-category count < 2^20, so the packed
+//   i64 weight sum is exact), divided at the output map (average.rs).
+// Explicit per-op path (no chained speculation yet — q4 is not a headline
+// config; correctness and coverage first).
+// ---------------------------------------------------------------------------
+
+static dbsp_status q4_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
+    dbsp_ctx *c = e->ctx;
+    engine_free_output(e);
+    DevBatch dA, dB;  // auction / bid deltas, keyed by auction id
+    TRY(build_deltas(e, d_ev, n, dA, dB, true));
+    // bilinear expansion against PREVIOUS traces:
+    //   dB ⋈ A_prev + dA ⋈ B_prev + dA ⋈ dB   (join.rs:217-292)
+    std::vector<DevBatch> outs;
+    TRY(join_vs_spine(c, dB, e->q4_a_int, DBSP_PROJ_Q4_BID_X_AUC, 0, outs));
+    TRY(join_vs_spine(c, dA, e->q4_b_int, DBSP_PROJ_Q4_AUC_X_BID, 0, outs));
+    if (dA.n > 0 && dB.n > 0) {
+        TraceArgs t{};
+        t.nb = 1;
+        t.k[0] = dB.k; t.v[0] = dB.v; t.w[0] = dB.w; t.n[0] = dB.n;
+        DevBatch o;
+        ScopedTimer timer(c, 2, (double)dA.n * 24.0);
+        TRY(dbspk::join_spine_rows(c->stream, dA.k, dA.v, dA.w, dA.n, t,
+                                   DBSP_PROJ_Q4_AUC_X_BID, 0, &o.k, &o.v,
+                                   &o.w, &o.n));
+        if (o.n > 0) outs.push_back(o);
+        else free_batch(c, o);
+    }
+    DevBatch dWinIn;
+    TRY(finalize_raw(c, outs, dWinIn));
+    DevBatch dWin{};
+    if (dWinIn.n > 0) {
+        // max input integral is kept consolidated: the Max aggregator reads
+        // the key's whole value run (max.rs:36-55)
+        {
+            DevBatch cpy, m;
+            TRY(copy_batch(c, dWinIn, cpy));
+            TRY(merge_batches(c, e->q4_maxin, cpy, m));
+            free_batch(c, e->q4_maxin);
+            free_batch(c, cpy);
+            e->q4_maxin = m;
+        }
+        uint64_t *keys = nullptr;
+        int64_t nk = 0;
+        TRY(dbspk::unique_keys(c->stream, dWinIn.k, dWinIn.n, &keys, &nk));
+        DevBatch raw;
+        {
+            ScopedTimer timer(c, 3, 0.0);
+            TRY(dbspk::agg_max_upsert_rows(
+                c->stream, keys, nk, e->q4_maxin.k, e->q4_maxin.v,
+                e->q4_maxin.w, e->q4_maxin.n, e->q4_maxout.k, e->q4_maxout.v,
+                e->q4_maxout.w, e->q4_maxout.n, &raw.k, &raw.v, &raw.w,
+                &raw.n));
+        }
+        HIP_CHECK_ST(dbspk::cache_free(keys, c->stream));
+        TRY(sort_consolidate_batch(c, raw, dWin));
+        if (dWin.n > 0) {
+            DevBatch cpy, m;
+            TRY(copy_batch(c, dWin, cpy));
+            TRY(merge_batches(c, e->q4_maxout, cpy, m));
+            free_batch(c, e->q4_maxout);
+            free_batch(c, cpy);
+            e->q4_maxout = m;
+        }
+    }
+    free_batch(c, dWinIn);
+    // average per category: weigh into the packed (sum<<20)+count weight,
+    // aggregate linearly over the integral spine, upsert, divide at output
+    if (dWin.n > 0) {
+        DevBatch wraw, dAvgIn;
+        TRY(alloc_batch(c, dWin.n, wraw, true));
+        TRY(dbspk::map_rows(c->stream, dWin.k, dWin.v, dWin.w, dWin.n, 5,
+                            wraw.k, wraw.v, wraw.w));
+        wraw.n = dWin.n;
+        TRY(sort_consolidate_batch(c, wraw, dAvgIn));
+        free_batch(c, dWin);
+        if (dAvgIn.n > 0) {
+            DevBatch cpy;
+            TRY(copy_batch(c, dAvgIn, cpy));
+            TRY(e->q4_avg_int.insert(c, cpy));
+            DevBatch upd;
+            TRY(agg_linear_spine(c, dAvgIn, e->q4_avg_int, e->q4_avgout,
+                                 upd));
+            free_batch(c, dAvgIn);
+            if (upd.n > 0) {
+                DevBatch cpy2;
+                TRY(copy_batch(c, upd, cpy2));
+                TRY(e->q4_avgout.insert(c, cpy2));
+                // output: divide the packed sums (two packed values can map
+                // to one average, so consolidate after the map)
+                TRY(map_sorted(c, upd, 6, e->output));
+            }
+            free_batch(c, upd);
+        }
+    } else {
+        free_batch(c, dWin);
+    }
+    // TraceAppend both input deltas
+    TRY(e->q4_a_int.insert(c, dA));
+    TRY(e->q4_b_int.insert(c, dB));
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    return DBSP_OK;
+}
+
+// ---------------------------------------------------------------------------
 // C5 (query 100): synthetic 1B-row OrdIndexedZSet x 10M-row delta incremental
 // join with f64 sum aggregate (BASELINE configs[4]).  Per
 // tick: delta ⋈ trace carrying the trace's f64 val (JoinTrace::eval,
@@ -3185,6 +3308,7 @@ extern "C" dbsp_status dbsp_engine_step_staged(dbsp_engine *e, int64_t lo,
     const dbsp_event *d_ev = e->d_events + lo;
     switch (e->query) {
         case 3: return q3_step(e, d_ev, hi - lo);
+        case 4: return q4_step(e, d_ev, hi - lo);
         case 5: return q5_step(e, d_ev, hi - lo);
         case 8: return q8_step(e, d_ev, hi - lo);
     }
@@ -3229,6 +3353,7 @@ extern "C" dbsp_status dbsp_engine_step(dbsp_engine *e, const dbsp_event *events
     dbsp_status st = DBSP_OK;
     switch (e->query) {
         case 3: st = q3_step(e, d_ev, n); break;
+        case 4: st = q4_step(e, d_ev, n); break;
         case 5: st = q5_step(e, d_ev, n); break;
         case 8: st = q8_step(e, d_ev, n); break;
         default: st = DBSP_ERR_INVALID;

@@ -1089,10 +1089,31 @@ __global__ void k_join_count(const uint64_t *dk, int64_t nd, const uint64_t *tk,
     }
 }
 
-__device__ inline void proj_out(int proj, uint64_t param, uint64_t k,
+// Returns whether the pair is a real match (join_func filters, e.g. q4's
+// bid-validity window): an invalid pair still occupies its counted output
+// slot but emits weight 0, which the tick's consolidate then drops.
+__device__ inline bool proj_out(int proj, uint64_t param, uint64_t k,
                                 uint64_t v1, uint64_t v2, uint64_t &hi,
                                 uint64_t &lo) {
     switch (proj) {
+        case DBSP_PROJ_Q4_BID_X_AUC: {
+            // delta = bid (v1 = bid_dt<<20 | price), trace = auction
+            // (v2 = a_dt<<28 | (expires-a_dt)<<4 | category-10); emit
+            // ((auction<<4)|cat, price) when the bid falls in the auction's
+            // validity window (q4.rs:58-68)
+            const uint64_t bid_dt = v1 >> 20, price = v1 & 0xFFFFFull;
+            const uint64_t a_dt = v2 >> 28, dur = (v2 >> 4) & 0xFFFFFFull;
+            hi = (k << 4) | (v2 & 0xFull);
+            lo = price;
+            return bid_dt >= a_dt && bid_dt <= a_dt + dur;
+        }
+        case DBSP_PROJ_Q4_AUC_X_BID: {  // sides swapped
+            const uint64_t bid_dt = v2 >> 20, price = v2 & 0xFFFFFull;
+            const uint64_t a_dt = v1 >> 28, dur = (v1 >> 4) & 0xFFFFFFull;
+            hi = (k << 4) | (v1 & 0xFull);
+            lo = price;
+            return bid_dt >= a_dt && bid_dt <= a_dt + dur;
+        }
         case DBSP_PROJ_HI_V2_LO_V1: hi = v2; lo = v1; break;
         case DBSP_PROJ_HI_V1_LO_V2: hi = v1; lo = v2; break;
         case DBSP_PROJ_HI_K_LO_V1V2: hi = k; lo = (v1 << 32) | (v2 & 0xFFFFFFFFull); break;
@@ -1112,6 +1133,7 @@ __device__ inline void proj_out(int proj, uint64_t param, uint64_t k,
         case DBSP_PROJ_HI_K_LO_V2: hi = k; lo = v2; break;
         default: hi = 0; lo = 0; break;
     }
+    return true;
 }
 
 // emit over the OUTPUT index space: balanced regardless of per-key skew
@@ -1135,10 +1157,10 @@ __global__ void k_join_emit(const uint64_t *dk, const uint64_t *dv,
         int64_t j = o - (int64_t)offsets[i];
         int64_t t = (int64_t)starts[i] + j;
         uint64_t hi_o, lo_o;
-        proj_out(proj, param, dk[i], dv[i], tv[t], hi_o, lo_o);
+        const bool valid = proj_out(proj, param, dk[i], dv[i], tv[t], hi_o, lo_o);
         ok[o] = hi_o;
         ov[o] = lo_o;
-        ow[o] = dw[i] * tw[t];
+        ow[o] = valid ? dw[i] * tw[t] : 0;
     }
 }
 
@@ -1668,10 +1690,11 @@ __global__ void k_join_emit_chain(const uint64_t *dk, const uint64_t *dv,
         int64_t start = lower_bound_k(t.k[b], tb_n, key);
         int64_t ti = start + j;
         uint64_t hi_o, lo_o;
-        proj_out(proj, param, key, dv[i], t.v[b][ti], hi_o, lo_o);
+        const bool valid =
+            proj_out(proj, param, key, dv[i], t.v[b][ti], hi_o, lo_o);
         ok[base + o] = hi_o;
         ov[base + o] = lo_o;
-        ow[base + o] = dw[i] * t.w[b][ti];
+        ow[base + o] = valid ? dw[i] * t.w[b][ti] : 0;
     }
 }
 
@@ -1698,10 +1721,11 @@ __global__ void k_join_emit_multi(const uint64_t *dk, const uint64_t *dv,
         int64_t start = lower_bound_k(t.k[b], t.n[b], key);
         int64_t ti = start + j;
         uint64_t hi_o, lo_o;
-        proj_out(proj, param, key, dv[i], t.v[b][ti], hi_o, lo_o);
+        const bool valid =
+            proj_out(proj, param, key, dv[i], t.v[b][ti], hi_o, lo_o);
         ok[o] = hi_o;
         ov[o] = lo_o;
-        ow[o] = dw[i] * t.w[b][ti];
+        ow[o] = valid ? dw[i] * t.w[b][ti] : 0;
     }
 }
 
@@ -2217,6 +2241,24 @@ __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
                 v1[pos1] = (e.f1 << 8) | ((e.f2 & 0xF) << 4) | (e.f3 & 0xF);
                 w1[pos1] = e.w;
             }
+        } else if (query == 4) {
+            // q4.rs:45-56: auctions by id (validity window + category packed
+            // into the val), bids by auction (dt + price packed)
+            const bool p0 = act && e.kind == 1;
+            const uint64_t pos0 = wave_append((unsigned long long *)c0, p0);
+            if (p0) {
+                k0[pos0] = e.f0;
+                v0[pos0] = (e.f3 << 28) | (((e.f4 - e.f3) & 0xFFFFFFull) << 4) |
+                           (e.f2 & 0xFull);
+                w0[pos0] = e.w;
+            }
+            const bool p1 = act && e.kind == 2;
+            const uint64_t pos1 = wave_append((unsigned long long *)c1, p1);
+            if (p1) {
+                k1[pos1] = e.f0;
+                v1[pos1] = (e.f3 << 20) | (e.f2 & 0xFFFFFull);
+                w1[pos1] = e.w;
+            }
         } else if (query == 5) {
             // q5.rs:79-83: bids by time
             const bool p0 = act && e.kind == 2;
@@ -2251,6 +2293,11 @@ __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
 // mode 1: (k,v) -> (v, 0)                             [q8 auctions map / q5 windowed-bids map]
 // mode 2: (k,v) -> (0, v)                             [q5 map_index ((),count)]
 // mode 3: (k,v) -> (v, k)                             [q5 by_count map_index]
+// mode 5: (k=auction<<4|cat, v=price) -> (cat, 0) with w' = w*(price<<20|1)
+//         [q4 average weigh: one linear pass accumulates (sum<<20)+count
+//          exactly — price < 2^20 and per-category count < 2^20]
+// mode 6: (k=cat, v=(sum<<20)|count) -> (cat, sum/count)  [q4 average output,
+//          integer division as the reference's isize avg (average.rs)]
 __global__ void k_map(const uint64_t *k, const uint64_t *v, const int64_t *w,
                       int64_t n, int mode, uint64_t *ok, uint64_t *ov,
                       int64_t *ow) {
@@ -2268,6 +2315,17 @@ __global__ void k_map(const uint64_t *k, const uint64_t *v, const int64_t *w,
                 double f = *(const double *)&vv;
                 double wp = f * (double)w[i];
                 rw = *(const int64_t *)&wp;
+                break;
+            }
+            case 5:
+                rk = kk & 0xFull;
+                rv = 0;
+                rw = w[i] * (int64_t)((vv << 20) | 1ull);
+                break;
+            case 6: {
+                const int64_t cnt = (int64_t)(vv & 0xFFFFFull);
+                rk = kk;
+                rv = cnt > 0 ? (vv >> 20) / (uint64_t)cnt : 0;
                 break;
             }
             default: rk = vv; rv = kk; break;

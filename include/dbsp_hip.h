@@ -91,6 +91,13 @@ typedef enum {
     DBSP_PROJ_HI_K_LO_V2V1   = 7, /* generic pair join, swapped side: lo=v2<<32|v1 */
     DBSP_PROJ_HI_K_LO_V2     = 8, /* (k, v2): keeps the join key, takes the trace val
                                      (full 64-bit; the C5 join -> f64-sum pipeline) */
+    DBSP_PROJ_Q4_BID_X_AUC   = 9, /* q4.rs:58-68 join_func: delta=bid
+                                     (v=bid_dt<<20|price), trace=auction
+                                     (v=a_dt<<28|(expires-a_dt)<<4|(cat&0xF)):
+                                     emit ((auction<<4)|cat, price) iff
+                                     a_dt <= bid_dt <= expires, else weight 0
+                                     (dropped by the consolidate) */
+    DBSP_PROJ_Q4_AUC_X_BID   = 10, /* q4 swapped side */
 } dbsp_proj;
 
 /* ======================================================================

@@ -112,6 +112,26 @@ inline void proj_emit(std::vector<Row> &out, dbsp_proj proj, uint64_t param,
         case DBSP_PROJ_HI_V1_LO_K: hi = v1; lo = k; break;
         case DBSP_PROJ_HI_K_LO_V2V1: hi = k; lo = (v2 << 32) | (v1 & 0xFFFFFFFFull); break;
         case DBSP_PROJ_HI_K_LO_V2: hi = k; lo = v2; break;
+        case DBSP_PROJ_Q4_BID_X_AUC: {
+            // q4.rs:58-68 join_func: delta = bid (v1 = bid_dt<<20|price),
+            // trace = auction (v2 = a_dt<<28|(expires-a_dt)<<4|cat&0xF);
+            // invalid pairs keep their slot with weight 0 (dropped by the
+            // consolidate), mirroring the GPU emit contract
+            const uint64_t bid_dt = v1 >> 20, price = v1 & 0xFFFFFull;
+            const uint64_t a_dt = v2 >> 28, dur = (v2 >> 4) & 0xFFFFFFull;
+            hi = (k << 4) | (v2 & 0xFull);
+            lo = price;
+            if (!(bid_dt >= a_dt && bid_dt <= a_dt + dur)) w = 0;
+            break;
+        }
+        case DBSP_PROJ_Q4_AUC_X_BID: {
+            const uint64_t bid_dt = v2 >> 20, price = v2 & 0xFFFFFull;
+            const uint64_t a_dt = v1 >> 28, dur = (v1 >> 4) & 0xFFFFFFull;
+            hi = (k << 4) | (v1 & 0xFull);
+            lo = price;
+            if (!(bid_dt >= a_dt && bid_dt <= a_dt + dur)) w = 0;
+            break;
+        }
     }
     out.push_back({hi, lo, w});
 }
@@ -304,6 +324,10 @@ struct Oracle {
     bool q8_have_prev_p = false, q8_have_prev_a = false;
     uint64_t q8_s0p = 0, q8_e0p = 0, q8_s0a = 0, q8_e0a = 0;
     uint64_t q8_wm = 0;
+    // q4 state
+    std::vector<Row> q4_a_int, q4_b_int;   // auctions / bids by auction id
+    std::vector<Row> q4_maxin, q4_maxout;  // max in/out integrals
+    std::vector<Row> q4_avg_int, q4_avgout;  // packed avg integral + output
     // q5 state
     std::vector<Row> bt_int;               // bids by time (trace for window)
     std::vector<Row> wb_int;               // windowed bids by auction (weighed integral)
@@ -362,6 +386,61 @@ std::vector<Row> q3_step(Oracle &o, const std::vector<dbsp_event> &ev) {
     join_raw(dP, o.a_int, DBSP_PROJ_HI_V1_LO_V2, 0, out);  // A_t ⋈ dP
     o.p_int = merge(o.p_int, dP);
     consolidate(out);
+    return out;
+}
+
+// queries/q4.rs: join (validity-filtered) -> Max per (auction,category) ->
+// Average per category (packed (sum<<20)+count linear aggregate; see the
+// engine's q4_step for the operator mapping)
+std::vector<Row> q4_step(Oracle &o, const std::vector<dbsp_event> &ev) {
+    std::vector<Row> dA, dB;
+    for (auto &e : ev) {
+        if (e.kind == 1)
+            dA.push_back({e.f0, (e.f3 << 28) | (((e.f4 - e.f3) & 0xFFFFFFull) << 4) |
+                                    (e.f2 & 0xFull),
+                          e.w});
+        if (e.kind == 2)
+            dB.push_back({e.f0, (e.f3 << 20) | (e.f2 & 0xFFFFFull), e.w});
+    }
+    consolidate(dA);
+    consolidate(dB);
+    std::vector<Row> dWinIn;
+    join_raw(dB, o.q4_a_int, DBSP_PROJ_Q4_BID_X_AUC, 0, dWinIn);  // dB ⋈ A_{t-1}
+    o.q4_b_int = merge(o.q4_b_int, dB);
+    join_raw(dA, o.q4_b_int, DBSP_PROJ_Q4_AUC_X_BID, 0, dWinIn);  // dA ⋈ B_t
+    o.q4_a_int = merge(o.q4_a_int, dA);
+    consolidate(dWinIn);
+    std::vector<Row> dWin;
+    if (!dWinIn.empty()) {
+        o.q4_maxin = merge(o.q4_maxin, dWinIn);
+        std::vector<uint64_t> keys;
+        for (auto &r : dWinIn)
+            if (keys.empty() || keys.back() != r.k) keys.push_back(r.k);
+        agg_max_upsert(keys, o.q4_maxin, o.q4_maxout, dWin);
+        consolidate(dWin);
+        o.q4_maxout = merge(o.q4_maxout, dWin);
+    }
+    // average per category: weigh to the packed (sum<<20)+count weight
+    std::vector<Row> dAvgIn;
+    for (auto &r : dWin)
+        dAvgIn.push_back({r.k & 0xFull, 0, r.w * (int64_t)((r.v << 20) | 1ull)});
+    consolidate(dAvgIn);
+    std::vector<Row> out;
+    if (!dAvgIn.empty()) {
+        o.q4_avg_int = merge(o.q4_avg_int, dAvgIn);
+        std::vector<uint64_t> keys;
+        for (auto &r : dAvgIn)
+            if (keys.empty() || keys.back() != r.k) keys.push_back(r.k);
+        std::vector<Row> upd;
+        agg_linear_upsert(keys, o.q4_avg_int, o.q4_avgout, upd);
+        consolidate(upd);
+        o.q4_avgout = merge(o.q4_avgout, upd);
+        for (auto &r : upd) {
+            const uint64_t cnt = r.v & 0xFFFFFull;
+            out.push_back({r.k, cnt ? (r.v >> 20) / cnt : 0, r.w});
+        }
+        consolidate(out);
+    }
     return out;
 }
 
@@ -673,6 +752,7 @@ int64_t oracle_query_step(void *h, const dbsp_event *events, int64_t n,
     std::vector<Row> res;
     switch (o->query) {
         case 3: res = q3_step(*o, ev); break;
+        case 4: res = q4_step(*o, ev); break;
         case 5: res = q5_step(*o, ev); break;
         case 8: res = q8_step(*o, ev); break;
         default: return -2;

@@ -229,3 +229,29 @@ def test_query_parity_fuzz_seeds(ctx, query, seed):
     boundaries that the sized sort/dense/window routing keys off."""
     evs = gen.generate(60_000, seed=seed)
     _run_parity(ctx, query, evs, tick=7_777, seed_note=f"+fuzz{seed}")
+
+
+def test_query_parity_generated_q4(ctx):
+    evs = gen.generate(100_000, seed=11)
+    _run_parity(ctx, 4, evs, tick=10_000)
+
+
+def test_query_parity_small_ticks_q4(ctx):
+    evs = gen.generate(2_000, seed=13)
+    _run_parity(ctx, 4, evs, tick=137)
+
+
+def test_golden_q4_on_gpu(ctx):
+    """The reference's q4 test (queries/q4.rs:94-239) through the engine."""
+    import sys
+    from dbsp_amd.engine import Engine
+    from test_oracle import _q4_golden_events
+    g = load_golden("q4_category_avg.json")
+    eng = Engine(ctx, query=4)
+    for t, tick in enumerate(g["ticks"]):
+        evs = _q4_golden_events(tick)
+        eng.step(evs)
+        got = zset(eng.output())
+        exp = {(int(k), int(v)): int(w) for k, v, w in tick["expected"]}
+        assert got == exp, f"tick {t}: {got} != {exp}"
+    eng.close()

@@ -380,3 +380,30 @@ def test_oracle_sharded_q3_unions_to_global():
     single.close()
     for q in shards:
         q.close()
+
+
+def _q4_golden_events(tick):
+    import numpy as np
+    from dbsp_amd import EVENT_DT
+    evs = np.zeros(len(tick["events"]), dtype=EVENT_DT)
+    for i, e in enumerate(tick["events"]):
+        if e["kind"] == 1:
+            evs[i] = (1, e["id"], 1, e["category"], e["date_time"],
+                      e["expires"], e["w"])
+        else:
+            evs[i] = (2, e["auction"], 1, e["price"], e["date_time"], 0,
+                      e["w"])
+    return evs
+
+
+def test_oracle_q4_reference_vectors():
+    """q4 replayed against the reference's own in-tree test
+    (queries/q4.rs:94-239): per-tick (category, avg) output deltas."""
+    g = load_golden("q4_category_avg.json")
+    q = oracle.Query(4)
+    for t, tick in enumerate(g["ticks"]):
+        out = q.step(_q4_golden_events(tick), cap=1 << 20)
+        got = zset(out)
+        exp = {(int(k), int(v)): int(w) for k, v, w in tick["expected"]}
+        assert got == exp, f"tick {t}: {got} != {exp}"
+    q.close()
