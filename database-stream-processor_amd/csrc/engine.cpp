@@ -1010,7 +1010,8 @@ struct dbsp_engine {
     DevBatch maxin_int, maxout_int, maxz_int;
     // q4 state (queries/q4.rs: join + per-auction Max + per-category Average)
     Spine q4_a_int, q4_b_int;      // auction / bid traces (keyed by auction)
-    DevBatch q4_maxin, q4_maxout;  // consolidated max in/out integrals
+    Spine q4_maxin_sp;             // max input integral (affected-key gather)
+    DevBatch q4_maxout;            // consolidated max output integral
     Spine q4_avg_int, q4_avgout;   // packed (sum<<20|count) integral + output
     // C5 state (query 100; BASELINE configs[4]: 1B-row indexed trace x
     // 10M-row delta incremental join + f64 sum aggregate)
@@ -1091,9 +1092,8 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
                      &e->wa_int, &e->bt_int, &e->wb_int, &e->counts_int,
                      &e->bc_int, &e->c5_trace, &e->c5_wint, &e->c5_out,
                      &e->q4_a_int, &e->q4_b_int, &e->q4_avg_int,
-                     &e->q4_avgout})
+                     &e->q4_avgout, &e->q4_maxin_sp})
         s->clear(c);
-    free_batch(c, e->q4_maxin);
     free_batch(c, e->q4_maxout);
     free_batch(c, e->maxin_int);
     free_batch(c, e->maxout_int);
@@ -3106,28 +3106,42 @@ static dbsp_status q4_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     TRY(finalize_raw(c, outs, dWinIn));
     DevBatch dWin{};
     if (dWinIn.n > 0) {
-        // max input integral is kept consolidated: the Max aggregator reads
-        // the key's whole value run (max.rs:36-55)
+        // max input lives in a SPINE (amortized log maintenance); the Max
+        // aggregator re-reads only the AFFECTED keys' value runs each tick —
+        // the reference's eval_key discipline (aggregate/mod.rs:479-547) —
+        // by gathering them with a unit-weight key join over the spine
+        // (weights multiply by 1, so the gather carries the trace weights)
+        // and consolidating just that slice before the max scan
+        // (max.rs:36-55 needs per-val TOTAL weights)
         {
-            DevBatch cpy, m;
+            DevBatch cpy;
             TRY(copy_batch(c, dWinIn, cpy));
-            TRY(merge_batches(c, e->q4_maxin, cpy, m));
-            free_batch(c, e->q4_maxin);
-            free_batch(c, cpy);
-            e->q4_maxin = m;
+            TRY(e->q4_maxin_sp.insert(c, cpy));
         }
         uint64_t *keys = nullptr;
         int64_t nk = 0;
         TRY(dbspk::unique_keys(c->stream, dWinIn.k, dWinIn.n, &keys, &nk));
+        DevBatch kb;
+        TRY(alloc_batch(c, nk, kb, true));
+        HIP_CHECK_ST(hipMemcpyAsync(kb.k, keys, nk * 8,
+                                    hipMemcpyDeviceToDevice, c->stream));
+        HIP_CHECK_ST(hipMemsetAsync(kb.v, 0, nk * 8, c->stream));
+        TRY(dbspk::fill_u64(c->stream, (uint64_t *)kb.w, 1, nk));
+        kb.n = nk;
+        std::vector<DevBatch> gouts;
+        TRY(join_vs_spine(c, kb, e->q4_maxin_sp, DBSP_PROJ_HI_K_LO_V2, 0,
+                          gouts));
+        DevBatch gathered;
+        TRY(finalize_raw(c, gouts, gathered));
         DevBatch raw;
         {
             ScopedTimer timer(c, 3, 0.0);
             TRY(dbspk::agg_max_upsert_rows(
-                c->stream, keys, nk, e->q4_maxin.k, e->q4_maxin.v,
-                e->q4_maxin.w, e->q4_maxin.n, e->q4_maxout.k, e->q4_maxout.v,
-                e->q4_maxout.w, e->q4_maxout.n, &raw.k, &raw.v, &raw.w,
-                &raw.n));
+                c->stream, keys, nk, gathered.k, gathered.v, gathered.w,
+                gathered.n, e->q4_maxout.k, e->q4_maxout.v, e->q4_maxout.w,
+                e->q4_maxout.n, &raw.k, &raw.v, &raw.w, &raw.n));
         }
+        free_batch(c, gathered);
         HIP_CHECK_ST(dbspk::cache_free(keys, c->stream));
         TRY(sort_consolidate_batch(c, raw, dWin));
         if (dWin.n > 0) {
