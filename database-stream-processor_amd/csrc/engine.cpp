@@ -3126,7 +3126,7 @@ static dbsp_status q4_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         HIP_CHECK_ST(hipMemcpyAsync(kb.k, keys, nk * 8,
                                     hipMemcpyDeviceToDevice, c->stream));
         HIP_CHECK_ST(hipMemsetAsync(kb.v, 0, nk * 8, c->stream));
-        TRY(dbspk::fill_u64(c->stream, (uint64_t *)kb.w, 1, nk));
+        dbspk::fill_u64(c->stream, (uint64_t *)kb.w, 1, nk);
         kb.n = nk;
         std::vector<DevBatch> gouts;
         TRY(join_vs_spine(c, kb, e->q4_maxin_sp, DBSP_PROJ_HI_K_LO_V2, 0,
