@@ -32,7 +32,7 @@ sys.path.insert(0, str(ROOT / "database-stream-processor_amd" / "python"))
 HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
 # Reference's published q3 number (9,936,407 events/s on an
 # unnamed 16-core CPU — the only published figure for this metric).
-PUBLISHED = {3: 9_936_407.0, 4: 9_768_487.0, 5: 9_906_875.0,
+PUBLISHED = {3: 9_936_407.0, 4: 9_768_487.0, 5: 9_906_875.0, 6: 9_829_942.0,
              8: 9_380_863.0, 0: 9_926_544.0}
 
 
@@ -113,8 +113,9 @@ def _shard_filter(evs, query, world, rank):
         # q3's flat_map anyway (queries/q3.rs:39-52)
         keep = ((kind == 0) & (evs["f0"] % world == rank)) | \
                ((kind == 1) & (evs["f1"] % world == rank))
-    elif query == 4:
-        # auctions and bids both by auction id (queries/q4.rs:45-56)
+    elif query in (4, 6):
+        # auctions and bids both by auction id (queries/q4.rs:45-56,
+        # q6.rs:46-57)
         keep = ((kind == 1) | (kind == 2)) & (evs["f0"] % world == rank)
     elif query == 5:
         # bids by auction id (queries/q5.rs:67-88)
@@ -373,7 +374,7 @@ def main():
     ap.add_argument("--steps", type=int, default=250)
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--query", default="3",
-                    choices=["0", "3", "4", "5", "8", "c5"])
+                    choices=["0", "3", "4", "5", "6", "8", "c5"])
     ap.add_argument("--tick", type=int, default=40_000)
     ap.add_argument("--seed", type=int, default=1)
     ap.add_argument("--c5-trace", type=int, default=1_000_000_000,

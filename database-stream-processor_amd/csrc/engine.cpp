@@ -1013,6 +1013,10 @@ struct dbsp_engine {
     Spine q4_maxin_sp;             // max input integral (affected-key gather)
     DevBatch q4_maxout;            // consolidated max output integral
     Spine q4_avg_int, q4_avgout;   // packed (sum<<20|count) integral + output
+    // q6 state (queries/q6.rs: join + per-(auction,seller) Max + per-seller
+    // last-10 average fold)
+    Spine q6_a_int, q6_b_int, q6_maxin_sp, q6_fold_sp;
+    DevBatch q6_maxout, q6_foldout;
     // C5 state (query 100; BASELINE configs[4]: 1B-row indexed trace x
     // 10M-row delta incremental join + f64 sum aggregate)
     Spine c5_trace;   // (k, f64-bits val, +1 i64) join-side trace
@@ -1062,7 +1066,7 @@ struct dbsp_engine {
 extern "C" dbsp_status dbsp_engine_create(dbsp_engine **out, dbsp_ctx *ctx,
                                           int query, int rank, int world) {
     if (query != 0 && query != 3 && query != 4 && query != 5 &&
-        query != 8 && query != 100)
+        query != 6 && query != 8 && query != 100)
         return DBSP_ERR_INVALID;
 
     dbsp_engine *e = new dbsp_engine();
@@ -1092,9 +1096,12 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
                      &e->wa_int, &e->bt_int, &e->wb_int, &e->counts_int,
                      &e->bc_int, &e->c5_trace, &e->c5_wint, &e->c5_out,
                      &e->q4_a_int, &e->q4_b_int, &e->q4_avg_int,
-                     &e->q4_avgout, &e->q4_maxin_sp})
+                     &e->q4_avgout, &e->q4_maxin_sp, &e->q6_a_int,
+                     &e->q6_b_int, &e->q6_maxin_sp, &e->q6_fold_sp})
         s->clear(c);
     free_batch(c, e->q4_maxout);
+    free_batch(c, e->q6_maxout);
+    free_batch(c, e->q6_foldout);
     free_batch(c, e->maxin_int);
     free_batch(c, e->maxout_int);
     free_batch(c, e->maxz_int);
@@ -1420,10 +1427,9 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
         // global row counts are known from the tick's event count alone
         if (want_two) {
             DevBatch s0, s1;
-            const int64_t h0 =
-                e->query == 4 ? 3 * n / 50 + 64 : n / 50 + 64;
-            const int64_t h1 =
-                e->query == 4 ? 46 * n / 50 + 64 : 3 * n / 50 + 64;
+            const bool ab = e->query == 4 || e->query == 6;
+            const int64_t h0 = ab ? 3 * n / 50 + 64 : n / 50 + 64;
+            const int64_t h1 = ab ? 46 * n / 50 + 64 : 3 * n / 50 + 64;
             TRY(shard_exchange_pair(c, d0, d1, s0, s1, h0, h1));
             d0 = s0;
             d1 = s1;
@@ -3192,6 +3198,124 @@ static dbsp_status q4_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
     return DBSP_OK;
 }
 
+// affected-key gather + keyed aggregate + upsert over a spine: the
+// reference's eval_key discipline (aggregate/mod.rs:479-547) — gather the
+// affected keys' value runs with a unit-weight key join (weights multiply by
+// one, so the gather carries the trace weights), consolidate the slice (the
+// aggregate needs per-val TOTAL weights), then run the MODE aggregate with
+// upsert retractions against the consolidated output integral.
+static dbsp_status agg_keyed_spine(dbsp_engine *e, const DevBatch &delta,
+                                   Spine &in_sp, DevBatch &out_int, int mode,
+                                   DevBatch &dOut) {
+    dbsp_ctx *c = e->ctx;
+    dOut = DevBatch{};
+    if (delta.n == 0) return DBSP_OK;
+    uint64_t *keys = nullptr;
+    int64_t nk = 0;
+    TRY(dbspk::unique_keys(c->stream, delta.k, delta.n, &keys, &nk));
+    DevBatch kb;
+    TRY(alloc_batch(c, nk, kb, true));
+    HIP_CHECK_ST(hipMemcpyAsync(kb.k, keys, nk * 8, hipMemcpyDeviceToDevice,
+                                c->stream));
+    HIP_CHECK_ST(hipMemsetAsync(kb.v, 0, nk * 8, c->stream));
+    dbspk::fill_u64(c->stream, (uint64_t *)kb.w, 1, nk);
+    kb.n = nk;
+    std::vector<DevBatch> gouts;
+    TRY(join_vs_spine(c, kb, in_sp, DBSP_PROJ_HI_K_LO_V2, 0, gouts));
+    DevBatch gathered;
+    TRY(finalize_raw(c, gouts, gathered));
+    DevBatch raw;
+    {
+        ScopedTimer timer(c, 3, 0.0);
+        if (mode == 1)
+            TRY(dbspk::agg_max_upsert_rows(
+                c->stream, keys, nk, gathered.k, gathered.v, gathered.w,
+                gathered.n, out_int.k, out_int.v, out_int.w, out_int.n,
+                &raw.k, &raw.v, &raw.w, &raw.n));
+        else
+            TRY(dbspk::agg_last10_upsert_rows(
+                c->stream, keys, nk, gathered.k, gathered.v, gathered.w,
+                gathered.n, out_int.k, out_int.v, out_int.w, out_int.n,
+                &raw.k, &raw.v, &raw.w, &raw.n));
+    }
+    free_batch(c, gathered);
+    HIP_CHECK_ST(dbspk::cache_free(keys, c->stream));
+    TRY(sort_consolidate_batch(c, raw, dOut));
+    if (dOut.n > 0) {
+        DevBatch cpy, m;
+        TRY(copy_batch(c, dOut, cpy));
+        TRY(merge_batches(c, out_int, cpy, m));
+        free_batch(c, out_int);
+        free_batch(c, cpy);
+        out_int = m;
+    }
+    return DBSP_OK;
+}
+
+// ---------------------------------------------------------------------------
+// q6 (queries/q6.rs): average winning-bid price of the last 10 closed
+// auctions per seller.  Same join shape as q4 on (auction, seller), Max per
+// (auction<<20|seller), then the VecDeque fold (q6.rs:96-110) as the MODE-2
+// last-10 average over the seller-keyed integral whose vals
+// (auction<<20)|price keep the reference's cursor order by auction id.
+// ---------------------------------------------------------------------------
+
+static dbsp_status q6_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
+    dbsp_ctx *c = e->ctx;
+    engine_free_output(e);
+    DevBatch dA, dB;
+    TRY(build_deltas(e, d_ev, n, dA, dB, true));
+    std::vector<DevBatch> outs;
+    TRY(join_vs_spine(c, dB, e->q6_a_int, DBSP_PROJ_Q6_BID_X_AUC, 0, outs));
+    TRY(join_vs_spine(c, dA, e->q6_b_int, DBSP_PROJ_Q6_AUC_X_BID, 0, outs));
+    if (dA.n > 0 && dB.n > 0) {
+        TraceArgs t{};
+        t.nb = 1;
+        t.k[0] = dB.k; t.v[0] = dB.v; t.w[0] = dB.w; t.n[0] = dB.n;
+        DevBatch o;
+        ScopedTimer timer(c, 2, (double)dA.n * 24.0);
+        TRY(dbspk::join_spine_rows(c->stream, dA.k, dA.v, dA.w, dA.n, t,
+                                   DBSP_PROJ_Q6_AUC_X_BID, 0, &o.k, &o.v,
+                                   &o.w, &o.n));
+        if (o.n > 0) outs.push_back(o);
+        else free_batch(c, o);
+    }
+    DevBatch dWinIn;
+    TRY(finalize_raw(c, outs, dWinIn));
+    DevBatch dWin{};
+    if (dWinIn.n > 0) {
+        DevBatch cpy;
+        TRY(copy_batch(c, dWinIn, cpy));
+        TRY(e->q6_maxin_sp.insert(c, cpy));
+        TRY(agg_keyed_spine(e, dWinIn, e->q6_maxin_sp, e->q6_maxout, 1, dWin));
+    }
+    free_batch(c, dWinIn);
+    if (dWin.n > 0) {
+        // map_index (q6.rs:92-94): key by seller, val (auction<<20)|price
+        DevBatch fraw, dFoldIn;
+        TRY(alloc_batch(c, dWin.n, fraw, true));
+        TRY(dbspk::map_rows(c->stream, dWin.k, dWin.v, dWin.w, dWin.n, 7,
+                            fraw.k, fraw.v, fraw.w));
+        fraw.n = dWin.n;
+        TRY(sort_consolidate_batch(c, fraw, dFoldIn));
+        free_batch(c, dWin);
+        if (dFoldIn.n > 0) {
+            DevBatch cpy;
+            TRY(copy_batch(c, dFoldIn, cpy));
+            TRY(e->q6_fold_sp.insert(c, cpy));
+            TRY(agg_keyed_spine(e, dFoldIn, e->q6_fold_sp, e->q6_foldout, 2,
+                                e->output));
+        }
+        free_batch(c, dFoldIn);
+    } else {
+        free_batch(c, dWin);
+    }
+    TRY(e->q6_a_int.insert(c, dA));
+    TRY(e->q6_b_int.insert(c, dB));
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    return DBSP_OK;
+}
+
 // ---------------------------------------------------------------------------
 // C5 (query 100): synthetic 1B-row OrdIndexedZSet x 10M-row delta incremental
 // join with f64 sum aggregate (BASELINE configs[4]).  Per
@@ -3323,6 +3447,7 @@ extern "C" dbsp_status dbsp_engine_step_staged(dbsp_engine *e, int64_t lo,
     switch (e->query) {
         case 3: return q3_step(e, d_ev, hi - lo);
         case 4: return q4_step(e, d_ev, hi - lo);
+        case 6: return q6_step(e, d_ev, hi - lo);
         case 5: return q5_step(e, d_ev, hi - lo);
         case 8: return q8_step(e, d_ev, hi - lo);
     }
@@ -3368,6 +3493,7 @@ extern "C" dbsp_status dbsp_engine_step(dbsp_engine *e, const dbsp_event *events
     switch (e->query) {
         case 3: st = q3_step(e, d_ev, n); break;
         case 4: st = q4_step(e, d_ev, n); break;
+        case 6: st = q6_step(e, d_ev, n); break;
         case 5: st = q5_step(e, d_ev, n); break;
         case 8: st = q8_step(e, d_ev, n); break;
         default: st = DBSP_ERR_INVALID;

@@ -1107,6 +1107,23 @@ __device__ inline bool proj_out(int proj, uint64_t param, uint64_t k,
             lo = price;
             return bid_dt >= a_dt && bid_dt <= a_dt + dur;
         }
+        case DBSP_PROJ_Q6_BID_X_AUC: {
+            // q6.rs:60-80: delta = bid (v1 = bid_dt<<20|price), trace =
+            // auction (v2 = a_dt<<34 | (expires-a_dt)<<20 | seller); emit
+            // ((auction<<20)|seller, price) in the validity window
+            const uint64_t bid_dt = v1 >> 20, price = v1 & 0xFFFFFull;
+            const uint64_t a_dt = v2 >> 36, dur = (v2 >> 20) & 0xFFFFull;
+            hi = (k << 20) | (v2 & 0xFFFFFull);
+            lo = price;
+            return bid_dt >= a_dt && bid_dt <= a_dt + dur;
+        }
+        case DBSP_PROJ_Q6_AUC_X_BID: {
+            const uint64_t bid_dt = v2 >> 20, price = v2 & 0xFFFFFull;
+            const uint64_t a_dt = v1 >> 36, dur = (v1 >> 20) & 0xFFFFull;
+            hi = (k << 20) | (v1 & 0xFFFFFull);
+            lo = price;
+            return bid_dt >= a_dt && bid_dt <= a_dt + dur;
+        }
         case DBSP_PROJ_Q4_AUC_X_BID: {  // sides swapped
             const uint64_t bid_dt = v2 >> 20, price = v2 & 0xFFFFFull;
             const uint64_t a_dt = v1 >> 28, dur = (v1 >> 4) & 0xFFFFFFull;
@@ -1840,7 +1857,11 @@ __global__ void k_distinct_count(const uint64_t *dk, const uint64_t *dv,
 // aggregate (linear / max) + upsert  (count/emit over delta keys)
 // ---------------------------------------------------------------------------
 
-template <bool MAX>
+// MODE: 0 = linear weight-sum (WeightedCount, aggregate/mod.rs:129-156),
+// 1 = Max (max.rs:36-55), 2 = q6's fold: average of the last <= 10 vals in
+// the key's run (queries/q6.rs:96-110 VecDeque fold in cursor order; the
+// price lives in the low 20 bits of the val)
+template <int MODE>
 __global__ void k_agg_count(const uint64_t *keys, int64_t nd,
                             const uint64_t *ik, const int64_t *iw, int64_t ni,
                             const uint64_t *ok_, int64_t no,
@@ -1853,7 +1874,7 @@ __global__ void k_agg_count(const uint64_t *keys, int64_t nd,
         int64_t ihi = upper_bound_k(ik, ni, key);
         uint64_t nv = 0;
         uint64_t hn = 0;
-        if (MAX) {
+        if (MODE != 0) {
             // consolidated trace: any present val has w != 0; max = last val
             if (ihi > ilo) { nv = 1; hn = 1; }
             if (hn) nv = 0;  // placeholder; real val read in emit via istart/iend
@@ -1867,12 +1888,12 @@ __global__ void k_agg_count(const uint64_t *keys, int64_t nd,
         counts[i] = (hn ? 1 : 0) + (ohi - olo);
         istart[i] = (uint64_t)ilo;
         ostart[i] = (uint64_t)olo;
-        newval[i] = MAX ? (ihi > ilo ? /* iend */ (uint64_t)ihi : 0) : nv;
+        newval[i] = MODE != 0 ? (ihi > ilo ? /* iend */ (uint64_t)ihi : 0) : nv;
         has_new[i] = hn;
     }
 }
 
-template <bool MAX>
+template <int MODE>
 __global__ void k_agg_emit(const uint64_t *keys, int64_t nd,
                            const uint64_t *iv, const uint64_t *ov_,
                            const int64_t *ow_, const uint64_t *offsets,
@@ -1888,8 +1909,19 @@ __global__ void k_agg_emit(const uint64_t *keys, int64_t nd,
         uint64_t emitted = 0;
         if (has_new[i]) {
             uint64_t nv;
-            if (MAX) {
+            if (MODE == 1) {
                 nv = iv[newval[i] - 1];  // last val of key's run (max.rs:40-52)
+            } else if (MODE == 2) {
+                // q6.rs:96-110: the VecDeque fold keeps the last <= 10 vals
+                // of the seller's cursor-ordered run; avg of their prices
+                // (val low 20 bits), integer division
+                const uint64_t ihi = newval[i], ilo = istart[i];
+                const uint64_t n10 =
+                    ihi - ilo < 10 ? ihi - ilo : (uint64_t)10;
+                uint64_t sum = 0;
+                for (uint64_t t = ihi - n10; t < ihi; t++)
+                    sum += iv[t] & 0xFFFFFull;
+                nv = sum / n10;
             } else {
                 nv = newval[i];
             }
@@ -2259,6 +2291,25 @@ __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
                 v1[pos1] = (e.f3 << 20) | (e.f2 & 0xFFFFFull);
                 w1[pos1] = e.w;
             }
+        } else if (query == 6) {
+            // q6.rs:46-57: auctions by id (seller + validity window packed:
+            // dt 28 bits from bit 36, duration 16 bits, seller 20 bits),
+            // bids by auction as q4
+            const bool p0 = act && e.kind == 1;
+            const uint64_t pos0 = wave_append((unsigned long long *)c0, p0);
+            if (p0) {
+                k0[pos0] = e.f0;
+                v0[pos0] = (e.f3 << 36) | (((e.f4 - e.f3) & 0xFFFFull) << 20) |
+                           (e.f1 & 0xFFFFFull);
+                w0[pos0] = e.w;
+            }
+            const bool p1 = act && e.kind == 2;
+            const uint64_t pos1 = wave_append((unsigned long long *)c1, p1);
+            if (p1) {
+                k1[pos1] = e.f0;
+                v1[pos1] = (e.f3 << 20) | (e.f2 & 0xFFFFFull);
+                w1[pos1] = e.w;
+            }
         } else if (query == 5) {
             // q5.rs:79-83: bids by time
             const bool p0 = act && e.kind == 2;
@@ -2328,6 +2379,12 @@ __global__ void k_map(const uint64_t *k, const uint64_t *v, const int64_t *w,
                 rv = cnt > 0 ? (vv >> 20) / (uint64_t)cnt : 0;
                 break;
             }
+            case 7:  // q6 map_index (q6.rs:92-94): winning bids keyed by
+                     // seller, val (auction<<20)|price keeps cursor order
+                     // by auction id for the last-10 fold
+                rk = kk & 0xFFFFFull;
+                rv = ((kk >> 20) << 20) | (vv & 0xFFFFFull);
+                break;
             default: rk = vv; rv = kk; break;
         }
         ok[i] = rk;
@@ -2847,7 +2904,7 @@ dbsp_status join_rows(hipStream_t s, const uint64_t *dk, const uint64_t *dv,
     return DBSP_OK;
 }
 
-template <bool MAX>
+template <int MODE>
 static dbsp_status agg_upsert_impl(hipStream_t s, const uint64_t *keys,
                                    int64_t nd, const uint64_t *ik,
                                    const uint64_t *iv, const int64_t *iw,
@@ -2866,7 +2923,7 @@ static dbsp_status agg_upsert_impl(hipStream_t s, const uint64_t *keys,
     HIP_CHECK(dbspk::cache_malloc((void **)&newval, nd * sizeof(uint64_t), s));
     HIP_CHECK(dbspk::cache_malloc((void **)&hasnew, nd * sizeof(uint64_t), s));
     HIP_CHECK(dbspk::cache_malloc((void **)&offsets, nd * sizeof(uint64_t), s));
-    k_agg_count<MAX><<<grid_for(nd), BLK, 0, s>>>(keys, nd, ik, iw, ni, tok, no,
+    k_agg_count<MODE><<<grid_for(nd), BLK, 0, s>>>(keys, nd, ik, iw, ni, tok, no,
                                                   counts, istart, ostart, newval,
                                                   hasnew);
     uint64_t nout = 0;
@@ -2877,7 +2934,7 @@ static dbsp_status agg_upsert_impl(hipStream_t s, const uint64_t *keys,
     HIP_CHECK(dbspk::cache_malloc((void **)&rv, nout * sizeof(uint64_t) + 8, s));
     HIP_CHECK(dbspk::cache_malloc((void **)&rw, nout * sizeof(int64_t) + 8, s));
     if (nout > 0)
-        k_agg_emit<MAX><<<grid_for(nd), BLK, 0, s>>>(keys, nd, iv, tov, tow,
+        k_agg_emit<MODE><<<grid_for(nd), BLK, 0, s>>>(keys, nd, iv, tov, tow,
                                                      offsets, counts, istart,
                                                      ostart, newval, hasnew, rk,
                                                      rv, rw);
@@ -2929,8 +2986,8 @@ dbsp_status agg_linear_upsert_rows(hipStream_t s, const uint64_t *keys,
                                    const uint64_t *tov, const int64_t *tow,
                                    int64_t no, uint64_t **ok, uint64_t **ov,
                                    int64_t **ow, int64_t *out_n) {
-    return agg_upsert_impl<false>(s, keys, nd, ik, iv, iw, ni, tok, tov, tow, no,
-                                  ok, ov, ow, out_n);
+    return agg_upsert_impl<0>(s, keys, nd, ik, iv, iw, ni, tok, tov, tow, no,
+                              ok, ov, ow, out_n);
 }
 
 dbsp_status agg_max_upsert_rows(hipStream_t s, const uint64_t *keys, int64_t nd,
@@ -2939,8 +2996,19 @@ dbsp_status agg_max_upsert_rows(hipStream_t s, const uint64_t *keys, int64_t nd,
                                 const uint64_t *tok, const uint64_t *tov,
                                 const int64_t *tow, int64_t no, uint64_t **ok,
                                 uint64_t **ov, int64_t **ow, int64_t *out_n) {
-    return agg_upsert_impl<true>(s, keys, nd, ik, iv, iw, ni, tok, tov, tow, no,
-                                 ok, ov, ow, out_n);
+    return agg_upsert_impl<1>(s, keys, nd, ik, iv, iw, ni, tok, tov, tow, no,
+                              ok, ov, ow, out_n);
+}
+
+dbsp_status agg_last10_upsert_rows(hipStream_t s, const uint64_t *keys,
+                                   int64_t nd, const uint64_t *ik,
+                                   const uint64_t *iv, const int64_t *iw,
+                                   int64_t ni, const uint64_t *tok,
+                                   const uint64_t *tov, const int64_t *tow,
+                                   int64_t no, uint64_t **ok, uint64_t **ov,
+                                   int64_t **ow, int64_t *out_n) {
+    return agg_upsert_impl<2>(s, keys, nd, ik, iv, iw, ni, tok, tov, tow, no,
+                              ok, ov, ow, out_n);
 }
 
 dbsp_status window_rows(hipStream_t s, const uint64_t *tk, const uint64_t *tv,

@@ -241,6 +241,14 @@ dbsp_status agg_linear_upsert_rows(hipStream_t s, const uint64_t *keys,
                                    int64_t no, uint64_t **ok, uint64_t **ov,
                                    int64_t **ow, int64_t *out_n);
 
+// q6's fold: avg of the last <= 10 vals in the key's run (cursor order)
+dbsp_status agg_last10_upsert_rows(hipStream_t s, const uint64_t *keys,
+                                   int64_t nd, const uint64_t *ik,
+                                   const uint64_t *iv, const int64_t *iw,
+                                   int64_t ni, const uint64_t *tok,
+                                   const uint64_t *tov, const int64_t *tow,
+                                   int64_t no, uint64_t **ok, uint64_t **ov,
+                                   int64_t **ow, int64_t *out_n);
 dbsp_status agg_max_upsert_rows(hipStream_t s, const uint64_t *keys, int64_t nd,
                                 const uint64_t *ik, const uint64_t *iv,
                                 const int64_t *iw, int64_t ni,

@@ -98,6 +98,11 @@ typedef enum {
                                      a_dt <= bid_dt <= expires, else weight 0
                                      (dropped by the consolidate) */
     DBSP_PROJ_Q4_AUC_X_BID   = 10, /* q4 swapped side */
+    DBSP_PROJ_Q6_BID_X_AUC   = 11, /* q6.rs:60-80: delta=bid, trace=auction
+                                      (v=a_dt<<36|(expires-a_dt)<<20|seller):
+                                      emit ((auction<<20)|seller, price) in
+                                      the validity window, else weight 0 */
+    DBSP_PROJ_Q6_AUC_X_BID   = 12, /* q6 swapped side */
 } dbsp_proj;
 
 /* ======================================================================

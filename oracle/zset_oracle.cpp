@@ -124,6 +124,24 @@ inline void proj_emit(std::vector<Row> &out, dbsp_proj proj, uint64_t param,
             if (!(bid_dt >= a_dt && bid_dt <= a_dt + dur)) w = 0;
             break;
         }
+        case DBSP_PROJ_Q6_BID_X_AUC: {
+            // q6.rs:60-80: delta = bid (v1 = bid_dt<<20|price), trace =
+            // auction (v2 = a_dt<<34 | (expires-a_dt)<<20 | seller)
+            const uint64_t bid_dt = v1 >> 20, price = v1 & 0xFFFFFull;
+            const uint64_t a_dt = v2 >> 36, dur = (v2 >> 20) & 0xFFFFull;
+            hi = (k << 20) | (v2 & 0xFFFFFull);
+            lo = price;
+            if (!(bid_dt >= a_dt && bid_dt <= a_dt + dur)) w = 0;
+            break;
+        }
+        case DBSP_PROJ_Q6_AUC_X_BID: {
+            const uint64_t bid_dt = v2 >> 20, price = v2 & 0xFFFFFull;
+            const uint64_t a_dt = v1 >> 36, dur = (v1 >> 20) & 0xFFFFull;
+            hi = (k << 20) | (v1 & 0xFFFFFull);
+            lo = price;
+            if (!(bid_dt >= a_dt && bid_dt <= a_dt + dur)) w = 0;
+            break;
+        }
         case DBSP_PROJ_Q4_AUC_X_BID: {
             const uint64_t bid_dt = v2 >> 20, price = v2 & 0xFFFFFull;
             const uint64_t a_dt = v1 >> 28, dur = (v1 >> 4) & 0xFFFFFFull;
@@ -328,6 +346,10 @@ struct Oracle {
     std::vector<Row> q4_a_int, q4_b_int;   // auctions / bids by auction id
     std::vector<Row> q4_maxin, q4_maxout;  // max in/out integrals
     std::vector<Row> q4_avg_int, q4_avgout;  // packed avg integral + output
+    // q6 state
+    std::vector<Row> q6_a_int, q6_b_int;
+    std::vector<Row> q6_maxin, q6_maxout;    // per-(auction,seller) max
+    std::vector<Row> q6_fold_in, q6_foldout; // seller-keyed last-10 fold
     // q5 state
     std::vector<Row> bt_int;               // bids by time (trace for window)
     std::vector<Row> wb_int;               // windowed bids by auction (weighed integral)
@@ -440,6 +462,70 @@ std::vector<Row> q4_step(Oracle &o, const std::vector<dbsp_event> &ev) {
             out.push_back({r.k, cnt ? (r.v >> 20) / cnt : 0, r.w});
         }
         consolidate(out);
+    }
+    return out;
+}
+
+// queries/q6.rs: join -> Max per (auction<<20|seller) -> per-seller average
+// of the last <= 10 winning bids (the VecDeque fold in cursor order,
+// q6.rs:96-110; vals (auction<<20)|price keep that order)
+std::vector<Row> q6_step(Oracle &o, const std::vector<dbsp_event> &ev) {
+    std::vector<Row> dA, dB;
+    for (auto &e : ev) {
+        if (e.kind == 1)
+            dA.push_back({e.f0, (e.f3 << 36) | (((e.f4 - e.f3) & 0xFFFFull) << 20) |
+                                    (e.f1 & 0xFFFFFull),
+                          e.w});
+        if (e.kind == 2)
+            dB.push_back({e.f0, (e.f3 << 20) | (e.f2 & 0xFFFFFull), e.w});
+    }
+    consolidate(dA);
+    consolidate(dB);
+    std::vector<Row> dWinIn;
+    join_raw(dB, o.q6_a_int, DBSP_PROJ_Q6_BID_X_AUC, 0, dWinIn);
+    o.q6_b_int = merge(o.q6_b_int, dB);
+    join_raw(dA, o.q6_b_int, DBSP_PROJ_Q6_AUC_X_BID, 0, dWinIn);
+    o.q6_a_int = merge(o.q6_a_int, dA);
+    consolidate(dWinIn);
+    std::vector<Row> dWin;
+    if (!dWinIn.empty()) {
+        o.q6_maxin = merge(o.q6_maxin, dWinIn);
+        std::vector<uint64_t> keys;
+        for (auto &r : dWinIn)
+            if (keys.empty() || keys.back() != r.k) keys.push_back(r.k);
+        agg_max_upsert(keys, o.q6_maxin, o.q6_maxout, dWin);
+        consolidate(dWin);
+        o.q6_maxout = merge(o.q6_maxout, dWin);
+    }
+    std::vector<Row> dFoldIn;
+    for (auto &r : dWin)
+        dFoldIn.push_back({r.k & 0xFFFFFull, ((r.k >> 20) << 20) | (r.v & 0xFFFFFull), r.w});
+    consolidate(dFoldIn);
+    std::vector<Row> out;
+    if (!dFoldIn.empty()) {
+        o.q6_fold_in = merge(o.q6_fold_in, dFoldIn);
+        std::vector<uint64_t> keys;
+        for (auto &r : dFoldIn)
+            if (keys.empty() || keys.back() != r.k) keys.push_back(r.k);
+        // per affected seller: avg of the last <= 10 vals' prices; then the
+        // upsert against the output integral
+        for (uint64_t key : keys) {
+            // gather the seller's current run
+            std::vector<uint64_t> vals;
+            for (auto &r : o.q6_fold_in)
+                if (r.k == key && r.w > 0) vals.push_back(r.v);
+            if (!vals.empty()) {
+                size_t n10 = vals.size() < 10 ? vals.size() : 10;
+                uint64_t sum = 0;
+                for (size_t t = vals.size() - n10; t < vals.size(); t++)
+                    sum += vals[t] & 0xFFFFFull;
+                out.push_back({key, sum / n10, 1});
+            }
+            for (auto &r : o.q6_foldout)
+                if (r.k == key) out.push_back({key, r.v, -r.w});
+        }
+        consolidate(out);
+        o.q6_foldout = merge(o.q6_foldout, out);
     }
     return out;
 }
@@ -753,6 +839,7 @@ int64_t oracle_query_step(void *h, const dbsp_event *events, int64_t n,
     switch (o->query) {
         case 3: res = q3_step(*o, ev); break;
         case 4: res = q4_step(*o, ev); break;
+        case 6: res = q6_step(*o, ev); break;
         case 5: res = q5_step(*o, ev); break;
         case 8: res = q8_step(*o, ev); break;
         default: return -2;

@@ -255,3 +255,23 @@ def test_golden_q4_on_gpu(ctx):
         exp = {(int(k), int(v)): int(w) for k, v, w in tick["expected"]}
         assert got == exp, f"tick {t}: {got} != {exp}"
     eng.close()
+
+
+def test_query_parity_generated_q6(ctx):
+    evs = gen.generate(100_000, seed=11)
+    _run_parity(ctx, 6, evs, tick=10_000)
+
+
+def test_golden_q6_on_gpu(ctx):
+    """The reference's q6 tests (queries/q6.rs) through the engine."""
+    from dbsp_amd.engine import Engine
+    from test_oracle import _q6_golden_events
+    g = load_golden("q6_seller_avg.json")
+    for case in g["cases"]:
+        eng = Engine(ctx, query=6)
+        for t, tick in enumerate(case["ticks"]):
+            eng.step(_q6_golden_events(tick))
+            got = zset(eng.output())
+            exp = {(int(k), int(v)): int(w) for k, v, w in tick["expected"]}
+            assert got == exp, f'{case["name"]} tick {t}: {got} != {exp}'
+        eng.close()

@@ -407,3 +407,31 @@ def test_oracle_q4_reference_vectors():
         exp = {(int(k), int(v)): int(w) for k, v, w in tick["expected"]}
         assert got == exp, f"tick {t}: {got} != {exp}"
     q.close()
+
+
+def _q6_golden_events(tick):
+    import numpy as np
+    from dbsp_amd import EVENT_DT
+    evs = np.zeros(len(tick["events"]), dtype=EVENT_DT)
+    for i, e in enumerate(tick["events"]):
+        if e["kind"] == 1:
+            evs[i] = (1, e["id"], e["seller"], 1, e["date_time"],
+                      e["expires"], e["w"])
+        else:
+            evs[i] = (2, e["auction"], 1, e["price"], e["date_time"], 0,
+                      e["w"])
+    return evs
+
+
+def test_oracle_q6_reference_vectors():
+    """q6 replayed against the reference's own in-tree tests (queries/q6.rs:
+    single auction, multiple auctions, >10-auction last-10 eviction)."""
+    g = load_golden("q6_seller_avg.json")
+    for case in g["cases"]:
+        q = oracle.Query(6)
+        for t, tick in enumerate(case["ticks"]):
+            out = q.step(_q6_golden_events(tick), cap=1 << 20)
+            got = zset(out)
+            exp = {(int(k), int(v)): int(w) for k, v, w in tick["expected"]}
+            assert got == exp, f'{case["name"]} tick {t}: {got} != {exp}'
+        q.close()
