@@ -1101,31 +1101,31 @@ __device__ inline bool proj_out(int proj, uint64_t param, uint64_t k,
             // (v2 = a_dt<<28 | (expires-a_dt)<<4 | category-10); emit
             // ((auction<<4)|cat, price) when the bid falls in the auction's
             // validity window (q4.rs:58-68)
-            const uint64_t bid_dt = v1 >> 20, price = v1 & 0xFFFFFull;
+            const uint64_t bid_dt = v1 >> 27, price = v1 & 0x7FFFFFFull;
             const uint64_t a_dt = v2 >> 28, dur = (v2 >> 4) & 0xFFFFFFull;
             hi = (k << 4) | (v2 & 0xFull);
             lo = price;
             return bid_dt >= a_dt && bid_dt <= a_dt + dur;
         }
         case DBSP_PROJ_Q6_BID_X_AUC: {
-            // q6.rs:60-80: delta = bid (v1 = bid_dt<<20|price), trace =
+            // q6.rs:60-80: delta = bid (v1 = bid_dt<<27|price), trace =
             // auction (v2 = a_dt<<34 | (expires-a_dt)<<20 | seller); emit
             // ((auction<<20)|seller, price) in the validity window
-            const uint64_t bid_dt = v1 >> 20, price = v1 & 0xFFFFFull;
+            const uint64_t bid_dt = v1 >> 27, price = v1 & 0x7FFFFFFull;
             const uint64_t a_dt = v2 >> 36, dur = (v2 >> 20) & 0xFFFFull;
             hi = (k << 20) | (v2 & 0xFFFFFull);
             lo = price;
             return bid_dt >= a_dt && bid_dt <= a_dt + dur;
         }
         case DBSP_PROJ_Q6_AUC_X_BID: {
-            const uint64_t bid_dt = v2 >> 20, price = v2 & 0xFFFFFull;
+            const uint64_t bid_dt = v2 >> 27, price = v2 & 0x7FFFFFFull;
             const uint64_t a_dt = v1 >> 36, dur = (v1 >> 20) & 0xFFFFull;
             hi = (k << 20) | (v1 & 0xFFFFFull);
             lo = price;
             return bid_dt >= a_dt && bid_dt <= a_dt + dur;
         }
         case DBSP_PROJ_Q4_AUC_X_BID: {  // sides swapped
-            const uint64_t bid_dt = v2 >> 20, price = v2 & 0xFFFFFull;
+            const uint64_t bid_dt = v2 >> 27, price = v2 & 0x7FFFFFFull;
             const uint64_t a_dt = v1 >> 28, dur = (v1 >> 4) & 0xFFFFFFull;
             hi = (k << 4) | (v1 & 0xFull);
             lo = price;
@@ -1920,7 +1920,7 @@ __global__ void k_agg_emit(const uint64_t *keys, int64_t nd,
                     ihi - ilo < 10 ? ihi - ilo : (uint64_t)10;
                 uint64_t sum = 0;
                 for (uint64_t t = ihi - n10; t < ihi; t++)
-                    sum += iv[t] & 0xFFFFFull;
+                    sum += iv[t] & 0x7FFFFFFull;
                 nv = sum / n10;
             } else {
                 nv = newval[i];
@@ -2288,7 +2288,7 @@ __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
             const uint64_t pos1 = wave_append((unsigned long long *)c1, p1);
             if (p1) {
                 k1[pos1] = e.f0;
-                v1[pos1] = (e.f3 << 20) | (e.f2 & 0xFFFFFull);
+                v1[pos1] = (e.f3 << 27) | (e.f2 & 0x7FFFFFFull);
                 w1[pos1] = e.w;
             }
         } else if (query == 6) {
@@ -2307,7 +2307,7 @@ __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
             const uint64_t pos1 = wave_append((unsigned long long *)c1, p1);
             if (p1) {
                 k1[pos1] = e.f0;
-                v1[pos1] = (e.f3 << 20) | (e.f2 & 0xFFFFFull);
+                v1[pos1] = (e.f3 << 27) | (e.f2 & 0x7FFFFFFull);
                 w1[pos1] = e.w;
             }
         } else if (query == 5) {
@@ -2344,10 +2344,10 @@ __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
 // mode 1: (k,v) -> (v, 0)                             [q8 auctions map / q5 windowed-bids map]
 // mode 2: (k,v) -> (0, v)                             [q5 map_index ((),count)]
 // mode 3: (k,v) -> (v, k)                             [q5 by_count map_index]
-// mode 5: (k=auction<<4|cat, v=price) -> (cat, 0) with w' = w*(price<<20|1)
+// mode 5: (k=auction<<4|cat, v=price) -> (cat, 0) with w' = w*(price<<18|1)
 //         [q4 average weigh: one linear pass accumulates (sum<<20)+count
-//          exactly — price < 2^20 and per-category count < 2^20]
-// mode 6: (k=cat, v=(sum<<20)|count) -> (cat, sum/count)  [q4 average output,
+//          exactly — price < 2^27 (price.rs) and count < 2^18]
+// mode 6: (k=cat, v=(sum<<18)|count) -> (cat, sum/count)  [q4 average output,
 //          integer division as the reference's isize avg (average.rs)]
 __global__ void k_map(const uint64_t *k, const uint64_t *v, const int64_t *w,
                       int64_t n, int mode, uint64_t *ok, uint64_t *ov,
@@ -2369,21 +2369,24 @@ __global__ void k_map(const uint64_t *k, const uint64_t *v, const int64_t *w,
                 break;
             }
             case 5:
+                // (sum<<18)+count packed weight: price <= 10^8 (price.rs),
+                // per-category sum <= ~1.2e13 -> sum<<18 < 2^62; count
+                // (auctions per category) < 2^18
                 rk = kk & 0xFull;
                 rv = 0;
-                rw = w[i] * (int64_t)((vv << 20) | 1ull);
+                rw = w[i] * (int64_t)((vv << 18) | 1ull);
                 break;
             case 6: {
-                const int64_t cnt = (int64_t)(vv & 0xFFFFFull);
+                const int64_t cnt = (int64_t)(vv & 0x3FFFFull);
                 rk = kk;
-                rv = cnt > 0 ? (vv >> 20) / (uint64_t)cnt : 0;
+                rv = cnt > 0 ? (vv >> 18) / (uint64_t)cnt : 0;
                 break;
             }
             case 7:  // q6 map_index (q6.rs:92-94): winning bids keyed by
                      // seller, val (auction<<20)|price keeps cursor order
                      // by auction id for the last-10 fold
                 rk = kk & 0xFFFFFull;
-                rv = ((kk >> 20) << 20) | (vv & 0xFFFFFull);
+                rv = ((kk >> 20) << 27) | (vv & 0x7FFFFFFull);
                 break;
             default: rk = vv; rv = kk; break;
         }

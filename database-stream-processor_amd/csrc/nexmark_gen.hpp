@@ -23,6 +23,7 @@
 // ids its US_CITIES table (people.rs:27-38, Phoenix=0), name ids are opaque
 // u32s standing for the generated first+last name combination.
 #pragma once
+#include <cmath>
 #include <cstdint>
 #include "../../include/dbsp_hip.h"
 
@@ -47,15 +48,39 @@ constexpr uint64_t NUM_IN_FLIGHT_AUCTIONS = 100;  // config.rs:141
 constexpr uint64_t NUM_US_STATES = 6, NUM_US_CITIES = 10;
 constexpr double DEFAULT_RATE = 10'000'000.0;   // events/s, config.rs:50
 
-struct Rng {  // splitmix64 (documented; replaces the reference's SmallRng)
+struct Rng {
+    // Raw stream: splitmix64 (documented; stands in for the reference bench's
+    // UNSEEDED ThreadRng — lib.rs:198 `create_generators_for_config::<ThreadRng>`
+    // — so no reproducible reference event stream exists to pin against).
+    // step=true switches to the reference generator tests' StepRng(0,1)
+    // (rand::rngs::mock::StepRng: raw values 0,1,2,...) so those tests'
+    // expected events can be asserted verbatim.
     uint64_t s;
+    bool step = false;
     uint64_t next() {
+        if (step) return s++;
         uint64_t z = (s += 0x9E3779B97F4A7C15ull);
         z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
         z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
         return z ^ (z >> 31);
     }
-    uint64_t range(uint64_t n) { return n ? next() % n : 0; }
+    // rand 0.8.5 UniformInt::sample_single (uniform_int_impl!, the
+    // widening-multiply + zone-rejection method) — the exact arithmetic the
+    // reference's rng.gen_range(0..n) performs on the raw draw
+    uint64_t range(uint64_t n) {
+        if (n == 0) return 0;
+        const uint64_t zone = (n << __builtin_clzll(n)) - 1;
+        for (;;) {
+            const uint64_t v = next();
+            const unsigned __int128 m = (unsigned __int128)v * n;
+            if ((uint64_t)m <= zone) return (uint64_t)(m >> 64);
+        }
+    }
+    // rand's f32 range sample on [0,1): 24 high mantissa bits of next_u32
+    // (xoshiro-style next_u32 = upper half of next_u64)
+    float f01() {
+        return (float)((uint32_t)(next() >> 32) >> 8) * (1.0f / 16777216.0f);
+    }
 };
 
 struct Generator {
@@ -114,6 +139,23 @@ struct Generator {
         return min_a + rng.range(last - min_a + 1);
     }
 
+    // generator/price.rs:9-11: ceil(10^(U(0,1)*6) * 100) — log-uniform in
+    // [100, 10^8]
+    uint64_t next_price() {
+        return (uint64_t)std::ceil(std::pow(10.0f, rng.f01() * 6.0f) * 100.0f);
+    }
+
+    // generator/auctions.rs:125-148: auction length with the in-flight
+    // horizon (events for NUM_IN_FLIGHT_AUCTIONS more auctions)
+    uint64_t next_auction_length_ms(uint64_t events_count, uint64_t ts) {
+        const uint64_t num_events_for_auctions =
+            NUM_IN_FLIGHT_AUCTIONS * TOTAL_PROP / AUCTION_PROP;  // 1666
+        const uint64_t future = timestamp_for(events_count + num_events_for_auctions);
+        const uint64_t horizon = future > ts ? future - ts : 0;
+        const uint64_t cap = horizon * 2 > 1 ? horizon * 2 : 1;
+        return 1 + rng.range(cap);
+    }
+
     // One event.  kind/f* layout per include/dbsp_hip.h.
     dbsp_event next() {
         uint64_t event_id = count;  // single generator: first_event_id = 0
@@ -122,50 +164,79 @@ struct Generator {
         dbsp_event e{};
         e.w = 1;
         if (rem < PERSON_PROP) {
-            // generator/people.rs:50-79
-            e.kind = 0;
-            e.f0 = last_base0_person_id(event_id) + FIRST_PERSON_ID;
-            e.f1 = rng.range(1000);  // name id (dictionary of ~1000 generated
-                                     // first+last combinations, people.rs:40-46)
-            e.f2 = rng.range(NUM_US_CITIES);      // city id
-            e.f3 = rng.range(NUM_US_STATES);      // state id
-            e.f4 = ts;
+            e = next_person(event_id, ts);
         } else if (rem < PERSON_PROP + AUCTION_PROP) {
-            // generator/auctions.rs:26-82
-            e.kind = 1;
-            e.f0 = last_base0_auction_id(event_id) + FIRST_AUCTION_ID;
-            uint64_t seller;
-            if (rng.range(CFG_HOT_SELLERS_RATIO) == 0)
-                seller = next_base0_person_id(event_id);
-            else
-                seller = (last_base0_person_id(event_id) / HOT_SELLER_RATIO_CONST) *
-                         HOT_SELLER_RATIO_CONST;
-            e.f1 = seller + FIRST_PERSON_ID;
-            e.f2 = FIRST_CATEGORY_ID + rng.range(NUM_CATEGORIES);
-            e.f3 = ts;
-            e.f4 = ts + 10'000;  // expires; unused by q0/q3/q5/q8
+            e = next_auction(count, event_id, ts);
         } else {
-            // generator/bids.rs:61-100
-            e.kind = 2;
-            uint64_t auction;
-            if (rng.range(CFG_HOT_AUCTION_RATIO) == 0)
-                auction = next_base0_auction_id(event_id);
-            else
-                auction = (last_base0_auction_id(event_id) / HOT_AUCTION_RATIO_CONST) *
-                          HOT_AUCTION_RATIO_CONST;
-            e.f0 = auction + FIRST_AUCTION_ID;
-            uint64_t bidder;
-            if (rng.range(CFG_HOT_BIDDERS_RATIO) == 0)
-                bidder = next_base0_person_id(event_id);
-            else
-                bidder = (last_base0_person_id(event_id) / HOT_BIDDER_RATIO_CONST) *
-                             HOT_BIDDER_RATIO_CONST + 1;
-            e.f1 = bidder + FIRST_PERSON_ID;
-            e.f2 = rng.range(10'000);  // price stand-in (price.rs)
-            e.f3 = ts;
-            e.f4 = 0;
+            e = next_bid(event_id, ts);
         }
         count++;
+        return e;
+    }
+
+    // generator/people.rs:50-79.  Draw order follows the reference for the
+    // fields carried here (name, city, state); the email / credit-card /
+    // extra string draws are not emitted and are skipped — with the test
+    // StepRng their gen_range results are all 0, so the reference's own
+    // generator tests remain assertable verbatim.
+    dbsp_event next_person(uint64_t event_id, uint64_t ts) {
+        dbsp_event e{};
+        e.w = 1;
+        e.kind = 0;
+        e.f0 = last_base0_person_id(event_id) + FIRST_PERSON_ID;
+        const uint64_t first = rng.range(11);  // FIRST_NAMES (people.rs:40-42)
+        const uint64_t last = rng.range(10);   // LAST_NAMES (people.rs:44-46)
+        e.f1 = first * 16 + last;              // name id = (first, last) pair
+        e.f2 = rng.range(NUM_US_CITIES);       // city id (Phoenix = 0)
+        e.f3 = rng.range(NUM_US_STATES);       // state id (AZ = 0)
+        e.f4 = ts;
+        return e;
+    }
+
+    // generator/auctions.rs:26-82 (draw order: seller, category, initial_bid
+    // — drawn, not emitted — then the length horizon draw)
+    dbsp_event next_auction(uint64_t events_count, uint64_t event_id,
+                            uint64_t ts) {
+        dbsp_event e{};
+        e.w = 1;
+        e.kind = 1;
+        e.f0 = last_base0_auction_id(event_id) + FIRST_AUCTION_ID;
+        uint64_t seller;
+        if (rng.range(CFG_HOT_SELLERS_RATIO) == 0)
+            seller = next_base0_person_id(event_id);
+        else
+            seller = (last_base0_person_id(event_id) / HOT_SELLER_RATIO_CONST) *
+                     HOT_SELLER_RATIO_CONST;
+        e.f1 = seller + FIRST_PERSON_ID;
+        e.f2 = FIRST_CATEGORY_ID + rng.range(NUM_CATEGORIES);
+        (void)next_price();  // initial_bid: drawn by the reference, not emitted
+        e.f3 = ts;
+        e.f4 = ts + next_auction_length_ms(events_count, ts);  // expires
+        return e;
+    }
+
+    // generator/bids.rs:60-100 (channel/url draws not emitted, skipped)
+    dbsp_event next_bid(uint64_t event_id, uint64_t ts) {
+        dbsp_event e{};
+        e.w = 1;
+        e.kind = 2;
+        uint64_t auction;
+        if (rng.range(CFG_HOT_AUCTION_RATIO) == 0)
+            auction = next_base0_auction_id(event_id);
+        else
+            auction = (last_base0_auction_id(event_id) / HOT_AUCTION_RATIO_CONST) *
+                      HOT_AUCTION_RATIO_CONST;
+        e.f0 = auction + FIRST_AUCTION_ID;
+        uint64_t bidder;
+        if (rng.range(CFG_HOT_BIDDERS_RATIO) == 0)
+            bidder = next_base0_person_id(event_id);
+        else
+            bidder = (last_base0_person_id(event_id) / HOT_BIDDER_RATIO_CONST) *
+                         HOT_BIDDER_RATIO_CONST + 1;
+        e.f1 = bidder + FIRST_PERSON_ID;
+        e.f2 = next_price();
+        e.f3 = ts;
+        e.f4 = 0;
         return e;
     }
 };

@@ -113,11 +113,11 @@ inline void proj_emit(std::vector<Row> &out, dbsp_proj proj, uint64_t param,
         case DBSP_PROJ_HI_K_LO_V2V1: hi = k; lo = (v2 << 32) | (v1 & 0xFFFFFFFFull); break;
         case DBSP_PROJ_HI_K_LO_V2: hi = k; lo = v2; break;
         case DBSP_PROJ_Q4_BID_X_AUC: {
-            // q4.rs:58-68 join_func: delta = bid (v1 = bid_dt<<20|price),
+            // q4.rs:58-68 join_func: delta = bid (v1 = bid_dt<<27|price),
             // trace = auction (v2 = a_dt<<28|(expires-a_dt)<<4|cat&0xF);
             // invalid pairs keep their slot with weight 0 (dropped by the
             // consolidate), mirroring the GPU emit contract
-            const uint64_t bid_dt = v1 >> 20, price = v1 & 0xFFFFFull;
+            const uint64_t bid_dt = v1 >> 27, price = v1 & 0x7FFFFFFull;
             const uint64_t a_dt = v2 >> 28, dur = (v2 >> 4) & 0xFFFFFFull;
             hi = (k << 4) | (v2 & 0xFull);
             lo = price;
@@ -125,9 +125,9 @@ inline void proj_emit(std::vector<Row> &out, dbsp_proj proj, uint64_t param,
             break;
         }
         case DBSP_PROJ_Q6_BID_X_AUC: {
-            // q6.rs:60-80: delta = bid (v1 = bid_dt<<20|price), trace =
+            // q6.rs:60-80: delta = bid (v1 = bid_dt<<27|price), trace =
             // auction (v2 = a_dt<<34 | (expires-a_dt)<<20 | seller)
-            const uint64_t bid_dt = v1 >> 20, price = v1 & 0xFFFFFull;
+            const uint64_t bid_dt = v1 >> 27, price = v1 & 0x7FFFFFFull;
             const uint64_t a_dt = v2 >> 36, dur = (v2 >> 20) & 0xFFFFull;
             hi = (k << 20) | (v2 & 0xFFFFFull);
             lo = price;
@@ -135,7 +135,7 @@ inline void proj_emit(std::vector<Row> &out, dbsp_proj proj, uint64_t param,
             break;
         }
         case DBSP_PROJ_Q6_AUC_X_BID: {
-            const uint64_t bid_dt = v2 >> 20, price = v2 & 0xFFFFFull;
+            const uint64_t bid_dt = v2 >> 27, price = v2 & 0x7FFFFFFull;
             const uint64_t a_dt = v1 >> 36, dur = (v1 >> 20) & 0xFFFFull;
             hi = (k << 20) | (v1 & 0xFFFFFull);
             lo = price;
@@ -143,7 +143,7 @@ inline void proj_emit(std::vector<Row> &out, dbsp_proj proj, uint64_t param,
             break;
         }
         case DBSP_PROJ_Q4_AUC_X_BID: {
-            const uint64_t bid_dt = v2 >> 20, price = v2 & 0xFFFFFull;
+            const uint64_t bid_dt = v2 >> 27, price = v2 & 0x7FFFFFFull;
             const uint64_t a_dt = v1 >> 28, dur = (v1 >> 4) & 0xFFFFFFull;
             hi = (k << 4) | (v1 & 0xFull);
             lo = price;
@@ -422,7 +422,7 @@ std::vector<Row> q4_step(Oracle &o, const std::vector<dbsp_event> &ev) {
                                     (e.f2 & 0xFull),
                           e.w});
         if (e.kind == 2)
-            dB.push_back({e.f0, (e.f3 << 20) | (e.f2 & 0xFFFFFull), e.w});
+            dB.push_back({e.f0, (e.f3 << 27) | (e.f2 & 0x7FFFFFFull), e.w});
     }
     consolidate(dA);
     consolidate(dB);
@@ -445,7 +445,7 @@ std::vector<Row> q4_step(Oracle &o, const std::vector<dbsp_event> &ev) {
     // average per category: weigh to the packed (sum<<20)+count weight
     std::vector<Row> dAvgIn;
     for (auto &r : dWin)
-        dAvgIn.push_back({r.k & 0xFull, 0, r.w * (int64_t)((r.v << 20) | 1ull)});
+        dAvgIn.push_back({r.k & 0xFull, 0, r.w * (int64_t)((r.v << 18) | 1ull)});
     consolidate(dAvgIn);
     std::vector<Row> out;
     if (!dAvgIn.empty()) {
@@ -458,8 +458,8 @@ std::vector<Row> q4_step(Oracle &o, const std::vector<dbsp_event> &ev) {
         consolidate(upd);
         o.q4_avgout = merge(o.q4_avgout, upd);
         for (auto &r : upd) {
-            const uint64_t cnt = r.v & 0xFFFFFull;
-            out.push_back({r.k, cnt ? (r.v >> 20) / cnt : 0, r.w});
+            const uint64_t cnt = r.v & 0x3FFFFull;
+            out.push_back({r.k, cnt ? (r.v >> 18) / cnt : 0, r.w});
         }
         consolidate(out);
     }
@@ -477,7 +477,7 @@ std::vector<Row> q6_step(Oracle &o, const std::vector<dbsp_event> &ev) {
                                     (e.f1 & 0xFFFFFull),
                           e.w});
         if (e.kind == 2)
-            dB.push_back({e.f0, (e.f3 << 20) | (e.f2 & 0xFFFFFull), e.w});
+            dB.push_back({e.f0, (e.f3 << 27) | (e.f2 & 0x7FFFFFFull), e.w});
     }
     consolidate(dA);
     consolidate(dB);
@@ -499,7 +499,7 @@ std::vector<Row> q6_step(Oracle &o, const std::vector<dbsp_event> &ev) {
     }
     std::vector<Row> dFoldIn;
     for (auto &r : dWin)
-        dFoldIn.push_back({r.k & 0xFFFFFull, ((r.k >> 20) << 20) | (r.v & 0xFFFFFull), r.w});
+        dFoldIn.push_back({r.k & 0xFFFFFull, ((r.k >> 20) << 27) | (r.v & 0x7FFFFFFull), r.w});
     consolidate(dFoldIn);
     std::vector<Row> out;
     if (!dFoldIn.empty()) {
@@ -518,7 +518,7 @@ std::vector<Row> q6_step(Oracle &o, const std::vector<dbsp_event> &ev) {
                 size_t n10 = vals.size() < 10 ? vals.size() : 10;
                 uint64_t sum = 0;
                 for (size_t t = vals.size() - n10; t < vals.size(); t++)
-                    sum += vals[t] & 0xFFFFFull;
+                    sum += vals[t] & 0x7FFFFFFull;
                 out.push_back({key, sum / n10, 1});
             }
             for (auto &r : o.q6_foldout)

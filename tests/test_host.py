@@ -87,3 +87,47 @@ def test_event_struct_layout():
     e["w"] = -3
     raw = e.tobytes()
     assert raw[0] == 2 and raw[48:56] == (-3).to_bytes(8, "little", signed=True)
+
+
+def test_generator_reference_steprng_vectors():
+    """The reference's OWN generator unit tests replayed through the
+    restated generator with their StepRng(0,1) (rand::rngs::mock::StepRng;
+    generator/mod.rs:147-159 make_test_generator, base_time 0):
+      - people.rs:163-184 test_next_person: event 105 -> id 1002,
+        name "Peter Shultz" (FIRST_NAMES[0], LAST_NAMES[0] -> name id 0),
+        city "Phoenix" (0), state "AZ" (0)
+      - auctions.rs:152-244 test_next_auction: id 1000, seller 1000,
+        category 10, expires = ts + 1 (length draw 1 with zeros);
+        test_next_auction_length -> 1
+      - bids.rs:132-170 test_next_bid cases: (event_id, auction, bidder) =
+        (0, 1000, 1000), (50*34+3, 1004, 1000), (50*1500, 5399, 1501),
+        price 100 (price.rs test: 10^0 * 100)
+    These pin the generator's deterministic skeleton and its rand-0.8.5
+    gen_range arithmetic; stream-level parity remains unpinnable because the
+    reference bench draws from an UNSEEDED ThreadRng (lib.rs:198)."""
+    import ctypes
+    from dbsp_amd import EVENT_DT, load_gen_lib
+    L = load_gen_lib()
+    L.dbsp_gen_unit.restype = ctypes.c_int64
+    L.dbsp_gen_unit.argtypes = [ctypes.c_int, ctypes.c_uint64,
+                                ctypes.c_uint64, ctypes.c_void_p]
+    out = np.zeros(1, dtype=EVENT_DT)
+    p = out.ctypes.data_as(ctypes.c_void_p)
+
+    L.dbsp_gen_unit(0, 105, 1_000_000_000_000, p)
+    e = out[0]
+    assert (e["kind"], e["f0"], e["f1"], e["f2"], e["f3"], e["f4"]) == \
+        (0, 1002, 0, 0, 0, 1_000_000_000_000)
+
+    L.dbsp_gen_unit(1, 0, 0, p)
+    e = out[0]
+    assert (e["kind"], e["f0"], e["f1"], e["f2"], e["f3"], e["f4"]) == \
+        (1, 1000, 1000, 10, 0, 1)
+
+    for event_id, auction, bidder in [(0, 1000, 1000),
+                                      (50 * 34 + 3, 1004, 1000),
+                                      (50 * 1500, 5399, 1501)]:
+        L.dbsp_gen_unit(2, event_id, 1_000_000_000_000, p)
+        e = out[0]
+        assert (e["kind"], e["f0"], e["f1"], e["f2"]) == \
+            (2, auction, bidder, 100), (event_id, tuple(e))
