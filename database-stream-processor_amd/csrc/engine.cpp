@@ -996,9 +996,14 @@ struct dbsp_engine {
     int query = 0;
     int rank = 0, world = 1;
 
-    // staged input events (resident in HBM before the timed region)
+    // staged input events (resident in HBM before the timed region) and
+    // their device column split (§8f3 columnizer, input.rs:591-721): the
+    // per-tick flatmaps read the SoA columns; the AoS copy is kept for the
+    // unstaged dbsp_engine_step path and q0
     dbsp_event *d_events = nullptr;
     int64_t n_events = 0;
+    int64_t n_staged = 0;
+    uint64_t *ev_cols[7] = {};  // kind,f0..f4,w
     std::vector<dbsp_event> h_events;  // host copy for the q0 CPU path
 
     // q3 state
@@ -1112,6 +1117,8 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
     }
     if (e->out_store.k) free_batch(c, e->out_store);
     if (e->d_events) (void)dbspk::cache_free(e->d_events, c->stream);
+    for (int i = 0; i < 7; i++)
+        if (e->ev_cols[i]) (void)dbspk::cache_free(e->ev_cols[i], c->stream);
     (void)hipStreamSynchronize(c->stream);
     delete e;
     return DBSP_OK;
@@ -1128,8 +1135,23 @@ extern "C" dbsp_status dbsp_engine_stage_events(dbsp_engine *e,
     HIP_CHECK_ST(dbspk::cache_malloc((void **)&e->d_events, n * sizeof(dbsp_event) + 64, c->stream));
     HIP_CHECK_ST(hipMemcpyAsync(e->d_events, events, n * sizeof(dbsp_event),
                                 hipMemcpyHostToDevice, c->stream));
+    // device column split (§8f3): one pass at stage time, outside the timed
+    // region; per-tick flatmaps then read coalesced SoA streams
+    for (int i = 0; i < 7; i++) {
+        if (e->ev_cols[i]) {
+            HIP_CHECK_ST(dbspk::cache_free(e->ev_cols[i], c->stream));
+            e->ev_cols[i] = nullptr;
+        }
+        HIP_CHECK_ST(dbspk::cache_malloc((void **)&e->ev_cols[i],
+                                         n * 8 + 64, c->stream));
+    }
+    TRY(dbspk::columnize_events(c->stream, e->d_events, n, e->ev_cols[0],
+                                e->ev_cols[1], e->ev_cols[2], e->ev_cols[3],
+                                e->ev_cols[4], e->ev_cols[5],
+                                (int64_t *)e->ev_cols[6]));
     HIP_CHECK_ST(hipStreamSynchronize(c->stream));
     e->n_events = n;
+    e->n_staged = n;
     if (e->query == 0)
         e->h_events.assign(events, events + n);
     return DBSP_OK;
@@ -1365,6 +1387,19 @@ static dbsp_status build_deltas_chain(dbsp_engine *e, const dbsp_event *d_ev,
                                       int64_t n, DevBatch &rawA, DevBatch &rawB,
                                       DevBatch &oA, DevBatch &oB);
 
+// If d_ev points into the staged event array, return its column-split view
+// (shifted to the same offset) so the flatmap reads coalesced SoA streams.
+static bool staged_cols(dbsp_engine *e, const dbsp_event *d_ev,
+                        dbspk::EventCols &out) {
+    if (!e->ev_cols[0] || !e->d_events || d_ev < e->d_events) return false;
+    const int64_t off = d_ev - e->d_events;
+    if (off < 0 || off >= e->n_staged) return false;
+    out = {e->ev_cols[0] + off, e->ev_cols[1] + off, e->ev_cols[2] + off,
+           e->ev_cols[3] + off, e->ev_cols[4] + off, e->ev_cols[5] + off,
+           (const int64_t *)(e->ev_cols[6] + off)};
+    return true;
+}
+
 static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
                                 int64_t n, DevBatch &d0, DevBatch &d1,
                                 bool want_two) {
@@ -1382,8 +1417,11 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
         TRY(alloc_batch(c, n > 0 ? n : 1, raw0, true));
         TRY(alloc_batch(c, n > 0 ? n : 1, raw1, true));
         int64_t n0 = 0, n1 = 0;
-        TRY(dbspk::flatmap_events(c->stream, d_ev, n, e->query, raw0.k, raw0.v,
-                                  raw0.w, &n0, raw1.k, raw1.v, raw1.w, &n1));
+        dbspk::EventCols cols{};
+        const bool hc = staged_cols(e, d_ev, cols);
+        TRY(dbspk::flatmap_events(c->stream, d_ev, hc ? &cols : nullptr, n,
+                                  e->query, raw0.k, raw0.v, raw0.w, &n0,
+                                  raw1.k, raw1.v, raw1.w, &n1));
         raw0.n = n0;
         raw1.n = n1;
         if (want_two && n0 <= 8192 && n1 <= 8192) {
@@ -1454,8 +1492,11 @@ static dbsp_status build_deltas_chain(dbsp_engine *e, const dbsp_event *d_ev,
     int64_t cap = n > 0 ? n : 1;
     TRY(alloc_batch(c, cap, rawA, true));
     TRY(alloc_batch(c, cap, rawB, true));
-    TRY(dbspk::flatmap_events_chain(c->stream, d_ev, n, e->query, rawA.k,
-                                    rawA.v, rawA.w, rawB.k, rawB.v, rawB.w,
+    dbspk::EventCols cols{};
+    const bool hc = staged_cols(e, d_ev, cols);
+    TRY(dbspk::flatmap_events_chain(c->stream, d_ev, hc ? &cols : nullptr, n,
+                                    e->query, rawA.k, rawA.v, rawA.w, rawB.k,
+                                    rawB.v, rawB.w,
                                     (uint64_t *)(c->d_len + 8)));
     DevBatch sA, sB;
     TRY(alloc_batch(c, cap, sA, true));
@@ -1501,8 +1542,11 @@ static dbsp_status build_deltas_chain_sharded(
     const int64_t cap = n > 0 ? n : 1;
     TRY(alloc_batch(c, cap, rawA, true));
     TRY(alloc_batch(c, cap, rawB, true));
-    TRY(dbspk::flatmap_events_chain(c->stream, d_ev, n, e->query, rawA.k,
-                                    rawA.v, rawA.w, rawB.k, rawB.v, rawB.w,
+    dbspk::EventCols cols{};
+    const bool hc = staged_cols(e, d_ev, cols);
+    TRY(dbspk::flatmap_events_chain(c->stream, d_ev, hc ? &cols : nullptr, n,
+                                    e->query, rawA.k, rawA.v, rawA.w, rawB.k,
+                                    rawB.v, rawB.w,
                                     (uint64_t *)(c->d_len + 8)));
     // frame capacities from the deterministic event mix (config.rs:128-143)
     const int64_t P0 = 2 * (n / 50 + 64) / world + 512;
@@ -2739,7 +2783,10 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         } else {
             TRY(alloc_batch(c, cap, raw0, true));
             TRY(alloc_batch(c, cap, raw1, true));
-            TRY(dbspk::flatmap_events_chain(c->stream, d_ev, n, e->query,
+            dbspk::EventCols cols{};
+            const bool hc = staged_cols(e, d_ev, cols);
+            TRY(dbspk::flatmap_events_chain(c->stream, d_ev,
+                                            hc ? &cols : nullptr, n, e->query,
                                             raw0.k, raw0.v, raw0.w, raw1.k,
                                             raw1.v, raw1.w,
                                             (uint64_t *)(c->d_len + 8)));
@@ -3033,12 +3080,14 @@ static dbsp_status q5_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             const int64_t ncap = e->next_n > 0 ? e->next_n : 1;
             dbsp_status st = alloc_batch(c, ncap, e->front.rawA, true);
             if (st == DBSP_OK) st = alloc_batch(c, ncap, e->front.rawB, true);
+            dbspk::EventCols fcols{};
+            const bool fhc = staged_cols(e, e->next_ev, fcols);
             if (st == DBSP_OK)
                 st = dbspk::flatmap_events_chain(
-                    c->stream, e->next_ev, e->next_n, e->query,
-                    e->front.rawA.k, e->front.rawA.v, e->front.rawA.w,
-                    e->front.rawB.k, e->front.rawB.v, e->front.rawB.w,
-                    (uint64_t *)(c->d_len + 8));
+                    c->stream, e->next_ev, fhc ? &fcols : nullptr, e->next_n,
+                    e->query, e->front.rawA.k, e->front.rawA.v,
+                    e->front.rawA.w, e->front.rawB.k, e->front.rawB.v,
+                    e->front.rawB.w, (uint64_t *)(c->d_len + 8));
             if (st == DBSP_OK)
                 st = dbspk::minmax_rows_chain(
                     c->stream, e->front.rawA.k, e->front.rawA.v, ncap,

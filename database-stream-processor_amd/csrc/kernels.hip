@@ -2244,7 +2244,24 @@ __device__ inline uint64_t wave_append(unsigned long long *ctr, bool pred) {
     return base + (uint64_t)__popcll(m & lt);
 }
 
-__global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
+__global__ void k_columnize(const dbsp_event *ev, int64_t n, uint64_t *kind,
+                            uint64_t *f0, uint64_t *f1, uint64_t *f2,
+                            uint64_t *f3, uint64_t *f4, int64_t *w) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const dbsp_event e = ev[i];
+        kind[i] = e.kind;
+        f0[i] = e.f0;
+        f1[i] = e.f1;
+        f2[i] = e.f2;
+        f3[i] = e.f3;
+        f4[i] = e.f4;
+        w[i] = e.w;
+    }
+}
+
+__global__ void k_flatmap(const dbsp_event *ev, dbspk::EventCols cols, int64_t n,
+                          int query,
                           uint64_t *c0, uint64_t *k0, uint64_t *v0, int64_t *w0,
                           uint64_t *c1, uint64_t *k1, uint64_t *v1, int64_t *w1) {
     // wave-uniform outer loop (wb is the wave's first row) so the ballots in
@@ -2257,7 +2274,15 @@ __global__ void k_flatmap(const dbsp_event *ev, int64_t n, int query,
         const int64_t i = wb + lane;
         dbsp_event e{};
         const bool act = i < n;
-        if (act) e = ev[i];
+        if (act) {
+            if (cols.kind) {  // columnized staging: 7 coalesced 8 B streams
+                e.kind = cols.kind[i]; e.f0 = cols.f0[i]; e.f1 = cols.f1[i];
+                e.f2 = cols.f2[i]; e.f3 = cols.f3[i]; e.f4 = cols.f4[i];
+                e.w = cols.w[i];
+            } else {
+                e = ev[i];
+            }
+        }
         if (query == 3) {
             // q3.rs:37-49
             const bool p0 = act && e.kind == 1 && e.f2 == 10;
@@ -3353,27 +3378,41 @@ dbsp_status frames_unpack_pair(hipStream_t s, const uint64_t *frame,
     return DBSP_OK;
 }
 
-dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
-                                 int64_t n, int query, uint64_t *k0,
-                                 uint64_t *v0, int64_t *w0, uint64_t *k1,
-                                 uint64_t *v1, int64_t *w1,
-                                 uint64_t *ctr /* 2 device slots */) {
-    HIP_CHECK(hipMemsetAsync(ctr, 0, 2 * sizeof(uint64_t), s));
+dbsp_status columnize_events(hipStream_t s, const dbsp_event *ev, int64_t n,
+                             uint64_t *kind, uint64_t *f0, uint64_t *f1,
+                             uint64_t *f2, uint64_t *f3, uint64_t *f4,
+                             int64_t *w) {
     if (n > 0)
-        k_flatmap<<<grid_for(n), BLK, 0, s>>>(ev, n, query, ctr, k0, v0, w0,
-                                              ctr + 1, k1, v1, w1);
+        k_columnize<<<grid_for(n), BLK, 0, s>>>(ev, n, kind, f0, f1, f2, f3,
+                                                f4, w);
     return DBSP_OK;
 }
 
-dbsp_status flatmap_events(hipStream_t s, const dbsp_event *ev, int64_t n,
-                           int query, uint64_t *k0, uint64_t *v0, int64_t *w0,
+dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
+                                 const EventCols *cols, int64_t n, int query,
+                                 uint64_t *k0, uint64_t *v0, int64_t *w0,
+                                 uint64_t *k1, uint64_t *v1, int64_t *w1,
+                                 uint64_t *ctr /* 2 device slots */) {
+    HIP_CHECK(hipMemsetAsync(ctr, 0, 2 * sizeof(uint64_t), s));
+    EventCols c{};
+    if (cols) c = *cols;
+    if (n > 0)
+        k_flatmap<<<grid_for(n), BLK, 0, s>>>(ev, c, n, query, ctr, k0, v0,
+                                              w0, ctr + 1, k1, v1, w1);
+    return DBSP_OK;
+}
+
+dbsp_status flatmap_events(hipStream_t s, const dbsp_event *ev,
+                           const EventCols *cols, int64_t n, int query, uint64_t *k0, uint64_t *v0, int64_t *w0,
                            int64_t *n0, uint64_t *k1, uint64_t *v1, int64_t *w1,
                            int64_t *n1) {
+    EventCols c{};
+    if (cols) c = *cols;
     uint64_t *ctr;
     HIP_CHECK(dbspk::cache_malloc((void **)&ctr, 2 * sizeof(uint64_t), s));
     HIP_CHECK(hipMemsetAsync(ctr, 0, 2 * sizeof(uint64_t), s));
     if (n > 0)
-        k_flatmap<<<grid_for(n), BLK, 0, s>>>(ev, n, query, ctr, k0, v0, w0,
+        k_flatmap<<<grid_for(n), BLK, 0, s>>>(ev, c, n, query, ctr, k0, v0, w0,
                                               ctr + 1, k1, v1, w1);
     uint64_t h_ctr[2];
     HIP_CHECK(hipMemcpyAsync(h_ctr, ctr, sizeof(h_ctr), hipMemcpyDeviceToHost, s));

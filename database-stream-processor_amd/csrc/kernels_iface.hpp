@@ -128,10 +128,22 @@ dbsp_status shard_rows_pair(hipStream_t s, const uint64_t *k0,
                             uint64_t *ok0, uint64_t *ov0, int64_t *ow0,
                             uint64_t *ok1, uint64_t *ov1, int64_t *ow1,
                             int64_t *h_off0, int64_t *h_off1);
+// device event columns (the §8f3 columnizer, input.rs:591-721): staged
+// events split once into SoA columns so the per-tick flatmaps read
+// coalesced 8 B streams instead of 56 B strided AoS structs
+struct EventCols {
+    const uint64_t *kind, *f0, *f1, *f2, *f3, *f4;
+    const int64_t *w;
+};
+dbsp_status columnize_events(hipStream_t s, const dbsp_event *ev, int64_t n,
+                             uint64_t *kind, uint64_t *f0, uint64_t *f1,
+                             uint64_t *f2, uint64_t *f3, uint64_t *f4,
+                             int64_t *w);
 dbsp_status flatmap_events_chain(hipStream_t s, const dbsp_event *ev,
-                                 int64_t n, int query, uint64_t *k0,
-                                 uint64_t *v0, int64_t *w0, uint64_t *k1,
-                                 uint64_t *v1, int64_t *w1, uint64_t *ctr);
+                                 const EventCols *cols, int64_t n, int query,
+                                 uint64_t *k0, uint64_t *v0, int64_t *w0,
+                                 uint64_t *k1, uint64_t *v1, int64_t *w1,
+                                 uint64_t *ctr);
 
 // fixed-frame pair exchange (engine exchange fast path): both partitioned
 // streams packed into per-peer segments of fixed capacity P0/P1 rows with a
@@ -291,10 +303,11 @@ dbsp_status shard_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
                        const int64_t *w, int64_t n, int nshards, uint64_t *ok,
                        uint64_t *ov, int64_t *ow, int64_t *h_offsets);
 
-dbsp_status flatmap_events(hipStream_t s, const dbsp_event *ev, int64_t n,
-                           int query, uint64_t *k0, uint64_t *v0, int64_t *w0,
-                           int64_t *n0, uint64_t *k1, uint64_t *v1, int64_t *w1,
-                           int64_t *n1);
+dbsp_status flatmap_events(hipStream_t s, const dbsp_event *ev,
+                           const EventCols *cols, int64_t n, int query,
+                           uint64_t *k0, uint64_t *v0, int64_t *w0,
+                           int64_t *n0, uint64_t *k1, uint64_t *v1,
+                           int64_t *w1, int64_t *n1);
 
 dbsp_status map_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
                      const int64_t *w, int64_t n, int mode, uint64_t *ok,
