@@ -420,7 +420,10 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
 static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
                                  const DevBatch &b, DevBatch &out) {
     ScopedTimer t(c, 1, (double)(a.n + b.n) * 48.0);
-    if (a.n + b.n <= 32768) {
+    // single-WG cutoff: beyond ~16k rows one workgroup serializes ~126 us
+    // (q4 trace) while the multi-block two-pass costs ~40 us; the host-length
+    // path here syncs anyway, so the pipeline split is free
+    if (a.n + b.n <= 16384) {
         // one launch + one length readback (merge-path diagonals in one WG)
         DevBatch res;
         TRY(alloc_batch(c, a.n + b.n, res));
