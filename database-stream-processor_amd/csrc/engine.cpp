@@ -595,6 +595,29 @@ extern "C" dbsp_status dbsp_join(dbsp_ctx *c, const dbsp_batch *delta,
     return DBSP_OK;
 }
 
+// Radix-tree rolling aggregate primitive (operator/time_series/radix_tree/
+// + rolling_aggregate.rs:235-280): per input row
+// (partition, ts, w) of a consolidated (partition, time)-sorted batch, the
+// weight sum over that partition's rows with time in [ts - width, ts]
+// (RelRange(Before(width), Before(0)).range_of, range.rs:93-110).  Built on
+// a flat radix-16 prefix-aggregate tree over the row array — the GPU-native
+// form of the reference's per-prefix aggregate nodes.
+extern "C" dbsp_status dbsp_rolling_agg(dbsp_ctx *c, const dbsp_batch *in,
+                                        uint64_t width, dbsp_batch *out) {
+    DevBatch res;
+    TRY(alloc_batch(c, in->len > 0 ? in->len : 1, res));
+    {
+        ScopedTimer t(c, 3, 0.0);
+        TRY(dbspk::rolling_agg_rows(c->stream, in->k, in->v, in->w, in->len,
+                                    width, res.k, res.v, res.w));
+    }
+    out->k = res.k;
+    out->v = res.v;
+    out->w = res.w;
+    out->len = in->len;
+    return DBSP_OK;
+}
+
 extern "C" dbsp_status dbsp_agg_linear_upsert(dbsp_ctx *c,
                                               const uint64_t *delta_keys,
                                               int64_t nd,

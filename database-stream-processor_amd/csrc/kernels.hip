@@ -3638,6 +3638,129 @@ dbsp_status unique_keys(hipStream_t s, const uint64_t *kk, int64_t n,
     return DBSP_OK;
 }
 
+
+// ---------------------------------------------------------------------------
+// Radix-tree rolling aggregate (operator/time_series/radix_tree/mod.rs:1-75,
+// rolling_aggregate.rs:235-280, range.rs:76-110 — SURVEY.md §8f4).  The
+// reference covers the timeline with an adaptive radix tree of per-prefix
+// aggregates, stored as indexed Z-sets, so a range aggregate visits O(log n)
+// nodes.  The MI355X-native restatement: the batch is already (partition,
+// time)-sorted, so the same O(log) range queries come from a FLAT radix-16
+// prefix-aggregate tree over the row array (levels of 16:1 weight-sum
+// reductions — "prefix-sum trees map well to GPU scans") and
+// a row's time range maps to a row range by binary search.  One query
+// thread per input row computes the rolling aggregate
+// range_of(ts) = [ts - width, ts] (RelRange::range_of, range.rs:93-110,
+// saturating at 0) within its partition — the per-row output of
+// partitioned_rolling_aggregate for the linear (weight-sum) aggregate.
+// ---------------------------------------------------------------------------
+
+#define RT_RADIX 16
+#define RT_MAX_LEVELS 16
+
+struct RtLevels {
+    int nl;
+    const int64_t *lv[RT_MAX_LEVELS];
+    int64_t n[RT_MAX_LEVELS];
+};
+
+__global__ void k_rt_reduce(const int64_t *in, int64_t n, int64_t *out,
+                            int64_t n_out) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n_out;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        int64_t s = 0;
+        const int64_t lo = i * RT_RADIX;
+        const int64_t hi = lo + RT_RADIX < n ? lo + RT_RADIX : n;
+        for (int64_t j = lo; j < hi; j++) s += in[j];
+        out[i] = s;
+    }
+}
+
+// range weight-sum over [a, b) using the tree: peel unaligned edges at each
+// level (<= 15 adds per level per side), then ascend — O(log16 n) work
+__device__ inline int64_t rt_range_sum(const int64_t *w, int64_t nw,
+                                       const RtLevels &t, int64_t a,
+                                       int64_t b) {
+    int64_t sum = 0;
+    const int64_t *cur = w;
+    int lvl = -1;
+    while (a < b) {
+        while ((a % RT_RADIX) != 0 && a < b) sum += cur[a++];
+        while ((b % RT_RADIX) != 0 && a < b) sum += cur[--b];
+        if (a >= b) break;
+        a /= RT_RADIX;
+        b /= RT_RADIX;
+        lvl++;
+        cur = t.lv[lvl];
+    }
+    return sum;
+}
+
+__global__ void k_rolling_agg(const uint64_t *k, const uint64_t *v,
+                              const int64_t *w, int64_t n, RtLevels t,
+                              uint64_t width, uint64_t *ok, uint64_t *ov,
+                              int64_t *ow) {
+    for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const uint64_t p = k[i], ts = v[i];
+        // partition row range
+        const int64_t plo = lower_bound_k(k, n, p);
+        int64_t phi = plo;
+        {
+            int64_t step = 1;
+            while (phi + step < n && k[phi + step] == p) { phi += step; step <<= 1; }
+            phi = upper_bound_k(k + phi, min(step, n - phi), p) + phi;
+        }
+        // time range [ts - width, ts] inclusive (range.rs:93-110, saturating)
+        const uint64_t t0 = ts >= width ? ts - width : 0;
+        int64_t qlo = plo, qhi = plo;
+        {   // lower_bound of t0 and upper_bound of ts within [plo, phi)
+            int64_t lo = plo, hi = phi;
+            while (lo < hi) {
+                int64_t mid = (lo + hi) / 2;
+                if (v[mid] < t0) lo = mid + 1; else hi = mid;
+            }
+            qlo = lo;
+            hi = phi;
+            while (lo < hi) {
+                int64_t mid = (lo + hi) / 2;
+                if (v[mid] <= ts) lo = mid + 1; else hi = mid;
+            }
+            qhi = lo;
+        }
+        ok[i] = p;
+        ov[i] = ts;
+        ow[i] = rt_range_sum(w, n, t, qlo, qhi);
+    }
+}
+
+dbsp_status rolling_agg_rows(hipStream_t s, const uint64_t *k,
+                             const uint64_t *v, const int64_t *w, int64_t n,
+                             uint64_t width, uint64_t *ok, uint64_t *ov,
+                             int64_t *ow) {
+    if (n <= 0) return DBSP_OK;
+    RtLevels t{};
+    int64_t cur_n = n;
+    const int64_t *cur = w;
+    int64_t *bufs[RT_MAX_LEVELS] = {};
+    while (cur_n > 1 && t.nl < RT_MAX_LEVELS) {
+        const int64_t nn = (cur_n + RT_RADIX - 1) / RT_RADIX;
+        HIP_CHECK(dbspk::cache_malloc((void **)&bufs[t.nl],
+                                      nn * sizeof(int64_t) + 8, s));
+        k_rt_reduce<<<grid_for(nn), BLK, 0, s>>>(cur, cur_n, bufs[t.nl], nn);
+        t.lv[t.nl] = bufs[t.nl];
+        t.n[t.nl] = nn;
+        cur = bufs[t.nl];
+        cur_n = nn;
+        t.nl++;
+    }
+    k_rolling_agg<<<grid_for(n), BLK, 0, s>>>(k, v, w, n, t, width, ok, ov,
+                                              ow);
+    for (int i = 0; i < t.nl; i++)
+        HIP_CHECK(dbspk::cache_free(bufs[i], s));
+    return DBSP_OK;
+}
+
 uint64_t host_xxh3_u64(uint64_t key, uint64_t seed) {
     // host copy of dev_xxh3_u64 (kept in sync; parity-tested against the oracle)
     const uint64_t SEC8_16 = 0x1cad21f72c81017cull ^ 0xdb979083e96dd4deull;

@@ -344,6 +344,13 @@ dbsp_status c5_gen_rows(hipStream_t s, int64_t n, uint64_t stride,
                         uint64_t jitter, uint64_t seed, int val_mode,
                         uint64_t *k, uint64_t *v, int64_t *w);
 
+// radix-tree rolling aggregate (§8f4): per input row (partition, ts, w),
+// the weight sum over the partition's rows with time in [ts-width, ts]
+dbsp_status rolling_agg_rows(hipStream_t s, const uint64_t *k,
+                             const uint64_t *v, const int64_t *w, int64_t n,
+                             uint64_t width, uint64_t *ok, uint64_t *ov,
+                             int64_t *ow);
+
 uint64_t host_xxh3_u64(uint64_t key, uint64_t seed);
 
 }  // namespace dbspk

@@ -112,6 +112,9 @@ def _L():
         L.dbsp_engine_output.argtypes = [vp, vp, i64, ctypes.POINTER(i64)]
         L.dbsp_engine_output_events.restype = i32
         L.dbsp_engine_output_events.argtypes = [vp, vp, i64, ctypes.POINTER(i64)]
+        L.dbsp_rolling_agg.restype = i32
+        L.dbsp_rolling_agg.argtypes = [vp, ctypes.POINTER(BatchStruct), u64,
+                                       ctypes.POINTER(BatchStruct)]
         L.dbsp_engine_c5_init.restype = i32
         L.dbsp_engine_c5_init.argtypes = [vp, i64, i64, u64]
         L.dbsp_engine_kernel_stats.restype = i32
@@ -348,6 +351,17 @@ class Ctx:
         for b in batches:
             self.free_batch(b)
         self.free_batch(out)
+        return res
+
+    def rolling_agg(self, rows, width):
+        inp = self.upload_rows(rows)
+        out = BatchStruct()
+        _check(self._lib.dbsp_rolling_agg(self._h, ctypes.byref(inp), width,
+                                          ctypes.byref(out)), "rolling")
+        self.sync()
+        res = self.download_rows(out)
+        for x in (inp, out):
+            self.free_batch(x)
         return res
 
     def shard_partition(self, rows, nshards):

@@ -33,6 +33,8 @@ def _o():
         L.oracle_window.argtypes = [vp, i64, vp, i64, ctypes.c_int, u64, u64, u64, u64, vp, i64]
         L.oracle_xxh3_u64.restype = u64
         L.oracle_xxh3_u64.argtypes = [u64, u64]
+        L.oracle_rolling_agg.restype = i64
+        L.oracle_rolling_agg.argtypes = [vp, i64, u64, vp]
         L.oracle_query_new.restype = vp
         L.oracle_query_new.argtypes = [ctypes.c_int]
         L.oracle_query_free.argtypes = [vp]
@@ -108,6 +110,13 @@ def window(trace, batch, have_prev, s0, e0, s1, e1, cap=None):
 
 def xxh3_u64(key, seed=0x7F95EF85BE33C337):
     return _o().oracle_xxh3_u64(key, seed)
+
+
+def rolling_agg(rows, width):
+    rows = np.ascontiguousarray(rows, dtype=ROW_DT)
+    out = np.empty(len(rows), dtype=ROW_DT)
+    n = _o().oracle_rolling_agg(_p(rows), len(rows), width, _p(out))
+    return out[:n].copy()
 
 
 class Query:

@@ -168,6 +168,14 @@ dbsp_status dbsp_join(dbsp_ctx *ctx, const dbsp_batch *delta,
  * Out: raw update rows: for each key with new aggregate s != 0 emit (key, s, +1);
  *      for each existing (key, v, w_sum != 0) in the output trace emit (key, v, -w_sum).
  *      Caller consolidates. */
+/* Radix-tree rolling aggregate (operator/time_series/radix_tree/mod.rs:1-75
+ * + rolling_aggregate.rs:235-280): per row (partition, ts, w)
+ * of a consolidated batch, the weight sum over the partition rows with time
+ * in [ts-width, ts] (RelRange::range_of, range.rs:93-110).  Implemented as a
+ * flat radix-16 prefix-aggregate tree over the sorted rows (O(log) query). */
+dbsp_status dbsp_rolling_agg(dbsp_ctx *ctx, const dbsp_batch *in,
+                             uint64_t width, dbsp_batch *out);
+
 dbsp_status dbsp_agg_linear_upsert(dbsp_ctx *ctx,
                                    const uint64_t *delta_keys, int64_t nd,
                                    const dbsp_batch *in_trace,

@@ -709,6 +709,22 @@ int64_t oracle_window(const dbsp_row *trace, int64_t nt, const dbsp_row *batch,
     return (int64_t)vo.size();
 }
 
+// naive rolling aggregate (the radix-tree operators' semantic oracle:
+// rolling_aggregate.rs:235-280 with the linear weight-sum aggregate;
+// range_of(ts) = [ts - width, ts], range.rs:93-110, saturating at 0)
+int64_t oracle_rolling_agg(const dbsp_row *in, int64_t n, uint64_t width,
+                           dbsp_row *out) {
+    for (int64_t i = 0; i < n; i++) {
+        const uint64_t p = in[i].k, ts = in[i].v;
+        const uint64_t t0 = ts >= width ? ts - width : 0;
+        int64_t s = 0;
+        for (int64_t j = 0; j < n; j++)
+            if (in[j].k == p && in[j].v >= t0 && in[j].v <= ts) s += in[j].w;
+        out[i] = {p, ts, s};
+    }
+    return n;
+}
+
 uint64_t oracle_xxh3_u64(uint64_t key, uint64_t seed) { return xxh3_u64(key, seed); }
 
 // Incremental distinct (operator/distinct.rs:404-462 at root scope, depth 1:

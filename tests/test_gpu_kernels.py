@@ -235,3 +235,27 @@ def test_distinct_reference_vectors(ctx):
         assert zset(integral) == zset(rows_of(
             [tuple(r) for r in t["integral"]])), f"tick {tick}"
         trace = oracle.merge(trace, delta)
+
+
+def test_rolling_agg_parity(ctx):
+    """Radix-tree rolling aggregate (radix_tree/mod.rs:1-75,
+    rolling_aggregate.rs:235-280): per-row [ts-width, ts] partition-local
+    weight sums over the flat radix-16 prefix-aggregate tree, bit-exact vs
+    the naive oracle scan.  Covers width 0 (self only), narrow, wide and
+    whole-history windows, single-partition and many-partition batches, and
+    sizes across the tree's level boundaries (16^k edges)."""
+    rng = np.random.default_rng(77)
+    for n, nparts, trange in [(1, 1, 10), (255, 4, 100), (256, 4, 100),
+                              (257, 4, 100), (4096, 1, 10_000),
+                              (65_537, 64, 100_000), (200_000, 500, 50_000)]:
+        rows = np.empty(n, dtype=ROW_DT)
+        rows["k"] = rng.integers(0, nparts, n)
+        rows["v"] = rng.integers(0, trange, n)
+        rows["w"] = rng.integers(-5, 6, n)
+        rows = oracle.consolidate(rows)
+        if len(rows) == 0:
+            continue
+        for width in [0, 3, trange // 7 + 1, 1 << 62]:
+            got = ctx.rolling_agg(rows, width)
+            exp = oracle.rolling_agg(rows, width)
+            assert np.array_equal(got, exp), (n, nparts, width)
