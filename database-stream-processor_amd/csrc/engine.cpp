@@ -75,8 +75,10 @@ struct dbsp_ctx {
     bool force_shard = false;  // exercise the full partition+alltoallv path
                                // even at world=1 (self-exchange; test hook)
     // persistent length scratch (device + pinned host)
-    int64_t *d_len = nullptr;   // 20 device length slots (8-15: chained
-                                // ticks; 16-17: framed-exchange totals)
+    int64_t *d_len = nullptr;   // 44 device length slots: two 18-slot tick
+                                // bases (0 and 18: chained-tick lengths,
+                                // sb+16/17 = framed-exchange totals) +
+                                // 36-39 spine-insert rounds
     int64_t *h_len = nullptr;
     int timer_depth = 0;  // ScopedTimer nesting guard (shared event pair)
     // per-tick transient bump arena (reset at each engine tick; falls back to
@@ -90,6 +92,7 @@ struct dbsp_ctx {
     size_t arena_half = 0;  // half size (arena_sz / 2); 0 = no split
     hipEvent_t ev_sync = nullptr;   // tick-end event (pre-front-launch point)
     hipEvent_t ev_sync2 = nullptr;  // early-wake event (post-count readback)
+    hipEvent_t ev_tick[2] = {nullptr, nullptr};  // pipelined-train events
 };
 
 static void *arena_alloc(dbsp_ctx *c, size_t bytes) {
@@ -138,18 +141,20 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     HIP_CHECK_ST(hipEventCreate(&c->ev1));
     HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_sync, hipEventDisableTiming));
     HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_sync2, hipEventDisableTiming));
+    HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_tick[0], hipEventDisableTiming));
+    HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_tick[1], hipEventDisableTiming));
     const char *p = getenv("DBSP_PROFILE");
     c->profile = p && p[0] == '1';
     const char *fs = getenv("DBSP_FORCE_SHARD");
     c->force_shard = fs && fs[0] == '1';
-    HIP_CHECK_ST(hipMalloc(&c->d_len, 20 * sizeof(int64_t)));
+    HIP_CHECK_ST(hipMalloc(&c->d_len, 44 * sizeof(int64_t)));
     c->arena_sz = (size_t)512 << 20;
     c->arena_half = c->arena_sz / 2;
     if (hipMalloc(&c->arena, c->arena_sz) != hipSuccess) {
         c->arena = nullptr;
         c->arena_sz = 0;
     }
-    HIP_CHECK_ST(hipHostMalloc(&c->h_len, 20 * sizeof(int64_t)));
+    HIP_CHECK_ST(hipHostMalloc(&c->h_len, 44 * sizeof(int64_t)));
     *out = c;
     return DBSP_OK;
 }
@@ -158,6 +163,8 @@ extern "C" dbsp_status dbsp_ctx_destroy(dbsp_ctx *c) {
     if (!c) return DBSP_OK;
     if (c->ev_sync) (void)hipEventDestroy(c->ev_sync);
     if (c->ev_sync2) (void)hipEventDestroy(c->ev_sync2);
+    if (c->ev_tick[0]) (void)hipEventDestroy(c->ev_tick[0]);
+    if (c->ev_tick[1]) (void)hipEventDestroy(c->ev_tick[1]);
     if (c->comm) ncclCommDestroy(c->comm);
     (void)hipStreamSynchronize(c->stream);
     if (c->d_len) (void)hipFree(c->d_len);
@@ -327,9 +334,11 @@ static dbsp_status sort_medium(dbsp_ctx *c, DevBatch raw, DevBatch &out) {
             results.push_back(res);
         }
         if (ma.np > 0) {
-            ma.d_len = c->d_len;
+            // dedicated insert slots (36-39): a pipelined next-tick train
+            // owns one of the two 18-slot tick bases at this point
+            ma.d_len = c->d_len + 36;
             TRY(dbspk::merge_small_batch(c->stream, ma));
-            HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 36, c->d_len + 36,
                                         ma.np * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
             HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -1017,6 +1026,29 @@ static dbsp_status shard_exchange(dbsp_ctx *c, DevBatch local, DevBatch &out,
 // engine: per-query operator DAGs (mirrors nexmark/src/queries/q{0,3,5,8}.rs)
 // ---------------------------------------------------------------------------
 
+struct Q3Plan {
+    int which;  // 0 = delta A (auctions), 1 = delta P (persons)
+    TraceArgs t;
+    int proj;
+    uint32_t *cnts;
+    uint64_t *offsets;
+    bool small;
+    bool dd;  // delta-vs-delta plan (trace length is tick-fresh)
+    int slot;
+};
+
+struct Q3Train {
+    bool pending = false;
+    const dbsp_event *ev = nullptr;
+    int64_t n = -1;
+    int sb = 0, evi = 0;
+    int next_sb = 18, next_evi = 0;
+    DevBatch rawA, rawP, oA, oP, comb_chain;
+    Q3Plan plans[3];
+    int np = 0;
+    size_t arena_base = 0, arena_off = 0;
+};
+
 struct dbsp_engine {
     dbsp_ctx *ctx = nullptr;
     int query = 0;
@@ -1069,10 +1101,10 @@ struct dbsp_engine {
     // path allocates nothing for it (output_is_store marks when output
     // aliases it and must not be freed)
     DevBatch output;
-    // pipelined front half of the NEXT tick (launched during this tick's
-    // tail via the spines_insert_pair hook, consumed by the next step; the
-    // (ev, n) pair is verified at consumption so an out-of-band step never
-    // uses a stale front)
+    // pipelined front half of the NEXT tick (q5/q8: launched during this
+    // tick's tail via the spines_insert_pair hook, consumed by the next
+    // step; the (ev, n) pair is verified at consumption so an out-of-band
+    // step never uses a stale front)
     struct {
         bool pending = false;
         const dbsp_event *ev = nullptr;
@@ -1080,6 +1112,9 @@ struct dbsp_engine {
         DevBatch rawA, rawB, oA, oB;
         size_t arena_base = 0, arena_off = 0;
     } front;
+    // q3's pipelined full TRAIN (see the q3 tick comment): the whole next
+    // tick's launch train enqueued at the end of the current step
+    Q3Train train;
     const dbsp_event *next_ev = nullptr;  // set by dbsp_engine_run_staged
     int64_t next_n = -1;
 
@@ -1140,6 +1175,11 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
     if (e->front.pending) {
         free_batch(c, e->front.oA);
         free_batch(c, e->front.oB);
+    }
+    if (e->train.pending) {
+        (void)hipStreamSynchronize(c->stream);
+        free_batch(c, e->train.oA);
+        free_batch(c, e->train.oP);
     }
     if (e->out_store.k) free_batch(c, e->out_store);
     if (e->d_events) (void)dbspk::cache_free(e->d_events, c->stream);
@@ -1411,7 +1451,7 @@ static dbsp_status shard_exchange_pair(dbsp_ctx *c, DevBatch l0, DevBatch l1,
 // flatmap events slice into up to two raw streams, then sort+consolidate
 static dbsp_status build_deltas_chain(dbsp_engine *e, const dbsp_event *d_ev,
                                       int64_t n, DevBatch &rawA, DevBatch &rawB,
-                                      DevBatch &oA, DevBatch &oB);
+                                      DevBatch &oA, DevBatch &oB, int sb = 0);
 
 // If d_ev points into the staged event array, return its column-split view
 // (shifted to the same offset) so the flatmap reads coalesced SoA streams.
@@ -1513,7 +1553,7 @@ static dbsp_status build_deltas(dbsp_engine *e, const dbsp_event *d_ev,
 // re-sorts through the sized paths if the speculation lost.
 static dbsp_status build_deltas_chain(dbsp_engine *e, const dbsp_event *d_ev,
                                       int64_t n, DevBatch &rawA, DevBatch &rawB,
-                                      DevBatch &oA, DevBatch &oB) {
+                                      DevBatch &oA, DevBatch &oB, int sb) {
     dbsp_ctx *c = e->ctx;
     int64_t cap = n > 0 ? n : 1;
     TRY(alloc_batch(c, cap, rawA, true));
@@ -1523,7 +1563,7 @@ static dbsp_status build_deltas_chain(dbsp_engine *e, const dbsp_event *d_ev,
     TRY(dbspk::flatmap_events_chain(c->stream, d_ev, hc ? &cols : nullptr, n,
                                     e->query, rawA.k, rawA.v, rawA.w, rawB.k,
                                     rawB.v, rawB.w,
-                                    (uint64_t *)(c->d_len + 8)));
+                                    (uint64_t *)(c->d_len + sb + 8)));
     DevBatch sA, sB;
     TRY(alloc_batch(c, cap, sA, true));
     TRY(alloc_batch(c, cap, sB, true));
@@ -1533,18 +1573,18 @@ static dbsp_status build_deltas_chain(dbsp_engine *e, const dbsp_event *d_ev,
     sa.nb = 2;
     sa.kin[0] = rawA.k; sa.vin[0] = rawA.v; sa.win[0] = rawA.w;
     sa.kin[1] = rawB.k; sa.vin[1] = rawB.v; sa.win[1] = rawB.w;
-    sa.n_dev[0] = c->d_len + 8;
-    sa.n_dev[1] = c->d_len + 9;
+    sa.n_dev[0] = c->d_len + sb + 8;
+    sa.n_dev[1] = c->d_len + sb + 9;
     sa.tk[0] = sA.k; sa.tv[0] = sA.v; sa.tw[0] = sA.w;
     sa.tk[1] = sB.k; sa.tv[1] = sB.v; sa.tw[1] = sB.w;
     sa.ok[0] = oA.k; sa.ov[0] = oA.v; sa.ow[0] = oA.w;
     sa.ok[1] = oB.k; sa.ov[1] = oB.v; sa.ow[1] = oB.w;
-    sa.d_len = c->d_len + 10;
+    sa.d_len = c->d_len + sb + 10;
     {
         ScopedTimer t(c, 0, (double)n * 48.0);
         TRY(dbspk::sort_cons_small_batch(c->stream, sa));
     }
-    oA.n = -1;  // pending: d_len[10], d_len[11]
+    oA.n = -1;  // pending: d_len[sb+10], d_len[sb+11]
     oB.n = -1;
     return DBSP_OK;
 }
@@ -1996,9 +2036,11 @@ static dbsp_status spines_insert_multi(dbsp_ctx *c, Spine *const *sps,
             }
         }
         if (ma.np > 0) {
-            ma.d_len = c->d_len;
+            // dedicated insert slots (36-39): a pipelined next-tick train
+            // owns one of the two 18-slot tick bases at this point
+            ma.d_len = c->d_len + 36;
             TRY(dbspk::merge_small_batch(c->stream, ma));
-            HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 36, c->d_len + 36,
                                         ma.np * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
             if (hook && !hook_fired) {
@@ -2011,7 +2053,7 @@ static dbsp_status spines_insert_multi(dbsp_ctx *c, Spine *const *sps,
             }
             for (int j = 0; j < nbatched; j++) {
                 Spine &sp = *owners[j];
-                results[j].n = c->h_len[j];
+                results[j].n = c->h_len[36 + j];
                 DevBatch top = sp.batches.back();
                 sp.batches.pop_back();
                 DevBatch below = sp.batches.back();
@@ -2075,55 +2117,199 @@ static TraceArgs trace_args_of(Spine &sp) {
 // from the second readback and replayed explicitly; a lost sort speculation
 // re-sorts through the sized paths.  Sharded ranks keep the explicit path
 // (the exchange needs host lengths).
-static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
+// ---- q3 tick (queries/q3.rs:35-63) ----
+//
+// Pipelined TRAINS (round 2): for run_staged loops the ENTIRE next tick's
+// launch train (flatmap -> speculative fused sorts -> 3-plan join count ->
+// emit bases -> chained emits -> output consolidate, plus both readbacks and
+// an event record) is enqueued at the END of the current step, right after
+// the spine inserts commit — at that point the post-insert spine state the
+// next tick's probes must read is final, so the train's TraceArgs are
+// correct by construction.  The next step then only WAITS on the train's
+// event and runs verdicts + inserts: the ~90 us of per-tick host enqueue
+// work (the round-1 trace showed 53 us gaps before the join probes and
+// 63 us before the insert merges — fat-kernarg hipLaunchKernel time, not
+// sync latency) overlaps the previous train's GPU execution.  Trains
+// ping-pong the 18-slot length bases (0 / 18), the tick events and the
+// arena halves; the spine-insert rounds use dedicated slots 36-39.
+// Per-tick stepping (no next_ev), sharded ranks, oversized ticks and lost
+// speculations all run the unpipelined body below, which is the round-1
+// path unchanged.
+
+static constexpr int64_t Q3_EMIT_CAP = 32768;
+
+// Build the three bilinear join plans against the CURRENT spines (callers
+// guarantee the spine state is the one tick t's probes must read) and
+// enqueue count(+scan).  spec: delta lengths still on device at d_len[sb+10/11].
+static dbsp_status q3_plan_count(dbsp_engine *e, DevBatch &dA, DevBatch &dP,
+                                 bool spec, int sb, int64_t n_cap,
+                                 Q3Plan plans[3], int &np, int &jca_np,
+                                 bool &arena_ok) {
     dbsp_ctx *c = e->ctx;
-    bool use_front = e->front.pending && e->front.ev == d_ev &&
-                     e->front.n == n;
-    if (e->front.pending && !use_front) {  // stale pipelined front: discard
-        free_batch(c, e->front.oA);
-        free_batch(c, e->front.oB);
-        e->front.pending = false;
+    np = 0;
+    arena_ok = true;
+    const bool haveA = spec || dA.n > 0, haveP = spec || dP.n > 0;
+    if (haveA && !e->p_int.batches.empty())
+        plans[np++] = {0, trace_args_of(e->p_int), DBSP_PROJ_HI_V2_LO_V1,
+                       nullptr, nullptr, false, false, -1};
+    if (haveP && !e->a_int.batches.empty())
+        plans[np++] = {1, trace_args_of(e->a_int), DBSP_PROJ_HI_V1_LO_V2,
+                       nullptr, nullptr, false, false, -1};
+    if (haveA && haveP) {
+        TraceArgs t{};
+        t.nb = 1;
+        t.k[0] = dP.k; t.v[0] = dP.v; t.w[0] = dP.w;
+        t.n[0] = spec ? 0 : dP.n;
+        plans[np++] = {0, t, DBSP_PROJ_HI_V2_LO_V1, nullptr, nullptr, false,
+                       true, -1};
     }
-    const bool shard_chain =
-        sharding_on(c) && c->world <= 8 && n <= 131072 && e->spec_fail < 3;
-    const bool chain =
-        use_front || shard_chain ||
-        (!sharding_on(c) && n <= 131072 && e->spec_fail < 3);
-    DevBatch dA, dP, rawA, rawP, recvA, recvP;
-    if (use_front) {
-        e->front.pending = false;
-        rawA = e->front.rawA;
-        rawP = e->front.rawB;
-        dA = e->front.oA;
-        dP = e->front.oB;
-    } else if (shard_chain) {
-        TRY(build_deltas_chain_sharded(e, d_ev, n, rawA, rawP, recvA, recvP,
-                                       dA, dP));
-    } else if (chain) {
-        TRY(build_deltas_chain(e, d_ev, n, rawA, rawP, dA, dP));
-    } else {
-        TRY(build_deltas(e, d_ev, n, dA, dP, true));
+    JoinCountArgs jca{};
+    for (int i = 0; i < np; i++) {
+        Q3Plan &pl = plans[i];
+        pl.slot = -1;
+        if (pl.t.nb == 0) continue;
+        DevBatch &d = pl.which == 0 ? dA : dP;
+        const int64_t nd_cap = spec ? n_cap : d.n;
+        pl.cnts = (uint32_t *)arena_alloc(c, (size_t)nd_cap * pl.t.nb * 4 + 8);
+        pl.offsets = (uint64_t *)arena_alloc(c, (size_t)(nd_cap + 1) * 8);
+        const bool bufs = pl.cnts && pl.offsets;
+        if (spec && !bufs) {
+            arena_ok = false;
+            break;
+        }
+        if (bufs && (spec || nd_cap <= 8192)) {
+            pl.small = true;
+            pl.slot = jca.np;
+            jca.dk[jca.np] = d.k;
+            jca.nd[jca.np] = spec ? 0 : d.n;
+            if (spec)
+                jca.nd_dev[jca.np] = c->d_len + sb + (pl.which == 0 ? 10 : 11);
+            if (spec && pl.dd) jca.tn_dev[jca.np] = c->d_len + sb + 11;
+            jca.t[jca.np] = pl.t;
+            jca.cnts[jca.np] = pl.cnts;
+            jca.offsets[jca.np] = pl.offsets;
+            jca.np++;
+        }
     }
-    // Bilinear expansion against PREVIOUS traces only, so all three joins are
-    // independent of the spine inserts:
-    //   out = dA join P_prev + A_prev join dP + dA join dP
-    // — identical to the reference's dA join trace(P) + trace(A) delayed join dP
-    // (operator/join.rs:217-292) by bilinearity.
+    jca_np = jca.np;
+    if (arena_ok) {
+        jca.d_total = c->d_len + sb;
+        if (jca.np > 0) TRY(dbspk::join_count_scan_batch(c->stream, jca));
+    }
+    return DBSP_OK;
+}
+
+// chained emits + output consolidate behind the device-side totals; the
+// verdicts (emit total / overflow flag / output length) land at
+// h_len[sb+12..15] via the second readback
+static dbsp_status q3_chain_emits(dbsp_engine *e, DevBatch &dA, DevBatch &dP,
+                                  Q3Plan plans[3], int np, int sb,
+                                  DevBatch &comb_chain, bool &ok) {
+    dbsp_ctx *c = e->ctx;
+    ok = false;
+    int jca_np = 0;
+    for (int i = 0; i < np; i++)
+        if (plans[i].slot >= 0) jca_np = std::max(jca_np, plans[i].slot + 1);
+    int64_t *bases = (int64_t *)arena_alloc(c, (jca_np + 1) * 8 + 8);
+    DevBatch scr;
+    if (!bases || alloc_batch(c, Q3_EMIT_CAP, comb_chain, true) != DBSP_OK ||
+        alloc_batch(c, Q3_EMIT_CAP, scr, true) != DBSP_OK)
+        return DBSP_OK;  // arena exhausted: caller keeps the explicit path
+    TRY(dbspk::emit_bases(c->stream, c->d_len + sb, jca_np, Q3_EMIT_CAP,
+                          bases, c->d_len + sb + 3, c->d_len + sb + 4));
+    for (int i = 0; i < np; i++) {
+        Q3Plan &pl = plans[i];
+        if (pl.slot < 0) continue;
+        DevBatch &d = pl.which == 0 ? dA : dP;
+        TRY(dbspk::join_emit_chain(
+            c->stream, d.k, d.v, d.w,
+            c->d_len + sb + (pl.which == 0 ? 10 : 11), pl.t,
+            pl.dd ? c->d_len + sb + 11 : nullptr, pl.cnts, pl.offsets,
+            c->d_len + sb + pl.slot, bases + pl.slot, c->d_len + sb + 4,
+            Q3_EMIT_CAP, pl.proj, 0, comb_chain.k, comb_chain.v,
+            comb_chain.w));
+    }
+    if (!e->out_store.k) {
+        e->out_cap = 8192;
+        TRY(alloc_batch(c, e->out_cap, e->out_store));
+    }
+    SortArgs sa{};
+    sa.nb = 1;
+    sa.kin[0] = comb_chain.k;
+    sa.vin[0] = comb_chain.v;
+    sa.win[0] = comb_chain.w;
+    sa.n_dev[0] = c->d_len + sb + 3;
+    sa.tk[0] = scr.k; sa.tv[0] = scr.v; sa.tw[0] = scr.w;
+    sa.ok[0] = e->out_store.k;
+    sa.ov[0] = e->out_store.v;
+    sa.ow[0] = e->out_store.w;
+    sa.d_len = c->d_len + sb + 6;
+    TRY(dbspk::sort_cons_small_batch(c->stream, sa));
+    HIP_CHECK_ST(hipMemcpyAsync(c->h_len + sb + 12, c->d_len + sb + 3,
+                                4 * sizeof(int64_t), hipMemcpyDeviceToHost,
+                                c->stream));
+    ok = true;
+    return DBSP_OK;
+}
+
+// explicit emits from prepared counts/offsets (the non-chained path and the
+// emit-overflow replay)
+static dbsp_status q3_emit_explicit(dbsp_engine *e, DevBatch &dA, DevBatch &dP,
+                                    Q3Plan plans[3], int np,
+                                    const int64_t slot_totals[3],
+                                    std::vector<DevBatch> &outs) {
+    dbsp_ctx *c = e->ctx;
+    ScopedTimer timer(c, 2, 0.0);
+    int64_t total_small = 0;
+    for (int i = 0; i < np; i++)
+        if (plans[i].t.nb > 0 && plans[i].small)
+            total_small += slot_totals[plans[i].slot];
+    DevBatch comb;
+    if (total_small > 0) TRY(alloc_batch(c, total_small, comb, true));
+    int64_t base = 0;
+    for (int i = 0; i < np; i++) {
+        Q3Plan &pl = plans[i];
+        if (pl.t.nb == 0) continue;
+        DevBatch &d = pl.which == 0 ? dA : dP;
+        if (pl.small) {
+            int64_t total = slot_totals[pl.slot];
+            if (total <= 0) continue;
+            TRY(dbspk::join_emit_prepared(c->stream, d.k, d.v, d.w, d.n, pl.t,
+                                          pl.cnts, pl.offsets, total, pl.proj,
+                                          0, comb.k + base, comb.v + base,
+                                          comb.w + base));
+            base += total;
+        } else {
+            DevBatch o;
+            TRY(dbspk::join_spine_rows(c->stream, d.k, d.v, d.w, d.n, pl.t,
+                                       pl.proj, 0, &o.k, &o.v, &o.w, &o.n));
+            if (o.n > 0) outs.push_back(o);
+            else free_batch(c, o);
+        }
+    }
+    if (total_small > 0) {
+        comb.n = total_small;
+        outs.push_back(comb);
+    } else if (comb.k) {
+        free_batch(c, comb);
+    }
+    return DBSP_OK;
+}
+
+static void q3_maybe_enqueue_train(dbsp_engine *e);
+
+// the unpipelined tick body (round-1 path): deltas are already built;
+// plans/count/emits run in-step at slot base 0
+static dbsp_status q3_body(dbsp_engine *e, const dbsp_event *d_ev, int64_t n,
+                           DevBatch dA, DevBatch dP, DevBatch rawA,
+                           DevBatch rawP, DevBatch recvA, DevBatch recvP,
+                           bool chain, bool shard_chain) {
+    dbsp_ctx *c = e->ctx;
     std::vector<DevBatch> outs;
-    bool chain_emits = false;       // emits + output sort queued pre-event
-    DevBatch comb_chain{};          // capacity emit buffer (arena)
+    bool chain_emits = false;
+    DevBatch comb_chain{};
     int64_t slot_totals[3] = {0, 0, 0};
-    constexpr int64_t EMIT_CAP = 32768;
-    struct Plan {
-        const DevBatch *delta;
-        TraceArgs t;
-        int proj;
-        uint32_t *cnts;
-        uint64_t *offsets;
-        bool small;
-        bool dd;  // delta-vs-delta plan (trace length is tick-fresh)
-        int slot;
-    } plans[3];
+    Q3Plan plans[3];
     int np = 0;
     {
         ScopedTimer timer(c, 2, (double)n * 24.0);
@@ -2131,122 +2317,25 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             TRY(e->p_int.consolidate_all(c));
         if ((int)e->a_int.batches.size() > MAX_TRACE_BATCHES)
             TRY(e->a_int.consolidate_all(c));
-        bool spec = chain;  // delta lengths still on the device
+        bool spec = chain;
         for (;;) {
-            np = 0;
-            const bool haveA = spec || dA.n > 0, haveP = spec || dP.n > 0;
-            if (haveA && !e->p_int.batches.empty())
-                plans[np++] = {&dA, trace_args_of(e->p_int),
-                               DBSP_PROJ_HI_V2_LO_V1, nullptr, nullptr,
-                               false, false, -1};
-            if (haveP && !e->a_int.batches.empty())
-                plans[np++] = {&dP, trace_args_of(e->a_int),
-                               DBSP_PROJ_HI_V1_LO_V2, nullptr, nullptr,
-                               false, false, -1};
-            if (haveA && haveP) {
-                TraceArgs t{};
-                t.nb = 1;
-                t.k[0] = dP.k; t.v[0] = dP.v; t.w[0] = dP.w;
-                t.n[0] = spec ? 0 : dP.n;
-                plans[np++] = {&dA, t, DBSP_PROJ_HI_V2_LO_V1, nullptr, nullptr,
-                               false, true, -1};
-            }
-            JoinCountArgs jca{};
+            int jca_np = 0;
             bool arena_ok = true;
-            for (int i = 0; i < np; i++) {
-                Plan &pl = plans[i];
-                pl.slot = -1;
-                if (pl.t.nb == 0) continue;
-                const int64_t nd_cap = spec ? n : pl.delta->n;
-                pl.cnts =
-                    (uint32_t *)arena_alloc(c, (size_t)nd_cap * pl.t.nb * 4 + 8);
-                pl.offsets = (uint64_t *)arena_alloc(c, (size_t)(nd_cap + 1) * 8);
-                const bool bufs = pl.cnts && pl.offsets;
-                if (spec && !bufs) {
-                    arena_ok = false;
-                    break;
-                }
-                if (bufs && (spec || nd_cap <= 8192)) {
-                    pl.small = true;
-                    pl.slot = jca.np;
-                    jca.dk[jca.np] = pl.delta->k;
-                    jca.nd[jca.np] = spec ? 0 : pl.delta->n;
-                    if (spec)
-                        jca.nd_dev[jca.np] =
-                            c->d_len + (pl.delta == &dA ? 10 : 11);
-                    if (spec && pl.dd) jca.tn_dev[jca.np] = c->d_len + 11;
-                    jca.t[jca.np] = pl.t;
-                    jca.cnts[jca.np] = pl.cnts;
-                    jca.offsets[jca.np] = pl.offsets;
-                    jca.np++;
-                }
-            }
+            TRY(q3_plan_count(e, dA, dP, spec, 0, n, plans, np, jca_np,
+                              arena_ok));
             if (spec && !arena_ok) {
-                // arena exhausted mid-speculation: fetch the real lengths and
-                // rebuild the plans the explicit way
                 HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
                                             18 * sizeof(int64_t),
                                             hipMemcpyDeviceToHost, c->stream));
                 HIP_CHECK_ST(hipStreamSynchronize(c->stream));
             } else {
-                jca.d_total = c->d_len;
-                if (jca.np > 0) TRY(dbspk::join_count_scan_batch(c->stream, jca));
                 HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
                                             18 * sizeof(int64_t),
                                             hipMemcpyDeviceToHost, c->stream));
-                if (spec && jca.np > 0) {
+                if (spec && jca_np > 0) {
                     (void)hipEventRecord(c->ev_sync2, c->stream);
-                    // chain the emits + output consolidate behind the
-                    // device-side totals (the host wakes at the event above,
-                    // before these run); the second readback lands their
-                    // verdicts at h_len[8..15): totals, combined total,
-                    // overflow flag, output length
-                    int64_t *bases =
-                        (int64_t *)arena_alloc(c, (jca.np + 1) * 8 + 8);
-                    DevBatch scr;
-                    if (bases &&
-                        alloc_batch(c, EMIT_CAP, comb_chain, true) == DBSP_OK &&
-                        alloc_batch(c, EMIT_CAP, scr, true) == DBSP_OK) {
-                        TRY(dbspk::emit_bases(c->stream, c->d_len, jca.np,
-                                              EMIT_CAP, bases, c->d_len + 3,
-                                              c->d_len + 4));
-                        for (int i = 0; i < np; i++) {
-                            Plan &pl = plans[i];
-                            if (pl.slot < 0) continue;
-                            TRY(dbspk::join_emit_chain(
-                                c->stream, pl.delta->k, pl.delta->v,
-                                pl.delta->w, jca.nd_dev[pl.slot], pl.t,
-                                jca.tn_dev[pl.slot], pl.cnts, pl.offsets,
-                                c->d_len + pl.slot, bases + pl.slot,
-                                c->d_len + 4, EMIT_CAP, pl.proj, 0,
-                                comb_chain.k, comb_chain.v, comb_chain.w));
-                        }
-                        if (!e->out_store.k) {
-                            e->out_cap = 8192;
-                            TRY(alloc_batch(c, e->out_cap, e->out_store));
-                        }
-                        SortArgs sa{};
-                        sa.nb = 1;
-                        sa.kin[0] = comb_chain.k;
-                        sa.vin[0] = comb_chain.v;
-                        sa.win[0] = comb_chain.w;
-                        sa.n_dev[0] = c->d_len + 3;
-                        sa.tk[0] = scr.k; sa.tv[0] = scr.v; sa.tw[0] = scr.w;
-                        sa.ok[0] = e->out_store.k;
-                        sa.ov[0] = e->out_store.v;
-                        sa.ow[0] = e->out_store.w;
-                        sa.d_len = c->d_len + 6;
-                        TRY(dbspk::sort_cons_small_batch(c->stream, sa));
-                        chain_emits = true;
-                    }
-                    // second readback: emit total / overflow flag / output
-                    // length land at h_len[12..15] (clear of the flatmap
-                    // counts and sort lengths the lost-speculation path
-                    // reads from the first copy)
-                    HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 12, c->d_len + 3,
-                                                4 * sizeof(int64_t),
-                                                hipMemcpyDeviceToHost,
-                                                c->stream));
+                    TRY(q3_chain_emits(e, dA, dP, plans, np, 0, comb_chain,
+                                       chain_emits));
                     (void)hipEventSynchronize(c->ev_sync2);
                 } else {
                     HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -2269,9 +2358,6 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
             free_batch(c, dA);
             free_batch(c, dP);
             if (shard_chain) {
-                // sharded recovery: frame overflow (totals -1, all ranks see
-                // it here) replays the dynamic exchange from the raw flatmap
-                // outputs; a plain sort overflow re-sorts the RECEIVED rows
                 if (c->h_len[16] < 0 || c->h_len[17] < 0) {
                     rawA.n = c->h_len[8];
                     rawP.n = c->h_len[9];
@@ -2289,8 +2375,6 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                 spec = false;
                 continue;
             }
-            // speculation lost: re-sort from the raw flatmap outputs with the
-            // now-known lengths, then redo the planning explicitly
             rawA.n = c->h_len[8];
             rawP.n = c->h_len[9];
             TRY(sort_consolidate_batch(c, rawA, dA));
@@ -2302,75 +2386,9 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         for (int i = 0; i < np; i++)
             if (plans[i].dd) plans[i].t.n[0] = dP.n;
     }
-    auto emit_explicit = [&](void) -> dbsp_status {
-        ScopedTimer timer(c, 2, 0.0);
-        int64_t total_small = 0;
-        for (int i = 0; i < np; i++)
-            if (plans[i].t.nb > 0 && plans[i].small)
-                total_small += slot_totals[plans[i].slot];
-        DevBatch comb;
-        if (total_small > 0) TRY(alloc_batch(c, total_small, comb, true));
-        int64_t base = 0;
-        for (int i = 0; i < np; i++) {
-            Plan &pl = plans[i];
-            if (pl.t.nb == 0) continue;
-            if (pl.small) {
-                int64_t total = slot_totals[pl.slot];
-                if (total <= 0) continue;
-                TRY(dbspk::join_emit_prepared(c->stream, pl.delta->k,
-                                              pl.delta->v, pl.delta->w,
-                                              pl.delta->n, pl.t, pl.cnts,
-                                              pl.offsets, total, pl.proj, 0,
-                                              comb.k + base, comb.v + base,
-                                              comb.w + base));
-                base += total;
-            } else {
-                DevBatch o;
-                TRY(dbspk::join_spine_rows(c->stream, pl.delta->k, pl.delta->v,
-                                           pl.delta->w, pl.delta->n, pl.t,
-                                           pl.proj, 0, &o.k, &o.v, &o.w, &o.n));
-                if (o.n > 0) outs.push_back(o);
-                else free_batch(c, o);
-            }
-        }
-        if (total_small > 0) {
-            comb.n = total_small;
-            outs.push_back(comb);
-        } else if (comb.k) {
-            free_batch(c, comb);
-        }
-        return DBSP_OK;
-    };
-    std::function<dbsp_status()> hook = [&]() -> dbsp_status {
-        // sharded ranks: no pipelined fronts (the front would enqueue an
-        // alltoall whose cross-rank ordering vs this tick's tail collectives
-        // is not guaranteed rank-uniform)
-        if (sharding_on(c) || !e->next_ev || e->next_n < 0 ||
-            e->next_n > 131072 || e->spec_fail >= 3)
-            return DBSP_OK;
-        const size_t save_base = c->arena_base, save_off = c->arena_off;
-        c->arena_base =
-            c->arena_half ? (save_base ? 0 : c->arena_half) : save_base;
-        c->arena_off = 0;
-        dbsp_status st = build_deltas_chain(e, e->next_ev, e->next_n,
-                                            e->front.rawA, e->front.rawB,
-                                            e->front.oA, e->front.oB);
-        e->front.arena_base = c->arena_base;
-        e->front.arena_off = c->arena_off;
-        c->arena_base = save_base;
-        c->arena_off = save_off;
-        if (st == DBSP_OK) {
-            e->front.pending = true;
-            e->front.ev = e->next_ev;
-            e->front.n = e->next_n;
-        }
-        return st;
-    };
     engine_free_output(e);
     if (chain_emits) {
-        // inserts first — they overlap the still-running emits/output sort —
-        // then read the emit verdicts from the second readback
-        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
+        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
         const int64_t flag = c->h_len[13];
         if (flag == 0) {
             const int64_t out_n = c->h_len[15];
@@ -2379,21 +2397,18 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
                 e->output.n = out_n;
                 e->output_is_store = true;
             } else {
-                // output larger than the fused sort: the capacity buffer
-                // holds the raw rows
                 comb_chain.n = c->h_len[12];
                 TRY(sort_consolidate_batch(c, comb_chain, e->output));
             }
         } else {
-            // combined emit output exceeded the capacity buffer: replay the
-            // emits explicitly (counts/offsets are still valid)
-            TRY(emit_explicit());
+            TRY(q3_emit_explicit(e, dA, dP, plans, np, slot_totals, outs));
             TRY(finalize_raw(c, outs, e->output));
         }
+        q3_maybe_enqueue_train(e);
         return DBSP_OK;
     }
     // explicit path (sharded ranks and lost speculations)
-    TRY(emit_explicit());
+    TRY(q3_emit_explicit(e, dA, dP, plans, np, slot_totals, outs));
     int64_t cat_n = 0;
     for (auto &b : outs) cat_n += b.n;
     bool async_final = cat_n > 0 && cat_n <= 8192;
@@ -2402,7 +2417,7 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         ScopedTimer t0(c, 0, (double)cat_n * 48.0);
         DevBatch cat, scratch;
         if (outs.size() == 1) {
-            cat = outs[0];  // the usual case: one combined emit buffer
+            cat = outs[0];
             outs.clear();
         } else {
             TRY(alloc_batch(c, cat_n, cat, true));
@@ -2434,36 +2449,193 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         sa.ok[0] = res.k; sa.ov[0] = res.v; sa.ow[0] = res.w;
         sa.d_len = c->d_len + 6;
         TRY(dbspk::sort_cons_small_batch(c->stream, sa));
-        // no sync yet — the spine inserts below sync the stream
     }
-    if (!sharding_on(c)) {
-        if (async_final)
-            HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6,
-                                        sizeof(int64_t),
-                                        hipMemcpyDeviceToHost, c->stream));
-        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
-        if (async_final) {
-            res.n = c->h_len[6];
-            e->output = res;
+    if (async_final)
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6,
+                                    sizeof(int64_t), hipMemcpyDeviceToHost,
+                                    c->stream));
+    TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
+    if (async_final) {
+        if (sharding_on(c)) HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        res.n = c->h_len[6];
+        e->output = res;
+        e->output_is_store = true;
+    } else {
+        TRY(finalize_raw(c, outs, e->output));
+    }
+    q3_maybe_enqueue_train(e);
+    return DBSP_OK;
+}
+
+// enqueue the NEXT tick's full train (called after this tick's inserts have
+// committed, so the spine state its probes bake in is final)
+static dbsp_status q3_enqueue_train(dbsp_engine *e, const dbsp_event *d_ev,
+                                    int64_t n) {
+    dbsp_ctx *c = e->ctx;
+    Q3Train &T = e->train;
+    const int sb = T.next_sb;
+    const int evi = T.next_evi;
+    const size_t save_base = c->arena_base, save_off = c->arena_off;
+    c->arena_base = c->arena_half ? (save_base ? 0 : c->arena_half) : save_base;
+    c->arena_off = 0;
+    auto bail = [&](void) {
+        free_batch(c, T.oA);
+        free_batch(c, T.oP);
+        T.oA = DevBatch{};
+        T.oP = DevBatch{};
+        c->arena_base = save_base;
+        c->arena_off = save_off;
+        T.pending = false;
+    };
+    dbsp_status st = build_deltas_chain(e, d_ev, n, T.rawA, T.rawP, T.oA,
+                                        T.oP, sb);
+    if (st != DBSP_OK) {
+        bail();
+        return st;
+    }
+    if ((int)e->p_int.batches.size() > MAX_TRACE_BATCHES)
+        TRY(e->p_int.consolidate_all(c));
+    if ((int)e->a_int.batches.size() > MAX_TRACE_BATCHES)
+        TRY(e->a_int.consolidate_all(c));
+    int jca_np = 0;
+    bool arena_ok = true;
+    {
+        ScopedTimer timer(c, 2, (double)n * 24.0);
+        TRY(q3_plan_count(e, T.oA, T.oP, true, sb, n, T.plans, T.np, jca_np,
+                          arena_ok));
+    }
+    if (!arena_ok || jca_np == 0) {
+        bail();
+        return DBSP_OK;  // fall back: next step runs the unpipelined body
+    }
+    HIP_CHECK_ST(hipMemcpyAsync(c->h_len + sb, c->d_len + sb,
+                                18 * sizeof(int64_t), hipMemcpyDeviceToHost,
+                                c->stream));
+    (void)hipEventRecord(c->ev_tick[evi], c->stream);
+    bool emits_ok = false;
+    TRY(q3_chain_emits(e, T.oA, T.oP, T.plans, T.np, sb, T.comb_chain,
+                       emits_ok));
+    if (!emits_ok) {
+        bail();
+        return DBSP_OK;
+    }
+    T.arena_base = c->arena_base;
+    T.arena_off = c->arena_off;
+    c->arena_base = save_base;
+    c->arena_off = save_off;
+    T.pending = true;
+    T.ev = d_ev;
+    T.n = n;
+    T.sb = sb;
+    T.evi = evi;
+    T.next_sb = sb == 0 ? 18 : 0;
+    T.next_evi = evi ^ 1;
+    return DBSP_OK;
+}
+
+static void q3_maybe_enqueue_train(dbsp_engine *e) {
+    dbsp_ctx *c = e->ctx;
+    static const bool train_on = []() {
+        const char *v = getenv("DBSP_Q3_TRAIN");
+        return !(v && v[0] == '0');
+    }();
+    if (!train_on || sharding_on(c) || e->train.pending || !e->next_ev ||
+        e->next_n <= 0 || e->next_n > 131072 || e->spec_fail >= 3)
+        return;
+    (void)q3_enqueue_train(e, e->next_ev, e->next_n);
+}
+
+// commit a pipelined train: wait its event, read verdicts, insert, publish
+static dbsp_status q3_commit_train(dbsp_engine *e) {
+    dbsp_ctx *c = e->ctx;
+    Q3Train T = e->train;
+    e->train.pending = false;
+    e->train.oA = DevBatch{};
+    e->train.oP = DevBatch{};
+    // commit-side transients go to the half the train is NOT using
+    c->arena_base =
+        c->arena_half ? (T.arena_base ? 0 : c->arena_half) : c->arena_base;
+    c->arena_off = 0;
+    (void)hipEventSynchronize(c->ev_tick[T.evi]);
+    int64_t H[18];
+    memcpy(H, c->h_len + T.sb, sizeof(H));
+    bool lost = H[10] < 0 || H[11] < 0;
+    for (int i = 0; i < T.np && !lost; i++)
+        if (T.plans[i].slot >= 0 && H[T.plans[i].slot] < 0) lost = true;
+    if (lost) {
+        e->spec_fail++;
+        free_batch(c, T.oA);
+        free_batch(c, T.oP);
+        // re-materialize the deltas from the raw flatmap outputs (arena of
+        // the train's half — untouched until the next enqueue) and replay
+        // the tick explicitly
+        DevBatch dA, dP;
+        T.rawA.n = H[8];
+        T.rawP.n = H[9];
+        TRY(sort_consolidate_batch(c, T.rawA, dA));
+        TRY(sort_consolidate_batch(c, T.rawP, dP));
+        return q3_body(e, T.ev, T.n, dA, dP, DevBatch{}, DevBatch{},
+                       DevBatch{}, DevBatch{}, false, false);
+    }
+    e->spec_fail = 0;
+    DevBatch dA = T.oA, dP = T.oP;
+    dA.n = H[10];
+    dP.n = H[11];
+    for (int i = 0; i < T.np; i++)
+        if (T.plans[i].dd) T.plans[i].t.n[0] = dP.n;
+    engine_free_output(e);
+    TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
+    // emit verdicts: the second readback (enqueued before the inserts'
+    // length copy) has completed by the inserts' first wait
+    const int64_t flag = c->h_len[T.sb + 13];
+    if (flag == 0) {
+        const int64_t out_n = c->h_len[T.sb + 15];
+        if (out_n >= 0) {
+            e->output = e->out_store;
+            e->output.n = out_n;
             e->output_is_store = true;
         } else {
-            TRY(finalize_raw(c, outs, e->output));
+            T.comb_chain.n = c->h_len[T.sb + 12];
+            TRY(sort_consolidate_batch(c, T.comb_chain, e->output));
         }
     } else {
-        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
-        if (async_final) {
-            HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6,
-                                        sizeof(int64_t),
-                                        hipMemcpyDeviceToHost, c->stream));
-            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
-            res.n = c->h_len[6];
-            e->output = res;
-            e->output_is_store = true;
-        } else {
-            TRY(finalize_raw(c, outs, e->output));
-        }
+        // combined emits exceeded the capacity buffer: replay explicitly
+        // from the still-valid counts/offsets
+        int64_t slot_totals[3] = {0, 0, 0};
+        for (int i = 0; i < T.np && i < 3; i++) slot_totals[i] = H[i];
+        std::vector<DevBatch> outs;
+        TRY(q3_emit_explicit(e, dA, dP, T.plans, T.np, slot_totals, outs));
+        TRY(finalize_raw(c, outs, e->output));
     }
+    q3_maybe_enqueue_train(e);
     return DBSP_OK;
+}
+
+static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
+    dbsp_ctx *c = e->ctx;
+    if (e->train.pending) {
+        if (e->train.ev == d_ev && e->train.n == n) return q3_commit_train(e);
+        // stale train (out-of-band step): drain its enqueued work, discard
+        (void)hipStreamSynchronize(c->stream);
+        free_batch(c, e->train.oA);
+        free_batch(c, e->train.oP);
+        e->train.pending = false;
+    }
+    const bool shard_chain =
+        sharding_on(c) && c->world <= 8 && n <= 131072 && e->spec_fail < 3;
+    const bool chain =
+        shard_chain || (!sharding_on(c) && n <= 131072 && e->spec_fail < 3);
+    DevBatch dA, dP, rawA, rawP, recvA, recvP;
+    if (shard_chain) {
+        TRY(build_deltas_chain_sharded(e, d_ev, n, rawA, rawP, recvA, recvP,
+                                       dA, dP));
+    } else if (chain) {
+        TRY(build_deltas_chain(e, d_ev, n, rawA, rawP, dA, dP));
+    } else {
+        TRY(build_deltas(e, d_ev, n, dA, dP, true));
+    }
+    return q3_body(e, d_ev, n, dA, dP, rawA, rawP, recvA, recvP, chain,
+                   shard_chain);
 }
 
 // ---- q8 tick (queries/q8.rs:48-93) ----
