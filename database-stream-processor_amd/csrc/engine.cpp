@@ -2389,6 +2389,9 @@ static dbsp_status q3_body(dbsp_engine *e, const dbsp_event *d_ev, int64_t n,
     engine_free_output(e);
     if (chain_emits) {
         TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
+        // the hook-less insert can exit without a tail sync (no merge
+        // rounds); the verdict readback below must be complete
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
         const int64_t flag = c->h_len[13];
         if (flag == 0) {
             const int64_t out_n = c->h_len[15];
@@ -2456,7 +2459,7 @@ static dbsp_status q3_body(dbsp_engine *e, const dbsp_event *d_ev, int64_t n,
                                     c->stream));
     TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
     if (async_final) {
-        if (sharding_on(c)) HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
         res.n = c->h_len[6];
         e->output = res;
         e->output_is_store = true;
@@ -2585,8 +2588,9 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
         if (T.plans[i].dd) T.plans[i].t.n[0] = dP.n;
     engine_free_output(e);
     TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
-    // emit verdicts: the second readback (enqueued before the inserts'
-    // length copy) has completed by the inserts' first wait
+    // the insert rounds may exit without a tail sync; the second readback's
+    // verdicts must be complete before reading them
+    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
     const int64_t flag = c->h_len[T.sb + 13];
     if (flag == 0) {
         const int64_t out_n = c->h_len[T.sb + 15];
