@@ -2388,10 +2388,17 @@ static dbsp_status q3_body(dbsp_engine *e, const dbsp_event *d_ev, int64_t n,
     }
     engine_free_output(e);
     if (chain_emits) {
-        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
-        // the hook-less insert can exit without a tail sync (no merge
-        // rounds); the verdict readback below must be complete
-        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+        // the hook enqueues the NEXT tick's train mid-insert (after this
+        // tick's merge launches, before their wait): probing the
+        // pre-cascade batch list is Z-set-identical to the merged state,
+        // and single-stream ordering keeps any recycled buffers safe.  The
+        // insert's event sync (recorded pre-hook) covers this tick's
+        // readbacks, so the verdict reads below are complete.
+        std::function<dbsp_status()> hook = [&]() -> dbsp_status {
+            q3_maybe_enqueue_train(e);
+            return DBSP_OK;
+        };
+        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
         const int64_t flag = c->h_len[13];
         if (flag == 0) {
             const int64_t out_n = c->h_len[15];
@@ -2407,7 +2414,6 @@ static dbsp_status q3_body(dbsp_engine *e, const dbsp_event *d_ev, int64_t n,
             TRY(q3_emit_explicit(e, dA, dP, plans, np, slot_totals, outs));
             TRY(finalize_raw(c, outs, e->output));
         }
-        q3_maybe_enqueue_train(e);
         return DBSP_OK;
     }
     // explicit path (sharded ranks and lost speculations)
@@ -2457,16 +2463,30 @@ static dbsp_status q3_body(dbsp_engine *e, const dbsp_event *d_ev, int64_t n,
         HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 6, c->d_len + 6,
                                     sizeof(int64_t), hipMemcpyDeviceToHost,
                                     c->stream));
-    TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
-    if (async_final) {
-        HIP_CHECK_ST(hipStreamSynchronize(c->stream));
-        res.n = c->h_len[6];
-        e->output = res;
-        e->output_is_store = true;
+    if (!sharding_on(c)) {
+        std::function<dbsp_status()> hook = [&]() -> dbsp_status {
+            q3_maybe_enqueue_train(e);
+            return DBSP_OK;
+        };
+        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
+        if (async_final) {
+            res.n = c->h_len[6];
+            e->output = res;
+            e->output_is_store = true;
+        } else {
+            TRY(finalize_raw(c, outs, e->output));
+        }
     } else {
-        TRY(finalize_raw(c, outs, e->output));
+        TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
+        if (async_final) {
+            HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+            res.n = c->h_len[6];
+            e->output = res;
+            e->output_is_store = true;
+        } else {
+            TRY(finalize_raw(c, outs, e->output));
+        }
     }
-    q3_maybe_enqueue_train(e);
     return DBSP_OK;
 }
 
@@ -2587,10 +2607,11 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
     for (int i = 0; i < T.np; i++)
         if (T.plans[i].dd) T.plans[i].t.n[0] = dP.n;
     engine_free_output(e);
-    TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
-    // the insert rounds may exit without a tail sync; the second readback's
-    // verdicts must be complete before reading them
-    HIP_CHECK_ST(hipStreamSynchronize(c->stream));
+    std::function<dbsp_status()> hook = [&]() -> dbsp_status {
+        q3_maybe_enqueue_train(e);
+        return DBSP_OK;
+    };
+    TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
     const int64_t flag = c->h_len[T.sb + 13];
     if (flag == 0) {
         const int64_t out_n = c->h_len[T.sb + 15];
@@ -2611,7 +2632,6 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
         TRY(q3_emit_explicit(e, dA, dP, T.plans, T.np, slot_totals, outs));
         TRY(finalize_raw(c, outs, e->output));
     }
-    q3_maybe_enqueue_train(e);
     return DBSP_OK;
 }
 
