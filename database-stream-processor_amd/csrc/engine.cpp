@@ -2210,25 +2210,35 @@ static dbsp_status q3_chain_emits(dbsp_engine *e, DevBatch &dA, DevBatch &dP,
     int jca_np = 0;
     for (int i = 0; i < np; i++)
         if (plans[i].slot >= 0) jca_np = std::max(jca_np, plans[i].slot + 1);
-    int64_t *bases = (int64_t *)arena_alloc(c, (jca_np + 1) * 8 + 8);
     DevBatch scr;
-    if (!bases || alloc_batch(c, Q3_EMIT_CAP, comb_chain, true) != DBSP_OK ||
+    if (alloc_batch(c, Q3_EMIT_CAP, comb_chain, true) != DBSP_OK ||
         alloc_batch(c, Q3_EMIT_CAP, scr, true) != DBSP_OK)
         return DBSP_OK;  // arena exhausted: caller keeps the explicit path
-    TRY(dbspk::emit_bases(c->stream, c->d_len + sb, jca_np, Q3_EMIT_CAP,
-                          bases, c->d_len + sb + 3, c->d_len + sb + 4));
+    dbspk::FusedEmitArgs fa{};
+    fa.np = jca_np;
     for (int i = 0; i < np; i++) {
         Q3Plan &pl = plans[i];
         if (pl.slot < 0) continue;
         DevBatch &d = pl.which == 0 ? dA : dP;
-        TRY(dbspk::join_emit_chain(
-            c->stream, d.k, d.v, d.w,
-            c->d_len + sb + (pl.which == 0 ? 10 : 11), pl.t,
-            pl.dd ? c->d_len + sb + 11 : nullptr, pl.cnts, pl.offsets,
-            c->d_len + sb + pl.slot, bases + pl.slot, c->d_len + sb + 4,
-            Q3_EMIT_CAP, pl.proj, 0, comb_chain.k, comb_chain.v,
-            comb_chain.w));
+        const int s2 = pl.slot;
+        fa.dk[s2] = d.k;
+        fa.dv[s2] = d.v;
+        fa.dw[s2] = d.w;
+        fa.nd_dev[s2] = c->d_len + sb + (pl.which == 0 ? 10 : 11);
+        fa.t[s2] = pl.t;
+        fa.tn_dev[s2] = pl.dd ? c->d_len + sb + 11 : nullptr;
+        fa.cnts[s2] = pl.cnts;
+        fa.offsets[s2] = pl.offsets;
+        fa.proj[s2] = pl.proj;
     }
+    fa.totals = c->d_len + sb;
+    fa.cap = Q3_EMIT_CAP;
+    fa.d_total = c->d_len + sb + 3;
+    fa.d_flag = c->d_len + sb + 4;
+    fa.ok = comb_chain.k;
+    fa.ov = comb_chain.v;
+    fa.ow = comb_chain.w;
+    TRY(dbspk::join_emit_fused(c->stream, fa));
     if (!e->out_store.k) {
         e->out_cap = 8192;
         TRY(alloc_batch(c, e->out_cap, e->out_store));

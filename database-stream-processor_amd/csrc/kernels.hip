@@ -1656,6 +1656,65 @@ __global__ void k_join_count_multi(const uint64_t *dk, int64_t nd, TraceArgs t,
 // (a lost speculation upstream, or the combined output exceeding the
 // capacity buffer) makes every consumer bail; the host re-emits explicitly.
 // ---------------------------------------------------------------------------
+// fused variant: bases + all (<=3) plans' emits in ONE launch — each of the
+// three separate chained emit launches cost ~5 us of dependent dispatch on
+// the q3 tick's critical path; every thread recomputes the 3-entry base
+// prefix from the (L2-hot) count totals instead
+__global__ void k_join_emit_fused(FusedEmitArgs a) {
+    int64_t bases[3];
+    int64_t acc = 0;
+    int okf = 1;
+    for (int i = 0; i < a.np; i++) {
+        const int64_t t = a.totals[i];
+        if (t < 0) {
+            okf = 0;
+            break;
+        }
+        bases[i] = acc;
+        acc += t;
+    }
+    if (acc > a.cap) okf = 0;
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        *a.d_total = okf ? acc : -1;
+        *a.d_flag = okf ? 0 : 1;
+    }
+    if (!okf) return;
+    for (int64_t o = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; o < acc;
+         o += (int64_t)gridDim.x * blockDim.x) {
+        int p = 0;
+        while (p + 1 < a.np && o >= bases[p + 1]) p++;
+        const int64_t nd = *a.nd_dev[p];
+        if (nd <= 0) continue;
+        const TraceArgs &t = a.t[p];
+        const uint32_t *cnts = a.cnts[p];
+        const uint64_t *offsets = a.offsets[p];
+        const int64_t tn0 = a.tn_dev[p] ? *a.tn_dev[p] : t.n[0];
+        const int64_t oo = o - bases[p];
+        int64_t lo = 0, hi = nd;
+        while (lo < hi) {
+            int64_t mid = (lo + hi) / 2;
+            if (offsets[mid] <= (uint64_t)oo) lo = mid + 1; else hi = mid;
+        }
+        int64_t i = lo - 1;
+        int64_t j = oo - (int64_t)offsets[i];
+        int b = 0;
+        while (j >= (int64_t)cnts[i * t.nb + b]) {
+            j -= cnts[i * t.nb + b];
+            b++;
+        }
+        const uint64_t key = a.dk[p][i];
+        const int64_t tb_n = b == 0 ? tn0 : t.n[b];
+        const int64_t start = lower_bound_k(t.k[b], tb_n, key);
+        const int64_t ti = start + j;
+        uint64_t hi_o, lo_o;
+        const bool valid = proj_out(a.proj[p], 0, key, a.dv[p][i],
+                                    t.v[b][ti], hi_o, lo_o);
+        a.ok[o] = hi_o;
+        a.ov[o] = lo_o;
+        a.ow[o] = valid ? a.dw[p][i] * t.w[b][ti] : 0;
+    }
+}
+
 __global__ void k_emit_bases(const int64_t *totals, int np, int64_t cap,
                              int64_t *bases, int64_t *out_total,
                              int64_t *out_flag) {
@@ -2887,6 +2946,11 @@ dbsp_status join_emit_chain(hipStream_t s, const uint64_t *dk,
     k_join_emit_chain<<<grid_for(grid_cap), BLK, 0, s>>>(
         dk, dv, dw, nd_dev, t, tn_dev, cnts, offsets, total_dev, base_dev,
         flag_dev, proj, param, ok, ov, ow);
+    return DBSP_OK;
+}
+
+dbsp_status join_emit_fused(hipStream_t s, const FusedEmitArgs &a) {
+    k_join_emit_fused<<<grid_for(a.cap), BLK, 0, s>>>(a);
     return DBSP_OK;
 }
 

@@ -54,6 +54,26 @@ struct MergeArgs {
     int64_t *d_len;
 };
 
+// fused chained-emit descriptor (bases + all plans' emits in one launch)
+struct FusedEmitArgs {
+    int np;
+    const uint64_t *dk[3], *dv[3];
+    const int64_t *dw[3];
+    const int64_t *nd_dev[3];
+    TraceArgs t[3];
+    const int64_t *tn_dev[3];
+    const uint32_t *cnts[3];
+    const uint64_t *offsets[3];
+    int proj[3];
+    const int64_t *totals;  // per-plan count totals (device)
+    int64_t cap;
+    int64_t *d_total;  // combined total out (or -1)
+    int64_t *d_flag;   // overflow flag out
+    uint64_t *ok, *ov;
+    int64_t *ow;
+};
+dbsp_status join_emit_fused(hipStream_t s, const FusedEmitArgs &a);
+
 struct JoinCountArgs {
     int np;
     const uint64_t *dk[3];
