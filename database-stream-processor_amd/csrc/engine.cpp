@@ -2214,7 +2214,7 @@ static dbsp_status q3_chain_emits(dbsp_engine *e, DevBatch &dA, DevBatch &dP,
     if (alloc_batch(c, Q3_EMIT_CAP, comb_chain, true) != DBSP_OK ||
         alloc_batch(c, Q3_EMIT_CAP, scr, true) != DBSP_OK)
         return DBSP_OK;  // arena exhausted: caller keeps the explicit path
-    dbspk::FusedEmitArgs fa{};
+    FusedEmitArgs fa{};
     fa.np = jca_np;
     for (int i = 0; i < np; i++) {
         Q3Plan &pl = plans[i];

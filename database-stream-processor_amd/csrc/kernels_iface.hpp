@@ -72,8 +72,6 @@ struct FusedEmitArgs {
     uint64_t *ok, *ov;
     int64_t *ow;
 };
-dbsp_status join_emit_fused(hipStream_t s, const FusedEmitArgs &a);
-
 struct JoinCountArgs {
     int np;
     const uint64_t *dk[3];
@@ -264,6 +262,8 @@ dbsp_status distinct_inc_rows(hipStream_t s, const uint64_t *dk,
                               const uint64_t *dv, const int64_t *dw,
                               int64_t nd, const TraceArgs &t, uint64_t **ok,
                               uint64_t **ov, int64_t **ow, int64_t *out_n);
+
+dbsp_status join_emit_fused(hipStream_t s, const FusedEmitArgs &a);
 
 dbsp_status agg_linear_upsert_rows(hipStream_t s, const uint64_t *keys,
                                    int64_t nd, const uint64_t *ik,
