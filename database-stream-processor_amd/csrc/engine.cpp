@@ -334,11 +334,9 @@ static dbsp_status sort_medium(dbsp_ctx *c, DevBatch raw, DevBatch &out) {
             results.push_back(res);
         }
         if (ma.np > 0) {
-            // dedicated insert slots (36-39): a pipelined next-tick train
-            // owns one of the two 18-slot tick bases at this point
-            ma.d_len = c->d_len + 36;
+            ma.d_len = c->d_len;
             TRY(dbspk::merge_small_batch(c->stream, ma));
-            HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 36, c->d_len + 36,
+            HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
                                         ma.np * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
             HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -2036,8 +2034,10 @@ static dbsp_status spines_insert_multi(dbsp_ctx *c, Spine *const *sps,
             }
         }
         if (ma.np > 0) {
-            // dedicated insert slots (36-39): a pipelined next-tick train
-            // owns one of the two 18-slot tick bases at this point
+            // dedicated insert slots (36-39): the hook below may enqueue the
+            // next tick's full train, whose 18-slot readback would otherwise
+            // overwrite h_len[0..3] before this round's merge lengths are
+            // consumed after the event wait
             ma.d_len = c->d_len + 36;
             TRY(dbspk::merge_small_batch(c->stream, ma));
             HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 36, c->d_len + 36,
