@@ -2579,7 +2579,19 @@ static void q3_maybe_enqueue_train(dbsp_engine *e) {
 }
 
 // commit a pipelined train: wait its event, read verdicts, insert, publish
+static inline double host_us() {
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return ts.tv_sec * 1e6 + ts.tv_nsec * 1e-3;
+}
+
 static dbsp_status q3_commit_train(dbsp_engine *e) {
+    static const bool prof = []() {
+        const char *v = getenv("DBSP_COMMIT_PROF");
+        return v && v[0] == '1';
+    }();
+    static int prof_n = 0;
+    double t0 = prof ? host_us() : 0;
     dbsp_ctx *c = e->ctx;
     Q3Train T = e->train;
     e->train.pending = false;
@@ -2590,6 +2602,7 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
         c->arena_half ? (T.arena_base ? 0 : c->arena_half) : c->arena_base;
     c->arena_off = 0;
     (void)hipEventSynchronize(c->ev_tick[T.evi]);
+    double t1 = prof ? host_us() : 0;
     int64_t H[18];
     memcpy(H, c->h_len + T.sb, sizeof(H));
     bool lost = H[10] < 0 || H[11] < 0;
@@ -2617,11 +2630,20 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
     for (int i = 0; i < T.np; i++)
         if (T.plans[i].dd) T.plans[i].t.n[0] = dP.n;
     engine_free_output(e);
+    double t2 = prof ? host_us() : 0;
+    double th0 = 0, th1 = 0;
     std::function<dbsp_status()> hook = [&]() -> dbsp_status {
+        th0 = prof ? host_us() : 0;
         q3_maybe_enqueue_train(e);
+        th1 = prof ? host_us() : 0;
         return DBSP_OK;
     };
     TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
+    double t3 = prof ? host_us() : 0;
+    if (prof && ++prof_n % 50 == 0)
+        fprintf(stderr,
+                "[commit] wait %.1f verdict %.1f insert(total %.1f, hook %.1f) us\n",
+                t1 - t0, t2 - t1, t3 - t2, th1 - th0);
     const int64_t flag = c->h_len[T.sb + 13];
     if (flag == 0) {
         const int64_t out_n = c->h_len[T.sb + 15];
