@@ -80,6 +80,7 @@ struct dbsp_ctx {
                                 // sb+16/17 = framed-exchange totals) +
                                 // 36-39 spine-insert rounds
     int64_t *h_len = nullptr;
+    int64_t *d_mid = nullptr;  // merge_mid per-WG count scratch (2 pairs)
     int timer_depth = 0;  // ScopedTimer nesting guard (shared event pair)
     // per-tick transient bump arena (reset at each engine tick; falls back to
     // the stream-ordered pool when exhausted)
@@ -93,6 +94,16 @@ struct dbsp_ctx {
     hipEvent_t ev_sync = nullptr;   // tick-end event (pre-front-launch point)
     hipEvent_t ev_sync2 = nullptr;  // early-wake event (post-count readback)
     hipEvent_t ev_tick[2] = {nullptr, nullptr};  // pipelined-train events
+    hipEvent_t ev_tail[2] = {nullptr, nullptr};  // train tail (post 2nd readback)
+    // deferred spine-insert round (train path): the last small-merge round's
+    // length readback is resolved at the NEXT insert/step on a near-empty
+    // stream instead of blocking behind the next tick's freshly enqueued
+    // train (h_len[36..39] hold its counts until then)
+    hipEvent_t ev_insert = nullptr;
+    int pend_n = 0;
+    void *pend_spine[4] = {};
+    uint64_t *pend_k[4] = {}, *pend_v[4] = {};
+    int64_t *pend_w[4] = {};
 };
 
 static void *arena_alloc(dbsp_ctx *c, size_t bytes) {
@@ -143,11 +154,15 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_sync2, hipEventDisableTiming));
     HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_tick[0], hipEventDisableTiming));
     HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_tick[1], hipEventDisableTiming));
+    HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_insert, hipEventDisableTiming));
+    HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_tail[0], hipEventDisableTiming));
+    HIP_CHECK_ST(hipEventCreateWithFlags(&c->ev_tail[1], hipEventDisableTiming));
     const char *p = getenv("DBSP_PROFILE");
     c->profile = p && p[0] == '1';
     const char *fs = getenv("DBSP_FORCE_SHARD");
     c->force_shard = fs && fs[0] == '1';
     HIP_CHECK_ST(hipMalloc(&c->d_len, 44 * sizeof(int64_t)));
+    HIP_CHECK_ST(hipMalloc(&c->d_mid, 2 * MERGE_MID_SCRATCH * sizeof(int64_t)));
     c->arena_sz = (size_t)512 << 20;
     c->arena_half = c->arena_sz / 2;
     if (hipMalloc(&c->arena, c->arena_sz) != hipSuccess) {
@@ -159,8 +174,14 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     return DBSP_OK;
 }
 
+static void drop_pending_insert(dbsp_ctx *c);
+
 extern "C" dbsp_status dbsp_ctx_destroy(dbsp_ctx *c) {
     if (!c) return DBSP_OK;
+    drop_pending_insert(c);
+    if (c->ev_insert) (void)hipEventDestroy(c->ev_insert);
+    if (c->ev_tail[0]) (void)hipEventDestroy(c->ev_tail[0]);
+    if (c->ev_tail[1]) (void)hipEventDestroy(c->ev_tail[1]);
     if (c->ev_sync) (void)hipEventDestroy(c->ev_sync);
     if (c->ev_sync2) (void)hipEventDestroy(c->ev_sync2);
     if (c->ev_tick[0]) (void)hipEventDestroy(c->ev_tick[0]);
@@ -168,6 +189,7 @@ extern "C" dbsp_status dbsp_ctx_destroy(dbsp_ctx *c) {
     if (c->comm) ncclCommDestroy(c->comm);
     (void)hipStreamSynchronize(c->stream);
     if (c->d_len) (void)hipFree(c->d_len);
+    if (c->d_mid) (void)hipFree(c->d_mid);
     dbspk::cache_trim(c->stream);
     (void)hipStreamSynchronize(c->stream);
     if (c->arena) (void)hipFree(c->arena);
@@ -1042,6 +1064,10 @@ struct Q3Train {
     int sb = 0, evi = 0;
     int next_sb = 18, next_evi = 0;
     DevBatch rawA, rawP, oA, oP, comb_chain;
+    // in-train accumulator merge results (acc side + this tick's delta side);
+    // lengths land in d_len[40+2*evi .. 41+2*evi] with the first readback
+    DevBatch res[2];
+    bool has_res = false;
     Q3Plan plans[3];
     int np = 0;
     size_t arena_base = 0, arena_off = 0;
@@ -1064,6 +1090,11 @@ struct dbsp_engine {
 
     // q3 state
     Spine a_int, p_int;
+    // train-path accumulator (memtable): mirrors the TOP batch of a_int /
+    // p_int when that batch was produced by the in-train (acc ⋈ delta)
+    // merge; n==0 means the top is not accumulator-owned (classic inserts,
+    // spills, fresh engine) and the next train merges from empty
+    DevBatch q3_acc[2];
     // q8 state
     Spine pt_int, at_int, wp_int, wa_int;
     // q5 state
@@ -1156,6 +1187,10 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
     if (e && e->d_wm) (void)hipFree(e->d_wm);
     if (!e) return DBSP_OK;
     dbsp_ctx *c = e->ctx;
+    // discard any deferred insert round before the spines it references go
+    // away (the merged pair is still in its spine's list, so dropping the
+    // result is safe for ANY engine's pending round — at worst re-merged)
+    drop_pending_insert(c);
     for (Spine *s : {&e->a_int, &e->p_int, &e->pt_int, &e->at_int, &e->wp_int,
                      &e->wa_int, &e->bt_int, &e->wb_int, &e->counts_int,
                      &e->bc_int, &e->c5_trace, &e->c5_wint, &e->c5_out,
@@ -1178,6 +1213,8 @@ extern "C" dbsp_status dbsp_engine_destroy(dbsp_engine *e) {
         (void)hipStreamSynchronize(c->stream);
         free_batch(c, e->train.oA);
         free_batch(c, e->train.oP);
+        free_batch(c, e->train.res[0]);
+        free_batch(c, e->train.res[1]);
     }
     if (e->out_store.k) free_batch(c, e->out_store);
     if (e->d_events) (void)dbspk::cache_free(e->d_events, c->stream);
@@ -1981,18 +2018,17 @@ static bool spine_needs_merge(Spine &s) {
 // recorded, and the first wait is hipEventSynchronize on that event — so
 // whatever the hook launches (the NEXT tick's front half) overlaps the
 // host-side tail of this tick instead of extending its sync.
-static dbsp_status spines_insert_multi(dbsp_ctx *c, Spine *const *sps,
-                                       DevBatch *bs, int ns,
-                                       const std::function<dbsp_status()> *hook
-                                       = nullptr) {
-    double bytes = 0;
-    for (int i = 0; i < ns; i++) bytes += (double)bs[i].n * 48.0;
-    ScopedTimer t(c, 1, bytes);
-    bool hook_fired = false;
-    for (int i = 0; i < ns; i++) {
-        if (bs[i].n > 0) sps[i]->batches.push_back(bs[i]);
-        else free_batch(c, bs[i]);
-    }
+// run top-two merge rounds over the given spines until quiescent.  With
+// `defer` set (train path), the FIRST batched small-merge round is enqueued,
+// the hook fired behind it, and the length readback STASHED in the ctx
+// instead of waited for: resolve_pending_insert() consumes it at the next
+// insert/step on a near-empty stream.  Without it, a round that runs after
+// the hook has enqueued the next tick's train would hipStreamSynchronize
+// behind that entire train (~150 us) — the dominant host stall of the
+// 40k-tick loop.
+static dbsp_status spine_rounds(dbsp_ctx *c, Spine *const *sps, int ns,
+                                const std::function<dbsp_status()> *hook,
+                                bool *hook_fired, bool defer) {
     while (true) {
         Spine *pending[4];
         int nps = 0;
@@ -2043,10 +2079,23 @@ static dbsp_status spines_insert_multi(dbsp_ctx *c, Spine *const *sps,
             HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 36, c->d_len + 36,
                                         ma.np * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
-            if (hook && !hook_fired) {
+            if (defer && hook && !*hook_fired) {
+                (void)hipEventRecord(c->ev_insert, c->stream);
+                TRY((*hook)());
+                *hook_fired = true;
+                c->pend_n = nbatched;
+                for (int j = 0; j < nbatched; j++) {
+                    c->pend_spine[j] = owners[j];
+                    c->pend_k[j] = results[j].k;
+                    c->pend_v[j] = results[j].v;
+                    c->pend_w[j] = results[j].w;
+                }
+                return DBSP_OK;  // pops + cascades resolved next tick
+            }
+            if (hook && !*hook_fired) {
                 (void)hipEventRecord(c->ev_sync, c->stream);
                 TRY((*hook)());
-                hook_fired = true;
+                *hook_fired = true;
                 (void)hipEventSynchronize(c->ev_sync);
             } else {
                 HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -2065,6 +2114,65 @@ static dbsp_status spines_insert_multi(dbsp_ctx *c, Spine *const *sps,
             }
         }
     }
+    return DBSP_OK;
+}
+
+// consume a deferred insert round: pop the merged pair, publish the result
+// (length from h_len[36..39], valid once ev_insert has fired), then run any
+// cascade rounds synchronously — the stream is near-empty at every call
+// site, so these cost only the merge kernels themselves
+static dbsp_status resolve_pending_insert(dbsp_ctx *c) {
+    if (!c->pend_n) return DBSP_OK;
+    (void)hipEventSynchronize(c->ev_insert);
+    Spine *owners[4];
+    int no = 0;
+    const int np = c->pend_n;
+    c->pend_n = 0;
+    for (int j = 0; j < np; j++) {
+        Spine &sp = *(Spine *)c->pend_spine[j];
+        DevBatch res{c->pend_k[j], c->pend_v[j], c->pend_w[j],
+                     c->h_len[36 + j]};
+        DevBatch top = sp.batches.back();
+        sp.batches.pop_back();
+        DevBatch below = sp.batches.back();
+        sp.batches.pop_back();
+        free_batch(c, top);
+        free_batch(c, below);
+        if (res.n > 0) sp.batches.push_back(res);
+        else free_batch(c, res);
+        bool dup = false;
+        for (int i = 0; i < no; i++) dup |= owners[i] == &sp;
+        if (!dup) owners[no++] = &sp;
+    }
+    return spine_rounds(c, owners, no, nullptr, nullptr, false);
+}
+
+// teardown guard: free a deferred round's result buffers without touching
+// the spines (their batch lists are freed by their own teardown)
+static void drop_pending_insert(dbsp_ctx *c) {
+    if (!c->pend_n) return;
+    (void)hipEventSynchronize(c->ev_insert);
+    for (int j = 0; j < c->pend_n; j++) {
+        DevBatch res{c->pend_k[j], c->pend_v[j], c->pend_w[j], 0};
+        free_batch(c, res);
+    }
+    c->pend_n = 0;
+}
+
+static dbsp_status spines_insert_multi(dbsp_ctx *c, Spine *const *sps,
+                                       DevBatch *bs, int ns,
+                                       const std::function<dbsp_status()> *hook
+                                       = nullptr, bool defer = false) {
+    double bytes = 0;
+    for (int i = 0; i < ns; i++) bytes += (double)bs[i].n * 48.0;
+    ScopedTimer t(c, 1, bytes);
+    TRY(resolve_pending_insert(c));
+    bool hook_fired = false;
+    for (int i = 0; i < ns; i++) {
+        if (bs[i].n > 0) sps[i]->batches.push_back(bs[i]);
+        else free_batch(c, bs[i]);
+    }
+    TRY(spine_rounds(c, sps, ns, hook, &hook_fired, defer));
     if (hook && !hook_fired) {
         (void)hipEventRecord(c->ev_sync, c->stream);
         TRY((*hook)());
@@ -2076,10 +2184,10 @@ static dbsp_status spines_insert_multi(dbsp_ctx *c, Spine *const *sps,
 static dbsp_status spines_insert_pair(dbsp_ctx *c, Spine &s1, DevBatch b1,
                                       Spine &s2, DevBatch b2,
                                       const std::function<dbsp_status()> *hook
-                                      = nullptr) {
+                                      = nullptr, bool defer = false) {
     Spine *sps[2] = {&s1, &s2};
     DevBatch bs[2] = {b1, b2};
-    return spines_insert_multi(c, sps, bs, 2, hook);
+    return spines_insert_multi(c, sps, bs, 2, hook, defer);
 }
 
 static void engine_free_output(dbsp_engine *e) {
@@ -2408,6 +2516,9 @@ static dbsp_status q3_body(dbsp_engine *e, const dbsp_event *d_ev, int64_t n,
             q3_maybe_enqueue_train(e);
             return DBSP_OK;
         };
+        // classic insert re-levels the spine: the accumulator identity is
+        // lost (a train enqueued by the hook merges from empty)
+        e->q3_acc[0] = e->q3_acc[1] = DevBatch{};
         TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
         const int64_t flag = c->h_len[13];
         if (flag == 0) {
@@ -2478,6 +2589,7 @@ static dbsp_status q3_body(dbsp_engine *e, const dbsp_event *d_ev, int64_t n,
             q3_maybe_enqueue_train(e);
             return DBSP_OK;
         };
+        e->q3_acc[0] = e->q3_acc[1] = DevBatch{};
         TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
         if (async_final) {
             res.n = c->h_len[6];
@@ -2487,6 +2599,7 @@ static dbsp_status q3_body(dbsp_engine *e, const dbsp_event *d_ev, int64_t n,
             TRY(finalize_raw(c, outs, e->output));
         }
     } else {
+        e->q3_acc[0] = e->q3_acc[1] = DevBatch{};
         TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP));
         if (async_final) {
             HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -2514,8 +2627,11 @@ static dbsp_status q3_enqueue_train(dbsp_engine *e, const dbsp_event *d_ev,
     auto bail = [&](void) {
         free_batch(c, T.oA);
         free_batch(c, T.oP);
+        free_batch(c, T.res[0]);
+        free_batch(c, T.res[1]);
         T.oA = DevBatch{};
         T.oP = DevBatch{};
+        T.has_res = false;
         c->arena_base = save_base;
         c->arena_off = save_off;
         T.pending = false;
@@ -2526,10 +2642,45 @@ static dbsp_status q3_enqueue_train(dbsp_engine *e, const dbsp_event *d_ev,
         bail();
         return st;
     }
-    if ((int)e->p_int.batches.size() > MAX_TRACE_BATCHES)
+    if ((int)e->p_int.batches.size() > MAX_TRACE_BATCHES) {
         TRY(e->p_int.consolidate_all(c));
-    if ((int)e->a_int.batches.size() > MAX_TRACE_BATCHES)
+        e->q3_acc[1] = DevBatch{};
+    }
+    if ((int)e->a_int.batches.size() > MAX_TRACE_BATCHES) {
         TRY(e->a_int.consolidate_all(c));
+        e->q3_acc[0] = DevBatch{};
+    }
+    // in-train accumulator merge: fold this tick's deltas into the per-spine
+    // memtable (one 2-pair single-WG launch; delta lengths read from the
+    // d_len slots the chain just wrote, so nothing here waits).  The commit
+    // swaps the spine top for the result — or, past the spill threshold,
+    // seals it as a ladder batch.  This removes the host-launched insert
+    // round that left the GPU idle ~45 us at every tick boundary.
+    {
+        MergeArgs ma{};
+        const int slot_base = 40 + 2 * evi;
+        for (int s = 0; s < 2; s++) {
+            DevBatch &acc = e->q3_acc[s];
+            DevBatch &delta = s == 0 ? T.oA : T.oP;
+            if (alloc_batch(c, acc.n + n, T.res[s]) != DBSP_OK) {
+                bail();
+                return DBSP_OK;
+            }
+            ma.ak[s] = acc.k; ma.av[s] = acc.v; ma.aw[s] = acc.w;
+            ma.na[s] = acc.n;
+            ma.bk[s] = delta.k; ma.bv[s] = delta.v; ma.bw[s] = delta.w;
+            ma.nb[s] = 0;
+            ma.dnb[s] = c->d_len + sb + (s == 0 ? 10 : 11);
+            ma.ok[s] = T.res[s].k; ma.ov[s] = T.res[s].v; ma.ow[s] = T.res[s].w;
+            ma.np++;
+        }
+        ma.d_len = c->d_len + slot_base;
+        TRY(dbspk::merge_mid_batch(c->stream, ma, c->d_mid));
+        HIP_CHECK_ST(hipMemcpyAsync(c->h_len + slot_base, c->d_len + slot_base,
+                                    2 * sizeof(int64_t),
+                                    hipMemcpyDeviceToHost, c->stream));
+        T.has_res = true;
+    }
     int jca_np = 0;
     bool arena_ok = true;
     {
@@ -2552,6 +2703,10 @@ static dbsp_status q3_enqueue_train(dbsp_engine *e, const dbsp_event *d_ev,
         bail();
         return DBSP_OK;
     }
+    // tail barrier: the commit's output-length reads (h_len[sb+12..15]) are
+    // written by the emit chain's SECOND readback, which ev_tick does not
+    // cover — the commit waits this event just before consuming them
+    (void)hipEventRecord(c->ev_tail[evi], c->stream);
     T.arena_base = c->arena_base;
     T.arena_off = c->arena_off;
     c->arena_base = save_base;
@@ -2597,6 +2752,9 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
     e->train.pending = false;
     e->train.oA = DevBatch{};
     e->train.oP = DevBatch{};
+    e->train.res[0] = DevBatch{};
+    e->train.res[1] = DevBatch{};
+    e->train.has_res = false;
     // commit-side transients go to the half the train is NOT using
     c->arena_base =
         c->arena_half ? (T.arena_base ? 0 : c->arena_half) : c->arena_base;
@@ -2612,6 +2770,8 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
         e->spec_fail++;
         free_batch(c, T.oA);
         free_batch(c, T.oP);
+        free_batch(c, T.res[0]);
+        free_batch(c, T.res[1]);
         // re-materialize the deltas from the raw flatmap outputs (arena of
         // the train's half — untouched until the next enqueue) and replay
         // the tick explicitly
@@ -2631,19 +2791,60 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
         if (T.plans[i].dd) T.plans[i].t.n[0] = dP.n;
     engine_free_output(e);
     double t2 = prof ? host_us() : 0;
+    // --- accumulator swap (the spine insert ran inside the train) ---
+    // The in-train merge produced res[s] = acc[s] ⋈ delta[s]; here we only
+    // swap pointers: pop the consumed accumulator, publish the result, and
+    // past the spill threshold seal it into the ladder.  No kernel launch
+    // on the common path, so the next train (hook below) starts within a
+    // few microseconds of the wake.
+    TRY(resolve_pending_insert(c));
+    constexpr int64_t Q3_ACC_SPILL = 32768;
+    const int slot_base = 40 + 2 * T.evi;
+    Spine *spill[2];
+    int nspill = 0;
+    for (int s = 0; s < 2; s++) {
+        Spine &sp = s == 0 ? e->a_int : e->p_int;
+        DevBatch res = T.res[s];
+        res.n = c->h_len[slot_base + s];
+        DevBatch acc = e->q3_acc[s];
+        if (acc.n > 0) {  // the top batch is the consumed accumulator
+            sp.batches.pop_back();
+            free_batch(c, acc);
+        }
+        e->q3_acc[s] = DevBatch{};
+        if (res.n > 0) {
+            sp.batches.push_back(res);
+            if (res.n > Q3_ACC_SPILL) {
+                spill[nspill++] = &sp;
+            } else {
+                e->q3_acc[s] = res;
+            }
+        } else {
+            free_batch(c, res);
+        }
+    }
     double th0 = 0, th1 = 0;
+    bool fired = false;
     std::function<dbsp_status()> hook = [&]() -> dbsp_status {
         th0 = prof ? host_us() : 0;
         q3_maybe_enqueue_train(e);
         th1 = prof ? host_us() : 0;
         return DBSP_OK;
     };
-    TRY(spines_insert_pair(c, e->a_int, dA, e->p_int, dP, &hook));
+    if (nspill > 0) {
+        // sealed batches re-level through the classic rounds; the readback
+        // is deferred (stash) so nothing blocks behind the new train
+        TRY(spine_rounds(c, spill, nspill, &hook, &fired, /*defer=*/true));
+    }
+    if (!fired) TRY(hook());
     double t3 = prof ? host_us() : 0;
     if (prof && ++prof_n % 50 == 0)
         fprintf(stderr,
-                "[commit] wait %.1f verdict %.1f insert(total %.1f, hook %.1f) us\n",
+                "[commit] wait %.1f verdict %.1f swap+hook(total %.1f, hook %.1f) us\n",
                 t1 - t0, t2 - t1, t3 - t2, th1 - th0);
+    // this tick's output lengths land with the emit chain's second
+    // readback, past ev_tick — wait the tail before consuming them
+    (void)hipEventSynchronize(c->ev_tail[T.evi]);
     const int64_t flag = c->h_len[T.sb + 13];
     if (flag == 0) {
         const int64_t out_n = c->h_len[T.sb + 15];
@@ -2664,6 +2865,10 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
         TRY(q3_emit_explicit(e, dA, dP, T.plans, T.np, slot_totals, outs));
         TRY(finalize_raw(c, outs, e->output));
     }
+    // the deltas were folded into res by the in-train merge (they are NOT
+    // in the spine); free once the output paths above are done with them
+    free_batch(c, dA);
+    free_batch(c, dP);
     return DBSP_OK;
 }
 
@@ -2675,6 +2880,9 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         (void)hipStreamSynchronize(c->stream);
         free_batch(c, e->train.oA);
         free_batch(c, e->train.oP);
+        free_batch(c, e->train.res[0]);
+        free_batch(c, e->train.res[1]);
+        e->train.has_res = false;
         e->train.pending = false;
     }
     const bool shard_chain =

@@ -1560,11 +1560,15 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_small(MergeArgs args)
     const int64_t na = args.na[p];
     const uint64_t *bk = args.bk[p], *bv = args.bv[p];
     const W *bw = (const W *)args.bw[p];
-    const int64_t nb = args.nb[p];
+    const int64_t nb = args.dnb[p] ? *args.dnb[p] : args.nb[p];
     uint64_t *ok = args.ok[p], *ov = args.ov[p];
     W *ow = (W *)args.ow[p];
     int64_t *out_len = args.d_len + p;
     const int tid = threadIdx.x;
+    if (nb < 0 || na < 0) {  // failed upstream speculation: poison the length
+        if (tid == 0) *out_len = -1;
+        return;
+    }
     const int64_t total = na + nb;
     const int64_t items = (total + FUSE_THREADS - 1) / FUSE_THREADS;
     int64_t d0 = min((int64_t)tid * items, total);
@@ -1605,6 +1609,137 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_small(MergeArgs args)
         }
     }
     if (tid == 0) *out_len = (int64_t)tot_out;
+}
+
+// fixed-grid multi-workgroup merge with DEVICE-side b length (the in-train
+// accumulator fold: the tick's delta length is only known on device).  Two
+// phases over MERGE_MID_WGS workgroups per pair: count writes per-WG output
+// counts to scratch; emit recomputes each WG's prefix from the tiny counts
+// array (no separate scan launch) and writes its segment.  Equal (k, v)
+// pairs never split across WGs (adjust_split), so per-segment consolidation
+// is exact.
+#define MERGE_MID_WGS 16
+static_assert(MERGE_MID_SCRATCH == MERGE_MID_WGS + 1,
+              "scratch layout mismatch with kernels_iface.hpp");
+
+template <typename W>
+__global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_mid_count(
+        MergeArgs args, int64_t *scratch /* np * (MERGE_MID_WGS+1) */) {
+    const int p = blockIdx.y;
+    const int wg = blockIdx.x;
+    int64_t *cnt_row = scratch + (int64_t)p * (MERGE_MID_WGS + 1);
+    const int64_t na = args.na[p];
+    const int64_t nb = args.dnb[p] ? *args.dnb[p] : args.nb[p];
+    if (na < 0 || nb < 0) {
+        if (threadIdx.x == 0) cnt_row[MERGE_MID_WGS] = -1;
+        return;
+    }
+    if (threadIdx.x == 0 && wg == 0) cnt_row[MERGE_MID_WGS] = 0;
+    const uint64_t *ak = args.ak[p], *av = args.av[p];
+    const W *aw = (const W *)args.aw[p];
+    const uint64_t *bk = args.bk[p], *bv = args.bv[p];
+    const W *bw = (const W *)args.bw[p];
+    const int64_t total = na + nb;
+    const int64_t per_wg = (total + MERGE_MID_WGS - 1) / MERGE_MID_WGS;
+    const int64_t w0 = min((int64_t)wg * per_wg, total);
+    const int64_t w1 = min(w0 + per_wg, total);
+    const int64_t items = (w1 - w0 + FUSE_THREADS - 1) / FUSE_THREADS;
+    int64_t d0 = min(w0 + (int64_t)threadIdx.x * items, w1);
+    int64_t d1 = min(d0 + items, w1);
+    int64_t ai, bi, ae, be;
+    merge_path(ak, av, na, bk, bv, nb, d0, ai, bi);
+    adjust_split(ak, av, bk, bv, na, nb, ai, bi);
+    merge_path(ak, av, na, bk, bv, nb, d1, ae, be);
+    adjust_split(ak, av, bk, bv, na, nb, ae, be);
+    uint32_t cnt = 0;
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(ak[i], av[i], bk[j], bv[j])) {
+                if (aw[i] + bw[j] != (W)0) cnt++;
+                i++; j++;
+            } else if (j >= be || (i < ae && row_lt(ak[i], av[i], bk[j], bv[j]))) {
+                cnt++; i++;
+            } else {
+                cnt++; j++;
+            }
+        }
+    }
+    __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
+    uint32_t wg_tot;
+    (void)fuse_scan(cnt, wave_tot, &wg_tot);
+    if (threadIdx.x == 0) cnt_row[wg] = (int64_t)wg_tot;
+}
+
+template <typename W>
+__global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_mid_emit(
+        MergeArgs args, const int64_t *scratch) {
+    const int p = blockIdx.y;
+    const int wg = blockIdx.x;
+    const int64_t *cnt_row = scratch + (int64_t)p * (MERGE_MID_WGS + 1);
+    int64_t *out_len = args.d_len + p;
+    if (cnt_row[MERGE_MID_WGS] < 0) {
+        if (threadIdx.x == 0 && wg == 0) *out_len = -1;
+        return;
+    }
+    const int64_t na = args.na[p];
+    const int64_t nb = args.dnb[p] ? *args.dnb[p] : args.nb[p];
+    const uint64_t *ak = args.ak[p], *av = args.av[p];
+    const W *aw = (const W *)args.aw[p];
+    const uint64_t *bk = args.bk[p], *bv = args.bv[p];
+    const W *bw = (const W *)args.bw[p];
+    uint64_t *ok = args.ok[p], *ov = args.ov[p];
+    W *ow = (W *)args.ow[p];
+    int64_t base = 0, all = 0;
+    for (int g = 0; g < MERGE_MID_WGS; g++) {
+        if (g < wg) base += cnt_row[g];
+        all += cnt_row[g];
+    }
+    if (threadIdx.x == 0 && wg == 0) *out_len = all;
+    const int64_t total = na + nb;
+    const int64_t per_wg = (total + MERGE_MID_WGS - 1) / MERGE_MID_WGS;
+    const int64_t w0 = min((int64_t)wg * per_wg, total);
+    const int64_t w1 = min(w0 + per_wg, total);
+    const int64_t items = (w1 - w0 + FUSE_THREADS - 1) / FUSE_THREADS;
+    int64_t d0 = min(w0 + (int64_t)threadIdx.x * items, w1);
+    int64_t d1 = min(d0 + items, w1);
+    int64_t ai, bi, ae, be;
+    merge_path(ak, av, na, bk, bv, nb, d0, ai, bi);
+    adjust_split(ak, av, bk, bv, na, nb, ai, bi);
+    merge_path(ak, av, na, bk, bv, nb, d1, ae, be);
+    adjust_split(ak, av, bk, bv, na, nb, ae, be);
+    uint32_t cnt = 0;
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(ak[i], av[i], bk[j], bv[j])) {
+                if (aw[i] + bw[j] != (W)0) cnt++;
+                i++; j++;
+            } else if (j >= be || (i < ae && row_lt(ak[i], av[i], bk[j], bv[j]))) {
+                cnt++; i++;
+            } else {
+                cnt++; j++;
+            }
+        }
+    }
+    __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
+    uint32_t seg_tot;
+    uint32_t off32 = fuse_scan(cnt, wave_tot, &seg_tot);
+    int64_t off = base + (int64_t)off32;
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(ak[i], av[i], bk[j], bv[j])) {
+                W s = aw[i] + bw[j];
+                if (s != (W)0) { ok[off] = ak[i]; ov[off] = av[i]; ow[off] = s; off++; }
+                i++; j++;
+            } else if (j >= be || (i < ae && row_lt(ak[i], av[i], bk[j], bv[j]))) {
+                ok[off] = ak[i]; ov[off] = av[i]; ow[off] = aw[i]; off++; i++;
+            } else {
+                ok[off] = bk[j]; ov[off] = bv[j]; ow[off] = bw[j]; off++; j++;
+            }
+        }
+    }
 }
 
 // ---------------------------------------------------------------------------
@@ -2883,6 +3018,15 @@ dbsp_status merge_small_batch_f64(hipStream_t s, const MergeArgs &args) {
     for (int i = 0; i < args.np; i++)
         if (args.na[i] + args.nb[i] > 4 * FUSE_MAX) return DBSP_ERR_INVALID;
     k_merge_small<double><<<dim3((uint32_t)args.np), FUSE_THREADS, 0, s>>>(args);
+    return DBSP_OK;
+}
+
+dbsp_status merge_mid_batch(hipStream_t s, const MergeArgs &args,
+                            int64_t *scratch) {
+    if (args.np == 0) return DBSP_OK;
+    dim3 grid(MERGE_MID_WGS, (uint32_t)args.np);
+    k_merge_mid_count<int64_t><<<grid, FUSE_THREADS, 0, s>>>(args, scratch);
+    k_merge_mid_emit<int64_t><<<grid, FUSE_THREADS, 0, s>>>(args, scratch);
     return DBSP_OK;
 }
 

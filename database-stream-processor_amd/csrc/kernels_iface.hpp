@@ -48,6 +48,10 @@ struct MergeArgs {
     const uint64_t *bv[MERGE_BATCH_MAX];
     const int64_t *bw[MERGE_BATCH_MAX];
     int64_t nb[MERGE_BATCH_MAX];
+    // optional device-side b length (the in-train accumulator merge: b is the
+    // tick's delta, whose consolidated length lives in a d_len slot); null =>
+    // use nb.  A negative device length (failed speculation) => emit -1.
+    const int64_t *dnb[MERGE_BATCH_MAX];
     uint64_t *ok[MERGE_BATCH_MAX];
     uint64_t *ov[MERGE_BATCH_MAX];
     int64_t *ow[MERGE_BATCH_MAX];
@@ -217,6 +221,11 @@ dbsp_status sort_cons_dense_chain(hipStream_t s, const uint64_t *k,
 // single-workgroup merge of two consolidated batches (na+nb <= 8192):
 // one launch, no host sync; length left in *d_len (device)
 dbsp_status merge_small_batch(hipStream_t s, const MergeArgs &args);
+// fixed-grid multi-WG merge with optional device-side b lengths (in-train
+// accumulator fold); scratch holds np * (MERGE_MID_SCRATCH) int64 slots
+#define MERGE_MID_SCRATCH 17
+dbsp_status merge_mid_batch(hipStream_t s, const MergeArgs &args,
+                            int64_t *scratch);
 
 // up to 3 single-workgroup join count+scan plans in ONE launch (nd <= 8192
 // each): per-plan per-row/per-batch cnts, exclusive offsets, totals to
