@@ -48,7 +48,13 @@
 
 namespace dbspk {
 namespace {
-constexpr size_t CACHE_MIN = 32u << 20;
+// 64 KB: originally only trace-scale buffers (>= 32 MB) were cached, but the
+// pipelined tick makes ~12 pool alloc/free calls per tick for ~0.1-3 MB
+// buffers, and those hipMallocAsync round-trips (~2-4 us each) were the
+// largest remaining host cost at the tick boundary — the free-list reuse is
+// ~0.2 us.  Footprint cost is bounded by the pow2 rounding (<= 2x per size
+// class) against 288 GB of HBM.
+constexpr size_t CACHE_MIN = 64u << 10;
 std::mutex g_cache_mu;
 std::unordered_map<void *, size_t> g_cache_live;
 std::multimap<size_t, void *> g_cache_free;
@@ -70,8 +76,10 @@ hipError_t cache_malloc(void **out, size_t bytes, hipStream_t s) {
         if (msb < bytes) bytes = msb << 1;
         std::lock_guard<std::mutex> g(g_cache_mu);
         auto it = g_cache_free.lower_bound(bytes);
-        if (it != g_cache_free.end() &&
-            it->first <= 2 * bytes + (64u << 20)) {
+        // small classes accept only <= 2x (a 64 MB slack there would let a
+        // trace-scale block be wasted on a tick-scale request)
+        const size_t slack = bytes >= (32u << 20) ? (64u << 20) : 0;
+        if (it != g_cache_free.end() && it->first <= 2 * bytes + slack) {
             *out = it->second;
             g_cache_live.emplace(it->second, it->first);
             g_cache_free.erase(it);
