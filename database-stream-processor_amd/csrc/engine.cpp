@@ -3845,6 +3845,10 @@ extern "C" dbsp_status dbsp_engine_c5_init(dbsp_engine *e, int64_t n_trace,
     if (!e || e->query != 100 || n_trace <= 0 || n_delta <= 0)
         return DBSP_ERR_INVALID;
     dbsp_ctx *c = e->ctx;
+    // trace-scale workload: keep tick-scale buffers on the stream-ordered
+    // pool so it can serve the per-tick multi-GB spine carves from retained
+    // slabs (see cache_small_set)
+    dbspk::cache_small_set(false, c->stream);
     e->c5_trace.clear(c);
     e->c5_wint.clear(c);
     e->c5_out.clear(c);

@@ -42,6 +42,7 @@
 // tick-scale buffers fine.  On allocation failure the cache is dropped and
 // the carve retried.
 // ---------------------------------------------------------------------------
+#include <atomic>
 #include <map>
 #include <mutex>
 #include <unordered_map>
@@ -54,7 +55,31 @@ namespace {
 // largest remaining host cost at the tick boundary — the free-list reuse is
 // ~0.2 us.  Footprint cost is bounded by the pow2 rounding (<= 2x per size
 // class) against 288 GB of HBM.
-constexpr size_t CACHE_MIN = 64u << 10;
+// Two cached bands: tick-scale blocks (64 KB .. small-ceiling) whose pool
+// round-trips dominate the pipelined tick's host cost, and trace-scale
+// blocks (>= 32 MB) whose pool re-carves cost ~100 ms/GB.  The band between
+// stays on the stream-ordered pool: C5-scale ticks allocate many MB-range
+// buffers and measured 2x slower with them free-listed.
+const size_t CACHE_MIN = []() -> size_t {
+    const char *v = getenv("DBSP_CACHE_MIN");  // diagnostic A/B knob
+    return v ? (size_t)atoll(v) : (64u << 10);
+}();
+const size_t CACHE_SMALL_CEIL = []() -> size_t {
+    const char *v = getenv("DBSP_CACHE_SMALL_CEIL");
+    return v ? (size_t)atoll(v) : (4u << 20);
+}();
+constexpr size_t CACHE_BIG_MIN = 32u << 20;
+// Small-band caching is SELF-DISABLING: if any carve fails (trace-scale
+// workloads near the memory limit — the hoarded small blocks keep the pool
+// from satisfying a multi-GB request), the failure path drops the free list
+// and turns the small band off for the rest of the process, so the cost is
+// one sync, not one per tick.  Tick-scale engines never fail and keep it.
+std::atomic<bool> g_small_cache_on{true};
+inline bool cacheable(size_t bytes) {
+    if (bytes >= CACHE_BIG_MIN) return true;
+    return bytes >= CACHE_MIN && bytes < CACHE_SMALL_CEIL &&
+           g_small_cache_on.load(std::memory_order_relaxed);
+}
 std::mutex g_cache_mu;
 std::unordered_map<void *, size_t> g_cache_live;
 std::multimap<size_t, void *> g_cache_free;
@@ -66,7 +91,7 @@ void cache_drop_locked(hipStream_t s) {
 }  // namespace
 
 hipError_t cache_malloc(void **out, size_t bytes, hipStream_t s) {
-    if (bytes >= CACHE_MIN) {
+    if (cacheable(bytes)) {
         // round carves up to the next power of two and accept cached blocks
         // up to 2x the request: a workload with growing buffers (spine
         // merges, trace cascades) then carves only log2(max/min) times
@@ -86,14 +111,51 @@ hipError_t cache_malloc(void **out, size_t bytes, hipStream_t s) {
             return hipSuccess;
         }
     }
+    if (bytes >= CACHE_BIG_MIN) {
+        // big-band carve: flush the small band back to the pool first —
+        // stream-ordered frees are reusable by this very carve, and a
+        // hoarded small band otherwise forces the pool to map fresh pages
+        // per multi-GB request (~120 ms each at trace scale).  Tick-scale
+        // engines do no big carves in steady state, so their band persists.
+        std::lock_guard<std::mutex> g(g_cache_mu);
+        for (auto it = g_cache_free.begin();
+             it != g_cache_free.end() && it->first < CACHE_BIG_MIN;) {
+            (void)hipFreeAsync(it->second, s);
+            it = g_cache_free.erase(it);
+        }
+    }
+    static const bool cstats = []() {
+        const char *v = getenv("DBSP_CACHE_STATS");
+        return v && v[0] == '1';
+    }();
+    double tc0 = 0;
+    if (cstats && bytes >= CACHE_BIG_MIN) {
+        struct timespec ts;
+        clock_gettime(CLOCK_MONOTONIC, &ts);
+        tc0 = ts.tv_sec * 1e3 + ts.tv_nsec * 1e-6;
+    }
     hipError_t e = hipMallocAsync(out, bytes, s);
+    if (cstats && bytes >= CACHE_BIG_MIN) {
+        struct timespec ts;
+        clock_gettime(CLOCK_MONOTONIC, &ts);
+        double dt = ts.tv_sec * 1e3 + ts.tv_nsec * 1e-6 - tc0;
+        size_t nsm = 0, live = 0;
+        {
+            std::lock_guard<std::mutex> g(g_cache_mu);
+            for (auto &kv : g_cache_live)
+                if (kv.second < CACHE_BIG_MIN) { nsm++; live += kv.second; }
+        }
+        fprintf(stderr, "[carve] %.1f MB in %.1f ms (small live: %zu blks %.1f MB)\n",
+                bytes / 1048576.0, dt, nsm, live / 1048576.0);
+    }
     if (e != hipSuccess) {
+        g_small_cache_on.store(false, std::memory_order_relaxed);
         std::lock_guard<std::mutex> g(g_cache_mu);
         cache_drop_locked(s);
         (void)hipStreamSynchronize(s);
         e = hipMallocAsync(out, bytes, s);
     }
-    if (e == hipSuccess && bytes >= CACHE_MIN) {
+    if (e == hipSuccess && cacheable(bytes)) {
         std::lock_guard<std::mutex> g(g_cache_mu);
         g_cache_live.emplace(*out, bytes);
     }
@@ -106,9 +168,15 @@ hipError_t cache_free(void *p, hipStream_t s) {
         std::lock_guard<std::mutex> g(g_cache_mu);
         auto it = g_cache_live.find(p);
         if (it != g_cache_live.end()) {
-            g_cache_free.emplace(it->second, p);
+            const size_t sz = it->second;
             g_cache_live.erase(it);
-            return hipSuccess;
+            if (sz < CACHE_BIG_MIN &&
+                !g_small_cache_on.load(std::memory_order_relaxed)) {
+                // small band disabled: return to the pool instead of hoarding
+            } else {
+                g_cache_free.emplace(sz, p);
+                return hipSuccess;
+            }
         }
     }
     return hipFreeAsync(p, s);
@@ -117,6 +185,25 @@ hipError_t cache_free(void *p, hipStream_t s) {
 void cache_trim(hipStream_t s) {
     std::lock_guard<std::mutex> g(g_cache_mu);
     cache_drop_locked(s);
+}
+
+// Explicit policy switch: trace-scale engines (C5's 1B-row spine) turn the
+// small band OFF at init — measured on MI355X, routing their tick-scale
+// churn away from the stream-ordered pool leaves the pool unable to serve
+// the per-tick multi-GB spine carves from retained slabs (0.3 ms -> 240 ms
+// per carve), tripling the step.  Tick-scale engines keep it on.
+void cache_small_set(bool on, hipStream_t s) {
+    if (getenv("DBSP_CACHE_STATS"))
+        fprintf(stderr, "[cache] small band %s\n", on ? "on" : "off");
+    g_small_cache_on.store(on, std::memory_order_relaxed);
+    if (!on) {
+        std::lock_guard<std::mutex> g(g_cache_mu);
+        for (auto it = g_cache_free.begin();
+             it != g_cache_free.end() && it->first < CACHE_BIG_MIN;) {
+            (void)hipFreeAsync(it->second, s);
+            it = g_cache_free.erase(it);
+        }
+    }
 }
 }  // namespace dbspk
 

@@ -99,6 +99,9 @@ namespace dbspk {
 hipError_t cache_malloc(void **out, size_t bytes, hipStream_t s);
 hipError_t cache_free(void *p, hipStream_t s);
 void cache_trim(hipStream_t s);
+// scope the tick-scale (64 KB..4 MB) free-list band: trace-scale engines
+// turn it off at init (see the definition for the measured pool interaction)
+void cache_small_set(bool on, hipStream_t s);
 }  // namespace dbspk
 
 namespace dbspk {
