@@ -1065,7 +1065,7 @@ struct Q3Train {
     int next_sb = 18, next_evi = 0;
     DevBatch rawA, rawP, oA, oP, comb_chain;
     // in-train accumulator merge results (acc side + this tick's delta side);
-    // lengths land in d_len[40+2*evi .. 41+2*evi] with the first readback
+    // lengths land in d_len[sb+16, sb+17] with the train's 18-slot readback
     DevBatch res[2];
     bool has_res = false;
     Q3Plan plans[3];
@@ -2658,7 +2658,10 @@ static dbsp_status q3_enqueue_train(dbsp_engine *e, const dbsp_event *d_ev,
     // round that left the GPU idle ~45 us at every tick boundary.
     {
         MergeArgs ma{};
-        const int slot_base = 40 + 2 * evi;
+        // result lengths ride in sb+16/17 (the framed-exchange totals slots,
+        // unused on the unsharded train path) so the train's single 18-slot
+        // readback carries them — no extra D2H
+        const int slot_base = sb + 16;
         for (int s = 0; s < 2; s++) {
             DevBatch &acc = e->q3_acc[s];
             DevBatch &delta = s == 0 ? T.oA : T.oP;
@@ -2675,10 +2678,9 @@ static dbsp_status q3_enqueue_train(dbsp_engine *e, const dbsp_event *d_ev,
             ma.np++;
         }
         ma.d_len = c->d_len + slot_base;
+        ScopedTimer timer(c, 1, (double)(e->q3_acc[0].n + e->q3_acc[1].n +
+                                         2 * n) * 48.0);
         TRY(dbspk::merge_mid_batch(c->stream, ma, c->d_mid));
-        HIP_CHECK_ST(hipMemcpyAsync(c->h_len + slot_base, c->d_len + slot_base,
-                                    2 * sizeof(int64_t),
-                                    hipMemcpyDeviceToHost, c->stream));
         T.has_res = true;
     }
     int jca_np = 0;
@@ -2799,13 +2801,12 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
     // few microseconds of the wake.
     TRY(resolve_pending_insert(c));
     constexpr int64_t Q3_ACC_SPILL = 32768;
-    const int slot_base = 40 + 2 * T.evi;
     Spine *spill[2];
     int nspill = 0;
     for (int s = 0; s < 2; s++) {
         Spine &sp = s == 0 ? e->a_int : e->p_int;
         DevBatch res = T.res[s];
-        res.n = c->h_len[slot_base + s];
+        res.n = H[16 + s];
         DevBatch acc = e->q3_acc[s];
         if (acc.n > 0) {  // the top batch is the consumed accumulator
             sp.batches.pop_back();

@@ -1713,7 +1713,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_small(MergeArgs args)
 // array (no separate scan launch) and writes its segment.  Equal (k, v)
 // pairs never split across WGs (adjust_split), so per-segment consolidation
 // is exact.
-#define MERGE_MID_WGS 16
+#define MERGE_MID_WGS 32
 static_assert(MERGE_MID_SCRATCH == MERGE_MID_WGS + 1,
               "scratch layout mismatch with kernels_iface.hpp");
 

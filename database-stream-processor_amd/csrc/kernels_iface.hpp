@@ -226,7 +226,7 @@ dbsp_status sort_cons_dense_chain(hipStream_t s, const uint64_t *k,
 dbsp_status merge_small_batch(hipStream_t s, const MergeArgs &args);
 // fixed-grid multi-WG merge with optional device-side b lengths (in-train
 // accumulator fold); scratch holds np * (MERGE_MID_SCRATCH) int64 slots
-#define MERGE_MID_SCRATCH 17
+#define MERGE_MID_SCRATCH 33
 dbsp_status merge_mid_batch(hipStream_t s, const MergeArgs &args,
                             int64_t *scratch);
 
