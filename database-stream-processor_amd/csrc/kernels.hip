@@ -609,8 +609,8 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     W *ow = (W *)args.ow[batch];
     int64_t *out_len = args.d_len + batch;
     const int64_t n_chain = args.n_dev[batch] ? *args.n_dev[batch] : -1;
-    __shared__ uint32_t bufA[FUSE_MAX];
-    __shared__ uint32_t bufB[FUSE_MAX];
+    __shared__ alignas(16) uint32_t bufA[FUSE_MAX];
+    __shared__ alignas(16) uint32_t bufB[FUSE_MAX];
     __shared__ uint32_t cnt[FUSE_MAX];  // ballot cells / f64 head positions
     __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
     __shared__ uint64_t smax[2];
@@ -632,6 +632,60 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
         return;
     }
 
+    const int rounds = (int)((n + FUSE_THREADS - 1) / FUSE_THREADS);  // <= 8
+    const int nwaves = FUSE_THREADS / WAVE;
+    uint32_t *src = bufA, *dst = bufB;
+
+    if (n <= 2048) {
+        // ---- bitonic fast path ----
+        // Sub-2k tick batches (delta sides, tick outputs) are where the
+        // radix path is pass-count-bound (~35 us regardless of n: ~16+
+        // digit passes of block-wide ballots and syncs).  A stable LDS
+        // bitonic over (k, v, idx) — idx as tiebreak keeps positions
+        // identical to the stable radix, so the f64 position-ordered sums
+        // stay bit-equal — runs them in ~10-20 us.  Pair-indexed so every
+        // thread does a real compare.  bufA/bufB are reused as the u64 key
+        // arrays and cnt as the index payload; the result lands in src in
+        // the radix entry layout (idx in the low 13 bits) so the
+        // consolidation below is shared by both paths.  Beyond 2k the pass
+        // count (log^2) overtakes the radix and the radix path stays.
+        uint64_t *lk = (uint64_t *)bufA;
+        uint64_t *lv = (uint64_t *)bufB;
+        uint32_t *lidx = cnt;
+        int64_t p2 = 1;
+        while (p2 < n) p2 <<= 1;
+        for (int64_t i = tid; i < p2; i += FUSE_THREADS) {
+            lk[i] = i < n ? kin[i] : ~0ull;
+            lv[i] = i < n ? vin[i] : ~0ull;
+            lidx[i] = (uint32_t)i;
+        }
+        __syncthreads();
+        const int64_t npairs = p2 >> 1;
+        for (int64_t kk = 2; kk <= p2; kk <<= 1) {
+            for (int64_t jj = kk >> 1, s = 63 - __clzll(kk >> 1); jj > 0;
+                 jj >>= 1, s--) {
+                for (int64_t p = tid; p < npairs; p += FUSE_THREADS) {
+                    const int64_t i = ((p >> s) << (s + 1)) | (p & (jj - 1));
+                    const int64_t l = i + jj;
+                    const bool up = (i & kk) == 0;
+                    const uint64_t ka = lk[i], kb = lk[l];
+                    const uint64_t va = lv[i], vb = lv[l];
+                    const uint32_t ia = lidx[i], ib = lidx[l];
+                    const bool gt =
+                        ka > kb ||
+                        (ka == kb && (va > vb || (va == vb && ia > ib)));
+                    if (gt == up) {
+                        lk[i] = kb; lk[l] = ka;
+                        lv[i] = vb; lv[l] = va;
+                        lidx[i] = ib; lidx[l] = ia;
+                    }
+                }
+                __syncthreads();
+            }
+        }
+        for (int64_t i = tid; i < n; i += FUSE_THREADS) src[i] = lidx[i];
+        __syncthreads();
+    } else {
     // ---- significant bits of (max-min) for k and v ----
     if (tid == 0) {
         smax[0] = 0; smax[1] = 0;
@@ -659,8 +713,6 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     while (vnibs < 16 && (vrange >> (4 * vnibs)) != 0) vnibs++;
     const int total_nibs = vnibs + knibs;
     const int vbits = 4 * vnibs;
-    const int rounds = (int)((n + FUSE_THREADS - 1) / FUSE_THREADS);  // <= 8
-    const int nwaves = FUSE_THREADS / WAVE;
 
     // bits [s, s+16) of the concatenated key ((k-kbase) << vbits | (v-vbase))
     auto key16_at = [&](uint32_t idx, int s) -> uint32_t {
@@ -678,7 +730,6 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
     };
 
     // ---- digit passes in groups of 4 sharing one entry packing ----
-    uint32_t *src = bufA, *dst = bufB;
     for (int base = 0; base < total_nibs; base += 4) {
         // (re)pack: entry = next-16-key-bits << 13 | idx  (one global gather)
         for (int64_t i = tid; i < n; i += FUSE_THREADS) {
@@ -758,6 +809,7 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_sort_cons_small(SortArgs ar
         for (int64_t i = tid; i < n; i += FUSE_THREADS) src[i] = (uint32_t)i;
         __syncthreads();
     }
+    }  // end radix path
 
     // ---- consolidate: head flags (full-key compares via one gather) ----
     // seg ids into dst (the free buffer); ballot cells per (round, wave)
