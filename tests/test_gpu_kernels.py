@@ -62,6 +62,15 @@ def test_sort_consolidate_parity(ctx):
     z["v"] = 9
     z["w"] = np.where(np.arange(1000) % 2 == 0, 1, -1)
     cases.append(z)
+    # bitonic fast-path band (n <= 2048): pow2 boundaries, duplicate-heavy
+    # (consolidation across thread ownership), cancel-heavy (zero-weight
+    # drop), and wide 64-bit keys (the bitonic compares full (k, v))
+    for n in (2, 3, 1023, 1024, 1025, 2047):
+        cases.append(_random_rows(rng, n, key_range=2**63, val_range=2**63))
+    dup = _random_rows(rng, 1500, key_range=40, val_range=3)
+    cases.append(dup)
+    canc = _random_rows(rng, 2000, key_range=50, val_range=2, w_range=1)
+    cases.append(canc)
     for r in cases:
         got = ctx.sort_consolidate(r)
         exp = oracle.consolidate(r)
