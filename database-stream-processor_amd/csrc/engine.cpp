@@ -104,6 +104,7 @@ struct dbsp_ctx {
     void *pend_spine[4] = {};
     uint64_t *pend_k[4] = {}, *pend_v[4] = {};
     int64_t *pend_w[4] = {};
+    int pend_slot[4] = {};  // h_len slot (36+s) holding each result length
 };
 
 static void *arena_alloc(dbsp_ctx *c, size_t bytes) {
@@ -162,7 +163,7 @@ extern "C" dbsp_status dbsp_ctx_create(dbsp_ctx **out, int device) {
     const char *fs = getenv("DBSP_FORCE_SHARD");
     c->force_shard = fs && fs[0] == '1';
     HIP_CHECK_ST(hipMalloc(&c->d_len, 44 * sizeof(int64_t)));
-    HIP_CHECK_ST(hipMalloc(&c->d_mid, 2 * MERGE_MID_SCRATCH * sizeof(int64_t)));
+    HIP_CHECK_ST(hipMalloc(&c->d_mid, 4 * MERGE_MID_SCRATCH * sizeof(int64_t)));
     c->arena_sz = (size_t)512 << 20;
     c->arena_half = c->arena_sz / 2;
     if (hipMalloc(&c->arena, c->arena_sz) != hipSuccess) {
@@ -340,30 +341,45 @@ static dbsp_status sort_medium(dbsp_ctx *c, DevBatch raw, DevBatch &out) {
     std::vector<DevBatch> cur(chunks.begin(), chunks.end());
     while (cur.size() > 1) {
         std::vector<DevBatch> next;
-        MergeArgs ma{};
+        MergeArgs ma{};   // tiny pairs (<= 4096): single-WG
+        MergeArgs mm{};   // mid pairs (<= 32768): fixed-grid multi-WG
         std::vector<DevBatch> results;
+        std::vector<int> slots;  // h_len slot per result (smalls then mids)
+        std::vector<bool> ismid;
         size_t i = 0;
-        for (; i + 1 < cur.size() && ma.np < MERGE_BATCH_MAX; i += 2) {
+        for (; i + 1 < cur.size() &&
+               (int)results.size() < MERGE_BATCH_MAX; i += 2) {
             DevBatch &a = cur[i], &b = cur[i + 1];
             if (a.n + b.n > 32768) break;  // fall through to merge_batches
             DevBatch res;
             bool final_round = (cur.size() == 2);
             TRY(alloc_batch(c, a.n + b.n, res, !final_round));
-            int p = ma.np++;
-            ma.ak[p] = a.k; ma.av[p] = a.v; ma.aw[p] = a.w; ma.na[p] = a.n;
-            ma.bk[p] = b.k; ma.bv[p] = b.v; ma.bw[p] = b.w; ma.nb[p] = b.n;
-            ma.ok[p] = res.k; ma.ov[p] = res.v; ma.ow[p] = res.w;
+            MergeArgs &mx = a.n + b.n <= 4096 ? ma : mm;
+            int p = mx.np++;
+            mx.ak[p] = a.k; mx.av[p] = a.v; mx.aw[p] = a.w; mx.na[p] = a.n;
+            mx.bk[p] = b.k; mx.bv[p] = b.v; mx.bw[p] = b.w; mx.nb[p] = b.n;
+            mx.ok[p] = res.k; mx.ov[p] = res.v; mx.ow[p] = res.w;
+            ismid.push_back(&mx == &mm);
             results.push_back(res);
         }
-        if (ma.np > 0) {
+        if (!results.empty()) {
+            {
+                int ps = 0, pm = ma.np;
+                for (bool m : ismid) slots.push_back(m ? pm++ : ps++);
+            }
             ma.d_len = c->d_len;
-            TRY(dbspk::merge_small_batch(c->stream, ma));
+            mm.d_len = c->d_len + ma.np;
+            if (ma.np > 0) TRY(dbspk::merge_small_batch(c->stream, ma));
+            if (mm.np > 0)
+                TRY(dbspk::merge_mid_batch(c->stream, mm, c->d_mid));
             HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len,
-                                        ma.np * sizeof(int64_t),
+                                        (ma.np + mm.np) * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
             HIP_CHECK_ST(hipStreamSynchronize(c->stream));
-            for (int p = 0; p < ma.np; p++) results[p].n = c->h_len[p];
-            for (size_t j = 0; j < (size_t)(2 * ma.np); j++) free_batch(c, cur[j]);
+            for (size_t p = 0; p < results.size(); p++)
+                results[p].n = c->h_len[slots[p]];
+            for (size_t j = 0; j < 2 * results.size(); j++)
+                free_batch(c, cur[j]);
             next.insert(next.end(), results.begin(), results.end());
         }
         // remaining (odd leftover or >32k pairs) via the generic path
@@ -449,11 +465,11 @@ static dbsp_status sort_consolidate_batch(dbsp_ctx *c, DevBatch raw, DevBatch &o
 static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
                                  const DevBatch &b, DevBatch &out) {
     ScopedTimer t(c, 1, (double)(a.n + b.n) * 48.0);
-    // single-WG cutoff: beyond ~16k rows one workgroup serializes ~126 us
-    // (q4 trace) while the multi-block two-pass costs ~40 us; the host-length
-    // path here syncs anyway, so the pipeline split is free
-    if (a.n + b.n <= 16384) {
-        // one launch + one length readback (merge-path diagonals in one WG)
+    // three bands: tiny pairs (<= 4k) in one single-WG launch; mid pairs
+    // (<= 32k) in the fixed-grid multi-WG merge (a 16k single-WG merge
+    // serializes ~126 us — q4's bid spine — where merge_mid takes ~12);
+    // larger through the multi-block two-pass
+    if (a.n + b.n <= 32768) {
         DevBatch res;
         TRY(alloc_batch(c, a.n + b.n, res));
         MergeArgs ma{};
@@ -462,7 +478,10 @@ static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
         ma.bk[0] = b.k; ma.bv[0] = b.v; ma.bw[0] = b.w; ma.nb[0] = b.n;
         ma.ok[0] = res.k; ma.ov[0] = res.v; ma.ow[0] = res.w;
         ma.d_len = c->d_len;
-        TRY(dbspk::merge_small_batch(c->stream, ma));
+        if (a.n + b.n <= 4096)
+            TRY(dbspk::merge_small_batch(c->stream, ma));
+        else
+            TRY(dbspk::merge_mid_batch(c->stream, ma, c->d_mid));
         HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -489,7 +508,10 @@ static dbsp_status merge_batches_f64(dbsp_ctx *c, const DevBatch &a,
         ma.bk[0] = b.k; ma.bv[0] = b.v; ma.bw[0] = b.w; ma.nb[0] = b.n;
         ma.ok[0] = res.k; ma.ov[0] = res.v; ma.ow[0] = res.w;
         ma.d_len = c->d_len;
-        TRY(dbspk::merge_small_batch_f64(c->stream, ma));
+        if (a.n + b.n <= 4096)
+            TRY(dbspk::merge_small_batch_f64(c->stream, ma));
+        else
+            TRY(dbspk::merge_mid_batch_f64(c->stream, ma, c->d_mid));
         HIP_CHECK_ST(hipMemcpyAsync(c->h_len, c->d_len, sizeof(int64_t),
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
@@ -2038,23 +2060,34 @@ static dbsp_status spine_rounds(dbsp_ctx *c, Spine *const *sps, int ns,
             if (!dup && spine_needs_merge(*sps[i])) pending[nps++] = sps[i];
         }
         if (nps == 0) break;
-        MergeArgs ma{};
+        // two batched launch classes by pair size: tiny pairs (<= 4096
+        // rows) go to the single-WG k_merge_small; mid pairs (<= 32768) to
+        // the fixed-grid multi-WG merge_mid — a 32k single-WG merge costs
+        // ~126 us where merge_mid does it in ~12 (q4's bid spine pushes
+        // ~37k rows/tick, so these dominate its tick)
+        MergeArgs ma{};   // small
+        MergeArgs mm{};   // mid
         DevBatch results[4];
         Spine *owners[4];
+        int slotofj[4];
+        bool midofj[4];
         int nbatched = 0;
         for (int i = 0; i < nps; i++) {
             Spine &sp = *pending[i];
             DevBatch &top = sp.batches.back();
             DevBatch &below = sp.batches[sp.batches.size() - 2];
-            if (top.n + below.n <= 32768 && ma.np < MERGE_BATCH_MAX) {
+            const int64_t tot = top.n + below.n;
+            if (tot <= 32768 && nbatched < MERGE_BATCH_MAX) {
                 DevBatch res;
-                TRY(alloc_batch(c, top.n + below.n, res));
-                int p = ma.np++;
-                ma.ak[p] = below.k; ma.av[p] = below.v; ma.aw[p] = below.w;
-                ma.na[p] = below.n;
-                ma.bk[p] = top.k; ma.bv[p] = top.v; ma.bw[p] = top.w;
-                ma.nb[p] = top.n;
-                ma.ok[p] = res.k; ma.ov[p] = res.v; ma.ow[p] = res.w;
+                TRY(alloc_batch(c, tot, res));
+                MergeArgs &mx = tot <= 4096 ? ma : mm;
+                int p = mx.np++;
+                mx.ak[p] = below.k; mx.av[p] = below.v; mx.aw[p] = below.w;
+                mx.na[p] = below.n;
+                mx.bk[p] = top.k; mx.bv[p] = top.v; mx.bw[p] = top.w;
+                mx.nb[p] = top.n;
+                mx.ok[p] = res.k; mx.ov[p] = res.v; mx.ow[p] = res.w;
+                midofj[nbatched] = tot > 4096;
                 results[nbatched] = res;
                 owners[nbatched] = &sp;
                 nbatched++;
@@ -2069,15 +2102,23 @@ static dbsp_status spine_rounds(dbsp_ctx *c, Spine *const *sps, int ns,
                 else free_batch(c, res);
             }
         }
-        if (ma.np > 0) {
+        if (nbatched > 0) {
             // dedicated insert slots (36-39): the hook below may enqueue the
             // next tick's full train, whose 18-slot readback would otherwise
             // overwrite h_len[0..3] before this round's merge lengths are
-            // consumed after the event wait
+            // consumed after the event wait.  Slots: small pairs first,
+            // then mid pairs.
+            {
+                int ps = 0, pm = ma.np;
+                for (int j = 0; j < nbatched; j++)
+                    slotofj[j] = midofj[j] ? pm++ : ps++;
+            }
             ma.d_len = c->d_len + 36;
-            TRY(dbspk::merge_small_batch(c->stream, ma));
+            mm.d_len = c->d_len + 36 + ma.np;
+            if (ma.np > 0) TRY(dbspk::merge_small_batch(c->stream, ma));
+            if (mm.np > 0) TRY(dbspk::merge_mid_batch(c->stream, mm, c->d_mid));
             HIP_CHECK_ST(hipMemcpyAsync(c->h_len + 36, c->d_len + 36,
-                                        ma.np * sizeof(int64_t),
+                                        (ma.np + mm.np) * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
             if (defer && hook && !*hook_fired) {
                 (void)hipEventRecord(c->ev_insert, c->stream);
@@ -2089,6 +2130,7 @@ static dbsp_status spine_rounds(dbsp_ctx *c, Spine *const *sps, int ns,
                     c->pend_k[j] = results[j].k;
                     c->pend_v[j] = results[j].v;
                     c->pend_w[j] = results[j].w;
+                    c->pend_slot[j] = slotofj[j];
                 }
                 return DBSP_OK;  // pops + cascades resolved next tick
             }
@@ -2102,7 +2144,7 @@ static dbsp_status spine_rounds(dbsp_ctx *c, Spine *const *sps, int ns,
             }
             for (int j = 0; j < nbatched; j++) {
                 Spine &sp = *owners[j];
-                results[j].n = c->h_len[36 + j];
+                results[j].n = c->h_len[36 + slotofj[j]];
                 DevBatch top = sp.batches.back();
                 sp.batches.pop_back();
                 DevBatch below = sp.batches.back();
@@ -2131,7 +2173,7 @@ static dbsp_status resolve_pending_insert(dbsp_ctx *c) {
     for (int j = 0; j < np; j++) {
         Spine &sp = *(Spine *)c->pend_spine[j];
         DevBatch res{c->pend_k[j], c->pend_v[j], c->pend_w[j],
-                     c->h_len[36 + j]};
+                     c->h_len[36 + c->pend_slot[j]]};
         DevBatch top = sp.batches.back();
         sp.batches.pop_back();
         DevBatch below = sp.batches.back();

@@ -3177,6 +3177,15 @@ dbsp_status merge_mid_batch(hipStream_t s, const MergeArgs &args,
     return DBSP_OK;
 }
 
+dbsp_status merge_mid_batch_f64(hipStream_t s, const MergeArgs &args,
+                                int64_t *scratch) {
+    if (args.np == 0) return DBSP_OK;
+    dim3 grid(MERGE_MID_WGS, (uint32_t)args.np);
+    k_merge_mid_count<double><<<grid, FUSE_THREADS, 0, s>>>(args, scratch);
+    k_merge_mid_emit<double><<<grid, FUSE_THREADS, 0, s>>>(args, scratch);
+    return DBSP_OK;
+}
+
 dbsp_status join_spine_rows(hipStream_t s, const uint64_t *dk,
                             const uint64_t *dv, const int64_t *dw, int64_t nd,
                             const TraceArgs &t, int proj, uint64_t param,

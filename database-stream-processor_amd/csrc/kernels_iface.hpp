@@ -229,6 +229,8 @@ dbsp_status merge_small_batch(hipStream_t s, const MergeArgs &args);
 #define MERGE_MID_SCRATCH 33
 dbsp_status merge_mid_batch(hipStream_t s, const MergeArgs &args,
                             int64_t *scratch);
+dbsp_status merge_mid_batch_f64(hipStream_t s, const MergeArgs &args,
+                                int64_t *scratch);
 
 // up to 3 single-workgroup join count+scan plans in ONE launch (nd <= 8192
 // each): per-plan per-row/per-batch cnts, exclusive offsets, totals to
