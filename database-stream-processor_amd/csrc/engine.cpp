@@ -1089,7 +1089,6 @@ struct Q3Train {
     // in-train accumulator merge results (acc side + this tick's delta side);
     // lengths land in d_len[sb+16, sb+17] with the train's 18-slot readback
     DevBatch res[2];
-    bool has_res = false;
     Q3Plan plans[3];
     int np = 0;
     size_t arena_base = 0, arena_off = 0;
@@ -2673,7 +2672,6 @@ static dbsp_status q3_enqueue_train(dbsp_engine *e, const dbsp_event *d_ev,
         free_batch(c, T.res[1]);
         T.oA = DevBatch{};
         T.oP = DevBatch{};
-        T.has_res = false;
         c->arena_base = save_base;
         c->arena_off = save_off;
         T.pending = false;
@@ -2723,7 +2721,6 @@ static dbsp_status q3_enqueue_train(dbsp_engine *e, const dbsp_event *d_ev,
         ScopedTimer timer(c, 1, (double)(e->q3_acc[0].n + e->q3_acc[1].n +
                                          2 * n) * 48.0);
         TRY(dbspk::merge_mid_batch(c->stream, ma, c->d_mid));
-        T.has_res = true;
     }
     int jca_np = 0;
     bool arena_ok = true;
@@ -2798,7 +2795,6 @@ static dbsp_status q3_commit_train(dbsp_engine *e) {
     e->train.oP = DevBatch{};
     e->train.res[0] = DevBatch{};
     e->train.res[1] = DevBatch{};
-    e->train.has_res = false;
     // commit-side transients go to the half the train is NOT using
     c->arena_base =
         c->arena_half ? (T.arena_base ? 0 : c->arena_half) : c->arena_base;
@@ -2925,8 +2921,7 @@ static dbsp_status q3_step(dbsp_engine *e, const dbsp_event *d_ev, int64_t n) {
         free_batch(c, e->train.oP);
         free_batch(c, e->train.res[0]);
         free_batch(c, e->train.res[1]);
-        e->train.has_res = false;
-        e->train.pending = false;
+            e->train.pending = false;
     }
     const bool shard_chain =
         sharding_on(c) && c->world <= 8 && n <= 131072 && e->spec_fail < 3;
