@@ -3059,7 +3059,10 @@ dbsp_status minmax_rows(hipStream_t s, const uint64_t *k, const uint64_t *v,
     HIP_CHECK(dbspk::cache_malloc((void **)&d, 4 * sizeof(uint64_t), s));
     const unsigned long long init[4] = {~0ull, ~0ull, 0ull, 0ull};
     HIP_CHECK(hipMemcpyAsync(d, init, sizeof(init), hipMemcpyHostToDevice, s));
-    k_minmax_rows<<<grid_for(n), BLK, 0, s>>>(k, v, n, d);
+    // cap the grid: each block ends with 8 global atomics on 4 hot words,
+    // and a full grid_for() grid serializes ~15 us on them alone
+    k_minmax_rows<<<dim3(std::min(grid_for(n).x, 64u)), BLK, 0, s>>>(k, v, n,
+                                                                     d);
     HIP_CHECK(hipMemcpyAsync(mm, d, 4 * sizeof(uint64_t),
                              hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
@@ -3076,7 +3079,8 @@ dbsp_status minmax_rows_chain(hipStream_t s, const uint64_t *k,
     const unsigned long long init[4] = {~0ull, ~0ull, 0ull, 0ull};
     HIP_CHECK(hipMemcpyAsync(mm_dev, init, sizeof(init), hipMemcpyHostToDevice,
                              s));
-    k_minmax_rows<<<grid_for(cap), BLK, 0, s>>>(k, v, 0, mm_dev, n_dev);
+    k_minmax_rows<<<dim3(std::min(grid_for(cap).x, 64u)), BLK, 0, s>>>(
+        k, v, 0, mm_dev, n_dev);
     return DBSP_OK;
 }
 
