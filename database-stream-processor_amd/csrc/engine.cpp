@@ -376,8 +376,10 @@ static dbsp_status sort_medium(dbsp_ctx *c, DevBatch raw, DevBatch &out) {
                                         (ma.np + mm.np) * sizeof(int64_t),
                                         hipMemcpyDeviceToHost, c->stream));
             HIP_CHECK_ST(hipStreamSynchronize(c->stream));
-            for (size_t p = 0; p < results.size(); p++)
+            for (size_t p = 0; p < results.size(); p++) {
                 results[p].n = c->h_len[slots[p]];
+                if (results[p].n < 0) return DBSP_ERR_INTERNAL;
+            }
             for (size_t j = 0; j < 2 * results.size(); j++)
                 free_batch(c, cur[j]);
             next.insert(next.end(), results.begin(), results.end());
@@ -486,6 +488,7 @@ static dbsp_status merge_batches(dbsp_ctx *c, const DevBatch &a,
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
         res.n = *c->h_len;
+        if (res.n < 0) return DBSP_ERR_INTERNAL;  // fused-barrier poison
         out = res;
         return DBSP_OK;
     }
@@ -516,6 +519,7 @@ static dbsp_status merge_batches_f64(dbsp_ctx *c, const DevBatch &a,
                                     hipMemcpyDeviceToHost, c->stream));
         HIP_CHECK_ST(hipStreamSynchronize(c->stream));
         res.n = *c->h_len;
+        if (res.n < 0) return DBSP_ERR_INTERNAL;  // fused-barrier poison
         out = res;
         return DBSP_OK;
     }
@@ -2144,6 +2148,7 @@ static dbsp_status spine_rounds(dbsp_ctx *c, Spine *const *sps, int ns,
             for (int j = 0; j < nbatched; j++) {
                 Spine &sp = *owners[j];
                 results[j].n = c->h_len[36 + slotofj[j]];
+                if (results[j].n < 0) return DBSP_ERR_INTERNAL;
                 DevBatch top = sp.batches.back();
                 sp.batches.pop_back();
                 DevBatch below = sp.batches.back();
@@ -2173,6 +2178,7 @@ static dbsp_status resolve_pending_insert(dbsp_ctx *c) {
         Spine &sp = *(Spine *)c->pend_spine[j];
         DevBatch res{c->pend_k[j], c->pend_v[j], c->pend_w[j],
                      c->h_len[36 + c->pend_slot[j]]};
+        if (res.n < 0) return DBSP_ERR_INTERNAL;
         DevBatch top = sp.batches.back();
         sp.batches.pop_back();
         DevBatch below = sp.batches.back();

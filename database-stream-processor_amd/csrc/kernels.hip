@@ -1889,6 +1889,113 @@ __global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_mid_emit(
     }
 }
 
+// single-launch variant: count, one-shot cross-WG barrier, emit.  The grid
+// is tiny (MERGE_MID_WGS x np <= 128 workgroups, all co-resident), so every
+// WG publishes its count tagged with the launch generation and spins for
+// its pair's 32 peers — a single barrier between uniform-work peers, NOT a
+// serial lookback chain.  Each thread keeps its merge-path bounds in
+// registers across the barrier, so the emit phase skips the splits the
+// two-launch version recomputes.  Generation tags make the persistent
+// scratch self-cleaning (no memset launch); a stale match would need a
+// pair row untouched for exactly 2^32 launches.
+template <typename W>
+__global__ __launch_bounds__(FUSE_THREADS, 4) void k_merge_mid_fused(
+        MergeArgs args, int64_t *scratch, uint32_t gen) {
+    const int p = blockIdx.y;
+    const int wg = blockIdx.x;
+    int64_t *cnt_row = scratch + (int64_t)p * (MERGE_MID_WGS + 1);
+    int64_t *out_len = args.d_len + p;
+    const int tid = threadIdx.x;
+    const int64_t na = args.na[p];
+    const int64_t nb = args.dnb[p] ? *args.dnb[p] : args.nb[p];
+    if (na < 0 || nb < 0) {  // whole pair bails uniformly: no spin to feed
+        if (tid == 0 && wg == 0) *out_len = -1;
+        return;
+    }
+    const uint64_t *ak = args.ak[p], *av = args.av[p];
+    const W *aw = (const W *)args.aw[p];
+    const uint64_t *bk = args.bk[p], *bv = args.bv[p];
+    const W *bw = (const W *)args.bw[p];
+    uint64_t *ok = args.ok[p], *ov = args.ov[p];
+    W *ow = (W *)args.ow[p];
+    const int64_t total = na + nb;
+    const int64_t per_wg = (total + MERGE_MID_WGS - 1) / MERGE_MID_WGS;
+    const int64_t w0 = min((int64_t)wg * per_wg, total);
+    const int64_t w1 = min(w0 + per_wg, total);
+    const int64_t items = (w1 - w0 + FUSE_THREADS - 1) / FUSE_THREADS;
+    int64_t d0 = min(w0 + (int64_t)tid * items, w1);
+    int64_t d1 = min(d0 + items, w1);
+    int64_t ai, bi, ae, be;
+    merge_path(ak, av, na, bk, bv, nb, d0, ai, bi);
+    adjust_split(ak, av, bk, bv, na, nb, ai, bi);
+    merge_path(ak, av, na, bk, bv, nb, d1, ae, be);
+    adjust_split(ak, av, bk, bv, na, nb, ae, be);
+    uint32_t cnt = 0;
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(ak[i], av[i], bk[j], bv[j])) {
+                if (aw[i] + bw[j] != (W)0) cnt++;
+                i++; j++;
+            } else if (j >= be || (i < ae && row_lt(ak[i], av[i], bk[j], bv[j]))) {
+                cnt++; i++;
+            } else {
+                cnt++; j++;
+            }
+        }
+    }
+    __shared__ uint32_t wave_tot[FUSE_THREADS / WAVE + 1];
+    uint32_t wg_tot;
+    uint32_t off32 = fuse_scan(cnt, wave_tot, &wg_tot);
+    if (tid == 0)
+        atomicExch((unsigned long long *)&cnt_row[wg],
+                   ((unsigned long long)gen << 32) | wg_tot);
+    __shared__ int64_t s_cnt[MERGE_MID_WGS];
+    __shared__ int s_dead;
+    if (tid == 0) s_dead = 0;
+    __syncthreads();
+    if (tid < MERGE_MID_WGS) {
+        // bounded spin: the grid is co-resident by construction, so this
+        // resolves in ~kernel-uniform time; the bound turns a scheduling
+        // assumption failure into a poisoned length (host fails loudly /
+        // the train replays) instead of a wedged GPU
+        unsigned long long v = 0;
+        for (long spins = 0;; spins++) {
+            v = atomicAdd((unsigned long long *)&cnt_row[tid], 0ull);
+            if ((uint32_t)(v >> 32) == (uint32_t)gen) break;
+            if (spins > (1l << 26)) { s_dead = 1; break; }
+            __builtin_amdgcn_s_sleep(1);
+        }
+        s_cnt[tid] = (int64_t)(uint32_t)v;
+    }
+    __syncthreads();
+    if (s_dead) {
+        if (tid == 0) *out_len = -1;
+        return;
+    }
+    int64_t base = 0, all = 0;
+    for (int g = 0; g < MERGE_MID_WGS; g++) {
+        if (g < wg) base += s_cnt[g];
+        all += s_cnt[g];
+    }
+    if (wg == 0 && tid == 0) *out_len = all;
+    int64_t off = base + (int64_t)off32;
+    {
+        int64_t i = ai, j = bi;
+        while (i < ae || j < be) {
+            if (i < ae && j < be && row_eq(ak[i], av[i], bk[j], bv[j])) {
+                W s = aw[i] + bw[j];
+                if (s != (W)0) { ok[off] = ak[i]; ov[off] = av[i]; ow[off] = s; off++; }
+                i++; j++;
+            } else if (j >= be || (i < ae && row_lt(ak[i], av[i], bk[j], bv[j]))) {
+                ok[off] = ak[i]; ov[off] = av[i]; ow[off] = aw[i]; off++; i++;
+            } else {
+                ok[off] = bk[j]; ov[off] = bv[j]; ow[off] = bw[j]; off++; j++;
+            }
+        }
+    }
+}
+
 // ---------------------------------------------------------------------------
 // multi-batch spine join: the reference reads a spine through a k-way
 // CursorList (trace/cursor/cursor_list.rs); join is linear in the trace, so
@@ -3172,21 +3279,28 @@ dbsp_status merge_small_batch_f64(hipStream_t s, const MergeArgs &args) {
     return DBSP_OK;
 }
 
+// launch generation for the fused barrier's tagged counts — shared across
+// the int64/f64 instantiations so two launchers can never reuse a tag the
+// other just published into the same scratch
+static std::atomic<uint32_t> g_mid_gen{0};
+
 dbsp_status merge_mid_batch(hipStream_t s, const MergeArgs &args,
                             int64_t *scratch) {
     if (args.np == 0) return DBSP_OK;
+    const uint32_t gen = ++g_mid_gen;
     dim3 grid(MERGE_MID_WGS, (uint32_t)args.np);
-    k_merge_mid_count<int64_t><<<grid, FUSE_THREADS, 0, s>>>(args, scratch);
-    k_merge_mid_emit<int64_t><<<grid, FUSE_THREADS, 0, s>>>(args, scratch);
+    k_merge_mid_fused<int64_t><<<grid, FUSE_THREADS, 0, s>>>(args, scratch,
+                                                             gen);
     return DBSP_OK;
 }
 
 dbsp_status merge_mid_batch_f64(hipStream_t s, const MergeArgs &args,
                                 int64_t *scratch) {
     if (args.np == 0) return DBSP_OK;
+    const uint32_t gen = ++g_mid_gen;
     dim3 grid(MERGE_MID_WGS, (uint32_t)args.np);
-    k_merge_mid_count<double><<<grid, FUSE_THREADS, 0, s>>>(args, scratch);
-    k_merge_mid_emit<double><<<grid, FUSE_THREADS, 0, s>>>(args, scratch);
+    k_merge_mid_fused<double><<<grid, FUSE_THREADS, 0, s>>>(args, scratch,
+                                                            gen);
     return DBSP_OK;
 }
 
