@@ -136,6 +136,68 @@ def test_window_first_tick_selects_batch(triples, lo, span):
     assert np.array_equal(oracle.consolidate(got.copy()), np.asarray(exp))
 
 
+@given(rows_strategy, st.integers(0, 12))
+@settings(**SETTINGS)
+def test_rolling_agg_matches_model(triples, width):
+    """Radix-tree rolling aggregate semantics (reference
+    rolling_aggregate.rs:235-280, range_of = [ts - width, ts] saturating at
+    0): per row, the sum of weights of same-partition rows whose timestamp
+    falls in the window."""
+    # unique (partition, ts) pairs with weight from the last occurrence
+    acc = {}
+    for k, v, w in triples:
+        acc[(k, v)] = w
+    rows = _rows(sorted((k, v, w) for (k, v), w in acc.items()))
+    got = oracle.rolling_agg(rows, width)
+    for k, ts, s in got:
+        t0 = max(0, int(ts) - width)
+        exp = sum(int(w) for kk, vv, w in rows
+                  if kk == k and t0 <= vv <= ts)
+        assert int(s) == exp
+
+
+# f64-weight variants: weights are f64 BIT PATTERNS in the w column;
+# dyadic values (multiples of 0.25) make every reduction order bit-exact,
+# so the i64-style equalities must hold verbatim
+f64_rows_strategy = st.lists(
+    st.tuples(st.integers(0, 24), st.integers(0, 6), st.integers(-8, 8)),
+    min_size=0, max_size=100,
+)
+
+
+def _rows_f64(triples):
+    out = np.empty(len(triples), dtype=ROW_DT)
+    for i, (k, v, q) in enumerate(triples):
+        out[i] = (k, v, np.float64(q * 0.25).view(np.int64))
+    return out
+
+
+def _model_consolidate_f64(triples):
+    acc = {}
+    for k, v, q in triples:
+        acc[(k, v)] = acc.get((k, v), 0) + q
+    items = sorted((k, v, q) for (k, v), q in acc.items() if q != 0)
+    return _rows_f64(items)
+
+
+@given(f64_rows_strategy)
+@settings(**SETTINGS)
+def test_consolidate_f64_matches_model(triples):
+    got = oracle.consolidate_f64(_rows_f64(triples))
+    exp = _model_consolidate_f64(triples)
+    assert np.array_equal(got, exp)
+
+
+@given(f64_rows_strategy, f64_rows_strategy)
+@settings(**SETTINGS)
+def test_merge_f64_is_consolidated_concat(a, b):
+    ca = oracle.consolidate_f64(_rows_f64(a))
+    cb = oracle.consolidate_f64(_rows_f64(b))
+    got = oracle.merge_f64(ca, cb)
+    exp = _model_consolidate_f64(a + b)
+    assert np.array_equal(got, exp)
+
+
 @given(rows_strategy, rows_strategy, st.integers(0, 20), st.integers(0, 24),
        st.integers(0, 10), st.integers(0, 10))
 @settings(**SETTINGS)
